@@ -1,0 +1,148 @@
+"""In-process fake Kubernetes API server (fastapi).
+
+Benchmark config #1 ("in-process fake k8s API on CPU") and the test
+harness both use this: it serves the list/patch/delete surface of
+kube/client.py over localhost, backed by in-memory pod/node stores, and
+supports fault injection (delayed / failed responses) for the failure-
+detection tests.
+
+The reference's tests mocked pykube instead [R: test/test_cluster.py —
+reconstructed]; a live fake server exercises the real HTTP client path.
+"""
+
+from __future__ import annotations
+
+import copy
+import threading
+import time
+from typing import Dict, List, Optional
+
+from fastapi import FastAPI, Request, Response
+from fastapi.responses import JSONResponse
+
+
+class FakeKubeState:
+    """Thread-safe in-memory object store."""
+
+    def __init__(self):
+        self.lock = threading.RLock()
+        self.pods: Dict[str, Dict] = {}   # "<ns>/<name>" -> obj
+        self.nodes: Dict[str, Dict] = {}
+        self.pdbs: Dict[str, Dict] = {}
+        self.fault_delay_s: float = 0.0
+        self.fail_next: int = 0           # fail the next N requests with 500
+        self.deleted_pods: List[str] = []
+        self.deleted_nodes: List[str] = []
+        self.patched_nodes: List[str] = []
+
+    def load(self, pods: List[Dict], nodes: List[Dict], pdbs: Optional[List[Dict]] = None):
+        with self.lock:
+            self.pods = {
+                f"{p['metadata'].get('namespace', 'default')}/{p['metadata']['name']}": copy.deepcopy(p)
+                for p in pods
+            }
+            self.nodes = {n["metadata"]["name"]: copy.deepcopy(n) for n in nodes}
+            self.pdbs = {d["metadata"]["name"]: copy.deepcopy(d) for d in (pdbs or [])}
+
+
+def build_app(state: FakeKubeState) -> FastAPI:
+    app = FastAPI()
+
+    @app.middleware("http")
+    async def faults(request: Request, call_next):
+        if state.fault_delay_s > 0:
+            time.sleep(state.fault_delay_s)
+        with state.lock:
+            if state.fail_next > 0:
+                state.fail_next -= 1
+                return JSONResponse({"message": "injected fault"}, status_code=500)
+        return await call_next(request)
+
+    @app.get("/api/v1/pods")
+    def list_pods():
+        with state.lock:
+            return {"kind": "PodList", "items": list(state.pods.values()), "metadata": {}}
+
+    @app.get("/api/v1/nodes")
+    def list_nodes():
+        with state.lock:
+            return {"kind": "NodeList", "items": list(state.nodes.values()), "metadata": {}}
+
+    @app.get("/apis/policy/v1/poddisruptionbudgets")
+    def list_pdbs():
+        with state.lock:
+            return {"kind": "PodDisruptionBudgetList", "items": list(state.pdbs.values()), "metadata": {}}
+
+    @app.patch("/api/v1/nodes/{name}")
+    async def patch_node(name: str, request: Request):
+        body = await request.json()
+        with state.lock:
+            node = state.nodes.get(name)
+            if node is None:
+                return JSONResponse({"message": "not found"}, status_code=404)
+            unsched = (body.get("spec") or {}).get("unschedulable")
+            if unsched is not None:
+                node.setdefault("spec", {})["unschedulable"] = bool(unsched)
+            state.patched_nodes.append(name)
+            return node
+
+    @app.delete("/api/v1/namespaces/{ns}/pods/{name}")
+    def delete_pod(ns: str, name: str):
+        with state.lock:
+            key = f"{ns}/{name}"
+            state.pods.pop(key, None)
+            state.deleted_pods.append(key)
+            return {"status": "Success"}
+
+    @app.delete("/api/v1/nodes/{name}")
+    def delete_node(name: str):
+        with state.lock:
+            state.nodes.pop(name, None)
+            state.deleted_nodes.append(name)
+            return {"status": "Success"}
+
+    return app
+
+
+class FakeKubeServer:
+    """Run the fake API in a background uvicorn thread on a free port."""
+
+    def __init__(self, state: Optional[FakeKubeState] = None, host: str = "127.0.0.1"):
+        import socket
+
+        import uvicorn
+
+        self.state = state or FakeKubeState()
+        self.app = build_app(self.state)
+        sock = socket.socket()
+        sock.bind((host, 0))
+        self.host, self.port = sock.getsockname()
+        sock.close()
+        config = uvicorn.Config(
+            self.app, host=host, port=self.port, log_level="error", lifespan="off"
+        )
+        self.server = uvicorn.Server(config)
+        self.thread = threading.Thread(target=self.server.run, daemon=True)
+
+    @property
+    def base_url(self) -> str:
+        return f"http://{self.host}:{self.port}"
+
+    def start(self, timeout: float = 10.0) -> "FakeKubeServer":
+        self.thread.start()
+        deadline = time.time() + timeout
+        while not self.server.started:
+            if time.time() > deadline:
+                raise RuntimeError("fake kube api failed to start")
+            time.sleep(0.01)
+        return self
+
+    def stop(self):
+        self.server.should_exit = True
+        self.thread.join(timeout=5)
+
+    def __enter__(self):
+        return self.start()
+
+    def __exit__(self, *exc):
+        self.stop()

@@ -1,0 +1,330 @@
+"""Cluster snapshot -> SoA encoding on a fixed integer lattice.
+
+The reference wraps every pod/node JSON in Python objects and does all
+scheduling arithmetic in interpreted dict algebra
+[R: autoscaler/kube.py, autoscaler/cluster.py — reconstructed]. This
+framework instead encodes the whole cluster into structure-of-arrays
+int32/int64 buffers once per cycle; the CPU oracle
+(:mod:`mi355x_autoscaler.sim.oracle`) and the HIP kernels
+(:mod:`mi355x_autoscaler.ops`) both consume THIS lattice, which is what
+makes CPU and GPU decisions bit-identical:
+
+  * cpu      -> millicores, int32 (requests already integral)
+  * memory   -> MiB, int32; requests round UP, allocatable rounds DOWN
+                (conservative: the simulator never over-packs vs exact
+                byte arithmetic)
+  * pods,gpu -> counts, int32
+
+Label/selector matching is hashed: every ``key=value`` pair becomes a
+64-bit FNV-1a hash; a pod's nodeSelector is a sorted, padded array of
+pair hashes and a combined set hash; node labels likewise. Subset match
+(every selector pair present in the node's label set) is then pure
+integer work — the form kernel A evaluates on-device.
+
+A C++ fast-path encoder (ops/csrc/snapshot_codec.cpp) implements the
+same lattice for high-throughput JSON ingest; this module is the
+canonical definition plus the numpy reference implementation.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+
+from ..models.agent_pool import AgentPool
+from ..models.kube import KubeNode, KubePod
+from ..models.resources import NUM_RESOURCES, ResourceVec
+
+MIB = 1024 * 1024
+
+# Padded-width limits for hashed label/selector sets. Pods with more
+# selector pairs than MAX_SELECTOR_PAIRS or nodes with more labels than
+# MAX_LABEL_PAIRS fall back to exact host-side matching (rare; counted
+# in the encode stats).
+MAX_SELECTOR_PAIRS = 8
+MAX_LABEL_PAIRS = 32
+
+_FNV_OFFSET = 0xCBF29CE484222325
+_FNV_PRIME = 0x100000001B3
+_MASK64 = (1 << 64) - 1
+
+
+def fnv1a64(data: bytes) -> int:
+    """64-bit FNV-1a. Mirrored bit-for-bit by the device hasher in
+    ops/hip/common.h so host- and device-computed hashes agree."""
+    h = _FNV_OFFSET
+    for b in data:
+        h ^= b
+        h = (h * _FNV_PRIME) & _MASK64
+    return h
+
+
+def hash_label_pair(key: str, value: str) -> int:
+    return fnv1a64(f"{key}\x1f{value}".encode("utf-8"))
+
+
+def _as_i64(h: int) -> int:
+    """Reinterpret uint64 hash as int64 for numpy storage."""
+    return h - (1 << 64) if h >= (1 << 63) else h
+
+
+class PodFlags:
+    """Bitfield for the per-pod flags column (uint16)."""
+
+    PENDING = 1 << 0        # unschedulable-pending -> scale-up input
+    MIRROR = 1 << 1
+    DAEMONSET = 1 << 2
+    CRITICAL = 1 << 3
+    REPLICATED = 1 << 4
+    DRAINABLE = 1 << 5      # may be evicted on scale-down
+    BLOCKS_DRAIN = 1 << 6   # pins its node against scale-down
+    SELECTOR_OVERFLOW = 1 << 7  # >MAX_SELECTOR_PAIRS pairs: host fallback
+    BOUND = 1 << 8          # has spec.nodeName resolving to a known node
+
+
+def quantize_request(v: ResourceVec) -> Tuple[int, int, int, int]:
+    """Request vector -> lattice (memory rounds UP)."""
+    return (v.cpu, -(-v.memory // MIB), v.pods, v.gpu)
+
+
+def quantize_allocatable(v: ResourceVec) -> Tuple[int, int, int, int]:
+    """Allocatable vector -> lattice (memory rounds DOWN)."""
+    return (v.cpu, v.memory // MIB, v.pods, v.gpu)
+
+
+@dataclass
+class ClusterSnapshot:
+    """SoA snapshot of pods, nodes and pools on the integer lattice.
+
+    Array shapes (P pods, N nodes, K pools):
+      pod_req        [P, 4] int32     quantized requests
+      pod_flags      [P]    uint16    PodFlags bits
+      pod_sel_hash   [P]    int64     combined selector-set hash (0 = empty)
+      pod_sel_pairs  [P, MAX_SELECTOR_PAIRS] int64  sorted, 0-padded
+      pod_sel_count  [P]    int8
+      pod_node       [P]    int32     bound node index or -1
+      pod_pdb        [P]    int32     PDB index or -1
+      node_alloc     [N, 4] int32     quantized allocatable
+      node_free      [N, 4] int32     alloc minus bound pods (filled by
+                                      compute_node_free / kernel)
+      node_labels    [N, MAX_LABEL_PAIRS] int64 sorted, 0-padded
+      node_label_count [N]  int8
+      node_unsched   [N]    uint8
+      node_pool      [N]    int32     pool index or -1
+      pool_unit      [K, 4] int32     capacity of ONE new node
+      pool_label_*               prototype labels of a NEW pool node
+      pdb_allowed    [D]    int32     disruptionsAllowed per PDB
+    """
+
+    pod_req: np.ndarray
+    pod_flags: np.ndarray
+    pod_sel_hash: np.ndarray
+    pod_sel_pairs: np.ndarray
+    pod_sel_count: np.ndarray
+    pod_node: np.ndarray
+    pod_pdb: np.ndarray
+
+    node_alloc: np.ndarray
+    node_free: np.ndarray
+    node_labels: np.ndarray
+    node_label_count: np.ndarray
+    node_unsched: np.ndarray
+    node_pool: np.ndarray
+
+    pool_unit: np.ndarray
+    pool_max: np.ndarray
+    pool_cur: np.ndarray
+    pool_labels: np.ndarray
+    pool_label_count: np.ndarray
+    pool_names: List[str] = field(default_factory=list)
+
+    pdb_allowed: np.ndarray = field(default_factory=lambda: np.zeros(0, np.int32))
+
+    # bookkeeping for decoding decisions back to objects
+    pod_uids: List[str] = field(default_factory=list)
+    pod_refs: List[Tuple[str, str]] = field(default_factory=list)  # (namespace, name)
+    node_names: List[str] = field(default_factory=list)
+
+    @property
+    def num_pods(self) -> int:
+        return int(self.pod_req.shape[0])
+
+    @property
+    def num_nodes(self) -> int:
+        return int(self.node_alloc.shape[0])
+
+    @property
+    def num_pools(self) -> int:
+        return int(self.pool_unit.shape[0])
+
+    def nbytes(self) -> int:
+        total = 0
+        for f in self.__dataclass_fields__:
+            v = getattr(self, f)
+            if isinstance(v, np.ndarray):
+                total += v.nbytes
+        return total
+
+
+def _hash_pairs(mapping: Dict[str, str], width: int) -> Tuple[np.ndarray, int, int]:
+    """dict -> (sorted padded int64 array, count, combined set hash)."""
+    hashes = sorted(hash_label_pair(k, v) for k, v in mapping.items())
+    combined = 0
+    for h in hashes:
+        combined ^= h  # order-independent set hash
+    out = np.zeros(width, np.int64)
+    cnt = min(len(hashes), width)
+    for i in range(cnt):
+        out[i] = _as_i64(hashes[i])
+    return out, len(hashes), _as_i64(combined)
+
+
+def encode_snapshot(
+    pods: Sequence[KubePod],
+    nodes: Sequence[KubeNode],
+    pools: Dict[str, AgentPool],
+    pdbs: Optional[Sequence] = None,
+) -> ClusterSnapshot:
+    """Encode wrapped k8s objects into the SoA lattice (numpy reference
+    implementation; the C++ codec is the high-throughput path)."""
+    P, N = len(pods), len(nodes)
+    pool_names = sorted(pools.keys())
+    pool_index = {name: i for i, name in enumerate(pool_names)}
+    K = len(pool_names)
+
+    pod_req = np.zeros((P, NUM_RESOURCES), np.int32)
+    pod_flags = np.zeros(P, np.uint16)
+    pod_sel_hash = np.zeros(P, np.int64)
+    pod_sel_pairs = np.zeros((P, MAX_SELECTOR_PAIRS), np.int64)
+    pod_sel_count = np.zeros(P, np.int8)
+    pod_node = np.full(P, -1, np.int32)
+    pod_pdb = np.full(P, -1, np.int32)
+
+    node_alloc = np.zeros((N, NUM_RESOURCES), np.int32)
+    node_labels = np.zeros((N, MAX_LABEL_PAIRS), np.int64)
+    node_label_count = np.zeros(N, np.int8)
+    node_unsched = np.zeros(N, np.uint8)
+    node_pool = np.full(N, -1, np.int32)
+
+    node_idx = {n.name: i for i, n in enumerate(nodes)}
+
+    for i, n in enumerate(nodes):
+        node_alloc[i] = quantize_allocatable(n.allocatable)
+        node_labels[i], cnt, _ = _hash_pairs(n.labels, MAX_LABEL_PAIRS)
+        node_label_count[i] = min(cnt, MAX_LABEL_PAIRS)
+        node_unsched[i] = 1 if n.unschedulable else 0
+        if n.pool_name in pool_index:
+            node_pool[i] = pool_index[n.pool_name]
+
+    # PDB matching: a PDB object carries {selector labels, allowed}.
+    pdbs = list(pdbs or [])
+    pdb_allowed = np.zeros(len(pdbs), np.int32)
+    pdb_selectors: List[Dict[str, str]] = []
+    for d, pdb in enumerate(pdbs):
+        pdb_allowed[d] = int(getattr(pdb, "disruptions_allowed", 0))
+        pdb_selectors.append(dict(getattr(pdb, "match_labels", {}) or {}))
+
+    for i, p in enumerate(pods):
+        pod_req[i] = quantize_request(p.resources)
+        pairs, cnt, combined = _hash_pairs(p.selectors, MAX_SELECTOR_PAIRS)
+        pod_sel_pairs[i] = pairs
+        pod_sel_count[i] = min(cnt, MAX_SELECTOR_PAIRS)
+        pod_sel_hash[i] = combined
+        f = 0
+        if p.is_pending_unschedulable:
+            f |= PodFlags.PENDING
+        if p.is_mirrored:
+            f |= PodFlags.MIRROR
+        if p.is_daemonset:
+            f |= PodFlags.DAEMONSET
+        if p.is_critical:
+            f |= PodFlags.CRITICAL
+        if p.is_replicated:
+            f |= PodFlags.REPLICATED
+        if p.is_drainable:
+            f |= PodFlags.DRAINABLE
+        if p.blocks_drain:
+            f |= PodFlags.BLOCKS_DRAIN
+        if cnt > MAX_SELECTOR_PAIRS:
+            f |= PodFlags.SELECTOR_OVERFLOW
+        if p.node_name and p.node_name in node_idx:
+            pod_node[i] = node_idx[p.node_name]
+            f |= PodFlags.BOUND
+        pod_flags[i] = f
+        for d, sel in enumerate(pdb_selectors):
+            if sel and all(p.labels.get(k) == v for k, v in sel.items()):
+                pod_pdb[i] = d
+                break
+
+    pool_unit = np.zeros((K, NUM_RESOURCES), np.int32)
+    pool_max = np.zeros(K, np.int32)
+    pool_cur = np.zeros(K, np.int32)
+    pool_labels = np.zeros((K, MAX_LABEL_PAIRS), np.int64)
+    pool_label_count = np.zeros(K, np.int8)
+    for name, k in pool_index.items():
+        pool = pools[name]
+        unit = pool.unit_capacity
+        if unit is not None:
+            pool_unit[k] = quantize_allocatable(unit)
+        pool_max[k] = pool.max_size
+        pool_cur[k] = pool.actual_capacity
+        pool_labels[k], cnt, _ = _hash_pairs(pool.prototype_labels(), MAX_LABEL_PAIRS)
+        pool_label_count[k] = min(cnt, MAX_LABEL_PAIRS)
+
+    snap = ClusterSnapshot(
+        pod_req=pod_req,
+        pod_flags=pod_flags,
+        pod_sel_hash=pod_sel_hash,
+        pod_sel_pairs=pod_sel_pairs,
+        pod_sel_count=pod_sel_count,
+        pod_node=pod_node,
+        pod_pdb=pod_pdb,
+        node_alloc=node_alloc,
+        node_free=node_alloc.copy(),
+        node_labels=node_labels,
+        node_label_count=node_label_count,
+        node_unsched=node_unsched,
+        node_pool=node_pool,
+        pool_unit=pool_unit,
+        pool_max=pool_max,
+        pool_cur=pool_cur,
+        pool_labels=pool_labels,
+        pool_label_count=pool_label_count,
+        pool_names=pool_names,
+        pdb_allowed=pdb_allowed,
+        pod_uids=[p.uid for p in pods],
+        pod_refs=[(p.namespace, p.name) for p in pods],
+        node_names=[n.name for n in nodes],
+    )
+    compute_node_free(snap)
+    return snap
+
+
+def compute_node_free(snap: ClusterSnapshot) -> None:
+    """node_free = allocatable - sum(requests of bound pods).
+
+    numpy reference of the device-side segmented subtraction (the GPU
+    path computes this with the CSR utilization kernel).
+    """
+    free = snap.node_alloc.astype(np.int64).copy()
+    bound = snap.pod_node >= 0
+    if bound.any():
+        np.subtract.at(free, snap.pod_node[bound], snap.pod_req[bound].astype(np.int64))
+    snap.node_free = np.clip(free, np.iinfo(np.int32).min, np.iinfo(np.int32).max).astype(np.int32)
+
+
+def selector_subset_match(
+    sel_pairs: np.ndarray, sel_count: int, label_pairs: np.ndarray, label_count: int
+) -> bool:
+    """True iff every selector pair hash appears in the label hash set.
+    Both arrays sorted ascending. Mirrors the device predicate."""
+    li = 0
+    for si in range(sel_count):
+        s = sel_pairs[si]
+        while li < label_count and label_pairs[li] < s:
+            li += 1
+        if li >= label_count or label_pairs[li] != s:
+            return False
+    return True

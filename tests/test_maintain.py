@@ -1,0 +1,96 @@
+import numpy as np
+
+from mi355x_autoscaler.kube.objects import PodDisruptionBudget
+from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools
+from mi355x_autoscaler.sim.oracle import OracleSimulator
+from mi355x_autoscaler.sim.synth import make_node_json, make_pod_json, scaledown_trace, wrap
+from mi355x_autoscaler.snapshot.encode import encode_snapshot
+
+
+def _snap(pods_json, nodes_json, pdbs=None):
+    pods, nodes = wrap(pods_json, nodes_json)
+    pools = group_nodes_into_pools(nodes)
+    return encode_snapshot(pods, nodes, pools, pdbs)
+
+
+def test_idle_node_is_candidate():
+    busy = make_node_json("agentpool1", 0)
+    idle = make_node_json("agentpool1", 1)
+    pod = make_pod_json("w", cpu_milli=1500, node_name=busy["metadata"]["name"])
+    snap = _snap([pod], [busy, idle])
+    rep = OracleSimulator(snap).maintain_report()
+    names = [c.node_name for c in rep.candidates]
+    assert idle["metadata"]["name"] in names
+    assert rep.node_idle[1] == 1
+    assert rep.node_util_permille[0] == 750  # 1500/2000
+
+
+def test_pinned_pod_blocks_drain():
+    node = make_node_json("agentpool1", 0)
+    bare = make_pod_json("bare", cpu_milli=100, node_name=node["metadata"]["name"], owner_kind=None)
+    snap = _snap([bare], [node])
+    rep = OracleSimulator(snap).maintain_report()
+    assert rep.node_blocked[0] == 1
+    assert rep.candidates == []
+
+
+def test_daemonset_and_mirror_pods_keep_node_idle():
+    node = make_node_json("agentpool1", 0)
+    ds = make_pod_json("ds", cpu_milli=100, node_name=node["metadata"]["name"], owner_kind="DaemonSet")
+    mirror = make_pod_json(
+        "m", cpu_milli=50, node_name=node["metadata"]["name"], owner_kind=None,
+        annotations={"kubernetes.io/config.mirror": "abc"},
+    )
+    snap = _snap([ds, mirror], [node])
+    rep = OracleSimulator(snap).maintain_report()
+    assert rep.node_idle[0] == 1
+    assert rep.node_blocked[0] == 0
+
+
+def test_pdb_zero_budget_blocks_drain():
+    node = make_node_json("agentpool1", 0)
+    pod = make_pod_json("db", cpu_milli=100, node_name=node["metadata"]["name"], labels={"app": "db"})
+    pdb0 = PodDisruptionBudget("g", match_labels={"app": "db"}, disruptions_allowed=0)
+    snap = _snap([pod], [node], [pdb0])
+    rep = OracleSimulator(snap).maintain_report()
+    assert rep.node_pdb_ok[0] == 0
+    assert rep.candidates == []
+
+    pdb1 = PodDisruptionBudget("g", match_labels={"app": "db"}, disruptions_allowed=1)
+    snap2 = _snap([pod], [node], [pdb1])
+    rep2 = OracleSimulator(snap2).maintain_report()
+    assert rep2.node_pdb_ok[0] == 1
+    assert len(rep2.candidates) == 1
+
+
+def test_under_utilized_vs_busy_threshold():
+    node = make_node_json("agentpool1", 0)
+    pod = make_pod_json("w", cpu_milli=500, node_name=node["metadata"]["name"])  # 250 permille
+    snap = _snap([pod], [node])
+    rep = OracleSimulator(snap).maintain_report(util_threshold_permille=300)
+    assert [c.node_name for c in rep.candidates] == [node["metadata"]["name"]]
+    rep2 = OracleSimulator(snap).maintain_report(util_threshold_permille=200)
+    assert rep2.candidates == []
+
+
+def test_candidates_sorted_lowest_util_first():
+    n0 = make_node_json("agentpool1", 0)
+    n1 = make_node_json("agentpool1", 1)
+    p = make_pod_json("w", cpu_milli=400, node_name=n1["metadata"]["name"])  # 200 permille
+    snap = _snap([p], [n0, n1])
+    rep = OracleSimulator(snap).maintain_report()
+    assert rep.candidates[0].node_name == n0["metadata"]["name"]  # idle (0) first
+
+
+def test_scaledown_trace_500_nodes():
+    pods_json, nodes_json, pdbs = scaledown_trace(n_nodes=100, seed=3)
+    snap = _snap(pods_json, nodes_json, pdbs)
+    rep = OracleSimulator(snap).maintain_report()
+    # every idle node (util level 0) must be a candidate
+    idle_count = int(rep.node_idle.sum())
+    assert idle_count >= 100 // 5  # util_levels has a 0 slot
+    assert len(rep.candidates) > 0
+    # candidates exclude every blocked / pdb-bad node
+    for c in rep.candidates:
+        assert rep.node_blocked[c.node_index] == 0
+        assert rep.node_pdb_ok[c.node_index] == 1

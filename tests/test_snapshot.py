@@ -1,0 +1,103 @@
+import numpy as np
+
+from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools
+from mi355x_autoscaler.snapshot.encode import (
+    MIB,
+    PodFlags,
+    encode_snapshot,
+    fnv1a64,
+    hash_label_pair,
+    selector_subset_match,
+)
+from mi355x_autoscaler.sim.synth import burst_trace, make_node_json, make_pod_json, wrap
+
+
+def _snap(pods_json, nodes_json, pdbs=None):
+    pods, nodes = wrap(pods_json, nodes_json)
+    pools = group_nodes_into_pools(nodes)
+    return encode_snapshot(pods, nodes, pools, pdbs)
+
+
+def test_fnv1a64_stability():
+    # pinned value: the HIP-side hasher must reproduce these exactly
+    assert fnv1a64(b"") == 0xCBF29CE484222325
+    assert fnv1a64(b"a") == 0xAF63DC4C8601EC8C
+    h1 = hash_label_pair("agentpool", "gpupool")
+    h2 = hash_label_pair("agentpool", "gpupool")
+    assert h1 == h2
+    assert hash_label_pair("a", "b") != hash_label_pair("a", "c")
+    # key/value boundary is unambiguous (a\x1fb vs ab)
+    assert hash_label_pair("ab", "") != hash_label_pair("a", "b")
+
+
+def test_quantization_rounding():
+    # request memory rounds UP to MiB, allocatable rounds DOWN
+    pod = make_pod_json("p", cpu_milli=100, mem_mib=0, pending=True)
+    pod["spec"]["containers"][0]["resources"]["requests"]["memory"] = "1000001"  # bytes
+    node = make_node_json("agentpool1", 0)
+    node["status"]["allocatable"]["memory"] = "1048577"  # 1 MiB + 1 byte
+    snap = _snap([pod], [node])
+    assert snap.pod_req[0, 1] == 1  # ceil(1000001 B / MiB)
+    assert snap.node_alloc[0, 1] == 1  # floor
+
+
+def test_flags_and_binding():
+    node = make_node_json("agentpool1", 0)
+    bound = make_pod_json("b", node_name=node["metadata"]["name"])
+    pend = make_pod_json("p", pending=True)
+    snap = _snap([bound, pend], [node])
+    assert snap.pod_flags[0] & PodFlags.BOUND
+    assert snap.pod_node[0] == 0
+    assert snap.pod_flags[1] & PodFlags.PENDING
+    assert snap.pod_node[1] == -1
+
+
+def test_node_free_subtracts_bound_pods():
+    node = make_node_json("agentpool1", 0, cpu_milli=2000, mem_mib=4096)
+    p1 = make_pod_json("a", cpu_milli=500, mem_mib=1024, node_name=node["metadata"]["name"])
+    p2 = make_pod_json("b", cpu_milli=300, mem_mib=512, node_name=node["metadata"]["name"])
+    snap = _snap([p1, p2], [node])
+    assert snap.node_free[0, 0] == 2000 - 800
+    assert snap.node_free[0, 1] == 4096 - 1536
+    assert snap.node_free[0, 2] == snap.node_alloc[0, 2] - 2
+
+
+def test_selector_subset_match_device_form():
+    node = make_node_json("gpupool", 0, gpu=8, instance_type="Standard_ND96_MI355X_v1")
+    pod = make_pod_json("g", gpu=1, node_selector={"accelerator": "amd-mi355x"}, pending=True)
+    bad = make_pod_json("x", node_selector={"accelerator": "tpu"}, pending=True)
+    snap = _snap([pod, bad], [node])
+    ok = selector_subset_match(
+        snap.pod_sel_pairs[0], int(snap.pod_sel_count[0]),
+        snap.node_labels[0], int(snap.node_label_count[0]),
+    )
+    assert ok
+    assert not selector_subset_match(
+        snap.pod_sel_pairs[1], int(snap.pod_sel_count[1]),
+        snap.node_labels[0], int(snap.node_label_count[0]),
+    )
+    # empty selector matches everything
+    empty = np.zeros(8, np.int64)
+    assert selector_subset_match(empty, 0, snap.node_labels[0], int(snap.node_label_count[0]))
+
+
+def test_pdb_assignment():
+    from mi355x_autoscaler.kube.objects import PodDisruptionBudget
+
+    node = make_node_json("agentpool1", 0)
+    p = make_pod_json("guarded", node_name=node["metadata"]["name"], labels={"app": "db"})
+    q = make_pod_json("free", node_name=node["metadata"]["name"], labels={"app": "web"})
+    pdb = PodDisruptionBudget(name="db-pdb", match_labels={"app": "db"}, disruptions_allowed=0)
+    snap = _snap([p, q], [node], [pdb])
+    assert snap.pod_pdb[0] == 0
+    assert snap.pod_pdb[1] == -1
+    assert snap.pdb_allowed[0] == 0
+
+
+def test_snapshot_scales_and_is_compact():
+    pods_json, nodes_json = burst_trace(2000, n_nodes_per_pool=(50,), seed=1)
+    snap = _snap(pods_json, nodes_json)
+    assert snap.num_pods == 2000
+    assert snap.num_nodes == 50
+    # SoA must stay lean: << 1 KiB per pod (288 GB HBM fits ~1e9 pods)
+    assert snap.nbytes() / snap.num_pods < 1024
