@@ -1,0 +1,49 @@
+// Python bindings for the MI355X autoscaler HIP kernels.
+// Kernels live in ops/hip/*.hip (CDNA4-native; no CUDA-compat paths).
+#include <torch/extension.h>
+
+void classify_keys(torch::Tensor pod_flags, torch::Tensor sel_hash,
+                   torch::Tensor pod_req, torch::Tensor key_sel,
+                   torch::Tensor key_req, torch::Tensor pending_flag);
+void node_free_subtract(torch::Tensor pod_node, torch::Tensor pod_req,
+                        torch::Tensor node_free);
+void fit_eligibility(torch::Tensor node_labels, torch::Tensor node_label_count,
+                     torch::Tensor node_unsched, torch::Tensor seg_sel,
+                     torch::Tensor seg_sel_count, torch::Tensor elig);
+void seg_pool_route(torch::Tensor seg_sel, torch::Tensor seg_sel_count,
+                    torch::Tensor seg_req, torch::Tensor pool_labels,
+                    torch::Tensor pool_label_count, torch::Tensor pool_unit,
+                    torch::Tensor seg_pool);
+void binpack_place(torch::Tensor free, torch::Tensor elig, torch::Tensor seg_req,
+                   torch::Tensor seg_count, torch::Tensor placed_out);
+void binpack_units(torch::Tensor seg_pool, torch::Tensor seg_req,
+                   torch::Tensor seg_count, torch::Tensor placed,
+                   torch::Tensor pool_unit, torch::Tensor pool_max_new,
+                   torch::Tensor units_free, torch::Tensor units_cnt,
+                   torch::Tensor placed_new, torch::Tensor unful,
+                   torch::Tensor pool_residual);
+void drain_flags(torch::Tensor pod_node, torch::Tensor pod_flags,
+                 torch::Tensor pod_pdb, torch::Tensor pdb_allowed,
+                 torch::Tensor blocked, torch::Tensor pdb_bad,
+                 torch::Tensor nontrivial, torch::Tensor drainable);
+void drain_finalize(torch::Tensor util, torch::Tensor node_alloc,
+                    torch::Tensor node_pool, torch::Tensor blocked,
+                    torch::Tensor pdb_bad, torch::Tensor nontrivial,
+                    int64_t util_threshold_permille, torch::Tensor util_permille,
+                    torch::Tensor candidate);
+void segsum_mfma(torch::Tensor values, torch::Tensor seg_of,
+                 torch::Tensor wi_win, torch::Tensor wi_start,
+                 torch::Tensor wi_end, int64_t num_segments, torch::Tensor out);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "MI355X-native autoscaler scheduling kernels (gfx950 HIP)";
+  m.def("classify_keys", &classify_keys, "kernel C: pod radix keys");
+  m.def("node_free_subtract", &node_free_subtract, "node free capacity");
+  m.def("fit_eligibility", &fit_eligibility, "kernel A: segment x node eligibility");
+  m.def("seg_pool_route", &seg_pool_route, "segment -> pool routing");
+  m.def("binpack_place", &binpack_place, "kernel B: FFD placement steps");
+  m.def("binpack_units", &binpack_units, "kernel B: units-needed phase");
+  m.def("drain_flags", &drain_flags, "kernel D: drain predicates");
+  m.def("drain_finalize", &drain_finalize, "kernel D: utilization + candidates");
+  m.def("segsum_mfma", &segsum_mfma, "MFMA segmented resource sums");
+}

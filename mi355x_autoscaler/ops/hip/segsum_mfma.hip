@@ -1,0 +1,112 @@
+// MFMA-backed segmented sums over resource vectors (gfx950).
+//
+// The north-star op [SURVEY §2.2 row B]: per-segment totals of int32
+// resource vectors (segments = nodes for utilization, = pools for
+// demand), computed on the matrix cores as
+//
+//   D[row][s] = Σ_p A[row][p] · B[p][s]
+//
+// with `v_mfma_f32_16x16x4_f32` (exact f32 MFMA at the vector rate —
+// guide §3: bit-exact fmaf chain, no TF32 on gfx950). Rows encode
+// (limb, resource): each int32 value is split into 16-bit limbs
+// (lo = v & 0xffff, hi = v >> 16), 8 of the 16 A-rows carry data. B is
+// the segment-indicator built in-register from the sorted segment ids.
+// Exactness: a 256-element accumulation chain keeps every partial
+// < 2^24 (lo ≤ 256·(2^16−1) < 2^24, hi ≤ 256·(2^15−1) < 2^23), so the
+// f32 MFMA result is the exact integer sum; limbs recombine into int64
+// with a device-scope atomicAdd per (segment, resource) per chunk.
+// Integer atomics commute exactly => deterministic for any chunk/shard
+// decomposition (the RCCL agreement requirement).
+//
+// Work decomposition: elements must arrive sorted by segment id.
+// Segments are tiled 16 per MFMA *window*; a work item = (window,
+// 256-element chunk of that window's element range). One wave per work
+// item: 64 MFMA instructions, then one 8-lane flush.
+#include "common.h"
+
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+extern "C" __global__ __launch_bounds__(WAVE) void segsum_mfma_kernel(
+    const int32_t* __restrict__ values,   // [M, NRES] sorted by seg
+    const int32_t* __restrict__ seg_of,   // [M] ascending
+    const int32_t* __restrict__ wi_win,   // [T] window per work item
+    const int64_t* __restrict__ wi_start, // [T] element start
+    const int64_t* __restrict__ wi_end,   // [T] element end (window bound)
+    int32_t S,                            // number of segments
+    int64_t T,
+    uint64_t* __restrict__ out)           // [S, NRES] (non-negative i64)
+{
+  const int64_t t = blockIdx.x;
+  if (t >= T) return;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int col = lane & 15;      // A row index i / B,D column j
+  const int kidx = lane >> 4;     // K slot 0..3 (one pod each)
+
+  const int win = wi_win[t];
+  const int64_t p_begin = wi_start[t];
+  const int64_t p_end = wi_end[t];
+  const int seg_base = win * 16;
+
+  // A-row decode for this lane: rows 0..7 = (limb = row>>2, res = row&3)
+  const int row = col;
+  const int res = row & 3;
+  const int limb = (row >> 2) & 1;
+  const bool live_row = row < 8;
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int64_t p0 = p_begin; p0 < p_end; p0 += 4) {
+    const int64_t p = p0 + kidx;
+    float a = 0.f, b = 0.f;
+    if (p < p_end) {
+      const int sloc = seg_of[p] - seg_base;
+      // B[k][j]: indicator (pod p belongs to segment column j)
+      b = (sloc == col) ? 1.f : 0.f;
+      // A[i][k]: limb value of pod p, row i
+      if (live_row) {
+        const uint32_t v = (uint32_t)values[p * NRES + res];
+        a = (float)(limb ? (v >> 16) : (v & 0xffffu));
+      }
+    }
+    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+  }
+
+  // C/D layout for 16x16x4: col = lane&15, row = (lane>>4)*4 + reg.
+  const int s = seg_base + col;
+  if (s < S) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int orow = (lane >> 4) * 4 + reg;
+      if (orow < 8) {
+        const int ores = orow & 3;
+        const int olimb = (orow >> 2) & 1;
+        const unsigned long long partial = (unsigned long long)(int64_t)acc[reg];
+        if (partial) {
+          atomicAdd(reinterpret_cast<unsigned long long*>(
+                        &out[(int64_t)s * NRES + ores]),
+                    olimb ? (partial << 16) : partial);
+        }
+      }
+    }
+  }
+}
+
+#include <torch/extension.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+
+// values/seg_of sorted by segment; work items precomputed on device by
+// ops/engine_gpu.py (window = seg // 16, chunks of <=256 elements that
+// never cross a window boundary). out must be zeroed.
+void segsum_mfma(torch::Tensor values, torch::Tensor seg_of,
+                 torch::Tensor wi_win, torch::Tensor wi_start,
+                 torch::Tensor wi_end, int64_t num_segments,
+                 torch::Tensor out) {
+  const int64_t T = wi_win.size(0);
+  if (T == 0) return;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(segsum_mfma_kernel, dim3((uint32_t)T), dim3(WAVE), 0,
+                     stream.stream(), values.data_ptr<int32_t>(),
+                     seg_of.data_ptr<int32_t>(), wi_win.data_ptr<int32_t>(),
+                     wi_start.data_ptr<int64_t>(), wi_end.data_ptr<int64_t>(),
+                     (int32_t)num_segments, T,
+                     reinterpret_cast<uint64_t*>(out.data_ptr<int64_t>()));
+}

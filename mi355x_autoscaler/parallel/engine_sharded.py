@@ -1,0 +1,249 @@
+"""Sharded controller engine: pods sharded across <=8 GPUs, one global
+scale decision agreed over RCCL.
+
+Protocol per decide():
+
+  1. each rank classifies ITS pod shard and builds local placement
+     segments (kernel C + device radix sort — the P-proportional work,
+     which is what scales with ranks);
+  2. `allgather_rows` of the per-rank segment rows (sel pairs + request
+     + count: 15 x i64 per segment, KB-sized, latency-bound);
+  3. every rank merges counts by key into the IDENTICAL global segment
+     list (same order as sim/oracle.py::merge_segment_counts);
+  4. every rank runs the node-table placement (kernels A/B) on the
+     replicated node table with the global counts -> decisions are
+     bit-identical by construction (node tables are replicated;
+     placement math is deterministic integer arithmetic);
+  5. a final all-reduce(MIN)/all-reduce(MAX) over the per-pool units
+     vector asserts the agreement (a cheap cross-rank determinism
+     check, the test SURVEY §4.5 demands).
+
+maintain(): per-rank partial per-node accumulators are all-reduced
+(SUM for util/counts, MAX for the blocked/pdb flags) before the
+finalize kernel — integer ops, exact under any shard split.
+
+Works identically over gloo (CPU oracle engine, CI-tested here with
+world_size 2) and nccl/RCCL (GPU engine on MI355X).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+
+from ..sim.decision import MaintainReport, ScaleDecision, Segment
+from ..sim.engine import CpuEngine, SimulationEngine
+from ..sim.oracle import OracleSimulator, _pack_req_key, build_segments
+from ..snapshot.encode import MAX_SELECTOR_PAIRS, ClusterSnapshot
+from . import dist as pdist
+
+_U64 = (1 << 64) - 1
+_ROW_COLS = MAX_SELECTOR_PAIRS + 7  # pairs(8) + count + req(4) + seg_count + sel_hash
+
+
+def _segments_to_rows(segs: List[Segment]) -> np.ndarray:
+    rows = np.zeros((len(segs), _ROW_COLS), np.int64)
+    for i, s in enumerate(segs):
+        rows[i, :MAX_SELECTOR_PAIRS] = s.sel_pairs
+        rows[i, MAX_SELECTOR_PAIRS] = s.sel_count
+        rows[i, MAX_SELECTOR_PAIRS + 1 : MAX_SELECTOR_PAIRS + 5] = s.req
+        rows[i, MAX_SELECTOR_PAIRS + 5] = s.count
+        rows[i, MAX_SELECTOR_PAIRS + 6] = s.sel_hash
+    return rows
+
+
+def _rows_to_segments(rows: np.ndarray) -> List[Segment]:
+    out = []
+    for r in rows:
+        out.append(
+            Segment(
+                sel_hash=int(r[MAX_SELECTOR_PAIRS + 6]),
+                sel_pairs=r[:MAX_SELECTOR_PAIRS].copy(),
+                sel_count=int(r[MAX_SELECTOR_PAIRS]),
+                req=r[MAX_SELECTOR_PAIRS + 1 : MAX_SELECTOR_PAIRS + 5].astype(np.int32),
+                count=int(r[MAX_SELECTOR_PAIRS + 5]),
+            )
+        )
+    return out
+
+
+def merge_rows(rows: np.ndarray) -> np.ndarray:
+    """Sum counts of identical (sel_hash, req-key) rows; order by
+    (sel_hash as uint64 asc, req desc) — vectorized equivalent of
+    sim/oracle.py::merge_segment_counts."""
+    if rows.shape[0] == 0:
+        return rows
+    sel_u = rows[:, MAX_SELECTOR_PAIRS + 6].astype(np.uint64)
+    keys = np.array(
+        [_pack_req_key(r[MAX_SELECTOR_PAIRS + 1 : MAX_SELECTOR_PAIRS + 5]) for r in rows],
+        dtype=np.uint64,
+    )
+    order = np.lexsort((np.iinfo(np.uint64).max - keys, sel_u))
+    rows = rows[order]
+    sel_u = sel_u[order]
+    keys = keys[order]
+    boundary = np.ones(rows.shape[0], bool)
+    boundary[1:] = (sel_u[1:] != sel_u[:-1]) | (keys[1:] != keys[:-1])
+    group_id = np.cumsum(boundary) - 1
+    n_groups = int(group_id[-1]) + 1
+    counts = np.zeros(n_groups, np.int64)
+    np.add.at(counts, group_id, rows[:, MAX_SELECTOR_PAIRS + 5])
+    out = rows[boundary].copy()
+    out[:, MAX_SELECTOR_PAIRS + 5] = counts
+    return out
+
+
+class ShardedEngine(SimulationEngine):
+    """Wraps a CpuEngine or GpuEngine with the RCCL decision agreement.
+
+    The snapshot passed to decide()/maintain() is the LOCAL shard:
+    pods local to this rank; nodes/pools REPLICATED and identical
+    across ranks (the caller shards pods, e.g. uid-hash mod world).
+    """
+
+    name = "sharded"
+
+    def __init__(self, inner: SimulationEngine, check_agreement: bool = True):
+        self.inner = inner
+        self.check_agreement = check_agreement
+        self.active = pdist.init_from_env()
+
+    # ------------------------------------------------------------------
+    def decide(self, snap: ClusterSnapshot, over_provision: int = 0) -> ScaleDecision:
+        device = getattr(self.inner, "device", None)
+        if isinstance(self.inner, CpuEngine) or device is None:
+            local = build_segments(snap)
+            rows = _segments_to_rows(local)
+            rows = pdist.allgather_rows(rows)
+            merged = merge_rows(rows)
+            segments = _rows_to_segments(merged)
+            node_free = None
+            if self.active:
+                # globalize free capacity: usage partials sum exactly
+                usage = torch.from_numpy(
+                    (snap.node_alloc.astype(np.int64) - snap.node_free.astype(np.int64))
+                )
+                pdist.allreduce_tensor(usage, "sum")
+                node_free = snap.node_alloc.astype(np.int64) - usage.numpy()
+            sim = OracleSimulator(snap, over_provision=over_provision)
+            decision = sim.fulfill_pending(segments, node_free=node_free)
+        else:
+            from ..ops.engine_gpu import DeviceSegments, build_segments_device, upload_snapshot
+
+            ds = upload_snapshot(snap, device)
+            segs = build_segments_device(ds, self.inner.ops)
+            rows = self._device_segments_to_rows(ds, segs)
+            rows = pdist.allgather_rows(rows, device=device)
+            merged = merge_rows(rows)
+            gsegs = self._rows_to_device_segments(merged, device)
+
+            def free_hook(free, dsnap):
+                if not self.active:
+                    return free
+                usage = (dsnap.node_alloc - free).to(torch.int64)
+                pdist.allreduce_tensor(usage, "sum")
+                return (dsnap.node_alloc.to(torch.int64) - usage).clamp(
+                    min=-(2**31), max=2**31 - 1
+                ).to(torch.int32)
+
+            decision = self.inner.decide_device(
+                ds, gsegs, snap.pool_names, over_provision=over_provision,
+                free_hook=free_hook,
+            )
+        if self.active and self.check_agreement:
+            self._assert_agreement(decision, snap)
+        return decision
+
+    @staticmethod
+    def _device_segments_to_rows(ds, segs) -> np.ndarray:
+        G = segs.G
+        rows = np.zeros((G, _ROW_COLS), np.int64)
+        if G:
+            rows[:, :MAX_SELECTOR_PAIRS] = segs.seg_sel_pairs.cpu().numpy()
+            rows[:, MAX_SELECTOR_PAIRS] = segs.seg_sel_count.cpu().numpy()
+            rows[:, MAX_SELECTOR_PAIRS + 1 : MAX_SELECTOR_PAIRS + 5] = segs.seg_req.cpu().numpy()
+            rows[:, MAX_SELECTOR_PAIRS + 5] = segs.seg_count.cpu().numpy()
+            rows[:, MAX_SELECTOR_PAIRS + 6] = segs.seg_sel_hash.cpu().numpy()
+        return rows
+
+    @staticmethod
+    def _rows_to_device_segments(rows: np.ndarray, device):
+        from ..ops.engine_gpu import DeviceSegments
+
+        G = rows.shape[0]
+        e64 = torch.zeros(0, dtype=torch.int64, device=device)
+        return DeviceSegments(
+            seg_sel_pairs=torch.from_numpy(
+                np.ascontiguousarray(rows[:, :MAX_SELECTOR_PAIRS])
+            ).to(device),
+            seg_sel_count=torch.from_numpy(
+                rows[:, MAX_SELECTOR_PAIRS].astype(np.int32)
+            ).to(device),
+            seg_sel_hash=torch.from_numpy(
+                np.ascontiguousarray(rows[:, MAX_SELECTOR_PAIRS + 6])
+            ).to(device),
+            seg_req=torch.from_numpy(
+                np.ascontiguousarray(rows[:, MAX_SELECTOR_PAIRS + 1 : MAX_SELECTOR_PAIRS + 5].astype(np.int32))
+            ).to(device),
+            seg_count=torch.from_numpy(
+                np.ascontiguousarray(rows[:, MAX_SELECTOR_PAIRS + 5])
+            ).to(device),
+            pod_sorted=e64,
+            pod_seg=e64,
+        )
+
+    def _assert_agreement(self, decision: ScaleDecision, snap: ClusterSnapshot):
+        """All-reduce the per-pool units vector with MIN and MAX; any
+        divergence across ranks is a hard error (determinism bug)."""
+        K = snap.num_pools
+        vec = torch.zeros(K + 3, dtype=torch.int64)
+        for k, name in enumerate(snap.pool_names):
+            vec[k] = decision.units_needed.get(name, 0)
+        vec[K] = decision.pods_placed_existing
+        vec[K + 1] = decision.pods_placed_new
+        vec[K + 2] = decision.pods_unfulfillable
+        device = getattr(self.inner, "device", None)
+        if device is not None and str(device).startswith("cuda"):
+            vec = vec.to(device)
+        lo = vec.clone()
+        hi = vec.clone()
+        pdist.allreduce_tensor(lo, "min")
+        pdist.allreduce_tensor(hi, "max")
+        if not torch.equal(lo, hi):
+            raise RuntimeError(
+                f"sharded decision divergence: min={lo.tolist()} max={hi.tolist()}"
+            )
+
+    # ------------------------------------------------------------------
+    def maintain(self, snap: ClusterSnapshot, util_threshold_permille: int = 300) -> MaintainReport:
+        device = getattr(self.inner, "device", None)
+        if isinstance(self.inner, CpuEngine) or device is None:
+            sim = OracleSimulator(snap)
+            partials = sim.maintain_partials()
+            if self.active:
+                # RAW partials reduce exactly: SUM for util/counts,
+                # MAX for flags; permille division happens AFTER.
+                t = {k: torch.from_numpy(v.astype(np.int64)) for k, v in partials.items()}
+                pdist.allreduce_tensor(t["util"], "sum")
+                pdist.allreduce_tensor(t["blocked"], "max")
+                pdist.allreduce_tensor(t["pdb_bad"], "max")
+                pdist.allreduce_tensor(t["nontrivial"], "sum")
+                pdist.allreduce_tensor(t["drainable"], "sum")
+                partials = {k: v.numpy() for k, v in t.items()}
+            return sim.finalize_maintain(partials, util_threshold_permille)
+
+        from ..ops.engine_gpu import upload_snapshot
+
+        ds = upload_snapshot(snap, device)
+        _, tensors = self.inner.maintain_device(ds, util_threshold_permille)
+        if self.active:
+            pdist.allreduce_tensor(tensors["util"], "sum")
+            pdist.allreduce_tensor(tensors["blocked"], "max")
+            pdist.allreduce_tensor(tensors["pdb_bad"], "max")
+            pdist.allreduce_tensor(tensors["nontrivial"], "sum")
+            pdist.allreduce_tensor(tensors["drainable"], "sum")
+        rep = self.inner.finalize_maintain(ds, tensors, util_threshold_permille)
+        rep.candidates = self.inner._candidates(rep, snap.node_names, snap, util_threshold_permille)
+        return rep

@@ -200,17 +200,26 @@ class OracleSimulator:
                 out.append(k)
         return out
 
-    def fulfill_pending(self, segments: Optional[List[Segment]] = None) -> ScaleDecision:
+    def fulfill_pending(
+        self,
+        segments: Optional[List[Segment]] = None,
+        node_free: Optional[np.ndarray] = None,
+    ) -> ScaleDecision:
         """The whole scale-up decision [R: cluster.py · fulfill_pending].
 
         Phase 1: place segments onto existing nodes (uniform steps).
         Phase 2: residual pods bin-pack into virtual new nodes of the
                  first candidate pool, FFD in global segment order.
+
+        ``node_free``: override for the free-capacity table (the
+        sharded engine passes the all-reduced global table when pods
+        are sharded across ranks).
         """
         snap = self.snap
         if segments is None:
             segments = build_segments(snap)
-        free = snap.node_free.astype(_I64).copy()
+        base_free = snap.node_free if node_free is None else node_free
+        free = base_free.astype(_I64).copy()
         decision = ScaleDecision()
         residuals: List[Tuple[Segment, int]] = []
 
@@ -286,14 +295,16 @@ class OracleSimulator:
     # ------------------------------------------------------------------
     # scale-down scoring
     # ------------------------------------------------------------------
-    def maintain_report(self, util_threshold_permille: int = 300) -> MaintainReport:
-        """Per-node drain scoring [R: cluster.py · maintain].
-        (Device form: kernel D drain_score.hip)"""
+    def maintain_partials(self) -> Dict[str, np.ndarray]:
+        """Per-node accumulators over THIS snapshot's pods (a shard's
+        partial when pods are sharded): raw integer sums/flags that
+        all-reduce exactly (SUM for util/counts, MAX for flags).
+        (Device form: drain_flags_kernel + segsum_mfma)"""
         snap = self.snap
         N = snap.num_nodes
         util = np.zeros((N, NUM_RESOURCES), dtype=_I64)
-        blocked = np.zeros(N, np.uint8)
-        pdb_bad = np.zeros(N, np.uint8)
+        blocked = np.zeros(N, np.int32)
+        pdb_bad = np.zeros(N, np.int32)
         nontrivial = np.zeros(N, np.int32)  # non-mirror/ds resident pods
         drainable_pods = np.zeros(N, np.int32)
 
@@ -312,6 +323,21 @@ class OracleSimulator:
             d = int(snap.pod_pdb[i])
             if d >= 0 and int(snap.pdb_allowed[d]) < 1:
                 pdb_bad[j] = 1
+        return dict(util=util, blocked=blocked, pdb_bad=pdb_bad,
+                    nontrivial=nontrivial, drainable=drainable_pods)
+
+    def finalize_maintain(
+        self, partials: Dict[str, np.ndarray], util_threshold_permille: int = 300
+    ) -> MaintainReport:
+        """Permille utilization + candidate list from (possibly
+        all-reduced) partials. (Device form: drain_finalize_kernel)"""
+        snap = self.snap
+        N = snap.num_nodes
+        util = partials["util"]
+        blocked = partials["blocked"]
+        pdb_bad = partials["pdb_bad"]
+        nontrivial = partials["nontrivial"]
+        drainable_pods = partials["drainable"]
 
         util_permille = np.zeros(N, np.int32)
         for j in range(N):
@@ -325,10 +351,10 @@ class OracleSimulator:
         idle = (nontrivial == 0).astype(np.uint8)
         report = MaintainReport(
             node_util_permille=util_permille,
-            node_blocked=blocked,
-            node_pdb_ok=(1 - pdb_bad).astype(np.uint8),
+            node_blocked=(blocked != 0).astype(np.uint8),
+            node_pdb_ok=(pdb_bad == 0).astype(np.uint8),
             node_idle=idle,
-            node_drainable_pods=drainable_pods,
+            node_drainable_pods=drainable_pods.astype(np.int32),
         )
         for j in np.argsort(util_permille, kind="stable"):
             j = int(j)
@@ -346,3 +372,7 @@ class OracleSimulator:
                 )
             )
         return report
+
+    def maintain_report(self, util_threshold_permille: int = 300) -> MaintainReport:
+        """Single-process scale-down scoring [R: cluster.py · maintain]."""
+        return self.finalize_maintain(self.maintain_partials(), util_threshold_permille)

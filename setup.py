@@ -1,0 +1,41 @@
+"""In-tree build of the gfx950 HIP extension.
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The built _hip_ops .so lands inside mi355x_autoscaler/ so it travels
+with the repo snapshot to GPU boxes (no JIT cache dependence).
+"""
+
+import os
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+
+ext = CUDAExtension(
+    name="mi355x_autoscaler._hip_ops",
+    sources=[
+        "mi355x_autoscaler/ops/csrc/bindings.cpp",
+        "mi355x_autoscaler/ops/hip/classify.hip",
+        "mi355x_autoscaler/ops/hip/node_free.hip",
+        "mi355x_autoscaler/ops/hip/fit_predicates.hip",
+        "mi355x_autoscaler/ops/hip/binpack.hip",
+        "mi355x_autoscaler/ops/hip/drain_score.hip",
+        "mi355x_autoscaler/ops/hip/segsum_mfma.hip",
+    ],
+    extra_compile_args={
+        "cxx": ["-O3", "-std=c++17"],
+        "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+    },
+)
+
+setup(
+    name="mi355x_autoscaler_ops",
+    version="0.1.0",
+    ext_modules=[ext],
+    cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
+)

@@ -1,0 +1,124 @@
+"""Multi-process (gloo, world_size=2) tests of the sharded decision
+agreement on CPU — the distributed path the driver's 8-GPU run uses,
+minus the device. Asserts the SURVEY §4.5 invariant: the sharded
+decision equals the single-process decision bit-for-bit."""
+
+import multiprocessing as mp
+import os
+import pickle
+
+import numpy as np
+import pytest
+
+from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools
+from mi355x_autoscaler.sim.engine import CpuEngine
+from mi355x_autoscaler.sim.oracle import OracleSimulator, build_segments
+from mi355x_autoscaler.sim.synth import burst_trace, scaledown_trace, wrap
+from mi355x_autoscaler.snapshot.encode import encode_snapshot
+
+WORLD = 2
+
+
+def _make_snap(rank=None, seed=11, n_pods=400):
+    pods_json, nodes_json = burst_trace(
+        n_pods,
+        n_nodes_per_pool=(5, 2),
+        pools=(("agentpool1", "Standard_D2_v2"), ("gpupool", "Standard_ND96_MI355X_v1")),
+        gpu_fraction=0.3,
+        seed=seed,
+        bound_pods_per_node=3,
+    )
+    if rank is not None:
+        # shard pods round-robin; keep every bound pod on its rank too
+        pods_json = pods_json[rank::WORLD]
+    pods, nodes = wrap(pods_json, nodes_json)
+    pools = group_nodes_into_pools(nodes)
+    return encode_snapshot(pods, nodes, pools)
+
+
+def _worker(rank, q, mode):
+    os.environ.update(
+        WORLD_SIZE=str(WORLD), RANK=str(rank), LOCAL_RANK=str(rank),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT="29617",
+    )
+    import torch.distributed as dist
+
+    from mi355x_autoscaler.parallel.engine_sharded import ShardedEngine
+
+    engine = ShardedEngine(CpuEngine())
+    snap = _make_snap(rank=rank)
+    try:
+        if mode == "decide":
+            d = engine.decide(snap)
+            q.put((rank, pickle.dumps(dict(
+                units=d.units_needed, pe=d.pods_placed_existing,
+                pn=d.pods_placed_new, un=d.pods_unfulfillable,
+            ))))
+        else:
+            rep = engine.maintain(snap)
+            q.put((rank, pickle.dumps(dict(
+                utilp=rep.node_util_permille.tolist(),
+                idle=rep.node_idle.tolist(),
+                cands=[c.node_name for c in rep.candidates],
+            ))))
+    finally:
+        dist.destroy_process_group()
+
+
+def _run_world(mode):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker, args=(r, q, mode)) for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, blob = q.get(timeout=120)
+        results[rank] = pickle.loads(blob)
+    for p in procs:
+        p.join(timeout=30)
+        assert p.exitcode == 0
+    return results
+
+
+def test_sharded_decide_matches_single_process():
+    results = _run_world("decide")
+    assert results[0] == results[1]  # rank agreement
+
+    # equality with the single-process global decision
+    snap = _make_snap(rank=None)
+    d = CpuEngine().decide(snap)
+    assert results[0]["units"] == d.units_needed
+    assert results[0]["pe"] == d.pods_placed_existing
+    assert results[0]["pn"] == d.pods_placed_new
+    assert results[0]["un"] == d.pods_unfulfillable
+
+
+def test_sharded_maintain_matches_single_process():
+    results = _run_world("maintain")
+    assert results[0] == results[1]
+    snap = _make_snap(rank=None)
+    rep = CpuEngine().maintain(snap)
+    assert results[0]["utilp"] == rep.node_util_permille.tolist()
+    assert results[0]["idle"] == rep.node_idle.tolist()
+    assert results[0]["cands"] == [c.node_name for c in rep.candidates]
+
+
+def test_merge_rows_vectorized_equals_reference():
+    from mi355x_autoscaler.parallel.engine_sharded import (
+        _rows_to_segments,
+        _segments_to_rows,
+        merge_rows,
+    )
+    from mi355x_autoscaler.sim.oracle import merge_segment_counts
+
+    snap = _make_snap(rank=None, seed=5, n_pods=300)
+    idx = np.arange(snap.num_pods)
+    shards = [build_segments(snap, idx[r::3]) for r in range(3)]
+    rows = np.concatenate([_segments_to_rows(s) for s in shards], axis=0)
+    merged = _rows_to_segments(merge_rows(rows))
+    ref = merge_segment_counts(shards)
+    assert len(merged) == len(ref)
+    for a, b in zip(merged, ref):
+        assert a.sel_hash == b.sel_hash and a.count == b.count
+        assert (a.req == b.req).all()
