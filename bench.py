@@ -1,0 +1,184 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: the 10k-pod scale-up burst decision cycle
+(BASELINE.json metric: scale-up decision latency + pods-scheduled/sec).
+
+One *step* = one full scheduling decision over a fresh host snapshot of
+the burst: H2D upload of the SoA, kernel C classify + device radix sort
+into segments, node-free atomics, kernel A eligibility (LDS-staged),
+MFMA per-pool demand sums, kernel B FFD placement + units phase, the
+RCCL segment-count agreement (world > 1) and the decision readback —
+i.e. everything between "pods are pending in the snapshot" and "every
+rank holds the identical new-pool-sizes vector". Encoding JSON -> SoA
+is the ingest phase (C++ codec), not part of the decision-latency
+metric, matching BASELINE.json's definition.
+
+Scaling is WEAK: each rank decides its own 10k-pod shard of a
+world-size x 10k global burst against a replicated node table
+(pods-per-GPU fixed). `value` is whole-job pods decided per second:
+world x pods_per_gpu / max-rank step time.
+
+Run:  python bench.py [--gpus N] [--steps K] [--warmup W]
+N>1 is launched by the driver via torch.distributed.run (one rank per
+GPU, RCCL). Falls back to the CPU oracle when no device is present
+(reports backend "cpu" — the reference-equivalent stand-in).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import time
+
+import numpy as np
+import torch
+
+from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools
+from mi355x_autoscaler.sim.engine import CpuEngine
+from mi355x_autoscaler.sim.synth import burst_trace, wrap
+from mi355x_autoscaler.snapshot.encode import encode_snapshot
+
+POOLS = (
+    ("agentpool1", "Standard_D2_v2"),
+    ("bigpool", "Standard_D4_v2"),
+    ("gpupool", "Standard_ND96_MI355X_v1"),
+)
+POOL_SIZES = (40, 30, 30)
+GPU_FRACTION = 0.35
+
+
+def build_snapshot(rank: int, pods_per_rank: int):
+    pods_json, nodes_json = burst_trace(
+        pods_per_rank,
+        n_nodes_per_pool=POOL_SIZES,
+        pools=POOLS,
+        gpu_fraction=GPU_FRACTION,
+        seed=1000 + rank,
+        bound_pods_per_node=2,
+    )
+    pods, nodes = wrap(pods_json, nodes_json)
+    pools = group_nodes_into_pools(nodes)
+    return encode_snapshot(pods, nodes, pools)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--pods-per-gpu", type=int, default=10000)
+    ap.add_argument("--backend", choices=["auto", "gpu", "cpu"], default="auto")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = max(world, args.gpus if world == 1 else world)
+
+    use_gpu = args.backend != "cpu" and torch.cuda.is_available()
+    if args.backend == "gpu" and not torch.cuda.is_available():
+        raise RuntimeError("--backend gpu requested but no ROCm device visible")
+
+    import torch.distributed as dist
+
+    if world > 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group(backend="nccl" if use_gpu else "gloo")
+        if use_gpu:
+            torch.cuda.set_device(local_rank)
+
+    if use_gpu:
+        from mi355x_autoscaler.ops.engine_gpu import GpuEngine
+
+        inner = GpuEngine(f"cuda:{local_rank}")
+    else:
+        inner = CpuEngine()
+
+    if world > 1:
+        from mi355x_autoscaler.parallel.engine_sharded import ShardedEngine
+
+        engine = ShardedEngine(inner, check_agreement=True)
+    else:
+        engine = inner
+
+    snap = build_snapshot(rank, args.pods_per_gpu)
+
+    def sync():
+        if use_gpu:
+            torch.cuda.synchronize()
+        if world > 1:
+            dist.barrier()
+            if use_gpu:
+                torch.cuda.synchronize()
+
+    # warmup (untimed)
+    decision = None
+    for _ in range(args.warmup):
+        decision = engine.decide(snap)
+
+    sync()
+    step_times = []
+    t_all0 = time.perf_counter()
+    for _ in range(args.steps):
+        t0 = time.perf_counter()
+        decision = engine.decide(snap)
+        if use_gpu:
+            torch.cuda.synchronize()
+        step_times.append(time.perf_counter() - t0)
+    sync()
+    t_all1 = time.perf_counter()
+
+    total_s_local = t_all1 - t_all0
+    # MAX over ranks of the timed window
+    if world > 1:
+        t = torch.tensor([total_s_local], dtype=torch.float64)
+        if use_gpu:
+            t = t.cuda()
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        total_s = float(t.item())
+    else:
+        total_s = total_s_local
+
+    ms_per_step = total_s / args.steps * 1000.0
+    pods_total = world * args.pods_per_gpu
+    pods_per_sec = pods_total / (total_s / args.steps)
+    p50_ms = statistics.median(step_times) * 1000.0
+
+    if rank == 0:
+        result = {
+            "metric": "pods_scheduled_per_sec",
+            "value": round(pods_per_sec, 1),
+            "unit": "pods/s",
+            "n_gpus": world if world > 1 else (1 if use_gpu else 0),
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "int32+exact-f32-mfma",
+            "data": "synthetic",
+            "config": {
+                "model": "10k-pod mixed CPU/GPU burst, 3 heterogeneous agent pools (BASELINE config #3)",
+                "pods_per_gpu": args.pods_per_gpu,
+                "nodes": sum(POOL_SIZES),
+                "pools": 3,
+                "gpu_pod_fraction": GPU_FRACTION,
+                "parallelism": f"pod-sharded controller x{world}" if world > 1 else "single",
+                "backend": "gpu" if use_gpu else "cpu-oracle",
+                "decision_p50_ms": round(p50_ms, 3),
+                "pods_placed": decision.pods_placed_existing + decision.pods_placed_new,
+                "pods_unfulfillable": decision.pods_unfulfillable,
+                "units_needed": decision.units_needed,
+            },
+        }
+        print(json.dumps(result))
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,185 @@
+"""GPU kernel tests: every HIP kernel and the whole GpuEngine pipeline
+against the CPU oracle, bit-for-bit (SURVEY §4 plan items 1, 4)."""
+
+import numpy as np
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+torch = pytest.importorskip("torch")
+
+from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools
+from mi355x_autoscaler.sim.engine import CpuEngine
+from mi355x_autoscaler.sim.oracle import OracleSimulator, build_segments
+from mi355x_autoscaler.sim.synth import burst_trace, make_node_json, make_pod_json, scaledown_trace, wrap
+from mi355x_autoscaler.snapshot.encode import encode_snapshot
+
+
+@pytest.fixture(scope="module")
+def engine():
+    from mi355x_autoscaler.ops.engine_gpu import GpuEngine
+
+    return GpuEngine("cuda:0")
+
+
+def _snap(pods_json, nodes_json, pdbs=None, max_size=100):
+    pods, nodes = wrap(pods_json, nodes_json)
+    pools = group_nodes_into_pools(nodes, max_size=max_size)
+    return encode_snapshot(pods, nodes, pools, pdbs)
+
+
+def _assert_decisions_equal(d_gpu, d_cpu):
+    assert d_gpu.units_needed == d_cpu.units_needed
+    assert d_gpu.new_pool_sizes == d_cpu.new_pool_sizes
+    assert d_gpu.pods_placed_existing == d_cpu.pods_placed_existing
+    assert d_gpu.pods_placed_new == d_cpu.pods_placed_new
+    assert d_gpu.pods_unfulfillable == d_cpu.pods_unfulfillable
+    np.testing.assert_array_equal(
+        d_gpu.pool_residual_totals, d_cpu.pool_residual_totals
+    )
+
+
+def test_native_extension_is_intree(engine):
+    import mi355x_autoscaler._hip_ops as ext
+
+    assert "mi355x_autoscaler" in ext.__file__
+    assert "site-packages" not in ext.__file__
+
+
+@pytest.mark.parametrize("seed", [0, 1, 2])
+@pytest.mark.parametrize(
+    "n_pods,pools,sizes,gpufrac",
+    [
+        (64, (("agentpool1", "Standard_D2_v2"),), (3,), 0.0),
+        (1000, (("gpupool", "Standard_ND96_MI355X_v1"),), (4,), 1.0),
+        (
+            5000,
+            (("agentpool1", "Standard_D2_v2"), ("bigpool", "Standard_D4_v2"),
+             ("gpupool", "Standard_ND96_MI355X_v1")),
+            (40, 30, 30),
+            0.35,
+        ),
+    ],
+)
+def test_decide_matches_oracle(engine, seed, n_pods, pools, sizes, gpufrac):
+    pods_json, nodes_json = burst_trace(
+        n_pods, n_nodes_per_pool=sizes, pools=pools, gpu_fraction=gpufrac,
+        seed=seed, bound_pods_per_node=3,
+    )
+    snap = _snap(pods_json, nodes_json)
+    _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))
+
+
+def test_decide_boundary_cases(engine):
+    # zero-request pods, exactly-full nodes, selector mismatch, pool at max
+    node = make_node_json("agentpool1", 0, cpu_milli=2000, mem_mib=4096, pods=5)
+    filler = make_pod_json("full", cpu_milli=2000, mem_mib=4096,
+                           node_name=node["metadata"]["name"])
+    pods = [filler]
+    pods += [make_pod_json(f"z{i}", cpu_milli=0, mem_mib=0, pending=True) for i in range(7)]
+    pods += [make_pod_json("sel", node_selector={"disk": "ssd"}, pending=True)]
+    pods += [make_pod_json(f"c{i}", cpu_milli=1000, mem_mib=512, pending=True) for i in range(9)]
+    snap = _snap(pods, [node], max_size=3)
+    _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))
+
+
+def test_decide_empty_pending(engine):
+    pods_json, nodes_json = burst_trace(0, n_nodes_per_pool=(3,), seed=0, bound_pods_per_node=2)
+    snap = _snap(pods_json, nodes_json)
+    d = engine.decide(snap)
+    assert not d.wants_scale_up()
+    assert d.pods_placed_existing == 0
+
+
+def test_over_provision_matches(engine):
+    pods_json, nodes_json = burst_trace(300, n_nodes_per_pool=(2,), seed=4)
+    snap = _snap(pods_json, nodes_json)
+    d_gpu = engine.decide(snap, over_provision=2)
+    d_cpu = CpuEngine().decide(snap, over_provision=2)
+    _assert_decisions_equal(d_gpu, d_cpu)
+
+
+@pytest.mark.parametrize("n_nodes", [10, 100, 500])
+def test_maintain_matches_oracle(engine, n_nodes):
+    pods_json, nodes_json, pdbs = scaledown_trace(n_nodes=n_nodes, seed=2)
+    snap = _snap(pods_json, nodes_json, pdbs)
+    rep_gpu = engine.maintain(snap)
+    rep_cpu = CpuEngine().maintain(snap)
+    np.testing.assert_array_equal(rep_gpu.node_util_permille, rep_cpu.node_util_permille)
+    np.testing.assert_array_equal(rep_gpu.node_blocked, rep_cpu.node_blocked)
+    np.testing.assert_array_equal(rep_gpu.node_pdb_ok, rep_cpu.node_pdb_ok)
+    np.testing.assert_array_equal(rep_gpu.node_idle, rep_cpu.node_idle)
+    np.testing.assert_array_equal(rep_gpu.node_drainable_pods, rep_cpu.node_drainable_pods)
+    assert [c.node_name for c in rep_gpu.candidates] == [c.node_name for c in rep_cpu.candidates]
+
+
+def test_segsum_mfma_exactness():
+    """MFMA segmented sums vs numpy over adversarial shapes: max-value
+    elements (limb-exactness bound), ragged segments, empty segments."""
+    from mi355x_autoscaler.ops import load_ops
+    from mi355x_autoscaler.ops.engine_gpu import segsum
+
+    ops = load_ops()
+    rng = np.random.default_rng(0)
+    for M, S in [(1, 1), (64, 3), (1000, 17), (100000, 999), (300000, 4)]:
+        seg = np.sort(rng.integers(0, S, M)).astype(np.int64)
+        vals = rng.integers(0, 2**31 - 1, size=(M, 4), dtype=np.int64).astype(np.int32)
+        # adversarial: a run of max-value elements in one segment
+        vals[: min(M, 300)] = 2**31 - 1
+        seg[: min(M, 300)] = 0
+        seg = np.sort(seg)
+        ref = np.zeros((S, 4), np.int64)
+        np.add.at(ref, seg, vals.astype(np.int64))
+        v = torch.from_numpy(vals).cuda()
+        s = torch.from_numpy(seg).cuda()
+        out = segsum(v, s, S, ops)
+        np.testing.assert_array_equal(out.cpu().numpy(), ref)
+
+
+def test_fit_eligibility_kernel_direct():
+    """Kernel A vs the host predicate on a label-heavy node set."""
+    from mi355x_autoscaler.ops import load_ops
+    from mi355x_autoscaler.ops.engine_gpu import build_segments_device, upload_snapshot
+    from mi355x_autoscaler.snapshot.encode import selector_subset_match
+
+    ops = load_ops()
+    pods_json, nodes_json = burst_trace(
+        500, n_nodes_per_pool=(77, 33),
+        pools=(("agentpool1", "Standard_D2_v2"), ("gpupool", "Standard_ND96_MI355X_v1")),
+        gpu_fraction=0.5, seed=9,
+    )
+    # mark some nodes unschedulable
+    for nd in nodes_json[::7]:
+        nd["spec"]["unschedulable"] = True
+    snap = _snap(pods_json, nodes_json)
+    ds = upload_snapshot(snap, torch.device("cuda:0"))
+    segs = build_segments_device(ds, ops)
+    G, N = segs.G, snap.num_nodes
+    W = (N + 63) // 64
+    elig = torch.zeros((G, W), dtype=torch.int64, device="cuda:0")
+    ops.fit_eligibility(ds.node_labels, ds.node_label_count, ds.node_unsched,
+                        segs.seg_sel_pairs.contiguous(), segs.seg_sel_count.contiguous(), elig)
+    got = elig.cpu().numpy().astype(np.uint64)
+    sel_pairs = segs.seg_sel_pairs.cpu().numpy()
+    sel_count = segs.seg_sel_count.cpu().numpy()
+    for g in range(G):
+        for j in range(N):
+            expect = (not snap.node_unsched[j]) and selector_subset_match(
+                sel_pairs[g], int(sel_count[g]), snap.node_labels[j],
+                int(snap.node_label_count[j]),
+            )
+            bit = (int(got[g, j // 64]) >> (j % 64)) & 1
+            assert bit == int(expect), (g, j)
+
+
+def test_gpu_pipeline_scales_10k(engine):
+    """Config #3 shape: 10k mixed pods, 3 heterogeneous pools."""
+    pods_json, nodes_json = burst_trace(
+        10000,
+        n_nodes_per_pool=(40, 30, 30),
+        pools=(("agentpool1", "Standard_D2_v2"), ("bigpool", "Standard_D4_v2"),
+               ("gpupool", "Standard_ND96_MI355X_v1")),
+        gpu_fraction=0.35, seed=7, bound_pods_per_node=2,
+    )
+    snap = _snap(pods_json, nodes_json)
+    _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))
