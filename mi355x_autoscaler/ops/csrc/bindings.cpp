@@ -34,6 +34,21 @@ void drain_finalize(torch::Tensor util, torch::Tensor node_alloc,
 void segsum_mfma(torch::Tensor values, torch::Tensor seg_of,
                  torch::Tensor wi_win, torch::Tensor wi_start,
                  torch::Tensor wi_end, int64_t num_segments, torch::Tensor out);
+void segment_hash(torch::Tensor pod_flags, torch::Tensor sel_hash,
+                  torch::Tensor pod_req, torch::Tensor slot_ks,
+                  torch::Tensor slot_kr, torch::Tensor slot_state,
+                  torch::Tensor slot_count, torch::Tensor slot_first,
+                  torch::Tensor slot_of_pod, torch::Tensor overflow);
+void segment_compact(torch::Tensor slot_ks, torch::Tensor slot_kr,
+                     torch::Tensor slot_state, torch::Tensor slot_count,
+                     torch::Tensor slot_first, torch::Tensor pod_sel_pairs,
+                     torch::Tensor pod_sel_count, torch::Tensor pod_req,
+                     torch::Tensor seg_sel_pairs, torch::Tensor seg_sel_count,
+                     torch::Tensor seg_sel_hash, torch::Tensor seg_req,
+                     torch::Tensor seg_count, torch::Tensor seg_first_pod,
+                     torch::Tensor slot2seg, torch::Tensor g_out);
+void pod_pool_map(torch::Tensor slot_of_pod, torch::Tensor slot2seg,
+                  torch::Tensor seg_pool, torch::Tensor pod_pool);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native autoscaler scheduling kernels (gfx950 HIP)";
@@ -46,4 +61,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("drain_flags", &drain_flags, "kernel D: drain predicates");
   m.def("drain_finalize", &drain_finalize, "kernel D: utilization + candidates");
   m.def("segsum_mfma", &segsum_mfma, "MFMA segmented resource sums");
+  m.def("segment_hash", &segment_hash, "kernel C: hash-aggregate pending pods");
+  m.def("segment_compact", &segment_compact, "kernel C: compact+sort segments");
+  m.def("pod_pool_map", &pod_pool_map, "pod -> routed pool map");
 }

@@ -1,20 +1,28 @@
 """GPU simulation engine: the HIP kernel pipeline on a device-resident
 cluster snapshot.
 
-Pipeline per decide() — all stages on-device, one small readback of the
-decision vector at the end:
+decide() pipeline — every stage on-device, ONE H2D copy in, ONE small
+D2H readback out:
 
-  upload SoA          host -> HBM (288 GB/GPU: million-pod traces fit)
-  classify_keys       kernel C  (radix keys realizing the oracle order)
-  sort + unique       device radix sort (torch) -> placement segments
-  node_free_subtract  free = allocatable - bound requests (atomics)
-  fit_eligibility     kernel A  (LDS-staged label sets, 64-bit ballots)
-  seg_pool_route      segment -> agent pool
-  segsum_mfma         per-pool pending demand (MFMA segmented sums)
-  binpack_place       kernel B  (FFD as uniform scan-threshold steps)
-  binpack_units       kernel B' (virtual new nodes -> units per pool)
+  packed upload        all SoA arrays in one pinned buffer -> HBM
+  segment_hash_build   lock-free hash aggregation of pending pods into
+                       (selector, request) segments  [kernel C]
+  segment_compact_sort single-block bitonic -> oracle segment order
+  node_free_subtract   free = allocatable - bound requests (atomics)
+  fit_eligibility      LDS-staged label sets, 64-bit ballots [kernel A]
+  seg_pool_route       segment -> agent pool
+  pod_pool_map+segsum  MFMA per-pool pending demand (f32 16x16x4
+                       segmented sums, exact integer limbs)
+  binpack_place        FFD as uniform scan-threshold steps  [kernel B]
+  binpack_units        virtual new nodes -> units per pool  [kernel B']
+  readback             one int64 blob: placed/new/unful per segment,
+                       units + residual per pool
 
-maintain() runs segsum_mfma (per-node utilization) + kernel D.
+The round-1 profile showed the torch sort/unique segment build + 36
+per-array H2D copies were 92% of the decision time; this layout
+replaces them with 2 kernels and 1 copy. A torch-based fallback path
+(`MI355X_AS_FORCE_TORCH_SEGBUILD=1`, or hash overflow: >2048 segments)
+keeps a second independent implementation for differential testing.
 
 Decisions are bit-identical to sim/oracle.py by construction (same
 lattice, same order, same integer arithmetic); tests/test_gpu_engine.py
@@ -23,8 +31,9 @@ asserts it on hardware.
 
 from __future__ import annotations
 
+import os
 from dataclasses import dataclass
-from typing import Dict, List, Optional
+from typing import Dict, List, Optional, Tuple
 
 import numpy as np
 import torch
@@ -36,11 +45,15 @@ from . import load_ops
 
 _PACK_MAX = 1 << 52
 _SEGSUM_CHUNK = 256  # exactness bound for the f32 MFMA limb chains
+SEG_MAX = 2048
+HASH_SLOTS = 8192
+_ALIGN = 64
 
 
 @dataclass
 class DeviceSnapshot:
-    """The SoA snapshot resident in HBM."""
+    """The SoA snapshot resident in HBM (views into one packed buffer
+    when uploaded via upload_snapshot_packed)."""
 
     pod_req: torch.Tensor        # [P,4] i32
     pod_flags: torch.Tensor      # [P]   i32
@@ -60,6 +73,8 @@ class DeviceSnapshot:
     pool_labels: torch.Tensor    # [K,32] i64
     pool_label_count: torch.Tensor  # [K] i32
     pdb_allowed: torch.Tensor    # [D] i32
+    pool_max_h: Optional[np.ndarray] = None
+    pool_cur_h: Optional[np.ndarray] = None
 
     @property
     def P(self) -> int:
@@ -74,33 +89,70 @@ class DeviceSnapshot:
         return int(self.pool_unit.shape[0])
 
 
+_FIELDS = [
+    # (name, source attr, np dtype, trailing shape)
+    ("pod_req", "pod_req", np.int32, (4,)),
+    ("pod_flags", "pod_flags", np.int32, ()),
+    ("pod_sel_hash", "pod_sel_hash", np.int64, ()),
+    ("pod_sel_pairs", "pod_sel_pairs", np.int64, (8,)),
+    ("pod_sel_count", "pod_sel_count", np.int32, ()),
+    ("pod_node", "pod_node", np.int32, ()),
+    ("pod_pdb", "pod_pdb", np.int32, ()),
+    ("node_alloc", "node_alloc", np.int32, (4,)),
+    ("node_labels", "node_labels", np.int64, (32,)),
+    ("node_label_count", "node_label_count", np.int32, ()),
+    ("node_unsched", "node_unsched", np.int32, ()),
+    ("node_pool", "node_pool", np.int32, ()),
+    ("pool_unit", "pool_unit", np.int32, (4,)),
+    ("pool_max", "pool_max", np.int32, ()),
+    ("pool_cur", "pool_cur", np.int32, ()),
+    ("pool_labels", "pool_labels", np.int64, (32,)),
+    ("pool_label_count", "pool_label_count", np.int32, ()),
+    ("pdb_allowed", "pdb_allowed", np.int32, ()),
+]
+
+_TORCH_DT = {np.int32: torch.int32, np.int64: torch.int64}
+
+
+class _PinnedPacker:
+    """Packs every snapshot array into one pinned host buffer so the
+    upload is a single DMA (the profile showed 36 copies/cycle)."""
+
+    def __init__(self):
+        self.buf: Optional[torch.Tensor] = None
+
+    def pack(self, snap: ClusterSnapshot, device) -> DeviceSnapshot:
+        offsets: List[Tuple[str, int, int, np.dtype, tuple]] = []
+        off = 0
+        arrays = {}
+        for name, attr, dt, tail in _FIELDS:
+            a = np.ascontiguousarray(getattr(snap, attr), dtype=dt)
+            arrays[name] = a
+            nb = a.nbytes
+            offsets.append((name, off, nb, dt, a.shape))
+            off += (nb + _ALIGN - 1) // _ALIGN * _ALIGN
+        total = max(off, _ALIGN)
+        if self.buf is None or self.buf.numel() < total:
+            pin = torch.cuda.is_available()
+            self.buf = torch.empty(total, dtype=torch.uint8, pin_memory=pin)
+        host = self.buf.numpy()
+        for name, o, nb, dt, shape in offsets:
+            host[o : o + nb] = arrays[name].view(np.uint8).reshape(-1)
+        dev = self.buf[:total].to(device, non_blocking=True)
+        views = {}
+        for name, o, nb, dt, shape in offsets:
+            t = dev[o : o + nb].view(_TORCH_DT[dt])
+            views[name] = t.view(*shape) if shape else t
+        return DeviceSnapshot(
+            **views,
+            pool_max_h=arrays["pool_max"].copy(),
+            pool_cur_h=arrays["pool_cur"].copy(),
+        )
+
+
 def upload_snapshot(snap: ClusterSnapshot, device) -> DeviceSnapshot:
-    def i32(a):
-        return torch.from_numpy(np.ascontiguousarray(a.astype(np.int32))).to(device, non_blocking=True)
-
-    def i64(a):
-        return torch.from_numpy(np.ascontiguousarray(a.astype(np.int64))).to(device, non_blocking=True)
-
-    return DeviceSnapshot(
-        pod_req=i32(snap.pod_req),
-        pod_flags=i32(snap.pod_flags),
-        pod_sel_hash=i64(snap.pod_sel_hash),
-        pod_sel_pairs=i64(snap.pod_sel_pairs),
-        pod_sel_count=i32(snap.pod_sel_count),
-        pod_node=i32(snap.pod_node),
-        pod_pdb=i32(snap.pod_pdb),
-        node_alloc=i32(snap.node_alloc),
-        node_labels=i64(snap.node_labels),
-        node_label_count=i32(snap.node_label_count),
-        node_unsched=i32(snap.node_unsched),
-        node_pool=i32(snap.node_pool),
-        pool_unit=i32(snap.pool_unit),
-        pool_max=i32(snap.pool_max),
-        pool_cur=i32(snap.pool_cur),
-        pool_labels=i64(snap.pool_labels),
-        pool_label_count=i32(snap.pool_label_count),
-        pdb_allowed=i32(snap.pdb_allowed),
-    )
+    """Unpooled upload (kept for tests and one-shot callers)."""
+    return _PinnedPacker().pack(snap, device)
 
 
 @dataclass
@@ -112,8 +164,11 @@ class DeviceSegments:
     seg_sel_hash: torch.Tensor   # [G] i64
     seg_req: torch.Tensor        # [G,4] i32
     seg_count: torch.Tensor      # [G] i64
-    pod_sorted: torch.Tensor     # [M] i64 pending pod ids, segment-sorted
-    pod_seg: torch.Tensor        # [M] i64 segment id per sorted pod
+    # pod -> segment mapping, one of:
+    slot_of_pod: Optional[torch.Tensor] = None  # [P] i32 (hash path)
+    slot2seg: Optional[torch.Tensor] = None     # [H] i32 (hash path)
+    pod_sorted: Optional[torch.Tensor] = None   # [M] i64 (torch path)
+    pod_seg: Optional[torch.Tensor] = None      # [M] i64 (torch path)
 
     @property
     def G(self) -> int:
@@ -121,6 +176,8 @@ class DeviceSegments:
 
 
 def build_segments_device(ds: DeviceSnapshot, ops) -> DeviceSegments:
+    """torch-sort fallback segment build (second implementation, used
+    on hash overflow and for differential tests)."""
     device = ds.pod_req.device
     P = ds.P
     key_sel = torch.empty(P, dtype=torch.int64, device=device)
@@ -144,14 +201,13 @@ def build_segments_device(ds: DeviceSnapshot, ops) -> DeviceSegments:
 
     ks = key_sel[pend_idx]
     kr = key_req[pend_idx]
-    # (sel asc, req-key asc[=req desc], pod idx asc): two stable passes
     order1 = torch.argsort(kr, stable=True)
     ks1 = ks[order1]
     order2 = torch.argsort(ks1, stable=True)
     order = order1[order2]
     pod_sorted = pend_idx[order].to(torch.int64)
 
-    keys2 = torch.stack([ks[order], kr[order]], dim=1)  # [M,2] sorted rows
+    keys2 = torch.stack([ks[order], kr[order]], dim=1)
     uniq, inverse, counts = torch.unique_consecutive(
         keys2, dim=0, return_inverse=True, return_counts=True
     )
@@ -173,13 +229,12 @@ def build_segments_device(ds: DeviceSnapshot, ops) -> DeviceSegments:
 
 
 def segsum_workitems(seg_of: torch.Tensor, num_segments: int):
-    """Build (window, start, end) work items for segsum_mfma over an
-    ascending segment-id array: windows of 16 segments, chunks of
-    <=_SEGSUM_CHUNK elements that never cross a window boundary."""
+    """(window, start, end) work items for segsum_mfma over an ASCENDING
+    segment-id array: 16-segment windows, <=256-element chunks."""
     device = seg_of.device
     W = (num_segments + 15) // 16
     win_of = torch.div(seg_of, 16, rounding_mode="floor")
-    cnt = torch.bincount(win_of, minlength=W)
+    cnt = torch.bincount(torch.clamp(win_of, min=0), minlength=W)
     starts = torch.zeros(W, dtype=torch.int64, device=device)
     if W > 1:
         starts[1:] = torch.cumsum(cnt, 0)[:-1]
@@ -199,15 +254,28 @@ def segsum_workitems(seg_of: torch.Tensor, num_segments: int):
     return wi_win.to(torch.int32), wi_start, wi_end
 
 
-def segsum(values: torch.Tensor, seg_of_sorted: torch.Tensor, num_segments: int, ops):
-    """MFMA segmented sum -> [S,4] int64 (values >= 0, seg ids sorted)."""
+def segsum(values: torch.Tensor, seg_of: torch.Tensor, num_segments: int, ops,
+           sorted_ids: bool = True):
+    """MFMA segmented sum -> [S,4] int64 (values >= 0; ids with -1 are
+    skipped). With S <= 16 there is a single window, so NO sortedness
+    is required and work items are plain chunks — the zero-sort fast
+    path the decide() demand sum uses."""
     device = values.device
     out = torch.zeros((num_segments, 4), dtype=torch.int64, device=device)
-    if values.shape[0] == 0 or num_segments == 0:
+    M = int(values.shape[0])
+    if M == 0 or num_segments == 0:
         return out
-    wi_win, wi_start, wi_end = segsum_workitems(seg_of_sorted, num_segments)
+    if num_segments <= 16:
+        T = (M + _SEGSUM_CHUNK - 1) // _SEGSUM_CHUNK
+        wi_win = torch.zeros(T, dtype=torch.int32, device=device)
+        wi_start = torch.arange(T, dtype=torch.int64, device=device) * _SEGSUM_CHUNK
+        wi_end = torch.minimum(wi_start + _SEGSUM_CHUNK,
+                               torch.tensor(M, dtype=torch.int64, device=device))
+    else:
+        assert sorted_ids, "segsum with >16 segments needs sorted ids"
+        wi_win, wi_start, wi_end = segsum_workitems(seg_of, num_segments)
     ops.segsum_mfma(
-        values.contiguous(), seg_of_sorted.to(torch.int32).contiguous(),
+        values.contiguous(), seg_of.to(torch.int32).contiguous(),
         wi_win.contiguous(), wi_start.contiguous(), wi_end.contiguous(),
         num_segments, out,
     )
@@ -222,12 +290,76 @@ class GpuEngine(SimulationEngine):
         if not torch.cuda.is_available():
             raise RuntimeError("GpuEngine requires a ROCm device (torch.cuda unavailable)")
         self.ops = load_ops()
+        self.packer = _PinnedPacker()
+        self.force_torch_segbuild = os.environ.get("MI355X_AS_FORCE_TORCH_SEGBUILD") == "1"
         self._last_demand: Optional[torch.Tensor] = None
+        d = self.device
+        # persistent hash workspace
+        self.ws_slot_ks = torch.zeros(HASH_SLOTS, dtype=torch.int64, device=d)
+        self.ws_slot_kr = torch.zeros(HASH_SLOTS, dtype=torch.int64, device=d)
+        self.ws_slot_state = torch.zeros(HASH_SLOTS, dtype=torch.int32, device=d)
+        self.ws_slot_count = torch.zeros(HASH_SLOTS, dtype=torch.int32, device=d)
+        self.ws_slot_first = torch.zeros(HASH_SLOTS, dtype=torch.int32, device=d)
+        self.ws_slot2seg = torch.zeros(HASH_SLOTS, dtype=torch.int32, device=d)
+        self.ws_stat = torch.zeros(3, dtype=torch.int32, device=d)  # [G, sortovf, hashovf]
+        self.ws_seg_sel_pairs = torch.zeros((SEG_MAX, 8), dtype=torch.int64, device=d)
+        self.ws_seg_sel_count = torch.zeros(SEG_MAX, dtype=torch.int32, device=d)
+        self.ws_seg_sel_hash = torch.zeros(SEG_MAX, dtype=torch.int64, device=d)
+        self.ws_seg_req = torch.zeros((SEG_MAX, 4), dtype=torch.int32, device=d)
+        self.ws_seg_count = torch.zeros(SEG_MAX, dtype=torch.int64, device=d)
+        self.ws_seg_first = torch.zeros(SEG_MAX, dtype=torch.int32, device=d)
+
+    # ------------------------------------------------------------------
+    def build_segments_hash(self, ds: DeviceSnapshot):
+        """Hash-aggregation segment build. Returns (DeviceSegments,
+        slot_of_pod) or None on overflow (caller falls back)."""
+        ops = self.ops
+        d = self.device
+        self.ws_slot_state.zero_()
+        self.ws_slot_count.zero_()
+        self.ws_slot_first.fill_(0x7F7F7F7F)
+        self.ws_stat.zero_()
+        slot_of_pod = torch.empty(ds.P, dtype=torch.int32, device=d)
+        overflow = self.ws_stat[2:3]
+        ops.segment_hash(
+            ds.pod_flags, ds.pod_sel_hash, ds.pod_req,
+            self.ws_slot_ks, self.ws_slot_kr, self.ws_slot_state,
+            self.ws_slot_count, self.ws_slot_first, slot_of_pod, overflow,
+        )
+        ops.segment_compact(
+            self.ws_slot_ks, self.ws_slot_kr, self.ws_slot_state,
+            self.ws_slot_count, self.ws_slot_first,
+            ds.pod_sel_pairs, ds.pod_sel_count, ds.pod_req,
+            self.ws_seg_sel_pairs, self.ws_seg_sel_count, self.ws_seg_sel_hash,
+            self.ws_seg_req, self.ws_seg_count, self.ws_seg_first,
+            self.ws_slot2seg, self.ws_stat[0:2],
+        )
+        stat = self.ws_stat.cpu().numpy()  # the ONE early sync (G needed for shapes)
+        G, sort_ovf, hash_ovf = int(stat[0]), int(stat[1]), int(stat[2])
+        if sort_ovf or hash_ovf:
+            return None
+        segs = DeviceSegments(
+            seg_sel_pairs=self.ws_seg_sel_pairs[:G],
+            seg_sel_count=self.ws_seg_sel_count[:G],
+            seg_sel_hash=self.ws_seg_sel_hash[:G],
+            seg_req=self.ws_seg_req[:G],
+            seg_count=self.ws_seg_count[:G],
+            slot_of_pod=slot_of_pod,
+            slot2seg=self.ws_slot2seg,
+        )
+        return segs
+
+    def build_segments(self, ds: DeviceSnapshot) -> DeviceSegments:
+        if not self.force_torch_segbuild:
+            segs = self.build_segments_hash(ds)
+            if segs is not None:
+                return segs
+        return build_segments_device(ds, self.ops)
 
     # ------------------------------------------------------------------
     def decide(self, snap: ClusterSnapshot, over_provision: int = 0) -> ScaleDecision:
-        ds = upload_snapshot(snap, self.device)
-        segs = build_segments_device(ds, self.ops)
+        ds = self.packer.pack(snap, self.device)
+        segs = self.build_segments(ds)
         return self.decide_device(ds, segs, snap.pool_names, over_provision)
 
     def decide_device(
@@ -236,7 +368,6 @@ class GpuEngine(SimulationEngine):
         segs: DeviceSegments,
         pool_names: List[str],
         over_provision: int = 0,
-        seg_count_override: Optional[torch.Tensor] = None,
         free_hook=None,
     ) -> ScaleDecision:
         ops = self.ops
@@ -246,8 +377,6 @@ class GpuEngine(SimulationEngine):
         if G == 0:
             decision.pool_residual_totals = np.zeros((K, 4), np.int64)
             return decision
-
-        seg_count = seg_count_override if seg_count_override is not None else segs.seg_count
 
         # free capacity (local bound pods; the sharded engine's hook
         # all-reduces usage across ranks so free is globally correct)
@@ -271,54 +400,55 @@ class GpuEngine(SimulationEngine):
             ds.pool_unit, seg_pool,
         )
 
-        # MFMA per-pool pending demand (local-shard pods)
-        if segs.pod_sorted.numel() and K > 0:
-            pod_pool = seg_pool.to(torch.int64)[segs.pod_seg]
-            keep = pod_pool >= 0
-            pp = pod_pool[keep]
-            vals = ds.pod_req[segs.pod_sorted[keep]]
-            order = torch.argsort(pp, stable=True)
-            demand = segsum(vals[order].contiguous(), pp[order], K, ops)
-        else:
-            demand = torch.zeros((K, 4), dtype=torch.int64, device=device)
-        self._last_demand = demand
+        # MFMA per-pool pending demand over the pod axis
+        self._last_demand = self._pool_demand(ds, segs, seg_pool, K)
 
         # FFD placement over existing nodes
-        placed = torch.zeros(G, dtype=torch.int64, device=device)
+        blob = torch.zeros(3 * G + 5 * max(K, 1), dtype=torch.int64, device=device)
+        placed = blob[0:G]
+        placed_new = blob[G : 2 * G]
+        unful = blob[2 * G : 3 * G]
+        units_cnt = blob[3 * G : 3 * G + K] if K else blob[3 * G : 3 * G]
+        pool_residual = blob[3 * G + K : 3 * G + 5 * K].view(max(K, 1) if K else 0, 4) \
+            if K else torch.zeros((0, 4), dtype=torch.int64, device=device)
         if N > 0:
-            ops.binpack_place(free, elig, segs.seg_req.contiguous(), seg_count.contiguous(), placed)
+            ops.binpack_place(free, elig, segs.seg_req.contiguous(),
+                              segs.seg_count.contiguous(), placed)
 
         # units phase
-        max_new = torch.clamp(ds.pool_max - ds.pool_cur, min=0)
-        maxu = max(int(max_new.max().item()) if K else 0, 1)
+        max_new_h = np.maximum(
+            (ds.pool_max_h if ds.pool_max_h is not None else ds.pool_max.cpu().numpy())
+            - (ds.pool_cur_h if ds.pool_cur_h is not None else ds.pool_cur.cpu().numpy()),
+            0,
+        ).astype(np.int32) if K else np.zeros(0, np.int32)
+        maxu = max(int(max_new_h.max()) if K else 0, 1)
         units_free = torch.zeros((max(K, 1), maxu, 4), dtype=torch.int64, device=device)
-        units_cnt = torch.zeros(max(K, 1), dtype=torch.int32, device=device)
-        placed_new = torch.zeros(G, dtype=torch.int64, device=device)
-        unful = torch.zeros(G, dtype=torch.int64, device=device)
-        pool_residual = torch.zeros((max(K, 1), 4), dtype=torch.int64, device=device)
-        ops.binpack_units(
-            seg_pool, segs.seg_req.contiguous(), seg_count.contiguous(), placed,
-            ds.pool_unit, max_new.to(torch.int32).contiguous(),
-            units_free, units_cnt, placed_new, unful, pool_residual,
-        )
+        max_new = torch.from_numpy(max_new_h).to(device)
+        if K:
+            ops.binpack_units(
+                seg_pool, segs.seg_req.contiguous(), segs.seg_count.contiguous(), placed,
+                ds.pool_unit, max_new, units_free, units_cnt,
+                placed_new, unful, pool_residual,
+            )
 
-        # single small readback
-        placed_h = placed.cpu().numpy()
-        placed_new_h = placed_new.cpu().numpy()
-        unful_h = unful.cpu().numpy()
-        units_h = units_cnt.cpu().numpy()
-        max_new_h = max_new.cpu().numpy()
-        pool_cur_h = ds.pool_cur.cpu().numpy()
-        residual_h = pool_residual.cpu().numpy().astype(np.int64)
-        sel_hash_h = segs.seg_sel_hash.cpu().numpy()
+        # single readback
+        blob_h = blob.cpu().numpy()
+        placed_h = blob_h[0:G]
+        placed_new_h = blob_h[G : 2 * G]
+        unful_h = blob_h[2 * G : 3 * G]
+        units_h = blob_h[3 * G : 3 * G + K]
+        residual_h = blob_h[3 * G + K : 3 * G + 5 * K].reshape(K, 4) if K else np.zeros((0, 4), np.int64)
+        pool_cur_h = ds.pool_cur_h if ds.pool_cur_h is not None else ds.pool_cur.cpu().numpy()
 
         decision.pods_placed_existing = int(placed_h.sum())
         decision.pods_placed_new = int(placed_new_h.sum())
         decision.pods_unfulfillable = int(unful_h.sum())
-        for g in np.nonzero(unful_h > 0)[0]:
-            decision.unfulfillable_detail.append(
-                (int(sel_hash_h[g]) & ((1 << 64) - 1), int(unful_h[g]))
-            )
+        if decision.pods_unfulfillable:
+            sel_hash_h = segs.seg_sel_hash.cpu().numpy()  # rare second readback
+            for g in np.nonzero(unful_h > 0)[0]:
+                decision.unfulfillable_detail.append(
+                    (int(sel_hash_h[g]) & ((1 << 64) - 1), int(unful_h[g]))
+                )
         for k in range(K):
             n_units = int(units_h[k])
             if n_units > 0 and over_provision > 0:
@@ -326,19 +456,44 @@ class GpuEngine(SimulationEngine):
             if n_units > 0:
                 decision.units_needed[pool_names[k]] = n_units
                 decision.new_pool_sizes[pool_names[k]] = int(pool_cur_h[k]) + n_units
-        decision.pool_residual_totals = residual_h[:K]
+        decision.pool_residual_totals = residual_h.copy()
         return decision
+
+    def _pool_demand(self, ds: DeviceSnapshot, segs: DeviceSegments,
+                     seg_pool: torch.Tensor, K: int) -> torch.Tensor:
+        """Per-pool pending demand via the MFMA segmented sum. K <= 16:
+        zero-sort fast path over the raw pod order."""
+        device = self.device
+        if K == 0:
+            return torch.zeros((0, 4), dtype=torch.int64, device=device)
+        if segs.slot_of_pod is not None:
+            pod_pool = torch.empty(ds.P, dtype=torch.int32, device=device)
+            self.ops.pod_pool_map(segs.slot_of_pod, segs.slot2seg, seg_pool, pod_pool)
+            if K <= 16:
+                return segsum(ds.pod_req, pod_pool, K, self.ops)
+            pp, order = torch.sort(pod_pool.to(torch.int64), stable=True)
+            keep = pp >= 0
+            return segsum(ds.pod_req[order][keep].contiguous(), pp[keep], K, self.ops)
+        if segs.pod_sorted is not None and segs.pod_sorted.numel():
+            pod_pool = seg_pool.to(torch.int64)[segs.pod_seg]
+            vals = ds.pod_req[segs.pod_sorted]
+            if K <= 16:
+                return segsum(vals.contiguous(), pod_pool, K, self.ops)
+            pp, order = torch.sort(pod_pool, stable=True)
+            keep = pp >= 0
+            return segsum(vals[order][keep].contiguous(), pp[keep], K, self.ops)
+        return torch.zeros((K, 4), dtype=torch.int64, device=device)
 
     # ------------------------------------------------------------------
     def maintain(self, snap: ClusterSnapshot, util_threshold_permille: int = 300) -> MaintainReport:
-        ds = upload_snapshot(snap, self.device)
+        ds = self.packer.pack(snap, self.device)
         rep, _ = self.maintain_device(ds, util_threshold_permille)
         rep.candidates = self._candidates(rep, snap.node_names, snap, util_threshold_permille)
         return rep
 
     def maintain_device(self, ds: DeviceSnapshot, util_threshold_permille: int = 300):
         """Device passes only; returns (MaintainReport without
-        candidates, device tensors dict) so the sharded engine can
+        candidates, device tensor dict) so the sharded engine can
         all-reduce the partials before finalizing."""
         ops = self.ops
         device = self.device
@@ -353,11 +508,13 @@ class GpuEngine(SimulationEngine):
         ops.drain_flags(ds.pod_node, ds.pod_flags, ds.pod_pdb, pdb_allowed,
                         blocked, pdb_bad, nontrivial, drainable)
 
-        bound = torch.nonzero(ds.pod_node >= 0, as_tuple=False).flatten()
-        if bound.numel() and N > 0:
-            nodes_of = ds.pod_node[bound].to(torch.int64)
-            order = torch.argsort(nodes_of, stable=True)
-            util = segsum(ds.pod_req[bound][order].contiguous(), nodes_of[order], N, self.ops)
+        if ds.P and N > 0:
+            if N <= 16:
+                util = segsum(ds.pod_req, ds.pod_node, N, self.ops)
+            else:
+                nodes_of, order = torch.sort(ds.pod_node.to(torch.int64), stable=True)
+                keep = nodes_of >= 0
+                util = segsum(ds.pod_req[order][keep].contiguous(), nodes_of[keep], N, self.ops)
         else:
             util = torch.zeros((N, 4), dtype=torch.int64, device=device)
 
@@ -380,10 +537,10 @@ class GpuEngine(SimulationEngine):
             )
         return MaintainReport(
             node_util_permille=util_permille.cpu().numpy(),
-            node_blocked=tensors["blocked"].cpu().numpy().astype(np.uint8),
+            node_blocked=(tensors["blocked"].cpu().numpy() != 0).astype(np.uint8),
             node_pdb_ok=(tensors["pdb_bad"].cpu().numpy() == 0).astype(np.uint8),
             node_idle=(tensors["nontrivial"].cpu().numpy() == 0).astype(np.uint8),
-            node_drainable_pods=tensors["drainable"].cpu().numpy(),
+            node_drainable_pods=tensors["drainable"].cpu().numpy().astype(np.int32),
         )
 
     @staticmethod

@@ -94,7 +94,7 @@ extern "C" __global__ __launch_bounds__(WAVE) void units_kernel(
     const int32_t* __restrict__ pool_unit,      // [K, NRES]
     const int32_t* __restrict__ pool_max_new,   // [K]
     int32_t G, int32_t K, int32_t MAXU,
-    int64_t* __restrict__ units_free, int32_t* __restrict__ units_cnt,
+    int64_t* __restrict__ units_free, int64_t* __restrict__ units_cnt,
     int64_t* __restrict__ placed_new, int64_t* __restrict__ unful,
     int64_t* __restrict__ pool_residual)
 {
@@ -121,7 +121,7 @@ extern "C" __global__ __launch_bounds__(WAVE) void units_kernel(
     const int64_t need0 = rem;
 
     // 1) fill already-open units in creation order (scan-threshold FFD)
-    int nopen = units_cnt[k];
+    int nopen = (int)units_cnt[k];
     for (int u0 = 0; u0 < nopen && rem > 0; u0 += WAVE) {
       const int u = u0 + lane;
       int64_t m = 0;
@@ -187,7 +187,7 @@ extern "C" __global__ __launch_bounds__(WAVE) void units_kernel(
       const int64_t absorbed =
           n_new * per_unit < rem ? (int64_t)n_new * per_unit : rem;
       rem -= absorbed;
-      if (lane == 0) units_cnt[k] = nopen + n_new;
+      if (lane == 0) units_cnt[k] = (int64_t)(nopen + n_new);
     }
     if (lane == 0) {
       placed_new[g] = need0 - rem;
@@ -228,7 +228,7 @@ void binpack_units(torch::Tensor seg_pool, torch::Tensor seg_req,
                      seg_count.data_ptr<int64_t>(), placed.data_ptr<int64_t>(),
                      pool_unit.data_ptr<int32_t>(),
                      pool_max_new.data_ptr<int32_t>(), G, K, MAXU,
-                     units_free.data_ptr<int64_t>(), units_cnt.data_ptr<int32_t>(),
+                     units_free.data_ptr<int64_t>(), units_cnt.data_ptr<int64_t>(),
                      placed_new.data_ptr<int64_t>(), unful.data_ptr<int64_t>(),
                      pool_residual.data_ptr<int64_t>());
 }

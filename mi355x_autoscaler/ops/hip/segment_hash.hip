@@ -1,0 +1,291 @@
+// Device-side segment aggregation — the "radix-hash pass" of SURVEY
+// §2.2 row C, as a lock-free hash table instead of a sort.
+//
+// Replaces the torch sort/unique/nonzero chain (~15 rocprim kernel
+// launches per cycle, 92% of the decision's GPU time in the round-1
+// profile) with two launches:
+//
+//   segment_hash_build: one pass over the pod SoA; pending pods CAS-
+//     claim open-addressed slots keyed by (selector-set hash, request
+//     pack key) and accumulate {count, min pod id} with device-scope
+//     integer atomics (exact, order-free). Inter-workgroup publication
+//     follows the agent-scope release/ready-flag discipline (guide §6
+//     G16): the claimer stores both keys with relaxed agent-scope
+//     atomics (L2-direct), then release-stores READY; probers read the
+//     keys with relaxed agent-scope atomic loads, so no per-probe
+//     acquire fence (no L1 involvement at all on the shared words).
+//     Spins are bounded; on bound/overflow a flag routes the host to
+//     the torch fallback path.
+//
+//   segment_compact_sort: ONE 1024-thread block compacts occupied
+//     slots into LDS and bitonic-sorts them by (sel-hash as uint64
+//     asc, request key desc) — the oracle's global segment order —
+//     then writes the segment SoA + the slot->segment LUT.
+//
+// Capacity: H slots (pow2), <= SEG_MAX segments. Real bursts have
+// O(10..100) distinct (selector, shape) pairs; 2048 is generous.
+#include "common.h"
+
+#define SEG_MAX 2048
+#define ST_EMPTY 0
+#define ST_CLAIMED 1
+#define ST_READY 2
+#define FIRST_INIT 0x7f7f7f7f
+#define MAX_SPIN 100000
+
+#define RLX_AGENT __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT
+
+__device__ inline uint32_t mix_hash(int64_t ks, int64_t kr) {
+  uint64_t x = (uint64_t)ks * 0x9E3779B97F4A7C15ull ^
+               (uint64_t)kr * 0xC2B2AE3D27D4EB4Full;
+  x ^= x >> 29;
+  x *= 0xBF58476D1CE4E5B9ull;
+  x ^= x >> 32;
+  return (uint32_t)x;
+}
+
+extern "C" __global__ void segment_hash_build(
+    const int32_t* __restrict__ pod_flags, const int64_t* __restrict__ sel_hash,
+    const int32_t* __restrict__ pod_req, int64_t P, int32_t H,
+    int64_t* __restrict__ slot_ks, int64_t* __restrict__ slot_kr,
+    int32_t* __restrict__ slot_state, int32_t* __restrict__ slot_count,
+    int32_t* __restrict__ slot_first, int32_t* __restrict__ slot_of_pod,
+    int32_t* __restrict__ overflow) {
+  const uint32_t mask = (uint32_t)H - 1u;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < P;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (!(pod_flags[i] & PF_PENDING)) {
+      slot_of_pod[i] = -1;
+      continue;
+    }
+    const int64_t ks = sel_hash[i];
+    int32_t rq[NRES];
+#pragma unroll
+    for (int r = 0; r < NRES; ++r) rq[r] = pod_req[i * NRES + r];
+    const int64_t kr = pack_req_key(rq);
+
+    uint32_t h = mix_hash(ks, kr) & mask;
+    int found = -1;
+    for (int probe = 0; probe < H; ++probe) {
+      int st = __hip_atomic_load(&slot_state[h], RLX_AGENT);
+      if (st == ST_EMPTY) {
+        int expected = ST_EMPTY;
+        if (__hip_atomic_compare_exchange_strong(
+                &slot_state[h], &expected, ST_CLAIMED, __ATOMIC_RELAXED,
+                __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)) {
+          __hip_atomic_store(&slot_ks[h], ks, RLX_AGENT);
+          __hip_atomic_store(&slot_kr[h], kr, RLX_AGENT);
+          __hip_atomic_store(&slot_state[h], ST_READY,
+                             __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+          found = (int)h;
+          break;
+        }
+        st = expected;  // observed state; handle CLAIMED/READY below
+      }
+      if (st == ST_CLAIMED) {
+        // claimer is between CAS and READY: spin briefly on this slot
+        int spins = 0;
+        while ((st = __hip_atomic_load(&slot_state[h], RLX_AGENT)) == ST_CLAIMED) {
+          __builtin_amdgcn_s_sleep(1);
+          if (++spins > MAX_SPIN) {
+            atomicOr(overflow, 2);
+            st = ST_READY;  // bail; comparison below will mismatch or not
+            break;
+          }
+        }
+      }
+      if (st == ST_READY) {
+        if (__hip_atomic_load(&slot_ks[h], RLX_AGENT) == ks &&
+            __hip_atomic_load(&slot_kr[h], RLX_AGENT) == kr) {
+          found = (int)h;
+          break;
+        }
+      }
+      h = (h + 1u) & mask;
+    }
+    if (found < 0) {
+      atomicOr(overflow, 1);
+      slot_of_pod[i] = -1;
+      continue;
+    }
+    atomicAdd(&slot_count[found], 1);
+    atomicMin(&slot_first[found], (int32_t)i);
+    slot_of_pod[i] = found;
+  }
+}
+
+// ---------------------------------------------------------------------------
+
+#define CS_BLOCK 1024
+
+extern "C" __global__ __launch_bounds__(CS_BLOCK) void segment_compact_sort(
+    const int64_t* __restrict__ slot_ks, const int64_t* __restrict__ slot_kr,
+    const int32_t* __restrict__ slot_state, const int32_t* __restrict__ slot_count,
+    const int32_t* __restrict__ slot_first, int32_t H,
+    const int64_t* __restrict__ pod_sel_pairs,  // [P, MAXSEL]
+    const int32_t* __restrict__ pod_sel_count,
+    const int32_t* __restrict__ pod_req,
+    // outputs
+    int64_t* __restrict__ seg_sel_pairs,  // [SEG_MAX, MAXSEL]
+    int32_t* __restrict__ seg_sel_count, int64_t* __restrict__ seg_sel_hash,
+    int32_t* __restrict__ seg_req, int64_t* __restrict__ seg_count,
+    int32_t* __restrict__ seg_first_pod,
+    int32_t* __restrict__ slot2seg,       // [H]
+    int32_t* __restrict__ g_out,          // [2]: G, overflow(G>SEG_MAX)
+    int32_t sort_dir_check)               // unused; keeps ABI room
+{
+  __shared__ uint64_t lds_ks[SEG_MAX];
+  __shared__ uint64_t lds_kr[SEG_MAX];
+  __shared__ int32_t lds_slot[SEG_MAX];
+  __shared__ int32_t lds_n;
+
+  const int tid = threadIdx.x;
+  if (tid == 0) lds_n = 0;
+  __syncthreads();
+
+  // compact occupied slots into LDS (order arbitrary; sort fixes it)
+  for (int h = tid; h < H; h += CS_BLOCK) {
+    if (slot_state[h] == ST_READY && slot_count[h] > 0) {
+      const int g = atomicAdd(&lds_n, 1);
+      if (g < SEG_MAX) {
+        // sel asc as uint64: flip the sign bit so int compare == uint
+        lds_ks[g] = (uint64_t)slot_ks[h] ^ 0x8000000000000000ull;
+        // req desc: invert the pack key
+        lds_kr[g] = ~(uint64_t)slot_kr[h];
+        lds_slot[g] = h;
+      }
+    }
+  }
+  __syncthreads();
+  int G = lds_n;
+  if (G > SEG_MAX) {
+    if (tid == 0) {
+      g_out[0] = 0;
+      g_out[1] = 1;
+    }
+    return;
+  }
+  // pad to pow2 with +inf keys
+  int NP = 1;
+  while (NP < G) NP <<= 1;
+  if (NP < 2) NP = 2;
+  for (int g = G + tid; g < NP; g += CS_BLOCK) {
+    lds_ks[g] = ~0ull;
+    lds_kr[g] = ~0ull;
+    lds_slot[g] = -1;
+  }
+  __syncthreads();
+
+  // bitonic sort ascending by (ks, kr)
+  for (int k = 2; k <= NP; k <<= 1) {
+    for (int j = k >> 1; j > 0; j >>= 1) {
+      for (int t = tid; t < NP; t += CS_BLOCK) {
+        const int ixj = t ^ j;
+        if (ixj > t) {
+          const bool up = ((t & k) == 0);
+          const bool gt = (lds_ks[t] > lds_ks[ixj]) ||
+                          (lds_ks[t] == lds_ks[ixj] && lds_kr[t] > lds_kr[ixj]);
+          if (gt == up) {
+            uint64_t a = lds_ks[t]; lds_ks[t] = lds_ks[ixj]; lds_ks[ixj] = a;
+            uint64_t b = lds_kr[t]; lds_kr[t] = lds_kr[ixj]; lds_kr[ixj] = b;
+            int32_t c = lds_slot[t]; lds_slot[t] = lds_slot[ixj]; lds_slot[ixj] = c;
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  // emit segment SoA in sorted order + slot->segment LUT
+  for (int g = tid; g < G; g += CS_BLOCK) {
+    const int h = lds_slot[g];
+    slot2seg[h] = g;
+    const int fp = slot_first[h];
+    seg_sel_hash[g] = (int64_t)(lds_ks[g] ^ 0x8000000000000000ull);
+    seg_count[g] = (int64_t)slot_count[h];
+    seg_first_pod[g] = fp;
+    seg_sel_count[g] = pod_sel_count[fp];
+#pragma unroll
+    for (int q = 0; q < MAXSEL; ++q)
+      seg_sel_pairs[(int64_t)g * MAXSEL + q] = pod_sel_pairs[(int64_t)fp * MAXSEL + q];
+#pragma unroll
+    for (int r = 0; r < NRES; ++r)
+      seg_req[g * NRES + r] = pod_req[(int64_t)fp * NRES + r];
+  }
+  if (tid == 0) {
+    g_out[0] = G;
+    g_out[1] = 0;
+  }
+}
+
+// pod -> routed-pool map (for the MFMA demand segsum): pool id of the
+// pod's segment, or -1 for non-pending pods.
+extern "C" __global__ void pod_pool_map_kernel(
+    const int32_t* __restrict__ slot_of_pod, const int32_t* __restrict__ slot2seg,
+    const int32_t* __restrict__ seg_pool, int64_t P,
+    int32_t* __restrict__ pod_pool) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < P;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int s = slot_of_pod[i];
+    pod_pool[i] = (s >= 0) ? seg_pool[slot2seg[s]] : -1;
+  }
+}
+
+#include <torch/extension.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+
+void segment_hash(torch::Tensor pod_flags, torch::Tensor sel_hash,
+                  torch::Tensor pod_req, torch::Tensor slot_ks,
+                  torch::Tensor slot_kr, torch::Tensor slot_state,
+                  torch::Tensor slot_count, torch::Tensor slot_first,
+                  torch::Tensor slot_of_pod, torch::Tensor overflow) {
+  const int64_t P = pod_flags.size(0);
+  const int H = (int)slot_ks.size(0);
+  if (P == 0) return;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const int block = 256;
+  const int grid = (int)std::min<int64_t>((P + block - 1) / block, 2048);
+  hipLaunchKernelGGL(segment_hash_build, dim3(grid), dim3(block), 0,
+                     stream.stream(), pod_flags.data_ptr<int32_t>(),
+                     sel_hash.data_ptr<int64_t>(), pod_req.data_ptr<int32_t>(),
+                     P, H, slot_ks.data_ptr<int64_t>(), slot_kr.data_ptr<int64_t>(),
+                     slot_state.data_ptr<int32_t>(), slot_count.data_ptr<int32_t>(),
+                     slot_first.data_ptr<int32_t>(), slot_of_pod.data_ptr<int32_t>(),
+                     overflow.data_ptr<int32_t>());
+}
+
+void segment_compact(torch::Tensor slot_ks, torch::Tensor slot_kr,
+                     torch::Tensor slot_state, torch::Tensor slot_count,
+                     torch::Tensor slot_first, torch::Tensor pod_sel_pairs,
+                     torch::Tensor pod_sel_count, torch::Tensor pod_req,
+                     torch::Tensor seg_sel_pairs, torch::Tensor seg_sel_count,
+                     torch::Tensor seg_sel_hash, torch::Tensor seg_req,
+                     torch::Tensor seg_count, torch::Tensor seg_first_pod,
+                     torch::Tensor slot2seg, torch::Tensor g_out) {
+  const int H = (int)slot_ks.size(0);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(segment_compact_sort, dim3(1), dim3(CS_BLOCK), 0,
+                     stream.stream(), slot_ks.data_ptr<int64_t>(),
+                     slot_kr.data_ptr<int64_t>(), slot_state.data_ptr<int32_t>(),
+                     slot_count.data_ptr<int32_t>(), slot_first.data_ptr<int32_t>(),
+                     H, pod_sel_pairs.data_ptr<int64_t>(),
+                     pod_sel_count.data_ptr<int32_t>(), pod_req.data_ptr<int32_t>(),
+                     seg_sel_pairs.data_ptr<int64_t>(),
+                     seg_sel_count.data_ptr<int32_t>(),
+                     seg_sel_hash.data_ptr<int64_t>(), seg_req.data_ptr<int32_t>(),
+                     seg_count.data_ptr<int64_t>(), seg_first_pod.data_ptr<int32_t>(),
+                     slot2seg.data_ptr<int32_t>(), g_out.data_ptr<int32_t>(), 0);
+}
+
+void pod_pool_map(torch::Tensor slot_of_pod, torch::Tensor slot2seg,
+                  torch::Tensor seg_pool, torch::Tensor pod_pool) {
+  const int64_t P = slot_of_pod.size(0);
+  if (P == 0) return;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const int block = 256;
+  const int grid = (int)std::min<int64_t>((P + block - 1) / block, 2048);
+  hipLaunchKernelGGL(pod_pool_map_kernel, dim3(grid), dim3(block), 0,
+                     stream.stream(), slot_of_pod.data_ptr<int32_t>(),
+                     slot2seg.data_ptr<int32_t>(), seg_pool.data_ptr<int32_t>(),
+                     P, pod_pool.data_ptr<int32_t>());
+}

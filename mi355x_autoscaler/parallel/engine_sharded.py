@@ -130,10 +130,8 @@ class ShardedEngine(SimulationEngine):
             sim = OracleSimulator(snap, over_provision=over_provision)
             decision = sim.fulfill_pending(segments, node_free=node_free)
         else:
-            from ..ops.engine_gpu import DeviceSegments, build_segments_device, upload_snapshot
-
-            ds = upload_snapshot(snap, device)
-            segs = build_segments_device(ds, self.inner.ops)
+            ds = self.inner.packer.pack(snap, device)
+            segs = self.inner.build_segments(ds)
             rows = self._device_segments_to_rows(ds, segs)
             rows = pdist.allgather_rows(rows, device=device)
             merged = merge_rows(rows)
@@ -234,9 +232,7 @@ class ShardedEngine(SimulationEngine):
                 partials = {k: v.numpy() for k, v in t.items()}
             return sim.finalize_maintain(partials, util_threshold_permille)
 
-        from ..ops.engine_gpu import upload_snapshot
-
-        ds = upload_snapshot(snap, device)
+        ds = self.inner.packer.pack(snap, device)
         _, tensors = self.inner.maintain_device(ds, util_threshold_permille)
         if self.active:
             pdist.allreduce_tensor(tensors["util"], "sum")

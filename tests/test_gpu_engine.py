@@ -183,3 +183,61 @@ def test_gpu_pipeline_scales_10k(engine):
     )
     snap = _snap(pods_json, nodes_json)
     _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))
+
+
+def test_segment_hash_path_equals_torch_path(engine):
+    """Differential test: hash-aggregation segment build vs the
+    torch-sort fallback produce identical segments and decisions."""
+    import os
+
+    from mi355x_autoscaler.ops.engine_gpu import GpuEngine
+
+    pods_json, nodes_json = burst_trace(
+        4000, n_nodes_per_pool=(20, 10),
+        pools=(("agentpool1", "Standard_D2_v2"), ("gpupool", "Standard_ND96_MI355X_v1")),
+        gpu_fraction=0.4, seed=13, bound_pods_per_node=2,
+    )
+    snap = _snap(pods_json, nodes_json)
+
+    ds = engine.packer.pack(snap, engine.device)
+    segs_hash = engine.build_segments_hash(ds)
+    assert segs_hash is not None
+    from mi355x_autoscaler.ops.engine_gpu import build_segments_device
+
+    segs_torch = build_segments_device(ds, engine.ops)
+    assert segs_hash.G == segs_torch.G
+    np.testing.assert_array_equal(
+        segs_hash.seg_sel_hash.cpu().numpy(), segs_torch.seg_sel_hash.cpu().numpy()
+    )
+    np.testing.assert_array_equal(
+        segs_hash.seg_req.cpu().numpy(), segs_torch.seg_req.cpu().numpy()
+    )
+    np.testing.assert_array_equal(
+        segs_hash.seg_count.cpu().numpy(), segs_torch.seg_count.cpu().numpy()
+    )
+
+    d_hash = engine.decide(snap)
+    try:
+        engine.force_torch_segbuild = True
+        d_torch = engine.decide(snap)
+    finally:
+        engine.force_torch_segbuild = False
+    _assert_decisions_equal(d_hash, d_torch)
+    _assert_decisions_equal(d_hash, CpuEngine().decide(snap))
+
+
+def test_segment_hash_many_segments(engine):
+    """Hundreds of distinct (selector, request) shapes still fit the
+    hash table and match the oracle."""
+    import random
+
+    rng = random.Random(0)
+    node = make_node_json("agentpool1", 0)
+    pods = []
+    for i in range(900):
+        pods.append(make_pod_json(
+            f"p{i}", cpu_milli=rng.randrange(1, 400), mem_mib=rng.randrange(1, 300),
+            pending=True,
+        ))
+    snap = _snap(pods, [node])
+    _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))
