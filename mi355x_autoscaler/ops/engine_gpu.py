@@ -294,14 +294,17 @@ class GpuEngine(SimulationEngine):
         self.force_torch_segbuild = os.environ.get("MI355X_AS_FORCE_TORCH_SEGBUILD") == "1"
         self._last_demand: Optional[torch.Tensor] = None
         d = self.device
-        # persistent hash workspace
+        # persistent hash workspace; the mutable i32 parts live in ONE
+        # buffer so per-decide reinit is a single fill kernel
         self.ws_slot_ks = torch.zeros(HASH_SLOTS, dtype=torch.int64, device=d)
         self.ws_slot_kr = torch.zeros(HASH_SLOTS, dtype=torch.int64, device=d)
-        self.ws_slot_state = torch.zeros(HASH_SLOTS, dtype=torch.int32, device=d)
-        self.ws_slot_count = torch.zeros(HASH_SLOTS, dtype=torch.int32, device=d)
-        self.ws_slot_first = torch.zeros(HASH_SLOTS, dtype=torch.int32, device=d)
+        self._ws_i32 = torch.zeros(3 * HASH_SLOTS + 3, dtype=torch.int32, device=d)
+        self.ws_slot_state = self._ws_i32[0:HASH_SLOTS]
+        self.ws_slot_count = self._ws_i32[HASH_SLOTS : 2 * HASH_SLOTS]
+        # stores INT32_MAX - min(pod id) under atomicMax => zero-init
+        self.ws_slot_first = self._ws_i32[2 * HASH_SLOTS : 3 * HASH_SLOTS]
+        self.ws_stat = self._ws_i32[3 * HASH_SLOTS :]  # [G, sortovf, hashovf]
         self.ws_slot2seg = torch.zeros(HASH_SLOTS, dtype=torch.int32, device=d)
-        self.ws_stat = torch.zeros(3, dtype=torch.int32, device=d)  # [G, sortovf, hashovf]
         self.ws_seg_sel_pairs = torch.zeros((SEG_MAX, 8), dtype=torch.int64, device=d)
         self.ws_seg_sel_count = torch.zeros(SEG_MAX, dtype=torch.int32, device=d)
         self.ws_seg_sel_hash = torch.zeros(SEG_MAX, dtype=torch.int64, device=d)
@@ -315,10 +318,7 @@ class GpuEngine(SimulationEngine):
         slot_of_pod) or None on overflow (caller falls back)."""
         ops = self.ops
         d = self.device
-        self.ws_slot_state.zero_()
-        self.ws_slot_count.zero_()
-        self.ws_slot_first.fill_(0x7F7F7F7F)
-        self.ws_stat.zero_()
+        self._ws_i32.zero_()
         slot_of_pod = torch.empty(ds.P, dtype=torch.int32, device=d)
         overflow = self.ws_stat[2:3]
         ops.segment_hash(
@@ -385,9 +385,9 @@ class GpuEngine(SimulationEngine):
         if free_hook is not None:
             free = free_hook(free, ds)
 
-        # eligibility + pool routing
+        # eligibility + pool routing (kernel A writes every word)
         W = (N + 63) // 64
-        elig = torch.zeros((G, max(W, 1)), dtype=torch.int64, device=device)
+        elig = torch.empty((G, max(W, 1)), dtype=torch.int64, device=device)
         if N > 0:
             ops.fit_eligibility(
                 ds.node_labels, ds.node_label_count, ds.node_unsched,
@@ -422,8 +422,9 @@ class GpuEngine(SimulationEngine):
             0,
         ).astype(np.int32) if K else np.zeros(0, np.int32)
         maxu = max(int(max_new_h.max()) if K else 0, 1)
-        units_free = torch.zeros((max(K, 1), maxu, 4), dtype=torch.int64, device=device)
-        max_new = torch.from_numpy(max_new_h).to(device)
+        units_free = torch.empty((max(K, 1), maxu, 4), dtype=torch.int64, device=device)
+        max_new = torch.clamp(ds.pool_max - ds.pool_cur, min=0) if K else \
+            torch.zeros(0, dtype=torch.int32, device=device)
         if K:
             ops.binpack_units(
                 seg_pool, segs.seg_req.contiguous(), segs.seg_count.contiguous(), placed,

@@ -24,6 +24,7 @@
 #include "common.h"
 
 #define BP_BLOCK 1024
+#define UNITS_MAX_POOLS 64
 
 extern "C" __global__ __launch_bounds__(BP_BLOCK) void binpack_kernel(
     int32_t* __restrict__ free_,                 // [N, NRES] mutated
@@ -93,12 +94,24 @@ extern "C" __global__ __launch_bounds__(WAVE) void units_kernel(
     const int64_t* __restrict__ seg_count, const int64_t* __restrict__ placed,
     const int32_t* __restrict__ pool_unit,      // [K, NRES]
     const int32_t* __restrict__ pool_max_new,   // [K]
-    int32_t G, int32_t K, int32_t MAXU,
+    int32_t G, int32_t K, int32_t MAXU, int32_t use_lds,
     int64_t* __restrict__ units_free, int64_t* __restrict__ units_cnt,
     int64_t* __restrict__ placed_new, int64_t* __restrict__ unful,
     int64_t* __restrict__ pool_residual)
 {
+  // units_free normally lives in LDS (K*MAXU*NRES int64; the engine
+  // sets use_lds=0 when that exceeds the carve) — the kernel is a
+  // single serial wave, so global-memory unit state was pure L2
+  // round-trip latency (29 us/decide in the round-1 profile).
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  int64_t* U = use_lds ? reinterpret_cast<int64_t*>(smem) : units_free;
+  __shared__ int64_t lds_resid[UNITS_MAX_POOLS * NRES];
+  __shared__ int32_t lds_cnt[UNITS_MAX_POOLS];
+
   const int lane = threadIdx.x & (WAVE - 1);
+  for (int k = lane; k < K; k += WAVE) lds_cnt[k] = 0;
+  for (int q = lane; q < K * NRES; q += WAVE) lds_resid[q] = 0;
+  __syncthreads();
 
   for (int g = 0; g < G; ++g) {
     int64_t rem = seg_count[g] - placed[g];
@@ -116,12 +129,12 @@ extern "C" __global__ __launch_bounds__(WAVE) void units_kernel(
 #pragma unroll
     for (int r = 0; r < NRES; ++r) rq[r] = seg_req[g * NRES + r];
     if (lane < NRES) {
-      pool_residual[k * NRES + lane] += rq[lane] * rem;
+      lds_resid[k * NRES + lane] += rq[lane] * rem;
     }
     const int64_t need0 = rem;
 
     // 1) fill already-open units in creation order (scan-threshold FFD)
-    int nopen = (int)units_cnt[k];
+    const int nopen = lds_cnt[k];
     for (int u0 = 0; u0 < nopen && rem > 0; u0 += WAVE) {
       const int u = u0 + lane;
       int64_t m = 0;
@@ -131,7 +144,7 @@ extern "C" __global__ __launch_bounds__(WAVE) void units_kernel(
         bool neg = false;
 #pragma unroll
         for (int r = 0; r < NRES; ++r) {
-          uf[r] = units_free[((int64_t)k * MAXU + u) * NRES + r];
+          uf[r] = U[((int64_t)k * MAXU + u) * NRES + r];
           if (uf[r] < 0) neg = true;
           if (rq[r] > 0) {
             int64_t q = uf[r] / rq[r];
@@ -147,7 +160,7 @@ extern "C" __global__ __launch_bounds__(WAVE) void units_kernel(
       if (u < nopen && take > 0) {
 #pragma unroll
         for (int r = 0; r < NRES; ++r) {
-          units_free[((int64_t)k * MAXU + u) * NRES + r] = uf[r] - take * rq[r];
+          U[((int64_t)k * MAXU + u) * NRES + r] = uf[r] - take * rq[r];
         }
       }
       const int64_t chunk_total =
@@ -179,7 +192,7 @@ extern "C" __global__ __launch_bounds__(WAVE) void units_kernel(
           take = take < 0 ? 0 : (take > per_unit ? per_unit : take);
 #pragma unroll
           for (int r = 0; r < NRES; ++r) {
-            units_free[((int64_t)k * MAXU + nopen + ui) * NRES + r] =
+            U[((int64_t)k * MAXU + nopen + ui) * NRES + r] =
                 (int64_t)pool_unit[k * NRES + r] - take * rq[r];
           }
         }
@@ -187,13 +200,17 @@ extern "C" __global__ __launch_bounds__(WAVE) void units_kernel(
       const int64_t absorbed =
           n_new * per_unit < rem ? (int64_t)n_new * per_unit : rem;
       rem -= absorbed;
-      if (lane == 0) units_cnt[k] = (int64_t)(nopen + n_new);
+      if (lane == 0) lds_cnt[k] = nopen + n_new;
+      __syncthreads();
     }
     if (lane == 0) {
       placed_new[g] = need0 - rem;
       unful[g] = rem;
     }
   }
+  __syncthreads();
+  for (int k = lane; k < K; k += WAVE) units_cnt[k] = (int64_t)lds_cnt[k];
+  for (int q = lane; q < K * NRES; q += WAVE) pool_residual[q] = lds_resid[q];
 }
 
 #include <torch/extension.h>
@@ -222,12 +239,16 @@ void binpack_units(torch::Tensor seg_pool, torch::Tensor seg_req,
   const int K = (int)pool_unit.size(0);
   const int MAXU = (int)units_free.size(1);
   if (G == 0) return;
+  TORCH_CHECK(K <= UNITS_MAX_POOLS, "units_kernel supports at most 64 pools");
+  const size_t lds_bytes = (size_t)K * MAXU * NRES * sizeof(int64_t);
+  const int use_lds = lds_bytes <= 64 * 1024 ? 1 : 0;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  hipLaunchKernelGGL(units_kernel, dim3(1), dim3(WAVE), 0, stream.stream(),
+  hipLaunchKernelGGL(units_kernel, dim3(1), dim3(WAVE),
+                     use_lds ? lds_bytes : 0, stream.stream(),
                      seg_pool.data_ptr<int32_t>(), seg_req.data_ptr<int32_t>(),
                      seg_count.data_ptr<int64_t>(), placed.data_ptr<int64_t>(),
                      pool_unit.data_ptr<int32_t>(),
-                     pool_max_new.data_ptr<int32_t>(), G, K, MAXU,
+                     pool_max_new.data_ptr<int32_t>(), G, K, MAXU, use_lds,
                      units_free.data_ptr<int64_t>(), units_cnt.data_ptr<int64_t>(),
                      placed_new.data_ptr<int64_t>(), unful.data_ptr<int64_t>(),
                      pool_residual.data_ptr<int64_t>());
