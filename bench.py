@@ -63,14 +63,38 @@ def build_snapshot(rank: int, pods_per_rank: int):
     return encode_snapshot(pods, nodes, pools)
 
 
+# BASELINE config #5: 1M-pod burst, snapshot resident in HBM. Nodes are
+# a 4k-node cluster; pool caps raised so the decision isn't truncated.
+BURST1M_POOL_SIZES = (2000, 1000, 1000)
+
+
+def build_snapshot_1m(rank: int, pods_per_rank: int):
+    from mi355x_autoscaler.sim.synth_soa import synth_burst_snapshot
+
+    snap = synth_burst_snapshot(
+        pods_per_rank,
+        n_nodes_per_pool=BURST1M_POOL_SIZES,
+        pools=POOLS,
+        gpu_fraction=GPU_FRACTION,
+        seed=1000 + rank,
+    )
+    snap.pool_max[:] = 1 << 20  # uncapped pools: full units computation
+    return snap
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--pods-per-gpu", type=int, default=10000)
+    ap.add_argument("--pods-per-gpu", type=int, default=None)
     ap.add_argument("--backend", choices=["auto", "gpu", "cpu"], default="auto")
+    ap.add_argument("--bench-config", choices=["burst10k", "burst1m"], default="burst10k",
+                    help="burst10k = BASELINE config #3 (default); burst1m = config #5 "
+                         "(1M pods, HBM-resident snapshot, re-decided each step)")
     args = ap.parse_args()
+    if args.pods_per_gpu is None:
+        args.pods_per_gpu = 1_000_000 if args.bench_config == "burst1m" else 10_000
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -103,7 +127,24 @@ def main():
     else:
         engine = inner
 
-    snap = build_snapshot(rank, args.pods_per_gpu)
+    resident = args.bench_config == "burst1m"
+    if resident:
+        snap = build_snapshot_1m(rank, args.pods_per_gpu)
+    else:
+        snap = build_snapshot(rank, args.pods_per_gpu)
+
+    if resident and use_gpu and world == 1:
+        # config #5: the snapshot stays resident in HBM; each step is a
+        # full re-decision (segment build + fit + pack) over it.
+        ds = inner.packer.pack(snap, inner.device)
+        torch.cuda.synchronize()
+
+        def decide_once():
+            segs = inner.build_segments(ds)
+            return inner.decide_device(ds, segs, snap.pool_names)
+    else:
+        def decide_once():
+            return engine.decide(snap)
 
     def sync():
         if use_gpu:
@@ -116,14 +157,14 @@ def main():
     # warmup (untimed)
     decision = None
     for _ in range(args.warmup):
-        decision = engine.decide(snap)
+        decision = decide_once()
 
     sync()
     step_times = []
     t_all0 = time.perf_counter()
     for _ in range(args.steps):
         t0 = time.perf_counter()
-        decision = engine.decide(snap)
+        decision = decide_once()
         if use_gpu:
             torch.cuda.synchronize()
         step_times.append(time.perf_counter() - t0)
@@ -161,9 +202,13 @@ def main():
             "dtype": "int32+exact-f32-mfma",
             "data": "synthetic",
             "config": {
-                "model": "10k-pod mixed CPU/GPU burst, 3 heterogeneous agent pools (BASELINE config #3)",
+                "model": (
+                    "1M-pod burst replay, HBM-resident snapshot (BASELINE config #5)"
+                    if resident else
+                    "10k-pod mixed CPU/GPU burst, 3 heterogeneous agent pools (BASELINE config #3)"
+                ),
                 "pods_per_gpu": args.pods_per_gpu,
-                "nodes": sum(POOL_SIZES),
+                "nodes": sum(BURST1M_POOL_SIZES if resident else POOL_SIZES),
                 "pools": 3,
                 "gpu_pod_fraction": GPU_FRACTION,
                 "parallelism": f"pod-sharded controller x{world}" if world > 1 else "single",

@@ -59,6 +59,15 @@ def _pack_req_key(req: np.ndarray) -> int:
     return (cpu << 31) | (mem << 8) | gpu
 
 
+def pack_req_keys(req: np.ndarray) -> np.ndarray:
+    """Vectorized _pack_req_key over a [M, 4] int array -> uint64[M]."""
+    r = req.astype(np.int64)
+    cpu = np.clip(r[:, 0], 0, (1 << 21) - 1).astype(np.uint64)
+    mem = np.clip(r[:, 1], 0, (1 << 23) - 1).astype(np.uint64)
+    gpu = np.clip(r[:, 3], 0, (1 << 8) - 1).astype(np.uint64)
+    return (cpu << np.uint64(31)) | (mem << np.uint64(8)) | gpu
+
+
 def build_segments(
     snap: ClusterSnapshot, pod_subset: Optional[np.ndarray] = None
 ) -> List[Segment]:
@@ -77,9 +86,7 @@ def build_segments(
         return []
 
     sel = snap.pod_sel_hash[pending].astype(np.uint64)
-    keys = np.array(
-        [_pack_req_key(snap.pod_req[i]) for i in pending], dtype=np.uint64
-    )
+    keys = pack_req_keys(snap.pod_req[pending])
     # order: sel asc, key desc, pod index asc (determinism of pod lists)
     order = np.lexsort((pending, np.iinfo(np.uint64).max - keys, sel))
     pending = pending[order]

@@ -241,3 +241,18 @@ def test_segment_hash_many_segments(engine):
         ))
     snap = _snap(pods, [node])
     _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))
+
+
+def test_burst1m_decide_matches_oracle(engine):
+    """BASELINE config #5 scale: 1M pending pods, 4k nodes, decisions
+    still bit-identical to the oracle."""
+    from mi355x_autoscaler.sim.synth_soa import synth_burst_snapshot
+
+    snap = synth_burst_snapshot(
+        1_000_000, n_nodes_per_pool=(2000, 1000, 1000), gpu_fraction=0.35, seed=3,
+    )
+    snap.pool_max[:] = 1 << 20
+    d_gpu = engine.decide(snap)
+    d_cpu = CpuEngine().decide(snap)
+    _assert_decisions_equal(d_gpu, d_cpu)
+    assert d_gpu.pods_placed_existing + d_gpu.pods_placed_new + d_gpu.pods_unfulfillable == 1_000_000
