@@ -19,7 +19,7 @@ void binpack_place(torch::Tensor free, torch::Tensor elig, torch::Tensor seg_req
 void binpack_units(torch::Tensor seg_pool, torch::Tensor seg_req,
                    torch::Tensor seg_count, torch::Tensor placed,
                    torch::Tensor pool_unit, torch::Tensor pool_max_new,
-                   torch::Tensor units_free, torch::Tensor units_cnt,
+                   torch::Tensor run_scratch, torch::Tensor units_cnt,
                    torch::Tensor placed_new, torch::Tensor unful,
                    torch::Tensor pool_residual);
 void drain_flags(torch::Tensor pod_node, torch::Tensor pod_flags,

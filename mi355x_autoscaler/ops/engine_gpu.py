@@ -421,14 +421,13 @@ class GpuEngine(SimulationEngine):
             - (ds.pool_cur_h if ds.pool_cur_h is not None else ds.pool_cur.cpu().numpy()),
             0,
         ).astype(np.int32) if K else np.zeros(0, np.int32)
-        maxu = max(int(max_new_h.max()) if K else 0, 1)
-        units_free = torch.empty((max(K, 1), maxu, 4), dtype=torch.int64, device=device)
         max_new = torch.clamp(ds.pool_max - ds.pool_cur, min=0) if K else \
             torch.zeros(0, dtype=torch.int32, device=device)
+        run_scratch = torch.empty((4 * G + 8) * 6, dtype=torch.int64, device=device)
         if K:
             ops.binpack_units(
                 seg_pool, segs.seg_req.contiguous(), segs.seg_count.contiguous(), placed,
-                ds.pool_unit, max_new, units_free, units_cnt,
+                ds.pool_unit, max_new, run_scratch, units_cnt,
                 placed_new, unful, pool_residual,
             )
 

@@ -89,128 +89,189 @@ extern "C" __global__ __launch_bounds__(BP_BLOCK) void binpack_kernel(
 //   units_free: [K, MAXU, NRES] int64 scratch (open virtual nodes)
 //   units_cnt:  [K] int32 (doubles as units_needed output)
 //   pool_residual: [K, NRES] int64 residual demand totals
+// Run-length units phase. Units opened in one batch share identical
+// residual vectors, and one placement step splits at most ONE run
+// (the partially-consumed one) into <=3 runs, so exact per-unit FFD
+// over an UNBOUNDED unit count is a walk over <=4G+K runs per pool —
+// O(G^2) total. (The per-unit dense formulation measured 15.6 ms/step
+// at 1M pods / 327k units; this is the algorithmic fix.) Semantics are
+// bit-identical to the oracle's unit-at-a-time loop: a unit takes
+// min(capacity, remaining), units in creation order.
+//
+// Run arena: res[RTOT][NRES] i64 | cnt[RTOT] i64 | next[RTOT] i64,
+// linked per pool (head/tail), carved from LDS when it fits else from
+// the global scratch tensor. The walk is lane-0 scalar — runs are few.
 extern "C" __global__ __launch_bounds__(WAVE) void units_kernel(
     const int32_t* __restrict__ seg_pool, const int32_t* __restrict__ seg_req,
     const int64_t* __restrict__ seg_count, const int64_t* __restrict__ placed,
     const int32_t* __restrict__ pool_unit,      // [K, NRES]
     const int32_t* __restrict__ pool_max_new,   // [K]
-    int32_t G, int32_t K, int32_t MAXU, int32_t use_lds,
-    int64_t* __restrict__ units_free, int64_t* __restrict__ units_cnt,
+    int32_t G, int32_t K, int32_t RTOT, int32_t use_lds,
+    int64_t* __restrict__ scratch,              // [RTOT * 6] when !use_lds
+    int64_t* __restrict__ units_cnt,            // [K] out
     int64_t* __restrict__ placed_new, int64_t* __restrict__ unful,
-    int64_t* __restrict__ pool_residual)
+    int64_t* __restrict__ pool_residual)        // [K, NRES] out
 {
-  // units_free normally lives in LDS (K*MAXU*NRES int64; the engine
-  // sets use_lds=0 when that exceeds the carve) — the kernel is a
-  // single serial wave, so global-memory unit state was pure L2
-  // round-trip latency (29 us/decide in the round-1 profile).
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  int64_t* U = use_lds ? reinterpret_cast<int64_t*>(smem) : units_free;
-  __shared__ int64_t lds_resid[UNITS_MAX_POOLS * NRES];
-  __shared__ int32_t lds_cnt[UNITS_MAX_POOLS];
+  int64_t* arena = use_lds ? reinterpret_cast<int64_t*>(smem) : scratch;
+  int64_t* run_res = arena;               // [RTOT][NRES]
+  int64_t* run_cnt = arena + (int64_t)RTOT * NRES;
+  int64_t* run_next = run_cnt + RTOT;
+
+  __shared__ int32_t head[UNITS_MAX_POOLS];
+  __shared__ int32_t tail[UNITS_MAX_POOLS];
+  __shared__ int64_t opened[UNITS_MAX_POOLS];
+  __shared__ int64_t resid[UNITS_MAX_POOLS * NRES];
 
   const int lane = threadIdx.x & (WAVE - 1);
-  for (int k = lane; k < K; k += WAVE) lds_cnt[k] = 0;
-  for (int q = lane; q < K * NRES; q += WAVE) lds_resid[q] = 0;
+  for (int k = lane; k < K; k += WAVE) {
+    head[k] = -1;
+    tail[k] = -1;
+    opened[k] = 0;
+  }
+  for (int q = lane; q < K * NRES; q += WAVE) resid[q] = 0;
   __syncthreads();
 
-  for (int g = 0; g < G; ++g) {
-    int64_t rem = seg_count[g] - placed[g];
-    if (lane == 0) {
+  if (lane == 0) {
+    int alloc = 0;
+    for (int g = 0; g < G; ++g) {
+      int64_t rem = seg_count[g] - placed[g];
       placed_new[g] = 0;
       unful[g] = 0;
-    }
-    if (rem <= 0) continue;
-    const int k = seg_pool[g];
-    if (k < 0) {
-      if (lane == 0) unful[g] = rem;
-      continue;
-    }
-    int64_t rq[NRES];
+      if (rem <= 0) continue;
+      const int k = seg_pool[g];
+      if (k < 0) {
+        unful[g] = rem;
+        continue;
+      }
+      if (alloc + 4 > RTOT) {  // arena bound: a segment allocates <=4 runs
+        unful[g] = rem;        // (unreachable when RTOT >= 4G+8; kept so a
+        continue;              //  bound bug can never corrupt the lists)
+      }
+      int64_t rq[NRES];
 #pragma unroll
-    for (int r = 0; r < NRES; ++r) rq[r] = seg_req[g * NRES + r];
-    if (lane < NRES) {
-      lds_resid[k * NRES + lane] += rq[lane] * rem;
-    }
-    const int64_t need0 = rem;
+      for (int r = 0; r < NRES; ++r) rq[r] = seg_req[g * NRES + r];
+#pragma unroll
+      for (int r = 0; r < NRES; ++r) resid[k * NRES + r] += rq[r] * rem;
+      const int64_t need0 = rem;
 
-    // 1) fill already-open units in creation order (scan-threshold FFD)
-    const int nopen = lds_cnt[k];
-    for (int u0 = 0; u0 < nopen && rem > 0; u0 += WAVE) {
-      const int u = u0 + lane;
-      int64_t m = 0;
-      int64_t uf[NRES];
-      if (u < nopen) {
-        int64_t mm = (int64_t)1 << 31;
+      // 1) walk open runs in creation order
+      for (int idx = head[k]; idx != -1 && rem > 0; idx = (int)run_next[idx]) {
+        int64_t m = (int64_t)1 << 31;
         bool neg = false;
 #pragma unroll
         for (int r = 0; r < NRES; ++r) {
-          uf[r] = U[((int64_t)k * MAXU + u) * NRES + r];
-          if (uf[r] < 0) neg = true;
+          const int64_t fr = run_res[(int64_t)idx * NRES + r];
+          if (fr < 0) neg = true;
           if (rq[r] > 0) {
-            int64_t q = uf[r] / rq[r];
-            mm = q < mm ? q : mm;
+            const int64_t q = fr / rq[r];
+            m = q < m ? q : m;
           }
         }
-        m = neg ? 0 : mm;
-      }
-      int64_t inc = wave_inclusive_scan_i64(m);
-      const int64_t ex = inc - m;
-      int64_t take = rem - ex;
-      take = take < 0 ? 0 : (take > m ? m : take);
-      if (u < nopen && take > 0) {
+        if (neg || m <= 0) continue;
+        const int64_t c = run_cnt[idx];
+        if (rem >= m * c) {  // every unit of the run takes m
 #pragma unroll
-        for (int r = 0; r < NRES; ++r) {
-          U[((int64_t)k * MAXU + u) * NRES + r] = uf[r] - take * rq[r];
+          for (int r = 0; r < NRES; ++r) run_res[(int64_t)idx * NRES + r] -= m * rq[r];
+          rem -= m * c;
+          continue;
         }
+        // split: nf full takers, maybe one partial, rest untouched
+        const int64_t nf = rem / m;
+        const int64_t part = rem - nf * m;
+        int64_t untouched = c - nf - (part > 0 ? 1 : 0);
+        // node idx becomes the full-takers run (or partial if nf==0)
+        const int64_t nxt = run_next[idx];
+        int cur = idx;
+        if (nf > 0) {
+          run_cnt[cur] = nf;
+#pragma unroll
+          for (int r = 0; r < NRES; ++r) run_res[(int64_t)cur * NRES + r] -= m * rq[r];
+        }
+        if (part > 0) {
+          if (nf > 0) {
+            const int nn = alloc++;
+            run_next[cur] = nn;
+            run_next[nn] = nxt;
+#pragma unroll
+            for (int r = 0; r < NRES; ++r)
+              run_res[(int64_t)nn * NRES + r] =
+                  run_res[(int64_t)idx * NRES + r] + m * rq[r] - part * rq[r];
+            run_cnt[nn] = 1;
+            if (tail[k] == cur) tail[k] = nn;
+            cur = nn;
+          } else {
+            run_cnt[cur] = 1;
+#pragma unroll
+            for (int r = 0; r < NRES; ++r) run_res[(int64_t)cur * NRES + r] -= part * rq[r];
+          }
+        }
+        if (untouched > 0) {
+          const int nn = alloc++;
+          run_next[cur] = nn;
+          run_next[nn] = nxt;
+#pragma unroll
+          for (int r = 0; r < NRES; ++r)
+            run_res[(int64_t)nn * NRES + r] =
+                run_res[(int64_t)idx * NRES + r] +
+                ((cur == idx && part > 0) ? part : m) * rq[r];
+          run_cnt[nn] = untouched;
+          if (tail[k] == cur) tail[k] = nn;
+        }
+        rem = 0;
+        break;
       }
-      const int64_t chunk_total =
-          __shfl((long long)inc, WAVE - 1, WAVE);
-      rem -= chunk_total < rem ? chunk_total : rem;
-    }
 
-    // 2) open new units while the pool cap allows
-    int64_t per_unit = (int64_t)1 << 31;
-    bool fits = true;
+      // 2) open new units while the pool cap allows
+      int64_t per_unit = (int64_t)1 << 31;
+      bool fits = true;
 #pragma unroll
-    for (int r = 0; r < NRES; ++r) {
-      const int64_t cap = pool_unit[k * NRES + r];
-      if (rq[r] > cap) fits = false;
-      if (rq[r] > 0) {
-        int64_t q = cap / rq[r];
-        per_unit = q < per_unit ? q : per_unit;
-      }
-    }
-    if (rem > 0 && fits && per_unit > 0) {
-      const int room = pool_max_new[k] - nopen;
-      int64_t want = (rem + per_unit - 1) / per_unit;
-      const int n_new = (int)(want < room ? want : (room < 0 ? 0 : room));
-      for (int u0 = 0; u0 < n_new; u0 += WAVE) {
-        const int ui = u0 + lane;
-        if (ui < n_new) {
-          const int64_t ex = (int64_t)ui * per_unit;
-          int64_t take = rem - ex;
-          take = take < 0 ? 0 : (take > per_unit ? per_unit : take);
-#pragma unroll
-          for (int r = 0; r < NRES; ++r) {
-            U[((int64_t)k * MAXU + nopen + ui) * NRES + r] =
-                (int64_t)pool_unit[k * NRES + r] - take * rq[r];
-          }
+      for (int r = 0; r < NRES; ++r) {
+        const int64_t cap = pool_unit[k * NRES + r];
+        if (rq[r] > cap) fits = false;
+        if (rq[r] > 0) {
+          const int64_t q = cap / rq[r];
+          per_unit = q < per_unit ? q : per_unit;
         }
       }
-      const int64_t absorbed =
-          n_new * per_unit < rem ? (int64_t)n_new * per_unit : rem;
-      rem -= absorbed;
-      if (lane == 0) lds_cnt[k] = nopen + n_new;
-      __syncthreads();
-    }
-    if (lane == 0) {
+      if (rem > 0 && fits && per_unit > 0) {
+        int64_t room = (int64_t)pool_max_new[k] - opened[k];
+        if (room < 0) room = 0;
+        int64_t nf = rem / per_unit;
+        nf = nf < room ? nf : room;
+        if (nf > 0) {
+          const int nn = alloc++;
+#pragma unroll
+          for (int r = 0; r < NRES; ++r)
+            run_res[(int64_t)nn * NRES + r] = (int64_t)pool_unit[k * NRES + r] - per_unit * rq[r];
+          run_cnt[nn] = nf;
+          run_next[nn] = -1;
+          if (tail[k] == -1) head[k] = nn; else run_next[tail[k]] = nn;
+          tail[k] = nn;
+          opened[k] += nf;
+          rem -= nf * per_unit;
+          room -= nf;
+        }
+        if (rem > 0 && room > 0) {  // one partial unit
+          const int nn = alloc++;
+#pragma unroll
+          for (int r = 0; r < NRES; ++r)
+            run_res[(int64_t)nn * NRES + r] = (int64_t)pool_unit[k * NRES + r] - rem * rq[r];
+          run_cnt[nn] = 1;
+          run_next[nn] = -1;
+          if (tail[k] == -1) head[k] = nn; else run_next[tail[k]] = nn;
+          tail[k] = nn;
+          opened[k] += 1;
+          rem = 0;
+        }
+      }
       placed_new[g] = need0 - rem;
       unful[g] = rem;
     }
   }
   __syncthreads();
-  for (int k = lane; k < K; k += WAVE) units_cnt[k] = (int64_t)lds_cnt[k];
-  for (int q = lane; q < K * NRES; q += WAVE) pool_residual[q] = lds_resid[q];
+  for (int k = lane; k < K; k += WAVE) units_cnt[k] = opened[k];
+  for (int q = lane; q < K * NRES; q += WAVE) pool_residual[q] = resid[q];
 }
 
 #include <torch/extension.h>
@@ -232,24 +293,25 @@ void binpack_place(torch::Tensor free, torch::Tensor elig, torch::Tensor seg_req
 void binpack_units(torch::Tensor seg_pool, torch::Tensor seg_req,
                    torch::Tensor seg_count, torch::Tensor placed,
                    torch::Tensor pool_unit, torch::Tensor pool_max_new,
-                   torch::Tensor units_free, torch::Tensor units_cnt,
+                   torch::Tensor run_scratch, torch::Tensor units_cnt,
                    torch::Tensor placed_new, torch::Tensor unful,
                    torch::Tensor pool_residual) {
   const int G = (int)seg_req.size(0);
   const int K = (int)pool_unit.size(0);
-  const int MAXU = (int)units_free.size(1);
   if (G == 0) return;
   TORCH_CHECK(K <= UNITS_MAX_POOLS, "units_kernel supports at most 64 pools");
-  const size_t lds_bytes = (size_t)K * MAXU * NRES * sizeof(int64_t);
-  const int use_lds = lds_bytes <= 64 * 1024 ? 1 : 0;
+  const int RTOT = (int)(run_scratch.numel() / 6);
+  TORCH_CHECK(RTOT >= 4 * G + 8, "run arena must hold 4G+8 runs");
+  const size_t lds_bytes = (size_t)RTOT * 6 * sizeof(int64_t);
+  const int use_lds = lds_bytes <= 96 * 1024 ? 1 : 0;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   hipLaunchKernelGGL(units_kernel, dim3(1), dim3(WAVE),
                      use_lds ? lds_bytes : 0, stream.stream(),
                      seg_pool.data_ptr<int32_t>(), seg_req.data_ptr<int32_t>(),
                      seg_count.data_ptr<int64_t>(), placed.data_ptr<int64_t>(),
                      pool_unit.data_ptr<int32_t>(),
-                     pool_max_new.data_ptr<int32_t>(), G, K, MAXU, use_lds,
-                     units_free.data_ptr<int64_t>(), units_cnt.data_ptr<int64_t>(),
+                     pool_max_new.data_ptr<int32_t>(), G, K, RTOT, use_lds,
+                     run_scratch.data_ptr<int64_t>(), units_cnt.data_ptr<int64_t>(),
                      placed_new.data_ptr<int64_t>(), unful.data_ptr<int64_t>(),
                      pool_residual.data_ptr<int64_t>());
 }
