@@ -38,6 +38,13 @@ from ..utils.notification import Notifier
 
 logger = logging.getLogger(__name__)
 
+
+def _codec_available() -> bool:
+    from ..snapshot.codec import codec_available
+
+    return codec_available()
+
+
 # node states [R: cluster.py — reconstructed enum strings are low
 # confidence; semantics preserved]
 STATE_BUSY = "busy"
@@ -67,6 +74,9 @@ class ClusterOptions:
     dry_run: bool = False
     drain_grace_seconds: int = 30
     max_drains_per_cycle: int = 4
+    # ingest pods via the C++ snapshot codec (JSON bytes -> SoA) when
+    # the extension is built; the Python path is the fallback/spec.
+    native_codec: bool = True
 
 
 @dataclass
@@ -124,21 +134,31 @@ class Cluster:
         now = time.time() if now is None else now
         result = CycleResult()
 
+        use_codec = self.opts.native_codec and _codec_available()
         with t.phase("ingest"):
-            pods_json = self.kube.list_pods()
+            if use_codec:
+                pods_body = self.kube.list_pods_raw()
+                pods_json = None
+            else:
+                pods_json = self.kube.list_pods()
             nodes_json = self.kube.list_nodes()
             pdbs_json = self.kube.list_pdbs()
-        pods = [KubePod(p) for p in pods_json]
         nodes = [KubeNode(n) for n in nodes_json]
         pdbs = [PodDisruptionBudget.from_json(d) for d in pdbs_json]
         pools = group_nodes_into_pools(
             nodes, ignore_pools=self.opts.ignore_pools, max_size=self.opts.max_pool_size
         )
-        result.pods_total = len(pods)
         result.nodes_total = len(nodes)
 
         with t.phase("encode"):
-            snap = encode_snapshot(pods, nodes, pools, pdbs)
+            if use_codec:
+                from ..snapshot.codec import encode_snapshot_fast
+
+                snap = encode_snapshot_fast(pods_body, nodes, pools, pdbs)
+            else:
+                pods = [KubePod(p) for p in pods_json]
+                snap = encode_snapshot(pods, nodes, pools, pdbs)
+        result.pods_total = snap.num_pods
         result.snapshot = snap
 
         # ---- scale-up --------------------------------------------------

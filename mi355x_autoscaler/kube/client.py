@@ -99,6 +99,11 @@ class KubeApiClient:
     def list_pods(self) -> List[Dict]:
         return self._list_all("/api/v1/pods")
 
+    def list_pods_raw(self) -> bytes:
+        """Raw PodList body for the native snapshot codec (single page;
+        the codec parses it without building Python dicts)."""
+        return self._req("GET", "/api/v1/pods").content
+
     def list_nodes(self) -> List[Dict]:
         return self._list_all("/api/v1/nodes")
 

@@ -49,6 +49,8 @@ void segment_compact(torch::Tensor slot_ks, torch::Tensor slot_kr,
                      torch::Tensor slot2seg, torch::Tensor g_out);
 void pod_pool_map(torch::Tensor slot_of_pod, torch::Tensor slot2seg,
                   torch::Tensor seg_pool, torch::Tensor pod_pool);
+pybind11::dict encode_pods_json(pybind11::bytes body, pybind11::dict node_index,
+                                pybind11::list pdb_selectors);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native autoscaler scheduling kernels (gfx950 HIP)";
@@ -64,4 +66,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("segment_hash", &segment_hash, "kernel C: hash-aggregate pending pods");
   m.def("segment_compact", &segment_compact, "kernel C: compact+sort segments");
   m.def("pod_pool_map", &pod_pool_map, "pod -> routed pool map");
+  m.def("encode_pods_json", &encode_pods_json,
+        "PodList JSON -> pod SoA columns (native snapshot codec)");
 }

@@ -1,0 +1,642 @@
+// High-throughput PodList JSON -> SoA snapshot encoder (C++/pybind11).
+//
+// Re-creation of the reference's ingest path (pykube JSON -> KubePod
+// wrapping [R: autoscaler/kube.py — reconstructed]) as a native codec
+// [SURVEY §2.2 row F]: a single pass over the API server's PodList
+// body materializes a tiny string_view DOM per item and emits the pod
+// columns of snapshot/encode.py's integer lattice. Every rule —
+// quantity parsing, MiB round-up, FNV-1a pair hashing, flag bits,
+// owner-kind sets — mirrors the Python encoder exactly; the
+// differential test (tests/test_codec.py) asserts byte-equality on
+// every column.
+//
+// The Python path is the semantic spec; this is the fast path the
+// control loop uses when the extension is built (ctl/loop.py).
+#include <torch/extension.h>
+
+#include <pybind11/numpy.h>
+
+#include <algorithm>
+#include <cctype>
+#include <cmath>
+#include <cstdint>
+#include <cstring>
+#include <deque>
+#include <memory>
+#include <stdexcept>
+#include <string>
+#include <string_view>
+#include <unordered_map>
+#include <vector>
+
+namespace {
+
+// ---- lattice constants (must match snapshot/encode.py) --------------------
+constexpr int NRES = 4;
+constexpr int MAXSEL = 8;
+constexpr int64_t MIB = 1024 * 1024;
+
+constexpr uint16_t PF_PENDING = 1 << 0;
+constexpr uint16_t PF_MIRROR = 1 << 1;
+constexpr uint16_t PF_DAEMONSET = 1 << 2;
+constexpr uint16_t PF_CRITICAL = 1 << 3;
+constexpr uint16_t PF_REPLICATED = 1 << 4;
+constexpr uint16_t PF_DRAINABLE = 1 << 5;
+constexpr uint16_t PF_BLOCKS_DRAIN = 1 << 6;
+constexpr uint16_t PF_SEL_OVERFLOW = 1 << 7;
+constexpr uint16_t PF_BOUND = 1 << 8;
+
+// FNV-1a over key + 0x1f + value, identical to hashing the
+// concatenated string (snapshot/encode.py::hash_label_pair).
+uint64_t hash_label_pair(std::string_view k, std::string_view v) {
+  uint64_t h = 0xCBF29CE484222325ull;
+  for (unsigned char c : k) {
+    h ^= c;
+    h *= 0x100000001B3ull;
+  }
+  h ^= 0x1f;
+  h *= 0x100000001B3ull;
+  for (unsigned char c : v) {
+    h ^= c;
+    h *= 0x100000001B3ull;
+  }
+  return h;
+}
+
+// ---- minimal JSON DOM (zero-copy string_views, pooled nodes) ------------
+struct JValue;
+using JMembers = std::vector<std::pair<std::string_view, JValue*>>;
+using JItems = std::vector<JValue*>;
+
+struct JValue {
+  enum Kind { Null, Bool, Num, Str, Obj, Arr } kind = Null;
+  bool b = false;
+  double num = 0;
+  std::string_view str;  // view into the body (or the decode side-buffer)
+  JMembers members;      // Obj
+  JItems items;          // Arr
+};
+
+class Parser {
+ public:
+  Parser(const char* p, size_t n) : p_(p), begin_(p), end_(p + n) {}
+  ~Parser() {
+    for (JValue* v : pool_) delete v;
+  }
+
+  JValue* parse_value() {
+    skip_ws();
+    if (p_ >= end_) fail("unexpected end");
+    switch (*p_) {
+      case '{':
+        return parse_object();
+      case '[':
+        return parse_array();
+      case '"': {
+        JValue* v = alloc();
+        v->kind = JValue::Str;
+        v->str = parse_string();
+        return v;
+      }
+      case 't':
+        expect("true");
+        return make_bool(true);
+      case 'f':
+        expect("false");
+        return make_bool(false);
+      case 'n':
+        expect("null");
+        return alloc();
+      default:
+        return parse_number();
+    }
+  }
+
+  void skip_ws() {
+    while (p_ < end_ && (*p_ == ' ' || *p_ == '\t' || *p_ == '\n' || *p_ == '\r')) ++p_;
+  }
+
+  bool at(char c) {
+    skip_ws();
+    return p_ < end_ && *p_ == c;
+  }
+
+  void consume(char c) {
+    skip_ws();
+    if (p_ >= end_ || *p_ != c) fail(std::string("expected '") + c + "'");
+    ++p_;
+  }
+
+  void expect(const char* lit) {
+    const size_t n = strlen(lit);
+    if ((size_t)(end_ - p_) < n || strncmp(p_, lit, n) != 0) fail("bad literal");
+    p_ += n;
+  }
+
+  std::string_view parse_string() {
+    consume('"');
+    const char* start = p_;
+    // fast path: no escapes -> zero-copy view into the body
+    while (p_ < end_ && *p_ != '"' && *p_ != '\\') ++p_;
+    if (p_ < end_ && *p_ == '"') {
+      std::string_view sv(start, (size_t)(p_ - start));
+      ++p_;
+      return sv;
+    }
+    // slow path: decode escapes into the (address-stable) side buffer
+    decoded_.emplace_back(start, (size_t)(p_ - start));
+    std::string& out = decoded_.back();
+    while (p_ < end_) {
+      char c = *p_++;
+      if (c == '"') return std::string_view(out);
+      if (c == '\\') {
+        if (p_ >= end_) break;
+        char e = *p_++;
+        switch (e) {
+          case '"': out += '"'; break;
+          case '\\': out += '\\'; break;
+          case '/': out += '/'; break;
+          case 'b': out += '\b'; break;
+          case 'f': out += '\f'; break;
+          case 'n': out += '\n'; break;
+          case 'r': out += '\r'; break;
+          case 't': out += '\t'; break;
+          case 'u': {
+            unsigned cp = parse_hex4();
+            if (cp >= 0xD800 && cp <= 0xDBFF && end_ - p_ >= 6 && p_[0] == '\\' &&
+                p_[1] == 'u') {
+              p_ += 2;
+              unsigned lo = parse_hex4();
+              cp = 0x10000 + ((cp - 0xD800) << 10) + (lo - 0xDC00);
+            }
+            if (cp < 0x80) out += (char)cp;
+            else if (cp < 0x800) {
+              out += (char)(0xC0 | (cp >> 6));
+              out += (char)(0x80 | (cp & 0x3F));
+            } else if (cp < 0x10000) {
+              out += (char)(0xE0 | (cp >> 12));
+              out += (char)(0x80 | ((cp >> 6) & 0x3F));
+              out += (char)(0x80 | (cp & 0x3F));
+            } else {
+              out += (char)(0xF0 | (cp >> 18));
+              out += (char)(0x80 | ((cp >> 12) & 0x3F));
+              out += (char)(0x80 | ((cp >> 6) & 0x3F));
+              out += (char)(0x80 | (cp & 0x3F));
+            }
+            break;
+          }
+          default:
+            fail("bad escape");
+        }
+      } else {
+        out += c;
+      }
+    }
+    fail("unterminated string");
+    return {};
+  }
+
+  unsigned parse_hex4() {
+    if (end_ - p_ < 4) fail("bad \\u escape");
+    unsigned cp = 0;
+    for (int i = 0; i < 4; ++i) {
+      char h = *p_++;
+      cp <<= 4;
+      if (h >= '0' && h <= '9') cp |= h - '0';
+      else if (h >= 'a' && h <= 'f') cp |= h - 'a' + 10;
+      else if (h >= 'A' && h <= 'F') cp |= h - 'A' + 10;
+      else fail("bad hex");
+    }
+    return cp;
+  }
+
+  JValue* parse_object() {
+    consume('{');
+    JValue* v = alloc();
+    v->kind = JValue::Obj;
+    skip_ws();
+    if (at('}')) {
+      ++p_;
+      return v;
+    }
+    while (true) {
+      skip_ws();
+      std::string_view key = parse_string();
+      consume(':');
+      JValue* val = parse_value();
+      v->members.emplace_back(key, val);
+      skip_ws();
+      if (at(',')) {
+        ++p_;
+        continue;
+      }
+      consume('}');
+      return v;
+    }
+  }
+
+  JValue* parse_array() {
+    consume('[');
+    JValue* v = alloc();
+    v->kind = JValue::Arr;
+    skip_ws();
+    if (at(']')) {
+      ++p_;
+      return v;
+    }
+    while (true) {
+      v->items.push_back(parse_value());
+      skip_ws();
+      if (at(',')) {
+        ++p_;
+        continue;
+      }
+      consume(']');
+      return v;
+    }
+  }
+
+  JValue* parse_number() {
+    char buf[64];
+    int n = 0;
+    while (p_ < end_ && n < 63 &&
+           (*p_ == '-' || *p_ == '+' || *p_ == '.' || *p_ == 'e' || *p_ == 'E' ||
+            (*p_ >= '0' && *p_ <= '9'))) {
+      buf[n++] = *p_++;
+    }
+    buf[n] = 0;
+    JValue* v = alloc();
+    v->kind = JValue::Num;
+    v->num = strtod(buf, nullptr);
+    return v;
+  }
+
+  // node pool: reused across items (reset_arena keeps capacity)
+  JValue* alloc() {
+    if (used_ == pool_.size()) pool_.push_back(new JValue);
+    JValue* v = pool_[used_++];
+    v->kind = JValue::Null;
+    v->b = false;
+    v->num = 0;
+    v->str = {};
+    v->members.clear();
+    v->items.clear();
+    return v;
+  }
+  JValue* make_bool(bool b) {
+    JValue* v = alloc();
+    v->kind = JValue::Bool;
+    v->b = b;
+    return v;
+  }
+  void reset_arena() {
+    used_ = 0;
+    decoded_.clear();
+  }
+
+  [[noreturn]] void fail(const std::string& msg) {
+    throw std::runtime_error("json parse error at offset " +
+                             std::to_string(p_ - begin_) + ": " + msg);
+  }
+
+  const char* p_;
+  const char* begin_;
+  const char* end_;
+  std::vector<JValue*> pool_;
+  size_t used_ = 0;
+  std::deque<std::string> decoded_;
+};
+
+const JValue* get(const JValue* obj, std::string_view key) {
+  if (!obj || obj->kind != JValue::Obj) return nullptr;
+  for (auto& m : obj->members)
+    if (m.first == key) return m.second;
+  return nullptr;
+}
+
+std::string_view get_str(const JValue* obj, std::string_view key) {
+  const JValue* v = get(obj, key);
+  return v && v->kind == JValue::Str ? v->str : std::string_view();
+}
+
+// ---- quantity parsing (mirrors models/resources.py) -----------------------
+double parse_quantity(const JValue* v) {
+  if (!v) return 0.0;
+  if (v->kind == JValue::Num) return v->num;
+  if (v->kind != JValue::Str) return 0.0;
+  std::string_view s = v->str;
+  if (s.empty()) return 0.0;
+  auto to_num = [](std::string_view sv) {
+    char buf[64];
+    const size_t n = std::min(sv.size(), sizeof(buf) - 1);
+    memcpy(buf, sv.data(), n);
+    buf[n] = 0;
+    return strtod(buf, nullptr);
+  };
+  static const std::pair<const char*, double> kBinary[] = {
+      {"Ki", 1024.0}, {"Mi", 1048576.0}, {"Gi", 1073741824.0},
+      {"Ti", 1099511627776.0}, {"Pi", 1125899906842624.0},
+      {"Ei", 1152921504606846976.0},
+  };
+  for (auto& [suf, mult] : kBinary) {
+    size_t sl = strlen(suf);
+    if (s.size() > sl && s.compare(s.size() - sl, sl, suf) == 0) {
+      return to_num(s.substr(0, s.size() - sl)) * mult;
+    }
+  }
+  char last = s.back();
+  double mult = 0;
+  switch (last) {
+    case 'n': mult = 1e-9; break;
+    case 'u': mult = 1e-6; break;
+    case 'm': mult = 1e-3; break;
+    case 'k': mult = 1e3; break;
+    case 'M': mult = 1e6; break;
+    case 'G': mult = 1e9; break;
+    case 'T': mult = 1e12; break;
+    case 'P': mult = 1e15; break;
+    case 'E': mult = 1e18; break;
+    default:
+      return to_num(s);
+  }
+  return to_num(s.substr(0, s.size() - 1)) * mult;
+}
+
+int64_t parse_cpu_milli(const JValue* v) {
+  double f = parse_quantity(v) * 1000.0;
+  return f > 0 ? (int64_t)(f + 0.5) : 0;
+}
+
+int64_t parse_mem_bytes(const JValue* v) {
+  double f = parse_quantity(v);
+  return f > 0 ? (int64_t)(f + 0.5) : 0;
+}
+
+bool is_gpu_key(std::string_view k) {
+  return k == "amd.com/gpu" || k == "alpha.kubernetes.io/nvidia-gpu" ||
+         k == "nvidia.com/gpu";
+}
+
+struct Req {
+  int64_t cpu = 0, mem = 0, pods = 0, gpu = 0;
+};
+
+Req sum_requests(const JValue* spec) {
+  Req total;
+  total.pods = 1;
+  const JValue* containers = get(spec, "containers");
+  if (containers && containers->kind == JValue::Arr) {
+    for (const JValue* c : containers->items) {
+      const JValue* req = get(get(c, "resources"), "requests");
+      if (!req || req->kind != JValue::Obj) continue;
+      for (auto& m : req->members) {
+        if (m.first == "cpu") total.cpu += parse_cpu_milli(m.second);
+        else if (m.first == "memory") total.mem += parse_mem_bytes(m.second);
+        else if (is_gpu_key(m.first)) total.gpu += (int64_t)parse_quantity(m.second);
+      }
+    }
+  }
+  const JValue* inits = get(spec, "initContainers");
+  if (inits && inits->kind == JValue::Arr) {
+    for (const JValue* c : inits->items) {
+      const JValue* req = get(get(c, "resources"), "requests");
+      if (!req || req->kind != JValue::Obj) continue;
+      Req iv;
+      for (auto& m : req->members) {
+        if (m.first == "cpu") iv.cpu = parse_cpu_milli(m.second);
+        else if (m.first == "memory") iv.mem = parse_mem_bytes(m.second);
+        else if (is_gpu_key(m.first)) iv.gpu += (int64_t)parse_quantity(m.second);
+      }
+      total.cpu = std::max(total.cpu, iv.cpu);
+      total.mem = std::max(total.mem, iv.mem);
+      total.gpu = std::max(total.gpu, iv.gpu);
+    }
+  }
+  return total;
+}
+
+bool kind_is_replicated(std::string_view k) {
+  return k == "ReplicationController" || k == "ReplicaSet" || k == "StatefulSet" ||
+         k == "Job" || k == "DeploymentConfig" || k == "DaemonSet";
+}
+
+bool ieq_true(std::string_view s) {
+  std::string t;
+  for (char c : s) t += (char)tolower((unsigned char)c);
+  return t == "true" || t == "'true'";
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// encode_pods(json_bytes, node_index: dict[str,int],
+//             pdb_selectors: list[dict[str,str]]) -> dict of arrays
+// ---------------------------------------------------------------------------
+pybind11::dict encode_pods_json(pybind11::bytes body, pybind11::dict node_index,
+                                pybind11::list pdb_selectors) {
+  namespace py = pybind11;
+  std::string_view sv = body;  // pybind keeps the bytes alive for the call
+
+  // host-side lookup tables
+  std::unordered_map<std::string, int32_t> node_idx;
+  for (auto item : node_index)
+    node_idx[py::cast<std::string>(item.first)] = py::cast<int32_t>(item.second);
+  std::vector<std::vector<std::pair<std::string, std::string>>> pdbs;
+  for (auto sel : pdb_selectors) {
+    std::vector<std::pair<std::string, std::string>> pairs;
+    for (auto kv : py::cast<py::dict>(sel))
+      pairs.emplace_back(py::cast<std::string>(kv.first), py::cast<std::string>(kv.second));
+    pdbs.push_back(std::move(pairs));
+  }
+
+  // column builders
+  std::vector<int32_t> pod_req;
+  std::vector<uint16_t> pod_flags;
+  std::vector<int64_t> pod_sel_hash;
+  std::vector<int64_t> pod_sel_pairs;
+  std::vector<int8_t> pod_sel_count;
+  std::vector<int32_t> pod_node;
+  std::vector<int32_t> pod_pdb;
+  py::list uids, refs;
+
+  Parser parser(sv.data(), sv.size());
+  // walk to items:[ without materializing the outer object
+  parser.skip_ws();
+  parser.consume('{');
+  bool found_items = false;
+  while (true) {
+    parser.skip_ws();
+    std::string_view key = parser.parse_string();
+    parser.consume(':');
+    if (key == "items") {
+      found_items = true;
+      break;
+    }
+    parser.parse_value();  // skip
+    parser.reset_arena();
+    parser.skip_ws();
+    if (parser.at(',')) {
+      ++parser.p_;
+      continue;
+    }
+    break;
+  }
+  if (!found_items) {
+    throw std::runtime_error("PodList has no items array");
+  }
+  parser.consume('[');
+  parser.skip_ws();
+  bool empty = parser.at(']');
+  if (empty) ++parser.p_;
+
+  while (!empty) {
+    const JValue* pod = parser.parse_value();
+    const JValue* meta = get(pod, "metadata");
+    const JValue* spec = get(pod, "spec");
+    const JValue* status = get(pod, "status");
+
+    // requests (quantize: cpu milli, memory ceil-MiB)
+    Req rq = sum_requests(spec);
+    pod_req.push_back((int32_t)std::min<int64_t>(rq.cpu, INT32_MAX));
+    pod_req.push_back((int32_t)std::min<int64_t>((rq.mem + MIB - 1) / MIB, INT32_MAX));
+    pod_req.push_back((int32_t)rq.pods);
+    pod_req.push_back((int32_t)std::min<int64_t>(rq.gpu, INT32_MAX));
+
+    // selectors
+    const JValue* nsel = get(spec, "nodeSelector");
+    std::vector<uint64_t> hashes;
+    if (nsel && nsel->kind == JValue::Obj) {
+      for (auto& m : nsel->members) {
+        if (m.second->kind == JValue::Str)
+          hashes.push_back(hash_label_pair(m.first, m.second->str));
+      }
+    }
+    std::sort(hashes.begin(), hashes.end());
+    uint64_t combined = 0;
+    for (uint64_t h : hashes) combined ^= h;
+    int cnt = (int)hashes.size();
+    for (int q = 0; q < MAXSEL; ++q)
+      pod_sel_pairs.push_back(q < cnt ? (int64_t)hashes[q] : 0);
+    pod_sel_count.push_back((int8_t)std::min(cnt, MAXSEL));
+    pod_sel_hash.push_back((int64_t)combined);
+
+    // flags
+    uint16_t f = 0;
+    const std::string_view phase = get_str(status, "phase");
+    const std::string_view node_name = get_str(spec, "nodeName");
+    if (phase == "Pending" && node_name.empty()) {
+      const JValue* conds = get(status, "conditions");
+      if (conds && conds->kind == JValue::Arr) {
+        for (const JValue* c : conds->items) {
+          if (get_str(c, "type") == "PodScheduled" &&
+              get_str(c, "status") == "False" &&
+              get_str(c, "reason") == "Unschedulable") {
+            f |= PF_PENDING;
+            break;
+          }
+        }
+      }
+    }
+    bool mirror = false, critical = false, daemonset = false, replicated = false;
+    bool evict_opt_out = false;
+    const JValue* annotations = get(meta, "annotations");
+    if (annotations && annotations->kind == JValue::Obj) {
+      for (auto& m : annotations->members) {
+        if (m.first == "kubernetes.io/config.mirror") mirror = true;
+        else if (m.first == "scheduler.alpha.kubernetes.io/critical-pod" &&
+                 m.second->kind == JValue::Str && ieq_true(m.second->str))
+          critical = true;
+        else if (m.first == "cluster-autoscaler.kubernetes.io/safe-to-evict" &&
+                 m.second->kind == JValue::Str) {
+          std::string t;
+          for (char ch : m.second->str) t += (char)tolower((unsigned char)ch);
+          if (t == "false") evict_opt_out = true;
+        }
+      }
+    }
+    const JValue* owners = get(meta, "ownerReferences");
+    if (owners && owners->kind == JValue::Arr) {
+      for (const JValue* o : owners->items) {
+        const std::string_view kind = get_str(o, "kind");
+        if (kind == "DaemonSet") daemonset = true;
+        if (kind_is_replicated(kind)) replicated = true;
+      }
+    }
+    if (mirror) f |= PF_MIRROR;
+    if (daemonset) f |= PF_DAEMONSET;
+    if (critical) f |= PF_CRITICAL;
+    if (replicated) f |= PF_REPLICATED;
+    // drainable / blocks-drain mirror models/kube.py exactly
+    bool drainable = !evict_opt_out && (replicated || mirror || daemonset) && !critical;
+    if (drainable) f |= PF_DRAINABLE;
+    bool blocks = !(mirror || daemonset) && !drainable;
+    if (blocks) f |= PF_BLOCKS_DRAIN;
+    if (cnt > MAXSEL) f |= PF_SEL_OVERFLOW;
+
+    int32_t nidx = -1;
+    if (!node_name.empty()) {
+      auto it = node_idx.find(std::string(node_name));
+      if (it != node_idx.end()) {
+        nidx = it->second;
+        f |= PF_BOUND;
+      }
+    }
+    pod_node.push_back(nidx);
+    pod_flags.push_back(f);
+
+    // PDB assignment: first PDB whose matchLabels are all present
+    int32_t pdb = -1;
+    if (!pdbs.empty()) {
+      const JValue* labels = get(meta, "labels");
+      for (size_t d = 0; d < pdbs.size(); ++d) {
+        if (pdbs[d].empty()) continue;
+        bool all = true;
+        for (auto& [k, v] : pdbs[d]) {
+          const JValue* lv = labels ? get(labels, k) : nullptr;
+          if (!lv || lv->kind != JValue::Str || lv->str != v) {
+            all = false;
+            break;
+          }
+        }
+        if (all) {
+          pdb = (int32_t)d;
+          break;
+        }
+      }
+    }
+    pod_pdb.push_back(pdb);
+
+    uids.append(std::string(get_str(meta, "uid")));
+    std::string ns(get_str(meta, "namespace"));
+    if (ns.empty()) ns = "default";
+    refs.append(py::make_tuple(ns, std::string(get_str(meta, "name"))));
+
+    parser.reset_arena();
+    parser.skip_ws();
+    if (parser.at(',')) {
+      ++parser.p_;
+      continue;
+    }
+    parser.consume(']');
+    break;
+  }
+
+  const ssize_t P = (ssize_t)pod_flags.size();
+  auto arr = [&](auto& vec, std::vector<ssize_t> shape) {
+    using T = typename std::remove_reference_t<decltype(vec)>::value_type;
+    py::array_t<T> out(shape);
+    std::memcpy(out.mutable_data(), vec.data(), vec.size() * sizeof(T));
+    return out;
+  };
+  py::dict result;
+  result["pod_req"] = arr(pod_req, {P, (ssize_t)NRES});
+  result["pod_flags"] = arr(pod_flags, {P});
+  result["pod_sel_hash"] = arr(pod_sel_hash, {P});
+  result["pod_sel_pairs"] = arr(pod_sel_pairs, {P, (ssize_t)MAXSEL});
+  result["pod_sel_count"] = arr(pod_sel_count, {P});
+  result["pod_node"] = arr(pod_node, {P});
+  result["pod_pdb"] = arr(pod_pdb, {P});
+  result["pod_uids"] = uids;
+  result["pod_refs"] = refs;
+  return result;
+}
