@@ -56,6 +56,8 @@ logger = logging.getLogger(__name__)
               help="scale-down utilization bound, per-mille of the worst axis")
 @click.option("--max-pool-size", default=100, show_default=True)
 @click.option("--prometheus-port", default=None, type=int, help="expose /metrics on this port")
+@click.option("--watch/--no-watch", default=False, show_default=True,
+              help="delta ingest: one full relist, then pod watch events patch the cached snapshot")
 @click.option("--max-cycles", default=None, type=int, hidden=True, help="stop after N cycles (tests)")
 def main(
     resource_group,
@@ -81,6 +83,7 @@ def main(
     utilization_threshold,
     max_pool_size,
     prometheus_port,
+    watch,
     max_cycles,
 ):
     """MI355X-native Kubernetes GPU-node autoscaler for ACS-Engine pools."""
@@ -122,6 +125,7 @@ def main(
         no_scale=no_scale,
         no_maintenance=no_maintenance,
         dry_run=dry_run,
+        watch=watch,
     )
     cluster = Cluster(
         kube,

@@ -77,6 +77,10 @@ class ClusterOptions:
     # ingest pods via the C++ snapshot codec (JSON bytes -> SoA) when
     # the extension is built; the Python path is the fallback/spec.
     native_codec: bool = True
+    # watch-based delta ingest: one full relist, then pod watch events
+    # patch the cached SoA (snapshot/delta.py); auto re-primes on node
+    # churn or a dropped watch stream.
+    watch: bool = False
 
 
 @dataclass
@@ -109,6 +113,7 @@ class Cluster:
         # idle bookkeeping across cycles: node name -> first-seen-idle ts
         self._idle_since: Dict[str, float] = {}
         self._cycle = 0
+        self._watch_state: Optional[Dict] = None
 
     # ------------------------------------------------------------------
     def loop(self, max_cycles: Optional[int] = None, debug: bool = False):
@@ -136,22 +141,25 @@ class Cluster:
 
         use_codec = self.opts.native_codec and _codec_available()
         with t.phase("ingest"):
-            if use_codec:
-                pods_body = self.kube.list_pods_raw()
-                pods_json = None
-            else:
-                pods_json = self.kube.list_pods()
             nodes_json = self.kube.list_nodes()
             pdbs_json = self.kube.list_pdbs()
-        nodes = [KubeNode(n) for n in nodes_json]
-        pdbs = [PodDisruptionBudget.from_json(d) for d in pdbs_json]
-        pools = group_nodes_into_pools(
-            nodes, ignore_pools=self.opts.ignore_pools, max_size=self.opts.max_pool_size
-        )
+            nodes = [KubeNode(n) for n in nodes_json]
+            pdbs = [PodDisruptionBudget.from_json(d) for d in pdbs_json]
+            pools = group_nodes_into_pools(
+                nodes, ignore_pools=self.opts.ignore_pools, max_size=self.opts.max_pool_size
+            )
+            pods_body = pods_json = None
+            if not self.opts.watch:
+                if use_codec:
+                    pods_body = self.kube.list_pods_raw()
+                else:
+                    pods_json = self.kube.list_pods()
         result.nodes_total = len(nodes)
 
         with t.phase("encode"):
-            if use_codec:
+            if self.opts.watch:
+                snap = self._watch_ingest(nodes, pools, pdbs, use_codec)
+            elif use_codec:
                 from ..snapshot.codec import encode_snapshot_fast
 
                 snap = encode_snapshot_fast(pods_body, nodes, pools, pdbs)
@@ -192,6 +200,74 @@ class Cluster:
         result.phase_times = t.end_cycle()
         self.metrics.count("cycles")
         return result
+
+    # ------------------------------------------------------------------
+    # watch-based delta ingest (host half)
+    # ------------------------------------------------------------------
+    def _watch_ingest(self, nodes, pools, pdbs, use_codec) -> ClusterSnapshot:
+        import queue as _q
+        import threading
+
+        from ..snapshot.delta import DeltaSnapshotCache
+
+        node_names = tuple(n.name for n in nodes)
+        ws = self._watch_state
+        if ws is not None and ws["broken"]:
+            logger.warning("watch stream dropped; re-priming from a full relist")
+            ws = self._watch_state = None
+        if ws is not None and ws["node_names"] != node_names:
+            logger.info("node set changed; re-priming the watch cache")
+            ws = self._watch_state = None
+
+        if ws is None:
+            if use_codec:
+                from ..snapshot.codec import encode_snapshot_fast
+
+                snap = encode_snapshot_fast(self.kube.list_pods_raw(), nodes, pools, pdbs)
+            else:
+                pods = [KubePod(p) for p in self.kube.list_pods()]
+                snap = encode_snapshot(pods, nodes, pools, pdbs)
+            cache = DeltaSnapshotCache()
+            cache.prime(
+                snap,
+                {n.name: i for i, n in enumerate(nodes)},
+                [dict(p.match_labels) for p in pdbs],
+            )
+            q: _q.SimpleQueue = _q.SimpleQueue()
+            th = threading.Thread(target=self._watch_pump, args=(q,), daemon=True)
+            th.start()
+            self._watch_state = {
+                "cache": cache, "q": q, "thread": th,
+                "node_names": node_names, "broken": False,
+            }
+            return snap
+
+        cache = ws["cache"]
+        applied = 0
+        while True:
+            try:
+                ev = ws["q"].get_nowait()
+            except _q.Empty:
+                break
+            if ev.get("type") == "__STREAM_END__":
+                ws["broken"] = True
+                break
+            cache.apply_event(ev)
+            applied += 1
+        cache.maybe_compact()
+        self.metrics.count("watch_events", applied)
+        base = encode_snapshot([], nodes, pools, pdbs)
+        return cache.build_snapshot(base)
+
+    def _watch_pump(self, q):
+        """Background thread: stream pod watch events into the queue;
+        a dropped stream posts a sentinel so the loop re-primes."""
+        try:
+            for ev in self.kube.watch("/api/v1/pods", resource_version="0"):
+                q.put(ev)
+        except Exception:
+            logger.exception("watch stream error")
+        q.put({"type": "__STREAM_END__"})
 
     # ------------------------------------------------------------------
     def classify_node_states(self, snap: ClusterSnapshot, report, now: float) -> List[str]:

@@ -22,7 +22,7 @@ from fastapi.responses import JSONResponse
 
 
 class FakeKubeState:
-    """Thread-safe in-memory object store."""
+    """Thread-safe in-memory object store with a watch event log."""
 
     def __init__(self):
         self.lock = threading.RLock()
@@ -31,9 +31,12 @@ class FakeKubeState:
         self.pdbs: Dict[str, Dict] = {}
         self.fault_delay_s: float = 0.0
         self.fail_next: int = 0           # fail the next N requests with 500
+        self.drop_watch_after: int = -1   # fault injection: cut streams
         self.deleted_pods: List[str] = []
         self.deleted_nodes: List[str] = []
         self.patched_nodes: List[str] = []
+        self.resource_version: int = 0
+        self.pod_events: List[Dict] = []  # [{rv, type, object}]
 
     def load(self, pods: List[Dict], nodes: List[Dict], pdbs: Optional[List[Dict]] = None):
         with self.lock:
@@ -43,6 +46,36 @@ class FakeKubeState:
             }
             self.nodes = {n["metadata"]["name"]: copy.deepcopy(n) for n in nodes}
             self.pdbs = {d["metadata"]["name"]: copy.deepcopy(d) for d in (pdbs or [])}
+            self.pod_events = []
+            self.resource_version += 1
+
+    # -- watch-visible mutations ---------------------------------------
+    def _emit(self, etype: str, obj: Dict):
+        self.resource_version += 1
+        obj = copy.deepcopy(obj)
+        obj.setdefault("metadata", {})["resourceVersion"] = str(self.resource_version)
+        self.pod_events.append(
+            {"rv": self.resource_version, "type": etype, "object": obj}
+        )
+
+    def add_pod(self, pod: Dict):
+        with self.lock:
+            key = f"{pod['metadata'].get('namespace', 'default')}/{pod['metadata']['name']}"
+            self.pods[key] = copy.deepcopy(pod)
+            self._emit("ADDED", pod)
+
+    def modify_pod(self, pod: Dict):
+        with self.lock:
+            key = f"{pod['metadata'].get('namespace', 'default')}/{pod['metadata']['name']}"
+            self.pods[key] = copy.deepcopy(pod)
+            self._emit("MODIFIED", pod)
+
+    def remove_pod(self, namespace: str, name: str):
+        with self.lock:
+            key = f"{namespace}/{name}"
+            pod = self.pods.pop(key, None)
+            if pod is not None:
+                self._emit("DELETED", pod)
 
 
 def build_app(state: FakeKubeState) -> FastAPI:
@@ -59,9 +92,39 @@ def build_app(state: FakeKubeState) -> FastAPI:
         return await call_next(request)
 
     @app.get("/api/v1/pods")
-    def list_pods():
+    def list_pods(watch: bool = False, resourceVersion: str = "0"):
+        if watch:
+            from starlette.responses import StreamingResponse
+
+            start_rv = int(resourceVersion or 0)
+
+            def stream():
+                import json as _json
+
+                sent = 0
+                last_rv = start_rv
+                while True:
+                    batch = []
+                    with state.lock:
+                        for e in state.pod_events:
+                            if e["rv"] > last_rv:
+                                batch.append(e)
+                    for e in batch:
+                        if state.drop_watch_after >= 0 and sent >= state.drop_watch_after:
+                            return  # injected disconnect
+                        last_rv = e["rv"]
+                        sent += 1
+                        yield _json.dumps({"type": e["type"], "object": e["object"]}) + "\n"
+                    if not batch:
+                        time.sleep(0.02)
+
+            return StreamingResponse(stream(), media_type="application/json")
         with state.lock:
-            return {"kind": "PodList", "items": list(state.pods.values()), "metadata": {}}
+            return {
+                "kind": "PodList",
+                "items": list(state.pods.values()),
+                "metadata": {"resourceVersion": str(state.resource_version)},
+            }
 
     @app.get("/api/v1/nodes")
     def list_nodes():
@@ -90,7 +153,9 @@ def build_app(state: FakeKubeState) -> FastAPI:
     def delete_pod(ns: str, name: str):
         with state.lock:
             key = f"{ns}/{name}"
-            state.pods.pop(key, None)
+            pod = state.pods.pop(key, None)
+            if pod is not None:
+                state._emit("DELETED", pod)
             state.deleted_pods.append(key)
             return {"status": "Success"}
 

@@ -51,6 +51,14 @@ void pod_pool_map(torch::Tensor slot_of_pod, torch::Tensor slot2seg,
                   torch::Tensor seg_pool, torch::Tensor pod_pool);
 pybind11::dict encode_pods_json(pybind11::bytes body, pybind11::dict node_index,
                                 pybind11::list pdb_selectors);
+void snapshot_scatter(torch::Tensor rows, torch::Tensor u_req,
+                      torch::Tensor u_flags, torch::Tensor u_sel_hash,
+                      torch::Tensor u_sel_pairs, torch::Tensor u_sel_count,
+                      torch::Tensor u_node, torch::Tensor u_pdb,
+                      torch::Tensor pod_req, torch::Tensor pod_flags,
+                      torch::Tensor pod_sel_hash, torch::Tensor pod_sel_pairs,
+                      torch::Tensor pod_sel_count, torch::Tensor pod_node,
+                      torch::Tensor pod_pdb);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native autoscaler scheduling kernels (gfx950 HIP)";
@@ -68,4 +76,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pod_pool_map", &pod_pool_map, "pod -> routed pool map");
   m.def("encode_pods_json", &encode_pods_json,
         "PodList JSON -> pod SoA columns (native snapshot codec)");
+  m.def("snapshot_scatter", &snapshot_scatter,
+        "kernel E: delta-apply rows into the resident snapshot");
 }

@@ -485,6 +485,30 @@ class GpuEngine(SimulationEngine):
         return torch.zeros((K, 4), dtype=torch.int64, device=device)
 
     # ------------------------------------------------------------------
+    def apply_pod_delta(self, ds: DeviceSnapshot, cache) -> DeviceSnapshot:
+        """Kernel E: scatter a DeltaSnapshotCache's dirty rows into the
+        resident device snapshot. Falls back to a full repack when the
+        pod-array capacity changed (grow/compact)."""
+        rows = cache.drain_dirty()
+        if ds.P != cache.size or (rows.size and int(rows.max()) >= ds.P):
+            return None  # capacity changed (grow/compact): caller repacks
+        if rows.size == 0:
+            return ds
+        d = self.device
+        to_i32 = lambda a: torch.from_numpy(np.ascontiguousarray(a[rows].astype(np.int32))).to(d)
+        to_i64 = lambda a: torch.from_numpy(np.ascontiguousarray(a[rows].astype(np.int64))).to(d)
+        r = torch.from_numpy(rows).to(d)
+        a = cache.arrays
+        self.ops.snapshot_scatter(
+            r, to_i32(a["pod_req"]), to_i32(a["pod_flags"]), to_i64(a["pod_sel_hash"]),
+            to_i64(a["pod_sel_pairs"]), to_i32(a["pod_sel_count"]),
+            to_i32(a["pod_node"]), to_i32(a["pod_pdb"]),
+            ds.pod_req, ds.pod_flags, ds.pod_sel_hash, ds.pod_sel_pairs,
+            ds.pod_sel_count, ds.pod_node, ds.pod_pdb,
+        )
+        return ds
+
+    # ------------------------------------------------------------------
     def maintain(self, snap: ClusterSnapshot, util_threshold_permille: int = 300) -> MaintainReport:
         ds = self.packer.pack(snap, self.device)
         rep, _ = self.maintain_device(ds, util_threshold_permille)

@@ -1,0 +1,244 @@
+"""Watch-based delta ingest: apply pod watch events to a cached SoA
+snapshot instead of re-listing/re-encoding the world every cycle.
+
+The reference re-listed everything per cycle [R: autoscaler/cluster.py
+· loop_logic — reconstructed]; at 1M pods the relist dominates wall
+time, so this cache is the host half of SURVEY §2.2 row E
+(snapshot_diff): events mutate SoA rows in place (tombstoned deletes,
+slot reuse, amortized compaction), and the per-cycle delta — the set
+of touched row indices — is what the device scatter kernel
+(ops/hip/snapshot_diff.hip) applies to the HBM-resident snapshot.
+
+Checkpoint/resume: `save()`/`load()` spill the cached SoA
+to an .npz so a 1M-pod replay can resume mid-burst.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Sequence, Set, Tuple
+
+import numpy as np
+
+from ..models.kube import KubePod
+from .encode import MAX_SELECTOR_PAIRS, ClusterSnapshot, PodFlags, encode_snapshot
+
+_GROW = 1.5
+_COMPACT_DEAD_FRACTION = 0.5
+
+_POD_COLS = ("pod_req", "pod_flags", "pod_sel_hash", "pod_sel_pairs",
+             "pod_sel_count", "pod_node", "pod_pdb")
+
+
+class DeltaSnapshotCache:
+    """Maintains the pod half of a ClusterSnapshot under watch events.
+
+    Node/pool arrays come from a (cheap) per-cycle re-encode of the few
+    node objects; pod columns live here and are patched row-wise.
+    """
+
+    def __init__(self):
+        self.cap = 0
+        self.size = 0
+        self.uid_row: Dict[str, int] = {}
+        self.free_rows: List[int] = []
+        self.dirty: Set[int] = set()
+        self.arrays: Dict[str, np.ndarray] = {}
+        self.pod_uids: List[str] = []
+        self.pod_refs: List[Tuple[str, str]] = []
+        self.node_index: Dict[str, int] = {}
+        self.pdb_selectors: List[Dict[str, str]] = []
+
+    # ------------------------------------------------------------------
+    def prime(self, snap: ClusterSnapshot, node_index: Dict[str, int],
+              pdb_selectors: Optional[List[Dict[str, str]]] = None):
+        """Seed from a full list/encode."""
+        self.size = snap.num_pods
+        self.cap = max(64, int(self.size * _GROW))
+        self.arrays = {}
+        for col in _POD_COLS:
+            src = getattr(snap, col)
+            shape = (self.cap,) + src.shape[1:]
+            a = np.zeros(shape, src.dtype)
+            a[: self.size] = src
+            self.arrays[col] = a
+        self.pod_uids = list(snap.pod_uids) + [""] * (self.cap - self.size)
+        self.pod_refs = list(snap.pod_refs) + [("", "")] * (self.cap - self.size)
+        self.uid_row = {u: i for i, u in enumerate(snap.pod_uids) if u}
+        self.free_rows = []
+        self.dirty = set()
+        self.node_index = dict(node_index)
+        self.pdb_selectors = list(pdb_selectors or [])
+
+    # ------------------------------------------------------------------
+    def _encode_row(self, pod: KubePod):
+        """One pod -> lattice row (same rules as encode_snapshot)."""
+        from .encode import _hash_pairs, quantize_request
+
+        req = np.array(quantize_request(pod.resources), np.int32)
+        pairs, cnt, combined = _hash_pairs(pod.selectors, MAX_SELECTOR_PAIRS)
+        f = 0
+        if pod.is_pending_unschedulable:
+            f |= PodFlags.PENDING
+        if pod.is_mirrored:
+            f |= PodFlags.MIRROR
+        if pod.is_daemonset:
+            f |= PodFlags.DAEMONSET
+        if pod.is_critical:
+            f |= PodFlags.CRITICAL
+        if pod.is_replicated:
+            f |= PodFlags.REPLICATED
+        if pod.is_drainable:
+            f |= PodFlags.DRAINABLE
+        if pod.blocks_drain:
+            f |= PodFlags.BLOCKS_DRAIN
+        if cnt > MAX_SELECTOR_PAIRS:
+            f |= PodFlags.SELECTOR_OVERFLOW
+        node = -1
+        if pod.node_name and pod.node_name in self.node_index:
+            node = self.node_index[pod.node_name]
+            f |= PodFlags.BOUND
+        pdb = -1
+        for d, sel in enumerate(self.pdb_selectors):
+            if sel and all(pod.labels.get(k) == v for k, v in sel.items()):
+                pdb = d
+                break
+        return req, pairs, min(cnt, MAX_SELECTOR_PAIRS), combined, f, node, pdb
+
+    def _write_row(self, row: int, pod: KubePod):
+        req, pairs, cnt, combined, f, node, pdb = self._encode_row(pod)
+        a = self.arrays
+        a["pod_req"][row] = req
+        a["pod_sel_pairs"][row] = pairs
+        a["pod_sel_count"][row] = cnt
+        a["pod_sel_hash"][row] = combined
+        a["pod_flags"][row] = f
+        a["pod_node"][row] = node
+        a["pod_pdb"][row] = pdb
+        self.pod_uids[row] = pod.uid
+        self.pod_refs[row] = (pod.namespace, pod.name)
+        self.dirty.add(row)
+
+    def _clear_row(self, row: int):
+        for col in _POD_COLS:
+            self.arrays[col][row] = 0
+        self.arrays["pod_node"][row] = -1
+        self.arrays["pod_pdb"][row] = -1
+        self.pod_uids[row] = ""
+        self.pod_refs[row] = ("", "")
+        self.dirty.add(row)
+
+    def _grow(self):
+        new_cap = max(64, int(self.cap * _GROW) + 1)
+        for col, a in self.arrays.items():
+            b = np.zeros((new_cap,) + a.shape[1:], a.dtype)
+            b[: self.cap] = a
+            self.arrays[col] = b
+        self.pod_uids += [""] * (new_cap - self.cap)
+        self.pod_refs += [("", "")] * (new_cap - self.cap)
+        self.cap = new_cap
+
+    # ------------------------------------------------------------------
+    def apply_event(self, event: Dict):
+        etype = event.get("type")
+        obj = event.get("object") or {}
+        pod = KubePod(obj)
+        if etype in ("ADDED", "MODIFIED"):
+            row = self.uid_row.get(pod.uid)
+            if row is None:
+                if self.free_rows:
+                    row = self.free_rows.pop()
+                else:
+                    if self.size >= self.cap:
+                        self._grow()
+                    row = self.size
+                    self.size += 1
+                self.uid_row[pod.uid] = row
+            self._write_row(row, pod)
+        elif etype == "DELETED":
+            row = self.uid_row.pop(pod.uid, None)
+            if row is not None:
+                self._clear_row(row)
+                self.free_rows.append(row)
+
+    def drain_dirty(self) -> np.ndarray:
+        """Rows touched since the last call (the device scatter set)."""
+        rows = np.array(sorted(self.dirty), np.int64)
+        self.dirty = set()
+        return rows
+
+    @property
+    def dead_fraction(self) -> float:
+        return len(self.free_rows) / max(1, self.size)
+
+    def maybe_compact(self) -> bool:
+        """Drop tombstones when they dominate (rebuild row map)."""
+        if self.dead_fraction < _COMPACT_DEAD_FRACTION:
+            return False
+        live = [i for i in range(self.size) if self.pod_uids[i]]
+        for col in _POD_COLS:
+            a = self.arrays[col]
+            a[: len(live)] = a[live]
+            a[len(live): self.size] = 0
+        self.pod_uids = [self.pod_uids[i] for i in live] + [""] * (self.cap - len(live))
+        self.pod_refs = [self.pod_refs[i] for i in live] + [("", "")] * (self.cap - len(live))
+        self.uid_row = {u: i for i, u in enumerate(self.pod_uids[: len(live)]) if u}
+        self.size = len(live)
+        self.free_rows = []
+        self.dirty = set(range(self.size))  # full rescatter after compact
+        return True
+
+    # ------------------------------------------------------------------
+    def build_snapshot(self, base: ClusterSnapshot) -> ClusterSnapshot:
+        """Combine cached pod columns with a freshly-encoded node/pool
+        side (ClusterSnapshot for the engines; tombstone rows are inert:
+        flags=0, node=-1)."""
+        import dataclasses
+
+        snap = dataclasses.replace(
+            base,
+            pod_req=self.arrays["pod_req"][: self.size],
+            pod_flags=self.arrays["pod_flags"][: self.size],
+            pod_sel_hash=self.arrays["pod_sel_hash"][: self.size],
+            pod_sel_pairs=self.arrays["pod_sel_pairs"][: self.size],
+            pod_sel_count=self.arrays["pod_sel_count"][: self.size],
+            pod_node=self.arrays["pod_node"][: self.size],
+            pod_pdb=self.arrays["pod_pdb"][: self.size],
+            pod_uids=self.pod_uids[: self.size],
+            pod_refs=self.pod_refs[: self.size],
+        )
+        from .encode import compute_node_free
+
+        compute_node_free(snap)
+        return snap
+
+    # ------------------------------------------------------------------
+    # checkpoint / resume (snapshot spill)
+    def save(self, path: str):
+        np.savez_compressed(
+            path,
+            size=self.size,
+            uids=np.array(self.pod_uids[: self.size], dtype=object),
+            refs=np.array(self.pod_refs[: self.size], dtype=object),
+            **{c: self.arrays[c][: self.size] for c in _POD_COLS},
+        )
+
+    def load(self, path: str, node_index: Dict[str, int],
+             pdb_selectors: Optional[List[Dict[str, str]]] = None):
+        data = np.load(path, allow_pickle=True)
+        self.size = int(data["size"])
+        self.cap = max(64, int(self.size * _GROW))
+        self.arrays = {}
+        for col in _POD_COLS:
+            src = data[col]
+            a = np.zeros((self.cap,) + src.shape[1:], src.dtype)
+            a[: self.size] = src
+            self.arrays[col] = a
+        uids = list(data["uids"])
+        refs = [tuple(r) for r in data["refs"]]
+        self.pod_uids = uids + [""] * (self.cap - self.size)
+        self.pod_refs = refs + [("", "")] * (self.cap - self.size)
+        self.uid_row = {u: i for i, u in enumerate(uids) if u}
+        self.free_rows = [i for i, u in enumerate(uids) if not u]
+        self.dirty = set(range(self.size))
+        self.node_index = dict(node_index)
+        self.pdb_selectors = list(pdb_selectors or [])

@@ -1,0 +1,103 @@
+"""DeltaSnapshotCache: watch-event delta ingest must converge to the
+same snapshot a full re-encode would produce, through adds, mods,
+deletes, slot reuse, compaction, and checkpoint/resume."""
+
+import numpy as np
+import pytest
+
+from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools
+from mi355x_autoscaler.models.kube import KubeNode, KubePod
+from mi355x_autoscaler.sim.engine import CpuEngine
+from mi355x_autoscaler.sim.synth import make_node_json, make_pod_json
+from mi355x_autoscaler.snapshot.delta import DeltaSnapshotCache
+from mi355x_autoscaler.snapshot.encode import encode_snapshot
+
+
+def _base(nodes):
+    pools = group_nodes_into_pools(nodes)
+    return encode_snapshot([], nodes, pools), pools
+
+
+def _full(pods_json, nodes):
+    pools = group_nodes_into_pools(nodes)
+    return encode_snapshot([KubePod(p) for p in pods_json], nodes, pools)
+
+
+def _assert_equivalent_decisions(snap_a, snap_b):
+    da = CpuEngine().decide(snap_a)
+    db = CpuEngine().decide(snap_b)
+    assert da.units_needed == db.units_needed
+    assert da.pods_placed_existing == db.pods_placed_existing
+    assert da.pods_unfulfillable == db.pods_unfulfillable
+    ra = CpuEngine().maintain(snap_a)
+    rb = CpuEngine().maintain(snap_b)
+    np.testing.assert_array_equal(ra.node_util_permille, rb.node_util_permille)
+
+
+def test_delta_add_modify_delete_converges():
+    nodes = [KubeNode(make_node_json("agentpool1", i)) for i in range(3)]
+    base, pools = _base(nodes)
+    initial = [make_pod_json(f"p{i}", cpu_milli=100, pending=True) for i in range(5)]
+    cache = DeltaSnapshotCache()
+    cache.prime(_full(initial, nodes), {n.name: i for i, n in enumerate(nodes)})
+
+    # adds, a modification, a bind, deletes
+    added = [make_pod_json(f"n{i}", cpu_milli=300, mem_mib=256, pending=True) for i in range(4)]
+    for p in added:
+        cache.apply_event({"type": "ADDED", "object": p})
+    mod = dict(initial[0])
+    mod_bound = make_pod_json("p0", cpu_milli=100, node_name=nodes[0].name)
+    cache.apply_event({"type": "MODIFIED", "object": mod_bound})
+    cache.apply_event({"type": "DELETED", "object": initial[1]})
+    cache.apply_event({"type": "DELETED", "object": initial[2]})
+    # slot reuse
+    readd = make_pod_json("fresh", cpu_milli=50, pending=True)
+    cache.apply_event({"type": "ADDED", "object": readd})
+
+    final_json = [mod_bound, initial[3], initial[4]] + added + [readd]
+    snap_delta = cache.build_snapshot(base)
+    snap_full = _full(final_json, nodes)
+    _assert_equivalent_decisions(snap_delta, snap_full)
+    assert len(cache.drain_dirty()) > 0
+    assert len(cache.drain_dirty()) == 0  # drained
+
+
+def test_delta_growth_and_compaction():
+    nodes = [KubeNode(make_node_json("agentpool1", 0))]
+    base, pools = _base(nodes)
+    cache = DeltaSnapshotCache()
+    cache.prime(_full([], nodes), {nodes[0].name: 0})
+    pods = [make_pod_json(f"g{i}", cpu_milli=10, pending=True) for i in range(300)]
+    for p in pods:
+        cache.apply_event({"type": "ADDED", "object": p})
+    assert cache.size == 300
+    for p in pods[:250]:
+        cache.apply_event({"type": "DELETED", "object": p})
+    assert cache.dead_fraction > 0.5
+    assert cache.maybe_compact()
+    assert cache.size == 50
+    snap = cache.build_snapshot(base)
+    snap_full = _full(pods[250:], nodes)
+    _assert_equivalent_decisions(snap, snap_full)
+
+
+def test_delta_checkpoint_resume(tmp_path):
+    nodes = [KubeNode(make_node_json("agentpool1", 0))]
+    base, pools = _base(nodes)
+    cache = DeltaSnapshotCache()
+    cache.prime(_full([], nodes), {nodes[0].name: 0})
+    pods = [make_pod_json(f"c{i}", cpu_milli=100, pending=True) for i in range(20)]
+    for p in pods:
+        cache.apply_event({"type": "ADDED", "object": p})
+    path = str(tmp_path / "spill.npz")
+    cache.save(path)
+
+    fresh = DeltaSnapshotCache()
+    fresh.load(path, {nodes[0].name: 0})
+    # resume mid-burst: apply more events on the restored cache
+    more = [make_pod_json(f"m{i}", cpu_milli=200, pending=True) for i in range(5)]
+    for p in more:
+        fresh.apply_event({"type": "ADDED", "object": p})
+    snap = fresh.build_snapshot(base)
+    snap_full = _full(pods + more, nodes)
+    _assert_equivalent_decisions(snap, snap_full)
