@@ -256,3 +256,51 @@ def test_burst1m_decide_matches_oracle(engine):
     d_cpu = CpuEngine().decide(snap)
     _assert_decisions_equal(d_gpu, d_cpu)
     assert d_gpu.pods_placed_existing + d_gpu.pods_placed_new + d_gpu.pods_unfulfillable == 1_000_000
+
+
+def test_snapshot_scatter_delta_matches_full_upload(engine):
+    """Kernel E: decisions on a delta-patched resident snapshot equal
+    decisions on a fresh full upload."""
+    from mi355x_autoscaler.models.kube import KubeNode, KubePod
+    from mi355x_autoscaler.snapshot.delta import DeltaSnapshotCache
+    from mi355x_autoscaler.snapshot.encode import encode_snapshot
+    from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools
+
+    nodes_json = [make_node_json("agentpool1", i) for i in range(4)]
+    nodes = [KubeNode(n) for n in nodes_json]
+    pools = group_nodes_into_pools(nodes)
+    initial = [make_pod_json(f"p{i}", cpu_milli=100, pending=True) for i in range(50)]
+    snap0 = encode_snapshot([KubePod(p) for p in initial], nodes, pools)
+
+    cache = DeltaSnapshotCache()
+    cache.prime(snap0, {n.name: i for i, n in enumerate(nodes)})
+    cache.drain_dirty()
+    ds = engine.packer.pack(snap0, engine.device)
+
+    # mutate: adds within capacity, a modify, a delete
+    for i in range(10):
+        cache.apply_event({"type": "ADDED", "object": make_pod_json(
+            f"d{i}", cpu_milli=400, mem_mib=256, pending=True)})
+    cache.apply_event({"type": "DELETED", "object": initial[0]})
+    cache.apply_event({"type": "MODIFIED", "object": make_pod_json(
+        "p1", cpu_milli=900, mem_mib=512, pending=True)})
+
+    base = encode_snapshot([], nodes, pools)
+    snap1 = cache.build_snapshot(base)
+
+    if cache.size <= ds.P:
+        ds2 = engine.apply_pod_delta(ds, cache)
+    else:
+        ds2 = None
+    if ds2 is None:
+        ds2 = engine.packer.pack(snap1, engine.device)
+    else:
+        # P may exceed live rows in snap1? sizes must match for compare
+        assert ds2.P == cache.size
+
+    segs = engine.build_segments(ds2)
+    d_delta = engine.decide_device(ds2, segs, snap1.pool_names)
+    d_full = CpuEngine().decide(snap1)
+    assert d_delta.units_needed == d_full.units_needed
+    assert d_delta.pods_placed_existing == d_full.pods_placed_existing
+    assert d_delta.pods_unfulfillable == d_full.pods_unfulfillable
