@@ -88,22 +88,45 @@ __device__ inline int probe_claim(int64_t ks, int64_t kr, int32_t H,
   return -1;
 }
 
-// Wave-aggregated build: a burst has few distinct (selector, request)
-// keys, so per-pod atomics on the handful of hot slots serialize at L2
-// (51 us for 10k pods in the round-1 profile). Leader election over
-// the wave's distinct keys turns 64 pod-atomics into one atomic per
-// distinct key per wave (~6x fewer in practice).
-extern "C" __global__ void segment_hash_build(
+// Two-level build: per-block LDS sub-table -> one global flush per
+// distinct key per block. A burst has few distinct (selector, request)
+// keys, so per-wave global atomics still serialized on the hot L2
+// lines (413 us at 1M pods); aggregating in LDS first cuts the global
+// atomic count from (waves x distinct) to (blocks x distinct).
+//
+// Intra-wave deadlock discipline: a lane spinning on another lane of
+// the SAME wave never terminates (lockstep exec mask), so every
+// claim/probe section runs leader-at-a-time via ballot/ffs — only one
+// lane of a wave is ever inside a probe loop.
+#define LDS_SLOTS 256
+
+extern "C" __global__ __launch_bounds__(256) void segment_hash_build(
     const int32_t* __restrict__ pod_flags, const int64_t* __restrict__ sel_hash,
     const int32_t* __restrict__ pod_req, int64_t P, int32_t H,
     int64_t* __restrict__ slot_ks, int64_t* __restrict__ slot_kr,
     int32_t* __restrict__ slot_state, int32_t* __restrict__ slot_count,
     int32_t* __restrict__ slot_first, int32_t* __restrict__ slot_of_pod,
     int32_t* __restrict__ overflow) {
-  const int lane = threadIdx.x & (WAVE - 1);
+  __shared__ int64_t l_ks[LDS_SLOTS];
+  __shared__ int64_t l_kr[LDS_SLOTS];
+  __shared__ int32_t l_state[LDS_SLOTS];
+  __shared__ int32_t l_count[LDS_SLOTS];
+  __shared__ int32_t l_first[LDS_SLOTS];
+  __shared__ int32_t l_gslot[LDS_SLOTS];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  for (int q = tid; q < LDS_SLOTS; q += blockDim.x) {
+    l_state[q] = ST_EMPTY;
+    l_count[q] = 0;
+    l_first[q] = 0;  // stores INT32_MAX - min(pod id) under atomicMax
+  }
+  __syncthreads();
+
+  // pass 1: aggregate this block's pods into the LDS table; stash the
+  // LDS slot id (or direct-global fallback) in slot_of_pod for pass 2.
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; ; i0 += stride) {
-    // every lane of the wave iterates in lockstep (uniform trip count)
+  for (int64_t i0 = (int64_t)blockIdx.x * blockDim.x + tid; ; i0 += stride) {
     if (__all(i0 >= P)) break;
     const bool live = i0 < P && (pod_flags[i0] & PF_PENDING);
     int64_t ks = 0, kr = 0;
@@ -117,31 +140,113 @@ extern "C" __global__ void segment_hash_build(
       slot_of_pod[i0] = -1;
     }
 
-    // leader election over distinct keys among live lanes
     uint64_t pending_lanes = __ballot(live);
-    int my_slot = -1;
+    int my_enc = -1;
     while (pending_lanes) {
       const int leader = __ffsll((unsigned long long)pending_lanes) - 1;
       const int64_t k0 = __shfl(ks, leader, WAVE);
       const int64_t r0 = __shfl(kr, leader, WAVE);
       const bool match = live && ks == k0 && kr == r0;
       const uint64_t match_mask = __ballot(match);
-      int slot = -1;
+      int enc = -1;
       if (lane == leader) {
-        slot = probe_claim(k0, r0, H, slot_ks, slot_kr, slot_state, overflow);
-        if (slot >= 0) {
-          atomicAdd(&slot_count[slot], __popcll((unsigned long long)match_mask));
-          // lanes ascend in pod id, so the leader holds the minimum
-          atomicMax(&slot_first[slot], INT32_MAX - (int32_t)i0);
+        // LDS find-or-claim (leader-at-a-time => no intra-wave spin)
+        uint32_t h = mix_hash(k0, r0) & (LDS_SLOTS - 1);
+        for (int probe = 0; probe < LDS_SLOTS; ++probe) {
+          int st = __hip_atomic_load(&l_state[h], __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_WORKGROUP);
+          if (st == ST_EMPTY) {
+            int expected = ST_EMPTY;
+            if (__hip_atomic_compare_exchange_strong(
+                    &l_state[h], &expected, ST_CLAIMED, __ATOMIC_RELAXED,
+                    __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_WORKGROUP)) {
+              l_ks[h] = k0;
+              l_kr[h] = r0;
+              __hip_atomic_store(&l_state[h], ST_READY, __ATOMIC_RELEASE,
+                                 __HIP_MEMORY_SCOPE_WORKGROUP);
+              enc = (int)h;
+              break;
+            }
+            st = expected;
+          }
+          if (st == ST_CLAIMED) {
+            while (__hip_atomic_load(&l_state[h], __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_WORKGROUP) == ST_CLAIMED) {
+              __builtin_amdgcn_s_sleep(1);
+            }
+            st = ST_READY;
+          }
+          if (st == ST_READY) {
+            __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "workgroup");
+            if (l_ks[h] == k0 && l_kr[h] == r0) {
+              enc = (int)h;
+              break;
+            }
+          }
+          h = (h + 1u) & (LDS_SLOTS - 1);
+        }
+        if (enc < 0) {
+          // LDS table full: direct global path for this key
+          const int g = probe_claim(k0, r0, H, slot_ks, slot_kr, slot_state, overflow);
+          if (g >= 0) {
+            atomicAdd(&slot_count[g], __popcll((unsigned long long)match_mask));
+            atomicMax(&slot_first[g], INT32_MAX - (int32_t)i0);
+            enc = -(g + 2);
+          } else {
+            atomicOr(overflow, 1);
+            enc = INT32_MIN;  // marks unresolvable
+          }
         } else {
-          atomicOr(overflow, 1);
+          atomicAdd(&l_count[enc], __popcll((unsigned long long)match_mask));
+          atomicMax(&l_first[enc], INT32_MAX - (int32_t)i0);
         }
       }
-      slot = __shfl(slot, leader, WAVE);
-      if (match) my_slot = slot;
+      enc = __shfl(enc, leader, WAVE);
+      if (match) my_enc = enc;
       pending_lanes &= ~match_mask;
     }
-    if (i0 < P && live) slot_of_pod[i0] = my_slot;
+    if (i0 < P && live) slot_of_pod[i0] = my_enc;
+  }
+  __syncthreads();
+
+  // flush: wave 0 resolves each occupied LDS slot to a global slot,
+  // leader-at-a-time (same intra-wave discipline).
+  if (tid < WAVE) {
+    for (int base = 0; base < LDS_SLOTS; base += WAVE) {
+      const int q = base + lane;
+      const bool occupied = l_state[q] == ST_READY && l_count[q] > 0;
+      uint64_t mask = __ballot(occupied);
+      int gres = -1;
+      while (mask) {
+        const int leader = __ffsll((unsigned long long)mask) - 1;
+        if (lane == leader) {
+          const int g = probe_claim(l_ks[q], l_kr[q], H, slot_ks, slot_kr,
+                                    slot_state, overflow);
+          if (g >= 0) {
+            atomicAdd(&slot_count[g], l_count[q]);
+            atomicMax(&slot_first[g], l_first[q]);
+            gres = g;
+          } else {
+            atomicOr(overflow, 1);
+            gres = INT32_MIN;
+          }
+        }
+        mask &= mask - 1;
+      }
+      if (occupied) l_gslot[q] = gres;
+    }
+  }
+  __syncthreads();
+
+  // pass 2: remap stashed LDS slot ids to global slot ids.
+  for (int64_t i0 = (int64_t)blockIdx.x * blockDim.x + tid; i0 < P; i0 += stride) {
+    const int v = slot_of_pod[i0];
+    if (v >= 0) {
+      const int g = l_gslot[v];
+      slot_of_pod[i0] = g == INT32_MIN ? -1 : g;
+    } else if (v != -1) {
+      slot_of_pod[i0] = v == INT32_MIN ? -1 : -(v + 2);
+    }
   }
 }
 
