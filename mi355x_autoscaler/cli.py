@@ -58,35 +58,33 @@ logger = logging.getLogger(__name__)
 @click.option("--prometheus-port", default=None, type=int, help="expose /metrics on this port")
 @click.option("--watch/--no-watch", default=False, show_default=True,
               help="delta ingest: one full relist, then pod watch events patch the cached snapshot")
+@click.option("--config", "config_path", default=None,
+              help="TOML config file; explicit CLI flags override file values")
 @click.option("--max-cycles", default=None, type=int, hidden=True, help="stop after N cycles (tests)")
-def main(
-    resource_group,
-    service_principal_app_id,
-    service_principal_secret,
-    service_principal_tenant_id,
-    kubeconfig,
-    sleep,
-    spare_agents,
-    idle_threshold,
-    over_provision,
-    ignore_pools,
-    no_scale,
-    no_maintenance,
-    slack_hook,
-    dry_run,
-    verbose,
-    debug,
-    backend,
-    gpus,
-    cloud,
-    kube_api,
-    utilization_threshold,
-    max_pool_size,
-    prometheus_port,
-    watch,
-    max_cycles,
+@click.pass_context
+def main(ctx,
+    **params,
 ):
     """MI355X-native Kubernetes GPU-node autoscaler for ACS-Engine pools."""
+    config_path = params.pop("config_path", None)
+    if config_path:
+        from .utils.config import apply_config_defaults, load_config
+
+        params = {k: v for k, v in apply_config_defaults(ctx, load_config(config_path)).items()}
+    (resource_group, service_principal_app_id, service_principal_secret,
+     service_principal_tenant_id, kubeconfig, sleep, spare_agents,
+     idle_threshold, over_provision, ignore_pools, no_scale, no_maintenance,
+     slack_hook, dry_run, verbose, debug, backend, gpus, cloud, kube_api,
+     utilization_threshold, max_pool_size, prometheus_port, watch, max_cycles) = (
+        params["resource_group"], params["service_principal_app_id"],
+        params["service_principal_secret"], params["service_principal_tenant_id"],
+        params["kubeconfig"], params["sleep"], params["spare_agents"],
+        params["idle_threshold"], params["over_provision"], params["ignore_pools"],
+        params["no_scale"], params["no_maintenance"], params["slack_hook"],
+        params["dry_run"], params["verbose"], params["debug"], params["backend"],
+        params["gpus"], params["cloud"], params["kube_api"],
+        params["utilization_threshold"], params["max_pool_size"],
+        params["prometheus_port"], params["watch"], params["max_cycles"])
     level = logging.DEBUG if debug else (logging.INFO if verbose else logging.WARNING)
     logging.basicConfig(level=level, format="%(asctime)s %(levelname)s %(name)s %(message)s")
 

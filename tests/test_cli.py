@@ -30,3 +30,36 @@ def test_cli_reference_flags_accepted():
                  "--ignore-pools", "--no-scale", "--no-maintenance", "--slack-hook",
                  "--dry-run", "--verbose", "--debug"):
         assert flag in result.output, flag
+
+
+def test_toml_config_file(tmp_path):
+    """TOML file supplies defaults; explicit CLI flags win."""
+    cfg = tmp_path / "autoscaler.toml"
+    cfg.write_text(
+        "[autoscaler]\nsleep = 42\nspare_agents = 3\ncloud = \"fake\"\n"
+        "no_scale = true\nno_maintenance = true\ndry_run = true\n"
+    )
+    with FakeKubeServer() as srv:
+        pods, nodes = burst_trace(5, n_nodes_per_pool=(2,), seed=1)
+        srv.state.load(pods, nodes)
+        runner = CliRunner()
+        result = runner.invoke(main, [
+            "--config", str(cfg), "--kube-api", srv.base_url,
+            "--max-cycles", "1", "--sleep", "0",  # CLI overrides file's 42
+        ])
+        assert result.exit_code == 0, result.output
+
+
+def test_toml_config_validation(tmp_path):
+    from mi355x_autoscaler.utils.config import load_config
+
+    bad = tmp_path / "bad.toml"
+    bad.write_text("[autoscaler]\nbackend = \"tpu\"\n")
+    import pytest as _pytest
+
+    with _pytest.raises(Exception):
+        load_config(str(bad))
+    good = tmp_path / "good.toml"
+    good.write_text("[autoscaler]\nbackend = \"gpu\"\ngpus = 8\nidle_threshold = 600\n")
+    cfg = load_config(str(good))
+    assert cfg.backend == "gpu" and cfg.gpus == 8 and cfg.idle_threshold == 600

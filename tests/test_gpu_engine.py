@@ -304,3 +304,22 @@ def test_snapshot_scatter_delta_matches_full_upload(engine):
     assert d_delta.units_needed == d_full.units_needed
     assert d_delta.pods_placed_existing == d_full.pods_placed_existing
     assert d_delta.pods_unfulfillable == d_full.pods_unfulfillable
+
+
+def test_sharded_engine_world1_gpu_path(engine):
+    """ShardedEngine(GpuEngine) at world 1 exercises the same
+    row-serialization + merge + decide_device path the driver's 8-GPU
+    run takes; decisions must equal the plain engine's."""
+    from mi355x_autoscaler.parallel.engine_sharded import ShardedEngine
+
+    pods_json, nodes_json = burst_trace(3000, n_nodes_per_pool=(10, 5),
+        pools=(("agentpool1", "Standard_D2_v2"), ("gpupool", "Standard_ND96_MI355X_v1")),
+        gpu_fraction=0.4, seed=21, bound_pods_per_node=2)
+    snap = _snap(pods_json, nodes_json)
+    sharded = ShardedEngine(engine, check_agreement=False)
+    d1 = sharded.decide(snap)
+    d2 = engine.decide(snap)
+    _assert_decisions_equal(d1, d2)
+    rep1 = sharded.maintain(snap)
+    rep2 = engine.maintain(snap)
+    np.testing.assert_array_equal(rep1.node_util_permille, rep2.node_util_permille)
