@@ -9,19 +9,21 @@ void node_free_subtract(torch::Tensor pod_node, torch::Tensor pod_req,
                         torch::Tensor node_free);
 void fit_eligibility(torch::Tensor node_labels, torch::Tensor node_label_count,
                      torch::Tensor node_unsched, torch::Tensor seg_sel,
-                     torch::Tensor seg_sel_count, torch::Tensor elig);
+                     torch::Tensor seg_sel_count, torch::Tensor g_dev,
+                     torch::Tensor elig);
 void seg_pool_route(torch::Tensor seg_sel, torch::Tensor seg_sel_count,
                     torch::Tensor seg_req, torch::Tensor pool_labels,
                     torch::Tensor pool_label_count, torch::Tensor pool_unit,
-                    torch::Tensor seg_pool);
+                    torch::Tensor g_dev, torch::Tensor seg_pool);
 void binpack_place(torch::Tensor free, torch::Tensor elig, torch::Tensor seg_req,
-                   torch::Tensor seg_count, torch::Tensor placed_out);
+                   torch::Tensor seg_count, torch::Tensor g_dev,
+                   torch::Tensor placed_out);
 void binpack_units(torch::Tensor seg_pool, torch::Tensor seg_req,
                    torch::Tensor seg_count, torch::Tensor placed,
                    torch::Tensor pool_unit, torch::Tensor pool_max_new,
-                   torch::Tensor run_scratch, torch::Tensor units_cnt,
-                   torch::Tensor placed_new, torch::Tensor unful,
-                   torch::Tensor pool_residual);
+                   torch::Tensor g_dev, torch::Tensor run_scratch,
+                   torch::Tensor units_cnt, torch::Tensor placed_new,
+                   torch::Tensor unful, torch::Tensor pool_residual);
 void drain_flags(torch::Tensor pod_node, torch::Tensor pod_flags,
                  torch::Tensor pod_pdb, torch::Tensor pdb_allowed,
                  torch::Tensor blocked, torch::Tensor pdb_bad,

@@ -311,6 +311,9 @@ class GpuEngine(SimulationEngine):
         self.ws_seg_req = torch.zeros((SEG_MAX, 4), dtype=torch.int32, device=d)
         self.ws_seg_count = torch.zeros(SEG_MAX, dtype=torch.int64, device=d)
         self.ws_seg_first = torch.zeros(SEG_MAX, dtype=torch.int32, device=d)
+        self.ws_seg_pool = torch.zeros(SEG_MAX, dtype=torch.int32, device=d)
+        self.ws_run_scratch = torch.empty((4 * SEG_MAX + 8) * 6, dtype=torch.int64, device=d)
+        self._elig_cache: Dict[int, torch.Tensor] = {}  # W -> [SEG_MAX, W]
 
     # ------------------------------------------------------------------
     def build_segments_hash(self, ds: DeviceSnapshot):
@@ -334,7 +337,7 @@ class GpuEngine(SimulationEngine):
             self.ws_seg_req, self.ws_seg_count, self.ws_seg_first,
             self.ws_slot2seg, self.ws_stat[0:2],
         )
-        stat = self.ws_stat.cpu().numpy()  # the ONE early sync (G needed for shapes)
+        stat = self.ws_stat.cpu().numpy()  # sync: G needed for host shapes
         G, sort_ovf, hash_ovf = int(stat[0]), int(stat[1]), int(stat[2])
         if sort_ovf or hash_ovf:
             return None
@@ -349,6 +352,130 @@ class GpuEngine(SimulationEngine):
         )
         return segs
 
+    def _enqueue_hash_segments(self, ds: DeviceSnapshot) -> torch.Tensor:
+        """Hash + compact WITHOUT any host sync; returns slot_of_pod.
+        G lives in ws_stat[0] on device; overflow flags in ws_stat[1:]."""
+        ops = self.ops
+        self._ws_i32.zero_()
+        slot_of_pod = torch.empty(ds.P, dtype=torch.int32, device=self.device)
+        ops.segment_hash(
+            ds.pod_flags, ds.pod_sel_hash, ds.pod_req,
+            self.ws_slot_ks, self.ws_slot_kr, self.ws_slot_state,
+            self.ws_slot_count, self.ws_slot_first, slot_of_pod, self.ws_stat[2:3],
+        )
+        ops.segment_compact(
+            self.ws_slot_ks, self.ws_slot_kr, self.ws_slot_state,
+            self.ws_slot_count, self.ws_slot_first,
+            ds.pod_sel_pairs, ds.pod_sel_count, ds.pod_req,
+            self.ws_seg_sel_pairs, self.ws_seg_sel_count, self.ws_seg_sel_hash,
+            self.ws_seg_req, self.ws_seg_count, self.ws_seg_first,
+            self.ws_slot2seg, self.ws_stat[0:2],
+        )
+        return slot_of_pod
+
+    def _elig_buffer(self, W: int) -> torch.Tensor:
+        W = max(W, 1)
+        buf = self._elig_cache.get(W)
+        if buf is None:
+            buf = torch.empty((SEG_MAX, W), dtype=torch.int64, device=self.device)
+            self._elig_cache = {W: buf}  # keep one shape resident
+        return buf
+
+    def _decide_fused(self, ds: DeviceSnapshot, pool_names: List[str],
+                      over_provision: int = 0) -> Optional[ScaleDecision]:
+        """The zero-mid-sync decision pipeline: every kernel reads G
+        from device memory; ONE sync at the result readback. Returns
+        None on hash/sort overflow (caller uses the torch path)."""
+        ops = self.ops
+        device = self.device
+        N, K = ds.N, ds.K
+        g_dev = self.ws_stat[0:1]
+
+        slot_of_pod = self._enqueue_hash_segments(ds)
+
+        free = ds.node_alloc.clone()
+        ops.node_free_subtract(ds.pod_node, ds.pod_req, free)
+
+        W = (N + 63) // 64
+        elig = self._elig_buffer(W)
+        if N > 0:
+            ops.fit_eligibility(ds.node_labels, ds.node_label_count,
+                                ds.node_unsched, self.ws_seg_sel_pairs,
+                                self.ws_seg_sel_count, g_dev, elig)
+        seg_pool = self.ws_seg_pool
+        ops.seg_pool_route(self.ws_seg_sel_pairs, self.ws_seg_sel_count,
+                           self.ws_seg_req, ds.pool_labels, ds.pool_label_count,
+                           ds.pool_unit, g_dev, seg_pool)
+
+        # MFMA per-pool demand (zero-sort path; skipped for >16 pools)
+        if K and K <= 16:
+            pod_pool = torch.empty(ds.P, dtype=torch.int32, device=device)
+            ops.pod_pool_map(slot_of_pod, self.ws_slot2seg, seg_pool, pod_pool)
+            self._last_demand = segsum(ds.pod_req, pod_pool, K, ops)
+        else:
+            self._last_demand = None
+
+        blob = torch.zeros(3 * SEG_MAX + 5 * max(K, 1), dtype=torch.int64, device=device)
+        placed = blob[0:SEG_MAX]
+        placed_new = blob[SEG_MAX : 2 * SEG_MAX]
+        unful = blob[2 * SEG_MAX : 3 * SEG_MAX]
+        units_cnt = blob[3 * SEG_MAX : 3 * SEG_MAX + K]
+        pool_residual = blob[3 * SEG_MAX + K : 3 * SEG_MAX + 5 * K]
+        if N > 0:
+            ops.binpack_place(free, elig, self.ws_seg_req, self.ws_seg_count,
+                              g_dev, placed)
+        max_new = torch.clamp(ds.pool_max - ds.pool_cur, min=0)
+        if K:
+            ops.binpack_units(seg_pool, self.ws_seg_req, self.ws_seg_count,
+                              placed, ds.pool_unit, max_new, g_dev,
+                              self.ws_run_scratch, units_cnt, placed_new,
+                              unful, pool_residual)
+
+        # THE sync: one blob + the 3-word stat
+        blob_h = blob.cpu().numpy()
+        stat_h = self.ws_stat.cpu().numpy()
+        G, sort_ovf, hash_ovf = int(stat_h[0]), int(stat_h[1]), int(stat_h[2])
+        if sort_ovf or hash_ovf:
+            return None
+        decision = ScaleDecision()
+        if G == 0:
+            decision.pool_residual_totals = np.zeros((K, 4), np.int64)
+            return decision
+        max_new_h = np.maximum(
+            (ds.pool_max_h - ds.pool_cur_h) if ds.pool_max_h is not None
+            else np.zeros(K, np.int64), 0)
+        return self._finish_decision(
+            decision, blob_h[0:G], blob_h[SEG_MAX : SEG_MAX + G],
+            blob_h[2 * SEG_MAX : 2 * SEG_MAX + G],
+            blob_h[3 * SEG_MAX : 3 * SEG_MAX + K],
+            blob_h[3 * SEG_MAX + K : 3 * SEG_MAX + 5 * K].reshape(max(K, 1) if K else 0, 4) if K else np.zeros((0, 4), np.int64),
+            max_new_h, ds.pool_cur_h, K, pool_names, over_provision,
+            lambda: self.ws_seg_sel_hash[:G].cpu().numpy(),
+        )
+
+    @staticmethod
+    def _finish_decision(decision, placed_h, placed_new_h, unful_h, units_h,
+                         residual_h, max_new_h, pool_cur_h, K, pool_names,
+                         over_provision, sel_hash_fn):
+        decision.pods_placed_existing = int(placed_h.sum())
+        decision.pods_placed_new = int(placed_new_h.sum())
+        decision.pods_unfulfillable = int(unful_h.sum())
+        if decision.pods_unfulfillable:
+            sel_hash_h = sel_hash_fn()
+            for g in np.nonzero(unful_h > 0)[0]:
+                decision.unfulfillable_detail.append(
+                    (int(sel_hash_h[g]) & ((1 << 64) - 1), int(unful_h[g]))
+                )
+        for k in range(K):
+            n_units = int(units_h[k])
+            if n_units > 0 and over_provision > 0:
+                n_units = min(n_units + over_provision, int(max_new_h[k]))
+            if n_units > 0:
+                decision.units_needed[pool_names[k]] = n_units
+                decision.new_pool_sizes[pool_names[k]] = int(pool_cur_h[k]) + n_units
+        decision.pool_residual_totals = np.array(residual_h, np.int64).copy()
+        return decision
+
     def build_segments(self, ds: DeviceSnapshot) -> DeviceSegments:
         if not self.force_torch_segbuild:
             segs = self.build_segments_hash(ds)
@@ -359,7 +486,11 @@ class GpuEngine(SimulationEngine):
     # ------------------------------------------------------------------
     def decide(self, snap: ClusterSnapshot, over_provision: int = 0) -> ScaleDecision:
         ds = self.packer.pack(snap, self.device)
-        segs = self.build_segments(ds)
+        if not self.force_torch_segbuild:
+            decision = self._decide_fused(ds, snap.pool_names, over_provision)
+            if decision is not None:
+                return decision
+        segs = build_segments_device(ds, self.ops)
         return self.decide_device(ds, segs, snap.pool_names, over_provision)
 
     def decide_device(
@@ -386,18 +517,20 @@ class GpuEngine(SimulationEngine):
             free = free_hook(free, ds)
 
         # eligibility + pool routing (kernel A writes every word)
+        g_dev = torch.full((1,), G, dtype=torch.int32, device=device)
         W = (N + 63) // 64
         elig = torch.empty((G, max(W, 1)), dtype=torch.int64, device=device)
         if N > 0:
             ops.fit_eligibility(
                 ds.node_labels, ds.node_label_count, ds.node_unsched,
-                segs.seg_sel_pairs.contiguous(), segs.seg_sel_count.contiguous(), elig,
+                segs.seg_sel_pairs.contiguous(), segs.seg_sel_count.contiguous(),
+                g_dev, elig,
             )
         seg_pool = torch.empty(G, dtype=torch.int32, device=device)
         ops.seg_pool_route(
             segs.seg_sel_pairs.contiguous(), segs.seg_sel_count.contiguous(),
             segs.seg_req.contiguous(), ds.pool_labels, ds.pool_label_count,
-            ds.pool_unit, seg_pool,
+            ds.pool_unit, g_dev, seg_pool,
         )
 
         # MFMA per-pool pending demand over the pod axis
@@ -413,7 +546,7 @@ class GpuEngine(SimulationEngine):
             if K else torch.zeros((0, 4), dtype=torch.int64, device=device)
         if N > 0:
             ops.binpack_place(free, elig, segs.seg_req.contiguous(),
-                              segs.seg_count.contiguous(), placed)
+                              segs.seg_count.contiguous(), g_dev, placed)
 
         # units phase
         max_new_h = np.maximum(
@@ -423,11 +556,14 @@ class GpuEngine(SimulationEngine):
         ).astype(np.int32) if K else np.zeros(0, np.int32)
         max_new = torch.clamp(ds.pool_max - ds.pool_cur, min=0) if K else \
             torch.zeros(0, dtype=torch.int32, device=device)
-        run_scratch = torch.empty((4 * G + 8) * 6, dtype=torch.int64, device=device)
         if K:
+            # merged sharded segment lists may exceed SEG_MAX
+            scratch = self.ws_run_scratch
+            if 4 * G + 8 > scratch.numel() // 6:
+                scratch = torch.empty((4 * G + 8) * 6, dtype=torch.int64, device=device)
             ops.binpack_units(
                 seg_pool, segs.seg_req.contiguous(), segs.seg_count.contiguous(), placed,
-                ds.pool_unit, max_new, run_scratch, units_cnt,
+                ds.pool_unit, max_new, g_dev, scratch, units_cnt,
                 placed_new, unful, pool_residual,
             )
 

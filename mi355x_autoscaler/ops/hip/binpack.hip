@@ -31,12 +31,13 @@ extern "C" __global__ __launch_bounds__(BP_BLOCK) void binpack_kernel(
     const uint64_t* __restrict__ elig,           // [G, W]
     const int32_t* __restrict__ seg_req,         // [G, NRES]
     const int64_t* __restrict__ seg_count,       // [G]
-    int32_t N, int32_t G,
+    int32_t N, const int32_t* __restrict__ g_ptr,
     int64_t* __restrict__ placed_out)            // [G]
 {
   __shared__ int64_t lds_scan[BP_BLOCK / WAVE + 1];
   const int tid = threadIdx.x;
   const int W = (N + WAVE - 1) / WAVE;
+  const int G = *g_ptr;
 
   for (int g = 0; g < G; ++g) {
     const int64_t need = seg_count[g];
@@ -106,17 +107,23 @@ extern "C" __global__ __launch_bounds__(WAVE) void units_kernel(
     const int64_t* __restrict__ seg_count, const int64_t* __restrict__ placed,
     const int32_t* __restrict__ pool_unit,      // [K, NRES]
     const int32_t* __restrict__ pool_max_new,   // [K]
-    int32_t G, int32_t K, int32_t RTOT, int32_t use_lds,
-    int64_t* __restrict__ scratch,              // [RTOT * 6] when !use_lds
+    const int32_t* __restrict__ g_ptr, int32_t K, int32_t RTOT, int32_t lds_cap,
+    int64_t* __restrict__ scratch,              // [RTOT * 6] global arena
     int64_t* __restrict__ units_cnt,            // [K] out
     int64_t* __restrict__ placed_new, int64_t* __restrict__ unful,
     int64_t* __restrict__ pool_residual)        // [K, NRES] out
 {
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  // arena tier chosen at run time: the LDS carve covers 4G+8 runs for
+  // typical G; bigger segment lists spill to the global scratch.
+  const int G = *g_ptr;
+  const int need = 4 * G + 8;
+  const bool use_lds = need <= lds_cap;
+  const int cap = use_lds ? lds_cap : RTOT;
   int64_t* arena = use_lds ? reinterpret_cast<int64_t*>(smem) : scratch;
-  int64_t* run_res = arena;               // [RTOT][NRES]
-  int64_t* run_cnt = arena + (int64_t)RTOT * NRES;
-  int64_t* run_next = run_cnt + RTOT;
+  int64_t* run_res = arena;               // [cap][NRES]
+  int64_t* run_cnt = arena + (int64_t)cap * NRES;
+  int64_t* run_next = run_cnt + cap;
 
   __shared__ int32_t head[UNITS_MAX_POOLS];
   __shared__ int32_t tail[UNITS_MAX_POOLS];
@@ -144,7 +151,7 @@ extern "C" __global__ __launch_bounds__(WAVE) void units_kernel(
         unful[g] = rem;
         continue;
       }
-      if (alloc + 4 > RTOT) {  // arena bound: a segment allocates <=4 runs
+      if (alloc + 4 > cap) {   // arena bound: a segment allocates <=4 runs
         unful[g] = rem;        // (unreachable when RTOT >= 4G+8; kept so a
         continue;              //  bound bug can never corrupt the lists)
       }
@@ -278,39 +285,38 @@ extern "C" __global__ __launch_bounds__(WAVE) void units_kernel(
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
 void binpack_place(torch::Tensor free, torch::Tensor elig, torch::Tensor seg_req,
-                   torch::Tensor seg_count, torch::Tensor placed_out) {
+                   torch::Tensor seg_count, torch::Tensor g_dev,
+                   torch::Tensor placed_out) {
   const int N = (int)free.size(0);
-  const int G = (int)seg_req.size(0);
-  if (G == 0) return;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   hipLaunchKernelGGL(binpack_kernel, dim3(1), dim3(BP_BLOCK), 0, stream.stream(),
                      free.data_ptr<int32_t>(),
                      reinterpret_cast<uint64_t*>(elig.data_ptr<int64_t>()),
                      seg_req.data_ptr<int32_t>(), seg_count.data_ptr<int64_t>(),
-                     N, G, placed_out.data_ptr<int64_t>());
+                     N, g_dev.data_ptr<int32_t>(), placed_out.data_ptr<int64_t>());
 }
 
 void binpack_units(torch::Tensor seg_pool, torch::Tensor seg_req,
                    torch::Tensor seg_count, torch::Tensor placed,
                    torch::Tensor pool_unit, torch::Tensor pool_max_new,
-                   torch::Tensor run_scratch, torch::Tensor units_cnt,
-                   torch::Tensor placed_new, torch::Tensor unful,
-                   torch::Tensor pool_residual) {
-  const int G = (int)seg_req.size(0);
+                   torch::Tensor g_dev, torch::Tensor run_scratch,
+                   torch::Tensor units_cnt, torch::Tensor placed_new,
+                   torch::Tensor unful, torch::Tensor pool_residual) {
   const int K = (int)pool_unit.size(0);
-  if (G == 0) return;
   TORCH_CHECK(K <= UNITS_MAX_POOLS, "units_kernel supports at most 64 pools");
   const int RTOT = (int)(run_scratch.numel() / 6);
-  TORCH_CHECK(RTOT >= 4 * G + 8, "run arena must hold 4G+8 runs");
-  const size_t lds_bytes = (size_t)RTOT * 6 * sizeof(int64_t);
-  const int use_lds = lds_bytes <= 96 * 1024 ? 1 : 0;
+  TORCH_CHECK(RTOT >= 4 * (int)seg_req.size(0) + 8,
+              "run arena must hold 4*SEG_MAX+8 runs");
+  const int lds_cap = 1300;  // 1300 runs * 48 B = 62.4 KiB carve
+  const size_t lds_bytes = (size_t)lds_cap * 6 * sizeof(int64_t);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  hipLaunchKernelGGL(units_kernel, dim3(1), dim3(WAVE),
-                     use_lds ? lds_bytes : 0, stream.stream(),
+  hipLaunchKernelGGL(units_kernel, dim3(1), dim3(WAVE), lds_bytes,
+                     stream.stream(),
                      seg_pool.data_ptr<int32_t>(), seg_req.data_ptr<int32_t>(),
                      seg_count.data_ptr<int64_t>(), placed.data_ptr<int64_t>(),
                      pool_unit.data_ptr<int32_t>(),
-                     pool_max_new.data_ptr<int32_t>(), G, K, RTOT, use_lds,
+                     pool_max_new.data_ptr<int32_t>(), g_dev.data_ptr<int32_t>(),
+                     K, RTOT, lds_cap,
                      run_scratch.data_ptr<int64_t>(), units_cnt.data_ptr<int64_t>(),
                      placed_new.data_ptr<int64_t>(), unful.data_ptr<int64_t>(),
                      pool_residual.data_ptr<int64_t>());

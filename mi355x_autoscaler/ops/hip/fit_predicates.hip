@@ -30,9 +30,10 @@ extern "C" __global__ __launch_bounds__(FIT_BLOCK) void fit_eligibility_kernel(
     const int32_t* __restrict__ node_unsched,     // [N]
     const int64_t* __restrict__ seg_sel,          // [G, MAXSEL] sorted
     const int32_t* __restrict__ seg_sel_count,    // [G]
-    int32_t N, int32_t G,
+    int32_t N, const int32_t* __restrict__ g_ptr,
     uint64_t* __restrict__ elig)                  // [G, W]
 {
+  const int G = *g_ptr;  // device-resident so the host never syncs for it
   __shared__ int64_t lds_labels[NODES_PER_BLOCK * MAXLAB];
   __shared__ int32_t lds_nlab[NODES_PER_BLOCK];
 
@@ -88,11 +89,11 @@ extern "C" __global__ void seg_pool_kernel(
     const int64_t* __restrict__ pool_labels,      // [K, MAXLAB]
     const int32_t* __restrict__ pool_label_count, // [K]
     const int32_t* __restrict__ pool_unit,        // [K, NRES]
-    int32_t G, int32_t K,
+    const int32_t* __restrict__ g_ptr, int32_t K,
     int32_t* __restrict__ seg_pool)               // [G] first pool or -1
 {
   const int g = blockIdx.x;
-  if (g >= G) return;
+  if (g >= *g_ptr) return;
   const int lane = threadIdx.x & (WAVE - 1);
   const int nsel = seg_sel_count[g];
   const int64_t* sel = seg_sel + (int64_t)g * MAXSEL;
@@ -126,10 +127,10 @@ extern "C" __global__ void seg_pool_kernel(
 
 void fit_eligibility(torch::Tensor node_labels, torch::Tensor node_label_count,
                      torch::Tensor node_unsched, torch::Tensor seg_sel,
-                     torch::Tensor seg_sel_count, torch::Tensor elig) {
+                     torch::Tensor seg_sel_count, torch::Tensor g_dev,
+                     torch::Tensor elig) {
   const int N = (int)node_labels.size(0);
-  const int G = (int)seg_sel.size(0);
-  if (N == 0 || G == 0) return;
+  if (N == 0) return;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const int grid = (N + NODES_PER_BLOCK - 1) / NODES_PER_BLOCK;
   hipLaunchKernelGGL(fit_eligibility_kernel, dim3(grid), dim3(FIT_BLOCK), 0,
@@ -137,24 +138,25 @@ void fit_eligibility(torch::Tensor node_labels, torch::Tensor node_label_count,
                      node_label_count.data_ptr<int32_t>(),
                      node_unsched.data_ptr<int32_t>(),
                      seg_sel.data_ptr<int64_t>(),
-                     seg_sel_count.data_ptr<int32_t>(), N, G,
+                     seg_sel_count.data_ptr<int32_t>(), N,
+                     g_dev.data_ptr<int32_t>(),
                      reinterpret_cast<uint64_t*>(elig.data_ptr<int64_t>()));
 }
 
 void seg_pool_route(torch::Tensor seg_sel, torch::Tensor seg_sel_count,
                     torch::Tensor seg_req, torch::Tensor pool_labels,
                     torch::Tensor pool_label_count, torch::Tensor pool_unit,
-                    torch::Tensor seg_pool) {
-  const int G = (int)seg_sel.size(0);
+                    torch::Tensor g_dev, torch::Tensor seg_pool) {
+  const int Gmax = (int)seg_sel.size(0);
   const int K = (int)pool_labels.size(0);
-  if (G == 0) return;
+  if (Gmax == 0) return;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  hipLaunchKernelGGL(seg_pool_kernel, dim3(G), dim3(WAVE), 0, stream.stream(),
+  hipLaunchKernelGGL(seg_pool_kernel, dim3(Gmax), dim3(WAVE), 0, stream.stream(),
                      seg_sel.data_ptr<int64_t>(),
                      seg_sel_count.data_ptr<int32_t>(),
                      seg_req.data_ptr<int32_t>(),
                      pool_labels.data_ptr<int64_t>(),
                      pool_label_count.data_ptr<int32_t>(),
-                     pool_unit.data_ptr<int32_t>(), G, K,
+                     pool_unit.data_ptr<int32_t>(), g_dev.data_ptr<int32_t>(), K,
                      seg_pool.data_ptr<int32_t>());
 }
