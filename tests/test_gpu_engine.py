@@ -157,8 +157,10 @@ def test_fit_eligibility_kernel_direct():
     G, N = segs.G, snap.num_nodes
     W = (N + 63) // 64
     elig = torch.zeros((G, W), dtype=torch.int64, device="cuda:0")
+    g_dev = torch.full((1,), G, dtype=torch.int32, device="cuda:0")
     ops.fit_eligibility(ds.node_labels, ds.node_label_count, ds.node_unsched,
-                        segs.seg_sel_pairs.contiguous(), segs.seg_sel_count.contiguous(), elig)
+                        segs.seg_sel_pairs.contiguous(), segs.seg_sel_count.contiguous(),
+                        g_dev, elig)
     got = elig.cpu().numpy().astype(np.uint64)
     sel_pairs = segs.seg_sel_pairs.cpu().numpy()
     sel_count = segs.seg_sel_count.cpu().numpy()
