@@ -1,0 +1,84 @@
+"""Shard rebalancing (row H): skewed shards even out over all_to_all
+and the global decision is unchanged (ownership-invariance)."""
+
+import multiprocessing as mp
+import os
+import pickle
+
+import numpy as np
+import pytest
+
+from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools
+from mi355x_autoscaler.sim.engine import CpuEngine
+from mi355x_autoscaler.sim.synth import burst_trace, wrap
+from mi355x_autoscaler.snapshot.encode import encode_snapshot
+
+WORLD = 2
+
+
+def _snap_for_rank(rank, skewed=True):
+    pods_json, nodes_json = burst_trace(400, n_nodes_per_pool=(4,), seed=8,
+                                        bound_pods_per_node=2)
+    if skewed:
+        # rank 0 gets 90% of the pods
+        cut = int(len(pods_json) * 0.9)
+        pods_json = pods_json[:cut] if rank == 0 else pods_json[cut:]
+    else:
+        pods_json = pods_json[rank::WORLD]
+    pods, nodes = wrap(pods_json, nodes_json)
+    return encode_snapshot(pods, nodes, group_nodes_into_pools(nodes))
+
+
+def _worker(rank, q):
+    os.environ.update(WORLD_SIZE=str(WORLD), RANK=str(rank), LOCAL_RANK=str(rank),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT="29631")
+    import torch.distributed as dist
+
+    from mi355x_autoscaler.parallel.engine_sharded import ShardedEngine
+    from mi355x_autoscaler.parallel.rebalance import rebalance_shard
+
+    try:
+        engine = ShardedEngine(CpuEngine())
+        snap = _snap_for_rank(rank)
+        before = snap.num_pods
+        d_before = engine.decide(snap)
+        snap2, did = rebalance_shard(snap, skew_threshold=1.2)
+        d_after = engine.decide(snap2)
+        q.put((rank, pickle.dumps(dict(
+            before=before, after=snap2.num_pods, did=did,
+            units_b=d_before.units_needed, units_a=d_after.units_needed,
+            pe_b=d_before.pods_placed_existing, pe_a=d_after.pods_placed_existing,
+        ))))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_rebalance_evens_shards_and_preserves_decision():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker, args=(r, q)) for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(WORLD):
+        rank, blob = q.get(timeout=180)
+        res[rank] = pickle.loads(blob)
+    for p in procs:
+        p.join(timeout=30)
+        assert p.exitcode == 0
+
+    assert res[0]["did"] and res[1]["did"]
+    total = res[0]["before"] + res[1]["before"]
+    assert res[0]["after"] + res[1]["after"] == total
+    assert abs(res[0]["after"] - res[1]["after"]) <= 1  # balanced
+    # ownership-invariance of the global decision
+    assert res[0]["units_b"] == res[0]["units_a"] == res[1]["units_a"]
+    assert res[0]["pe_b"] == res[0]["pe_a"]
+
+
+def test_rebalance_noop_when_balanced():
+    from mi355x_autoscaler.parallel.rebalance import rebalance_shard
+
+    snap = _snap_for_rank(0, skewed=False)
+    out, did = rebalance_shard(snap)  # world == 1 here
+    assert not did and out is snap
