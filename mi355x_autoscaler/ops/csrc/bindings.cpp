@@ -47,7 +47,8 @@ void segment_compact(torch::Tensor slot_ks, torch::Tensor slot_kr,
                      torch::Tensor pod_sel_count, torch::Tensor pod_req,
                      torch::Tensor seg_sel_pairs, torch::Tensor seg_sel_count,
                      torch::Tensor seg_sel_hash, torch::Tensor seg_req,
-                     torch::Tensor seg_count, torch::Tensor seg_first_pod,
+                     torch::Tensor seg_count, torch::Tensor seg_key,
+                     torch::Tensor seg_first_pod,
                      torch::Tensor slot2seg, torch::Tensor g_out);
 void pod_pool_map(torch::Tensor slot_of_pod, torch::Tensor slot2seg,
                   torch::Tensor seg_pool, torch::Tensor pod_pool);

@@ -43,7 +43,7 @@ from ..sim.engine import SimulationEngine
 from ..snapshot.encode import ClusterSnapshot, PodFlags
 from . import load_ops
 
-_PACK_MAX = 1 << 52
+_PACK_MAX = 1 << 56
 _SEGSUM_CHUNK = 256  # exactness bound for the f32 MFMA limb chains
 SEG_MAX = 2048
 HASH_SLOTS = 8192
@@ -164,6 +164,7 @@ class DeviceSegments:
     seg_sel_hash: torch.Tensor   # [G] i64
     seg_req: torch.Tensor        # [G,4] i32
     seg_count: torch.Tensor      # [G] i64
+    seg_key: Optional[torch.Tensor] = None  # [G] i64 raw pack key (merge id)
     # pod -> segment mapping, one of:
     slot_of_pod: Optional[torch.Tensor] = None  # [P] i32 (hash path)
     slot2seg: Optional[torch.Tensor] = None     # [H] i32 (hash path)
@@ -223,6 +224,7 @@ def build_segments_device(ds: DeviceSnapshot, ops) -> DeviceSegments:
         seg_sel_hash=ds.pod_sel_hash[first_pod],
         seg_req=ds.pod_req[first_pod],
         seg_count=counts.to(torch.int64),
+        seg_key=_PACK_MAX - uniq[:, 1],  # un-invert the classify key
         pod_sorted=pod_sorted,
         pod_seg=inverse.to(torch.int64),
     )
@@ -310,6 +312,7 @@ class GpuEngine(SimulationEngine):
         self.ws_seg_sel_hash = torch.zeros(SEG_MAX, dtype=torch.int64, device=d)
         self.ws_seg_req = torch.zeros((SEG_MAX, 4), dtype=torch.int32, device=d)
         self.ws_seg_count = torch.zeros(SEG_MAX, dtype=torch.int64, device=d)
+        self.ws_seg_key = torch.zeros(SEG_MAX, dtype=torch.int64, device=d)
         self.ws_seg_first = torch.zeros(SEG_MAX, dtype=torch.int32, device=d)
         self.ws_seg_pool = torch.zeros(SEG_MAX, dtype=torch.int32, device=d)
         self.ws_run_scratch = torch.empty((4 * SEG_MAX + 8) * 6, dtype=torch.int64, device=d)
@@ -334,8 +337,8 @@ class GpuEngine(SimulationEngine):
             self.ws_slot_count, self.ws_slot_first,
             ds.pod_sel_pairs, ds.pod_sel_count, ds.pod_req,
             self.ws_seg_sel_pairs, self.ws_seg_sel_count, self.ws_seg_sel_hash,
-            self.ws_seg_req, self.ws_seg_count, self.ws_seg_first,
-            self.ws_slot2seg, self.ws_stat[0:2],
+            self.ws_seg_req, self.ws_seg_count, self.ws_seg_key,
+            self.ws_seg_first, self.ws_slot2seg, self.ws_stat[0:2],
         )
         stat = self.ws_stat.cpu().numpy()  # sync: G needed for host shapes
         G, sort_ovf, hash_ovf = int(stat[0]), int(stat[1]), int(stat[2])
@@ -347,6 +350,7 @@ class GpuEngine(SimulationEngine):
             seg_sel_hash=self.ws_seg_sel_hash[:G],
             seg_req=self.ws_seg_req[:G],
             seg_count=self.ws_seg_count[:G],
+            seg_key=self.ws_seg_key[:G],
             slot_of_pod=slot_of_pod,
             slot2seg=self.ws_slot2seg,
         )
@@ -368,8 +372,8 @@ class GpuEngine(SimulationEngine):
             self.ws_slot_count, self.ws_slot_first,
             ds.pod_sel_pairs, ds.pod_sel_count, ds.pod_req,
             self.ws_seg_sel_pairs, self.ws_seg_sel_count, self.ws_seg_sel_hash,
-            self.ws_seg_req, self.ws_seg_count, self.ws_seg_first,
-            self.ws_slot2seg, self.ws_stat[0:2],
+            self.ws_seg_req, self.ws_seg_count, self.ws_seg_key,
+            self.ws_seg_first, self.ws_slot2seg, self.ws_stat[0:2],
         )
         return slot_of_pod
 

@@ -16,7 +16,7 @@
 // one place, bit-identical to sim/oracle.py::build_segments.
 #include "common.h"
 
-#define PACK_MAX ((int64_t)1 << 52)
+#define PACK_MAX ((int64_t)1 << 56)
 
 extern "C" __global__ void classify_keys_kernel(
     const int32_t* __restrict__ pod_flags, const int64_t* __restrict__ sel_hash,
@@ -33,7 +33,7 @@ extern "C" __global__ void classify_keys_kernel(
       int32_t rq[NRES];
 #pragma unroll
       for (int r = 0; r < NRES; ++r) rq[r] = pod_req[i * NRES + r];
-      key_req[i] = PACK_MAX - pack_req_key(rq);
+      key_req[i] = PACK_MAX - pack_req_key(rq, f & PF_CRITICAL);
     } else {
       key_sel[i] = INT64_MAX;
       key_req[i] = INT64_MAX;

@@ -26,14 +26,18 @@
 #define PF_BOUND (1 << 8)
 
 // Request sort-key packing — must match sim/oracle.py::_pack_req_key.
-__device__ __host__ inline int64_t pack_req_key(const int32_t req[NRES]) {
+// Critical pods [R: reference critical-pod annotation] outrank every
+// non-critical shape within their selector group: bit 55 sits above
+// the resource fields, so descending key order places them first.
+__device__ __host__ inline int64_t pack_req_key(const int32_t req[NRES],
+                                                int critical) {
   int64_t cpu = req[0] < 0 ? 0 : (int64_t)req[0];
   int64_t mem = req[1] < 0 ? 0 : (int64_t)req[1];
   int64_t gpu = req[3] < 0 ? 0 : (int64_t)req[3];
   if (cpu > (1 << 21) - 1) cpu = (1 << 21) - 1;
   if (mem > (1 << 23) - 1) mem = (1 << 23) - 1;
   if (gpu > (1 << 8) - 1) gpu = (1 << 8) - 1;
-  return (cpu << 31) | (mem << 8) | gpu;
+  return ((int64_t)(critical ? 1 : 0) << 55) | (cpu << 31) | (mem << 8) | gpu;
 }
 
 // Selector-subset predicate over sorted pair-hash arrays — device form of

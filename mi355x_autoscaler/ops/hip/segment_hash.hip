@@ -128,14 +128,15 @@ extern "C" __global__ __launch_bounds__(256) void segment_hash_build(
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i0 = (int64_t)blockIdx.x * blockDim.x + tid; ; i0 += stride) {
     if (__all(i0 >= P)) break;
-    const bool live = i0 < P && (pod_flags[i0] & PF_PENDING);
+    const int pf = i0 < P ? pod_flags[i0] : 0;
+    const bool live = i0 < P && (pf & PF_PENDING);
     int64_t ks = 0, kr = 0;
     if (live) {
       ks = sel_hash[i0];
       int32_t rq[NRES];
 #pragma unroll
       for (int r = 0; r < NRES; ++r) rq[r] = pod_req[i0 * NRES + r];
-      kr = pack_req_key(rq);
+      kr = pack_req_key(rq, pf & PF_CRITICAL);
     } else if (i0 < P) {
       slot_of_pod[i0] = -1;
     }
@@ -265,7 +266,7 @@ extern "C" __global__ __launch_bounds__(CS_BLOCK) void segment_compact_sort(
     int64_t* __restrict__ seg_sel_pairs,  // [SEG_MAX, MAXSEL]
     int32_t* __restrict__ seg_sel_count, int64_t* __restrict__ seg_sel_hash,
     int32_t* __restrict__ seg_req, int64_t* __restrict__ seg_count,
-    int32_t* __restrict__ seg_first_pod,
+    int64_t* __restrict__ seg_key, int32_t* __restrict__ seg_first_pod,
     int32_t* __restrict__ slot2seg,       // [H]
     int32_t* __restrict__ g_out,          // [2]: G, overflow(G>SEG_MAX)
     int32_t sort_dir_check)               // unused; keeps ABI room
@@ -341,6 +342,7 @@ extern "C" __global__ __launch_bounds__(CS_BLOCK) void segment_compact_sort(
     const int fp = INT32_MAX - slot_first[h];
     seg_sel_hash[g] = (int64_t)(lds_ks[g] ^ 0x8000000000000000ull);
     seg_count[g] = (int64_t)slot_count[h];
+    seg_key[g] = (int64_t)~lds_kr[g];  // un-invert: the raw pack key
     seg_first_pod[g] = fp;
     seg_sel_count[g] = pod_sel_count[fp];
 #pragma unroll
@@ -398,7 +400,8 @@ void segment_compact(torch::Tensor slot_ks, torch::Tensor slot_kr,
                      torch::Tensor pod_sel_count, torch::Tensor pod_req,
                      torch::Tensor seg_sel_pairs, torch::Tensor seg_sel_count,
                      torch::Tensor seg_sel_hash, torch::Tensor seg_req,
-                     torch::Tensor seg_count, torch::Tensor seg_first_pod,
+                     torch::Tensor seg_count, torch::Tensor seg_key,
+                     torch::Tensor seg_first_pod,
                      torch::Tensor slot2seg, torch::Tensor g_out) {
   const int H = (int)slot_ks.size(0);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
@@ -411,7 +414,8 @@ void segment_compact(torch::Tensor slot_ks, torch::Tensor slot_kr,
                      seg_sel_pairs.data_ptr<int64_t>(),
                      seg_sel_count.data_ptr<int32_t>(),
                      seg_sel_hash.data_ptr<int64_t>(), seg_req.data_ptr<int32_t>(),
-                     seg_count.data_ptr<int64_t>(), seg_first_pod.data_ptr<int32_t>(),
+                     seg_count.data_ptr<int64_t>(), seg_key.data_ptr<int64_t>(),
+                     seg_first_pod.data_ptr<int32_t>(),
                      slot2seg.data_ptr<int32_t>(), g_out.data_ptr<int32_t>(), 0);
 }
 

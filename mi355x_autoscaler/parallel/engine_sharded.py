@@ -40,7 +40,7 @@ from ..snapshot.encode import MAX_SELECTOR_PAIRS, ClusterSnapshot
 from . import dist as pdist
 
 _U64 = (1 << 64) - 1
-_ROW_COLS = MAX_SELECTOR_PAIRS + 7  # pairs(8) + count + req(4) + seg_count + sel_hash
+_ROW_COLS = MAX_SELECTOR_PAIRS + 8  # pairs(8) + count + req(4) + seg_count + sel_hash + req_key
 
 
 def _segments_to_rows(segs: List[Segment]) -> np.ndarray:
@@ -51,6 +51,7 @@ def _segments_to_rows(segs: List[Segment]) -> np.ndarray:
         rows[i, MAX_SELECTOR_PAIRS + 1 : MAX_SELECTOR_PAIRS + 5] = s.req
         rows[i, MAX_SELECTOR_PAIRS + 5] = s.count
         rows[i, MAX_SELECTOR_PAIRS + 6] = s.sel_hash
+        rows[i, MAX_SELECTOR_PAIRS + 7] = s.req_key
     return rows
 
 
@@ -64,6 +65,7 @@ def _rows_to_segments(rows: np.ndarray) -> List[Segment]:
                 sel_count=int(r[MAX_SELECTOR_PAIRS]),
                 req=r[MAX_SELECTOR_PAIRS + 1 : MAX_SELECTOR_PAIRS + 5].astype(np.int32),
                 count=int(r[MAX_SELECTOR_PAIRS + 5]),
+                req_key=int(r[MAX_SELECTOR_PAIRS + 7]),
             )
         )
     return out
@@ -76,10 +78,7 @@ def merge_rows(rows: np.ndarray) -> np.ndarray:
     if rows.shape[0] == 0:
         return rows
     sel_u = rows[:, MAX_SELECTOR_PAIRS + 6].astype(np.uint64)
-    keys = np.array(
-        [_pack_req_key(r[MAX_SELECTOR_PAIRS + 1 : MAX_SELECTOR_PAIRS + 5]) for r in rows],
-        dtype=np.uint64,
-    )
+    keys = rows[:, MAX_SELECTOR_PAIRS + 7].astype(np.uint64)
     order = np.lexsort((np.iinfo(np.uint64).max - keys, sel_u))
     rows = rows[order]
     sel_u = sel_u[order]
@@ -164,6 +163,7 @@ class ShardedEngine(SimulationEngine):
             rows[:, MAX_SELECTOR_PAIRS + 1 : MAX_SELECTOR_PAIRS + 5] = segs.seg_req.cpu().numpy()
             rows[:, MAX_SELECTOR_PAIRS + 5] = segs.seg_count.cpu().numpy()
             rows[:, MAX_SELECTOR_PAIRS + 6] = segs.seg_sel_hash.cpu().numpy()
+            rows[:, MAX_SELECTOR_PAIRS + 7] = segs.seg_key.cpu().numpy()
         return rows
 
     @staticmethod
@@ -187,6 +187,9 @@ class ShardedEngine(SimulationEngine):
             ).to(device),
             seg_count=torch.from_numpy(
                 np.ascontiguousarray(rows[:, MAX_SELECTOR_PAIRS + 5])
+            ).to(device),
+            seg_key=torch.from_numpy(
+                np.ascontiguousarray(rows[:, MAX_SELECTOR_PAIRS + 7])
             ).to(device),
             pod_sorted=e64,
             pod_seg=e64,

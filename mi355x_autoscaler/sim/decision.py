@@ -21,6 +21,7 @@ class Segment:
     sel_count: int
     req: np.ndarray               # [4] int32 quantized request
     count: int                    # number of pods in the segment
+    req_key: int = 0              # pack_req_key(req, critical) — the merge key
     pod_indices: Optional[np.ndarray] = None  # global pod ids (local shard)
 
 

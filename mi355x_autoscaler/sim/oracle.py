@@ -47,25 +47,30 @@ from .decision import DrainCandidate, MaintainReport, ScaleDecision, Segment
 _I64 = np.int64
 
 
-def _pack_req_key(req: np.ndarray) -> int:
+def _pack_req_key(req: np.ndarray, critical: bool = False) -> int:
     """Pack a request vector into one uint64 sort key such that
-    descending key order == descending (cpu, memory, gpu) tuple order.
+    descending key order == descending (critical, cpu, memory, gpu)
+    tuple order — critical pods [R: reference critical-pod annotation]
+    outrank every non-critical shape within their selector group.
     cpu milli <= 2^21 (2096 cores), mem MiB <= 2^23 (8 TiB), gpu <= 2^8.
     Values clamp into their fields; the pods axis is the constant 1 for
-    a single pod and is excluded."""
+    a single pod and is excluded. Must match ops/hip/common.h."""
     cpu = min(int(req[0]), (1 << 21) - 1)
     mem = min(int(req[1]), (1 << 23) - 1)
     gpu = min(int(req[3]), (1 << 8) - 1)
-    return (cpu << 31) | (mem << 8) | gpu
+    return ((1 << 55) if critical else 0) | (cpu << 31) | (mem << 8) | gpu
 
 
-def pack_req_keys(req: np.ndarray) -> np.ndarray:
+def pack_req_keys(req: np.ndarray, critical: Optional[np.ndarray] = None) -> np.ndarray:
     """Vectorized _pack_req_key over a [M, 4] int array -> uint64[M]."""
     r = req.astype(np.int64)
     cpu = np.clip(r[:, 0], 0, (1 << 21) - 1).astype(np.uint64)
     mem = np.clip(r[:, 1], 0, (1 << 23) - 1).astype(np.uint64)
     gpu = np.clip(r[:, 3], 0, (1 << 8) - 1).astype(np.uint64)
-    return (cpu << np.uint64(31)) | (mem << np.uint64(8)) | gpu
+    key = (cpu << np.uint64(31)) | (mem << np.uint64(8)) | gpu
+    if critical is not None:
+        key |= critical.astype(np.uint64) << np.uint64(55)
+    return key
 
 
 def build_segments(
@@ -86,7 +91,8 @@ def build_segments(
         return []
 
     sel = snap.pod_sel_hash[pending].astype(np.uint64)
-    keys = pack_req_keys(snap.pod_req[pending])
+    crit = (snap.pod_flags[pending] & PodFlags.CRITICAL) != 0
+    keys = pack_req_keys(snap.pod_req[pending], crit)
     # order: sel asc, key desc, pod index asc (determinism of pod lists)
     order = np.lexsort((pending, np.iinfo(np.uint64).max - keys, sel))
     pending = pending[order]
@@ -105,6 +111,7 @@ def build_segments(
                     sel_count=int(snap.pod_sel_count[p0]),
                     req=snap.pod_req[p0].copy(),
                     count=int(i - start),
+                    req_key=int(keys[start]),
                     pod_indices=pending[start:i].copy(),
                 )
             )
@@ -119,7 +126,7 @@ def merge_segment_counts(per_rank: List[List[Segment]]) -> List[Segment]:
     table: Dict[Tuple[int, int], Segment] = {}
     for segs in per_rank:
         for s in segs:
-            k = ((int(s.sel_hash) & ((1 << 64) - 1)), _pack_req_key(s.req))
+            k = ((int(s.sel_hash) & ((1 << 64) - 1)), int(s.req_key))
             if k in table:
                 table[k].count += s.count
             else:
@@ -129,6 +136,7 @@ def merge_segment_counts(per_rank: List[List[Segment]]) -> List[Segment]:
                     sel_count=s.sel_count,
                     req=s.req.copy(),
                     count=s.count,
+                    req_key=s.req_key,
                     pod_indices=None,
                 )
     order = sorted(table.keys(), key=lambda k: (k[0], -k[1]))

@@ -159,3 +159,27 @@ def test_segment_merge_matches_global_build():
     d_merged = OracleSimulator(snap).fulfill_pending(merged)
     assert d_whole.units_needed == d_merged.units_needed
     assert d_whole.pods_placed_existing == d_merged.pods_placed_existing
+
+
+def test_critical_pods_place_first():
+    """Critical pods [R: reference critical-pod annotation] outrank
+    non-critical shapes in their selector group: with capacity for only
+    one pod, the critical one wins even though it is smaller."""
+    node = make_node_json("agentpool1", 0, cpu_milli=2000, mem_mib=7168)
+    filler = make_pod_json("filler", cpu_milli=1500, mem_mib=100,
+                           node_name=node["metadata"]["name"])
+    big = make_pod_json("big", cpu_milli=500, mem_mib=512, pending=True)
+    crit = make_pod_json(
+        "crit", cpu_milli=300, mem_mib=128, pending=True,
+        annotations={"scheduler.alpha.kubernetes.io/critical-pod": "true"},
+    )
+    snap = _snap([filler, big, crit], [node], max_size=1)  # no room to grow
+    d = OracleSimulator(snap).fulfill_pending()
+    # only the critical pod fits the 500m of remaining capacity first;
+    # 'big' (500m) then cannot fit (200m left) and the pool is capped.
+    assert d.pods_placed_existing == 1
+    assert d.pods_unfulfillable == 1
+    # and the placed one must be the critical pod: re-check by removing it
+    snap2 = _snap([filler, big], [node], max_size=1)
+    d2 = OracleSimulator(snap2).fulfill_pending()
+    assert d2.pods_placed_existing == 1  # big fits when critical absent
