@@ -36,6 +36,14 @@ extern "C" __global__ __launch_bounds__(WAVE) void segsum_mfma_kernel(
     int64_t T,
     uint64_t* __restrict__ out)           // [S, NRES] (non-negative i64)
 {
+  // The whole 256-element chunk is staged cooperatively into LDS first
+  // (4 coalesced dwordx4 loads + 1 dword load per lane, all in flight
+  // together), then the MFMA loop feeds from LDS broadcasts — the
+  // fragment-at-a-time global loads of v1 serialized on L2 latency
+  // (220 us at 1M pods).
+  __shared__ int32_t l_vals[256 * NRES];
+  __shared__ int32_t l_segs[256];
+
   const int64_t t = blockIdx.x;
   if (t >= T) return;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -47,25 +55,37 @@ extern "C" __global__ __launch_bounds__(WAVE) void segsum_mfma_kernel(
   const int64_t p_end = wi_end[t];
   const int seg_base = win * 16;
 
+  // stage: lane covers pods {lane, lane+64, lane+128, lane+192}
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    const int pl = lane + 64 * q;
+    const int64_t p = p_begin + pl;
+    if (p < p_end) {
+      const int4 v = reinterpret_cast<const int4*>(values)[p];
+      reinterpret_cast<int4*>(l_vals)[pl] = v;
+      l_segs[pl] = seg_of[p] - seg_base;
+    } else {
+      reinterpret_cast<int4*>(l_vals)[pl] = int4{0, 0, 0, 0};
+      l_segs[pl] = -1;  // no column matches: contributes zero
+    }
+  }
+  __syncthreads();
+
   // A-row decode for this lane: rows 0..7 = (limb = row>>2, res = row&3)
   const int row = col;
   const int res = row & 3;
   const int limb = (row >> 2) & 1;
   const bool live_row = row < 8;
 
+  const int steps = (int)((p_end - p_begin + 3) >> 2);
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-  for (int64_t p0 = p_begin; p0 < p_end; p0 += 4) {
-    const int64_t p = p0 + kidx;
-    float a = 0.f, b = 0.f;
-    if (p < p_end) {
-      const int sloc = seg_of[p] - seg_base;
-      // B[k][j]: indicator (pod p belongs to segment column j)
-      b = (sloc == col) ? 1.f : 0.f;
-      // A[i][k]: limb value of pod p, row i
-      if (live_row) {
-        const uint32_t v = (uint32_t)values[p * NRES + res];
-        a = (float)(limb ? (v >> 16) : (v & 0xffffu));
-      }
+  for (int s = 0; s < steps; ++s) {
+    const int pl = s * 4 + kidx;
+    const float b = (l_segs[pl] == col) ? 1.f : 0.f;
+    float a = 0.f;
+    if (live_row) {
+      const uint32_t v = (uint32_t)l_vals[pl * NRES + res];
+      a = (float)(limb ? (v >> 16) : (v & 0xffffu));
     }
     acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
   }
