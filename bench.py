@@ -89,9 +89,11 @@ def main():
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--pods-per-gpu", type=int, default=None)
     ap.add_argument("--backend", choices=["auto", "gpu", "cpu"], default="auto")
-    ap.add_argument("--bench-config", choices=["burst10k", "burst1m"], default="burst10k",
+    ap.add_argument("--bench-config", choices=["burst10k", "burst1m", "maintain500"],
+                    default="burst10k",
                     help="burst10k = BASELINE config #3 (default); burst1m = config #5 "
-                         "(1M pods, HBM-resident snapshot, re-decided each step)")
+                         "(1M pods, HBM-resident snapshot); maintain500 = config #4 "
+                         "(drain scoring over 500 under-utilized MI355X nodes with PDBs)")
     args = ap.parse_args()
     if args.pods_per_gpu is None:
         args.pods_per_gpu = 1_000_000 if args.bench_config == "burst1m" else 10_000
@@ -128,12 +130,23 @@ def main():
         engine = inner
 
     resident = args.bench_config == "burst1m"
+    maintain_mode = args.bench_config == "maintain500"
     if resident:
         snap = build_snapshot_1m(rank, args.pods_per_gpu)
+    elif maintain_mode:
+        from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools as _group
+        from mi355x_autoscaler.sim.synth import scaledown_trace
+
+        pods_json, nodes_json, pdbs = scaledown_trace(n_nodes=500, seed=1000 + rank)
+        pods, nodes = wrap(pods_json, nodes_json)
+        snap = encode_snapshot(pods, nodes, _group(nodes, max_size=1000), pdbs)
     else:
         snap = build_snapshot(rank, args.pods_per_gpu)
 
-    if resident and use_gpu and world == 1:
+    if maintain_mode:
+        def decide_once():
+            return engine.maintain(snap)
+    elif resident and use_gpu and world == 1:
         # config #5: the snapshot stays resident in HBM; each step is a
         # full re-decision (segment build + fit + pack) over it.
         ds = inner.packer.pack(snap, inner.device)
@@ -183,7 +196,7 @@ def main():
         total_s = total_s_local
 
     ms_per_step = total_s / args.steps * 1000.0
-    pods_total = world * args.pods_per_gpu
+    pods_total = world * (snap.num_pods if maintain_mode else args.pods_per_gpu)
     pods_per_sec = pods_total / (total_s / args.steps)
     p50_ms = statistics.median(step_times) * 1000.0
 
@@ -205,6 +218,8 @@ def main():
                 "model": (
                     "1M-pod burst replay, HBM-resident snapshot (BASELINE config #5)"
                     if resident else
+                    "scale-down drain scoring, 500 under-utilized MI355X nodes + PDBs (BASELINE config #4)"
+                    if maintain_mode else
                     "10k-pod mixed CPU/GPU burst, 3 heterogeneous agent pools (BASELINE config #3)"
                 ),
                 "pods_per_gpu": args.pods_per_gpu,
@@ -214,9 +229,15 @@ def main():
                 "parallelism": f"pod-sharded controller x{world}" if world > 1 else "single",
                 "backend": "gpu" if use_gpu else "cpu-oracle",
                 "decision_p50_ms": round(p50_ms, 3),
-                "pods_placed": decision.pods_placed_existing + decision.pods_placed_new,
-                "pods_unfulfillable": decision.pods_unfulfillable,
-                "units_needed": decision.units_needed,
+                **(
+                    {"drain_candidates": len(decision.candidates)}
+                    if maintain_mode else
+                    {
+                        "pods_placed": decision.pods_placed_existing + decision.pods_placed_new,
+                        "pods_unfulfillable": decision.pods_unfulfillable,
+                        "units_needed": decision.units_needed,
+                    }
+                ),
             },
         }
         print(json.dumps(result))

@@ -325,3 +325,26 @@ def test_sharded_engine_world1_gpu_path(engine):
     rep1 = sharded.maintain(snap)
     rep2 = engine.maintain(snap)
     np.testing.assert_array_equal(rep1.node_util_permille, rep2.node_util_permille)
+
+
+@pytest.mark.parametrize("n_pods,n_nodes", [
+    (1, 1), (64, 3), (1000, 500), (10000, 10000), (100000, 10000),
+])
+def test_shape_class_sweep(engine, n_pods, n_nodes):
+    """SURVEY §4 kernel shape classes: pods x nodes grid including the
+    10k-node eligibility/binpack capacity paths."""
+    from mi355x_autoscaler.sim.synth_soa import synth_burst_snapshot
+
+    per_pool = max(n_nodes // 3, 1)
+    snap = synth_burst_snapshot(
+        n_pods,
+        n_nodes_per_pool=(per_pool, per_pool, n_nodes - 2 * per_pool)
+        if n_nodes >= 3 else (n_nodes,),
+        pools=(("agentpool1", "Standard_D2_v2"), ("bigpool", "Standard_D4_v2"),
+               ("gpupool", "Standard_ND96_MI355X_v1"))[: (3 if n_nodes >= 3 else 1)],
+        gpu_fraction=0.3 if n_nodes >= 3 else 0.0,
+        seed=n_pods + n_nodes,
+        bound_pods_per_node=1 if n_nodes <= 500 else 0,
+    )
+    snap.pool_max[:] = 1 << 20
+    _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))
