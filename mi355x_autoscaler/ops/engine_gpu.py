@@ -709,13 +709,16 @@ class GpuEngine(SimulationEngine):
 
     @staticmethod
     def _candidates(rep: MaintainReport, node_names, snap, util_threshold_permille):
+        ok = (
+            (rep.node_blocked == 0)
+            & (rep.node_pdb_ok == 1)
+            & (snap.node_pool >= 0)
+            & ((rep.node_idle == 1) | (rep.node_util_permille < util_threshold_permille))
+        )
+        order = np.argsort(rep.node_util_permille, kind="stable")
         out = []
-        for j in np.argsort(rep.node_util_permille, kind="stable"):
+        for j in order[ok[order]]:
             j = int(j)
-            if rep.node_blocked[j] or not rep.node_pdb_ok[j] or snap.node_pool[j] < 0:
-                continue
-            if rep.node_util_permille[j] >= util_threshold_permille and not rep.node_idle[j]:
-                continue
             out.append(
                 DrainCandidate(
                     node_index=j,

@@ -354,14 +354,17 @@ class OracleSimulator:
         nontrivial = partials["nontrivial"]
         drainable_pods = partials["drainable"]
 
-        util_permille = np.zeros(N, np.int32)
-        for j in range(N):
-            worst = 0
-            for r in range(NUM_RESOURCES):
-                alloc = int(snap.node_alloc[j, r])
-                if alloc > 0 and r != 2:  # the pods axis is not "utilization"
-                    worst = max(worst, int(util[j, r]) * 1000 // alloc)
-            util_permille[j] = worst
+        # worst-axis permille, vectorized (pods axis is occupancy, not
+        # utilization); integer floor division matches the device kernel
+        alloc = snap.node_alloc.astype(_I64)
+        u = util.astype(_I64)
+        worst = np.zeros(N, dtype=_I64)
+        for r in (0, 1, 3):
+            a = alloc[:, r]
+            with np.errstate(divide="ignore", invalid="ignore"):
+                perm = np.where(a > 0, u[:, r] * 1000 // np.maximum(a, 1), 0)
+            worst = np.maximum(worst, perm)
+        util_permille = worst.astype(np.int32)
 
         idle = (nontrivial == 0).astype(np.uint8)
         report = MaintainReport(
@@ -371,12 +374,16 @@ class OracleSimulator:
             node_idle=idle,
             node_drainable_pods=drainable_pods.astype(np.int32),
         )
-        for j in np.argsort(util_permille, kind="stable"):
+        # vectorized candidate predicate; python touches candidates only
+        ok = (
+            (report.node_blocked == 0)
+            & (report.node_pdb_ok == 1)
+            & (snap.node_pool >= 0)
+            & ((idle == 1) | (util_permille < util_threshold_permille))
+        )
+        order = np.argsort(util_permille, kind="stable")
+        for j in order[ok[order]]:
             j = int(j)
-            if blocked[j] or pdb_bad[j] or snap.node_pool[j] < 0:
-                continue
-            if util_permille[j] >= util_threshold_permille and not idle[j]:
-                continue
             report.candidates.append(
                 DrainCandidate(
                     node_index=j,
