@@ -21,8 +21,12 @@
 #include <cmath>
 #include <cstdint>
 #include <cstring>
+#include <atomic>
+#include <chrono>
 #include <deque>
 #include <memory>
+#include <mutex>
+#include <thread>
 #include <stdexcept>
 #include <string>
 #include <string_view>
@@ -294,6 +298,13 @@ class Parser {
     decoded_.clear();
   }
 
+  void reset_to(const char* p, size_t n) {
+    p_ = p;
+    begin_ = p;
+    end_ = p + n;
+    reset_arena();
+  }
+
   [[noreturn]] void fail(const std::string& msg) {
     throw std::runtime_error("json parse error at offset " +
                              std::to_string(p_ - begin_) + ": " + msg);
@@ -429,27 +440,16 @@ bool ieq_true(std::string_view s) {
 }  // namespace
 
 // ---------------------------------------------------------------------------
-// encode_pods(json_bytes, node_index: dict[str,int],
-//             pdb_selectors: list[dict[str,str]]) -> dict of arrays
+// Parallel encode: boundary pre-scan -> per-thread parse (GIL released)
+// -> column concatenation.
 // ---------------------------------------------------------------------------
-pybind11::dict encode_pods_json(pybind11::bytes body, pybind11::dict node_index,
-                                pybind11::list pdb_selectors) {
-  namespace py = pybind11;
-  std::string_view sv = body;  // pybind keeps the bytes alive for the call
 
-  // host-side lookup tables
+struct Ctx {
   std::unordered_map<std::string, int32_t> node_idx;
-  for (auto item : node_index)
-    node_idx[py::cast<std::string>(item.first)] = py::cast<int32_t>(item.second);
   std::vector<std::vector<std::pair<std::string, std::string>>> pdbs;
-  for (auto sel : pdb_selectors) {
-    std::vector<std::pair<std::string, std::string>> pairs;
-    for (auto kv : py::cast<py::dict>(sel))
-      pairs.emplace_back(py::cast<std::string>(kv.first), py::cast<std::string>(kv.second));
-    pdbs.push_back(std::move(pairs));
-  }
+};
 
-  // column builders
+struct Columns {
   std::vector<int32_t> pod_req;
   std::vector<uint16_t> pod_flags;
   std::vector<int64_t> pod_sel_hash;
@@ -457,186 +457,314 @@ pybind11::dict encode_pods_json(pybind11::bytes body, pybind11::dict node_index,
   std::vector<int8_t> pod_sel_count;
   std::vector<int32_t> pod_node;
   std::vector<int32_t> pod_pdb;
-  py::list uids, refs;
+  std::vector<std::string> uids, ns, names;
 
-  Parser parser(sv.data(), sv.size());
-  // walk to items:[ without materializing the outer object
-  parser.skip_ws();
-  parser.consume('{');
-  bool found_items = false;
-  while (true) {
-    parser.skip_ws();
-    std::string_view key = parser.parse_string();
-    parser.consume(':');
-    if (key == "items") {
-      found_items = true;
-      break;
-    }
-    parser.parse_value();  // skip
-    parser.reset_arena();
-    parser.skip_ws();
-    if (parser.at(',')) {
-      ++parser.p_;
-      continue;
-    }
-    break;
+  void reserve(size_t n) {
+    pod_req.reserve(n * NRES);
+    pod_flags.reserve(n);
+    pod_sel_hash.reserve(n);
+    pod_sel_pairs.reserve(n * MAXSEL);
+    pod_sel_count.reserve(n);
+    pod_node.reserve(n);
+    pod_pdb.reserve(n);
+    uids.reserve(n);
+    ns.reserve(n);
+    names.reserve(n);
   }
-  if (!found_items) {
-    throw std::runtime_error("PodList has no items array");
+};
+
+void encode_one(const JValue* pod, const Ctx& ctx, Columns& out) {
+  const JValue* meta = get(pod, "metadata");
+  const JValue* spec = get(pod, "spec");
+  const JValue* status = get(pod, "status");
+
+  // requests (quantize: cpu milli, memory ceil-MiB)
+  Req rq = sum_requests(spec);
+  out.pod_req.push_back((int32_t)std::min<int64_t>(rq.cpu, INT32_MAX));
+  out.pod_req.push_back((int32_t)std::min<int64_t>((rq.mem + MIB - 1) / MIB, INT32_MAX));
+  out.pod_req.push_back((int32_t)rq.pods);
+  out.pod_req.push_back((int32_t)std::min<int64_t>(rq.gpu, INT32_MAX));
+
+  // selectors
+  const JValue* nsel = get(spec, "nodeSelector");
+  std::vector<uint64_t> hashes;
+  if (nsel && nsel->kind == JValue::Obj) {
+    for (auto& m : nsel->members) {
+      if (m.second->kind == JValue::Str)
+        hashes.push_back(hash_label_pair(m.first, m.second->str));
+    }
   }
-  parser.consume('[');
-  parser.skip_ws();
-  bool empty = parser.at(']');
-  if (empty) ++parser.p_;
+  std::sort(hashes.begin(), hashes.end());
+  uint64_t combined = 0;
+  for (uint64_t h : hashes) combined ^= h;
+  int cnt = (int)hashes.size();
+  for (int q = 0; q < MAXSEL; ++q)
+    out.pod_sel_pairs.push_back(q < cnt ? (int64_t)hashes[q] : 0);
+  out.pod_sel_count.push_back((int8_t)std::min(cnt, MAXSEL));
+  out.pod_sel_hash.push_back((int64_t)combined);
 
-  while (!empty) {
-    const JValue* pod = parser.parse_value();
-    const JValue* meta = get(pod, "metadata");
-    const JValue* spec = get(pod, "spec");
-    const JValue* status = get(pod, "status");
-
-    // requests (quantize: cpu milli, memory ceil-MiB)
-    Req rq = sum_requests(spec);
-    pod_req.push_back((int32_t)std::min<int64_t>(rq.cpu, INT32_MAX));
-    pod_req.push_back((int32_t)std::min<int64_t>((rq.mem + MIB - 1) / MIB, INT32_MAX));
-    pod_req.push_back((int32_t)rq.pods);
-    pod_req.push_back((int32_t)std::min<int64_t>(rq.gpu, INT32_MAX));
-
-    // selectors
-    const JValue* nsel = get(spec, "nodeSelector");
-    std::vector<uint64_t> hashes;
-    if (nsel && nsel->kind == JValue::Obj) {
-      for (auto& m : nsel->members) {
-        if (m.second->kind == JValue::Str)
-          hashes.push_back(hash_label_pair(m.first, m.second->str));
-      }
-    }
-    std::sort(hashes.begin(), hashes.end());
-    uint64_t combined = 0;
-    for (uint64_t h : hashes) combined ^= h;
-    int cnt = (int)hashes.size();
-    for (int q = 0; q < MAXSEL; ++q)
-      pod_sel_pairs.push_back(q < cnt ? (int64_t)hashes[q] : 0);
-    pod_sel_count.push_back((int8_t)std::min(cnt, MAXSEL));
-    pod_sel_hash.push_back((int64_t)combined);
-
-    // flags
-    uint16_t f = 0;
-    const std::string_view phase = get_str(status, "phase");
-    const std::string_view node_name = get_str(spec, "nodeName");
-    if (phase == "Pending" && node_name.empty()) {
-      const JValue* conds = get(status, "conditions");
-      if (conds && conds->kind == JValue::Arr) {
-        for (const JValue* c : conds->items) {
-          if (get_str(c, "type") == "PodScheduled" &&
-              get_str(c, "status") == "False" &&
-              get_str(c, "reason") == "Unschedulable") {
-            f |= PF_PENDING;
-            break;
-          }
-        }
-      }
-    }
-    bool mirror = false, critical = false, daemonset = false, replicated = false;
-    bool evict_opt_out = false;
-    const JValue* annotations = get(meta, "annotations");
-    if (annotations && annotations->kind == JValue::Obj) {
-      for (auto& m : annotations->members) {
-        if (m.first == "kubernetes.io/config.mirror") mirror = true;
-        else if (m.first == "scheduler.alpha.kubernetes.io/critical-pod" &&
-                 m.second->kind == JValue::Str && ieq_true(m.second->str))
-          critical = true;
-        else if (m.first == "cluster-autoscaler.kubernetes.io/safe-to-evict" &&
-                 m.second->kind == JValue::Str) {
-          std::string t;
-          for (char ch : m.second->str) t += (char)tolower((unsigned char)ch);
-          if (t == "false") evict_opt_out = true;
-        }
-      }
-    }
-    const JValue* owners = get(meta, "ownerReferences");
-    if (owners && owners->kind == JValue::Arr) {
-      for (const JValue* o : owners->items) {
-        const std::string_view kind = get_str(o, "kind");
-        if (kind == "DaemonSet") daemonset = true;
-        if (kind_is_replicated(kind)) replicated = true;
-      }
-    }
-    if (mirror) f |= PF_MIRROR;
-    if (daemonset) f |= PF_DAEMONSET;
-    if (critical) f |= PF_CRITICAL;
-    if (replicated) f |= PF_REPLICATED;
-    // drainable / blocks-drain mirror models/kube.py exactly
-    bool drainable = !evict_opt_out && (replicated || mirror || daemonset) && !critical;
-    if (drainable) f |= PF_DRAINABLE;
-    bool blocks = !(mirror || daemonset) && !drainable;
-    if (blocks) f |= PF_BLOCKS_DRAIN;
-    if (cnt > MAXSEL) f |= PF_SEL_OVERFLOW;
-
-    int32_t nidx = -1;
-    if (!node_name.empty()) {
-      auto it = node_idx.find(std::string(node_name));
-      if (it != node_idx.end()) {
-        nidx = it->second;
-        f |= PF_BOUND;
-      }
-    }
-    pod_node.push_back(nidx);
-    pod_flags.push_back(f);
-
-    // PDB assignment: first PDB whose matchLabels are all present
-    int32_t pdb = -1;
-    if (!pdbs.empty()) {
-      const JValue* labels = get(meta, "labels");
-      for (size_t d = 0; d < pdbs.size(); ++d) {
-        if (pdbs[d].empty()) continue;
-        bool all = true;
-        for (auto& [k, v] : pdbs[d]) {
-          const JValue* lv = labels ? get(labels, k) : nullptr;
-          if (!lv || lv->kind != JValue::Str || lv->str != v) {
-            all = false;
-            break;
-          }
-        }
-        if (all) {
-          pdb = (int32_t)d;
+  // flags
+  uint16_t f = 0;
+  const std::string_view phase = get_str(status, "phase");
+  const std::string_view node_name = get_str(spec, "nodeName");
+  if (phase == "Pending" && node_name.empty()) {
+    const JValue* conds = get(status, "conditions");
+    if (conds && conds->kind == JValue::Arr) {
+      for (const JValue* c : conds->items) {
+        if (get_str(c, "type") == "PodScheduled" &&
+            get_str(c, "status") == "False" &&
+            get_str(c, "reason") == "Unschedulable") {
+          f |= PF_PENDING;
           break;
         }
       }
     }
-    pod_pdb.push_back(pdb);
+  }
+  bool mirror = false, critical = false, daemonset = false, replicated = false;
+  bool evict_opt_out = false;
+  const JValue* annotations = get(meta, "annotations");
+  if (annotations && annotations->kind == JValue::Obj) {
+    for (auto& m : annotations->members) {
+      if (m.first == "kubernetes.io/config.mirror") mirror = true;
+      else if (m.first == "scheduler.alpha.kubernetes.io/critical-pod" &&
+               m.second->kind == JValue::Str && ieq_true(m.second->str))
+        critical = true;
+      else if (m.first == "cluster-autoscaler.kubernetes.io/safe-to-evict" &&
+               m.second->kind == JValue::Str) {
+        std::string t;
+        for (char ch : m.second->str) t += (char)tolower((unsigned char)ch);
+        if (t == "false") evict_opt_out = true;
+      }
+    }
+  }
+  const JValue* owners = get(meta, "ownerReferences");
+  if (owners && owners->kind == JValue::Arr) {
+    for (const JValue* o : owners->items) {
+      const std::string_view kind = get_str(o, "kind");
+      if (kind == "DaemonSet") daemonset = true;
+      if (kind_is_replicated(kind)) replicated = true;
+    }
+  }
+  if (mirror) f |= PF_MIRROR;
+  if (daemonset) f |= PF_DAEMONSET;
+  if (critical) f |= PF_CRITICAL;
+  if (replicated) f |= PF_REPLICATED;
+  // drainable / blocks-drain mirror models/kube.py exactly
+  bool drainable = !evict_opt_out && (replicated || mirror || daemonset) && !critical;
+  if (drainable) f |= PF_DRAINABLE;
+  bool blocks = !(mirror || daemonset) && !drainable;
+  if (blocks) f |= PF_BLOCKS_DRAIN;
+  if (cnt > MAXSEL) f |= PF_SEL_OVERFLOW;
 
-    uids.append(std::string(get_str(meta, "uid")));
-    std::string ns(get_str(meta, "namespace"));
-    if (ns.empty()) ns = "default";
-    refs.append(py::make_tuple(ns, std::string(get_str(meta, "name"))));
+  int32_t nidx = -1;
+  if (!node_name.empty()) {
+    auto it = ctx.node_idx.find(std::string(node_name));
+    if (it != ctx.node_idx.end()) {
+      nidx = it->second;
+      f |= PF_BOUND;
+    }
+  }
+  out.pod_node.push_back(nidx);
+  out.pod_flags.push_back(f);
 
-    parser.reset_arena();
-    parser.skip_ws();
-    if (parser.at(',')) {
-      ++parser.p_;
+  // PDB assignment: first PDB whose matchLabels are all present
+  int32_t pdb = -1;
+  if (!ctx.pdbs.empty()) {
+    const JValue* labels = get(meta, "labels");
+    for (size_t d = 0; d < ctx.pdbs.size(); ++d) {
+      if (ctx.pdbs[d].empty()) continue;
+      bool all = true;
+      for (auto& [k, v] : ctx.pdbs[d]) {
+        const JValue* lv = labels ? get(labels, k) : nullptr;
+        if (!lv || lv->kind != JValue::Str || lv->str != v) {
+          all = false;
+          break;
+        }
+      }
+      if (all) {
+        pdb = (int32_t)d;
+        break;
+      }
+    }
+  }
+  out.pod_pdb.push_back(pdb);
+
+  out.uids.emplace_back(get_str(meta, "uid"));
+  std::string ns(get_str(meta, "namespace"));
+  if (ns.empty()) ns = "default";
+  out.ns.push_back(std::move(ns));
+  out.names.emplace_back(get_str(meta, "name"));
+}
+
+// String-aware scan for the byte ranges of the items-array elements.
+// Returns {items_found, [ (begin, end) per item ]}.
+std::pair<bool, std::vector<std::pair<size_t, size_t>>> scan_items(
+    std::string_view sv) {
+  std::vector<std::pair<size_t, size_t>> spans;
+  // locate "items" key at depth 1 with a full structural walk
+  int depth = 0;
+  bool in_str = false, esc = false;
+  bool items_next = false;     // saw "items" key at depth 1, awaiting '['
+  size_t i = 0, n = sv.size();
+  size_t item_begin = 0;
+  int items_depth = -1;        // depth of the items array once entered
+  for (; i < n; ++i) {
+    const char c = sv[i];
+    if (in_str) {
+      // memchr to the closing quote; a quote preceded by an odd number
+      // of backslashes is escaped (strings are most of the bytes)
+      if (esc) { esc = false; continue; }
+      size_t j = i;
+      while (true) {
+        const void* q = memchr(sv.data() + j, '"', n - j);
+        if (!q) { i = n; in_str = false; break; }
+        size_t qi = (const char*)q - sv.data();
+        size_t bs = 0;
+        while (qi >= j + bs + 1 && sv[qi - bs - 1] == '\\') ++bs;
+        if ((bs & 1) == 0) { i = qi; in_str = false; break; }
+        j = qi + 1;
+      }
       continue;
     }
-    parser.consume(']');
-    break;
+    switch (c) {
+      case '"': {
+        if (depth == 1 && items_depth < 0) {
+          // possible key start: match "items" followed by ':'
+          if (n - i >= 7 && sv.compare(i, 7, "\"items\"") == 0) {
+            size_t j = i + 7;
+            while (j < n && (sv[j] == ' ' || sv[j] == '\t' || sv[j] == '\n' || sv[j] == '\r')) ++j;
+            if (j < n && sv[j] == ':') items_next = true;
+          }
+        }
+        in_str = true;
+        break;
+      }
+      case '{':
+        if (items_depth >= 0 && depth == items_depth) item_begin = i;
+        ++depth;
+        break;
+      case '}':
+        --depth;
+        if (items_depth >= 0 && depth == items_depth) spans.emplace_back(item_begin, i + 1);
+        break;
+      case '[':
+        if (items_next && items_depth < 0) {
+          items_depth = depth + 1;
+          items_next = false;
+        }
+        ++depth;
+        break;
+      case ']':
+        --depth;
+        if (items_depth >= 0 && depth == items_depth - 1) return {true, spans};
+        break;
+      default:
+        break;
+    }
+  }
+  return {items_depth >= 0, spans};
+}
+
+pybind11::dict encode_pods_json(pybind11::bytes body, pybind11::dict node_index,
+                                pybind11::list pdb_selectors) {
+  namespace py = pybind11;
+  std::string_view sv = body;  // pybind keeps the bytes alive for the call
+
+  Ctx ctx;
+  for (auto item : node_index)
+    ctx.node_idx[py::cast<std::string>(item.first)] = py::cast<int32_t>(item.second);
+  for (auto sel : pdb_selectors) {
+    std::vector<std::pair<std::string, std::string>> pairs;
+    for (auto kv : py::cast<py::dict>(sel))
+      pairs.emplace_back(py::cast<std::string>(kv.first), py::cast<std::string>(kv.second));
+    ctx.pdbs.push_back(std::move(pairs));
   }
 
-  const ssize_t P = (ssize_t)pod_flags.size();
-  auto arr = [&](auto& vec, std::vector<ssize_t> shape) {
-    using T = typename std::remove_reference_t<decltype(vec)>::value_type;
-    py::array_t<T> out(shape);
-    std::memcpy(out.mutable_data(), vec.data(), vec.size() * sizeof(T));
-    return out;
+  const bool dbg = getenv("MI355X_CODEC_DEBUG") != nullptr;
+  auto t0 = std::chrono::steady_clock::now();
+  std::vector<Columns> shards;
+  {
+    py::gil_scoped_release nogil;  // phases below are pure C++
+
+    auto [found, spans] = scan_items(sv);
+    if (!found) throw std::runtime_error("PodList has no items array");
+    const size_t M = spans.size();
+
+    unsigned hw = std::thread::hardware_concurrency();
+    size_t T = std::max<size_t>(1, std::min<size_t>({hw ? hw : 1, 8, (M + 4095) / 4096}));
+    shards.resize(T);
+    std::vector<std::thread> threads;
+    std::atomic<bool> failed{false};
+    std::string error_msg;
+    std::mutex err_mu;
+    for (size_t t = 0; t < T; ++t) {
+      threads.emplace_back([&, t]() {
+        const size_t lo = M * t / T, hi = M * (t + 1) / T;
+        Columns& out = shards[t];
+        out.reserve(hi - lo);
+        try {
+          Parser p(nullptr, 0);
+          for (size_t k = lo; k < hi; ++k) {
+            p.reset_to(sv.data() + spans[k].first, spans[k].second - spans[k].first);
+            const JValue* pod = p.parse_value();
+            encode_one(pod, ctx, out);
+          }
+        } catch (const std::exception& e) {
+          std::lock_guard<std::mutex> g(err_mu);
+          failed = true;
+          error_msg = e.what();
+        }
+      });
+    }
+    for (auto& th : threads) th.join();
+    if (failed) throw std::runtime_error("codec parse failed: " + error_msg);
+  }
+  auto t1 = std::chrono::steady_clock::now();
+
+  // concatenate under the GIL into numpy arrays + python lists
+  size_t P = 0;
+  for (auto& s : shards) P += s.pod_flags.size();
+  auto make = [&](auto getter, int width, auto tag) {
+    using T = decltype(tag);
+    py::array_t<T> arr(width == 1 ? std::vector<ssize_t>{(ssize_t)P}
+                                  : std::vector<ssize_t>{(ssize_t)P, (ssize_t)width});
+    T* dst = arr.mutable_data();
+    for (auto& s : shards) {
+      auto& v = getter(s);
+      std::memcpy(dst, v.data(), v.size() * sizeof(T));
+      dst += v.size();
+    }
+    return arr;
   };
   py::dict result;
-  result["pod_req"] = arr(pod_req, {P, (ssize_t)NRES});
-  result["pod_flags"] = arr(pod_flags, {P});
-  result["pod_sel_hash"] = arr(pod_sel_hash, {P});
-  result["pod_sel_pairs"] = arr(pod_sel_pairs, {P, (ssize_t)MAXSEL});
-  result["pod_sel_count"] = arr(pod_sel_count, {P});
-  result["pod_node"] = arr(pod_node, {P});
-  result["pod_pdb"] = arr(pod_pdb, {P});
-  result["pod_uids"] = uids;
-  result["pod_refs"] = refs;
+  result["pod_req"] = make([](Columns& c) -> auto& { return c.pod_req; }, NRES, int32_t{});
+  result["pod_flags"] = make([](Columns& c) -> auto& { return c.pod_flags; }, 1, uint16_t{});
+  result["pod_sel_hash"] = make([](Columns& c) -> auto& { return c.pod_sel_hash; }, 1, int64_t{});
+  result["pod_sel_pairs"] = make([](Columns& c) -> auto& { return c.pod_sel_pairs; }, MAXSEL, int64_t{});
+  result["pod_sel_count"] = make([](Columns& c) -> auto& { return c.pod_sel_count; }, 1, int8_t{});
+  result["pod_node"] = make([](Columns& c) -> auto& { return c.pod_node; }, 1, int32_t{});
+  result["pod_pdb"] = make([](Columns& c) -> auto& { return c.pod_pdb; }, 1, int32_t{});
+  PyObject* uids = PyList_New((Py_ssize_t)P);
+  PyObject* refs = PyList_New((Py_ssize_t)P);
+  Py_ssize_t w = 0;
+  for (auto& s : shards)
+    for (size_t i = 0; i < s.uids.size(); ++i, ++w) {
+      PyList_SET_ITEM(uids, w,
+                      PyUnicode_FromStringAndSize(s.uids[i].data(), (Py_ssize_t)s.uids[i].size()));
+      PyObject* tup = PyTuple_New(2);
+      PyTuple_SET_ITEM(tup, 0,
+                       PyUnicode_FromStringAndSize(s.ns[i].data(), (Py_ssize_t)s.ns[i].size()));
+      PyTuple_SET_ITEM(tup, 1,
+                       PyUnicode_FromStringAndSize(s.names[i].data(), (Py_ssize_t)s.names[i].size()));
+      PyList_SET_ITEM(refs, w, tup);
+    }
+  result["pod_uids"] = py::reinterpret_steal<py::list>(uids);
+  result["pod_refs"] = py::reinterpret_steal<py::list>(refs);
+  if (dbg) {
+    auto t2 = std::chrono::steady_clock::now();
+    fprintf(stderr, "[codec] parse %.3fs gil-build %.3fs\n",
+            std::chrono::duration<double>(t1 - t0).count(),
+            std::chrono::duration<double>(t2 - t1).count());
+  }
   return result;
 }
