@@ -54,6 +54,8 @@ void pod_pool_map(torch::Tensor slot_of_pod, torch::Tensor slot2seg,
                   torch::Tensor seg_pool, torch::Tensor pod_pool);
 pybind11::dict encode_pods_json(pybind11::bytes body, pybind11::dict node_index,
                                 pybind11::list pdb_selectors);
+void pack_into(pybind11::buffer buf, const std::vector<int64_t>& offsets,
+               pybind11::list arrays);
 void snapshot_scatter(torch::Tensor rows, torch::Tensor u_req,
                       torch::Tensor u_flags, torch::Tensor u_sel_hash,
                       torch::Tensor u_sel_pairs, torch::Tensor u_sel_count,
@@ -81,4 +83,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "PodList JSON -> pod SoA columns (native snapshot codec)");
   m.def("snapshot_scatter", &snapshot_scatter,
         "kernel E: delta-apply rows into the resident snapshot");
+  m.def("pack_into", &pack_into, "memcpy arrays into the staging buffer");
 }

@@ -451,10 +451,10 @@ struct Ctx {
 
 struct Columns {
   std::vector<int32_t> pod_req;
-  std::vector<uint16_t> pod_flags;
+  std::vector<int32_t> pod_flags;
   std::vector<int64_t> pod_sel_hash;
   std::vector<int64_t> pod_sel_pairs;
-  std::vector<int8_t> pod_sel_count;
+  std::vector<int32_t> pod_sel_count;
   std::vector<int32_t> pod_node;
   std::vector<int32_t> pod_pdb;
   std::vector<std::string> uids, ns, names;
@@ -500,11 +500,11 @@ void encode_one(const JValue* pod, const Ctx& ctx, Columns& out) {
   int cnt = (int)hashes.size();
   for (int q = 0; q < MAXSEL; ++q)
     out.pod_sel_pairs.push_back(q < cnt ? (int64_t)hashes[q] : 0);
-  out.pod_sel_count.push_back((int8_t)std::min(cnt, MAXSEL));
+  out.pod_sel_count.push_back((int32_t)std::min(cnt, MAXSEL));
   out.pod_sel_hash.push_back((int64_t)combined);
 
   // flags
-  uint16_t f = 0;
+  int32_t f = 0;
   const std::string_view phase = get_str(status, "phase");
   const std::string_view node_name = get_str(spec, "nodeName");
   if (phase == "Pending" && node_name.empty()) {
@@ -738,10 +738,10 @@ pybind11::dict encode_pods_json(pybind11::bytes body, pybind11::dict node_index,
   };
   py::dict result;
   result["pod_req"] = make([](Columns& c) -> auto& { return c.pod_req; }, NRES, int32_t{});
-  result["pod_flags"] = make([](Columns& c) -> auto& { return c.pod_flags; }, 1, uint16_t{});
+  result["pod_flags"] = make([](Columns& c) -> auto& { return c.pod_flags; }, 1, int32_t{});
   result["pod_sel_hash"] = make([](Columns& c) -> auto& { return c.pod_sel_hash; }, 1, int64_t{});
   result["pod_sel_pairs"] = make([](Columns& c) -> auto& { return c.pod_sel_pairs; }, MAXSEL, int64_t{});
-  result["pod_sel_count"] = make([](Columns& c) -> auto& { return c.pod_sel_count; }, 1, int8_t{});
+  result["pod_sel_count"] = make([](Columns& c) -> auto& { return c.pod_sel_count; }, 1, int32_t{});
   result["pod_node"] = make([](Columns& c) -> auto& { return c.pod_node; }, 1, int32_t{});
   result["pod_pdb"] = make([](Columns& c) -> auto& { return c.pod_pdb; }, 1, int32_t{});
   PyObject* uids = PyList_New((Py_ssize_t)P);
@@ -767,4 +767,28 @@ pybind11::dict encode_pods_json(pybind11::bytes body, pybind11::dict node_index,
             std::chrono::duration<double>(t2 - t1).count());
   }
   return result;
+}
+
+
+// ---------------------------------------------------------------------------
+// pack_into: memcpy a list of contiguous numpy arrays into a host
+// staging buffer at given byte offsets — the 18-field python packing
+// loop cost ~380 us/cycle at 10k pods; this is one call + 18 memcpys.
+// ---------------------------------------------------------------------------
+void pack_into(pybind11::buffer buf, const std::vector<int64_t>& offsets,
+               pybind11::list arrays) {
+  namespace py = pybind11;
+  py::buffer_info dst = buf.request(true);
+  char* base = static_cast<char*>(dst.ptr);
+  const int64_t cap = dst.size * dst.itemsize;
+  const size_t n = py::len(arrays);
+  if (offsets.size() != n) throw std::runtime_error("offsets/arrays mismatch");
+  for (size_t i = 0; i < n; ++i) {
+    py::buffer src_b = py::cast<py::buffer>(arrays[i]);
+    py::buffer_info src = src_b.request(false);
+    int64_t nb = src.size * src.itemsize;
+    // contiguity: itemsize-stride rows (ascontiguousarray upstream)
+    if (offsets[i] + nb > cap) throw std::runtime_error("pack overflow");
+    std::memcpy(base + offsets[i], src.ptr, (size_t)nb);
+  }
 }

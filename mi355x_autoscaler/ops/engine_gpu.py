@@ -116,37 +116,71 @@ _TORCH_DT = {np.int32: torch.int32, np.int64: torch.int64}
 
 class _PinnedPacker:
     """Packs every snapshot array into one pinned host buffer so the
-    upload is a single DMA (the profile showed 36 copies/cycle)."""
+    upload is a single DMA. Steady state (same array shapes as the
+    previous cycle — the common case for a controller): the device
+    buffer and its typed views are REUSED and the per-cycle cost is 18
+    host memcpys + one async H2D `copy_` (the naive per-field path
+    measured 417 us of pure Python/torch overhead at 10k pods)."""
 
     def __init__(self):
         self.buf: Optional[torch.Tensor] = None
+        self._sig = None
+        self._dev: Optional[torch.Tensor] = None
+        self._views = None
+        self._offsets = None
+        self._total = 0
+        self._ops = load_ops()
+
+    def _layout(self, snap: ClusterSnapshot):
+        offsets = []
+        off = 0
+        for name, attr, dt, tail in _FIELDS:
+            a = getattr(snap, attr)
+            nb = int(a.size) * np.dtype(dt).itemsize
+            offsets.append((name, attr, off, nb, dt, a.shape))
+            off += (nb + _ALIGN - 1) // _ALIGN * _ALIGN
+        return offsets, max(off, _ALIGN)
 
     def pack(self, snap: ClusterSnapshot, device) -> DeviceSnapshot:
-        offsets: List[Tuple[str, int, int, np.dtype, tuple]] = []
-        off = 0
-        arrays = {}
-        for name, attr, dt, tail in _FIELDS:
-            a = np.ascontiguousarray(getattr(snap, attr), dtype=dt)
-            arrays[name] = a
-            nb = a.nbytes
-            offsets.append((name, off, nb, dt, a.shape))
-            off += (nb + _ALIGN - 1) // _ALIGN * _ALIGN
-        total = max(off, _ALIGN)
-        if self.buf is None or self.buf.numel() < total:
+        offsets, total = self._layout(snap)
+        sig = tuple(shape for _, _, _, _, _, shape in offsets)
+
+        rebuild = (
+            self._sig != sig
+            or self._dev is None
+            or self._dev.device != torch.device(device)
+        )
+        # exact-size buffers: the upload is an unsliced tensor copy_
+        if self.buf is None or self.buf.numel() != total:
             pin = torch.cuda.is_available()
             self.buf = torch.empty(total, dtype=torch.uint8, pin_memory=pin)
-        host = self.buf.numpy()
-        for name, o, nb, dt, shape in offsets:
-            host[o : o + nb] = arrays[name].view(np.uint8).reshape(-1)
-        dev = self.buf[:total].to(device, non_blocking=True)
-        views = {}
-        for name, o, nb, dt, shape in offsets:
-            t = dev[o : o + nb].view(_TORCH_DT[dt])
-            views[name] = t.view(*shape) if shape else t
+            rebuild = True
+        arrays = [np.ascontiguousarray(getattr(snap, attr), dtype=dt)
+                  for name, attr, o, nb, dt, shape in offsets]
+        self._ops.pack_into(self.buf.numpy(), [o for _, _, o, _, _, _ in offsets], arrays)
+
+        is_cuda = torch.device(device).type == "cuda"
+        if rebuild:
+            # on CPU (tests) the views alias the staging buffer itself —
+            # no copy; on a GPU the upload is ONE pinned->HBM DMA
+            self._dev = (
+                torch.empty(total, dtype=torch.uint8, device=device)
+                if is_cuda else self.buf
+            )
+            self._views = {}
+            for name, attr, o, nb, dt, shape in offsets:
+                t = self._dev[o : o + nb].view(_TORCH_DT[dt])
+                self._views[name] = t.view(*shape) if shape else t
+            self._offsets = offsets
+            self._total = total
+            self._sig = sig
+        if is_cuda:
+            self._dev.copy_(self.buf, non_blocking=True)
+
         return DeviceSnapshot(
-            **views,
-            pool_max_h=arrays["pool_max"].copy(),
-            pool_cur_h=arrays["pool_cur"].copy(),
+            **self._views,
+            pool_max_h=np.ascontiguousarray(snap.pool_max, dtype=np.int32).copy(),
+            pool_cur_h=np.ascontiguousarray(snap.pool_cur, dtype=np.int32).copy(),
         )
 
 

@@ -45,10 +45,10 @@ def pack_rows(snap: ClusterSnapshot, idx: np.ndarray) -> np.ndarray:
 def unpack_rows(rows: np.ndarray):
     return dict(
         pod_req=rows[:, 0:4].astype(np.int32),
-        pod_flags=rows[:, 4].astype(np.uint16),
+        pod_flags=rows[:, 4].astype(np.int32),
         pod_sel_hash=rows[:, 5].copy(),
         pod_sel_pairs=rows[:, 6 : 6 + MAX_SELECTOR_PAIRS].copy(),
-        pod_sel_count=rows[:, 14].astype(np.int8),
+        pod_sel_count=rows[:, 14].astype(np.int32),
         pod_node=rows[:, 15].astype(np.int32),
         pod_pdb=rows[:, 16].astype(np.int32),
     )

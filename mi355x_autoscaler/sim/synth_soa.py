@@ -68,10 +68,10 @@ def synth_burst_snapshot(
     P_bound = bound_pods_per_node * len(nodes)
     P = n_pending + P_bound
     pod_req = np.zeros((P, 4), np.int32)
-    pod_flags = np.zeros(P, np.uint16)
+    pod_flags = np.zeros(P, np.int32)
     pod_sel_hash = np.zeros(P, np.int64)
     pod_sel_pairs = np.zeros((P, MAX_SELECTOR_PAIRS), np.int64)
-    pod_sel_count = np.zeros(P, np.int8)
+    pod_sel_count = np.zeros(P, np.int32)
     pod_node = np.full(P, -1, np.int32)
     pod_pdb = np.full(P, -1, np.int32)
 
@@ -83,7 +83,7 @@ def synth_burst_snapshot(
     is_gpu = shape_gpu[sh] > 0
     pod_sel_hash[:n_pending] = np.where(is_gpu, gpu_sel_i64, 0)
     pod_sel_pairs[:n_pending, 0] = np.where(is_gpu, gpu_sel_i64, 0)
-    pod_sel_count[:n_pending] = is_gpu.astype(np.int8)
+    pod_sel_count[:n_pending] = is_gpu.astype(np.int32)
     pod_flags[:n_pending] = (
         PodFlags.PENDING | PodFlags.REPLICATED | PodFlags.DRAINABLE
     )

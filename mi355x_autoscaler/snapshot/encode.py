@@ -100,18 +100,18 @@ class ClusterSnapshot:
 
     Array shapes (P pods, N nodes, K pools):
       pod_req        [P, 4] int32     quantized requests
-      pod_flags      [P]    uint16    PodFlags bits
+      pod_flags      [P]    int32     PodFlags bits
       pod_sel_hash   [P]    int64     combined selector-set hash (0 = empty)
       pod_sel_pairs  [P, MAX_SELECTOR_PAIRS] int64  sorted, 0-padded
-      pod_sel_count  [P]    int8
+      pod_sel_count  [P]    int32
       pod_node       [P]    int32     bound node index or -1
       pod_pdb        [P]    int32     PDB index or -1
       node_alloc     [N, 4] int32     quantized allocatable
       node_free      [N, 4] int32     alloc minus bound pods (filled by
                                       compute_node_free / kernel)
       node_labels    [N, MAX_LABEL_PAIRS] int64 sorted, 0-padded
-      node_label_count [N]  int8
-      node_unsched   [N]    uint8
+      node_label_count [N]  int32
+      node_unsched   [N]    int32
       node_pool      [N]    int32     pool index or -1
       pool_unit      [K, 4] int32     capacity of ONE new node
       pool_label_*               prototype labels of a NEW pool node
@@ -195,17 +195,17 @@ def encode_snapshot(
     K = len(pool_names)
 
     pod_req = np.zeros((P, NUM_RESOURCES), np.int32)
-    pod_flags = np.zeros(P, np.uint16)
+    pod_flags = np.zeros(P, np.int32)
     pod_sel_hash = np.zeros(P, np.int64)
     pod_sel_pairs = np.zeros((P, MAX_SELECTOR_PAIRS), np.int64)
-    pod_sel_count = np.zeros(P, np.int8)
+    pod_sel_count = np.zeros(P, np.int32)
     pod_node = np.full(P, -1, np.int32)
     pod_pdb = np.full(P, -1, np.int32)
 
     node_alloc = np.zeros((N, NUM_RESOURCES), np.int32)
     node_labels = np.zeros((N, MAX_LABEL_PAIRS), np.int64)
-    node_label_count = np.zeros(N, np.int8)
-    node_unsched = np.zeros(N, np.uint8)
+    node_label_count = np.zeros(N, np.int32)
+    node_unsched = np.zeros(N, np.int32)
     node_pool = np.full(N, -1, np.int32)
 
     node_idx = {n.name: i for i, n in enumerate(nodes)}
@@ -262,7 +262,7 @@ def encode_snapshot(
     pool_max = np.zeros(K, np.int32)
     pool_cur = np.zeros(K, np.int32)
     pool_labels = np.zeros((K, MAX_LABEL_PAIRS), np.int64)
-    pool_label_count = np.zeros(K, np.int8)
+    pool_label_count = np.zeros(K, np.int32)
     for name, k in pool_index.items():
         pool = pools[name]
         unit = pool.unit_capacity
