@@ -29,10 +29,8 @@ import argparse
 import json
 import os
 import statistics
-import sys
 import time
 
-import numpy as np
 import torch
 
 from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools

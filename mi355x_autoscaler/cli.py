@@ -12,7 +12,6 @@ MI355X-native additions namespaced as new flags (--backend, --gpus,
 from __future__ import annotations
 
 import logging
-import sys
 
 import click
 

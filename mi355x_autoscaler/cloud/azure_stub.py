@@ -13,9 +13,9 @@ BASELINE.json sanctions.
 from __future__ import annotations
 
 import logging
-from typing import Dict, Optional
+from typing import Dict
 
-from .provider import CloudProvider, ScaleEvent
+from .provider import CloudProvider
 from .template_processing import prepare_scale_deployment
 
 logger = logging.getLogger(__name__)

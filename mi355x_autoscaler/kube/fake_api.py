@@ -17,7 +17,7 @@ import threading
 import time
 from typing import Dict, List, Optional
 
-from fastapi import FastAPI, Request, Response
+from fastapi import FastAPI, Request
 from fastapi.responses import JSONResponse
 
 

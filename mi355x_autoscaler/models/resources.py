@@ -19,7 +19,7 @@ MI355X-native deltas vs the reference:
 
 from __future__ import annotations
 
-from typing import Dict, Iterable, Mapping
+from typing import Iterable, Mapping
 
 # Canonical internal axes, in fixed order. This order is ALSO the
 # resource-row order of the device SoA snapshot (snapshot/encode.py) and

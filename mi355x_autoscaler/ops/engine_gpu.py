@@ -33,14 +33,14 @@ from __future__ import annotations
 
 import os
 from dataclasses import dataclass
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 import numpy as np
 import torch
 
 from ..sim.decision import DrainCandidate, MaintainReport, ScaleDecision
 from ..sim.engine import SimulationEngine
-from ..snapshot.encode import ClusterSnapshot, PodFlags
+from ..snapshot.encode import ClusterSnapshot
 from . import load_ops
 
 _PACK_MAX = 1 << 56

@@ -12,7 +12,7 @@ rebalancing (row H, chunked).
 from __future__ import annotations
 
 import os
-from typing import List, Optional
+
 
 import numpy as np
 import torch

@@ -22,7 +22,7 @@ at most one cycle. The recovery contract:
 
 from __future__ import annotations
 
-from typing import List, Sequence
+from typing import Sequence
 
 import numpy as np
 

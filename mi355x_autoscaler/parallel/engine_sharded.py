@@ -28,14 +28,14 @@ world_size 2) and nccl/RCCL (GPU engine on MI355X).
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import List
 
 import numpy as np
 import torch
 
 from ..sim.decision import MaintainReport, ScaleDecision, Segment
 from ..sim.engine import CpuEngine, SimulationEngine
-from ..sim.oracle import OracleSimulator, _pack_req_key, build_segments
+from ..sim.oracle import OracleSimulator, build_segments
 from ..snapshot.encode import MAX_SELECTOR_PAIRS, ClusterSnapshot
 from . import dist as pdist
 

@@ -17,7 +17,7 @@ invariance is what tests/test_rebalance.py asserts.
 
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import numpy as np
 import torch

@@ -10,10 +10,10 @@ snapshot — that property is CI-tested.
 
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import Optional
 
 from ..snapshot.encode import ClusterSnapshot
-from .decision import MaintainReport, ScaleDecision, Segment
+from .decision import MaintainReport, ScaleDecision
 from .oracle import OracleSimulator, build_segments
 
 

@@ -6,7 +6,7 @@ SAME hashes/flags the JSON path would produce (asserted in tests)."""
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Sequence, Tuple
 
 import numpy as np
 

@@ -12,9 +12,8 @@ is not built (CPU-only dev boxes).
 from __future__ import annotations
 
 import json
-from typing import Dict, List, Optional, Sequence
+from typing import Dict, Optional, Sequence
 
-import numpy as np
 
 from ..models.agent_pool import AgentPool
 from ..models.kube import KubeNode

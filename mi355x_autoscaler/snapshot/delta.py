@@ -15,7 +15,7 @@ to an .npz so a 1M-pod replay can resume mid-burst.
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Sequence, Set, Tuple
+from typing import Dict, List, Optional, Set, Tuple
 
 import numpy as np
 

@@ -13,7 +13,7 @@ command-line flags always win over file values.
 
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import Optional
 
 from pydantic import BaseModel, Field, field_validator
 
