@@ -31,10 +31,17 @@ ext = CUDAExtension(
         "mi355x_autoscaler/ops/hip/snapshot_diff.hip",
     ],
     extra_compile_args={
-        "cxx": ["-O3", "-std=c++17"],
-        "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+        "cxx": ["-O3", "-std=c++17"]
+        + (["-fsanitize=address", "-fno-omit-frame-pointer"]
+           if os.environ.get("MI355X_AS_ASAN") == "1" else []),
+        "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"]
+        + (["-fsanitize=address", "-fno-omit-frame-pointer"]
+           if os.environ.get("MI355X_AS_ASAN") == "1" else []),
     },
 )
+# MI355X_AS_ASAN=1: host-side AddressSanitizer build of the extension
+# (SURVEY §5 race-detection plan); run pytest with
+# LD_PRELOAD=$(gcc -print-file-name=libasan.so).
 
 setup(
     name="mi355x_autoscaler_ops",

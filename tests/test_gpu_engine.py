@@ -348,3 +348,20 @@ def test_shape_class_sweep(engine, n_pods, n_nodes):
     )
     snap.pool_max[:] = 1 << 20
     _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))
+
+
+def test_decide_is_deterministic_across_repeats(engine):
+    """Race screen: the lock-free hash build + atomics must produce the
+    SAME decision on every repeat (integer ops commute exactly; slot
+    assignment varies but the sorted segment list must not)."""
+    pods_json, nodes_json = burst_trace(
+        20000, n_nodes_per_pool=(40, 30, 30),
+        pools=(("agentpool1", "Standard_D2_v2"), ("bigpool", "Standard_D4_v2"),
+               ("gpupool", "Standard_ND96_MI355X_v1")),
+        gpu_fraction=0.35, seed=33, bound_pods_per_node=2,
+    )
+    snap = _snap(pods_json, nodes_json)
+    first = engine.decide(snap)
+    for _ in range(15):
+        d = engine.decide(snap)
+        _assert_decisions_equal(d, first)

@@ -19,7 +19,7 @@ from mi355x_autoscaler.snapshot.encode import encode_snapshot
 WORLD = 2
 
 
-def _make_snap(rank=None, seed=11, n_pods=400):
+def _make_snap(rank=None, seed=11, n_pods=400, world=WORLD):
     pods_json, nodes_json = burst_trace(
         n_pods,
         n_nodes_per_pool=(5, 2),
@@ -30,23 +30,23 @@ def _make_snap(rank=None, seed=11, n_pods=400):
     )
     if rank is not None:
         # shard pods round-robin; keep every bound pod on its rank too
-        pods_json = pods_json[rank::WORLD]
+        pods_json = pods_json[rank::world]
     pods, nodes = wrap(pods_json, nodes_json)
     pools = group_nodes_into_pools(nodes)
     return encode_snapshot(pods, nodes, pools)
 
 
-def _worker(rank, q, mode):
+def _worker(rank, q, mode, world=WORLD, port="29617"):
     os.environ.update(
-        WORLD_SIZE=str(WORLD), RANK=str(rank), LOCAL_RANK=str(rank),
-        MASTER_ADDR="127.0.0.1", MASTER_PORT="29617",
+        WORLD_SIZE=str(world), RANK=str(rank), LOCAL_RANK=str(rank),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=port,
     )
     import torch.distributed as dist
 
     from mi355x_autoscaler.parallel.engine_sharded import ShardedEngine
 
     engine = ShardedEngine(CpuEngine())
-    snap = _make_snap(rank=rank)
+    snap = _make_snap(rank=rank, world=world)
     try:
         if mode == "decide":
             d = engine.decide(snap)
@@ -65,15 +65,15 @@ def _worker(rank, q, mode):
         dist.destroy_process_group()
 
 
-def _run_world(mode):
+def _run_world(mode, world=WORLD, port="29617"):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_worker, args=(r, q, mode)) for r in range(WORLD)]
+    procs = [ctx.Process(target=_worker, args=(r, q, mode, world, port)) for r in range(world)]
     for p in procs:
         p.start()
     results = {}
-    for _ in range(WORLD):
-        rank, blob = q.get(timeout=120)
+    for _ in range(world):
+        rank, blob = q.get(timeout=180)
         results[rank] = pickle.loads(blob)
     for p in procs:
         p.join(timeout=30)
@@ -122,3 +122,16 @@ def test_merge_rows_vectorized_equals_reference():
     for a, b in zip(merged, ref):
         assert a.sel_hash == b.sel_hash and a.count == b.count
         assert (a.req == b.req).all()
+
+
+def test_sharded_decide_world4():
+    """The driver's 8-GPU run uses world sizes we cannot reach here;
+    world 4 over gloo covers the >2-rank merge/agreement paths."""
+    results = _run_world("decide", world=4, port="29653")
+    assert results[0] == results[1] == results[2] == results[3]
+    snap = _make_snap(rank=None)
+    from mi355x_autoscaler.sim.engine import CpuEngine as _C
+
+    d = _C().decide(snap)
+    assert results[0]["units"] == d.units_needed
+    assert results[0]["pe"] == d.pods_placed_existing
