@@ -74,6 +74,10 @@ class ClusterOptions:
     dry_run: bool = False
     drain_grace_seconds: int = 30
     max_drains_per_cycle: int = 4
+    # suppress scale-down while a scale-up is in flight (the reference
+    # ran both every cycle, which oscillates at low idle thresholds —
+    # kubernetes/cluster-autoscaler's behavior is the safer default)
+    scale_down_during_scale_up: bool = False
     # ingest pods via the C++ snapshot codec (JSON bytes -> SoA) when
     # the extension is built; the Python path is the fallback/spec.
     native_codec: bool = True
@@ -192,7 +196,8 @@ class Cluster:
             self.notifier.notify_failed_to_scale(sel_hash, count)
 
         # ---- scale-down ------------------------------------------------
-        if not self.opts.no_maintenance:
+        suppress = decision.wants_scale_up() and not self.opts.scale_down_during_scale_up
+        if not self.opts.no_maintenance and not suppress:
             with t.phase("maintain"):
                 drained = self.maintain(snap, pools, nodes, now=now)
             result.drained_nodes = drained
