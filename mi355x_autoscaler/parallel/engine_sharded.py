@@ -155,16 +155,21 @@ class ShardedEngine(SimulationEngine):
 
     @staticmethod
     def _device_segments_to_rows(ds, segs) -> np.ndarray:
+        """Serialize the segment SoA into host rows with ONE device->host
+        copy (a per-column .cpu() costs a sync each; this path runs every
+        sharded decide)."""
         G = segs.G
-        rows = np.zeros((G, _ROW_COLS), np.int64)
-        if G:
-            rows[:, :MAX_SELECTOR_PAIRS] = segs.seg_sel_pairs.cpu().numpy()
-            rows[:, MAX_SELECTOR_PAIRS] = segs.seg_sel_count.cpu().numpy()
-            rows[:, MAX_SELECTOR_PAIRS + 1 : MAX_SELECTOR_PAIRS + 5] = segs.seg_req.cpu().numpy()
-            rows[:, MAX_SELECTOR_PAIRS + 5] = segs.seg_count.cpu().numpy()
-            rows[:, MAX_SELECTOR_PAIRS + 6] = segs.seg_sel_hash.cpu().numpy()
-            rows[:, MAX_SELECTOR_PAIRS + 7] = segs.seg_key.cpu().numpy()
-        return rows
+        if not G:
+            return np.zeros((0, _ROW_COLS), np.int64)
+        dev_rows = torch.empty((G, _ROW_COLS), dtype=torch.int64,
+                               device=segs.seg_req.device)
+        dev_rows[:, :MAX_SELECTOR_PAIRS] = segs.seg_sel_pairs
+        dev_rows[:, MAX_SELECTOR_PAIRS] = segs.seg_sel_count.to(torch.int64)
+        dev_rows[:, MAX_SELECTOR_PAIRS + 1 : MAX_SELECTOR_PAIRS + 5] = segs.seg_req.to(torch.int64)
+        dev_rows[:, MAX_SELECTOR_PAIRS + 5] = segs.seg_count
+        dev_rows[:, MAX_SELECTOR_PAIRS + 6] = segs.seg_sel_hash
+        dev_rows[:, MAX_SELECTOR_PAIRS + 7] = segs.seg_key
+        return dev_rows.cpu().numpy()
 
     @staticmethod
     def _rows_to_device_segments(rows: np.ndarray, device):
