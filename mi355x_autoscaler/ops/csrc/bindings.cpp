@@ -56,6 +56,24 @@ pybind11::dict encode_pods_json(pybind11::bytes body, pybind11::dict node_index,
                                 pybind11::list pdb_selectors);
 void pack_into(pybind11::buffer buf, const std::vector<int64_t>& offsets,
                pybind11::list arrays);
+void pool_max_new(torch::Tensor pool_max, torch::Tensor pool_cur,
+                  torch::Tensor out);
+void fused_decide_launch(
+    torch::Tensor pod_req, torch::Tensor pod_flags, torch::Tensor pod_sel_hash,
+    torch::Tensor pod_sel_pairs, torch::Tensor pod_sel_count,
+    torch::Tensor pod_node, torch::Tensor node_alloc, torch::Tensor node_labels,
+    torch::Tensor node_label_count, torch::Tensor node_unsched,
+    torch::Tensor pool_unit, torch::Tensor pool_labels,
+    torch::Tensor pool_label_count, torch::Tensor pool_max, torch::Tensor pool_cur,
+    torch::Tensor ws_i32, torch::Tensor slot_ks, torch::Tensor slot_kr,
+    torch::Tensor slot2seg,
+    torch::Tensor seg_sel_pairs, torch::Tensor seg_sel_count,
+    torch::Tensor seg_sel_hash, torch::Tensor seg_req, torch::Tensor seg_count,
+    torch::Tensor seg_key, torch::Tensor seg_first,
+    torch::Tensor slot_of_pod, torch::Tensor free, torch::Tensor elig,
+    torch::Tensor seg_pool, torch::Tensor max_new, torch::Tensor run_scratch,
+    torch::Tensor blob, torch::Tensor pod_pool, int64_t hash_slots,
+    int64_t seg_max, int64_t want_pod_pool);
 void snapshot_scatter(torch::Tensor rows, torch::Tensor u_req,
                       torch::Tensor u_flags, torch::Tensor u_sel_hash,
                       torch::Tensor u_sel_pairs, torch::Tensor u_sel_count,
@@ -84,4 +102,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("snapshot_scatter", &snapshot_scatter,
         "kernel E: delta-apply rows into the resident snapshot");
   m.def("pack_into", &pack_into, "memcpy arrays into the staging buffer");
+  m.def("pool_max_new", &pool_max_new, "clamp(pool_max - pool_cur, 0)");
+  m.def("fused_decide_launch", &fused_decide_launch,
+        "enqueue the whole scale-up pipeline in one call");
 }

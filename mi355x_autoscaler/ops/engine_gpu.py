@@ -328,6 +328,8 @@ class GpuEngine(SimulationEngine):
         self.ops = load_ops()
         self.packer = _PinnedPacker()
         self.force_torch_segbuild = os.environ.get("MI355X_AS_FORCE_TORCH_SEGBUILD") == "1"
+        self.force_py_pipeline = os.environ.get("MI355X_AS_PY_PIPELINE") == "1"
+        self._free_cache: Dict[int, torch.Tensor] = {}
         self._last_demand: Optional[torch.Tensor] = None
         d = self.device
         # persistent hash workspace; the mutable i32 parts live in ONE
@@ -422,52 +424,47 @@ class GpuEngine(SimulationEngine):
     def _decide_fused(self, ds: DeviceSnapshot, pool_names: List[str],
                       over_provision: int = 0) -> Optional[ScaleDecision]:
         """The zero-mid-sync decision pipeline: every kernel reads G
-        from device memory; ONE sync at the result readback. Returns
-        None on hash/sort overflow (caller uses the torch path)."""
+        from device memory; ONE sync at the result readback. The whole
+        launch sequence is enqueued by the native orchestrator
+        (ops/csrc/orchestrator.cpp) in a single Python->C++ call;
+        MI355X_AS_PY_PIPELINE=1 selects the op-by-op Python sequencing
+        (differential-tested equal). Returns None on hash/sort overflow
+        (caller falls back to the torch path)."""
         ops = self.ops
         device = self.device
         N, K = ds.N, ds.K
-        g_dev = self.ws_stat[0:1]
-
-        slot_of_pod = self._enqueue_hash_segments(ds)
-
-        free = ds.node_alloc.clone()
-        ops.node_free_subtract(ds.pod_node, ds.pod_req, free)
 
         W = (N + 63) // 64
         elig = self._elig_buffer(W)
-        if N > 0:
-            ops.fit_eligibility(ds.node_labels, ds.node_label_count,
-                                ds.node_unsched, self.ws_seg_sel_pairs,
-                                self.ws_seg_sel_count, g_dev, elig)
-        seg_pool = self.ws_seg_pool
-        ops.seg_pool_route(self.ws_seg_sel_pairs, self.ws_seg_sel_count,
-                           self.ws_seg_req, ds.pool_labels, ds.pool_label_count,
-                           ds.pool_unit, g_dev, seg_pool)
+        slot_of_pod = torch.empty(ds.P, dtype=torch.int32, device=device)
+        blob = torch.empty(3 * SEG_MAX + 5 * max(K, 1), dtype=torch.int64, device=device)
+        max_new = torch.empty(max(K, 1), dtype=torch.int32, device=device)
+        want_pod_pool = bool(K and K <= 16)
+        pod_pool = torch.empty(ds.P if want_pod_pool else 0, dtype=torch.int32,
+                               device=device)
+
+        if self.force_py_pipeline:
+            self._enqueue_py(ds, slot_of_pod, elig, blob, max_new, pod_pool,
+                             want_pod_pool)
+        else:
+            ops.fused_decide_launch(
+                ds.pod_req, ds.pod_flags, ds.pod_sel_hash, ds.pod_sel_pairs,
+                ds.pod_sel_count, ds.pod_node, ds.node_alloc, ds.node_labels,
+                ds.node_label_count, ds.node_unsched, ds.pool_unit,
+                ds.pool_labels, ds.pool_label_count, ds.pool_max, ds.pool_cur,
+                self._ws_i32, self.ws_slot_ks, self.ws_slot_kr, self.ws_slot2seg,
+                self.ws_seg_sel_pairs, self.ws_seg_sel_count, self.ws_seg_sel_hash,
+                self.ws_seg_req, self.ws_seg_count, self.ws_seg_key, self.ws_seg_first,
+                slot_of_pod, self._free_buffer(N), elig, self.ws_seg_pool, max_new,
+                self.ws_run_scratch, blob, pod_pool, HASH_SLOTS, SEG_MAX,
+                1 if want_pod_pool else 0,
+            )
 
         # MFMA per-pool demand (zero-sort path; skipped for >16 pools)
-        if K and K <= 16:
-            pod_pool = torch.empty(ds.P, dtype=torch.int32, device=device)
-            ops.pod_pool_map(slot_of_pod, self.ws_slot2seg, seg_pool, pod_pool)
+        if want_pod_pool:
             self._last_demand = segsum(ds.pod_req, pod_pool, K, ops)
         else:
             self._last_demand = None
-
-        blob = torch.zeros(3 * SEG_MAX + 5 * max(K, 1), dtype=torch.int64, device=device)
-        placed = blob[0:SEG_MAX]
-        placed_new = blob[SEG_MAX : 2 * SEG_MAX]
-        unful = blob[2 * SEG_MAX : 3 * SEG_MAX]
-        units_cnt = blob[3 * SEG_MAX : 3 * SEG_MAX + K]
-        pool_residual = blob[3 * SEG_MAX + K : 3 * SEG_MAX + 5 * K]
-        if N > 0:
-            ops.binpack_place(free, elig, self.ws_seg_req, self.ws_seg_count,
-                              g_dev, placed)
-        max_new = torch.clamp(ds.pool_max - ds.pool_cur, min=0)
-        if K:
-            ops.binpack_units(seg_pool, self.ws_seg_req, self.ws_seg_count,
-                              placed, ds.pool_unit, max_new, g_dev,
-                              self.ws_run_scratch, units_cnt, placed_new,
-                              unful, pool_residual)
 
         # THE sync: one blob + the 3-word stat
         blob_h = blob.cpu().numpy()
@@ -490,6 +487,62 @@ class GpuEngine(SimulationEngine):
             max_new_h, ds.pool_cur_h, K, pool_names, over_provision,
             lambda: self.ws_seg_sel_hash[:G].cpu().numpy(),
         )
+
+    def _free_buffer(self, N: int) -> torch.Tensor:
+        buf = self._free_cache.get(N)
+        if buf is None:
+            buf = torch.empty((max(N, 1), 4), dtype=torch.int32, device=self.device)
+            self._free_cache = {N: buf}
+        return buf
+
+    def _enqueue_py(self, ds, slot_of_pod, elig, blob, max_new, pod_pool,
+                    want_pod_pool):
+        """Op-by-op Python sequencing of the same pipeline (reference
+        for the native orchestrator; also exercised in tests)."""
+        ops = self.ops
+        self._ws_i32.zero_()
+        blob.zero_()
+        ops.segment_hash(
+            ds.pod_flags, ds.pod_sel_hash, ds.pod_req,
+            self.ws_slot_ks, self.ws_slot_kr, self.ws_slot_state,
+            self.ws_slot_count, self.ws_slot_first, slot_of_pod, self.ws_stat[2:3],
+        )
+        ops.segment_compact(
+            self.ws_slot_ks, self.ws_slot_kr, self.ws_slot_state,
+            self.ws_slot_count, self.ws_slot_first,
+            ds.pod_sel_pairs, ds.pod_sel_count, ds.pod_req,
+            self.ws_seg_sel_pairs, self.ws_seg_sel_count, self.ws_seg_sel_hash,
+            self.ws_seg_req, self.ws_seg_count, self.ws_seg_key,
+            self.ws_seg_first, self.ws_slot2seg, self.ws_stat[0:2],
+        )
+        g_dev = self.ws_stat[0:1]
+        N, K = ds.N, ds.K
+        free = self._free_buffer(N)
+        free.copy_(ds.node_alloc)
+        ops.node_free_subtract(ds.pod_node, ds.pod_req, free)
+        if N > 0:
+            ops.fit_eligibility(ds.node_labels, ds.node_label_count,
+                                ds.node_unsched, self.ws_seg_sel_pairs,
+                                self.ws_seg_sel_count, g_dev, elig)
+        ops.seg_pool_route(self.ws_seg_sel_pairs, self.ws_seg_sel_count,
+                           self.ws_seg_req, ds.pool_labels, ds.pool_label_count,
+                           ds.pool_unit, g_dev, self.ws_seg_pool)
+        if want_pod_pool:
+            ops.pod_pool_map(slot_of_pod, self.ws_slot2seg, self.ws_seg_pool, pod_pool)
+        placed = blob[0:SEG_MAX]
+        placed_new = blob[SEG_MAX : 2 * SEG_MAX]
+        unful = blob[2 * SEG_MAX : 3 * SEG_MAX]
+        if N > 0:
+            ops.binpack_place(free, elig, self.ws_seg_req, self.ws_seg_count,
+                              g_dev, placed)
+        if K:
+            units_cnt = blob[3 * SEG_MAX : 3 * SEG_MAX + K]
+            pool_residual = blob[3 * SEG_MAX + K : 3 * SEG_MAX + 5 * K]
+            ops.pool_max_new(ds.pool_max, ds.pool_cur, max_new)
+            ops.binpack_units(self.ws_seg_pool, self.ws_seg_req, self.ws_seg_count,
+                              placed, ds.pool_unit, max_new, g_dev,
+                              self.ws_run_scratch, units_cnt, placed_new,
+                              unful, pool_residual)
 
     @staticmethod
     def _finish_decision(decision, placed_h, placed_new_h, unful_h, units_h,

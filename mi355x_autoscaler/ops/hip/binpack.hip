@@ -281,6 +281,18 @@ extern "C" __global__ __launch_bounds__(WAVE) void units_kernel(
   for (int q = lane; q < K * NRES; q += WAVE) pool_residual[q] = resid[q];
 }
 
+// pool_max_new = clamp(pool_max - pool_cur, 0) — device-side so the
+// orchestrated pipeline never syncs for it.
+extern "C" __global__ void pool_max_new_kernel(
+    const int32_t* __restrict__ pool_max, const int32_t* __restrict__ pool_cur,
+    int32_t K, int32_t* __restrict__ out) {
+  const int k = blockIdx.x * blockDim.x + threadIdx.x;
+  if (k < K) {
+    const int32_t d = pool_max[k] - pool_cur[k];
+    out[k] = d > 0 ? d : 0;
+  }
+}
+
 #include <torch/extension.h>
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
@@ -320,4 +332,14 @@ void binpack_units(torch::Tensor seg_pool, torch::Tensor seg_req,
                      run_scratch.data_ptr<int64_t>(), units_cnt.data_ptr<int64_t>(),
                      placed_new.data_ptr<int64_t>(), unful.data_ptr<int64_t>(),
                      pool_residual.data_ptr<int64_t>());
+}
+
+void pool_max_new(torch::Tensor pool_max, torch::Tensor pool_cur,
+                  torch::Tensor out) {
+  const int K = (int)pool_max.size(0);
+  if (K == 0) return;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(pool_max_new_kernel, dim3((K + 63) / 64), dim3(64), 0,
+                     stream.stream(), pool_max.data_ptr<int32_t>(),
+                     pool_cur.data_ptr<int32_t>(), K, out.data_ptr<int32_t>());
 }

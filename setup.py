@@ -21,6 +21,7 @@ ext = CUDAExtension(
     sources=[
         "mi355x_autoscaler/ops/csrc/bindings.cpp",
         "mi355x_autoscaler/ops/csrc/snapshot_codec.cpp",
+        "mi355x_autoscaler/ops/csrc/orchestrator.cpp",
         "mi355x_autoscaler/ops/hip/classify.hip",
         "mi355x_autoscaler/ops/hip/node_free.hip",
         "mi355x_autoscaler/ops/hip/fit_predicates.hip",

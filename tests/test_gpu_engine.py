@@ -365,3 +365,23 @@ def test_decide_is_deterministic_across_repeats(engine):
     for _ in range(15):
         d = engine.decide(snap)
         _assert_decisions_equal(d, first)
+
+
+def test_native_orchestrator_equals_python_pipeline(engine):
+    """The one-call C++ orchestrator and the op-by-op Python sequencing
+    of the fused pipeline must produce identical decisions."""
+    pods_json, nodes_json = burst_trace(
+        8000, n_nodes_per_pool=(40, 30, 30),
+        pools=(("agentpool1", "Standard_D2_v2"), ("bigpool", "Standard_D4_v2"),
+               ("gpupool", "Standard_ND96_MI355X_v1")),
+        gpu_fraction=0.35, seed=44, bound_pods_per_node=2,
+    )
+    snap = _snap(pods_json, nodes_json)
+    d_native = engine.decide(snap)
+    try:
+        engine.force_py_pipeline = True
+        d_py = engine.decide(snap)
+    finally:
+        engine.force_py_pipeline = False
+    _assert_decisions_equal(d_native, d_py)
+    _assert_decisions_equal(d_native, CpuEngine().decide(snap))
