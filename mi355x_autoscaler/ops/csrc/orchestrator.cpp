@@ -125,4 +125,9 @@ void fused_decide_launch(
                   g_dev, run_scratch, units_cnt, placed_new, unful,
                   pool_residual);
   }
+  // piggyback the 3-word stat (G, sort-ovf, hash-ovf) onto the blob
+  // tail so the host needs exactly ONE readback
+  hipMemcpyAsync(blob.data_ptr<int64_t>() + (blob.numel() - 2),
+                 stat.data_ptr<int32_t>(), 3 * sizeof(int32_t),
+                 hipMemcpyDeviceToDevice, stream);
 }

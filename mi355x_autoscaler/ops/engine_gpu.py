@@ -44,6 +44,7 @@ from ..snapshot.encode import ClusterSnapshot
 from . import load_ops
 
 _PACK_MAX = 1 << 56
+_FASTPATH_WI: dict = {}  # (device, M) -> cached segsum chunk work items
 _SEGSUM_CHUNK = 256  # exactness bound for the f32 MFMA limb chains
 SEG_MAX = 2048
 HASH_SLOTS = 8192
@@ -303,10 +304,16 @@ def segsum(values: torch.Tensor, seg_of: torch.Tensor, num_segments: int, ops,
         return out
     if num_segments <= 16:
         T = (M + _SEGSUM_CHUNK - 1) // _SEGSUM_CHUNK
-        wi_win = torch.zeros(T, dtype=torch.int32, device=device)
-        wi_start = torch.arange(T, dtype=torch.int64, device=device) * _SEGSUM_CHUNK
-        wi_end = torch.minimum(wi_start + _SEGSUM_CHUNK,
-                               torch.tensor(M, dtype=torch.int64, device=device))
+        cached = _FASTPATH_WI.get((device, M))
+        if cached is None:
+            wi_win = torch.zeros(T, dtype=torch.int32, device=device)
+            wi_start = torch.arange(T, dtype=torch.int64, device=device) * _SEGSUM_CHUNK
+            wi_end = torch.minimum(wi_start + _SEGSUM_CHUNK,
+                                   torch.tensor(M, dtype=torch.int64, device=device))
+            _FASTPATH_WI.clear()
+            _FASTPATH_WI[(device, M)] = (wi_win, wi_start, wi_end)
+        else:
+            wi_win, wi_start, wi_end = cached
     else:
         assert sorted_ids, "segsum with >16 segments needs sorted ids"
         wi_win, wi_start, wi_end = segsum_workitems(seg_of, num_segments)
@@ -437,7 +444,9 @@ class GpuEngine(SimulationEngine):
         W = (N + 63) // 64
         elig = self._elig_buffer(W)
         slot_of_pod = torch.empty(ds.P, dtype=torch.int32, device=device)
-        blob = torch.empty(3 * SEG_MAX + 5 * max(K, 1), dtype=torch.int64, device=device)
+        # +2 tail slots: the orchestrator copies the 3-word i32 stat there
+        blob = torch.empty(3 * SEG_MAX + 5 * max(K, 1) + 2, dtype=torch.int64,
+                           device=device)
         max_new = torch.empty(max(K, 1), dtype=torch.int32, device=device)
         want_pod_pool = bool(K and K <= 16)
         pod_pool = torch.empty(ds.P if want_pod_pool else 0, dtype=torch.int32,
@@ -466,9 +475,12 @@ class GpuEngine(SimulationEngine):
         else:
             self._last_demand = None
 
-        # THE sync: one blob + the 3-word stat
+        # THE sync: one readback (stat rides on the blob tail)
         blob_h = blob.cpu().numpy()
-        stat_h = self.ws_stat.cpu().numpy()
+        if self.force_py_pipeline:
+            stat_h = self.ws_stat.cpu().numpy()
+        else:
+            stat_h = blob_h[-2:].view(np.int32)
         G, sort_ovf, hash_ovf = int(stat_h[0]), int(stat_h[1]), int(stat_h[2])
         if sort_ovf or hash_ovf:
             return None
