@@ -176,7 +176,20 @@ class Cluster:
         # ---- scale-up --------------------------------------------------
         t_decide0 = time.perf_counter()
         with t.phase("decide"):
-            decision = self.engine.decide(snap, over_provision=self.opts.over_provision)
+            ws = self._watch_state
+            if (
+                self.opts.watch
+                and ws is not None
+                and hasattr(self.engine, "decide_resident")
+            ):
+                # resident-controller path: pod SoA stays in HBM; watch
+                # deltas scatter in (kernel E), node/pool side re-uploads
+                base = encode_snapshot([], ws["nodes"], ws["pools"], ws["pdbs"])
+                decision = self.engine.decide_resident(
+                    ws["cache"], base, over_provision=self.opts.over_provision
+                )
+            else:
+                decision = self.engine.decide(snap, over_provision=self.opts.over_provision)
         self.metrics.observe_latency(time.perf_counter() - t_decide0)
         result.decision = decision
         from ..snapshot.encode import PodFlags
@@ -244,10 +257,12 @@ class Cluster:
             self._watch_state = {
                 "cache": cache, "q": q, "thread": th,
                 "node_names": node_names, "broken": False,
+                "nodes": nodes, "pools": pools, "pdbs": pdbs,
             }
             return snap
 
         cache = ws["cache"]
+        ws["nodes"], ws["pools"], ws["pdbs"] = nodes, pools, pdbs
         applied = 0
         while True:
             try:

@@ -90,8 +90,8 @@ class DeviceSnapshot:
         return int(self.pool_unit.shape[0])
 
 
-_FIELDS = [
-    # (name, source attr, np dtype, trailing shape)
+# (name, source attr, np dtype, trailing shape)
+_POD_FIELDS = [
     ("pod_req", "pod_req", np.int32, (4,)),
     ("pod_flags", "pod_flags", np.int32, ()),
     ("pod_sel_hash", "pod_sel_hash", np.int64, ()),
@@ -99,6 +99,8 @@ _FIELDS = [
     ("pod_sel_count", "pod_sel_count", np.int32, ()),
     ("pod_node", "pod_node", np.int32, ()),
     ("pod_pdb", "pod_pdb", np.int32, ()),
+]
+_REST_FIELDS = [
     ("node_alloc", "node_alloc", np.int32, (4,)),
     ("node_labels", "node_labels", np.int64, (32,)),
     ("node_label_count", "node_label_count", np.int32, ()),
@@ -111,6 +113,7 @@ _FIELDS = [
     ("pool_label_count", "pool_label_count", np.int32, ()),
     ("pdb_allowed", "pdb_allowed", np.int32, ()),
 ]
+_FIELDS = _POD_FIELDS + _REST_FIELDS
 
 _TORCH_DT = {np.int32: torch.int32, np.int64: torch.int64}
 
@@ -123,7 +126,8 @@ class _PinnedPacker:
     host memcpys + one async H2D `copy_` (the naive per-field path
     measured 417 us of pure Python/torch overhead at 10k pods)."""
 
-    def __init__(self):
+    def __init__(self, fields=None):
+        self.fields = fields if fields is not None else _FIELDS
         self.buf: Optional[torch.Tensor] = None
         self._sig = None
         self._dev: Optional[torch.Tensor] = None
@@ -135,14 +139,14 @@ class _PinnedPacker:
     def _layout(self, snap: ClusterSnapshot):
         offsets = []
         off = 0
-        for name, attr, dt, tail in _FIELDS:
+        for name, attr, dt, tail in self.fields:
             a = getattr(snap, attr)
             nb = int(a.size) * np.dtype(dt).itemsize
             offsets.append((name, attr, off, nb, dt, a.shape))
             off += (nb + _ALIGN - 1) // _ALIGN * _ALIGN
         return offsets, max(off, _ALIGN)
 
-    def pack(self, snap: ClusterSnapshot, device) -> DeviceSnapshot:
+    def pack_views(self, snap, device) -> dict:
         offsets, total = self._layout(snap)
         sig = tuple(shape for _, _, _, _, _, shape in offsets)
 
@@ -177,9 +181,12 @@ class _PinnedPacker:
             self._sig = sig
         if is_cuda:
             self._dev.copy_(self.buf, non_blocking=True)
+        return self._views
 
+    def pack(self, snap: ClusterSnapshot, device) -> DeviceSnapshot:
+        views = self.pack_views(snap, device)
         return DeviceSnapshot(
-            **self._views,
+            **views,
             pool_max_h=np.ascontiguousarray(snap.pool_max, dtype=np.int32).copy(),
             pool_cur_h=np.ascontiguousarray(snap.pool_cur, dtype=np.int32).copy(),
         )
@@ -337,6 +344,8 @@ class GpuEngine(SimulationEngine):
         self.force_torch_segbuild = os.environ.get("MI355X_AS_FORCE_TORCH_SEGBUILD") == "1"
         self.force_py_pipeline = os.environ.get("MI355X_AS_PY_PIPELINE") == "1"
         self._free_cache: Dict[int, torch.Tensor] = {}
+        self._resident: Optional[dict] = None
+        self._rest_packer = _PinnedPacker(_REST_FIELDS)
         self._last_demand: Optional[torch.Tensor] = None
         d = self.device
         # persistent hash workspace; the mutable i32 parts live in ONE
@@ -724,15 +733,11 @@ class GpuEngine(SimulationEngine):
         return torch.zeros((K, 4), dtype=torch.int64, device=device)
 
     # ------------------------------------------------------------------
-    def apply_pod_delta(self, ds: DeviceSnapshot, cache) -> DeviceSnapshot:
-        """Kernel E: scatter a DeltaSnapshotCache's dirty rows into the
-        resident device snapshot. Falls back to a full repack when the
-        pod-array capacity changed (grow/compact)."""
-        rows = cache.drain_dirty()
-        if ds.P != cache.size or (rows.size and int(rows.max()) >= ds.P):
-            return None  # capacity changed (grow/compact): caller repacks
+    def _scatter_rows(self, pod_views: dict, cache, rows: np.ndarray):
+        """Kernel E: scatter the cache's dirty rows into the resident
+        pod columns (one gather on host, one H2D per column batch)."""
         if rows.size == 0:
-            return ds
+            return
         d = self.device
         to_i32 = lambda a: torch.from_numpy(np.ascontiguousarray(a[rows].astype(np.int32))).to(d)
         to_i64 = lambda a: torch.from_numpy(np.ascontiguousarray(a[rows].astype(np.int64))).to(d)
@@ -742,10 +747,62 @@ class GpuEngine(SimulationEngine):
             r, to_i32(a["pod_req"]), to_i32(a["pod_flags"]), to_i64(a["pod_sel_hash"]),
             to_i64(a["pod_sel_pairs"]), to_i32(a["pod_sel_count"]),
             to_i32(a["pod_node"]), to_i32(a["pod_pdb"]),
-            ds.pod_req, ds.pod_flags, ds.pod_sel_hash, ds.pod_sel_pairs,
-            ds.pod_sel_count, ds.pod_node, ds.pod_pdb,
+            pod_views["pod_req"], pod_views["pod_flags"], pod_views["pod_sel_hash"],
+            pod_views["pod_sel_pairs"], pod_views["pod_sel_count"],
+            pod_views["pod_node"], pod_views["pod_pdb"],
         )
+
+    def apply_pod_delta(self, ds: DeviceSnapshot, cache) -> Optional[DeviceSnapshot]:
+        """Scatter a DeltaSnapshotCache's dirty rows into a resident
+        snapshot whose pod axis matches the cache size; None when the
+        capacity changed (caller repacks)."""
+        rows = cache.drain_dirty()
+        if ds.P != cache.size or (rows.size and int(rows.max()) >= ds.P):
+            return None
+        views = dict(pod_req=ds.pod_req, pod_flags=ds.pod_flags,
+                     pod_sel_hash=ds.pod_sel_hash, pod_sel_pairs=ds.pod_sel_pairs,
+                     pod_sel_count=ds.pod_sel_count, pod_node=ds.pod_node,
+                     pod_pdb=ds.pod_pdb)
+        self._scatter_rows(views, cache, rows)
         return ds
+
+    # ------------------------------------------------------------------
+    def decide_resident(self, cache, base: ClusterSnapshot,
+                        over_provision: int = 0) -> ScaleDecision:
+        """Resident-controller decide: the pod SoA stays in HBM at the
+        watch cache's CAPACITY (tombstone rows are inert); per cycle the
+        dirty rows scatter in (kernel E) and the small node/pool side
+        re-uploads — no full snapshot upload. `base` is the per-cycle
+        node/pool-side snapshot (zero pods)."""
+        import types
+
+        cap = cache.cap
+        res = self._resident
+        if res is None or res["cap"] != cap:
+            pod_side = types.SimpleNamespace(
+                **{name: cache.arrays[name] for name, *_ in _POD_FIELDS}
+            )
+            packer = _PinnedPacker(_POD_FIELDS)
+            pod_views = packer.pack_views(pod_side, self.device)
+            self._resident = res = {"cap": cap, "packer": packer,
+                                    "pod_views": pod_views}
+            cache.drain_dirty()
+        else:
+            self._scatter_rows(res["pod_views"], cache, cache.drain_dirty())
+
+        rest_views = self._rest_packer.pack_views(base, self.device)
+        ds = DeviceSnapshot(
+            **res["pod_views"], **rest_views,
+            pool_max_h=np.ascontiguousarray(base.pool_max, dtype=np.int32).copy(),
+            pool_cur_h=np.ascontiguousarray(base.pool_cur, dtype=np.int32).copy(),
+        )
+        decision = self._decide_fused(ds, base.pool_names, over_provision)
+        if decision is None:  # hash overflow: fall back via a full snapshot
+            snap = cache.build_snapshot(base)
+            full = self.packer.pack(snap, self.device)
+            segs = build_segments_device(full, self.ops)
+            return self.decide_device(full, segs, base.pool_names, over_provision)
+        return decision
 
     # ------------------------------------------------------------------
     def maintain(self, snap: ClusterSnapshot, util_threshold_permille: int = 300) -> MaintainReport:

@@ -385,3 +385,45 @@ def test_native_orchestrator_equals_python_pipeline(engine):
         engine.force_py_pipeline = False
     _assert_decisions_equal(d_native, d_py)
     _assert_decisions_equal(d_native, CpuEngine().decide(snap))
+
+
+def test_decide_resident_matches_oracle_through_churn(engine):
+    """Resident-controller mode (kernel E end-to-end): prime, churn via
+    watch events (adds/mods/deletes, capacity growth), and every
+    decision over the HBM-resident snapshot equals the oracle."""
+    import random
+
+    from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools
+    from mi355x_autoscaler.models.kube import KubeNode, KubePod
+    from mi355x_autoscaler.snapshot.delta import DeltaSnapshotCache
+    from mi355x_autoscaler.snapshot.encode import encode_snapshot
+
+    rng = random.Random(5)
+    nodes_json = [make_node_json("agentpool1", i) for i in range(5)]
+    nodes = [KubeNode(n) for n in nodes_json]
+    pools = group_nodes_into_pools(nodes)
+    initial = [make_pod_json(f"p{i}", cpu_milli=100, pending=True) for i in range(200)]
+    snap0 = encode_snapshot([KubePod(p) for p in initial], nodes, pools)
+
+    cache = DeltaSnapshotCache()
+    cache.prime(snap0, {n.name: i for i, n in enumerate(nodes)})
+    engine._resident = None
+    live = {p["metadata"]["name"]: p for p in initial}
+
+    for cycle in range(8):
+        for _ in range(rng.randrange(5, 120)):  # growth forces cap changes
+            p = make_pod_json(f"c{cycle}-{rng.randrange(10**6)}",
+                              cpu_milli=rng.choice([100, 400, 900]),
+                              mem_mib=rng.choice([64, 512]), pending=True)
+            cache.apply_event({"type": "ADDED", "object": p})
+            live[p["metadata"]["name"]] = p
+        for name in rng.sample(sorted(live), min(len(live), rng.randrange(0, 40))):
+            cache.apply_event({"type": "DELETED", "object": live.pop(name)})
+        cache.maybe_compact()
+
+        base = encode_snapshot([], nodes, pools)
+        d_gpu = engine.decide_resident(cache, base)
+        d_cpu = CpuEngine().decide(cache.build_snapshot(base))
+        assert d_gpu.units_needed == d_cpu.units_needed, cycle
+        assert d_gpu.pods_placed_existing == d_cpu.pods_placed_existing, cycle
+        assert d_gpu.pods_unfulfillable == d_cpu.pods_unfulfillable, cycle
