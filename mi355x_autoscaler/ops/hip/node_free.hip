@@ -16,7 +16,10 @@ extern "C" __global__ void node_free_kernel(
     if (j < 0) continue;
 #pragma unroll
     for (int r = 0; r < NRES; ++r) {
-      atomicSub(&node_free[(int64_t)j * NRES + r], pod_req[i * NRES + r]);
+      const int32_t v = pod_req[i * NRES + r];
+      // zero-valued atomics still serialize on the line; skip them
+      // (tombstone/padding rows and sparse request vectors)
+      if (v != 0) atomicSub(&node_free[(int64_t)j * NRES + r], v);
     }
   }
 }

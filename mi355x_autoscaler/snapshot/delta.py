@@ -60,6 +60,8 @@ class DeltaSnapshotCache:
             shape = (self.cap,) + src.shape[1:]
             a = np.zeros(shape, src.dtype)
             a[: self.size] = src
+            if col in ("pod_node", "pod_pdb"):
+                a[self.size :] = -1  # padding rows must be UNBOUND
             self.arrays[col] = a
         self.pod_uids = list(snap.pod_uids) + [""] * (self.cap - self.size)
         self.pod_refs = list(snap.pod_refs) + [("", "")] * (self.cap - self.size)
@@ -132,6 +134,8 @@ class DeltaSnapshotCache:
         for col, a in self.arrays.items():
             b = np.zeros((new_cap,) + a.shape[1:], a.dtype)
             b[: self.cap] = a
+            if col in ("pod_node", "pod_pdb"):
+                b[self.cap :] = -1
             self.arrays[col] = b
         self.pod_uids += [""] * (new_cap - self.cap)
         self.pod_refs += [("", "")] * (new_cap - self.cap)
@@ -232,6 +236,8 @@ class DeltaSnapshotCache:
             src = data[col]
             a = np.zeros((self.cap,) + src.shape[1:], src.dtype)
             a[: self.size] = src
+            if col in ("pod_node", "pod_pdb"):
+                a[self.size :] = -1
             self.arrays[col] = a
         uids = list(data["uids"])
         refs = [tuple(r) for r in data["refs"]]
