@@ -497,12 +497,19 @@ class GpuEngine(SimulationEngine):
         if G == 0:
             decision.pool_residual_totals = np.zeros((K, 4), np.int64)
             return decision
+        placed_h = blob_h[0:G]
+        unful_h = blob_h[2 * SEG_MAX : 2 * SEG_MAX + G]
+        if K == 0:
+            # no agent pools at all: everything unplaced is unfulfillable
+            # (the units kernel is skipped, so derive it from the counts)
+            counts_h = self.ws_seg_count[:G].cpu().numpy()
+            unful_h = counts_h - placed_h
         max_new_h = np.maximum(
             (ds.pool_max_h - ds.pool_cur_h) if ds.pool_max_h is not None
             else np.zeros(K, np.int64), 0)
         return self._finish_decision(
-            decision, blob_h[0:G], blob_h[SEG_MAX : SEG_MAX + G],
-            blob_h[2 * SEG_MAX : 2 * SEG_MAX + G],
+            decision, placed_h, blob_h[SEG_MAX : SEG_MAX + G],
+            unful_h,
             blob_h[3 * SEG_MAX : 3 * SEG_MAX + K],
             blob_h[3 * SEG_MAX + K : 3 * SEG_MAX + 5 * K].reshape(max(K, 1) if K else 0, 4) if K else np.zeros((0, 4), np.int64),
             max_new_h, ds.pool_cur_h, K, pool_names, over_provision,
@@ -688,6 +695,9 @@ class GpuEngine(SimulationEngine):
         residual_h = blob_h[3 * G + K : 3 * G + 5 * K].reshape(K, 4) if K else np.zeros((0, 4), np.int64)
         pool_cur_h = ds.pool_cur_h if ds.pool_cur_h is not None else ds.pool_cur.cpu().numpy()
 
+        if K == 0:
+            counts_h = segs.seg_count.cpu().numpy()
+            unful_h = counts_h - placed_h
         decision.pods_placed_existing = int(placed_h.sum())
         decision.pods_placed_new = int(placed_new_h.sum())
         decision.pods_unfulfillable = int(unful_h.sum())

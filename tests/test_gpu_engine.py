@@ -427,3 +427,19 @@ def test_decide_resident_matches_oracle_through_churn(engine):
         assert d_gpu.units_needed == d_cpu.units_needed, cycle
         assert d_gpu.pods_placed_existing == d_cpu.pods_placed_existing, cycle
         assert d_gpu.pods_unfulfillable == d_cpu.pods_unfulfillable, cycle
+
+
+def test_no_pools_cluster(engine):
+    """Edge: every node is a master (no agent pools) — pending pods
+    that fit nowhere must count unfulfillable, not silently vanish."""
+    master = {"metadata": {"name": "k8s-master-12345678-0", "uid": "m0",
+                           "labels": {}},
+              "spec": {}, "status": {"capacity": {"cpu": "2", "memory": "4Gi",
+                                                  "pods": "10"}}}
+    pods = [make_pod_json(f"p{i}", cpu_milli=500, pending=True) for i in range(5)]
+    snap = _snap(pods, [master])
+    assert snap.num_pools == 0
+    d_gpu = engine.decide(snap)
+    d_cpu = CpuEngine().decide(snap)
+    assert d_gpu.pods_unfulfillable == d_cpu.pods_unfulfillable == 5
+    assert d_gpu.pods_placed_existing == d_cpu.pods_placed_existing
