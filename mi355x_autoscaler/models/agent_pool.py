@@ -20,6 +20,11 @@ from .resources import ResourceVec
 # ceiling [R: README.md known limitations — reconstructed].
 DEFAULT_MAX_POOL_SIZE = 100
 
+# "k8s-master-<suffix>-<idx>" matches the agent naming convention but is
+# the control plane — never an agent pool to scale (the reference's
+# template processing strips master resources for the same reason).
+ALWAYS_IGNORED_POOLS = frozenset({"master"})
+
 
 class AgentPool:
     def __init__(
@@ -106,7 +111,7 @@ def group_nodes_into_pools(
     (e.g. masters) are skipped.
     [R: autoscaler/cluster.py · pool grouping — reconstructed]
     """
-    ignore = set(ignore_pools or [])
+    ignore = set(ignore_pools or []) | ALWAYS_IGNORED_POOLS
     pools: Dict[str, AgentPool] = {}
     for node in nodes:
         pool = node.pool_name
