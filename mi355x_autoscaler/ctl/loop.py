@@ -173,6 +173,22 @@ class Cluster:
         result.pods_total = snap.num_pods
         result.snapshot = snap
 
+        # sharded controller: every rank lists the FULL cluster; the
+        # decide/maintain inputs are this rank's uid-hash shard (the
+        # engine all-reduces the per-segment counts / node partials),
+        # and only rank 0 actuates.
+        from ..parallel import dist as pdist
+
+        world = pdist.world_size()
+        rank = pdist.rank()
+        if world > 1:
+            from ..parallel.elastic import shard_snapshot
+
+            engine_snap = shard_snapshot(snap, rank, world)
+        else:
+            engine_snap = snap
+        is_rank0 = rank == 0
+
         # ---- scale-up --------------------------------------------------
         t_decide0 = time.perf_counter()
         with t.phase("decide"):
@@ -180,6 +196,7 @@ class Cluster:
             if (
                 self.opts.watch
                 and ws is not None
+                and world == 1
                 and hasattr(self.engine, "decide_resident")
             ):
                 # resident-controller path: pod SoA stays in HBM; watch
@@ -189,7 +206,7 @@ class Cluster:
                     ws["cache"], base, over_provision=self.opts.over_provision
                 )
             else:
-                decision = self.engine.decide(snap, over_provision=self.opts.over_provision)
+                decision = self.engine.decide(engine_snap, over_provision=self.opts.over_provision)
         self.metrics.observe_latency(time.perf_counter() - t_decide0)
         result.decision = decision
         from ..snapshot.encode import PodFlags
@@ -198,21 +215,23 @@ class Cluster:
         self.metrics.gauge("pods_pending", result.pods_pending)
         self.metrics.count("pods_scheduled", decision.pods_placed_existing + decision.pods_placed_new)
 
-        if decision.wants_scale_up() and not self.opts.no_scale:
+        if decision.wants_scale_up() and not self.opts.no_scale and is_rank0:
             with t.phase("actuate"):
                 target = self.provider.current_pool_sizes()
                 target.update(decision.new_pool_sizes)
                 self.provider.scale_pools(target, dry_run=self.opts.dry_run)
             self.metrics.count("scale_ups")
             self.notifier.notify_scale_up(decision.units_needed, result.pods_pending)
-        for sel_hash, count in decision.unfulfillable_detail:
-            self.notifier.notify_failed_to_scale(sel_hash, count)
+        if is_rank0:
+            for sel_hash, count in decision.unfulfillable_detail:
+                self.notifier.notify_failed_to_scale(sel_hash, count)
 
         # ---- scale-down ------------------------------------------------
         suppress = decision.wants_scale_up() and not self.opts.scale_down_during_scale_up
         if not self.opts.no_maintenance and not suppress:
             with t.phase("maintain"):
-                drained = self.maintain(snap, pools, nodes, now=now)
+                drained = self.maintain(engine_snap, pools, nodes, now=now,
+                                        actuation_snap=snap, actuate=is_rank0)
             result.drained_nodes = drained
 
         result.phase_times = t.end_cycle()
@@ -332,11 +351,16 @@ class Cluster:
         pools,
         nodes: List[KubeNode],
         now: Optional[float] = None,
+        actuation_snap: Optional[ClusterSnapshot] = None,
+        actuate: bool = True,
     ) -> List[str]:
         """Scale-down pass [R: cluster.py · maintain — reconstructed]:
-        score nodes (engine kernel D), classify states, cordon+drain+
-        delete eligible nodes, reserving spare agents."""
+        score nodes (engine kernel D — a collective when sharded),
+        classify states, cordon+drain+delete eligible nodes, reserving
+        spare agents. ``snap`` may be a pod shard (nodes replicated);
+        ``actuation_snap`` is the full snapshot rank 0 drains from."""
         now = time.time() if now is None else now
+        actuation_snap = snap if actuation_snap is None else actuation_snap
         report = self.engine.maintain(snap, self.opts.utilization_threshold_permille)
         states = self.classify_node_states(snap, report, now)
         node_by_name = {n.name: n for n in nodes}
@@ -354,8 +378,11 @@ class Cluster:
             node = node_by_name.get(name)
             if node is None:
                 continue
+            if not actuate:
+                drained.append(name)  # agreed decision; rank 0 executes
+                continue
             try:
-                self._drain_and_delete(snap, node, j)
+                self._drain_and_delete(actuation_snap, node, j)
                 drained.append(name)
                 self.metrics.count("scale_downs")
             except Exception:
