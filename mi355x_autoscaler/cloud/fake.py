@@ -34,6 +34,7 @@ class FakeCloudProvider(CloudProvider):
         self.node_boot_delay_s = node_boot_delay_s
         self.events: List[ScaleEvent] = []
         self.fail_next: int = 0
+        self.terminated_vms: set = set()
 
     def _maybe_fail(self):
         if self.fail_next > 0:
@@ -83,6 +84,7 @@ class FakeCloudProvider(CloudProvider):
             self.events.append(ScaleEvent(kind="delete_node", node_name=node_name, dry_run=dry_run))
             if dry_run:
                 return True
+            self.terminated_vms.add(node_name)
             # pool bookkeeping: k8s-<pool>-<suffix>-<idx>
             parts = node_name.split("-")
             if len(parts) >= 4 and parts[0] == "k8s":
@@ -94,3 +96,7 @@ class FakeCloudProvider(CloudProvider):
     def current_pool_sizes(self) -> Dict[str, int]:
         with self.lock:
             return dict(self.pool_sizes)
+
+    def vm_exists(self, instance_id: str) -> bool:
+        with self.lock:
+            return instance_id not in self.terminated_vms

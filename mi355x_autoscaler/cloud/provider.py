@@ -41,3 +41,9 @@ class CloudProvider(abc.ABC):
     @abc.abstractmethod
     def current_pool_sizes(self) -> Dict[str, int]:
         """Pool sizes as the cloud believes them (deployment record)."""
+
+    def vm_exists(self, instance_id: str) -> Optional[bool]:
+        """Whether the backing VM still exists; None = unknown (the
+        instance-terminated node state falls back to 'busy' handling).
+        [R: ec2 ancestor's instance-terminated state — reconstructed]"""
+        return None

@@ -54,6 +54,7 @@ STATE_UNDER_UTILIZED_DRAINABLE = "under-utilized-drainable"
 STATE_UNDER_UTILIZED_UNDRAINABLE = "under-utilized-undrainable"
 STATE_GRACE_PERIOD = "grace-period"
 STATE_SPARE_AGENT = "spare-agent"
+STATE_INSTANCE_TERMINATED = "instance-terminated"
 
 
 @dataclass
@@ -365,6 +366,19 @@ class Cluster:
         states = self.classify_node_states(snap, report, now)
         node_by_name = {n.name: n for n in nodes}
         drained: List[str] = []
+
+        # instance-terminated cleanup: the backing VM is gone but the
+        # Node object lingers — delete the orphan (rank 0 / actuating
+        # caller only). [R: ec2 ancestor instance-terminated state]
+        if actuate and not self.opts.dry_run:
+            for node in nodes:
+                if self.provider.vm_exists(node.instance_id) is False:
+                    logger.warning("node %s backing VM is gone; removing orphan", node.name)
+                    try:
+                        self.kube.delete_node(node.name)
+                        self._idle_since.pop(node.name, None)
+                    except Exception:
+                        logger.exception("orphan node cleanup failed for %s", node.name)
 
         drainable_states = (STATE_IDLE_SCHEDULABLE, STATE_IDLE_UNSCHEDULABLE, STATE_UNDER_UTILIZED_DRAINABLE)
         # drain lowest-utilization first (report.candidates is util-sorted)

@@ -121,3 +121,17 @@ def test_decision_latency_recorded(server):
     cluster.loop_logic()
     assert cluster.metrics.counters["last_decision_latency_s"] > 0
     assert cluster.metrics.timer.history[-1]["total"] > 0
+
+
+def test_instance_terminated_orphan_cleanup(server):
+    """A Node whose backing VM was deleted out-of-band (the ec2
+    ancestor's instance-terminated state) is removed from the API."""
+    nodes = [make_node_json("agentpool1", i) for i in range(3)]
+    pods = [make_pod_json("w", cpu_milli=1500, node_name=nodes[0]["metadata"]["name"])]
+    server.state.load(pods, nodes)
+    cluster, provider = _cluster(server, no_scale=True, max_drains_per_cycle=0)
+    orphan = nodes[2]["metadata"]["name"]
+    provider.terminated_vms.add(orphan)  # VM deleted outside the autoscaler
+    cluster.loop_logic()
+    assert orphan in server.state.deleted_nodes
+    assert orphan not in server.state.nodes
