@@ -182,7 +182,9 @@ class DeltaSnapshotCache:
         for col in _POD_COLS:
             a = self.arrays[col]
             a[: len(live)] = a[live]
-            a[len(live): self.size] = 0
+            # vacated rows must be inert: flags 0 AND unbound (-1) — a
+            # zero pod_node would phantom-bind them to node 0
+            a[len(live): self.size] = -1 if col in ("pod_node", "pod_pdb") else 0
         self.pod_uids = [self.pod_uids[i] for i in live] + [""] * (self.cap - len(live))
         self.pod_refs = [self.pod_refs[i] for i in live] + [("", "")] * (self.cap - len(live))
         self.uid_row = {u: i for i, u in enumerate(self.pod_uids[: len(live)]) if u}
