@@ -156,16 +156,20 @@ class OracleSimulator:
     # ------------------------------------------------------------------
     def node_eligibility(self, seg: Segment) -> np.ndarray:
         """[N] bool: schedulable node whose labels contain the selector.
-        (Device form: kernel A fit_predicates.hip)"""
+        (Device form: kernel A fit_predicates.hip.) Vectorized subset
+        test: selector pair s is present on node j iff any of the
+        node's padded label hashes equals s — semantically identical to
+        selector_subset_match (tested directly in test_snapshot.py),
+        but numpy-parallel over nodes."""
         snap = self.snap
-        out = np.zeros(snap.num_nodes, dtype=bool)
-        for j in range(snap.num_nodes):
-            if snap.node_unsched[j]:
-                continue
-            out[j] = selector_subset_match(
-                seg.sel_pairs, seg.sel_count, snap.node_labels[j], int(snap.node_label_count[j])
-            )
-        return out
+        sched = snap.node_unsched == 0
+        if seg.sel_count == 0:
+            return sched.copy()
+        sel = seg.sel_pairs[: seg.sel_count]
+        # [N, L, S]: node label l equals selector pair s; 0-padded label
+        # slots never match a real pair hash (0 is not a valid FNV pair)
+        present = (snap.node_labels[:, :, None] == sel[None, None, :]).any(axis=1)
+        return sched & present.all(axis=1)
 
     @staticmethod
     def _per_node_capacity(free: np.ndarray, req: np.ndarray, elig: np.ndarray) -> np.ndarray:
