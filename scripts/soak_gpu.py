@@ -1,6 +1,7 @@
-# 120-cycle churn soak on hardware: watch-mode loop + GPU engine with
+# long churn soak (cycles from argv, default 120) on hardware: watch-mode loop + GPU engine with
 # pod add/remove churn and periodic scale-up/down, asserting stability.
 import random, sys, time
+CYCLES = int(sys.argv[1]) if len(sys.argv) > 1 else 120
 sys.path.insert(0, "/root/repo")
 from mi355x_autoscaler.cloud.fake import FakeCloudProvider
 from mi355x_autoscaler.ctl.loop import Cluster, ClusterOptions
@@ -23,21 +24,28 @@ with FakeKubeServer() as srv:
     cluster = Cluster(kube, provider, gpu, options=opts)
     added = 0
     t0 = time.time()
-    for cycle in range(120):
+    for cycle in range(CYCLES):
         # churn: add a burst, delete some pods
         for _ in range(rng.randrange(0, 30)):
             srv.state.add_pod(make_pod_json(f"churn-{added}", cpu_milli=rng.choice([100, 300, 700]),
                                             mem_mib=rng.choice([64, 256, 1024]), pending=True))
             added += 1
+        # occasionally delete random churned pods (exercise tombstones)
+        if cycle % 7 == 3:
+            with srv.state.lock:
+                keys = [k for k in srv.state.pods if "churn-" in k][:20]
+            for k in keys:
+                ns, name = k.split("/", 1)
+                srv.state.remove_pod(ns, name)
         time.sleep(0.03)
         res = cluster.loop_logic()
         # every 20 cycles cross-check the decision against the CPU oracle
-        if cycle % 20 == 0 and res.snapshot is not None:
+        if cycle % 25 == 0 and res.snapshot is not None:
             d_cpu = cpu_check.decide(res.snapshot)
             assert d_cpu.units_needed == res.decision.units_needed, (
                 cycle, d_cpu.units_needed, res.decision.units_needed)
     dt = time.time() - t0
-    print(f"soak ok: 120 cycles, {added} churned pods, {dt:.1f}s "
-          f"({dt/120*1000:.1f} ms/cycle), pools={provider.pool_sizes}, "
+    print(f"soak ok: {CYCLES} cycles, {added} churned pods, {dt:.1f}s "
+          f"({dt/CYCLES*1000:.1f} ms/cycle), pools={provider.pool_sizes}, "
           f"scale_ups={sum(1 for e in provider.events if e.kind=='scale_up')}, "
           f"deletes={sum(1 for e in provider.events if e.kind=='delete_node')}")
