@@ -28,6 +28,8 @@ world_size 2) and nccl/RCCL (GPU engine on MI355X).
 
 from __future__ import annotations
 
+import time
+
 from typing import List
 
 import numpy as np
@@ -108,6 +110,8 @@ class ShardedEngine(SimulationEngine):
         self.inner = inner
         self.check_agreement = check_agreement
         self.active = pdist.init_from_env()
+        # SURVEY §5 observability: cumulative collective wall time
+        self.comm_stats = {"allgather_s": 0.0, "allreduce_s": 0.0, "calls": 0}
 
     # ------------------------------------------------------------------
     def decide(self, snap: ClusterSnapshot, over_provision: int = 0) -> ScaleDecision:
@@ -115,7 +119,9 @@ class ShardedEngine(SimulationEngine):
         if isinstance(self.inner, CpuEngine) or device is None:
             local = build_segments(snap)
             rows = _segments_to_rows(local)
+            t0 = time.perf_counter()
             rows = pdist.allgather_rows(rows)
+            self.comm_stats["allgather_s"] += time.perf_counter() - t0
             merged = merge_rows(rows)
             segments = _rows_to_segments(merged)
             node_free = None
@@ -132,7 +138,9 @@ class ShardedEngine(SimulationEngine):
             ds = self.inner.packer.pack(snap, device)
             segs = self.inner.build_segments(ds)
             rows = self._device_segments_to_rows(ds, segs)
+            t0 = time.perf_counter()
             rows = pdist.allgather_rows(rows, device=device)
+            self.comm_stats["allgather_s"] += time.perf_counter() - t0
             merged = merge_rows(rows)
             gsegs = self._rows_to_device_segments(merged, device)
 
@@ -150,7 +158,10 @@ class ShardedEngine(SimulationEngine):
                 free_hook=free_hook,
             )
         if self.active and self.check_agreement:
+            t0 = time.perf_counter()
             self._assert_agreement(decision, snap)
+            self.comm_stats["allreduce_s"] += time.perf_counter() - t0
+        self.comm_stats["calls"] += 1
         return decision
 
     @staticmethod
