@@ -251,6 +251,62 @@ extern "C" __global__ __launch_bounds__(256) void segment_hash_build(
   }
 }
 
+// Small-P variant: below ~64k pods the LDS sub-table's fixed costs
+// (table init, per-block flush) exceed the contention they avoid, so
+// the wave-aggregated direct-global form wins; the host dispatches on
+// P (segment_hash launcher below).
+extern "C" __global__ void segment_hash_build_small(
+    const int32_t* __restrict__ pod_flags, const int64_t* __restrict__ sel_hash,
+    const int32_t* __restrict__ pod_req, int64_t P, int32_t H,
+    int64_t* __restrict__ slot_ks, int64_t* __restrict__ slot_kr,
+    int32_t* __restrict__ slot_state, int32_t* __restrict__ slot_count,
+    int32_t* __restrict__ slot_first, int32_t* __restrict__ slot_of_pod,
+    int32_t* __restrict__ overflow) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; ; i0 += stride) {
+    if (__all(i0 >= P)) break;
+    const int pf = i0 < P ? pod_flags[i0] : 0;
+    const bool live = i0 < P && (pf & PF_PENDING);
+    int64_t ks = 0, kr = 0;
+    if (live) {
+      ks = sel_hash[i0];
+      int32_t rq[NRES];
+#pragma unroll
+      for (int r = 0; r < NRES; ++r) rq[r] = pod_req[i0 * NRES + r];
+      kr = pack_req_key(rq, pf & PF_CRITICAL);
+    } else if (i0 < P) {
+      slot_of_pod[i0] = -1;
+    }
+
+    // leader election over distinct keys among live lanes: one global
+    // probe + one count/first atomic pair per distinct key per wave
+    uint64_t pending_lanes = __ballot(live);
+    int my_slot = -1;
+    while (pending_lanes) {
+      const int leader = __ffsll((unsigned long long)pending_lanes) - 1;
+      const int64_t k0 = __shfl(ks, leader, WAVE);
+      const int64_t r0 = __shfl(kr, leader, WAVE);
+      const bool match = live && ks == k0 && kr == r0;
+      const uint64_t match_mask = __ballot(match);
+      int slot = -1;
+      if (lane == leader) {
+        slot = probe_claim(k0, r0, H, slot_ks, slot_kr, slot_state, overflow);
+        if (slot >= 0) {
+          atomicAdd(&slot_count[slot], __popcll((unsigned long long)match_mask));
+          atomicMax(&slot_first[slot], INT32_MAX - (int32_t)i0);
+        } else {
+          atomicOr(overflow, 1);
+        }
+      }
+      slot = __shfl(slot, leader, WAVE);
+      if (match) my_slot = slot;
+      pending_lanes &= ~match_mask;
+    }
+    if (i0 < P && live) slot_of_pod[i0] = my_slot;
+  }
+}
+
 // ---------------------------------------------------------------------------
 
 #define CS_BLOCK 1024
@@ -385,7 +441,8 @@ void segment_hash(torch::Tensor pod_flags, torch::Tensor sel_hash,
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const int block = 256;
   const int grid = (int)std::min<int64_t>((P + block - 1) / block, 2048);
-  hipLaunchKernelGGL(segment_hash_build, dim3(grid), dim3(block), 0,
+  auto kfn = P < 65536 ? segment_hash_build_small : segment_hash_build;
+  hipLaunchKernelGGL(kfn, dim3(grid), dim3(block), 0,
                      stream.stream(), pod_flags.data_ptr<int32_t>(),
                      sel_hash.data_ptr<int64_t>(), pod_req.data_ptr<int32_t>(),
                      P, H, slot_ks.data_ptr<int64_t>(), slot_kr.data_ptr<int64_t>(),
