@@ -119,6 +119,9 @@ class Cluster:
         self._idle_since: Dict[str, float] = {}
         self._cycle = 0
         self._watch_state: Optional[Dict] = None
+        # node/pool-side base-snapshot cache for the watch-resident path
+        # (re-encoding 1k+ nodes in Python costs ~100 ms/cycle otherwise)
+        self._base_cache: Optional[tuple] = None
 
     # ------------------------------------------------------------------
     def loop(self, max_cycles: Optional[int] = None, debug: bool = False):
@@ -202,7 +205,7 @@ class Cluster:
             ):
                 # resident-controller path: pod SoA stays in HBM; watch
                 # deltas scatter in (kernel E), node/pool side re-uploads
-                base = encode_snapshot([], ws["nodes"], ws["pools"], ws["pdbs"])
+                base = self._base_snapshot(ws["nodes"], ws["pools"], ws["pdbs"])
                 decision = self.engine.decide_resident(
                     ws["cache"], base, over_provision=self.opts.over_provision
                 )
@@ -238,6 +241,24 @@ class Cluster:
         result.phase_times = t.end_cycle()
         self.metrics.count("cycles")
         return result
+
+    def _base_snapshot(self, nodes, pools, pdbs) -> ClusterSnapshot:
+        """Node/pool-side snapshot (zero pods), cached behind a content
+        fingerprint so unchanged node sets skip the Python re-encode."""
+        fp = hash((
+            tuple(
+                (n.name, n.unschedulable, n.allocatable.as_tuple(),
+                 frozenset(n.labels.items()))
+                for n in nodes
+            ),
+            tuple((p.name, p.disruptions_allowed,
+                   frozenset(p.match_labels.items())) for p in pdbs),
+        ))
+        if self._base_cache is not None and self._base_cache[0] == fp:
+            return self._base_cache[1]
+        base = encode_snapshot([], nodes, pools, pdbs)
+        self._base_cache = (fp, base)
+        return base
 
     # ------------------------------------------------------------------
     # watch-based delta ingest (SURVEY row E, host half)
@@ -296,7 +317,7 @@ class Cluster:
             applied += 1
         cache.maybe_compact()
         self.metrics.count("watch_events", applied)
-        base = encode_snapshot([], nodes, pools, pdbs)
+        base = self._base_snapshot(nodes, pools, pdbs)
         return cache.build_snapshot(base)
 
     def _watch_pump(self, q):
