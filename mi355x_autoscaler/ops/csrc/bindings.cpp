@@ -58,6 +58,18 @@ void pack_into(pybind11::buffer buf, const std::vector<int64_t>& offsets,
                pybind11::list arrays);
 void pool_max_new(torch::Tensor pool_max, torch::Tensor pool_cur,
                   torch::Tensor out);
+void fused_tail_launch(
+    torch::Tensor pod_req, torch::Tensor pod_node, torch::Tensor node_alloc,
+    torch::Tensor node_labels, torch::Tensor node_label_count,
+    torch::Tensor node_unsched, torch::Tensor pool_unit,
+    torch::Tensor pool_labels, torch::Tensor pool_label_count,
+    torch::Tensor pool_max, torch::Tensor pool_cur,
+    torch::Tensor stat,
+    torch::Tensor seg_sel_pairs, torch::Tensor seg_sel_count,
+    torch::Tensor seg_req, torch::Tensor seg_count,
+    torch::Tensor free, torch::Tensor elig, torch::Tensor seg_pool,
+    torch::Tensor max_new, torch::Tensor run_scratch, torch::Tensor blob,
+    int64_t seg_max, int64_t want_free);
 void fused_decide_launch(
     torch::Tensor pod_req, torch::Tensor pod_flags, torch::Tensor pod_sel_hash,
     torch::Tensor pod_sel_pairs, torch::Tensor pod_sel_count,
@@ -105,4 +117,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pool_max_new", &pool_max_new, "clamp(pool_max - pool_cur, 0)");
   m.def("fused_decide_launch", &fused_decide_launch,
         "enqueue the whole scale-up pipeline in one call");
+  m.def("fused_tail_launch", &fused_tail_launch,
+        "enqueue the post-segment-build pipeline (sharded merged path)");
 }

@@ -131,3 +131,60 @@ void fused_decide_launch(
                  stat.data_ptr<int32_t>(), 3 * sizeof(int32_t),
                  hipMemcpyDeviceToDevice, stream);
 }
+
+
+// Tail-only variant for the SHARDED path: the merged global segment
+// list (all-gathered + merged by the host) is already in the seg
+// workspace and stat[0] = G; this enqueues everything after the
+// segment build. want_free=0 means the caller already produced `free`
+// (the cross-rank usage all-reduce sits between subtract and place).
+void fused_tail_launch(
+    torch::Tensor pod_req, torch::Tensor pod_node, torch::Tensor node_alloc,
+    torch::Tensor node_labels, torch::Tensor node_label_count,
+    torch::Tensor node_unsched, torch::Tensor pool_unit,
+    torch::Tensor pool_labels, torch::Tensor pool_label_count,
+    torch::Tensor pool_max, torch::Tensor pool_cur,
+    torch::Tensor stat,            // [3] i32: G, 0, 0 (host-set)
+    torch::Tensor seg_sel_pairs, torch::Tensor seg_sel_count,
+    torch::Tensor seg_req, torch::Tensor seg_count,
+    torch::Tensor free, torch::Tensor elig, torch::Tensor seg_pool,
+    torch::Tensor max_new, torch::Tensor run_scratch, torch::Tensor blob,
+    int64_t seg_max, int64_t want_free) {
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  const int64_t SM = seg_max;
+  const int64_t K = pool_unit.size(0);
+  const int64_t N = node_alloc.size(0);
+  auto g_dev = stat.narrow(0, 0, 1);
+
+  hipMemsetAsync(blob.data_ptr<int64_t>(), 0,
+                 (size_t)blob.numel() * sizeof(int64_t), stream);
+  if (want_free) {
+    hipMemcpyAsync(free.data_ptr<int32_t>(), node_alloc.data_ptr<int32_t>(),
+                   (size_t)node_alloc.numel() * sizeof(int32_t),
+                   hipMemcpyDeviceToDevice, stream);
+    node_free_subtract(pod_node, pod_req, free);
+  }
+  if (N > 0) {
+    fit_eligibility(node_labels, node_label_count, node_unsched, seg_sel_pairs,
+                    seg_sel_count, g_dev, elig);
+  }
+  seg_pool_route(seg_sel_pairs, seg_sel_count, seg_req, pool_labels,
+                 pool_label_count, pool_unit, g_dev, seg_pool);
+  auto placed = blob.narrow(0, 0, SM);
+  auto placed_new = blob.narrow(0, SM, SM);
+  auto unful = blob.narrow(0, 2 * SM, SM);
+  if (N > 0) {
+    binpack_place(free, elig, seg_req, seg_count, g_dev, placed);
+  }
+  if (K > 0) {
+    auto units_cnt = blob.narrow(0, 3 * SM, K);
+    auto pool_residual = blob.narrow(0, 3 * SM + K, 4 * K);
+    pool_max_new(pool_max, pool_cur, max_new);
+    binpack_units(seg_pool, seg_req, seg_count, placed, pool_unit, max_new,
+                  g_dev, run_scratch, units_cnt, placed_new, unful,
+                  pool_residual);
+  }
+  hipMemcpyAsync(blob.data_ptr<int64_t>() + (blob.numel() - 2),
+                 stat.data_ptr<int32_t>(), 3 * sizeof(int32_t),
+                 hipMemcpyDeviceToDevice, stream);
+}

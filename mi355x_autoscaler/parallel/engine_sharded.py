@@ -142,7 +142,6 @@ class ShardedEngine(SimulationEngine):
             rows = pdist.allgather_rows(rows, device=device)
             self.comm_stats["allgather_s"] += time.perf_counter() - t0
             merged = merge_rows(rows)
-            gsegs = self._rows_to_device_segments(merged, device)
 
             def free_hook(free, dsnap):
                 if not self.active:
@@ -153,10 +152,17 @@ class ShardedEngine(SimulationEngine):
                     min=-(2**31), max=2**31 - 1
                 ).to(torch.int32)
 
-            decision = self.inner.decide_device(
-                ds, gsegs, snap.pool_names, over_provision=over_provision,
-                free_hook=free_hook,
+            hook = free_hook if self.active else None
+            decision = self.inner.decide_merged_rows(
+                ds, merged, snap.pool_names, over_provision=over_provision,
+                free_hook=hook,
             )
+            if decision is None:  # merged list exceeds the workspace
+                gsegs = self._rows_to_device_segments(merged, device)
+                decision = self.inner.decide_device(
+                    ds, gsegs, snap.pool_names, over_provision=over_provision,
+                    free_hook=free_hook,
+                )
         if self.active and self.check_agreement:
             t0 = time.perf_counter()
             self._assert_agreement(decision, snap)
