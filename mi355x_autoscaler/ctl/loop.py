@@ -305,27 +305,46 @@ class Cluster:
         cache = ws["cache"]
         ws["nodes"], ws["pools"], ws["pdbs"] = nodes, pools, pdbs
         applied = 0
+        raw_lines: List[bytes] = []
         while True:
             try:
                 ev = ws["q"].get_nowait()
             except _q.Empty:
                 break
+            if isinstance(ev, bytes):
+                raw_lines.append(ev)
+                continue
             if ev.get("type") == "__STREAM_END__":
                 ws["broken"] = True
                 break
             cache.apply_event(ev)
             applied += 1
+        if raw_lines:
+            # C++ batch decode of the whole cycle's events at once
+            from mi355x_autoscaler import _hip_ops
+
+            decoded = _hip_ops.decode_watch_events(
+                b"\n".join(raw_lines) + b"\n", cache.node_index,
+                list(cache.pdb_selectors),
+            )
+            applied += cache.apply_batch(decoded)
         cache.maybe_compact()
         self.metrics.count("watch_events", applied)
         base = self._base_snapshot(nodes, pools, pdbs)
         return cache.build_snapshot(base)
 
     def _watch_pump(self, q):
-        """Background thread: stream pod watch events into the queue;
-        a dropped stream posts a sentinel so the loop re-primes."""
+        """Background thread: stream pod watch events into the queue —
+        raw JSONL lines when the C++ batch decoder is available,
+        parsed dicts otherwise; a dropped stream posts a sentinel so
+        the loop re-primes."""
         try:
-            for ev in self.kube.watch("/api/v1/pods", resource_version="0"):
-                q.put(ev)
+            if self.opts.native_codec and _codec_available():
+                for line in self.kube.watch_raw("/api/v1/pods", resource_version="0"):
+                    q.put(line)
+            else:
+                for ev in self.kube.watch("/api/v1/pods", resource_version="0"):
+                    q.put(ev)
         except Exception:
             logger.exception("watch stream error")
         q.put({"type": "__STREAM_END__"})

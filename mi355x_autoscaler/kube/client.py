@@ -117,6 +117,12 @@ class KubeApiClient:
 
     def watch(self, path: str, resource_version: Optional[str] = None) -> Iterator[Dict]:
         """Yield watch events (``{"type": ..., "object": ...}``)."""
+        for line in self.watch_raw(path, resource_version):
+            yield json.loads(line)
+
+    def watch_raw(self, path: str, resource_version: Optional[str] = None) -> Iterator[bytes]:
+        """Raw JSONL watch lines — fed to the C++ batch event decoder
+        without building Python dicts."""
         params = {"watch": "true"}
         if resource_version:
             params["resourceVersion"] = resource_version
@@ -125,7 +131,7 @@ class KubeApiClient:
         )
         for line in resp.iter_lines():
             if line:
-                yield json.loads(line)
+                yield line
 
     # -- mutations (cordon / drain / delete) ------------------------------
     def set_unschedulable(self, node_name: str, value: bool) -> None:

@@ -56,6 +56,9 @@ pybind11::dict encode_pods_json(pybind11::bytes body, pybind11::dict node_index,
                                 pybind11::list pdb_selectors);
 void pack_into(pybind11::buffer buf, const std::vector<int64_t>& offsets,
                pybind11::list arrays);
+pybind11::dict decode_watch_events(pybind11::bytes body,
+                                   pybind11::dict node_index,
+                                   pybind11::list pdb_selectors);
 void pool_max_new(torch::Tensor pool_max, torch::Tensor pool_cur,
                   torch::Tensor out);
 void fused_tail_launch(
@@ -114,6 +117,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("snapshot_scatter", &snapshot_scatter,
         "kernel E: delta-apply rows into the resident snapshot");
   m.def("pack_into", &pack_into, "memcpy arrays into the staging buffer");
+  m.def("decode_watch_events", &decode_watch_events,
+        "JSONL watch stream -> event types + pod SoA columns");
   m.def("pool_max_new", &pool_max_new, "clamp(pool_max - pool_cur, 0)");
   m.def("fused_decide_launch", &fused_decide_launch,
         "enqueue the whole scale-up pipeline in one call");

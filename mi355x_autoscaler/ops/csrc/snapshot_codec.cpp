@@ -792,3 +792,138 @@ void pack_into(pybind11::buffer buf, const std::vector<int64_t>& offsets,
     std::memcpy(base + offsets[i], src.ptr, (size_t)nb);
   }
 }
+
+
+// ---------------------------------------------------------------------------
+// decode_watch_events: JSONL watch stream ({"type":..,"object":{pod}})
+// -> event types + the same pod columns as encode_pods_json, so the
+// delta cache can batch-apply instead of per-event Python encoding.
+// ---------------------------------------------------------------------------
+pybind11::dict decode_watch_events(pybind11::bytes body,
+                                   pybind11::dict node_index,
+                                   pybind11::list pdb_selectors) {
+  namespace py = pybind11;
+  std::string_view sv = body;
+
+  Ctx ctx;
+  for (auto item : node_index)
+    ctx.node_idx[py::cast<std::string>(item.first)] = py::cast<int32_t>(item.second);
+  for (auto sel : pdb_selectors) {
+    std::vector<std::pair<std::string, std::string>> pairs;
+    for (auto kv : py::cast<py::dict>(sel))
+      pairs.emplace_back(py::cast<std::string>(kv.first), py::cast<std::string>(kv.second));
+    ctx.pdbs.push_back(std::move(pairs));
+  }
+
+  struct EvShard {
+    Columns cols;
+    std::vector<int8_t> types;
+  };
+  std::vector<EvShard> shards;
+  {
+    py::gil_scoped_release nogil;
+    // line spans (memchr scan), then thread-parallel per-line parse
+    std::vector<std::pair<size_t, size_t>> lines;
+    size_t pos = 0, n = sv.size();
+    while (pos < n) {
+      const void* q = memchr(sv.data() + pos, '\n', n - pos);
+      size_t eol = q ? (const char*)q - sv.data() : n;
+      if (eol > pos) lines.emplace_back(pos, eol);
+      pos = eol + 1;
+    }
+    const size_t M = lines.size();
+    unsigned hw = std::thread::hardware_concurrency();
+    size_t T = std::max<size_t>(1, std::min<size_t>({hw ? hw : 1, 8, (M + 2047) / 2048}));
+    shards.resize(T);
+    std::vector<std::thread> threads;
+    std::atomic<bool> failed{false};
+    std::string error_msg;
+    std::mutex err_mu;
+    for (size_t t = 0; t < T; ++t) {
+      threads.emplace_back([&, t]() {
+        const size_t lo = M * t / T, hi = M * (t + 1) / T;
+        EvShard& sh = shards[t];
+        sh.cols.reserve(hi - lo);
+        try {
+          Parser p(nullptr, 0);
+          for (size_t k = lo; k < hi; ++k) {
+            p.reset_to(sv.data() + lines[k].first,
+                       lines[k].second - lines[k].first);
+            p.skip_ws();
+            if (p.p_ >= p.end_) continue;
+            const JValue* ev = p.parse_value();
+            const std::string_view ty = get_str(ev, "type");
+            const JValue* pod = get(ev, "object");
+            if (pod && pod->kind == JValue::Obj) {
+              encode_one(pod, ctx, sh.cols);
+              sh.types.push_back(ty == "DELETED" ? 1
+                                 : (ty == "ADDED" || ty == "MODIFIED") ? 0
+                                                                       : -1);
+            }
+          }
+        } catch (const std::exception& e) {
+          std::lock_guard<std::mutex> g(err_mu);
+          failed = true;
+          error_msg = e.what();
+        }
+      });
+    }
+    for (auto& th : threads) th.join();
+    if (failed) throw std::runtime_error("watch decode failed: " + error_msg);
+  }
+  Columns out;
+  std::vector<int8_t> types;
+  {
+    size_t total = 0;
+    for (auto& sh : shards) total += sh.types.size();
+    out.reserve(total);
+    types.reserve(total);
+    for (auto& sh : shards) {
+      auto app = [](auto& dst, auto& src) { dst.insert(dst.end(), src.begin(), src.end()); };
+      app(out.pod_req, sh.cols.pod_req);
+      app(out.pod_flags, sh.cols.pod_flags);
+      app(out.pod_sel_hash, sh.cols.pod_sel_hash);
+      app(out.pod_sel_pairs, sh.cols.pod_sel_pairs);
+      app(out.pod_sel_count, sh.cols.pod_sel_count);
+      app(out.pod_node, sh.cols.pod_node);
+      app(out.pod_pdb, sh.cols.pod_pdb);
+      for (auto& u : sh.cols.uids) out.uids.push_back(std::move(u));
+      for (auto& x : sh.cols.ns) out.ns.push_back(std::move(x));
+      for (auto& x : sh.cols.names) out.names.push_back(std::move(x));
+      app(types, sh.types);
+    }
+  }
+
+  const ssize_t P = (ssize_t)out.pod_flags.size();
+  auto make = [&](auto& vec, int width, auto tag) {
+    using T = decltype(tag);
+    py::array_t<T> arr(width == 1 ? std::vector<ssize_t>{P}
+                                  : std::vector<ssize_t>{P, (ssize_t)width});
+    std::memcpy(arr.mutable_data(), vec.data(), vec.size() * sizeof(T));
+    return arr;
+  };
+  py::dict result;
+  result["types"] = make(types, 1, int8_t{});
+  result["pod_req"] = make(out.pod_req, NRES, int32_t{});
+  result["pod_flags"] = make(out.pod_flags, 1, int32_t{});
+  result["pod_sel_hash"] = make(out.pod_sel_hash, 1, int64_t{});
+  result["pod_sel_pairs"] = make(out.pod_sel_pairs, MAXSEL, int64_t{});
+  result["pod_sel_count"] = make(out.pod_sel_count, 1, int32_t{});
+  result["pod_node"] = make(out.pod_node, 1, int32_t{});
+  result["pod_pdb"] = make(out.pod_pdb, 1, int32_t{});
+  PyObject* uids = PyList_New(P);
+  PyObject* refs = PyList_New(P);
+  for (ssize_t i = 0; i < P; ++i) {
+    PyList_SET_ITEM(uids, i,
+                    PyUnicode_FromStringAndSize(out.uids[i].data(), (Py_ssize_t)out.uids[i].size()));
+    PyObject* tup = PyTuple_New(2);
+    PyTuple_SET_ITEM(tup, 0,
+                     PyUnicode_FromStringAndSize(out.ns[i].data(), (Py_ssize_t)out.ns[i].size()));
+    PyTuple_SET_ITEM(tup, 1,
+                     PyUnicode_FromStringAndSize(out.names[i].data(), (Py_ssize_t)out.names[i].size()));
+    PyList_SET_ITEM(refs, i, tup);
+  }
+  result["pod_uids"] = py::reinterpret_steal<py::list>(uids);
+  result["pod_refs"] = py::reinterpret_steal<py::list>(refs);
+  return result;
+}

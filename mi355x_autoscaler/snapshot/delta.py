@@ -164,6 +164,57 @@ class DeltaSnapshotCache:
                 self._clear_row(row)
                 self.free_rows.append(row)
 
+    def apply_batch(self, decoded: Dict) -> int:
+        """Apply a C++-decoded watch batch (decode_watch_events output):
+        per-event row resolution stays Python (a dict op), the column
+        writes are vectorized. Returns events applied."""
+        types = decoded["types"]
+        n = int(types.shape[0])
+        if n == 0:
+            return 0
+        uids = decoded["pod_uids"]
+        refs = decoded["pod_refs"]
+        rows = np.empty(n, np.int64)
+        upsert_idx: List[int] = []
+        for i in range(n):
+            t = types[i]
+            uid = uids[i]
+            if t == 1:  # DELETED
+                row = self.uid_row.pop(uid, None)
+                if row is not None:
+                    self._clear_row(row)
+                    self.free_rows.append(row)
+                continue
+            if t != 0:
+                continue
+            row = self.uid_row.get(uid)
+            if row is None:
+                if self.free_rows:
+                    row = self.free_rows.pop()
+                else:
+                    if self.size >= self.cap:
+                        self._grow()
+                    row = self.size
+                    self.size += 1
+                self.uid_row[uid] = row
+            rows[len(upsert_idx)] = row
+            upsert_idx.append(i)
+            self.pod_uids[row] = uid
+            self.pod_refs[row] = tuple(refs[i])
+            self.dirty.add(row)
+        if upsert_idx:
+            sel = np.asarray(upsert_idx, np.int64)
+            dst = rows[: len(upsert_idx)]
+            a = self.arrays
+            a["pod_req"][dst] = decoded["pod_req"][sel]
+            a["pod_flags"][dst] = decoded["pod_flags"][sel]
+            a["pod_sel_hash"][dst] = decoded["pod_sel_hash"][sel]
+            a["pod_sel_pairs"][dst] = decoded["pod_sel_pairs"][sel]
+            a["pod_sel_count"][dst] = decoded["pod_sel_count"][sel]
+            a["pod_node"][dst] = decoded["pod_node"][sel]
+            a["pod_pdb"][dst] = decoded["pod_pdb"][sel]
+        return n
+
     def drain_dirty(self) -> np.ndarray:
         """Rows touched since the last call (the device scatter set)."""
         rows = np.array(sorted(self.dirty), np.int64)

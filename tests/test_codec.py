@@ -113,3 +113,52 @@ def test_codec_decision_parity():
     d2 = CpuEngine().decide(fast)
     assert d1.units_needed == d2.units_needed
     assert d1.pods_placed_existing == d2.pods_placed_existing
+
+
+def test_decode_watch_events_batch_matches_per_event():
+    """C++ batch decode + vectorized apply == per-event Python apply."""
+    import mi355x_autoscaler._hip_ops as ops
+    from mi355x_autoscaler.models.kube import KubeNode
+    from mi355x_autoscaler.snapshot.delta import DeltaSnapshotCache
+
+    nodes = [KubeNode(make_node_json("agentpool1", i)) for i in range(3)]
+    pools = group_nodes_into_pools(nodes)
+    base = encode_snapshot([], nodes, pools)
+    node_index = {n.name: i for i, n in enumerate(nodes)}
+
+    initial = [make_pod_json(f"p{i}", cpu_milli=100, pending=True) for i in range(10)]
+    from mi355x_autoscaler.models.kube import KubePod
+
+    snap0 = encode_snapshot([KubePod(p) for p in initial], nodes, pools)
+
+    events = []
+    for i in range(30):
+        events.append({"type": "ADDED", "object": make_pod_json(
+            f"n{i}", cpu_milli=200 + i, mem_mib=64, pending=True)})
+    events.append({"type": "MODIFIED", "object": make_pod_json(
+        "p0", cpu_milli=999, node_name=nodes[0].name)})
+    events.append({"type": "DELETED", "object": initial[1]})
+    events.append({"type": "BOOKMARK", "object": initial[2]})  # ignored kind
+    blob = ("\n".join(json.dumps(e) for e in events) + "\n").encode()
+
+    # path A: per-event python
+    ca = DeltaSnapshotCache()
+    ca.prime(snap0, node_index)
+    for e in events:
+        if e["type"] in ("ADDED", "MODIFIED", "DELETED"):
+            ca.apply_event(e)
+
+    # path B: C++ batch decode + vectorized apply
+    cb = DeltaSnapshotCache()
+    cb.prime(snap0, node_index)
+    decoded = ops.decode_watch_events(blob, node_index, [])
+    cb.apply_batch(decoded)
+
+    sa = ca.build_snapshot(base)
+    sb = cb.build_snapshot(base)
+    for col in POD_COLS:
+        order_a = np.argsort(np.array(sa.pod_uids, dtype=object).astype(str))
+        order_b = np.argsort(np.array(sb.pod_uids, dtype=object).astype(str))
+        np.testing.assert_array_equal(
+            getattr(sa, col)[order_a], getattr(sb, col)[order_b], err_msg=col)
+    assert sorted(u for u in sa.pod_uids if u) == sorted(u for u in sb.pod_uids if u)
