@@ -81,16 +81,11 @@ extern "C" __global__ __launch_bounds__(BP_BLOCK) void binpack_kernel(
 // Units phase — residual pods bin-pack into virtual new nodes of the
 // segment's routed pool, FFD in global segment order, capped at the
 // pool's max size [R: autoscaler/cluster.py units-needed + agent_pool
-// max — reconstructed]. Work is tiny (G x <=max_new units); one wave
-// runs it unit-parallel with the same scan-threshold math the oracle
-// uses sequentially, keeping device-resident state so the decision
+// max — reconstructed], keeping device-resident state so the decision
 // vector (units per pool) never round-trips to the host before the
 // RCCL all-reduce.
 //
-//   units_free: [K, MAXU, NRES] int64 scratch (open virtual nodes)
-//   units_cnt:  [K] int32 (doubles as units_needed output)
-//   pool_residual: [K, NRES] int64 residual demand totals
-// Run-length units phase. Units opened in one batch share identical
+// Run-length representation: units opened in one batch share identical
 // residual vectors, and one placement step splits at most ONE run
 // (the partially-consumed one) into <=3 runs, so exact per-unit FFD
 // over an UNBOUNDED unit count is a walk over <=4G+K runs per pool —
@@ -234,10 +229,10 @@ extern "C" __global__ __launch_bounds__(WAVE) void units_kernel(
       bool fits = true;
 #pragma unroll
       for (int r = 0; r < NRES; ++r) {
-        const int64_t cap = pool_unit[k * NRES + r];
-        if (rq[r] > cap) fits = false;
+        const int64_t ucap = pool_unit[k * NRES + r];  // not the arena cap
+        if (rq[r] > ucap) fits = false;
         if (rq[r] > 0) {
-          const int64_t q = cap / rq[r];
+          const int64_t q = ucap / rq[r];
           per_unit = q < per_unit ? q : per_unit;
         }
       }
