@@ -369,6 +369,71 @@ class GpuEngine(SimulationEngine):
         self.ws_seg_pool = torch.zeros(SEG_MAX, dtype=torch.int32, device=d)
         self.ws_run_scratch = torch.empty((4 * SEG_MAX + 8) * 6, dtype=torch.int64, device=d)
         self._elig_cache: Dict[int, torch.Tensor] = {}  # W -> [SEG_MAX, W]
+        # hipGraph capture of the fused pipeline (OPT-IN, measured
+        # slower in situ): the native orchestrator already collapsed
+        # Python dispatch, so the pipeline is GPU-bound, not launch-
+        # bound — isolated replay is 0.097 ms vs 0.091 ms eager on
+        # MI355X/ROCm 7.2, and inside the full decide (pinned H2D +
+        # readback on the same stream) replay regressed 0.22 -> 0.89 ms
+        # with multi-ms spikes (profiles/hipgraph_probe.txt). Kept as
+        # MI355X_AS_HIPGRAPH=1 with warm-up-then-capture keyed by the
+        # full data_ptr tuple; capture failure reverts to eager.
+        self._graph_ok = (os.environ.get("MI355X_AS_HIPGRAPH") == "1"
+                          and hasattr(torch.cuda, "CUDAGraph"))
+        self._graph_cache: Dict[tuple, object] = {}
+        self._decide_bufs: Dict[tuple, dict] = {}
+
+    def _decide_buffers(self, P: int, K: int, want_pod_pool: bool) -> dict:
+        """Per-shape persistent decide outputs (slot_of_pod / blob /
+        max_new / pod_pool): every element is fully rewritten each
+        decide (blob via the orchestrator's memset), so reuse is safe —
+        and stable pointers are what makes hipGraph replay legal."""
+        bkey = (P, K, want_pod_pool)
+        bufs = self._decide_bufs.get(bkey)
+        if bufs is None:
+            d = self.device
+            bufs = {
+                "slot_of_pod": torch.empty(P, dtype=torch.int32, device=d),
+                "blob": torch.empty(3 * SEG_MAX + 5 * max(K, 1) + 2,
+                                    dtype=torch.int64, device=d),
+                "max_new": torch.empty(max(K, 1), dtype=torch.int32, device=d),
+                "pod_pool": torch.empty(P if want_pod_pool else 0,
+                                        dtype=torch.int32, device=d),
+            }
+            if len(self._decide_bufs) > 4:
+                self._decide_bufs.clear()
+            self._decide_bufs[bkey] = bufs
+        return bufs
+
+    def _launch_graphed(self, fn, tensors: tuple, scalars: tuple):
+        """Run fn(*tensors, *scalars) through a cached hipGraph when the
+        pointer-set has been seen before (warm-up eagerly once, capture
+        on second sight, replay afterwards)."""
+        if not self._graph_ok:
+            fn(*tensors, *scalars)
+            return
+        gkey = (fn.__name__, scalars) + tuple(t.data_ptr() for t in tensors)
+        ent = self._graph_cache.get(gkey)
+        if ent is None:
+            fn(*tensors, *scalars)  # warm-up: module/kernel lazy init
+            if len(self._graph_cache) > 8:
+                self._graph_cache.clear()
+            self._graph_cache[gkey] = "warm"
+        elif ent == "warm":
+            try:
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    fn(*tensors, *scalars)
+                g.replay()
+                self._graph_cache[gkey] = g
+            except Exception as e:  # capture unsupported: stay eager
+                logger = __import__("logging").getLogger(__name__)
+                logger.warning("hipGraph capture failed (%s); eager enqueue", e)
+                self._graph_ok = False
+                self._graph_cache.clear()
+                fn(*tensors, *scalars)
+        else:
+            ent.replay()
 
     # ------------------------------------------------------------------
     def build_segments_hash(self, ds: DeviceSnapshot):
@@ -452,30 +517,30 @@ class GpuEngine(SimulationEngine):
 
         W = (N + 63) // 64
         elig = self._elig_buffer(W)
-        slot_of_pod = torch.empty(ds.P, dtype=torch.int32, device=device)
-        # +2 tail slots: the orchestrator copies the 3-word i32 stat there
-        blob = torch.empty(3 * SEG_MAX + 5 * max(K, 1) + 2, dtype=torch.int64,
-                           device=device)
-        max_new = torch.empty(max(K, 1), dtype=torch.int32, device=device)
         want_pod_pool = bool(K and K <= 16)
-        pod_pool = torch.empty(ds.P if want_pod_pool else 0, dtype=torch.int32,
-                               device=device)
+        bufs = self._decide_buffers(ds.P, K, want_pod_pool)
+        slot_of_pod = bufs["slot_of_pod"]
+        # +2 tail slots: the orchestrator copies the 3-word i32 stat there
+        blob = bufs["blob"]
+        max_new = bufs["max_new"]
+        pod_pool = bufs["pod_pool"]
 
         if self.force_py_pipeline:
             self._enqueue_py(ds, slot_of_pod, elig, blob, max_new, pod_pool,
                              want_pod_pool)
         else:
-            ops.fused_decide_launch(
-                ds.pod_req, ds.pod_flags, ds.pod_sel_hash, ds.pod_sel_pairs,
-                ds.pod_sel_count, ds.pod_node, ds.node_alloc, ds.node_labels,
-                ds.node_label_count, ds.node_unsched, ds.pool_unit,
-                ds.pool_labels, ds.pool_label_count, ds.pool_max, ds.pool_cur,
-                self._ws_i32, self.ws_slot_ks, self.ws_slot_kr, self.ws_slot2seg,
-                self.ws_seg_sel_pairs, self.ws_seg_sel_count, self.ws_seg_sel_hash,
-                self.ws_seg_req, self.ws_seg_count, self.ws_seg_key, self.ws_seg_first,
-                slot_of_pod, self._free_buffer(N), elig, self.ws_seg_pool, max_new,
-                self.ws_run_scratch, blob, pod_pool, HASH_SLOTS, SEG_MAX,
-                1 if want_pod_pool else 0,
+            self._launch_graphed(
+                ops.fused_decide_launch,
+                (ds.pod_req, ds.pod_flags, ds.pod_sel_hash, ds.pod_sel_pairs,
+                 ds.pod_sel_count, ds.pod_node, ds.node_alloc, ds.node_labels,
+                 ds.node_label_count, ds.node_unsched, ds.pool_unit,
+                 ds.pool_labels, ds.pool_label_count, ds.pool_max, ds.pool_cur,
+                 self._ws_i32, self.ws_slot_ks, self.ws_slot_kr, self.ws_slot2seg,
+                 self.ws_seg_sel_pairs, self.ws_seg_sel_count, self.ws_seg_sel_hash,
+                 self.ws_seg_req, self.ws_seg_count, self.ws_seg_key, self.ws_seg_first,
+                 slot_of_pod, self._free_buffer(N), elig, self.ws_seg_pool, max_new,
+                 self.ws_run_scratch, blob, pod_pool),
+                (HASH_SLOTS, SEG_MAX, 1 if want_pod_pool else 0),
             )
 
         # MFMA per-pool demand (zero-sort path; skipped for >16 pools)
@@ -814,20 +879,24 @@ class GpuEngine(SimulationEngine):
         if free_hook is not None:
             free.copy_(ds.node_alloc)
             self.ops.node_free_subtract(ds.pod_node, ds.pod_req, free)
-            free = free_hook(free, ds)
+            hooked = free_hook(free, ds)
+            if hooked is not free:  # pin the pointer for graph replay
+                free.copy_(hooked)
             want_free = 0
         W = (N + 63) // 64
         elig = self._elig_buffer(W)
-        blob = torch.empty(3 * SEG_MAX + 5 * max(K, 1) + 2, dtype=torch.int64,
-                           device=device)
-        max_new = torch.empty(max(K, 1), dtype=torch.int32, device=device)
-        self.ops.fused_tail_launch(
-            ds.pod_req, ds.pod_node, ds.node_alloc, ds.node_labels,
-            ds.node_label_count, ds.node_unsched, ds.pool_unit, ds.pool_labels,
-            ds.pool_label_count, ds.pool_max, ds.pool_cur, self.ws_stat,
-            self.ws_seg_sel_pairs, self.ws_seg_sel_count, self.ws_seg_req,
-            self.ws_seg_count, free, elig, self.ws_seg_pool, max_new,
-            self.ws_run_scratch, blob, SEG_MAX, want_free,
+        bufs = self._decide_buffers(1, K, False)
+        blob = bufs["blob"]
+        max_new = bufs["max_new"]
+        self._launch_graphed(
+            self.ops.fused_tail_launch,
+            (ds.pod_req, ds.pod_node, ds.node_alloc, ds.node_labels,
+             ds.node_label_count, ds.node_unsched, ds.pool_unit, ds.pool_labels,
+             ds.pool_label_count, ds.pool_max, ds.pool_cur, self.ws_stat,
+             self.ws_seg_sel_pairs, self.ws_seg_sel_count, self.ws_seg_req,
+             self.ws_seg_count, free, elig, self.ws_seg_pool, max_new,
+             self.ws_run_scratch, blob),
+            (SEG_MAX, want_free),
         )
         self._last_demand = None
 

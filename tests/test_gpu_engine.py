@@ -446,3 +446,27 @@ def test_no_pools_cluster(engine):
     assert d_gpu.pods_placed_existing == d_cpu.pods_placed_existing == 4
     assert d_gpu.pods_unfulfillable == d_cpu.pods_unfulfillable == 1
     assert not d_gpu.wants_scale_up()
+
+
+def test_hipgraph_optin_replay_matches_eager_and_oracle(engine, monkeypatch):
+    """hipGraph capture is opt-in (measured slower in situ on ROCm 7.2 —
+    see engine_gpu.__init__ and profiles/hipgraph_probe.txt); when
+    enabled, warm-up -> capture -> replay must still produce
+    bit-identical decisions, and the graph cache must actually be
+    exercised."""
+    from mi355x_autoscaler.ops.engine_gpu import GpuEngine
+
+    assert not engine._graph_ok  # default is eager
+    monkeypatch.setenv("MI355X_AS_HIPGRAPH", "1")
+    graphed = GpuEngine("cuda:0")
+    assert graphed._graph_ok
+    pods_json, nodes_json = burst_trace(
+        2000, n_nodes_per_pool=(5,), seed=11, bound_pods_per_node=2)
+    snap = _snap(pods_json, nodes_json)
+    ref = CpuEngine().decide(snap)
+    for _ in range(4):
+        _assert_decisions_equal(graphed.decide(snap), ref)
+        _assert_decisions_equal(engine.decide(snap), ref)
+    if graphed._graph_ok:  # capture may legitimately disable itself
+        assert any(not isinstance(v, str) for v in graphed._graph_cache.values()), \
+            "expected at least one captured hipGraph after repeated decides"
