@@ -125,8 +125,13 @@ void fused_decide_launch(
                   g_dev, run_scratch, units_cnt, placed_new, unful,
                   pool_residual);
   }
-  // piggyback the 3-word stat (G, sort-ovf, hash-ovf) onto the blob
-  // tail so the host needs exactly ONE readback
+  // piggyback the segment sel-hash column and the 3-word stat
+  // (G, sort-ovf, hash-ovf) onto the blob tail so the host needs
+  // exactly ONE readback (the sel-hash feeds unfulfillable_detail —
+  // without the mirror that path cost a second D2H sync per decide)
+  hipMemcpyAsync(blob.data_ptr<int64_t>() + (blob.numel() - 2 - SM),
+                 seg_sel_hash.data_ptr<int64_t>(), (size_t)SM * sizeof(int64_t),
+                 hipMemcpyDeviceToDevice, stream);
   hipMemcpyAsync(blob.data_ptr<int64_t>() + (blob.numel() - 2),
                  stat.data_ptr<int32_t>(), 3 * sizeof(int32_t),
                  hipMemcpyDeviceToDevice, stream);

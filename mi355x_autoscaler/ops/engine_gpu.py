@@ -392,10 +392,15 @@ class GpuEngine(SimulationEngine):
         bufs = self._decide_bufs.get(bkey)
         if bufs is None:
             d = self.device
+            nblob = 3 * SEG_MAX + 5 * max(K, 1) + SEG_MAX + 2
             bufs = {
                 "slot_of_pod": torch.empty(P, dtype=torch.int32, device=d),
-                "blob": torch.empty(3 * SEG_MAX + 5 * max(K, 1) + 2,
-                                    dtype=torch.int64, device=d),
+                # layout: placed|placed_new|unful [3*SM] · units [K] ·
+                # residual [4K] · slack · seg_sel_hash mirror [SM] ·
+                # stat tail [2] — ONE readback serves the whole decision
+                "blob": torch.empty(nblob, dtype=torch.int64, device=d),
+                "blob_host": torch.empty(nblob, dtype=torch.int64,
+                                         pin_memory=torch.cuda.is_available()),
                 "max_new": torch.empty(max(K, 1), dtype=torch.int32, device=d),
                 "pod_pool": torch.empty(P if want_pod_pool else 0,
                                         dtype=torch.int32, device=d),
@@ -549,8 +554,10 @@ class GpuEngine(SimulationEngine):
         else:
             self._last_demand = None
 
-        # THE sync: one readback (stat rides on the blob tail)
-        blob_h = blob.cpu().numpy()
+        # THE sync: one readback into the pinned mirror (stat and the
+        # segment sel-hash column ride on the blob tail)
+        bufs["blob_host"].copy_(blob)
+        blob_h = bufs["blob_host"].numpy()
         if self.force_py_pipeline:
             stat_h = self.ws_stat.cpu().numpy()
         else:
@@ -578,7 +585,7 @@ class GpuEngine(SimulationEngine):
             blob_h[3 * SEG_MAX : 3 * SEG_MAX + K],
             blob_h[3 * SEG_MAX + K : 3 * SEG_MAX + 5 * K].reshape(max(K, 1) if K else 0, 4) if K else np.zeros((0, 4), np.int64),
             max_new_h, ds.pool_cur_h, K, pool_names, over_provision,
-            lambda: self.ws_seg_sel_hash[:G].cpu().numpy(),
+            lambda: blob_h[-2 - SEG_MAX : -2],
         )
 
     def _free_buffer(self, N: int) -> torch.Tensor:
@@ -636,6 +643,7 @@ class GpuEngine(SimulationEngine):
                               placed, ds.pool_unit, max_new, g_dev,
                               self.ws_run_scratch, units_cnt, placed_new,
                               unful, pool_residual)
+        blob[-2 - SEG_MAX : -2].copy_(self.ws_seg_sel_hash)
 
     @staticmethod
     def _finish_decision(decision, placed_h, placed_new_h, unful_h, units_h,
@@ -900,7 +908,8 @@ class GpuEngine(SimulationEngine):
         )
         self._last_demand = None
 
-        blob_h = blob.cpu().numpy()
+        bufs["blob_host"].copy_(blob)
+        blob_h = bufs["blob_host"].numpy()
         decision = ScaleDecision()
         placed_h = blob_h[0:G]
         unful_h = blob_h[2 * SEG_MAX : 2 * SEG_MAX + G]
