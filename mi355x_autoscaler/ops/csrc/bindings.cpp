@@ -20,10 +20,14 @@ void binpack_place(torch::Tensor free, torch::Tensor elig, torch::Tensor seg_req
                    torch::Tensor placed_out);
 void binpack_units(torch::Tensor seg_pool, torch::Tensor seg_req,
                    torch::Tensor seg_count, torch::Tensor placed,
-                   torch::Tensor pool_unit, torch::Tensor pool_max_new,
-                   torch::Tensor g_dev, torch::Tensor run_scratch,
-                   torch::Tensor units_cnt, torch::Tensor placed_new,
-                   torch::Tensor unful, torch::Tensor pool_residual);
+                   torch::Tensor pool_unit, torch::Tensor pool_max,
+                   torch::Tensor pool_cur, torch::Tensor g_dev,
+                   torch::Tensor run_scratch, torch::Tensor units_cnt,
+                   torch::Tensor placed_new, torch::Tensor unful,
+                   torch::Tensor pool_residual,
+                   c10::optional<torch::Tensor> seg_sel_hash,
+                   c10::optional<torch::Tensor> stat,
+                   c10::optional<torch::Tensor> blob);
 void drain_flags(torch::Tensor pod_node, torch::Tensor pod_flags,
                  torch::Tensor pod_pdb, torch::Tensor pdb_allowed,
                  torch::Tensor blocked, torch::Tensor pdb_bad,
@@ -59,8 +63,6 @@ void pack_into(pybind11::buffer buf, const std::vector<int64_t>& offsets,
 pybind11::dict decode_watch_events(pybind11::bytes body,
                                    pybind11::dict node_index,
                                    pybind11::list pdb_selectors);
-void pool_max_new(torch::Tensor pool_max, torch::Tensor pool_cur,
-                  torch::Tensor out);
 void fused_tail_launch(
     torch::Tensor pod_req, torch::Tensor pod_node, torch::Tensor node_alloc,
     torch::Tensor node_labels, torch::Tensor node_label_count,
@@ -69,9 +71,10 @@ void fused_tail_launch(
     torch::Tensor pool_max, torch::Tensor pool_cur,
     torch::Tensor stat,
     torch::Tensor seg_sel_pairs, torch::Tensor seg_sel_count,
+    torch::Tensor seg_sel_hash,
     torch::Tensor seg_req, torch::Tensor seg_count,
     torch::Tensor free, torch::Tensor elig, torch::Tensor seg_pool,
-    torch::Tensor max_new, torch::Tensor run_scratch, torch::Tensor blob,
+    torch::Tensor run_scratch, torch::Tensor blob,
     int64_t seg_max, int64_t want_free);
 void fused_decide_launch(
     torch::Tensor pod_req, torch::Tensor pod_flags, torch::Tensor pod_sel_hash,
@@ -86,7 +89,7 @@ void fused_decide_launch(
     torch::Tensor seg_sel_hash, torch::Tensor seg_req, torch::Tensor seg_count,
     torch::Tensor seg_key, torch::Tensor seg_first,
     torch::Tensor slot_of_pod, torch::Tensor free, torch::Tensor elig,
-    torch::Tensor seg_pool, torch::Tensor max_new, torch::Tensor run_scratch,
+    torch::Tensor seg_pool, torch::Tensor run_scratch,
     torch::Tensor blob, torch::Tensor pod_pool, int64_t hash_slots,
     int64_t seg_max, int64_t want_pod_pool);
 void snapshot_scatter(torch::Tensor rows, torch::Tensor u_req,
@@ -105,7 +108,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fit_eligibility", &fit_eligibility, "kernel A: segment x node eligibility");
   m.def("seg_pool_route", &seg_pool_route, "segment -> pool routing");
   m.def("binpack_place", &binpack_place, "kernel B: FFD placement steps");
-  m.def("binpack_units", &binpack_units, "kernel B: units-needed phase");
+  m.def("binpack_units", &binpack_units, "kernel B: units-needed phase",
+        pybind11::arg("seg_pool"), pybind11::arg("seg_req"),
+        pybind11::arg("seg_count"), pybind11::arg("placed"),
+        pybind11::arg("pool_unit"), pybind11::arg("pool_max"),
+        pybind11::arg("pool_cur"), pybind11::arg("g_dev"),
+        pybind11::arg("run_scratch"), pybind11::arg("units_cnt"),
+        pybind11::arg("placed_new"), pybind11::arg("unful"),
+        pybind11::arg("pool_residual"),
+        pybind11::arg("seg_sel_hash") = pybind11::none(),
+        pybind11::arg("stat") = pybind11::none(),
+        pybind11::arg("blob") = pybind11::none());
   m.def("drain_flags", &drain_flags, "kernel D: drain predicates");
   m.def("drain_finalize", &drain_finalize, "kernel D: utilization + candidates");
   m.def("segsum_mfma", &segsum_mfma, "MFMA segmented resource sums");
@@ -119,7 +132,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_into", &pack_into, "memcpy arrays into the staging buffer");
   m.def("decode_watch_events", &decode_watch_events,
         "JSONL watch stream -> event types + pod SoA columns");
-  m.def("pool_max_new", &pool_max_new, "clamp(pool_max - pool_cur, 0)");
   m.def("fused_decide_launch", &fused_decide_launch,
         "enqueue the whole scale-up pipeline in one call");
   m.def("fused_tail_launch", &fused_tail_launch,

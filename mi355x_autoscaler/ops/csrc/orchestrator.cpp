@@ -40,12 +40,14 @@ void binpack_place(torch::Tensor free, torch::Tensor elig, torch::Tensor seg_req
                    torch::Tensor placed_out);
 void binpack_units(torch::Tensor seg_pool, torch::Tensor seg_req,
                    torch::Tensor seg_count, torch::Tensor placed,
-                   torch::Tensor pool_unit, torch::Tensor pool_max_new,
-                   torch::Tensor g_dev, torch::Tensor run_scratch,
-                   torch::Tensor units_cnt, torch::Tensor placed_new,
-                   torch::Tensor unful, torch::Tensor pool_residual);
-void pool_max_new(torch::Tensor pool_max, torch::Tensor pool_cur,
-                  torch::Tensor out);
+                   torch::Tensor pool_unit, torch::Tensor pool_max,
+                   torch::Tensor pool_cur, torch::Tensor g_dev,
+                   torch::Tensor run_scratch, torch::Tensor units_cnt,
+                   torch::Tensor placed_new, torch::Tensor unful,
+                   torch::Tensor pool_residual,
+                   c10::optional<torch::Tensor> seg_sel_hash,
+                   c10::optional<torch::Tensor> stat,
+                   c10::optional<torch::Tensor> blob);
 void pod_pool_map(torch::Tensor slot_of_pod, torch::Tensor slot2seg,
                   torch::Tensor seg_pool, torch::Tensor pod_pool);
 
@@ -66,7 +68,7 @@ void fused_decide_launch(
     torch::Tensor seg_key, torch::Tensor seg_first,
     // per-decide buffers
     torch::Tensor slot_of_pod, torch::Tensor free, torch::Tensor elig,
-    torch::Tensor seg_pool, torch::Tensor max_new, torch::Tensor run_scratch,
+    torch::Tensor seg_pool, torch::Tensor run_scratch,
     torch::Tensor blob, torch::Tensor pod_pool, int64_t hash_slots,
     int64_t seg_max, int64_t want_pod_pool) {
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
@@ -75,11 +77,15 @@ void fused_decide_launch(
   const int64_t K = pool_unit.size(0);
   const int64_t N = node_alloc.size(0);
 
-  // workspace reinit + blob zero: two async memsets
+  // workspace reinit; the blob needs zeroing only when a producer
+  // kernel is skipped (binpack writes placed[:G], units writes the
+  // rest + the readback epilogue)
   hipMemsetAsync(ws_i32.data_ptr<int32_t>(), 0,
                  (size_t)ws_i32.numel() * sizeof(int32_t), stream);
-  hipMemsetAsync(blob.data_ptr<int64_t>(), 0,
-                 (size_t)blob.numel() * sizeof(int64_t), stream);
+  if (N == 0 || K == 0) {
+    hipMemsetAsync(blob.data_ptr<int64_t>(), 0,
+                   (size_t)blob.numel() * sizeof(int64_t), stream);
+  }
 
   auto slot_state = ws_i32.narrow(0, 0, H);
   auto slot_count = ws_i32.narrow(0, H, H);
@@ -120,21 +126,22 @@ void fused_decide_launch(
   if (K > 0) {
     auto units_cnt = blob.narrow(0, 3 * SM, K);
     auto pool_residual = blob.narrow(0, 3 * SM + K, 4 * K);
-    pool_max_new(pool_max, pool_cur, max_new);
-    binpack_units(seg_pool, seg_req, seg_count, placed, pool_unit, max_new,
-                  g_dev, run_scratch, units_cnt, placed_new, unful,
-                  pool_residual);
+    // units also runs the readback epilogue: segment sel-hash column +
+    // 3-word stat (G, sort-ovf, hash-ovf) land on the blob tail so the
+    // host needs exactly ONE readback (the sel-hash feeds
+    // unfulfillable_detail — without the mirror that path cost a
+    // second D2H sync per decide)
+    binpack_units(seg_pool, seg_req, seg_count, placed, pool_unit, pool_max,
+                  pool_cur, g_dev, run_scratch, units_cnt, placed_new, unful,
+                  pool_residual, seg_sel_hash.narrow(0, 0, SM), stat, blob);
+  } else {
+    hipMemcpyAsync(blob.data_ptr<int64_t>() + (blob.numel() - 2 - SM),
+                   seg_sel_hash.data_ptr<int64_t>(), (size_t)SM * sizeof(int64_t),
+                   hipMemcpyDeviceToDevice, stream);
+    hipMemcpyAsync(blob.data_ptr<int64_t>() + (blob.numel() - 2),
+                   stat.data_ptr<int32_t>(), 3 * sizeof(int32_t),
+                   hipMemcpyDeviceToDevice, stream);
   }
-  // piggyback the segment sel-hash column and the 3-word stat
-  // (G, sort-ovf, hash-ovf) onto the blob tail so the host needs
-  // exactly ONE readback (the sel-hash feeds unfulfillable_detail —
-  // without the mirror that path cost a second D2H sync per decide)
-  hipMemcpyAsync(blob.data_ptr<int64_t>() + (blob.numel() - 2 - SM),
-                 seg_sel_hash.data_ptr<int64_t>(), (size_t)SM * sizeof(int64_t),
-                 hipMemcpyDeviceToDevice, stream);
-  hipMemcpyAsync(blob.data_ptr<int64_t>() + (blob.numel() - 2),
-                 stat.data_ptr<int32_t>(), 3 * sizeof(int32_t),
-                 hipMemcpyDeviceToDevice, stream);
 }
 
 
@@ -151,9 +158,10 @@ void fused_tail_launch(
     torch::Tensor pool_max, torch::Tensor pool_cur,
     torch::Tensor stat,            // [3] i32: G, 0, 0 (host-set)
     torch::Tensor seg_sel_pairs, torch::Tensor seg_sel_count,
+    torch::Tensor seg_sel_hash,
     torch::Tensor seg_req, torch::Tensor seg_count,
     torch::Tensor free, torch::Tensor elig, torch::Tensor seg_pool,
-    torch::Tensor max_new, torch::Tensor run_scratch, torch::Tensor blob,
+    torch::Tensor run_scratch, torch::Tensor blob,
     int64_t seg_max, int64_t want_free) {
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   const int64_t SM = seg_max;
@@ -161,8 +169,10 @@ void fused_tail_launch(
   const int64_t N = node_alloc.size(0);
   auto g_dev = stat.narrow(0, 0, 1);
 
-  hipMemsetAsync(blob.data_ptr<int64_t>(), 0,
-                 (size_t)blob.numel() * sizeof(int64_t), stream);
+  if (N == 0 || K == 0) {
+    hipMemsetAsync(blob.data_ptr<int64_t>(), 0,
+                   (size_t)blob.numel() * sizeof(int64_t), stream);
+  }
   if (want_free) {
     hipMemcpyAsync(free.data_ptr<int32_t>(), node_alloc.data_ptr<int32_t>(),
                    (size_t)node_alloc.numel() * sizeof(int32_t),
@@ -184,12 +194,12 @@ void fused_tail_launch(
   if (K > 0) {
     auto units_cnt = blob.narrow(0, 3 * SM, K);
     auto pool_residual = blob.narrow(0, 3 * SM + K, 4 * K);
-    pool_max_new(pool_max, pool_cur, max_new);
-    binpack_units(seg_pool, seg_req, seg_count, placed, pool_unit, max_new,
-                  g_dev, run_scratch, units_cnt, placed_new, unful,
-                  pool_residual);
+    binpack_units(seg_pool, seg_req, seg_count, placed, pool_unit, pool_max,
+                  pool_cur, g_dev, run_scratch, units_cnt, placed_new, unful,
+                  pool_residual, seg_sel_hash.narrow(0, 0, SM), stat, blob);
+  } else {
+    hipMemcpyAsync(blob.data_ptr<int64_t>() + (blob.numel() - 2),
+                   stat.data_ptr<int32_t>(), 3 * sizeof(int32_t),
+                   hipMemcpyDeviceToDevice, stream);
   }
-  hipMemcpyAsync(blob.data_ptr<int64_t>() + (blob.numel() - 2),
-                 stat.data_ptr<int32_t>(), 3 * sizeof(int32_t),
-                 hipMemcpyDeviceToDevice, stream);
 }

@@ -367,7 +367,7 @@ class GpuEngine(SimulationEngine):
         self.ws_seg_key = torch.zeros(SEG_MAX, dtype=torch.int64, device=d)
         self.ws_seg_first = torch.zeros(SEG_MAX, dtype=torch.int32, device=d)
         self.ws_seg_pool = torch.zeros(SEG_MAX, dtype=torch.int32, device=d)
-        self.ws_run_scratch = torch.empty((4 * SEG_MAX + 8) * 6, dtype=torch.int64, device=d)
+        self.ws_run_scratch = torch.empty((4 * SEG_MAX + 32) * 6, dtype=torch.int64, device=d)
         self._elig_cache: Dict[int, torch.Tensor] = {}  # W -> [SEG_MAX, W]
         # hipGraph capture of the fused pipeline (OPT-IN, measured
         # slower in situ): the native orchestrator already collapsed
@@ -385,7 +385,7 @@ class GpuEngine(SimulationEngine):
 
     def _decide_buffers(self, P: int, K: int, want_pod_pool: bool) -> dict:
         """Per-shape persistent decide outputs (slot_of_pod / blob /
-        max_new / pod_pool): every element is fully rewritten each
+        pod_pool): every element is fully rewritten each
         decide (blob via the orchestrator's memset), so reuse is safe —
         and stable pointers are what makes hipGraph replay legal."""
         bkey = (P, K, want_pod_pool)
@@ -401,7 +401,6 @@ class GpuEngine(SimulationEngine):
                 "blob": torch.empty(nblob, dtype=torch.int64, device=d),
                 "blob_host": torch.empty(nblob, dtype=torch.int64,
                                          pin_memory=torch.cuda.is_available()),
-                "max_new": torch.empty(max(K, 1), dtype=torch.int32, device=d),
                 "pod_pool": torch.empty(P if want_pod_pool else 0,
                                         dtype=torch.int32, device=d),
             }
@@ -527,11 +526,10 @@ class GpuEngine(SimulationEngine):
         slot_of_pod = bufs["slot_of_pod"]
         # +2 tail slots: the orchestrator copies the 3-word i32 stat there
         blob = bufs["blob"]
-        max_new = bufs["max_new"]
         pod_pool = bufs["pod_pool"]
 
         if self.force_py_pipeline:
-            self._enqueue_py(ds, slot_of_pod, elig, blob, max_new, pod_pool,
+            self._enqueue_py(ds, slot_of_pod, elig, blob, pod_pool,
                              want_pod_pool)
         else:
             self._launch_graphed(
@@ -543,7 +541,7 @@ class GpuEngine(SimulationEngine):
                  self._ws_i32, self.ws_slot_ks, self.ws_slot_kr, self.ws_slot2seg,
                  self.ws_seg_sel_pairs, self.ws_seg_sel_count, self.ws_seg_sel_hash,
                  self.ws_seg_req, self.ws_seg_count, self.ws_seg_key, self.ws_seg_first,
-                 slot_of_pod, self._free_buffer(N), elig, self.ws_seg_pool, max_new,
+                 slot_of_pod, self._free_buffer(N), elig, self.ws_seg_pool,
                  self.ws_run_scratch, blob, pod_pool),
                 (HASH_SLOTS, SEG_MAX, 1 if want_pod_pool else 0),
             )
@@ -595,7 +593,7 @@ class GpuEngine(SimulationEngine):
             self._free_cache = {N: buf}
         return buf
 
-    def _enqueue_py(self, ds, slot_of_pod, elig, blob, max_new, pod_pool,
+    def _enqueue_py(self, ds, slot_of_pod, elig, blob, pod_pool,
                     want_pod_pool):
         """Op-by-op Python sequencing of the same pipeline (reference
         for the native orchestrator; also exercised in tests)."""
@@ -638,12 +636,13 @@ class GpuEngine(SimulationEngine):
         if K:
             units_cnt = blob[3 * SEG_MAX : 3 * SEG_MAX + K]
             pool_residual = blob[3 * SEG_MAX + K : 3 * SEG_MAX + 5 * K]
-            ops.pool_max_new(ds.pool_max, ds.pool_cur, max_new)
             ops.binpack_units(self.ws_seg_pool, self.ws_seg_req, self.ws_seg_count,
-                              placed, ds.pool_unit, max_new, g_dev,
-                              self.ws_run_scratch, units_cnt, placed_new,
-                              unful, pool_residual)
-        blob[-2 - SEG_MAX : -2].copy_(self.ws_seg_sel_hash)
+                              placed, ds.pool_unit, ds.pool_max, ds.pool_cur,
+                              g_dev, self.ws_run_scratch, units_cnt, placed_new,
+                              unful, pool_residual, self.ws_seg_sel_hash,
+                              self.ws_stat, blob)
+        else:
+            blob[-2 - SEG_MAX : -2].copy_(self.ws_seg_sel_hash)
 
     @staticmethod
     def _finish_decision(decision, placed_h, placed_new_h, unful_h, units_h,
@@ -746,16 +745,14 @@ class GpuEngine(SimulationEngine):
             - (ds.pool_cur_h if ds.pool_cur_h is not None else ds.pool_cur.cpu().numpy()),
             0,
         ).astype(np.int32) if K else np.zeros(0, np.int32)
-        max_new = torch.clamp(ds.pool_max - ds.pool_cur, min=0) if K else \
-            torch.zeros(0, dtype=torch.int32, device=device)
         if K:
             # merged sharded segment lists may exceed SEG_MAX
             scratch = self.ws_run_scratch
-            if 4 * G + 8 > scratch.numel() // 6:
-                scratch = torch.empty((4 * G + 8) * 6, dtype=torch.int64, device=device)
+            if 4 * G + 32 > scratch.numel() // 6:
+                scratch = torch.empty((4 * G + 32) * 6, dtype=torch.int64, device=device)
             ops.binpack_units(
                 seg_pool, segs.seg_req.contiguous(), segs.seg_count.contiguous(), placed,
-                ds.pool_unit, max_new, g_dev, scratch, units_cnt,
+                ds.pool_unit, ds.pool_max, ds.pool_cur, g_dev, scratch, units_cnt,
                 placed_new, unful, pool_residual,
             )
 
@@ -895,14 +892,13 @@ class GpuEngine(SimulationEngine):
         elig = self._elig_buffer(W)
         bufs = self._decide_buffers(1, K, False)
         blob = bufs["blob"]
-        max_new = bufs["max_new"]
         self._launch_graphed(
             self.ops.fused_tail_launch,
             (ds.pod_req, ds.pod_node, ds.node_alloc, ds.node_labels,
              ds.node_label_count, ds.node_unsched, ds.pool_unit, ds.pool_labels,
              ds.pool_label_count, ds.pool_max, ds.pool_cur, self.ws_stat,
-             self.ws_seg_sel_pairs, self.ws_seg_sel_count, self.ws_seg_req,
-             self.ws_seg_count, free, elig, self.ws_seg_pool, max_new,
+             self.ws_seg_sel_pairs, self.ws_seg_sel_count, self.ws_seg_sel_hash,
+             self.ws_seg_req, self.ws_seg_count, free, elig, self.ws_seg_pool,
              self.ws_run_scratch, blob),
             (SEG_MAX, want_free),
         )

@@ -441,7 +441,14 @@ void segment_hash(torch::Tensor pod_flags, torch::Tensor sel_hash,
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const int block = 256;
   const int grid = (int)std::min<int64_t>((P + block - 1) / block, 2048);
-  auto kfn = P < 65536 ? segment_hash_build_small : segment_hash_build;
+  // crossover between the direct wave-aggregated form and the LDS
+  // sub-table form; measured ~64k on MI355X, overridable for tuning
+  static int small_max = -1;
+  if (small_max < 0) {
+    const char* e = getenv("MI355X_AS_HASH_SMALL_MAX");
+    small_max = e ? atoi(e) : 65536;
+  }
+  auto kfn = P < small_max ? segment_hash_build_small : segment_hash_build;
   hipLaunchKernelGGL(kfn, dim3(grid), dim3(block), 0,
                      stream.stream(), pod_flags.data_ptr<int32_t>(),
                      sel_hash.data_ptr<int64_t>(), pod_req.data_ptr<int32_t>(),
