@@ -19,7 +19,7 @@
 // HBM label traffic is paid once per block, not once per segment.
 //
 // Grid: ceil(N/256) blocks. Output: [G, W] int64 (W = ceil(N/64)).
-#include "common.h"
+#include "pipeline_device.h"
 
 #define FIT_BLOCK 256
 #define NODES_PER_BLOCK FIT_BLOCK
@@ -36,46 +36,9 @@ extern "C" __global__ __launch_bounds__(FIT_BLOCK) void fit_eligibility_kernel(
   const int G = *g_ptr;  // device-resident so the host never syncs for it
   __shared__ int64_t lds_labels[NODES_PER_BLOCK * MAXLAB];
   __shared__ int32_t lds_nlab[NODES_PER_BLOCK];
-
-  const int tid = threadIdx.x;
-  const int lane = tid & (WAVE - 1);
-  const int wid = tid / WAVE;
-  const int node0 = blockIdx.x * NODES_PER_BLOCK;
-  const int my_node = node0 + tid;
-  const int W = (N + WAVE - 1) / WAVE;
-
-  // stage this block's node labels into LDS, b128 rows: thread t copies
-  // node t's row (coalesced across the row via 4-wide int64 chunks).
-  if (my_node < N) {
-    const int64_t* src = node_labels + (int64_t)my_node * MAXLAB;
-    int64_t* dst = lds_labels + (int64_t)tid * MAXLAB;
-#pragma unroll
-    for (int q = 0; q < MAXLAB; ++q) dst[q] = src[q];
-    int nl = node_label_count[my_node];
-    // fold schedulability into the count: unschedulable => impossible
-    lds_nlab[tid] = node_unsched[my_node] ? -1 : nl;
-  } else {
-    lds_nlab[tid] = -1;
-  }
-  __syncthreads();
-
-  // each wave owns word (node0/64 + wid); loop over all segments
-  const int my_word = node0 / WAVE + wid;
-  const int lds_off = wid * WAVE + lane;  // this lane's node slot in LDS
-
-  for (int g = 0; g < G; ++g) {
-    const int nsel = seg_sel_count[g];
-    const int64_t* sel = seg_sel + (int64_t)g * MAXSEL;  // wave-uniform
-    bool ok = false;
-    const int nlab = lds_nlab[lds_off];
-    if (nlab >= 0) {
-      ok = sel_subset(sel, nsel, lds_labels + (int64_t)lds_off * MAXLAB, nlab);
-    }
-    const uint64_t ballot = __ballot(ok);
-    if (lane == 0 && my_word < W) {
-      elig[(int64_t)g * W + my_word] = ballot;
-    }
-  }
+  dev_fit_tile(blockIdx.x * NODES_PER_BLOCK, node_labels, node_label_count,
+               node_unsched, seg_sel, seg_sel_count, N, G, elig, lds_labels,
+               lds_nlab);
 }
 
 // Segment -> agent-pool routing: for each segment, the FIRST pool
@@ -94,32 +57,8 @@ extern "C" __global__ void seg_pool_kernel(
 {
   const int g = blockIdx.x;
   if (g >= *g_ptr) return;
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int nsel = seg_sel_count[g];
-  const int64_t* sel = seg_sel + (int64_t)g * MAXSEL;
-  int32_t rq[NRES];
-#pragma unroll
-  for (int r = 0; r < NRES; ++r) rq[r] = seg_req[g * NRES + r];
-
-  int best = INT32_MAX;
-  for (int k0 = 0; k0 < K; k0 += WAVE) {
-    const int k = k0 + lane;
-    bool ok = false;
-    if (k < K) {
-      ok = sel_subset(sel, nsel, pool_labels + (int64_t)k * MAXLAB,
-                      pool_label_count[k]);
-      if (ok) {
-#pragma unroll
-        for (int r = 0; r < NRES; ++r) ok &= rq[r] <= pool_unit[k * NRES + r];
-      }
-    }
-    const uint64_t ballot = __ballot(ok);
-    if (ballot) {
-      best = k0 + __ffsll((unsigned long long)ballot) - 1;
-      break;
-    }
-  }
-  if (lane == 0) seg_pool[g] = best == INT32_MAX ? -1 : best;
+  dev_route_one(g, K, seg_sel, seg_sel_count, seg_req, pool_labels,
+                pool_label_count, pool_unit, seg_pool);
 }
 
 #include <torch/extension.h>

@@ -5,23 +5,12 @@
 // reconstructed]. Integer atomics are exact and commutative, so the
 // result is deterministic under any pod order / shard split; sharded
 // ranks sum partial deltas and all-reduce (parallel/engine_sharded.py).
-#include "common.h"
+#include "pipeline_device.h"
 
 extern "C" __global__ void node_free_kernel(
     const int32_t* __restrict__ pod_node, const int32_t* __restrict__ pod_req,
     int64_t P, int32_t* __restrict__ node_free) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < P;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    const int j = pod_node[i];
-    if (j < 0) continue;
-#pragma unroll
-    for (int r = 0; r < NRES; ++r) {
-      const int32_t v = pod_req[i * NRES + r];
-      // zero-valued atomics still serialize on the line; skip them
-      // (tombstone/padding rows and sparse request vectors)
-      if (v != 0) atomicSub(&node_free[(int64_t)j * NRES + r], v);
-    }
-  }
+  dev_node_usage_subtract(pod_node, pod_req, P, node_free);
 }
 
 #include <torch/extension.h>

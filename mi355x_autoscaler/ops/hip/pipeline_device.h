@@ -1,0 +1,624 @@
+// Shared device bodies of the scale-up decision pipeline.
+//
+// Each phase of the decide (hash-aggregate -> compact+sort -> free ->
+// eligibility -> route -> FFD place -> units) lives here as a
+// __device__ function with the block/grid mapping passed in. Two
+// callers instantiate them:
+//
+//   * the standalone kernels (segment_hash.hip, binpack.hip,
+//     fit_predicates.hip, node_free.hip) — one launch per phase,
+//     used by the classic orchestrated pipeline, the op-by-op Python
+//     reference path, and the large-P regime;
+//   * mega_decide.hip — ONE cooperative launch running every phase
+//     with grid.sync() barriers, for the small/medium-P regime where
+//     the ~13 per-launch fixed costs dominate the decision.
+//
+// Sharing the bodies keeps the two paths bit-identical by
+// construction (same arithmetic, same order); tests assert equality
+// against the CPU oracle on hardware.
+#pragma once
+
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// hash-aggregate constants (shared by segment_hash.hip and the mega path)
+#define SEG_MAX 2048
+#define ST_EMPTY 0
+#define ST_CLAIMED 1
+#define ST_READY 2
+#define MAX_SPIN 100000
+#define RLX_AGENT __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT
+
+__device__ inline uint32_t mix_hash(int64_t ks, int64_t kr) {
+  uint64_t x = (uint64_t)ks * 0x9E3779B97F4A7C15ull ^
+               (uint64_t)kr * 0xC2B2AE3D27D4EB4Full;
+  x ^= x >> 29;
+  x *= 0xBF58476D1CE4E5B9ull;
+  x ^= x >> 32;
+  return (uint32_t)x;
+}
+
+// Probe/claim one key; returns the slot or -1 (overflow). Agent-scope
+// release/ready-flag discipline (guide §6 G16); spins bounded.
+__device__ inline int probe_claim(int64_t ks, int64_t kr, int32_t H,
+                                  int64_t* slot_ks, int64_t* slot_kr,
+                                  int32_t* slot_state, int32_t* overflow) {
+  const uint32_t mask = (uint32_t)H - 1u;
+  uint32_t h = mix_hash(ks, kr) & mask;
+  for (int probe = 0; probe < H; ++probe) {
+    int st = __hip_atomic_load(&slot_state[h], RLX_AGENT);
+    if (st == ST_EMPTY) {
+      int expected = ST_EMPTY;
+      if (__hip_atomic_compare_exchange_strong(
+              &slot_state[h], &expected, ST_CLAIMED, __ATOMIC_RELAXED,
+              __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)) {
+        __hip_atomic_store(&slot_ks[h], ks, RLX_AGENT);
+        __hip_atomic_store(&slot_kr[h], kr, RLX_AGENT);
+        __hip_atomic_store(&slot_state[h], ST_READY,
+                           __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+        return (int)h;
+      }
+      st = expected;  // observed state; handle CLAIMED/READY below
+    }
+    if (st == ST_CLAIMED) {
+      // claimer is between CAS and READY: spin briefly on this slot
+      int spins = 0;
+      while ((st = __hip_atomic_load(&slot_state[h], RLX_AGENT)) == ST_CLAIMED) {
+        __builtin_amdgcn_s_sleep(1);
+        if (++spins > MAX_SPIN) {
+          atomicOr(overflow, 2);
+          st = ST_READY;  // bail; comparison below will mismatch or not
+          break;
+        }
+      }
+    }
+    if (st == ST_READY) {
+      if (__hip_atomic_load(&slot_ks[h], RLX_AGENT) == ks &&
+          __hip_atomic_load(&slot_kr[h], RLX_AGENT) == kr) {
+        return (int)h;
+      }
+    }
+    h = (h + 1u) & mask;
+  }
+  return -1;
+}
+
+// Wave-aggregated direct-global hash build (the small-P form): one
+// probe + one count/first atomic pair per distinct key per wave.
+// Grid-strided over pods; safe at any grid shape.
+__device__ inline void dev_hash_build_small(
+    const int32_t* __restrict__ pod_flags, const int64_t* __restrict__ sel_hash,
+    const int32_t* __restrict__ pod_req, int64_t P, int32_t H,
+    int64_t* __restrict__ slot_ks, int64_t* __restrict__ slot_kr,
+    int32_t* __restrict__ slot_state, int32_t* __restrict__ slot_count,
+    int32_t* __restrict__ slot_first, int32_t* __restrict__ slot_of_pod,
+    int32_t* __restrict__ overflow) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; ; i0 += stride) {
+    if (__all(i0 >= P)) break;
+    const int pf = i0 < P ? pod_flags[i0] : 0;
+    const bool live = i0 < P && (pf & PF_PENDING);
+    int64_t ks = 0, kr = 0;
+    if (live) {
+      ks = sel_hash[i0];
+      int32_t rq[NRES];
+#pragma unroll
+      for (int r = 0; r < NRES; ++r) rq[r] = pod_req[i0 * NRES + r];
+      kr = pack_req_key(rq, pf & PF_CRITICAL);
+    } else if (i0 < P) {
+      slot_of_pod[i0] = -1;
+    }
+
+    uint64_t pending_lanes = __ballot(live);
+    int my_slot = -1;
+    while (pending_lanes) {
+      const int leader = __ffsll((unsigned long long)pending_lanes) - 1;
+      const int64_t k0 = __shfl(ks, leader, WAVE);
+      const int64_t r0 = __shfl(kr, leader, WAVE);
+      const bool match = live && ks == k0 && kr == r0;
+      const uint64_t match_mask = __ballot(match);
+      int slot = -1;
+      if (lane == leader) {
+        slot = probe_claim(k0, r0, H, slot_ks, slot_kr, slot_state, overflow);
+        if (slot >= 0) {
+          atomicAdd(&slot_count[slot], __popcll((unsigned long long)match_mask));
+          atomicMax(&slot_first[slot], INT32_MAX - (int32_t)i0);
+        } else {
+          atomicOr(overflow, 1);
+        }
+      }
+      slot = __shfl(slot, leader, WAVE);
+      if (match) my_slot = slot;
+      pending_lanes &= ~match_mask;
+    }
+    if (i0 < P && live) slot_of_pod[i0] = my_slot;
+  }
+}
+
+// Single-block compact + bitonic sort into the oracle's global segment
+// order (sel-hash as uint64 asc, request key desc). LDS arrays are
+// caller-provided: lds_ks/lds_kr [SEG_MAX] u64, lds_slot [SEG_MAX]
+// i32, lds_n one i32. Runs at any block size.
+__device__ inline void dev_compact_sort(
+    const int64_t* __restrict__ slot_ks, const int64_t* __restrict__ slot_kr,
+    const int32_t* __restrict__ slot_state, const int32_t* __restrict__ slot_count,
+    const int32_t* __restrict__ slot_first, int32_t H,
+    const int64_t* __restrict__ pod_sel_pairs, const int32_t* __restrict__ pod_sel_count,
+    const int32_t* __restrict__ pod_req,
+    int64_t* __restrict__ seg_sel_pairs, int32_t* __restrict__ seg_sel_count,
+    int64_t* __restrict__ seg_sel_hash, int32_t* __restrict__ seg_req,
+    int64_t* __restrict__ seg_count, int64_t* __restrict__ seg_key,
+    int32_t* __restrict__ seg_first_pod, int32_t* __restrict__ slot2seg,
+    int32_t* __restrict__ g_out,
+    uint64_t* lds_ks, uint64_t* lds_kr, int32_t* lds_slot, int32_t* lds_n) {
+  const int tid = threadIdx.x;
+  const int B = blockDim.x;
+  if (tid == 0) *lds_n = 0;
+  __syncthreads();
+
+  // compact occupied slots into LDS (order arbitrary; sort fixes it)
+  for (int h = tid; h < H; h += B) {
+    if (slot_state[h] == ST_READY && slot_count[h] > 0) {
+      const int g = atomicAdd(lds_n, 1);
+      if (g < SEG_MAX) {
+        // sel asc as uint64: flip the sign bit so int compare == uint
+        lds_ks[g] = (uint64_t)slot_ks[h] ^ 0x8000000000000000ull;
+        // req desc: invert the pack key
+        lds_kr[g] = ~(uint64_t)slot_kr[h];
+        lds_slot[g] = h;
+      }
+    }
+  }
+  __syncthreads();
+  int G = *lds_n;
+  if (G > SEG_MAX) {
+    if (tid == 0) {
+      g_out[0] = 0;
+      g_out[1] = 1;
+    }
+    return;
+  }
+  // pad to pow2 with +inf keys
+  int NP = 1;
+  while (NP < G) NP <<= 1;
+  if (NP < 2) NP = 2;
+  for (int g = G + tid; g < NP; g += B) {
+    lds_ks[g] = ~0ull;
+    lds_kr[g] = ~0ull;
+    lds_slot[g] = -1;
+  }
+  __syncthreads();
+
+  // bitonic sort ascending by (ks, kr)
+  for (int k = 2; k <= NP; k <<= 1) {
+    for (int j = k >> 1; j > 0; j >>= 1) {
+      for (int t = tid; t < NP; t += B) {
+        const int ixj = t ^ j;
+        if (ixj > t) {
+          const bool up = ((t & k) == 0);
+          const bool gt = (lds_ks[t] > lds_ks[ixj]) ||
+                          (lds_ks[t] == lds_ks[ixj] && lds_kr[t] > lds_kr[ixj]);
+          if (gt == up) {
+            uint64_t a = lds_ks[t]; lds_ks[t] = lds_ks[ixj]; lds_ks[ixj] = a;
+            uint64_t b = lds_kr[t]; lds_kr[t] = lds_kr[ixj]; lds_kr[ixj] = b;
+            int32_t c = lds_slot[t]; lds_slot[t] = lds_slot[ixj]; lds_slot[ixj] = c;
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  // emit segment SoA in sorted order + slot->segment LUT
+  for (int g = tid; g < G; g += B) {
+    const int h = lds_slot[g];
+    slot2seg[h] = g;
+    // slot_first stores INT32_MAX - min(pod id) under atomicMax so the
+    // workspace zeroes with one memset; decode here.
+    const int fp = INT32_MAX - slot_first[h];
+    seg_sel_hash[g] = (int64_t)(lds_ks[g] ^ 0x8000000000000000ull);
+    seg_count[g] = (int64_t)slot_count[h];
+    seg_key[g] = (int64_t)~lds_kr[g];  // un-invert: the raw pack key
+    seg_first_pod[g] = fp;
+    seg_sel_count[g] = pod_sel_count[fp];
+#pragma unroll
+    for (int q = 0; q < MAXSEL; ++q)
+      seg_sel_pairs[(int64_t)g * MAXSEL + q] = pod_sel_pairs[(int64_t)fp * MAXSEL + q];
+#pragma unroll
+    for (int r = 0; r < NRES; ++r)
+      seg_req[g * NRES + r] = pod_req[(int64_t)fp * NRES + r];
+  }
+  if (tid == 0) {
+    g_out[0] = G;
+    g_out[1] = 0;
+  }
+}
+
+// free-capacity usage subtraction, grid-strided over pods.
+__device__ inline void dev_node_usage_subtract(
+    const int32_t* __restrict__ pod_node, const int32_t* __restrict__ pod_req,
+    int64_t P, int32_t* __restrict__ node_free) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < P;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int j = pod_node[i];
+    if (j < 0) continue;
+#pragma unroll
+    for (int r = 0; r < NRES; ++r) {
+      const int32_t v = pod_req[i * NRES + r];
+      // zero-valued atomics still serialize on the line; skip them
+      // (tombstone/padding rows and sparse request vectors)
+      if (v != 0) atomicSub(&node_free[(int64_t)j * NRES + r], v);
+    }
+  }
+}
+
+// One 256-node eligibility tile: stage the tile's label sets into LDS,
+// then emit 64-bit ballots for every segment. lds_labels holds
+// FIT_TILE*MAXLAB i64, lds_nlab FIT_TILE i32. Block size must be
+// FIT_TILE (== 256).
+#define FIT_TILE 256
+
+__device__ inline void dev_fit_tile(
+    int node0,
+    const int64_t* __restrict__ node_labels, const int32_t* __restrict__ node_label_count,
+    const int32_t* __restrict__ node_unsched,
+    const int64_t* __restrict__ seg_sel, const int32_t* __restrict__ seg_sel_count,
+    int32_t N, int G, uint64_t* __restrict__ elig,
+    int64_t* lds_labels, int32_t* lds_nlab) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid / WAVE;
+  const int my_node = node0 + tid;
+  const int W = (N + WAVE - 1) / WAVE;
+
+  // stage this tile's node labels into LDS: thread t copies node t's row
+  if (my_node < N) {
+    const int64_t* src = node_labels + (int64_t)my_node * MAXLAB;
+    int64_t* dst = lds_labels + (int64_t)tid * MAXLAB;
+#pragma unroll
+    for (int q = 0; q < MAXLAB; ++q) dst[q] = src[q];
+    int nl = node_label_count[my_node];
+    // fold schedulability into the count: unschedulable => impossible
+    lds_nlab[tid] = node_unsched[my_node] ? -1 : nl;
+  } else {
+    lds_nlab[tid] = -1;
+  }
+  __syncthreads();
+
+  // each wave owns word (node0/64 + wid); loop over all segments
+  const int my_word = node0 / WAVE + wid;
+  const int lds_off = wid * WAVE + lane;  // this lane's node slot in LDS
+
+  for (int g = 0; g < G; ++g) {
+    const int nsel = seg_sel_count[g];
+    const int64_t* sel = seg_sel + (int64_t)g * MAXSEL;  // wave-uniform
+    bool ok = false;
+    const int nlab = lds_nlab[lds_off];
+    if (nlab >= 0) {
+      ok = sel_subset(sel, nsel, lds_labels + (int64_t)lds_off * MAXLAB, nlab);
+    }
+    const uint64_t ballot = __ballot(ok);
+    if (lane == 0 && my_word < W) {
+      elig[(int64_t)g * W + my_word] = ballot;
+    }
+  }
+  __syncthreads();  // tile reuse: LDS free for the next tile
+}
+
+// Segment -> pool routing for ONE segment; a full wave cooperates
+// (lanes fan out over pools). Matches the oracle's first-fit pool
+// choice (ascending pool index == sorted pool name).
+__device__ inline void dev_route_one(
+    int g, int32_t K,
+    const int64_t* __restrict__ seg_sel, const int32_t* __restrict__ seg_sel_count,
+    const int32_t* __restrict__ seg_req,
+    const int64_t* __restrict__ pool_labels, const int32_t* __restrict__ pool_label_count,
+    const int32_t* __restrict__ pool_unit, int32_t* __restrict__ seg_pool) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int nsel = seg_sel_count[g];
+  const int64_t* sel = seg_sel + (int64_t)g * MAXSEL;
+  int32_t rq[NRES];
+#pragma unroll
+  for (int r = 0; r < NRES; ++r) rq[r] = seg_req[g * NRES + r];
+
+  int best = INT32_MAX;
+  for (int k0 = 0; k0 < K; k0 += WAVE) {
+    const int k = k0 + lane;
+    bool ok = false;
+    if (k < K) {
+      ok = sel_subset(sel, nsel, pool_labels + (int64_t)k * MAXLAB,
+                      pool_label_count[k]);
+      if (ok) {
+#pragma unroll
+        for (int r = 0; r < NRES; ++r) ok &= rq[r] <= pool_unit[k * NRES + r];
+      }
+    }
+    const uint64_t ballot = __ballot(ok);
+    if (ballot) {
+      best = k0 + __ffsll((unsigned long long)ballot) - 1;
+      break;
+    }
+  }
+  if (lane == 0) seg_pool[g] = best == INT32_MAX ? -1 : best;
+}
+
+// pod -> routed-pool map (for the MFMA demand segsum), grid-strided.
+__device__ inline void dev_pod_pool_map(
+    const int32_t* __restrict__ slot_of_pod, const int32_t* __restrict__ slot2seg,
+    const int32_t* __restrict__ seg_pool, int64_t P,
+    int32_t* __restrict__ pod_pool) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < P;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int s = slot_of_pod[i];
+    pod_pool[i] = (s >= 0) ? seg_pool[slot2seg[s]] : -1;
+  }
+}
+
+// Single-block FFD placement over existing nodes (kernel B core).
+// lds_scan must hold BLOCK/WAVE + 1 int64 slots.
+template <int BLOCK>
+__device__ inline void dev_binpack(
+    int32_t* __restrict__ free_, const uint64_t* __restrict__ elig,
+    const int32_t* __restrict__ seg_req, const int64_t* __restrict__ seg_count,
+    int32_t N, int G, int64_t* __restrict__ placed_out, int64_t* lds_scan) {
+  const int tid = threadIdx.x;
+  const int W = (N + WAVE - 1) / WAVE;
+
+  for (int g = 0; g < G; ++g) {
+    const int64_t need = seg_count[g];
+    if (need <= 0) {
+      if (tid == 0) placed_out[g] = 0;
+      continue;
+    }
+    int32_t rq[NRES];
+#pragma unroll
+    for (int r = 0; r < NRES; ++r) rq[r] = seg_req[g * NRES + r];
+    const uint64_t* erow = elig + (int64_t)g * W;
+
+    int64_t carry = 0;  // pods placeable on nodes [0, base)
+    for (int base = 0; base < N && carry < need; base += BLOCK) {
+      const int j = base + tid;
+      int64_t m = 0;
+      int32_t f[NRES];
+      if (j < N && ((erow[j / WAVE] >> (j & (WAVE - 1))) & 1ull)) {
+#pragma unroll
+        for (int r = 0; r < NRES; ++r) f[r] = free_[(int64_t)j * NRES + r];
+        m = per_node_capacity(f, rq);
+      }
+      int64_t chunk_total;
+      const int64_t scan = block_exclusive_scan_i64<BLOCK>(m, lds_scan, &chunk_total);
+      const int64_t offset = carry + scan;
+      int64_t take = need - offset;
+      take = take < 0 ? 0 : (take > m ? m : take);
+      if (take > 0) {
+#pragma unroll
+        for (int r = 0; r < NRES; ++r) {
+          free_[(int64_t)j * NRES + r] = f[r] - (int32_t)(take * rq[r]);
+        }
+      }
+      carry += chunk_total;
+    }
+    if (tid == 0) placed_out[g] = carry < need ? carry : need;
+    __syncthreads();
+  }
+}
+
+// Units phase (kernel B'): run-length FFD into virtual new nodes,
+// pools wave-parallel, inline max-new, optional fused readback
+// epilogue. Block size must be UNITS_BLOCK (== 256). `lds_arena`
+// (lds_cap runs * 6 i64) may be null when lds_cap == 0 — then the
+// global scratch is used unconditionally.
+#define UNITS_MAX_POOLS 64
+#define UNITS_BLOCK 256
+#define UNITS_NW (UNITS_BLOCK / WAVE)
+#define UNITS_PAD 8  // arena slack per wave
+
+__device__ inline void dev_units(
+    const int32_t* __restrict__ seg_pool, const int32_t* __restrict__ seg_req,
+    const int64_t* __restrict__ seg_count, const int64_t* __restrict__ placed,
+    const int32_t* __restrict__ pool_unit,
+    const int32_t* __restrict__ pool_max, const int32_t* __restrict__ pool_cur,
+    int G, int32_t K, int32_t RTOT, int32_t lds_cap,
+    int64_t* __restrict__ scratch, int64_t* lds_arena,
+    int64_t* __restrict__ units_cnt,
+    int64_t* __restrict__ placed_new, int64_t* __restrict__ unful,
+    int64_t* __restrict__ pool_residual,
+    const int64_t* __restrict__ seg_sel_hash, const int32_t* __restrict__ stat,
+    int64_t* __restrict__ blob_mirror, int32_t* __restrict__ blob_stat,
+    int32_t SM) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid / WAVE;
+
+  __shared__ int32_t head[UNITS_MAX_POOLS];
+  __shared__ int32_t tailp[UNITS_MAX_POOLS];
+  __shared__ int64_t opened[UNITS_MAX_POOLS];
+  __shared__ int64_t resid[UNITS_MAX_POOLS * NRES];
+  __shared__ int32_t wave_off[UNITS_NW + 1];
+
+  for (int k = tid; k < K; k += UNITS_BLOCK) {
+    head[k] = -1;
+    tailp[k] = -1;
+    opened[k] = 0;
+  }
+  for (int q = tid; q < K * NRES; q += UNITS_BLOCK) resid[q] = 0;
+
+  // per-wave arena need: 4 runs per owned segment + fixed slack
+  int cnt = 0;
+  for (int g = lane; g < G; g += WAVE) {
+    const int k = seg_pool[g];
+    if (k >= 0 && (k % UNITS_NW) == w) ++cnt;
+  }
+#pragma unroll
+  for (int off = WAVE / 2; off; off >>= 1) cnt += __shfl_down(cnt, off, WAVE);
+  if (lane == 0) wave_off[w] = 4 * cnt + UNITS_PAD;
+  __syncthreads();
+  if (tid == 0) {
+    int off = 0;
+    for (int i = 0; i < UNITS_NW; ++i) {
+      const int n = wave_off[i];
+      wave_off[i] = off;
+      off += n;
+    }
+    wave_off[UNITS_NW] = off;
+  }
+  __syncthreads();
+  const int total_need = wave_off[UNITS_NW];
+  const bool use_lds = lds_arena != nullptr && total_need <= lds_cap;
+  const int cap = use_lds ? lds_cap : RTOT;
+  int64_t* arena = use_lds ? lds_arena : scratch;
+  int64_t* run_res = arena;               // [cap][NRES]
+  int64_t* run_cnt = arena + (int64_t)cap * NRES;
+  int64_t* run_next = run_cnt + cap;
+  const int a_hi = wave_off[w + 1];
+
+  if (lane == 0) {
+    int alloc = wave_off[w];
+    for (int g = 0; g < G; ++g) {
+      const int k = seg_pool[g];
+      const bool mine = (k >= 0) ? ((k % UNITS_NW) == w) : (w == 0);
+      if (!mine) continue;
+      int64_t rem = seg_count[g] - placed[g];
+      placed_new[g] = 0;
+      unful[g] = 0;
+      if (rem <= 0) continue;
+      if (k < 0) {
+        unful[g] = rem;
+        continue;
+      }
+      if (alloc + 4 > a_hi) {  // arena bound: a segment allocates <=4 runs
+        unful[g] = rem;        // (unreachable when the scratch honors the
+        continue;              //  4G+pad sizing; kept so a bound bug can
+      }                        //  never corrupt the lists)
+      int64_t rq[NRES];
+#pragma unroll
+      for (int r = 0; r < NRES; ++r) rq[r] = seg_req[g * NRES + r];
+#pragma unroll
+      for (int r = 0; r < NRES; ++r) resid[k * NRES + r] += rq[r] * rem;
+      const int64_t need0 = rem;
+
+      // 1) walk open runs in creation order
+      for (int idx = head[k]; idx != -1 && rem > 0; idx = (int)run_next[idx]) {
+        int64_t m = (int64_t)1 << 31;
+        bool neg = false;
+#pragma unroll
+        for (int r = 0; r < NRES; ++r) {
+          const int64_t fr = run_res[(int64_t)idx * NRES + r];
+          if (fr < 0) neg = true;
+          if (rq[r] > 0) {
+            const int64_t q = fr / rq[r];
+            m = q < m ? q : m;
+          }
+        }
+        if (neg || m <= 0) continue;
+        const int64_t c = run_cnt[idx];
+        if (rem >= m * c) {  // every unit of the run takes m
+#pragma unroll
+          for (int r = 0; r < NRES; ++r) run_res[(int64_t)idx * NRES + r] -= m * rq[r];
+          rem -= m * c;
+          continue;
+        }
+        // split: nf full takers, maybe one partial, rest untouched
+        const int64_t nf = rem / m;
+        const int64_t part = rem - nf * m;
+        int64_t untouched = c - nf - (part > 0 ? 1 : 0);
+        // node idx becomes the full-takers run (or partial if nf==0)
+        const int64_t nxt = run_next[idx];
+        int cur = idx;
+        if (nf > 0) {
+          run_cnt[cur] = nf;
+#pragma unroll
+          for (int r = 0; r < NRES; ++r) run_res[(int64_t)cur * NRES + r] -= m * rq[r];
+        }
+        if (part > 0) {
+          if (nf > 0) {
+            const int nn = alloc++;
+            run_next[cur] = nn;
+            run_next[nn] = nxt;
+#pragma unroll
+            for (int r = 0; r < NRES; ++r)
+              run_res[(int64_t)nn * NRES + r] =
+                  run_res[(int64_t)idx * NRES + r] + m * rq[r] - part * rq[r];
+            run_cnt[nn] = 1;
+            if (tailp[k] == cur) tailp[k] = nn;
+            cur = nn;
+          } else {
+            run_cnt[cur] = 1;
+#pragma unroll
+            for (int r = 0; r < NRES; ++r) run_res[(int64_t)cur * NRES + r] -= part * rq[r];
+          }
+        }
+        if (untouched > 0) {
+          const int nn = alloc++;
+          run_next[cur] = nn;
+          run_next[nn] = nxt;
+#pragma unroll
+          for (int r = 0; r < NRES; ++r)
+            run_res[(int64_t)nn * NRES + r] =
+                run_res[(int64_t)idx * NRES + r] +
+                ((cur == idx && part > 0) ? part : m) * rq[r];
+          run_cnt[nn] = untouched;
+          if (tailp[k] == cur) tailp[k] = nn;
+        }
+        rem = 0;
+        break;
+      }
+
+      // 2) open new units while the pool cap allows
+      int64_t per_unit = (int64_t)1 << 31;
+      bool fits = true;
+#pragma unroll
+      for (int r = 0; r < NRES; ++r) {
+        const int64_t ucap = pool_unit[k * NRES + r];  // not the arena cap
+        if (rq[r] > ucap) fits = false;
+        if (rq[r] > 0) {
+          const int64_t q = ucap / rq[r];
+          per_unit = q < per_unit ? q : per_unit;
+        }
+      }
+      if (rem > 0 && fits && per_unit > 0) {
+        int32_t mn = pool_max[k] - pool_cur[k];  // inline max-new
+        if (mn < 0) mn = 0;
+        int64_t room = (int64_t)mn - opened[k];
+        if (room < 0) room = 0;
+        int64_t nf = rem / per_unit;
+        nf = nf < room ? nf : room;
+        if (nf > 0) {
+          const int nn = alloc++;
+#pragma unroll
+          for (int r = 0; r < NRES; ++r)
+            run_res[(int64_t)nn * NRES + r] = (int64_t)pool_unit[k * NRES + r] - per_unit * rq[r];
+          run_cnt[nn] = nf;
+          run_next[nn] = -1;
+          if (tailp[k] == -1) head[k] = nn; else run_next[tailp[k]] = nn;
+          tailp[k] = nn;
+          opened[k] += nf;
+          rem -= nf * per_unit;
+          room -= nf;
+        }
+        if (rem > 0 && room > 0) {  // one partial unit
+          const int nn = alloc++;
+#pragma unroll
+          for (int r = 0; r < NRES; ++r)
+            run_res[(int64_t)nn * NRES + r] = (int64_t)pool_unit[k * NRES + r] - rem * rq[r];
+          run_cnt[nn] = 1;
+          run_next[nn] = -1;
+          if (tailp[k] == -1) head[k] = nn; else run_next[tailp[k]] = nn;
+          tailp[k] = nn;
+          opened[k] += 1;
+          rem = 0;
+        }
+      }
+      placed_new[g] = need0 - rem;
+      unful[g] = rem;
+    }
+  }
+  __syncthreads();
+  for (int k = tid; k < K; k += UNITS_BLOCK) units_cnt[k] = opened[k];
+  for (int q = tid; q < K * NRES; q += UNITS_BLOCK) pool_residual[q] = resid[q];
+  if (blob_mirror != nullptr) {
+    for (int i = tid; i < SM; i += UNITS_BLOCK) blob_mirror[i] = seg_sel_hash[i];
+    if (tid < 3) blob_stat[tid] = stat[tid];
+  }
+}

@@ -24,69 +24,7 @@
 //
 // Capacity: H slots (pow2), <= SEG_MAX segments. Real bursts have
 // O(10..100) distinct (selector, shape) pairs; 2048 is generous.
-#include "common.h"
-
-#define SEG_MAX 2048
-#define ST_EMPTY 0
-#define ST_CLAIMED 1
-#define ST_READY 2
-#define FIRST_INIT 0x7f7f7f7f
-#define MAX_SPIN 100000
-
-#define RLX_AGENT __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT
-
-__device__ inline uint32_t mix_hash(int64_t ks, int64_t kr) {
-  uint64_t x = (uint64_t)ks * 0x9E3779B97F4A7C15ull ^
-               (uint64_t)kr * 0xC2B2AE3D27D4EB4Full;
-  x ^= x >> 29;
-  x *= 0xBF58476D1CE4E5B9ull;
-  x ^= x >> 32;
-  return (uint32_t)x;
-}
-
-// Probe/claim one key; returns the slot or -1 (overflow).
-__device__ inline int probe_claim(int64_t ks, int64_t kr, int32_t H,
-                                  int64_t* slot_ks, int64_t* slot_kr,
-                                  int32_t* slot_state, int32_t* overflow) {
-  const uint32_t mask = (uint32_t)H - 1u;
-  uint32_t h = mix_hash(ks, kr) & mask;
-  for (int probe = 0; probe < H; ++probe) {
-    int st = __hip_atomic_load(&slot_state[h], RLX_AGENT);
-    if (st == ST_EMPTY) {
-      int expected = ST_EMPTY;
-      if (__hip_atomic_compare_exchange_strong(
-              &slot_state[h], &expected, ST_CLAIMED, __ATOMIC_RELAXED,
-              __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)) {
-        __hip_atomic_store(&slot_ks[h], ks, RLX_AGENT);
-        __hip_atomic_store(&slot_kr[h], kr, RLX_AGENT);
-        __hip_atomic_store(&slot_state[h], ST_READY,
-                           __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
-        return (int)h;
-      }
-      st = expected;  // observed state; handle CLAIMED/READY below
-    }
-    if (st == ST_CLAIMED) {
-      // claimer is between CAS and READY: spin briefly on this slot
-      int spins = 0;
-      while ((st = __hip_atomic_load(&slot_state[h], RLX_AGENT)) == ST_CLAIMED) {
-        __builtin_amdgcn_s_sleep(1);
-        if (++spins > MAX_SPIN) {
-          atomicOr(overflow, 2);
-          st = ST_READY;  // bail; comparison below will mismatch or not
-          break;
-        }
-      }
-    }
-    if (st == ST_READY) {
-      if (__hip_atomic_load(&slot_ks[h], RLX_AGENT) == ks &&
-          __hip_atomic_load(&slot_kr[h], RLX_AGENT) == kr) {
-        return (int)h;
-      }
-    }
-    h = (h + 1u) & mask;
-  }
-  return -1;
-}
+#include "pipeline_device.h"
 
 // Two-level build: per-block LDS sub-table -> one global flush per
 // distinct key per block. A burst has few distinct (selector, request)
@@ -262,49 +200,9 @@ extern "C" __global__ void segment_hash_build_small(
     int32_t* __restrict__ slot_state, int32_t* __restrict__ slot_count,
     int32_t* __restrict__ slot_first, int32_t* __restrict__ slot_of_pod,
     int32_t* __restrict__ overflow) {
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; ; i0 += stride) {
-    if (__all(i0 >= P)) break;
-    const int pf = i0 < P ? pod_flags[i0] : 0;
-    const bool live = i0 < P && (pf & PF_PENDING);
-    int64_t ks = 0, kr = 0;
-    if (live) {
-      ks = sel_hash[i0];
-      int32_t rq[NRES];
-#pragma unroll
-      for (int r = 0; r < NRES; ++r) rq[r] = pod_req[i0 * NRES + r];
-      kr = pack_req_key(rq, pf & PF_CRITICAL);
-    } else if (i0 < P) {
-      slot_of_pod[i0] = -1;
-    }
-
-    // leader election over distinct keys among live lanes: one global
-    // probe + one count/first atomic pair per distinct key per wave
-    uint64_t pending_lanes = __ballot(live);
-    int my_slot = -1;
-    while (pending_lanes) {
-      const int leader = __ffsll((unsigned long long)pending_lanes) - 1;
-      const int64_t k0 = __shfl(ks, leader, WAVE);
-      const int64_t r0 = __shfl(kr, leader, WAVE);
-      const bool match = live && ks == k0 && kr == r0;
-      const uint64_t match_mask = __ballot(match);
-      int slot = -1;
-      if (lane == leader) {
-        slot = probe_claim(k0, r0, H, slot_ks, slot_kr, slot_state, overflow);
-        if (slot >= 0) {
-          atomicAdd(&slot_count[slot], __popcll((unsigned long long)match_mask));
-          atomicMax(&slot_first[slot], INT32_MAX - (int32_t)i0);
-        } else {
-          atomicOr(overflow, 1);
-        }
-      }
-      slot = __shfl(slot, leader, WAVE);
-      if (match) my_slot = slot;
-      pending_lanes &= ~match_mask;
-    }
-    if (i0 < P && live) slot_of_pod[i0] = my_slot;
-  }
+  dev_hash_build_small(pod_flags, sel_hash, pod_req, P, H, slot_ks, slot_kr,
+                       slot_state, slot_count, slot_first, slot_of_pod,
+                       overflow);
 }
 
 // ---------------------------------------------------------------------------
@@ -331,87 +229,11 @@ extern "C" __global__ __launch_bounds__(CS_BLOCK) void segment_compact_sort(
   __shared__ uint64_t lds_kr[SEG_MAX];
   __shared__ int32_t lds_slot[SEG_MAX];
   __shared__ int32_t lds_n;
-
-  const int tid = threadIdx.x;
-  if (tid == 0) lds_n = 0;
-  __syncthreads();
-
-  // compact occupied slots into LDS (order arbitrary; sort fixes it)
-  for (int h = tid; h < H; h += CS_BLOCK) {
-    if (slot_state[h] == ST_READY && slot_count[h] > 0) {
-      const int g = atomicAdd(&lds_n, 1);
-      if (g < SEG_MAX) {
-        // sel asc as uint64: flip the sign bit so int compare == uint
-        lds_ks[g] = (uint64_t)slot_ks[h] ^ 0x8000000000000000ull;
-        // req desc: invert the pack key
-        lds_kr[g] = ~(uint64_t)slot_kr[h];
-        lds_slot[g] = h;
-      }
-    }
-  }
-  __syncthreads();
-  int G = lds_n;
-  if (G > SEG_MAX) {
-    if (tid == 0) {
-      g_out[0] = 0;
-      g_out[1] = 1;
-    }
-    return;
-  }
-  // pad to pow2 with +inf keys
-  int NP = 1;
-  while (NP < G) NP <<= 1;
-  if (NP < 2) NP = 2;
-  for (int g = G + tid; g < NP; g += CS_BLOCK) {
-    lds_ks[g] = ~0ull;
-    lds_kr[g] = ~0ull;
-    lds_slot[g] = -1;
-  }
-  __syncthreads();
-
-  // bitonic sort ascending by (ks, kr)
-  for (int k = 2; k <= NP; k <<= 1) {
-    for (int j = k >> 1; j > 0; j >>= 1) {
-      for (int t = tid; t < NP; t += CS_BLOCK) {
-        const int ixj = t ^ j;
-        if (ixj > t) {
-          const bool up = ((t & k) == 0);
-          const bool gt = (lds_ks[t] > lds_ks[ixj]) ||
-                          (lds_ks[t] == lds_ks[ixj] && lds_kr[t] > lds_kr[ixj]);
-          if (gt == up) {
-            uint64_t a = lds_ks[t]; lds_ks[t] = lds_ks[ixj]; lds_ks[ixj] = a;
-            uint64_t b = lds_kr[t]; lds_kr[t] = lds_kr[ixj]; lds_kr[ixj] = b;
-            int32_t c = lds_slot[t]; lds_slot[t] = lds_slot[ixj]; lds_slot[ixj] = c;
-          }
-        }
-      }
-      __syncthreads();
-    }
-  }
-
-  // emit segment SoA in sorted order + slot->segment LUT
-  for (int g = tid; g < G; g += CS_BLOCK) {
-    const int h = lds_slot[g];
-    slot2seg[h] = g;
-    // slot_first stores INT32_MAX - min(pod id) under atomicMax so the
-    // workspace zeroes with one memset; decode here.
-    const int fp = INT32_MAX - slot_first[h];
-    seg_sel_hash[g] = (int64_t)(lds_ks[g] ^ 0x8000000000000000ull);
-    seg_count[g] = (int64_t)slot_count[h];
-    seg_key[g] = (int64_t)~lds_kr[g];  // un-invert: the raw pack key
-    seg_first_pod[g] = fp;
-    seg_sel_count[g] = pod_sel_count[fp];
-#pragma unroll
-    for (int q = 0; q < MAXSEL; ++q)
-      seg_sel_pairs[(int64_t)g * MAXSEL + q] = pod_sel_pairs[(int64_t)fp * MAXSEL + q];
-#pragma unroll
-    for (int r = 0; r < NRES; ++r)
-      seg_req[g * NRES + r] = pod_req[(int64_t)fp * NRES + r];
-  }
-  if (tid == 0) {
-    g_out[0] = G;
-    g_out[1] = 0;
-  }
+  dev_compact_sort(slot_ks, slot_kr, slot_state, slot_count, slot_first, H,
+                   pod_sel_pairs, pod_sel_count, pod_req, seg_sel_pairs,
+                   seg_sel_count, seg_sel_hash, seg_req, seg_count, seg_key,
+                   seg_first_pod, slot2seg, g_out, lds_ks, lds_kr, lds_slot,
+                   &lds_n);
 }
 
 // pod -> routed-pool map (for the MFMA demand segsum): pool id of the
@@ -420,11 +242,7 @@ extern "C" __global__ void pod_pool_map_kernel(
     const int32_t* __restrict__ slot_of_pod, const int32_t* __restrict__ slot2seg,
     const int32_t* __restrict__ seg_pool, int64_t P,
     int32_t* __restrict__ pod_pool) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < P;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    const int s = slot_of_pod[i];
-    pod_pool[i] = (s >= 0) ? seg_pool[slot2seg[s]] : -1;
-  }
+  dev_pod_pool_map(slot_of_pod, slot2seg, seg_pool, P, pod_pool);
 }
 
 #include <torch/extension.h>
