@@ -50,6 +50,21 @@ void binpack_units(torch::Tensor seg_pool, torch::Tensor seg_req,
                    c10::optional<torch::Tensor> blob);
 void pod_pool_map(torch::Tensor slot_of_pod, torch::Tensor slot2seg,
                   torch::Tensor seg_pool, torch::Tensor pod_pool);
+bool mega_decide_launch(
+    torch::Tensor pod_req, torch::Tensor pod_flags, torch::Tensor pod_sel_hash,
+    torch::Tensor pod_sel_pairs, torch::Tensor pod_sel_count,
+    torch::Tensor pod_node, torch::Tensor node_alloc, torch::Tensor node_labels,
+    torch::Tensor node_label_count, torch::Tensor node_unsched,
+    torch::Tensor pool_unit, torch::Tensor pool_labels,
+    torch::Tensor pool_label_count, torch::Tensor pool_max,
+    torch::Tensor pool_cur, torch::Tensor ws_i32, torch::Tensor slot_ks,
+    torch::Tensor slot_kr, torch::Tensor slot2seg, torch::Tensor seg_sel_pairs,
+    torch::Tensor seg_sel_count, torch::Tensor seg_sel_hash,
+    torch::Tensor seg_req, torch::Tensor seg_count, torch::Tensor seg_key,
+    torch::Tensor seg_first, torch::Tensor slot_of_pod, torch::Tensor free,
+    torch::Tensor elig, torch::Tensor seg_pool, torch::Tensor run_scratch,
+    torch::Tensor blob, torch::Tensor pod_pool, int64_t hash_slots,
+    int64_t seg_max, int64_t want_pod_pool);
 
 void fused_decide_launch(
     // snapshot (device views)
@@ -76,6 +91,32 @@ void fused_decide_launch(
   const int64_t SM = seg_max;
   const int64_t K = pool_unit.size(0);
   const int64_t N = node_alloc.size(0);
+  const int64_t P = pod_flags.size(0);
+
+  // small/medium-P regime: ONE cooperative launch runs every phase
+  // (mega_decide.hip); falls through to the classic multi-launch
+  // pipeline on large P, degenerate shapes, or when cooperative
+  // launch is unavailable. MI355X_AS_NO_MEGA=1 forces classic.
+  static int mega_env = -1;
+  static int64_t mega_max_p = 0;
+  if (mega_env < 0) {
+    mega_env = getenv("MI355X_AS_NO_MEGA") ? 0 : 1;
+    const char* mp = getenv("MI355X_AS_MEGA_MAX_P");
+    mega_max_p = mp ? atoll(mp) : 131072;
+  }
+  if (mega_env && N > 0 && K > 0 && P > 0 && P <= mega_max_p && N <= 65536) {
+    if (mega_decide_launch(pod_req, pod_flags, pod_sel_hash, pod_sel_pairs,
+                           pod_sel_count, pod_node, node_alloc, node_labels,
+                           node_label_count, node_unsched, pool_unit,
+                           pool_labels, pool_label_count, pool_max, pool_cur,
+                           ws_i32, slot_ks, slot_kr, slot2seg, seg_sel_pairs,
+                           seg_sel_count, seg_sel_hash, seg_req, seg_count,
+                           seg_key, seg_first, slot_of_pod, free, elig,
+                           seg_pool, run_scratch, blob, pod_pool, hash_slots,
+                           seg_max, want_pod_pool)) {
+      return;
+    }
+  }
 
   // workspace reinit; the blob needs zeroing only when a producer
   // kernel is skipped (binpack writes placed[:G], units writes the

@@ -29,6 +29,7 @@ ext = CUDAExtension(
         "mi355x_autoscaler/ops/hip/drain_score.hip",
         "mi355x_autoscaler/ops/hip/segsum_mfma.hip",
         "mi355x_autoscaler/ops/hip/segment_hash.hip",
+        "mi355x_autoscaler/ops/hip/mega_decide.hip",
         "mi355x_autoscaler/ops/hip/snapshot_diff.hip",
     ],
     extra_compile_args={
