@@ -270,13 +270,24 @@ class Cluster:
         from ..snapshot.delta import DeltaSnapshotCache
 
         node_names = tuple(n.name for n in nodes)
+        pdb_fp = tuple((p.name, frozenset(p.match_labels.items())) for p in pdbs)
         ws = self._watch_state
         if ws is not None and ws["broken"]:
             logger.warning("watch stream dropped; re-priming from a full relist")
             ws = self._watch_state = None
-        if ws is not None and ws["node_names"] != node_names:
-            logger.info("node set changed; re-priming the watch cache")
+        if ws is not None and ws.get("pdb_fp") != pdb_fp:
+            # PDB selectors feed the cached pod_pdb column; a changed
+            # set invalidates it (rare — full re-prime)
+            logger.info("PDB set changed; re-priming the watch cache")
             ws = self._watch_state = None
+        if ws is not None and ws["node_names"] != node_names:
+            # node set changed (scale-up materialized / node deleted):
+            # remap the cached pod->node indices in place — NO pod
+            # relist. New nodes list at the end in the common case, so
+            # the remap usually touches zero rows.
+            n = ws["cache"].remap_nodes({nd.name: i for i, nd in enumerate(nodes)})
+            logger.info("node set changed; remapped %d pod bindings in place", n)
+            ws["node_names"] = node_names
 
         if ws is None:
             if use_codec:
@@ -297,7 +308,7 @@ class Cluster:
             th.start()
             self._watch_state = {
                 "cache": cache, "q": q, "thread": th,
-                "node_names": node_names, "broken": False,
+                "node_names": node_names, "pdb_fp": pdb_fp, "broken": False,
                 "nodes": nodes, "pools": pools, "pdbs": pdbs,
             }
             return snap

@@ -42,6 +42,7 @@ class DeltaSnapshotCache:
         self.uid_row: Dict[str, int] = {}
         self.free_rows: List[int] = []
         self.dirty: Set[int] = set()
+        self.dirty_all = False  # cheaper than a size-wide dirty set
         self.arrays: Dict[str, np.ndarray] = {}
         self.pod_uids: List[str] = []
         self.pod_refs: List[Tuple[str, str]] = []
@@ -68,6 +69,7 @@ class DeltaSnapshotCache:
         self.uid_row = {u: i for i, u in enumerate(snap.pod_uids) if u}
         self.free_rows = []
         self.dirty = set()
+        self.dirty_all = False
         self.node_index = dict(node_index)
         self.pdb_selectors = list(pdb_selectors or [])
 
@@ -217,9 +219,48 @@ class DeltaSnapshotCache:
 
     def drain_dirty(self) -> np.ndarray:
         """Rows touched since the last call (the device scatter set)."""
+        if self.dirty_all:
+            self.dirty_all = False
+            self.dirty = set()
+            return np.arange(self.size, dtype=np.int64)
         rows = np.array(sorted(self.dirty), np.int64)
         self.dirty = set()
         return rows
+
+    # ------------------------------------------------------------------
+    def remap_nodes(self, new_node_index: Dict[str, int]) -> int:
+        """The node SET changed (scale-up materialized, node deleted):
+        rewrite pod_node indices in place instead of re-priming from a
+        full pod relist (74 ms at 1M pods). Bound pods whose node
+        vanished become unbound and lose the BOUND flag — exactly what
+        a relist would encode for a node name missing from the index.
+        Returns the number of pod rows whose binding changed."""
+        old = self.node_index
+        remap = np.full(max(len(old), 1), -1, np.int32)
+        for name, i in old.items():
+            j = new_node_index.get(name)
+            if j is not None:
+                remap[i] = j
+        pn = self.arrays["pod_node"][: self.size]
+        bound_idx = np.nonzero(pn >= 0)[0]
+        if bound_idx.size:
+            newv = remap[pn[bound_idx]]
+            changed = newv != pn[bound_idx]
+            nchanged = int(changed.sum())
+            if nchanged:
+                rows = bound_idx[changed]
+                pn[rows] = newv[changed]
+                lost = rows[newv[changed] < 0]
+                if lost.size:
+                    self.arrays["pod_flags"][lost] &= ~int(PodFlags.BOUND)
+                if rows.size > max(1024, self.size // 8):
+                    self.dirty_all = True
+                else:
+                    self.dirty.update(int(r) for r in rows)
+        else:
+            nchanged = 0
+        self.node_index = dict(new_node_index)
+        return nchanged
 
     @property
     def dead_fraction(self) -> float:
@@ -241,7 +282,8 @@ class DeltaSnapshotCache:
         self.uid_row = {u: i for i, u in enumerate(self.pod_uids[: len(live)]) if u}
         self.size = len(live)
         self.free_rows = []
-        self.dirty = set(range(self.size))  # full rescatter after compact
+        self.dirty = set()
+        self.dirty_all = True  # full rescatter after compact
         return True
 
     # ------------------------------------------------------------------
@@ -298,6 +340,7 @@ class DeltaSnapshotCache:
         self.pod_refs = refs + [("", "")] * (self.cap - self.size)
         self.uid_row = {u: i for i, u in enumerate(uids) if u}
         self.free_rows = [i for i, u in enumerate(uids) if not u]
-        self.dirty = set(range(self.size))
+        self.dirty = set()
+        self.dirty_all = True
         self.node_index = dict(node_index)
         self.pdb_selectors = list(pdb_selectors or [])

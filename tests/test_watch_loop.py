@@ -76,15 +76,87 @@ def test_watch_stream_drop_reprimes(server):
     assert res.pods_pending == 8  # nothing lost despite the drop
 
 
-def test_watch_node_churn_reprimes(server):
-    pods, nodes = burst_trace(4, n_nodes_per_pool=(3,), seed=2)
+def test_watch_node_churn_remaps_in_place(server):
+    """Node additions/removals must NOT re-prime (relist every pod):
+    the cache remaps pod->node indices in place and decisions stay
+    identical to a fresh full encode."""
+    pods, nodes = burst_trace(4, n_nodes_per_pool=(3,), seed=2,
+                              bound_pods_per_node=2)
     server.state.load(pods, nodes)
     cluster, _ = _cluster(server, no_maintenance=True, no_scale=True)
     cluster.loop_logic()
     first_state = cluster._watch_state
-    # add a node out of band: indices shift -> cache must re-prime
+    first_cache = first_state["cache"]
+    # add a node out of band: the watch cache must survive
     server.state.nodes["k8s-agentpool1-12345678-9"] = make_node_json("agentpool1", 9)
     res = cluster.loop_logic()
-    assert cluster._watch_state is not first_state
+    assert cluster._watch_state is first_state  # NOT re-primed
+    assert cluster._watch_state["cache"] is first_cache
     assert res.nodes_total == 4
     assert res.pods_pending == 4
+
+    # remove a node that has bound pods: indices of later nodes shift
+    # and the orphaned bindings must clear, exactly as a relist would
+    victim = sorted(server.state.nodes)[0]
+    del server.state.nodes[victim]
+    res2 = cluster.loop_logic()
+    assert cluster._watch_state is first_state
+    snap = res2.snapshot
+    from mi355x_autoscaler.kube.client import KubeApiClient as _K
+    from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools
+    from mi355x_autoscaler.models.kube import KubeNode, KubePod
+    from mi355x_autoscaler.snapshot.encode import encode_snapshot
+
+    kube = _K(server.base_url)
+    fresh_nodes = [KubeNode(n) for n in kube.list_nodes()]
+    fresh = encode_snapshot([KubePod(p) for p in kube.list_pods()],
+                            fresh_nodes, group_nodes_into_pools(fresh_nodes))
+    # same node table and identical per-node free capacity
+    assert snap.node_names == fresh.node_names
+    import numpy as np
+
+    order_a = np.argsort(np.array([u or "" for u in snap.pod_uids], dtype=object).astype(str))
+    order_b = np.argsort(np.array([u or "" for u in fresh.pod_uids], dtype=object).astype(str))
+    live_a = np.array([bool(u) for u in snap.pod_uids])[order_a]
+    np.testing.assert_array_equal(
+        snap.pod_node[order_a][live_a],
+        fresh.pod_node[order_b][np.array([bool(u) for u in fresh.pod_uids])[order_b]])
+    np.testing.assert_array_equal(snap.node_free, fresh.node_free)
+    d1 = CpuEngine().decide(snap)
+    d2 = CpuEngine().decide(fresh)
+    assert d1.units_needed == d2.units_needed
+    assert d1.pods_placed_existing == d2.pods_placed_existing
+
+
+def test_remap_nodes_unit():
+    """DeltaSnapshotCache.remap_nodes: shifted indices rewrite, lost
+    nodes unbind + clear BOUND, dirty rows are tracked."""
+    import numpy as np
+
+    from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools
+    from mi355x_autoscaler.models.kube import KubeNode, KubePod
+    from mi355x_autoscaler.snapshot.delta import DeltaSnapshotCache
+    from mi355x_autoscaler.snapshot.encode import PodFlags, encode_snapshot
+
+    nodes = [KubeNode(make_node_json("agentpool1", i)) for i in range(3)]
+    names = [n.name for n in nodes]
+    pods = [make_pod_json(f"b{i}", cpu_milli=100, node_name=names[i]) for i in range(3)]
+    pods.append(make_pod_json("p", cpu_milli=50, pending=True))
+    snap = encode_snapshot([KubePod(p) for p in pods], nodes,
+                           group_nodes_into_pools(nodes))
+    cache = DeltaSnapshotCache()
+    cache.prime(snap, {n: i for i, n in enumerate(names)})
+    cache.drain_dirty()
+
+    # drop node 0; node 1 -> 0, node 2 -> 1; new node appended at 2
+    new_index = {names[1]: 0, names[2]: 1, "k8s-agentpool1-12345678-7": 2}
+    changed = cache.remap_nodes(new_index)
+    assert changed == 3  # all three bound rows changed
+    pn = cache.arrays["pod_node"][: cache.size]
+    fl = cache.arrays["pod_flags"][: cache.size]
+    # rows 0..2 were the bound pods, in pod order
+    assert pn[0] == -1 and not (fl[0] & PodFlags.BOUND)
+    assert pn[1] == 0 and (fl[1] & PodFlags.BOUND)
+    assert pn[2] == 1 and (fl[2] & PodFlags.BOUND)
+    rows = cache.drain_dirty()
+    assert set(rows.tolist()) >= {0, 1, 2}
