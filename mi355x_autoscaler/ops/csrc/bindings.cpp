@@ -1,6 +1,7 @@
 // Python bindings for the MI355X autoscaler HIP kernels.
 // Kernels live in ops/hip/*.hip (CDNA4-native; no CUDA-compat paths).
 #include <torch/extension.h>
+#include <pybind11/numpy.h>
 
 void classify_keys(torch::Tensor pod_flags, torch::Tensor sel_hash,
                    torch::Tensor pod_req, torch::Tensor key_sel,
@@ -63,6 +64,11 @@ void pack_into(pybind11::buffer buf, const std::vector<int64_t>& offsets,
 pybind11::dict decode_watch_events(pybind11::bytes body,
                                    pybind11::dict node_index,
                                    pybind11::list pdb_selectors);
+pybind11::tuple resolve_batch_rows(
+    pybind11::array_t<int8_t, pybind11::array::c_style | pybind11::array::forcecast> types,
+    pybind11::list uids, pybind11::list refs, pybind11::dict uid_row,
+    pybind11::list free_rows, int64_t size, int64_t cap,
+    pybind11::list pod_uids, pybind11::list pod_refs);
 void fused_tail_launch(
     torch::Tensor pod_req, torch::Tensor pod_node, torch::Tensor node_alloc,
     torch::Tensor node_labels, torch::Tensor node_label_count,
@@ -132,6 +138,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_into", &pack_into, "memcpy arrays into the staging buffer");
   m.def("decode_watch_events", &decode_watch_events,
         "JSONL watch stream -> event types + pod SoA columns");
+  m.def("resolve_batch_rows", &resolve_batch_rows,
+        "watch batch -> target rows (C-speed uid/row bookkeeping)");
   m.def("fused_decide_launch", &fused_decide_launch,
         "enqueue the whole scale-up pipeline in one call");
   m.def("fused_tail_launch", &fused_tail_launch,

@@ -168,8 +168,55 @@ class DeltaSnapshotCache:
 
     def apply_batch(self, decoded: Dict) -> int:
         """Apply a C++-decoded watch batch (decode_watch_events output):
-        per-event row resolution stays Python (a dict op), the column
-        writes are vectorized. Returns events applied."""
+        row resolution runs in C (resolve_batch_rows) when the native
+        extension is present, the column writes are vectorized; the
+        Python loop below remains the reference/fallback (differential-
+        tested). Returns events applied."""
+        types = decoded["types"]
+        n = int(types.shape[0])
+        if n == 0:
+            return 0
+        resolve = None
+        try:
+            import torch  # noqa: F401  (extension links against libtorch)
+
+            from mi355x_autoscaler import _hip_ops
+
+            resolve = _hip_ops.resolve_batch_rows
+        except (ImportError, AttributeError):
+            pass
+        if resolve is not None:
+            # pre-grow so capacity can never run out mid-batch
+            n_upserts = int((types == 0).sum())
+            while (self.cap - self.size) + len(self.free_rows) < n_upserts:
+                self._grow()
+            rows, sel, cleared, new_size = resolve(
+                types, decoded["pod_uids"], decoded["pod_refs"],
+                self.uid_row, self.free_rows, self.size, self.cap,
+                self.pod_uids, self.pod_refs,
+            )
+            self.size = int(new_size)
+            if cleared.size:
+                for col in _POD_COLS:
+                    self.arrays[col][cleared] = 0
+                self.arrays["pod_node"][cleared] = -1
+                self.arrays["pod_pdb"][cleared] = -1
+                self.dirty.update(cleared.tolist())
+            if rows.size:
+                a = self.arrays
+                a["pod_req"][rows] = decoded["pod_req"][sel]
+                a["pod_flags"][rows] = decoded["pod_flags"][sel]
+                a["pod_sel_hash"][rows] = decoded["pod_sel_hash"][sel]
+                a["pod_sel_pairs"][rows] = decoded["pod_sel_pairs"][sel]
+                a["pod_sel_count"][rows] = decoded["pod_sel_count"][sel]
+                a["pod_node"][rows] = decoded["pod_node"][sel]
+                a["pod_pdb"][rows] = decoded["pod_pdb"][sel]
+                self.dirty.update(rows.tolist())
+            return n
+        return self._apply_batch_py(decoded)
+
+    def _apply_batch_py(self, decoded: Dict) -> int:
+        """Pure-Python reference for apply_batch's row resolution."""
         types = decoded["types"]
         n = int(types.shape[0])
         if n == 0:

@@ -162,3 +162,52 @@ def test_decode_watch_events_batch_matches_per_event():
         np.testing.assert_array_equal(
             getattr(sa, col)[order_a], getattr(sb, col)[order_b], err_msg=col)
     assert sorted(u for u in sa.pod_uids if u) == sorted(u for u in sb.pod_uids if u)
+
+
+def test_resolve_batch_rows_matches_python_loop():
+    """C++ row resolution == the Python reference loop on adversarial
+    orderings: duplicate MODIFIED, ADD-then-DELETE, DELETE-then-re-ADD
+    (row reuse within one batch)."""
+    import mi355x_autoscaler._hip_ops as ops
+    from mi355x_autoscaler.models.kube import KubeNode, KubePod
+    from mi355x_autoscaler.snapshot.delta import DeltaSnapshotCache
+
+    nodes = [KubeNode(make_node_json("agentpool1", i)) for i in range(2)]
+    pools = group_nodes_into_pools(nodes)
+    node_index = {n.name: i for i, n in enumerate(nodes)}
+    initial = [make_pod_json(f"p{i}", cpu_milli=100, pending=True) for i in range(5)]
+    snap0 = encode_snapshot([KubePod(p) for p in initial], nodes, pools)
+    base = encode_snapshot([], nodes, pools)
+
+    events = []
+    events.append({"type": "MODIFIED", "object": make_pod_json("p0", cpu_milli=111, pending=True)})
+    events.append({"type": "MODIFIED", "object": make_pod_json("p0", cpu_milli=222, pending=True)})
+    events.append({"type": "ADDED", "object": make_pod_json("tmp", cpu_milli=300, pending=True)})
+    events.append({"type": "DELETED", "object": make_pod_json("tmp")})
+    events.append({"type": "DELETED", "object": initial[1]})
+    events.append({"type": "ADDED", "object": make_pod_json("reuse", cpu_milli=400, pending=True)})
+    events.append({"type": "DELETED", "object": make_pod_json("ghost-never-seen")})
+    blob = ("\n".join(json.dumps(e) for e in events) + "\n").encode()
+    decoded = ops.decode_watch_events(blob, node_index, [])
+
+    caches = []
+    for use_cpp in (True, False):
+        c = DeltaSnapshotCache()
+        c.prime(snap0, node_index)
+        c.drain_dirty()
+        if use_cpp:
+            c.apply_batch(decoded)
+        else:
+            c._apply_batch_py(decoded)
+        caches.append(c)
+    ca, cb = caches
+    assert sorted(u for u in ca.pod_uids[: ca.size] if u) == \
+           sorted(u for u in cb.pod_uids[: cb.size] if u)
+    assert ca.uid_row == cb.uid_row
+    assert sorted(ca.free_rows) == sorted(cb.free_rows)
+    sa, sb = ca.build_snapshot(base), cb.build_snapshot(base)
+    for col in POD_COLS:
+        np.testing.assert_array_equal(getattr(sa, col), getattr(sb, col), err_msg=col)
+    # the duplicate-MODIFIED pod must hold the LAST value
+    row = ca.uid_row[[u for u in ca.pod_uids if u and "p0" in u][0]]
+    assert ca.arrays["pod_req"][row][0] == 222
