@@ -37,7 +37,7 @@ namespace {
 
 // ---- lattice constants (must match snapshot/encode.py) --------------------
 constexpr int NRES = 4;
-constexpr int MAXSEL = 8;
+constexpr int MAXSEL = 16;
 constexpr int64_t MIB = 1024 * 1024;
 
 constexpr uint16_t PF_PENDING = 1 << 0;

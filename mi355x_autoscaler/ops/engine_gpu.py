@@ -40,6 +40,7 @@ import torch
 
 from ..sim.decision import DrainCandidate, MaintainReport, ScaleDecision
 from ..sim.engine import SimulationEngine
+from ..snapshot.encode import MAX_SELECTOR_PAIRS as _MAXSEL
 from ..snapshot.encode import ClusterSnapshot
 from . import load_ops
 
@@ -59,7 +60,7 @@ class DeviceSnapshot:
     pod_req: torch.Tensor        # [P,4] i32
     pod_flags: torch.Tensor      # [P]   i32
     pod_sel_hash: torch.Tensor   # [P]   i64
-    pod_sel_pairs: torch.Tensor  # [P,8] i64
+    pod_sel_pairs: torch.Tensor  # [P,MAXSEL] i64
     pod_sel_count: torch.Tensor  # [P]   i32
     pod_node: torch.Tensor       # [P]   i32
     pod_pdb: torch.Tensor        # [P]   i32
@@ -95,7 +96,7 @@ _POD_FIELDS = [
     ("pod_req", "pod_req", np.int32, (4,)),
     ("pod_flags", "pod_flags", np.int32, ()),
     ("pod_sel_hash", "pod_sel_hash", np.int64, ()),
-    ("pod_sel_pairs", "pod_sel_pairs", np.int64, (8,)),
+    ("pod_sel_pairs", "pod_sel_pairs", np.int64, (_MAXSEL,)),
     ("pod_sel_count", "pod_sel_count", np.int32, ()),
     ("pod_node", "pod_node", np.int32, ()),
     ("pod_pdb", "pod_pdb", np.int32, ()),
@@ -359,7 +360,7 @@ class GpuEngine(SimulationEngine):
         self.ws_slot_first = self._ws_i32[2 * HASH_SLOTS : 3 * HASH_SLOTS]
         self.ws_stat = self._ws_i32[3 * HASH_SLOTS :]  # [G, sortovf, hashovf]
         self.ws_slot2seg = torch.zeros(HASH_SLOTS, dtype=torch.int32, device=d)
-        self.ws_seg_sel_pairs = torch.zeros((SEG_MAX, 8), dtype=torch.int64, device=d)
+        self.ws_seg_sel_pairs = torch.zeros((SEG_MAX, _MAXSEL), dtype=torch.int64, device=d)
         self.ws_seg_sel_count = torch.zeros(SEG_MAX, dtype=torch.int32, device=d)
         self.ws_seg_sel_hash = torch.zeros(SEG_MAX, dtype=torch.int64, device=d)
         self.ws_seg_req = torch.zeros((SEG_MAX, 4), dtype=torch.int32, device=d)

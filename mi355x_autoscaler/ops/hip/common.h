@@ -10,7 +10,7 @@
 #include <stdint.h>
 
 #define NRES 4            // cpu_milli, mem_mib, pods, gpu — fixed row order
-#define MAXSEL 8          // padded selector pair-hash width (pods)
+#define MAXSEL 16         // padded selector pair-hash width (pods)
 #define MAXLAB 32         // padded label pair-hash width (nodes/pools)
 #define WAVE 64
 

@@ -43,7 +43,7 @@ MIB = 1024 * 1024
 # selector pairs than MAX_SELECTOR_PAIRS or nodes with more labels than
 # MAX_LABEL_PAIRS fall back to exact host-side matching (rare; counted
 # in the encode stats).
-MAX_SELECTOR_PAIRS = 8
+MAX_SELECTOR_PAIRS = 16
 MAX_LABEL_PAIRS = 32
 
 _FNV_OFFSET = 0xCBF29CE484222325

@@ -470,3 +470,18 @@ def test_hipgraph_optin_replay_matches_eager_and_oracle(engine, monkeypatch):
     if graphed._graph_ok:  # capture may legitimately disable itself
         assert any(not isinstance(v, str) for v in graphed._graph_cache.values()), \
             "expected at least one captured hipGraph after repeated decides"
+
+
+def test_wide_selector_decide_matches_oracle(engine):
+    """12-pair selector sets (old width was 8) through the full GPU
+    pipeline: bit-equal to the oracle, correct node discrimination."""
+    sel = {f"k{i:02d}": f"v{i}" for i in range(12)}
+    full = make_node_json("agentpool1", 0)
+    full["metadata"]["labels"].update(sel)
+    partial = make_node_json("agentpool1", 1)
+    partial["metadata"]["labels"].update({k: sel[k] for k in list(sel)[:11]})
+    pods = [make_pod_json(f"w{i}", cpu_milli=100, pending=True, node_selector=sel)
+            for i in range(5)]
+    pods += [make_pod_json("plain", cpu_milli=100, pending=True)]
+    snap = _snap(pods, [full, partial])
+    _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))

@@ -143,3 +143,31 @@ def test_synth_soa_million_scale_fast():
     assert snap.num_pods == 1_000_000
     assert dt < 10.0
     assert snap.nbytes() < 300 * 1024 * 1024
+
+
+def test_wide_selector_sets_match_exactly():
+    """12 selector pairs (> the old 8-pair width) must match exactly —
+    no truncation over-match; >16 still sets SELECTOR_OVERFLOW."""
+    from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools
+    from mi355x_autoscaler.models.kube import KubeNode, KubePod
+    from mi355x_autoscaler.sim.engine import CpuEngine
+    from mi355x_autoscaler.sim.synth import make_node_json, make_pod_json
+    from mi355x_autoscaler.snapshot.encode import PodFlags, encode_snapshot
+
+    sel = {f"k{i:02d}": f"v{i}" for i in range(12)}
+    node_full = make_node_json("agentpool1", 0)
+    node_full["metadata"]["labels"].update(sel)
+    node_partial = make_node_json("agentpool1", 1)
+    node_partial["metadata"]["labels"].update({k: sel[k] for k in list(sel)[:11]})
+    nodes = [KubeNode(node_full), KubeNode(node_partial)]
+    pod = make_pod_json("wide", cpu_milli=100, pending=True, node_selector=sel)
+    snap = encode_snapshot([KubePod(pod)], nodes, group_nodes_into_pools(nodes))
+    assert snap.pod_sel_count[0] == 12
+    assert not (snap.pod_flags[0] & PodFlags.SELECTOR_OVERFLOW)
+    d = CpuEngine().decide(snap)
+    assert d.pods_placed_existing == 1  # fits node_full only
+
+    sel17 = {f"k{i:02d}": f"v{i}" for i in range(17)}
+    pod17 = make_pod_json("wider", cpu_milli=100, pending=True, node_selector=sel17)
+    snap17 = encode_snapshot([KubePod(pod17)], nodes, group_nodes_into_pools(nodes))
+    assert snap17.pod_flags[0] & PodFlags.SELECTOR_OVERFLOW
