@@ -24,6 +24,16 @@ def load_ops():
     try:
         from mi355x_autoscaler import _hip_ops as ext  # type: ignore
 
+        if hasattr(ext, "compiled_constants"):
+            from ..snapshot.encode import MAX_SELECTOR_PAIRS
+
+            cc = ext.compiled_constants()
+            if cc["MAXSEL"] != MAX_SELECTOR_PAIRS:
+                raise RuntimeError(
+                    f"_hip_ops compiled with MAXSEL={cc['MAXSEL']} but the "
+                    f"lattice uses {MAX_SELECTOR_PAIRS}: stale build — "
+                    "rebuild with `python setup.py build_ext --inplace`"
+                )
         _hip_ops = ext
         return ext
     except ImportError as e:  # pragma: no cover

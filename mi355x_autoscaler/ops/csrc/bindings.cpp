@@ -140,6 +140,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "JSONL watch stream -> event types + pod SoA columns");
   m.def("resolve_batch_rows", &resolve_batch_rows,
         "watch batch -> target rows (C-speed uid/row bookkeeping)");
+  m.def("compiled_constants", [] {
+    pybind11::dict d;
+    d["MAXSEL"] = 16;   // must match ops/hip/common.h and encode.py
+    d["MAXLAB"] = 32;
+    d["SEG_MAX"] = 2048;
+    return d;
+  }, "lattice constants this extension was compiled with");
   m.def("fused_decide_launch", &fused_decide_launch,
         "enqueue the whole scale-up pipeline in one call");
   m.def("fused_tail_launch", &fused_tail_launch,

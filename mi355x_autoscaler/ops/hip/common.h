@@ -42,14 +42,17 @@ __device__ __host__ inline int64_t pack_req_key(const int32_t req[NRES],
 
 // Selector-subset predicate over sorted pair-hash arrays — device form of
 // snapshot/encode.py::selector_subset_match. `sel` is wave-uniform in
-// every caller; `lab` may live in LDS or global.
+// every caller; `lab` may live in LDS or global. The producer sorts
+// the raw 64-bit FNV hashes UNSIGNED before the two's-complement int64
+// cast, so the walk compares unsigned (a signed walk silently fails
+// whenever a selector mixes positive- and negative-cast hashes).
 __device__ inline bool sel_subset(const int64_t* __restrict__ sel, int nsel,
                                   const int64_t* __restrict__ lab, int nlab) {
   int li = 0;
   for (int si = 0; si < nsel; ++si) {
-    const int64_t s = sel[si];
-    while (li < nlab && lab[li] < s) ++li;
-    if (li >= nlab || lab[li] != s) return false;
+    const uint64_t s = (uint64_t)sel[si];
+    while (li < nlab && (uint64_t)lab[li] < s) ++li;
+    if (li >= nlab || (uint64_t)lab[li] != s) return false;
   }
   return true;
 }

@@ -319,12 +319,16 @@ def selector_subset_match(
     sel_pairs: np.ndarray, sel_count: int, label_pairs: np.ndarray, label_count: int
 ) -> bool:
     """True iff every selector pair hash appears in the label hash set.
-    Both arrays sorted ascending. Mirrors the device predicate."""
+    Both arrays sorted ascending AS UNSIGNED 64-bit (the producer sorts
+    the raw FNV hashes before the two's-complement int64 cast), so the
+    two-pointer walk must compare unsigned too. Mirrors the device
+    predicate (common.h sel_subset)."""
+    _M = (1 << 64) - 1
     li = 0
     for si in range(sel_count):
-        s = sel_pairs[si]
-        while li < label_count and label_pairs[li] < s:
+        s = int(sel_pairs[si]) & _M
+        while li < label_count and (int(label_pairs[li]) & _M) < s:
             li += 1
-        if li >= label_count or label_pairs[li] != s:
+        if li >= label_count or (int(label_pairs[li]) & _M) != s:
             return False
     return True

@@ -171,3 +171,47 @@ def test_wide_selector_sets_match_exactly():
     pod17 = make_pod_json("wider", cpu_milli=100, pending=True, node_selector=sel17)
     snap17 = encode_snapshot([KubePod(pod17)], nodes, group_nodes_into_pools(nodes))
     assert snap17.pod_flags[0] & PodFlags.SELECTOR_OVERFLOW
+
+
+def test_subset_match_mixed_sign_hashes():
+    """The pair hashes are sorted UNSIGNED; the subset walk must compare
+    unsigned too (a 12-pair selector mixes positive- and negative-cast
+    int64 hashes — the old signed walk failed on exactly that)."""
+    from mi355x_autoscaler.snapshot.encode import _hash_pairs, selector_subset_match
+
+    sel = {f"k{i:02d}": f"v{i}" for i in range(12)}
+    labels = dict(sel)
+    labels.update({f"extra{i}": "x" for i in range(5)})
+    sp, sc, _ = _hash_pairs(sel, 16)
+    lp, lc, _ = _hash_pairs(labels, 32)
+    assert any(int(x) < 0 for x in sp[:sc]) and any(int(x) > 0 for x in sp[:sc])
+    assert selector_subset_match(sp, sc, lp, min(lc, 32))
+    # removing one selected label must break the match
+    labels2 = dict(labels)
+    del labels2["k07"]
+    lp2, lc2, _ = _hash_pairs(labels2, 32)
+    assert not selector_subset_match(sp, sc, lp2, min(lc2, 32))
+
+
+def test_pool_routing_wide_selector():
+    """Oracle pool routing (selector_subset_match walk) with a 12-pair
+    selector: the pool whose prototype labels carry all pairs routes."""
+    from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools
+    from mi355x_autoscaler.models.kube import KubeNode, KubePod
+    from mi355x_autoscaler.sim.engine import CpuEngine
+    from mi355x_autoscaler.sim.synth import make_node_json, make_pod_json
+    from mi355x_autoscaler.snapshot.encode import encode_snapshot
+
+    sel = {f"k{i:02d}": f"v{i}" for i in range(12)}
+    node = make_node_json("agentpool1", 0, cpu_milli=2000, mem_mib=2048)
+    node["metadata"]["labels"].update(sel)
+    nodes = [KubeNode(node)]
+    # more demand than one node holds -> must route to the pool
+    pods = [make_pod_json(f"w{i}", cpu_milli=1500, pending=True, node_selector=sel)
+            for i in range(6)]
+    snap = encode_snapshot([KubePod(p) for p in pods], nodes,
+                           group_nodes_into_pools(nodes))
+    d = CpuEngine().decide(snap)
+    assert d.pods_placed_existing == 1
+    assert d.units_needed.get("agentpool1", 0) >= 1
+    assert d.pods_unfulfillable == 0
