@@ -16,6 +16,27 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 
 ROOT = os.path.dirname(os.path.abspath(__file__))
 
+
+def _invalidate_stale_objects():
+    """distutils tracks NO header dependencies for hipcc-compiled .hip
+    sources: editing common.h / pipeline_device.h relinks stale objects
+    silently (which once shipped kernels with two different MAXSEL
+    widths in one .so). Delete any object older than the newest header
+    so header edits force recompilation."""
+    import glob
+
+    hdrs = glob.glob(os.path.join(ROOT, "mi355x_autoscaler", "ops", "**", "*.h"),
+                     recursive=True)
+    if not hdrs:
+        return
+    newest = max(os.path.getmtime(h) for h in hdrs)
+    for o in glob.glob(os.path.join(ROOT, "build", "**", "*.o"), recursive=True):
+        if os.path.getmtime(o) < newest:
+            os.remove(o)
+
+
+_invalidate_stale_objects()
+
 ext = CUDAExtension(
     name="mi355x_autoscaler._hip_ops",
     sources=[
