@@ -348,6 +348,11 @@ class GpuEngine(SimulationEngine):
         self._resident: Optional[dict] = None
         self._rest_packer = _PinnedPacker(_REST_FIELDS)
         self._last_demand: Optional[torch.Tensor] = None
+        # MFMA demand segsum runs on a side stream, off the decision's
+        # critical path (it fed only the pool-demand metric yet sat
+        # between the pipeline and the readback: ~200 us at 1M pods)
+        self._side_stream = torch.cuda.Stream(self.device) if torch.cuda.is_available() else None
+        self._side_done: Optional[torch.Tensor] = None  # cuda Event when pending
         d = self.device
         # persistent hash workspace; the mutable i32 parts live in ONE
         # buffer so per-decide reinit is a single fill kernel
@@ -523,6 +528,11 @@ class GpuEngine(SimulationEngine):
         W = (N + 63) // 64
         elig = self._elig_buffer(W)
         want_pod_pool = bool(K and K <= 16)
+        if self._side_done is not None:
+            # pod_pool / demand buffers are rewritten below: order the
+            # main stream after the previous cycle's side-stream segsum
+            torch.cuda.current_stream().wait_event(self._side_done)
+            self._side_done = None
         bufs = self._decide_buffers(ds.P, K, want_pod_pool)
         slot_of_pod = bufs["slot_of_pod"]
         # +2 tail slots: the orchestrator copies the 3-word i32 stat there
@@ -547,9 +557,23 @@ class GpuEngine(SimulationEngine):
                 (HASH_SLOTS, SEG_MAX, 1 if want_pod_pool else 0),
             )
 
-        # MFMA per-pool demand (zero-sort path; skipped for >16 pools)
+        # MFMA per-pool demand (zero-sort path; skipped for >16 pools) —
+        # enqueued on the side stream AFTER the pipeline's pod_pool map,
+        # overlapping the blob readback + host decision finish. The next
+        # decide waits on it (stream-level) before rewriting pod_pool.
         if want_pod_pool:
-            self._last_demand = segsum(ds.pod_req, pod_pool, K, ops)
+            if self._side_stream is not None:
+                ev = torch.cuda.Event()
+                ev.record()
+                with torch.cuda.stream(self._side_stream):
+                    self._side_stream.wait_event(ev)
+                    self._last_demand = segsum(ds.pod_req, pod_pool, K, ops)
+                    done = torch.cuda.Event()
+                    done.record(self._side_stream)
+                    self._side_done = done
+                self._last_demand.record_stream(self._side_stream)
+            else:
+                self._last_demand = segsum(ds.pod_req, pod_pool, K, ops)
         else:
             self._last_demand = None
 
@@ -586,6 +610,14 @@ class GpuEngine(SimulationEngine):
             max_new_h, ds.pool_cur_h, K, pool_names, over_provision,
             lambda: blob_h[-2 - SEG_MAX : -2],
         )
+
+    def pool_demand(self) -> Optional[torch.Tensor]:
+        """Per-pool pending demand totals from the MFMA segsum of the
+        last decide (None when >16 pools). Synchronizes the side
+        stream before handing the tensor out."""
+        if self._side_done is not None:
+            self._side_done.synchronize()
+        return self._last_demand
 
     def _free_buffer(self, N: int) -> torch.Tensor:
         buf = self._free_cache.get(N)
