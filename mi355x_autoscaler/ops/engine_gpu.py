@@ -409,6 +409,11 @@ class GpuEngine(SimulationEngine):
                                          pin_memory=torch.cuda.is_available()),
                 "pod_pool": torch.empty(P if want_pod_pool else 0,
                                         dtype=torch.int32, device=d),
+                # second buffer: resident mode alternates so the side-
+                # stream demand segsum of cycle n never collides with
+                # cycle n+1's pod_pool_map
+                "pod_pool_b": torch.empty(P if want_pod_pool else 0,
+                                          dtype=torch.int32, device=d),
             }
             if len(self._decide_bufs) > 4:
                 self._decide_bufs.clear()
@@ -513,7 +518,8 @@ class GpuEngine(SimulationEngine):
         return buf
 
     def _decide_fused(self, ds: DeviceSnapshot, pool_names: List[str],
-                      over_provision: int = 0) -> Optional[ScaleDecision]:
+                      over_provision: int = 0,
+                      stable_pods: bool = False) -> Optional[ScaleDecision]:
         """The zero-mid-sync decision pipeline: every kernel reads G
         from device memory; ONE sync at the result readback. The whole
         launch sequence is enqueued by the native orchestrator
@@ -528,16 +534,23 @@ class GpuEngine(SimulationEngine):
         W = (N + 63) // 64
         elig = self._elig_buffer(W)
         want_pod_pool = bool(K and K <= 16)
-        if self._side_done is not None:
-            # pod_pool / demand buffers are rewritten below: order the
-            # main stream after the previous cycle's side-stream segsum
+        if self._side_done is not None and not stable_pods:
+            # pod_pool / pod_req are rewritten below: order the main
+            # stream after the previous cycle's side-stream segsum.
+            # (Resident mode double-buffers pod_pool and waits at the
+            # scatter instead, where the wait has ~a full readback of
+            # natural slack.)
             torch.cuda.current_stream().wait_event(self._side_done)
             self._side_done = None
         bufs = self._decide_buffers(ds.P, K, want_pod_pool)
         slot_of_pod = bufs["slot_of_pod"]
         # +2 tail slots: the orchestrator copies the 3-word i32 stat there
         blob = bufs["blob"]
-        pod_pool = bufs["pod_pool"]
+        if stable_pods and want_pod_pool:
+            self._pp_flip = not getattr(self, "_pp_flip", False)
+            pod_pool = bufs["pod_pool_b"] if self._pp_flip else bufs["pod_pool"]
+        else:
+            pod_pool = bufs["pod_pool"]
 
         if self.force_py_pipeline:
             self._enqueue_py(ds, slot_of_pod, elig, blob, pod_pool,
@@ -978,6 +991,12 @@ class GpuEngine(SimulationEngine):
                                     "pod_views": pod_views}
             cache.drain_dirty()
         else:
+            if self._side_done is not None:
+                # the scatter rewrites pod columns the side-stream
+                # demand segsum may still read; stream-level wait here
+                # (never a host sync)
+                torch.cuda.current_stream().wait_event(self._side_done)
+                self._side_done = None
             self._scatter_rows(res["pod_views"], cache, cache.drain_dirty())
 
         rest_views = self._rest_packer.pack_views(base, self.device)
@@ -986,7 +1005,8 @@ class GpuEngine(SimulationEngine):
             pool_max_h=np.ascontiguousarray(base.pool_max, dtype=np.int32).copy(),
             pool_cur_h=np.ascontiguousarray(base.pool_cur, dtype=np.int32).copy(),
         )
-        decision = self._decide_fused(ds, base.pool_names, over_provision)
+        decision = self._decide_fused(ds, base.pool_names, over_provision,
+                                      stable_pods=True)
         if decision is None:  # hash overflow: fall back via a full snapshot
             snap = cache.build_snapshot(base)
             full = self.packer.pack(snap, self.device)
