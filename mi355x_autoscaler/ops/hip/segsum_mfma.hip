@@ -20,8 +20,12 @@
 //
 // Work decomposition: elements must arrive sorted by segment id.
 // Segments are tiled 16 per MFMA *window*; a work item = (window,
-// 256-element chunk of that window's element range). One wave per work
-// item: 64 MFMA instructions, then one 8-lane flush.
+// 256-element chunk of that window's element range). Each wave walks
+// work items grid-strided, accumulating the exact per-chunk integer
+// partials in int64 REGISTERS across chunks of the same window and
+// flushing once per window — the flush-per-chunk form serialized
+// ~30k atomics onto K*NRES hot words (202 us at 1M pods, the whole
+// kernel; the MFMA math itself is microseconds).
 #include "common.h"
 
 typedef float f32x4 __attribute__((ext_vector_type(4)));
@@ -44,32 +48,9 @@ extern "C" __global__ __launch_bounds__(WAVE) void segsum_mfma_kernel(
   __shared__ int32_t l_vals[256 * NRES];
   __shared__ int32_t l_segs[256];
 
-  const int64_t t = blockIdx.x;
-  if (t >= T) return;
   const int lane = threadIdx.x & (WAVE - 1);
   const int col = lane & 15;      // A row index i / B,D column j
   const int kidx = lane >> 4;     // K slot 0..3 (one pod each)
-
-  const int win = wi_win[t];
-  const int64_t p_begin = wi_start[t];
-  const int64_t p_end = wi_end[t];
-  const int seg_base = win * 16;
-
-  // stage: lane covers pods {lane, lane+64, lane+128, lane+192}
-#pragma unroll
-  for (int q = 0; q < 4; ++q) {
-    const int pl = lane + 64 * q;
-    const int64_t p = p_begin + pl;
-    if (p < p_end) {
-      const int4 v = reinterpret_cast<const int4*>(values)[p];
-      reinterpret_cast<int4*>(l_vals)[pl] = v;
-      l_segs[pl] = seg_of[p] - seg_base;
-    } else {
-      reinterpret_cast<int4*>(l_vals)[pl] = int4{0, 0, 0, 0};
-      l_segs[pl] = -1;  // no column matches: contributes zero
-    }
-  }
-  __syncthreads();
 
   // A-row decode for this lane: rows 0..7 = (limb = row>>2, res = row&3)
   const int row = col;
@@ -77,37 +58,74 @@ extern "C" __global__ __launch_bounds__(WAVE) void segsum_mfma_kernel(
   const int limb = (row >> 2) & 1;
   const bool live_row = row < 8;
 
-  const int steps = (int)((p_end - p_begin + 3) >> 2);
-  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-  for (int s = 0; s < steps; ++s) {
-    const int pl = s * 4 + kidx;
-    const float b = (l_segs[pl] == col) ? 1.f : 0.f;
-    float a = 0.f;
-    if (live_row) {
-      const uint32_t v = (uint32_t)l_vals[pl * NRES + res];
-      a = (float)(limb ? (v >> 16) : (v & 0xffffu));
-    }
-    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
-  }
+  int cur_win = -1;
+  long long p64[4] = {0, 0, 0, 0};  // exact int64 window partials
 
-  // C/D layout for 16x16x4: col = lane&15, row = (lane>>4)*4 + reg.
-  const int s = seg_base + col;
-  if (s < S) {
+  auto flush = [&](int win) {
+    const int s = win * 16 + col;
+    if (s < S) {
 #pragma unroll
-    for (int reg = 0; reg < 4; ++reg) {
-      const int orow = (lane >> 4) * 4 + reg;
-      if (orow < 8) {
-        const int ores = orow & 3;
-        const int olimb = (orow >> 2) & 1;
-        const unsigned long long partial = (unsigned long long)(int64_t)acc[reg];
-        if (partial) {
-          atomicAdd(reinterpret_cast<unsigned long long*>(
-                        &out[(int64_t)s * NRES + ores]),
-                    olimb ? (partial << 16) : partial);
+      for (int reg = 0; reg < 4; ++reg) {
+        const int orow = (lane >> 4) * 4 + reg;
+        if (orow < 8) {
+          const int ores = orow & 3;
+          const int olimb = (orow >> 2) & 1;
+          const unsigned long long partial = (unsigned long long)p64[reg];
+          if (partial) {
+            atomicAdd(reinterpret_cast<unsigned long long*>(
+                          &out[(int64_t)s * NRES + ores]),
+                      olimb ? (partial << 16) : partial);
+          }
         }
       }
     }
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) p64[reg] = 0;
+  };
+
+  for (int64_t t = blockIdx.x; t < T; t += gridDim.x) {
+    const int win = wi_win[t];
+    if (win != cur_win) {
+      if (cur_win >= 0) flush(cur_win);
+      cur_win = win;
+    }
+    const int64_t p_begin = wi_start[t];
+    const int64_t p_end = wi_end[t];
+    const int seg_base = win * 16;
+
+    // stage: lane covers pods {lane, lane+64, lane+128, lane+192}
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int pl = lane + 64 * q;
+      const int64_t p = p_begin + pl;
+      if (p < p_end) {
+        const int4 v = reinterpret_cast<const int4*>(values)[p];
+        reinterpret_cast<int4*>(l_vals)[pl] = v;
+        l_segs[pl] = seg_of[p] - seg_base;
+      } else {
+        reinterpret_cast<int4*>(l_vals)[pl] = int4{0, 0, 0, 0};
+        l_segs[pl] = -1;  // no column matches: contributes zero
+      }
+    }
+    __syncthreads();
+
+    const int steps = (int)((p_end - p_begin + 3) >> 2);
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    for (int sstep = 0; sstep < steps; ++sstep) {
+      const int pl = sstep * 4 + kidx;
+      const float b = (l_segs[pl] == col) ? 1.f : 0.f;
+      float a = 0.f;
+      if (live_row) {
+        const uint32_t v = (uint32_t)l_vals[pl * NRES + res];
+        a = (float)(limb ? (v >> 16) : (v & 0xffffu));
+      }
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+    }
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) p64[reg] += (long long)acc[reg];
+    __syncthreads();  // LDS reused by the next chunk
   }
+  if (cur_win >= 0) flush(cur_win);
 }
 
 #include <torch/extension.h>
@@ -123,7 +141,10 @@ void segsum_mfma(torch::Tensor values, torch::Tensor seg_of,
   const int64_t T = wi_win.size(0);
   if (T == 0) return;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  hipLaunchKernelGGL(segsum_mfma_kernel, dim3((uint32_t)T), dim3(WAVE), 0,
+  // enough single-wave blocks to fill the SIMD slots; each walks
+  // work items grid-strided with register-accumulated partials
+  const uint32_t grid = (uint32_t)std::min<int64_t>(T, 1024);
+  hipLaunchKernelGGL(segsum_mfma_kernel, dim3(grid), dim3(WAVE), 0,
                      stream.stream(), values.data_ptr<int32_t>(),
                      seg_of.data_ptr<int32_t>(), wi_win.data_ptr<int32_t>(),
                      wi_start.data_ptr<int64_t>(), wi_end.data_ptr<int64_t>(),
