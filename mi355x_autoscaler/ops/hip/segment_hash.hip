@@ -272,6 +272,12 @@ void segment_hash(torch::Tensor pod_flags, torch::Tensor sel_hash,
     // with wave count, the scan itself is bandwidth-trivial, so run
     // ~4 pods per thread instead of 1 (18.2 -> measured lower at 10k)
     grid = (int)std::max<int64_t>(1, std::min<int64_t>((P + 1023) / 1024, 2048));
+  } else {
+    // LDS-table form: the per-BLOCK flush serializes on the global
+    // slots (2048 blocks x ~distinct keys = 26k same-line atomics at
+    // 1M pods); ~16 pods per thread cuts the flush count 8x while the
+    // grid-strided scan stays bandwidth-bound
+    grid = (int)std::max<int64_t>(8, std::min<int64_t>((P + 4095) / 4096, 2048));
   }
   hipLaunchKernelGGL(kfn, dim3(grid), dim3(block), 0,
                      stream.stream(), pod_flags.data_ptr<int32_t>(),
