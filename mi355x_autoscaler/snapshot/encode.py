@@ -46,6 +46,25 @@ MIB = 1024 * 1024
 MAX_SELECTOR_PAIRS = 16
 MAX_LABEL_PAIRS = 32
 
+
+_label_overflow_warned = False
+
+
+def _warn_label_overflow(name: str, cnt: int) -> None:
+    """Nodes with more label pairs than the lattice width match on the
+    first MAX_LABEL_PAIRS (unsigned-sorted) only — consistent across
+    CPU/GPU engines but potentially narrower than exact k8s semantics.
+    Surface it loudly once instead of truncating silently."""
+    global _label_overflow_warned
+    if not _label_overflow_warned:
+        import logging
+
+        logging.getLogger(__name__).warning(
+            "node %s has %d label pairs (> lattice width %d): selector "
+            "matching uses the first %d sorted pairs", name, cnt,
+            MAX_LABEL_PAIRS, MAX_LABEL_PAIRS)
+        _label_overflow_warned = True
+
 _FNV_OFFSET = 0xCBF29CE484222325
 _FNV_PRIME = 0x100000001B3
 _MASK64 = (1 << 64) - 1
@@ -214,6 +233,8 @@ def encode_snapshot(
         node_alloc[i] = quantize_allocatable(n.allocatable)
         node_labels[i], cnt, _ = _hash_pairs(n.labels, MAX_LABEL_PAIRS)
         node_label_count[i] = min(cnt, MAX_LABEL_PAIRS)
+        if cnt > MAX_LABEL_PAIRS:
+            _warn_label_overflow(n.name, cnt)
         node_unsched[i] = 1 if n.unschedulable else 0
         if n.pool_name in pool_index:
             node_pool[i] = pool_index[n.pool_name]
