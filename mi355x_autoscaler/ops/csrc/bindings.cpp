@@ -8,6 +8,8 @@ void classify_keys(torch::Tensor pod_flags, torch::Tensor sel_hash,
                    torch::Tensor key_req, torch::Tensor pending_flag);
 void node_free_subtract(torch::Tensor pod_node, torch::Tensor pod_req,
                         torch::Tensor node_free);
+void node_usage_sum(torch::Tensor pod_node, torch::Tensor pod_req,
+                    torch::Tensor usage);
 void fit_eligibility(torch::Tensor node_labels, torch::Tensor node_label_count,
                      torch::Tensor node_unsched, torch::Tensor seg_sel,
                      torch::Tensor seg_sel_count, torch::Tensor g_dev,
@@ -111,6 +113,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native autoscaler scheduling kernels (gfx950 HIP)";
   m.def("classify_keys", &classify_keys, "kernel C: pod radix keys");
   m.def("node_free_subtract", &node_free_subtract, "node free capacity");
+  m.def("node_usage_sum", &node_usage_sum, "per-node int64 usage totals");
   m.def("fit_eligibility", &fit_eligibility, "kernel A: segment x node eligibility");
   m.def("seg_pool_route", &seg_pool_route, "segment -> pool routing");
   m.def("binpack_place", &binpack_place, "kernel B: FFD placement steps");
