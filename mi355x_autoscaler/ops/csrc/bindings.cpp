@@ -59,6 +59,10 @@ void segment_compact(torch::Tensor slot_ks, torch::Tensor slot_kr,
                      torch::Tensor slot2seg, torch::Tensor g_out);
 void pod_pool_map(torch::Tensor slot_of_pod, torch::Tensor slot2seg,
                   torch::Tensor seg_pool, torch::Tensor pod_pool);
+void rows_hash(torch::Tensor r_hash, torch::Tensor r_key, torch::Tensor r_count,
+               torch::Tensor slot_ks, torch::Tensor slot_kr,
+               torch::Tensor slot_state, torch::Tensor slot_count,
+               torch::Tensor slot_first, torch::Tensor overflow);
 pybind11::dict encode_pods_json(pybind11::bytes body, pybind11::dict node_index,
                                 pybind11::list pdb_selectors);
 void pack_into(pybind11::buffer buf, const std::vector<int64_t>& offsets,
@@ -134,6 +138,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("segment_hash", &segment_hash, "kernel C: hash-aggregate pending pods");
   m.def("segment_compact", &segment_compact, "kernel C: compact+sort segments");
   m.def("pod_pool_map", &pod_pool_map, "pod -> routed pool map");
+  m.def("rows_hash", &rows_hash, "hash-accumulate gathered segment rows");
   m.def("encode_pods_json", &encode_pods_json,
         "PodList JSON -> pod SoA columns (native snapshot codec)");
   m.def("snapshot_scatter", &snapshot_scatter,

@@ -970,6 +970,119 @@ class GpuEngine(SimulationEngine):
         )
 
     # ------------------------------------------------------------------
+    def rows_device(self, segs: DeviceSegments) -> torch.Tensor:
+        """[G, MAXSEL+8] int64 device rows (the sharded wire format of
+        parallel/engine_sharded.py) straight from the segment
+        workspace — no D2H round-trip."""
+        G = segs.G
+        if G == 0:
+            return torch.zeros((0, _MAXSEL + 8), dtype=torch.int64,
+                               device=self.device)
+        return torch.cat([
+            segs.seg_sel_pairs[:G].reshape(G, _MAXSEL),
+            segs.seg_sel_count[:G].to(torch.int64).unsqueeze(1),
+            segs.seg_req[:G].to(torch.int64).reshape(G, 4),
+            segs.seg_count[:G].unsqueeze(1),
+            segs.seg_sel_hash[:G].unsqueeze(1),
+            segs.seg_key[:G].unsqueeze(1),
+        ], dim=1)
+
+    def decide_rows_device(self, ds: DeviceSnapshot, gathered: torch.Tensor,
+                           pool_names: List[str], over_provision: int = 0,
+                           free_hook=None) -> Optional[ScaleDecision]:
+        """Sharded DEVICE merge: hash-accumulate the all-gathered
+        segment rows on device (rows_hash kernel; padded rows have
+        count 0 and are skipped), compact+sort into the global segment
+        order, and run the fused tail — no host-side merge, serialize,
+        or workspace reload. Returns None on hash/sort overflow
+        (deterministic across ranks; every rank falls back together)."""
+        from ..snapshot.encode import MAX_SELECTOR_PAIRS as MS
+
+        ops = self.ops
+        N, K = ds.N, ds.K
+        R = int(gathered.shape[0])
+        if R == 0:
+            d = ScaleDecision()
+            d.pool_residual_totals = np.zeros((K, 4), np.int64)
+            return d
+        self._ws_i32.zero_()
+        pairs = gathered[:, :MS].contiguous()
+        selc = gathered[:, MS].to(torch.int32)
+        req = gathered[:, MS + 1 : MS + 5].to(torch.int32).contiguous()
+        rcnt = gathered[:, MS + 5].contiguous()
+        rhash = gathered[:, MS + 6].contiguous()
+        rkey = gathered[:, MS + 7].contiguous()
+        ops.rows_hash(rhash, rkey, rcnt, self.ws_slot_ks, self.ws_slot_kr,
+                      self.ws_slot_state, self.ws_slot_count,
+                      self.ws_slot_first, self.ws_stat[2:3])
+        ops.segment_compact(self.ws_slot_ks, self.ws_slot_kr, self.ws_slot_state,
+                            self.ws_slot_count, self.ws_slot_first,
+                            pairs, selc, req,
+                            self.ws_seg_sel_pairs, self.ws_seg_sel_count,
+                            self.ws_seg_sel_hash, self.ws_seg_req,
+                            self.ws_seg_count, self.ws_seg_key,
+                            self.ws_seg_first, self.ws_slot2seg,
+                            self.ws_stat[0:2])
+
+        free = self._free_buffer(N)
+        want_free = 1
+        if free_hook is not None:
+            free.copy_(ds.node_alloc)
+            self.ops.node_free_subtract(ds.pod_node, ds.pod_req, free)
+            hooked = free_hook(free, ds)
+            if hooked is not free:
+                free.copy_(hooked)
+            want_free = 0
+        W = (N + 63) // 64
+        elig = self._elig_buffer(W)
+        bufs = self._decide_buffers(1, K, False)
+        blob = bufs["blob"]
+        self._launch_graphed(
+            self.ops.fused_tail_launch,
+            (ds.pod_req, ds.pod_node, ds.node_alloc, ds.node_labels,
+             ds.node_label_count, ds.node_unsched, ds.pool_unit, ds.pool_labels,
+             ds.pool_label_count, ds.pool_max, ds.pool_cur, self.ws_stat,
+             self.ws_seg_sel_pairs, self.ws_seg_sel_count, self.ws_seg_sel_hash,
+             self.ws_seg_req, self.ws_seg_count, free, elig, self.ws_seg_pool,
+             self.ws_run_scratch, blob),
+            (SEG_MAX, want_free),
+        )
+        self._last_demand = None
+
+        bufs["blob_host"].copy_(blob)
+        blob_h = bufs["blob_host"].numpy()
+        if K > 0:
+            stat_h = blob_h[-2:].view(np.int32)
+            G, sort_ovf, hash_ovf = int(stat_h[0]), int(stat_h[1]), int(stat_h[2])
+        else:
+            stat_h = self.ws_stat.cpu().numpy()
+            G, sort_ovf, hash_ovf = int(stat_h[0]), int(stat_h[1]), int(stat_h[2])
+        if sort_ovf or hash_ovf:
+            return None
+        decision = ScaleDecision()
+        if G == 0:
+            decision.pool_residual_totals = np.zeros((K, 4), np.int64)
+            return decision
+        placed_h = blob_h[0:G]
+        unful_h = blob_h[2 * SEG_MAX : 2 * SEG_MAX + G]
+        if K == 0:
+            counts_h = self.ws_seg_count[:G].cpu().numpy()
+            unful_h = counts_h - placed_h
+            sel_hash_fn = lambda: self.ws_seg_sel_hash[:G].cpu().numpy()
+        else:
+            sel_hash_fn = lambda: blob_h[-2 - SEG_MAX : -2]
+        max_new_h = np.maximum(
+            (ds.pool_max_h - ds.pool_cur_h) if ds.pool_max_h is not None
+            else np.zeros(K, np.int64), 0)
+        return self._finish_decision(
+            decision, placed_h, blob_h[SEG_MAX : SEG_MAX + G], unful_h,
+            blob_h[3 * SEG_MAX : 3 * SEG_MAX + K],
+            blob_h[3 * SEG_MAX + K : 3 * SEG_MAX + 5 * K].reshape(max(K, 1) if K else 0, 4) if K else np.zeros((0, 4), np.int64),
+            max_new_h, ds.pool_cur_h, K, pool_names, over_provision,
+            sel_hash_fn,
+        )
+
+    # ------------------------------------------------------------------
     def decide_resident(self, cache, base: ClusterSnapshot,
                         over_provision: int = 0) -> ScaleDecision:
         """Resident-controller decide: the pod SoA stays in HBM at the

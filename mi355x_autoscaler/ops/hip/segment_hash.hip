@@ -205,6 +205,57 @@ extern "C" __global__ void segment_hash_build_small(
                        overflow);
 }
 
+// Sharded merge: hash-accumulate GATHERED SEGMENT ROWS (one row per
+// (selector-set, request) segment per rank, counts pre-aggregated)
+// into the same table the pod build uses — the host-side numpy merge
+// cost ~180 us/rank of serialize + merge + workspace reload at the
+// 8-GPU weak-scaling point. Padded rows (count <= 0) are skipped, so
+// the all-gather needs no size compaction. Same wave-leader discipline;
+// matched lanes' counts reduce in-wave before ONE atomic pair.
+extern "C" __global__ void rows_hash_kernel(
+    const int64_t* __restrict__ r_hash, const int64_t* __restrict__ r_key,
+    const int64_t* __restrict__ r_count, int64_t R, int32_t H,
+    int64_t* __restrict__ slot_ks, int64_t* __restrict__ slot_kr,
+    int32_t* __restrict__ slot_state, int32_t* __restrict__ slot_count,
+    int32_t* __restrict__ slot_first, int32_t* __restrict__ overflow) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; ; i0 += stride) {
+    if (__all(i0 >= R)) break;
+    const bool live = i0 < R && r_count[i0] > 0;
+    const int64_t ks = live ? r_hash[i0] : 0;
+    const int64_t kr = live ? r_key[i0] : 0;
+    const int cnt = live ? (int)r_count[i0] : 0;
+
+    uint64_t pending = __ballot(live);
+    while (pending) {
+      const int leader = __ffsll((unsigned long long)pending) - 1;
+      const int64_t k0 = __shfl(ks, leader, WAVE);
+      const int64_t r0 = __shfl(kr, leader, WAVE);
+      const bool match = live && ks == k0 && kr == r0;
+      const uint64_t mm = __ballot(match);
+      int contrib = match ? cnt : 0;
+#pragma unroll
+      for (int off = WAVE / 2; off; off >>= 1)
+        contrib += __shfl_down(contrib, off, WAVE);
+      const int total = __shfl(contrib, 0, WAVE);
+      if (lane == leader) {
+        const int slot = probe_claim(k0, r0, H, slot_ks, slot_kr, slot_state,
+                                     overflow);
+        if (slot >= 0) {
+          atomicAdd(&slot_count[slot], total);
+          // any matched row is a valid representative (identical
+          // pairs/req for the same key); keep the min index
+          atomicMax(&slot_first[slot], INT32_MAX - (int32_t)i0);
+        } else {
+          atomicOr(overflow, 1);
+        }
+      }
+      pending &= ~mm;
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 
 #define CS_BLOCK 1024
@@ -324,4 +375,22 @@ void pod_pool_map(torch::Tensor slot_of_pod, torch::Tensor slot2seg,
                      stream.stream(), slot_of_pod.data_ptr<int32_t>(),
                      slot2seg.data_ptr<int32_t>(), seg_pool.data_ptr<int32_t>(),
                      P, pod_pool.data_ptr<int32_t>());
+}
+
+void rows_hash(torch::Tensor r_hash, torch::Tensor r_key, torch::Tensor r_count,
+               torch::Tensor slot_ks, torch::Tensor slot_kr,
+               torch::Tensor slot_state, torch::Tensor slot_count,
+               torch::Tensor slot_first, torch::Tensor overflow) {
+  const int64_t R = r_hash.size(0);
+  const int H = (int)slot_ks.size(0);
+  if (R == 0) return;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const int block = 256;
+  const int grid = (int)std::max<int64_t>(1, std::min<int64_t>((R + 1023) / 1024, 256));
+  hipLaunchKernelGGL(rows_hash_kernel, dim3(grid), dim3(block), 0,
+                     stream.stream(), r_hash.data_ptr<int64_t>(),
+                     r_key.data_ptr<int64_t>(), r_count.data_ptr<int64_t>(), R, H,
+                     slot_ks.data_ptr<int64_t>(), slot_kr.data_ptr<int64_t>(),
+                     slot_state.data_ptr<int32_t>(), slot_count.data_ptr<int32_t>(),
+                     slot_first.data_ptr<int32_t>(), overflow.data_ptr<int32_t>());
 }

@@ -75,6 +75,34 @@ def allgather_rows(rows: np.ndarray, device=None) -> np.ndarray:
     return np.concatenate(chunks, axis=0)
 
 
+def backend() -> str:
+    """Active torch.distributed backend name ('' when uninitialized)."""
+    try:
+        return dist.get_backend() if dist.is_initialized() else ""
+    except Exception:
+        return ""
+
+
+def allgather_rows_device(rows: torch.Tensor) -> torch.Tensor:
+    """Device-resident variant of allgather_rows: rows stay on the
+    GPU end to end (one scalar MAX all-reduce for the pad size, one
+    all_gather). Padded rows carry count 0 and are skipped by the
+    rows_hash kernel, so no per-rank size readback is needed."""
+    if world_size() == 1:
+        return rows
+    n = torch.tensor([rows.shape[0]], dtype=torch.int64, device=rows.device)
+    dist.all_reduce(n, op=dist.ReduceOp.MAX)
+    max_n = int(n.item())
+    if max_n == 0:
+        return rows[:0]
+    pad = torch.zeros((max_n, rows.shape[1]), dtype=torch.int64,
+                      device=rows.device)
+    pad[: rows.shape[0]] = rows
+    outs = [torch.zeros_like(pad) for _ in range(world_size())]
+    dist.all_gather(outs, pad)
+    return torch.cat(outs, dim=0)
+
+
 def allreduce_tensor(t: torch.Tensor, op: str = "sum") -> torch.Tensor:
     """In-place all-reduce; returns t. ops: sum | max | min."""
     if world_size() == 1:

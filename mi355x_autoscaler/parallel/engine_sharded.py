@@ -137,11 +137,6 @@ class ShardedEngine(SimulationEngine):
         else:
             ds = self.inner.packer.pack(snap, device)
             segs = self.inner.build_segments(ds)
-            rows = self._device_segments_to_rows(ds, segs)
-            t0 = time.perf_counter()
-            rows = pdist.allgather_rows(rows, device=device)
-            self.comm_stats["allgather_s"] += time.perf_counter() - t0
-            merged = merge_rows(rows)
 
             def free_hook(free, dsnap):
                 if not self.active:
@@ -153,16 +148,42 @@ class ShardedEngine(SimulationEngine):
                 ).to(torch.int32)
 
             hook = free_hook if self.active else None
-            decision = self.inner.decide_merged_rows(
-                ds, merged, snap.pool_names, over_provision=over_provision,
-                free_hook=hook,
+            decision = None
+            # device merge: rows never leave HBM (the host merge cost
+            # ~180 us/rank of serialize + numpy merge + ws reload);
+            # needs device collectives (nccl/RCCL) or world 1
+            device_merge = (
+                hasattr(self.inner, "decide_rows_device")
+                and str(device).startswith("cuda")
+                and (not self.active or pdist.backend() == "nccl")
             )
-            if decision is None:  # merged list exceeds the workspace
-                gsegs = self._rows_to_device_segments(merged, device)
-                decision = self.inner.decide_device(
-                    ds, gsegs, snap.pool_names, over_provision=over_provision,
-                    free_hook=free_hook,
+            if device_merge:
+                rows_dev = self.inner.rows_device(segs)
+                t0 = time.perf_counter()
+                gathered = pdist.allgather_rows_device(rows_dev)
+                self.comm_stats["allgather_s"] += time.perf_counter() - t0
+                decision = self.inner.decide_rows_device(
+                    ds, gathered, snap.pool_names,
+                    over_provision=over_provision, free_hook=hook,
                 )
+                # overflow is a function of the merged global rows, so
+                # every rank observes it together and falls back together
+            if decision is None:
+                rows = self._device_segments_to_rows(ds, segs)
+                t0 = time.perf_counter()
+                rows = pdist.allgather_rows(rows, device=device)
+                self.comm_stats["allgather_s"] += time.perf_counter() - t0
+                merged = merge_rows(rows)
+                decision = self.inner.decide_merged_rows(
+                    ds, merged, snap.pool_names, over_provision=over_provision,
+                    free_hook=hook,
+                )
+                if decision is None:  # merged list exceeds the workspace
+                    gsegs = self._rows_to_device_segments(merged, device)
+                    decision = self.inner.decide_device(
+                        ds, gsegs, snap.pool_names, over_provision=over_provision,
+                        free_hook=free_hook,
+                    )
         if self.active and self.check_agreement:
             t0 = time.perf_counter()
             self._assert_agreement(decision, snap)

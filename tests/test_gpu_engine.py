@@ -485,3 +485,35 @@ def test_wide_selector_decide_matches_oracle(engine):
     pods += [make_pod_json("plain", cpu_milli=100, pending=True)]
     snap = _snap(pods, [full, partial])
     _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))
+
+
+def test_decide_rows_device_matches_host_merge(engine):
+    """Sharded DEVICE merge (rows_hash + compact on device) must equal
+    the host-side numpy merge path on identical gathered rows —
+    including padded (count 0) rows from the size-aligned all-gather
+    and cross-rank duplicates."""
+    import torch
+
+    from mi355x_autoscaler.parallel.engine_sharded import merge_rows
+
+    pods_json, nodes_json = burst_trace(
+        3000, n_nodes_per_pool=(5, 4),
+        pools=(("agentpool1", "Standard_D2_v2"), ("gpupool", "Standard_ND96_MI355X_v1")),
+        gpu_fraction=0.4, seed=7, bound_pods_per_node=2)
+    snap = _snap(pods_json, nodes_json)
+    ds = engine.packer.pack(snap, engine.device)
+    segs = engine.build_segments(ds)
+    rows_dev = engine.rows_device(segs)
+    rows_h = rows_dev.cpu().numpy()
+
+    # simulate a 3-rank gather with per-rank padding
+    pad = torch.zeros((2, rows_dev.shape[1]), dtype=torch.int64,
+                      device=engine.device)
+    gathered = torch.cat([rows_dev, pad, rows_dev, pad, rows_dev, pad], dim=0)
+    d_dev = engine.decide_rows_device(ds, gathered, snap.pool_names)
+
+    merged = merge_rows(np.concatenate([rows_h, rows_h, rows_h], axis=0))
+    d_host = engine.decide_merged_rows(ds, merged, snap.pool_names)
+    _assert_decisions_equal(d_dev, d_host)
+    assert d_dev.pods_placed_existing + d_dev.pods_placed_new + \
+        d_dev.pods_unfulfillable == 3 * 3000
