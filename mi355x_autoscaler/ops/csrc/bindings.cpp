@@ -56,13 +56,15 @@ void segment_compact(torch::Tensor slot_ks, torch::Tensor slot_kr,
                      torch::Tensor seg_sel_hash, torch::Tensor seg_req,
                      torch::Tensor seg_count, torch::Tensor seg_key,
                      torch::Tensor seg_first_pod,
-                     torch::Tensor slot2seg, torch::Tensor g_out);
+                     torch::Tensor slot2seg, torch::Tensor g_out,
+                     c10::optional<torch::Tensor> rows_out);
 void pod_pool_map(torch::Tensor slot_of_pod, torch::Tensor slot2seg,
                   torch::Tensor seg_pool, torch::Tensor pod_pool);
-void rows_hash(torch::Tensor r_hash, torch::Tensor r_key, torch::Tensor r_count,
-               torch::Tensor slot_ks, torch::Tensor slot_kr,
+void rows_hash(torch::Tensor rows, torch::Tensor slot_ks, torch::Tensor slot_kr,
                torch::Tensor slot_state, torch::Tensor slot_count,
                torch::Tensor slot_first, torch::Tensor overflow);
+void rows_unpack(torch::Tensor rows, torch::Tensor pairs, torch::Tensor selc,
+                 torch::Tensor req);
 pybind11::dict encode_pods_json(pybind11::bytes body, pybind11::dict node_index,
                                 pybind11::list pdb_selectors);
 void pack_into(pybind11::buffer buf, const std::vector<int64_t>& offsets,
@@ -136,9 +138,19 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("drain_finalize", &drain_finalize, "kernel D: utilization + candidates");
   m.def("segsum_mfma", &segsum_mfma, "MFMA segmented resource sums");
   m.def("segment_hash", &segment_hash, "kernel C: hash-aggregate pending pods");
-  m.def("segment_compact", &segment_compact, "kernel C: compact+sort segments");
+  m.def("segment_compact", &segment_compact, "kernel C: compact+sort segments",
+        pybind11::arg("slot_ks"), pybind11::arg("slot_kr"),
+        pybind11::arg("slot_state"), pybind11::arg("slot_count"),
+        pybind11::arg("slot_first"), pybind11::arg("pod_sel_pairs"),
+        pybind11::arg("pod_sel_count"), pybind11::arg("pod_req"),
+        pybind11::arg("seg_sel_pairs"), pybind11::arg("seg_sel_count"),
+        pybind11::arg("seg_sel_hash"), pybind11::arg("seg_req"),
+        pybind11::arg("seg_count"), pybind11::arg("seg_key"),
+        pybind11::arg("seg_first_pod"), pybind11::arg("slot2seg"),
+        pybind11::arg("g_out"), pybind11::arg("rows_out") = pybind11::none());
   m.def("pod_pool_map", &pod_pool_map, "pod -> routed pool map");
   m.def("rows_hash", &rows_hash, "hash-accumulate gathered segment rows");
+  m.def("rows_unpack", &rows_unpack, "wire rows -> typed segment columns");
   m.def("encode_pods_json", &encode_pods_json,
         "PodList JSON -> pod SoA columns (native snapshot codec)");
   m.def("snapshot_scatter", &snapshot_scatter,

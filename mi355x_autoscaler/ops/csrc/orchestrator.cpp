@@ -24,7 +24,8 @@ void segment_compact(torch::Tensor slot_ks, torch::Tensor slot_kr,
                      torch::Tensor seg_sel_hash, torch::Tensor seg_req,
                      torch::Tensor seg_count, torch::Tensor seg_key,
                      torch::Tensor seg_first_pod,
-                     torch::Tensor slot2seg, torch::Tensor g_out);
+                     torch::Tensor slot2seg, torch::Tensor g_out,
+                     c10::optional<torch::Tensor> rows_out);
 void node_free_subtract(torch::Tensor pod_node, torch::Tensor pod_req,
                         torch::Tensor node_free);
 void fit_eligibility(torch::Tensor node_labels, torch::Tensor node_label_count,
@@ -140,7 +141,7 @@ void fused_decide_launch(
   segment_compact(slot_ks, slot_kr, slot_state, slot_count, slot_first,
                   pod_sel_pairs, pod_sel_count, pod_req, seg_sel_pairs,
                   seg_sel_count, seg_sel_hash, seg_req, seg_count, seg_key,
-                  seg_first, slot2seg, stat.narrow(0, 0, 2));
+                  seg_first, slot2seg, stat.narrow(0, 0, 2), c10::nullopt);
 
   // free = alloc - bound requests
   hipMemcpyAsync(free.data_ptr<int32_t>(), node_alloc.data_ptr<int32_t>(),

@@ -373,6 +373,9 @@ class GpuEngine(SimulationEngine):
         self.ws_seg_key = torch.zeros(SEG_MAX, dtype=torch.int64, device=d)
         self.ws_seg_first = torch.zeros(SEG_MAX, dtype=torch.int32, device=d)
         self.ws_seg_pool = torch.zeros(SEG_MAX, dtype=torch.int32, device=d)
+        # sharded wire rows emitted by segment_compact (MAXSEL+8 cols)
+        self.ws_seg_rows = torch.zeros((SEG_MAX, _MAXSEL + 8), dtype=torch.int64, device=d)
+        self._rows_ws: Optional[dict] = None  # gathered-row unpack buffers
         self.ws_run_scratch = torch.empty((4 * SEG_MAX + 32) * 6, dtype=torch.int64, device=d)
         self._elig_cache: Dict[int, torch.Tensor] = {}  # W -> [SEG_MAX, W]
         # hipGraph capture of the fused pipeline (OPT-IN, measured
@@ -471,6 +474,7 @@ class GpuEngine(SimulationEngine):
             self.ws_seg_sel_pairs, self.ws_seg_sel_count, self.ws_seg_sel_hash,
             self.ws_seg_req, self.ws_seg_count, self.ws_seg_key,
             self.ws_seg_first, self.ws_slot2seg, self.ws_stat[0:2],
+            self.ws_seg_rows,
         )
         stat = self.ws_stat.cpu().numpy()  # sync: G needed for host shapes
         G, sort_ovf, hash_ovf = int(stat[0]), int(stat[1]), int(stat[2])
@@ -972,20 +976,10 @@ class GpuEngine(SimulationEngine):
     # ------------------------------------------------------------------
     def rows_device(self, segs: DeviceSegments) -> torch.Tensor:
         """[G, MAXSEL+8] int64 device rows (the sharded wire format of
-        parallel/engine_sharded.py) straight from the segment
-        workspace — no D2H round-trip."""
-        G = segs.G
-        if G == 0:
-            return torch.zeros((0, _MAXSEL + 8), dtype=torch.int64,
-                               device=self.device)
-        return torch.cat([
-            segs.seg_sel_pairs[:G].reshape(G, _MAXSEL),
-            segs.seg_sel_count[:G].to(torch.int64).unsqueeze(1),
-            segs.seg_req[:G].to(torch.int64).reshape(G, 4),
-            segs.seg_count[:G].unsqueeze(1),
-            segs.seg_sel_hash[:G].unsqueeze(1),
-            segs.seg_key[:G].unsqueeze(1),
-        ], dim=1)
+        parallel/engine_sharded.py): emitted by segment_compact into
+        ws_seg_rows during build_segments_hash — zero extra dispatches,
+        no D2H round-trip."""
+        return self.ws_seg_rows[: segs.G]
 
     def decide_rows_device(self, ds: DeviceSnapshot, gathered: torch.Tensor,
                            pool_names: List[str], over_provision: int = 0,
@@ -1006,13 +1000,20 @@ class GpuEngine(SimulationEngine):
             d.pool_residual_totals = np.zeros((K, 4), np.int64)
             return d
         self._ws_i32.zero_()
-        pairs = gathered[:, :MS].contiguous()
-        selc = gathered[:, MS].to(torch.int32)
-        req = gathered[:, MS + 1 : MS + 5].to(torch.int32).contiguous()
-        rcnt = gathered[:, MS + 5].contiguous()
-        rhash = gathered[:, MS + 6].contiguous()
-        rkey = gathered[:, MS + 7].contiguous()
-        ops.rows_hash(rhash, rkey, rcnt, self.ws_slot_ks, self.ws_slot_kr,
+        rw = self._rows_ws
+        if rw is None or rw["pairs"].shape[0] < R:
+            cap = max(R, 4 * SEG_MAX)
+            rw = self._rows_ws = {
+                "pairs": torch.empty((cap, MS), dtype=torch.int64, device=self.device),
+                "selc": torch.empty(cap, dtype=torch.int32, device=self.device),
+                "req": torch.empty((cap, 4), dtype=torch.int32, device=self.device),
+            }
+        pairs = rw["pairs"][:R]
+        selc = rw["selc"][:R]
+        req = rw["req"][:R]
+        gathered = gathered.contiguous()
+        ops.rows_unpack(gathered, pairs, selc, req)
+        ops.rows_hash(gathered, self.ws_slot_ks, self.ws_slot_kr,
                       self.ws_slot_state, self.ws_slot_count,
                       self.ws_slot_first, self.ws_stat[2:3])
         ops.segment_compact(self.ws_slot_ks, self.ws_slot_kr, self.ws_slot_state,
@@ -1022,7 +1023,7 @@ class GpuEngine(SimulationEngine):
                             self.ws_seg_sel_hash, self.ws_seg_req,
                             self.ws_seg_count, self.ws_seg_key,
                             self.ws_seg_first, self.ws_slot2seg,
-                            self.ws_stat[0:2])
+                            self.ws_stat[0:2], None)
 
         free = self._free_buffer(N)
         want_free = 1

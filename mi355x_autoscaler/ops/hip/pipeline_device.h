@@ -151,7 +151,8 @@ __device__ inline void dev_compact_sort(
     int64_t* __restrict__ seg_count, int64_t* __restrict__ seg_key,
     int32_t* __restrict__ seg_first_pod, int32_t* __restrict__ slot2seg,
     int32_t* __restrict__ g_out,
-    uint64_t* lds_ks, uint64_t* lds_kr, int32_t* lds_slot, int32_t* lds_n) {
+    uint64_t* lds_ks, uint64_t* lds_kr, int32_t* lds_slot, int32_t* lds_n,
+    int64_t* __restrict__ rows_out = nullptr) {  // [SEG_MAX, MAXSEL+8] wire rows
   const int tid = threadIdx.x;
   const int B = blockDim.x;
   if (tid == 0) *lds_n = 0;
@@ -228,6 +229,19 @@ __device__ inline void dev_compact_sort(
 #pragma unroll
     for (int r = 0; r < NRES; ++r)
       seg_req[g * NRES + r] = pod_req[(int64_t)fp * NRES + r];
+    if (rows_out != nullptr) {
+      // sharded wire format (engine_sharded._ROW_COLS): emitted here so
+      // the all-gather source needs no torch-op assembly
+      int64_t* row = rows_out + (int64_t)g * (MAXSEL + 8);
+#pragma unroll
+      for (int q = 0; q < MAXSEL; ++q) row[q] = pod_sel_pairs[(int64_t)fp * MAXSEL + q];
+      row[MAXSEL] = pod_sel_count[fp];
+#pragma unroll
+      for (int r = 0; r < NRES; ++r) row[MAXSEL + 1 + r] = pod_req[(int64_t)fp * NRES + r];
+      row[MAXSEL + 5] = (int64_t)slot_count[h];
+      row[MAXSEL + 6] = seg_sel_hash[g];
+      row[MAXSEL + 7] = seg_key[g];
+    }
   }
   if (tid == 0) {
     g_out[0] = G;

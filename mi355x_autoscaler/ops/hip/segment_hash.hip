@@ -213,19 +213,21 @@ extern "C" __global__ void segment_hash_build_small(
 // the all-gather needs no size compaction. Same wave-leader discipline;
 // matched lanes' counts reduce in-wave before ONE atomic pair.
 extern "C" __global__ void rows_hash_kernel(
-    const int64_t* __restrict__ r_hash, const int64_t* __restrict__ r_key,
-    const int64_t* __restrict__ r_count, int64_t R, int32_t H,
+    const int64_t* __restrict__ rows,    // [R, MAXSEL+8] gathered wire rows
+    int64_t R, int32_t H,
     int64_t* __restrict__ slot_ks, int64_t* __restrict__ slot_kr,
     int32_t* __restrict__ slot_state, int32_t* __restrict__ slot_count,
     int32_t* __restrict__ slot_first, int32_t* __restrict__ overflow) {
+  const int C = MAXSEL + 8;
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; ; i0 += stride) {
     if (__all(i0 >= R)) break;
-    const bool live = i0 < R && r_count[i0] > 0;
-    const int64_t ks = live ? r_hash[i0] : 0;
-    const int64_t kr = live ? r_key[i0] : 0;
-    const int cnt = live ? (int)r_count[i0] : 0;
+    const int64_t* row = rows + i0 * C;
+    const bool live = i0 < R && row[MAXSEL + 5] > 0;
+    const int64_t ks = live ? row[MAXSEL + 6] : 0;
+    const int64_t kr = live ? row[MAXSEL + 7] : 0;
+    const int cnt = live ? (int)row[MAXSEL + 5] : 0;
 
     uint64_t pending = __ballot(live);
     while (pending) {
@@ -274,7 +276,7 @@ extern "C" __global__ __launch_bounds__(CS_BLOCK) void segment_compact_sort(
     int64_t* __restrict__ seg_key, int32_t* __restrict__ seg_first_pod,
     int32_t* __restrict__ slot2seg,       // [H]
     int32_t* __restrict__ g_out,          // [2]: G, overflow(G>SEG_MAX)
-    int32_t sort_dir_check)               // unused; keeps ABI room
+    int64_t* __restrict__ rows_out)       // optional [SEG_MAX, MAXSEL+8]
 {
   __shared__ uint64_t lds_ks[SEG_MAX];
   __shared__ uint64_t lds_kr[SEG_MAX];
@@ -284,7 +286,7 @@ extern "C" __global__ __launch_bounds__(CS_BLOCK) void segment_compact_sort(
                    pod_sel_pairs, pod_sel_count, pod_req, seg_sel_pairs,
                    seg_sel_count, seg_sel_hash, seg_req, seg_count, seg_key,
                    seg_first_pod, slot2seg, g_out, lds_ks, lds_kr, lds_slot,
-                   &lds_n);
+                   &lds_n, rows_out);
 }
 
 // pod -> routed-pool map (for the MFMA demand segsum): pool id of the
@@ -347,9 +349,11 @@ void segment_compact(torch::Tensor slot_ks, torch::Tensor slot_kr,
                      torch::Tensor seg_sel_hash, torch::Tensor seg_req,
                      torch::Tensor seg_count, torch::Tensor seg_key,
                      torch::Tensor seg_first_pod,
-                     torch::Tensor slot2seg, torch::Tensor g_out) {
+                     torch::Tensor slot2seg, torch::Tensor g_out,
+                     c10::optional<torch::Tensor> rows_out) {
   const int H = (int)slot_ks.size(0);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  int64_t* rows_ptr = rows_out.has_value() ? rows_out->data_ptr<int64_t>() : nullptr;
   hipLaunchKernelGGL(segment_compact_sort, dim3(1), dim3(CS_BLOCK), 0,
                      stream.stream(), slot_ks.data_ptr<int64_t>(),
                      slot_kr.data_ptr<int64_t>(), slot_state.data_ptr<int32_t>(),
@@ -361,7 +365,8 @@ void segment_compact(torch::Tensor slot_ks, torch::Tensor slot_kr,
                      seg_sel_hash.data_ptr<int64_t>(), seg_req.data_ptr<int32_t>(),
                      seg_count.data_ptr<int64_t>(), seg_key.data_ptr<int64_t>(),
                      seg_first_pod.data_ptr<int32_t>(),
-                     slot2seg.data_ptr<int32_t>(), g_out.data_ptr<int32_t>(), 0);
+                     slot2seg.data_ptr<int32_t>(), g_out.data_ptr<int32_t>(),
+                     rows_ptr);
 }
 
 void pod_pool_map(torch::Tensor slot_of_pod, torch::Tensor slot2seg,
@@ -377,20 +382,50 @@ void pod_pool_map(torch::Tensor slot_of_pod, torch::Tensor slot2seg,
                      P, pod_pool.data_ptr<int32_t>());
 }
 
-void rows_hash(torch::Tensor r_hash, torch::Tensor r_key, torch::Tensor r_count,
-               torch::Tensor slot_ks, torch::Tensor slot_kr,
+void rows_hash(torch::Tensor rows, torch::Tensor slot_ks, torch::Tensor slot_kr,
                torch::Tensor slot_state, torch::Tensor slot_count,
                torch::Tensor slot_first, torch::Tensor overflow) {
-  const int64_t R = r_hash.size(0);
+  const int64_t R = rows.size(0);
   const int H = (int)slot_ks.size(0);
   if (R == 0) return;
+  TORCH_CHECK(rows.size(1) == MAXSEL + 8, "rows must be [R, MAXSEL+8]");
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const int block = 256;
   const int grid = (int)std::max<int64_t>(1, std::min<int64_t>((R + 1023) / 1024, 256));
   hipLaunchKernelGGL(rows_hash_kernel, dim3(grid), dim3(block), 0,
-                     stream.stream(), r_hash.data_ptr<int64_t>(),
-                     r_key.data_ptr<int64_t>(), r_count.data_ptr<int64_t>(), R, H,
+                     stream.stream(), rows.data_ptr<int64_t>(), R, H,
                      slot_ks.data_ptr<int64_t>(), slot_kr.data_ptr<int64_t>(),
                      slot_state.data_ptr<int32_t>(), slot_count.data_ptr<int32_t>(),
                      slot_first.data_ptr<int32_t>(), overflow.data_ptr<int32_t>());
+}
+
+// gathered wire rows -> the typed "pod" columns segment_compact reads
+// (ONE kernel instead of ~8 torch slice/cast dispatches per step)
+extern "C" __global__ void rows_unpack_kernel(
+    const int64_t* __restrict__ rows, int64_t R,
+    int64_t* __restrict__ pairs, int32_t* __restrict__ selc,
+    int32_t* __restrict__ req) {
+  const int C = MAXSEL + 8;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < R;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t* row = rows + i * C;
+#pragma unroll
+    for (int q = 0; q < MAXSEL; ++q) pairs[i * MAXSEL + q] = row[q];
+    selc[i] = (int32_t)row[MAXSEL];
+#pragma unroll
+    for (int r = 0; r < NRES; ++r) req[i * NRES + r] = (int32_t)row[MAXSEL + 1 + r];
+  }
+}
+
+void rows_unpack(torch::Tensor rows, torch::Tensor pairs, torch::Tensor selc,
+                 torch::Tensor req) {
+  const int64_t R = rows.size(0);
+  if (R == 0) return;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const int block = 256;
+  const int grid = (int)std::max<int64_t>(1, std::min<int64_t>((R + 255) / 256, 1024));
+  hipLaunchKernelGGL(rows_unpack_kernel, dim3(grid), dim3(block), 0,
+                     stream.stream(), rows.data_ptr<int64_t>(), R,
+                     pairs.data_ptr<int64_t>(), selc.data_ptr<int32_t>(),
+                     req.data_ptr<int32_t>());
 }
