@@ -373,8 +373,10 @@ class GpuEngine(SimulationEngine):
         self.ws_seg_key = torch.zeros(SEG_MAX, dtype=torch.int64, device=d)
         self.ws_seg_first = torch.zeros(SEG_MAX, dtype=torch.int32, device=d)
         self.ws_seg_pool = torch.zeros(SEG_MAX, dtype=torch.int32, device=d)
-        # sharded wire rows emitted by segment_compact (MAXSEL+8 cols)
-        self.ws_seg_rows = torch.zeros((SEG_MAX, _MAXSEL + 8), dtype=torch.int64, device=d)
+        # sharded wire rows emitted by segment_compact (MAXSEL+8 cols);
+        # row SEG_MAX is the poison slot (negative count = local hash
+        # overflow, carried inside the fixed-size gather)
+        self.ws_seg_rows = torch.zeros((SEG_MAX + 1, _MAXSEL + 8), dtype=torch.int64, device=d)
         self._rows_ws: Optional[dict] = None  # gathered-row unpack buffers
         self.ws_run_scratch = torch.empty((4 * SEG_MAX + 32) * 6, dtype=torch.int64, device=d)
         self._elig_cache: Dict[int, torch.Tensor] = {}  # W -> [SEG_MAX, W]
@@ -492,9 +494,14 @@ class GpuEngine(SimulationEngine):
         )
         return segs
 
-    def _enqueue_hash_segments(self, ds: DeviceSnapshot) -> torch.Tensor:
+    def _enqueue_hash_segments(self, ds: DeviceSnapshot,
+                               emit_rows: bool = False) -> torch.Tensor:
         """Hash + compact WITHOUT any host sync; returns slot_of_pod.
-        G lives in ws_stat[0] on device; overflow flags in ws_stat[1:]."""
+        G lives in ws_stat[0] on device; overflow flags in ws_stat[1:].
+        With emit_rows the compact also writes the sharded wire rows
+        (stale rows zero-counted) and the poison slot carries the local
+        hash-overflow flag — the whole sharded gather source with no
+        host-visible G."""
         ops = self.ops
         self._ws_i32.zero_()
         slot_of_pod = torch.empty(ds.P, dtype=torch.int32, device=self.device)
@@ -510,7 +517,13 @@ class GpuEngine(SimulationEngine):
             self.ws_seg_sel_pairs, self.ws_seg_sel_count, self.ws_seg_sel_hash,
             self.ws_seg_req, self.ws_seg_count, self.ws_seg_key,
             self.ws_seg_first, self.ws_slot2seg, self.ws_stat[0:2],
+            self.ws_seg_rows if emit_rows else None,
         )
+        if emit_rows:
+            from ..snapshot.encode import MAX_SELECTOR_PAIRS as MS
+
+            self.ws_seg_rows[SEG_MAX, MS + 5].copy_(
+                self.ws_stat[2].to(torch.int64).neg())
         return slot_of_pod
 
     def _elig_buffer(self, W: int) -> torch.Tensor:

@@ -243,6 +243,12 @@ __device__ inline void dev_compact_sort(
       row[MAXSEL + 7] = seg_key[g];
     }
   }
+  if (rows_out != nullptr) {
+    // stale rows from earlier cycles must read as padding (count 0) so
+    // a FIXED-size gather needs no host-visible G
+    for (int g = G + tid; g < SEG_MAX; g += B)
+      rows_out[(int64_t)g * (MAXSEL + 8) + MAXSEL + 5] = 0;
+  }
   if (tid == 0) {
     g_out[0] = G;
     g_out[1] = 0;

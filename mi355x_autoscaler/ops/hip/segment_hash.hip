@@ -224,6 +224,12 @@ extern "C" __global__ void rows_hash_kernel(
   for (int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; ; i0 += stride) {
     if (__all(i0 >= R)) break;
     const int64_t* row = rows + i0 * C;
+    if (i0 < R && row[MAXSEL + 5] < 0) {
+      // poison row: some rank's LOCAL hash build overflowed; the
+      // gather is symmetric so every rank sees it and falls back
+      // together (no extra collective)
+      atomicOr(overflow, 4);
+    }
     const bool live = i0 < R && row[MAXSEL + 5] > 0;
     const int64_t ks = live ? row[MAXSEL + 6] : 0;
     const int64_t kr = live ? row[MAXSEL + 7] : 0;

@@ -103,6 +103,17 @@ def allgather_rows_device(rows: torch.Tensor) -> torch.Tensor:
     return torch.cat(outs, dim=0)
 
 
+def allgather_fixed_device(t: torch.Tensor) -> torch.Tensor:
+    """All-gather equal-shaped device tensors — ZERO host syncs (no
+    size exchange; padding carries count 0, overflow rides the poison
+    row). Returns the rank-order concatenation."""
+    if world_size() == 1:
+        return t
+    outs = [torch.zeros_like(t) for _ in range(world_size())]
+    dist.all_gather(outs, t.contiguous())
+    return torch.cat(outs, dim=0)
+
+
 def allreduce_tensor(t: torch.Tensor, op: str = "sum") -> torch.Tensor:
     """In-place all-reduce; returns t. ops: sum | max | min."""
     if world_size() == 1:

@@ -136,7 +136,7 @@ class ShardedEngine(SimulationEngine):
             decision = sim.fulfill_pending(segments, node_free=node_free)
         else:
             ds = self.inner.packer.pack(snap, device)
-            segs = self.inner.build_segments(ds)
+            segs = None
 
             def free_hook(free, dsnap):
                 if not self.active:
@@ -158,17 +158,21 @@ class ShardedEngine(SimulationEngine):
                 and (not self.active or pdist.backend() == "nccl")
             )
             if device_merge:
-                rows_dev = self.inner.rows_device(segs)
+                # no-sync local build: G stays on device, the FIXED-size
+                # gather needs no size exchange, and a rank's local hash
+                # overflow rides the poison row so every rank falls back
+                # together
+                self.inner._enqueue_hash_segments(ds, emit_rows=True)
                 t0 = time.perf_counter()
-                gathered = pdist.allgather_rows_device(rows_dev)
+                gathered = pdist.allgather_fixed_device(self.inner.ws_seg_rows)
                 self.comm_stats["allgather_s"] += time.perf_counter() - t0
                 decision = self.inner.decide_rows_device(
                     ds, gathered, snap.pool_names,
                     over_provision=over_provision, free_hook=hook,
                 )
-                # overflow is a function of the merged global rows, so
-                # every rank observes it together and falls back together
             if decision is None:
+                if segs is None:
+                    segs = self.inner.build_segments(ds)
                 rows = self._device_segments_to_rows(ds, segs)
                 t0 = time.perf_counter()
                 rows = pdist.allgather_rows(rows, device=device)
