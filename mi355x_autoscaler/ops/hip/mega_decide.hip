@@ -75,12 +75,25 @@ extern "C" __global__ __launch_bounds__(MEGA_BLOCK) void mega_decide_kernel(
   for (int64_t i = gtid; i < (int64_t)N * NRES; i += gsz) free_[i] = node_alloc[i];
   grid.sync();
 
-  // P1: hash-aggregate + usage subtraction (both grid-strided over pods)
-  dev_hash_build_small(pod_flags, pod_sel_hash, pod_req, P, H, slot_ks,
-                       slot_kr, slot_state, slot_count, slot_first,
-                       slot_of_pod, stat + 2);
-  dev_node_usage_subtract(pod_node, pod_req, P, free_);
-  grid.sync();
+  // P1: hash-aggregate + usage subtraction. The hash serializes on
+  // per-key atomics, so only ~1 block per 1024 pods participates
+  // (contention scales with wave count, the scan is bandwidth-trivial;
+  // same rule as the standalone launcher); the full grid handles the
+  // usage subtraction.
+  {
+    int hb = (int)((P + 1023) / 1024);
+    if (hb < 1) hb = 1;
+    if (hb > (int)gridDim.x) hb = gridDim.x;
+    if ((int)blockIdx.x < hb) {
+      dev_hash_build_small_ex(pod_flags, pod_sel_hash, pod_req, P, H, slot_ks,
+                              slot_kr, slot_state, slot_count, slot_first,
+                              slot_of_pod, stat + 2,
+                              (int64_t)blockIdx.x * blockDim.x + threadIdx.x,
+                              (int64_t)hb * blockDim.x);
+    }
+    dev_node_usage_subtract(pod_node, pod_req, P, free_);
+    grid.sync();
+  }
 
   // P2: compact + sort (block 0; LDS carve: 40 KiB)
   if (blockIdx.x == 0) {

@@ -86,16 +86,15 @@ __device__ inline int probe_claim(int64_t ks, int64_t kr, int32_t H,
 // Wave-aggregated direct-global hash build (the small-P form): one
 // probe + one count/first atomic pair per distinct key per wave.
 // Grid-strided over pods; safe at any grid shape.
-__device__ inline void dev_hash_build_small(
+__device__ inline void dev_hash_build_small_ex(
     const int32_t* __restrict__ pod_flags, const int64_t* __restrict__ sel_hash,
     const int32_t* __restrict__ pod_req, int64_t P, int32_t H,
     int64_t* __restrict__ slot_ks, int64_t* __restrict__ slot_kr,
     int32_t* __restrict__ slot_state, int32_t* __restrict__ slot_count,
     int32_t* __restrict__ slot_first, int32_t* __restrict__ slot_of_pod,
-    int32_t* __restrict__ overflow) {
+    int32_t* __restrict__ overflow, int64_t start, int64_t stride) {
   const int lane = threadIdx.x & (WAVE - 1);
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; ; i0 += stride) {
+  for (int64_t i0 = start; ; i0 += stride) {
     if (__all(i0 >= P)) break;
     const int pf = i0 < P ? pod_flags[i0] : 0;
     const bool live = i0 < P && (pf & PF_PENDING);
@@ -134,6 +133,20 @@ __device__ inline void dev_hash_build_small(
     }
     if (i0 < P && live) slot_of_pod[i0] = my_slot;
   }
+}
+
+__device__ inline void dev_hash_build_small(
+    const int32_t* __restrict__ pod_flags, const int64_t* __restrict__ sel_hash,
+    const int32_t* __restrict__ pod_req, int64_t P, int32_t H,
+    int64_t* __restrict__ slot_ks, int64_t* __restrict__ slot_kr,
+    int32_t* __restrict__ slot_state, int32_t* __restrict__ slot_count,
+    int32_t* __restrict__ slot_first, int32_t* __restrict__ slot_of_pod,
+    int32_t* __restrict__ overflow) {
+  dev_hash_build_small_ex(pod_flags, sel_hash, pod_req, P, H, slot_ks, slot_kr,
+                          slot_state, slot_count, slot_first, slot_of_pod,
+                          overflow,
+                          (int64_t)blockIdx.x * blockDim.x + threadIdx.x,
+                          (int64_t)gridDim.x * blockDim.x);
 }
 
 // Single-block compact + bitonic sort into the oracle's global segment
