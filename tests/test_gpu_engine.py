@@ -517,3 +517,28 @@ def test_decide_rows_device_matches_host_merge(engine):
     _assert_decisions_equal(d_dev, d_host)
     assert d_dev.pods_placed_existing + d_dev.pods_placed_new + \
         d_dev.pods_unfulfillable == 3 * 3000
+
+
+def test_decide_rows_device_poison_falls_back(engine):
+    """A poison row (negative count = some rank's local hash overflow)
+    must make decide_rows_device return None — the sharded engine then
+    falls back on every rank together."""
+    import torch
+
+    from mi355x_autoscaler.snapshot.encode import MAX_SELECTOR_PAIRS as MS
+
+    pods_json, nodes_json = burst_trace(500, n_nodes_per_pool=(3,), seed=3,
+                                        bound_pods_per_node=2)
+    snap = _snap(pods_json, nodes_json)
+    ds = engine.packer.pack(snap, engine.device)
+    segs = engine.build_segments(ds)
+    rows = engine.rows_device(segs).clone()
+    poison = torch.zeros((1, rows.shape[1]), dtype=torch.int64,
+                         device=engine.device)
+    poison[0, MS + 5] = -1
+    gathered = torch.cat([rows, poison], dim=0)
+    assert engine.decide_rows_device(ds, gathered, snap.pool_names) is None
+    # without the poison the same rows decide normally
+    d = engine.decide_rows_device(ds, rows, snap.pool_names)
+    assert d is not None
+    _assert_decisions_equal(d, CpuEngine().decide(snap))
