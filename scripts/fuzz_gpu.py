@@ -23,7 +23,7 @@ from mi355x_autoscaler.sim.engine import CpuEngine
 from mi355x_autoscaler.sim.synth import burst_trace, wrap
 from mi355x_autoscaler.snapshot.encode import encode_snapshot
 
-ITERS = int(sys.argv[1]) if len(sys.argv) > 1 else 60
+ITERS = 60
 
 POOL_MENU = [
     ("agentpool1", "Standard_D2_v2"),
@@ -41,11 +41,12 @@ def check_equal(d_gpu, d_cpu, ctx):
     np.testing.assert_array_equal(d_gpu.pool_residual_totals, d_cpu.pool_residual_totals)
 
 
-def main():
+def main(iters=None):
     eng = GpuEngine("cuda:0")
     cpu = CpuEngine()
     rng = random.Random(12345)
-    for it in range(ITERS):
+    n_iters = ITERS if iters is None else iters
+    for it in range(n_iters):
         npools = rng.randint(1, 4)
         pools = tuple(POOL_MENU[:npools])
         sizes = tuple(rng.randint(1, 12) for _ in range(npools))
@@ -89,8 +90,8 @@ def main():
             d_sh = eng.decide_rows_device(ds_full, gathered, snap.pool_names)
             assert d_sh is not None, ctx
             check_equal(d_sh, ref, ctx + f" sharded W={W}")
-    print(f"fuzz ok: {ITERS} iterations (incl. simulated sharded merges)")
+    print(f"fuzz ok: {n_iters} iterations (incl. simulated sharded merges)")
 
 
 if __name__ == "__main__":
-    main()
+    main(int(sys.argv[1]) if len(sys.argv) > 1 else None)

@@ -542,3 +542,20 @@ def test_decide_rows_device_poison_falls_back(engine):
     d = engine.decide_rows_device(ds, rows, snap.pool_names)
     assert d is not None
     _assert_decisions_equal(d, CpuEngine().decide(snap))
+
+
+def test_randomized_differential_fuzz(engine):
+    """Randomized shapes (pools, sizes, selector widths, criticals,
+    zero-requests) + simulated multi-rank sharded merges, all asserted
+    bit-equal to the CPU oracle (scripts/fuzz_gpu.py runs the longer
+    form)."""
+    import importlib.util
+    import os
+
+    spec = importlib.util.spec_from_file_location(
+        "fuzz_gpu", os.path.join(os.path.dirname(__file__), "..",
+                                 "scripts", "fuzz_gpu.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    mod.ITERS = 12
+    mod.main()
