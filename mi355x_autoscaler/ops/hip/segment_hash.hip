@@ -79,13 +79,43 @@ extern "C" __global__ __launch_bounds__(256) void segment_hash_build(
       slot_of_pod[i0] = -1;
     }
 
-    uint64_t pending_lanes = __ballot(live);
+    // fast path: per-lane DIRECT probe of already-claimed LDS slots —
+    // reads are deadlock-free (only claims need the leader-at-a-time
+    // discipline), and after the first occurrence of each key per
+    // block every pod hits here, skipping the serialized ballot loop
+    // (the per-wave distinct-key dance was the bulk of the 137 us
+    // hash at 1M pods)
     int my_enc = -1;
+    if (live) {
+      uint32_t h = mix_hash(ks, kr) & (LDS_SLOTS - 1);
+      for (int probe = 0; probe < 8; ++probe) {
+        const int st = __hip_atomic_load(&l_state[h], __ATOMIC_RELAXED,
+                                         __HIP_MEMORY_SCOPE_WORKGROUP);
+        if (st == ST_EMPTY) break;  // key not claimed yet: slow path
+        if (st == ST_READY) {
+          __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "workgroup");
+          if (l_ks[h] == ks && l_kr[h] == kr) {
+            my_enc = (int)h;
+            break;
+          }
+        }
+        // CLAIMED or mismatch: next slot (never spin — claimer may be
+        // a lane of this very wave)
+        h = (h + 1u) & (LDS_SLOTS - 1);
+      }
+      if (my_enc >= 0) {
+        __hip_atomic_fetch_add(&l_count[my_enc], 1, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_WORKGROUP);
+        atomicMax(&l_first[my_enc], INT32_MAX - (int32_t)i0);
+      }
+    }
+
+    uint64_t pending_lanes = __ballot(live && my_enc < 0);
     while (pending_lanes) {
       const int leader = __ffsll((unsigned long long)pending_lanes) - 1;
       const int64_t k0 = __shfl(ks, leader, WAVE);
       const int64_t r0 = __shfl(kr, leader, WAVE);
-      const bool match = live && ks == k0 && kr == r0;
+      const bool match = live && my_enc < 0 && ks == k0 && kr == r0;
       const uint64_t match_mask = __ballot(match);
       int enc = -1;
       if (lane == leader) {
