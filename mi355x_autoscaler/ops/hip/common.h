@@ -66,7 +66,9 @@ __device__ inline int64_t per_node_capacity(const int32_t f[NRES],
   for (int r = 0; r < NRES; ++r) {
     if (f[r] < 0) return 0;
     if (rq[r] > 0) {
-      int64_t q = (int64_t)f[r] / (int64_t)rq[r];
+      // both operands proven non-negative: 32-bit unsigned divide
+      // (the i64 form was the hot scalar op of the single-CU FFD walk)
+      int64_t q = (int64_t)((uint32_t)f[r] / (uint32_t)rq[r]);
       m = q < m ? q : m;
     }
   }
