@@ -162,7 +162,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "watch batch -> target rows (C-speed uid/row bookkeeping)");
   m.def("compiled_constants", [] {
     pybind11::dict d;
-    d["MAXSEL"] = 16;   // must match ops/hip/common.h and encode.py
+    d["MAXSEL"] = 32;   // must match ops/hip/common.h and encode.py
     d["MAXLAB"] = 32;
     d["SEG_MAX"] = 2048;
     return d;

@@ -37,7 +37,9 @@ namespace {
 
 // ---- lattice constants (must match snapshot/encode.py) --------------------
 constexpr int NRES = 4;
-constexpr int MAXSEL = 16;
+constexpr int MAXSEL = 32;
+// int64 view of uint64 max — mirrors encode.py::SELECTOR_POISON
+constexpr int64_t SELECTOR_POISON = -1;
 constexpr int64_t MIB = 1024 * 1024;
 
 constexpr uint16_t PF_PENDING = 1 << 0;
@@ -500,6 +502,8 @@ void encode_one(const JValue* pod, const Ctx& ctx, Columns& out) {
   int cnt = (int)hashes.size();
   for (int q = 0; q < MAXSEL; ++q)
     out.pod_sel_pairs.push_back(q < cnt ? (int64_t)hashes[q] : 0);
+  if (cnt > MAXSEL)  // wider than the lattice: poison -> matches nothing
+    out.pod_sel_pairs[out.pod_sel_pairs.size() - 1] = SELECTOR_POISON;
   out.pod_sel_count.push_back((int32_t)std::min(cnt, MAXSEL));
   out.pod_sel_hash.push_back((int64_t)combined);
 

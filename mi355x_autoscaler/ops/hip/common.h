@@ -10,7 +10,10 @@
 #include <stdint.h>
 
 #define NRES 4            // cpu_milli, mem_mib, pods, gpu — fixed row order
-#define MAXSEL 16         // padded selector pair-hash width (pods)
+#define MAXSEL 32         // padded selector pair-hash width (pods); ==
+                          // MAXLAB so subset matching is exact for every
+                          // representable selector (>32 pairs: poisoned
+                          // with 0xFFFF…F, matches nothing — encode.py)
 #define MAXLAB 32         // padded label pair-hash width (nodes/pools)
 #define WAVE 64
 

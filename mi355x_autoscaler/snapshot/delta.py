@@ -79,7 +79,8 @@ class DeltaSnapshotCache:
         from .encode import _hash_pairs, quantize_request
 
         req = np.array(quantize_request(pod.resources), np.int32)
-        pairs, cnt, combined = _hash_pairs(pod.selectors, MAX_SELECTOR_PAIRS)
+        pairs, cnt, combined = _hash_pairs(
+            pod.selectors, MAX_SELECTOR_PAIRS, poison_overflow=True)
         f = 0
         if pod.is_pending_unschedulable:
             f |= PodFlags.PENDING

@@ -39,12 +39,21 @@ from ..models.resources import NUM_RESOURCES, ResourceVec
 
 MIB = 1024 * 1024
 
-# Padded-width limits for hashed label/selector sets. Pods with more
-# selector pairs than MAX_SELECTOR_PAIRS or nodes with more labels than
-# MAX_LABEL_PAIRS fall back to exact host-side matching (rare; counted
-# in the encode stats).
-MAX_SELECTOR_PAIRS = 16
+# Padded-width limits for hashed label/selector sets. With
+# MAX_SELECTOR_PAIRS == MAX_LABEL_PAIRS, selector matching is EXACT for
+# every selector the node-label lattice can represent: a selector wider
+# than MAX_LABEL_PAIRS can never be a subset of a representable label
+# set, so such pods get a poison pair (below) and match nothing —
+# flagged SELECTOR_OVERFLOW. Nodes with more labels than MAX_LABEL_PAIRS
+# match on the first MAX_LABEL_PAIRS sorted pairs (loud warning).
+MAX_SELECTOR_PAIRS = 32
 MAX_LABEL_PAIRS = 32
+
+# uint64 0xFFFF…F as int64: sorts LAST in the unsigned pair order, and no
+# real FNV-1a label hash equals it in practice, so a selector carrying it
+# fails every subset walk (host selector_subset_match and device
+# sel_subset alike).
+SELECTOR_POISON = -1
 
 
 _label_overflow_warned = False
@@ -99,7 +108,7 @@ class PodFlags:
     REPLICATED = 1 << 4
     DRAINABLE = 1 << 5      # may be evicted on scale-down
     BLOCKS_DRAIN = 1 << 6   # pins its node against scale-down
-    SELECTOR_OVERFLOW = 1 << 7  # >MAX_SELECTOR_PAIRS pairs: host fallback
+    SELECTOR_OVERFLOW = 1 << 7  # >MAX_SELECTOR_PAIRS pairs: poisoned, matches nothing
     BOUND = 1 << 8          # has spec.nodeName resolving to a known node
 
 
@@ -187,8 +196,17 @@ class ClusterSnapshot:
         return total
 
 
-def _hash_pairs(mapping: Dict[str, str], width: int) -> Tuple[np.ndarray, int, int]:
-    """dict -> (sorted padded int64 array, count, combined set hash)."""
+def _hash_pairs(
+    mapping: Dict[str, str], width: int, poison_overflow: bool = False
+) -> Tuple[np.ndarray, int, int]:
+    """dict -> (sorted padded int64 array, count, combined set hash).
+
+    poison_overflow (pod selectors): when the set is wider than the
+    lattice, the last slot becomes SELECTOR_POISON so the truncated
+    selector can never spuriously subset-match a label set — exact
+    "matches nothing" semantics (a >width selector cannot be a subset
+    of a <=width label set). Node labels do NOT poison (they truncate
+    with a loud warning: labels are the superset side of the walk)."""
     hashes = sorted(hash_label_pair(k, v) for k, v in mapping.items())
     combined = 0
     for h in hashes:
@@ -197,6 +215,8 @@ def _hash_pairs(mapping: Dict[str, str], width: int) -> Tuple[np.ndarray, int, i
     cnt = min(len(hashes), width)
     for i in range(cnt):
         out[i] = _as_i64(hashes[i])
+    if poison_overflow and len(hashes) > width:
+        out[width - 1] = SELECTOR_POISON  # max uint64: stays sorted-last
     return out, len(hashes), _as_i64(combined)
 
 
@@ -249,7 +269,8 @@ def encode_snapshot(
 
     for i, p in enumerate(pods):
         pod_req[i] = quantize_request(p.resources)
-        pairs, cnt, combined = _hash_pairs(p.selectors, MAX_SELECTOR_PAIRS)
+        pairs, cnt, combined = _hash_pairs(
+            p.selectors, MAX_SELECTOR_PAIRS, poison_overflow=True)
         pod_sel_pairs[i] = pairs
         pod_sel_count[i] = min(cnt, MAX_SELECTOR_PAIRS)
         pod_sel_hash[i] = combined

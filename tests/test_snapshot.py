@@ -147,7 +147,7 @@ def test_synth_soa_million_scale_fast():
 
 def test_wide_selector_sets_match_exactly():
     """12 selector pairs (> the old 8-pair width) must match exactly —
-    no truncation over-match; >16 still sets SELECTOR_OVERFLOW."""
+    no truncation over-match; >32 sets SELECTOR_OVERFLOW + poison."""
     from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools
     from mi355x_autoscaler.models.kube import KubeNode, KubePod
     from mi355x_autoscaler.sim.engine import CpuEngine
@@ -167,10 +167,34 @@ def test_wide_selector_sets_match_exactly():
     d = CpuEngine().decide(snap)
     assert d.pods_placed_existing == 1  # fits node_full only
 
+    # 17 pairs sat above the old 16-wide lattice; now matches exactly.
     sel17 = {f"k{i:02d}": f"v{i}" for i in range(17)}
+    node17 = make_node_json("agentpool1", 2)
+    node17["metadata"]["labels"].update(sel17)
+    nodes17 = [KubeNode(node17), nodes[1]]
     pod17 = make_pod_json("wider", cpu_milli=100, pending=True, node_selector=sel17)
-    snap17 = encode_snapshot([KubePod(pod17)], nodes, group_nodes_into_pools(nodes))
-    assert snap17.pod_flags[0] & PodFlags.SELECTOR_OVERFLOW
+    snap17 = encode_snapshot(
+        [KubePod(pod17)], nodes17, group_nodes_into_pools(nodes17))
+    assert snap17.pod_sel_count[0] == 17
+    assert not (snap17.pod_flags[0] & PodFlags.SELECTOR_OVERFLOW)
+    d17 = CpuEngine().decide(snap17)
+    assert d17.pods_placed_existing == 1  # fits node17 only
+
+    # >32 pairs cannot subset-match any representable (<=32-label) node:
+    # flagged SELECTOR_OVERFLOW and poisoned so it matches NOTHING, even
+    # a node whose first 32 sorted labels cover the truncated pairs.
+    sel33 = {f"k{i:02d}": f"v{i}" for i in range(33)}
+    node33 = make_node_json("agentpool1", 3)
+    node33["metadata"]["labels"].update(sel33)  # >32 labels: truncates too
+    nodes33 = [KubeNode(node33)]
+    pod33 = make_pod_json("widest", cpu_milli=100, pending=True, node_selector=sel33)
+    snap33 = encode_snapshot(
+        [KubePod(pod33)], nodes33, group_nodes_into_pools(nodes33))
+    assert snap33.pod_flags[0] & PodFlags.SELECTOR_OVERFLOW
+    from mi355x_autoscaler.snapshot.encode import SELECTOR_POISON
+    assert snap33.pod_sel_pairs[0, -1] == SELECTOR_POISON
+    d33 = CpuEngine().decide(snap33)
+    assert d33.pods_placed_existing == 0
 
 
 def test_subset_match_mixed_sign_hashes():
