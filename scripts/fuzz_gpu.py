@@ -60,7 +60,10 @@ def main(iters=None):
         for p in pods_json:
             r = rng.random()
             if r < 0.05:
-                p["spec"]["nodeSelector"] = {f"k{i}": f"v{i}" for i in range(rng.randint(9, 14))}
+                # wide widths straddle the 32-pair lattice edge: >32 takes
+                # the poison (matches-nothing) path on both engines
+                w = rng.choice([9, 14, 31, 32, 33, 40])
+                p["spec"]["nodeSelector"] = {f"k{i}": f"v{i}" for i in range(w)}
             elif r < 0.10:
                 p["spec"]["containers"][0]["resources"]["requests"] = {}
             elif r < 0.15:

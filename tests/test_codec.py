@@ -98,6 +98,28 @@ def test_codec_empty_list():
     _compare([], [node])
 
 
+def test_codec_selector_widths_and_overflow_poison():
+    """Widths around the 32-pair lattice edge: both encoders must agree
+    column-exactly, including the >32 poison (matches-nothing) row."""
+    from mi355x_autoscaler.snapshot.encode import (
+        MAX_SELECTOR_PAIRS, SELECTOR_POISON, PodFlags)
+
+    node = make_node_json("agentpool1", 0)
+    pods = []
+    for w in (31, 32, 33, 40):
+        sel = {f"k{i:02d}": f"v{i}" for i in range(w)}
+        pods.append(make_pod_json(f"w{w}", cpu_milli=50, pending=True,
+                                  node_selector=sel))
+    snap = _compare(pods, [node])
+    assert list(snap.pod_sel_count) == [31, 32, 32, 32]
+    for row, w in enumerate((31, 32, 33, 40)):
+        if w > MAX_SELECTOR_PAIRS:
+            assert snap.pod_flags[row] & PodFlags.SELECTOR_OVERFLOW
+            assert snap.pod_sel_pairs[row, -1] == SELECTOR_POISON
+        else:
+            assert not (snap.pod_flags[row] & PodFlags.SELECTOR_OVERFLOW)
+
+
 def test_codec_decision_parity():
     """End goal: identical decisions through either ingest path."""
     from mi355x_autoscaler.sim.engine import CpuEngine

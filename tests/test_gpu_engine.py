@@ -487,6 +487,23 @@ def test_wide_selector_decide_matches_oracle(engine):
     _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))
 
 
+def test_overflow_selector_matches_nothing_on_gpu(engine):
+    """>32-pair selectors are poisoned at encode (exact matches-nothing
+    semantics): the GPU pipeline must agree with the oracle and place
+    zero such pods, even on a node whose first 32 sorted labels cover
+    the truncated pair array."""
+    sel33 = {f"k{i:02d}": f"v{i}" for i in range(33)}
+    covered = make_node_json("agentpool1", 0)
+    covered["metadata"]["labels"].update(sel33)
+    pods = [make_pod_json(f"o{i}", cpu_milli=100, pending=True,
+                          node_selector=sel33) for i in range(3)]
+    pods += [make_pod_json("plain", cpu_milli=100, pending=True)]
+    snap = _snap(pods, [covered])
+    d_gpu, d_cpu = engine.decide(snap), CpuEngine().decide(snap)
+    _assert_decisions_equal(d_gpu, d_cpu)
+    assert d_gpu.pods_placed_existing == 1  # the plain pod only
+
+
 def test_decide_rows_device_matches_host_merge(engine):
     """Sharded DEVICE merge (rows_hash + compact on device) must equal
     the host-side numpy merge path on identical gathered rows —
