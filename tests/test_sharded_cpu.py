@@ -19,7 +19,7 @@ from mi355x_autoscaler.snapshot.encode import encode_snapshot
 WORLD = 2
 
 
-def _make_snap(rank=None, seed=11, n_pods=400, world=WORLD):
+def _make_snap(rank=None, seed=11, n_pods=400, world=WORLD, overflow=False):
     pods_json, nodes_json = burst_trace(
         n_pods,
         n_nodes_per_pool=(5, 2),
@@ -28,6 +28,16 @@ def _make_snap(rank=None, seed=11, n_pods=400, world=WORLD):
         seed=seed,
         bound_pods_per_node=3,
     )
+    if overflow:
+        # >32-pair selectors (poisoned, match nothing) riding through the
+        # sharded wire rows; added pre-shard so every world sees the same
+        # global pod set
+        from mi355x_autoscaler.sim.synth import make_pod_json
+
+        for j in range(4):
+            sel = {f"k{i:02d}": f"v{i}" for i in range(33 + j)}
+            pods_json.append(make_pod_json(
+                f"ovf{j}", cpu_milli=100, pending=True, node_selector=sel))
     if rank is not None:
         # shard pods round-robin; keep every bound pod on its rank too
         pods_json = pods_json[rank::world]
@@ -46,9 +56,9 @@ def _worker(rank, q, mode, world=WORLD, port="29617"):
     from mi355x_autoscaler.parallel.engine_sharded import ShardedEngine
 
     engine = ShardedEngine(CpuEngine())
-    snap = _make_snap(rank=rank, world=world)
+    snap = _make_snap(rank=rank, world=world, overflow=(mode == "decide_ovf"))
     try:
-        if mode == "decide":
+        if mode.startswith("decide"):
             d = engine.decide(snap)
             q.put((rank, pickle.dumps(dict(
                 units=d.units_needed, pe=d.pods_placed_existing,
@@ -92,6 +102,19 @@ def test_sharded_decide_matches_single_process():
     assert results[0]["pe"] == d.pods_placed_existing
     assert results[0]["pn"] == d.pods_placed_new
     assert results[0]["un"] == d.pods_unfulfillable
+
+
+def test_sharded_decide_with_overflow_selectors():
+    """Poisoned (>32-pair) selector groups through the sharded merge:
+    ranks agree and match the single-process decision; the overflow
+    pods land in pods_unfulfillable on both paths."""
+    results = _run_world("decide_ovf", port="29661")
+    assert results[0] == results[1]
+    snap = _make_snap(rank=None, overflow=True)
+    d = CpuEngine().decide(snap)
+    assert results[0]["units"] == d.units_needed
+    assert results[0]["un"] == d.pods_unfulfillable
+    assert d.pods_unfulfillable >= 4
 
 
 def test_sharded_maintain_matches_single_process():
