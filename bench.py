@@ -80,6 +80,25 @@ def build_snapshot_1m(rank: int, pods_per_rank: int):
     return snap
 
 
+# BASELINE config #2: 1k pending amd.com/gpu=1 pods, one MI355X pool.
+GPU1K_POOLS = (("gpupool", "Standard_ND96_MI355X_v1"),)
+GPU1K_POOL_SIZES = (8,)
+
+
+def build_snapshot_gpu1k(rank: int, pods_per_rank: int):
+    pods_json, nodes_json = burst_trace(
+        pods_per_rank,
+        n_nodes_per_pool=GPU1K_POOL_SIZES,
+        pools=GPU1K_POOLS,
+        gpu_fraction=1.0,
+        seed=1000 + rank,
+        bound_pods_per_node=2,
+    )
+    pods, nodes = wrap(pods_json, nodes_json)
+    pools = group_nodes_into_pools(nodes)
+    return encode_snapshot(pods, nodes, pools)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -87,14 +106,17 @@ def main():
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--pods-per-gpu", type=int, default=None)
     ap.add_argument("--backend", choices=["auto", "gpu", "cpu"], default="auto")
-    ap.add_argument("--bench-config", choices=["burst10k", "burst1m", "maintain500"],
+    ap.add_argument("--bench-config",
+                    choices=["burst10k", "burst1m", "maintain500", "gpu1k"],
                     default="burst10k",
                     help="burst10k = BASELINE config #3 (default); burst1m = config #5 "
                          "(1M pods, HBM-resident snapshot); maintain500 = config #4 "
-                         "(drain scoring over 500 under-utilized MI355X nodes with PDBs)")
+                         "(drain scoring over 500 under-utilized MI355X nodes with PDBs); "
+                         "gpu1k = config #2 (1k amd.com/gpu=1 pods, single MI355X pool)")
     args = ap.parse_args()
     if args.pods_per_gpu is None:
-        args.pods_per_gpu = 1_000_000 if args.bench_config == "burst1m" else 10_000
+        args.pods_per_gpu = {"burst1m": 1_000_000, "gpu1k": 1_000}.get(
+            args.bench_config, 10_000)
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -138,6 +160,8 @@ def main():
         pods_json, nodes_json, pdbs = scaledown_trace(n_nodes=500, seed=1000 + rank)
         pods, nodes = wrap(pods_json, nodes_json)
         snap = encode_snapshot(pods, nodes, _group(nodes, max_size=1000), pdbs)
+    elif args.bench_config == "gpu1k":
+        snap = build_snapshot_gpu1k(rank, args.pods_per_gpu)
     else:
         snap = build_snapshot(rank, args.pods_per_gpu)
 
@@ -218,12 +242,18 @@ def main():
                     if resident else
                     "scale-down drain scoring, 500 under-utilized MI355X nodes + PDBs (BASELINE config #4)"
                     if maintain_mode else
+                    "1k amd.com/gpu=1 pods, single MI355X agent pool (BASELINE config #2)"
+                    if args.bench_config == "gpu1k" else
                     "10k-pod mixed CPU/GPU burst, 3 heterogeneous agent pools (BASELINE config #3)"
                 ),
                 "pods_per_gpu": args.pods_per_gpu,
-                "nodes": sum(BURST1M_POOL_SIZES if resident else POOL_SIZES),
-                "pools": 3,
-                "gpu_pod_fraction": GPU_FRACTION,
+                "nodes": sum(
+                    BURST1M_POOL_SIZES if resident
+                    else GPU1K_POOL_SIZES if args.bench_config == "gpu1k"
+                    else POOL_SIZES),
+                "pools": 1 if args.bench_config == "gpu1k" else 3,
+                "gpu_pod_fraction": (
+                    1.0 if args.bench_config == "gpu1k" else GPU_FRACTION),
                 "parallelism": f"pod-sharded controller x{world}" if world > 1 else "single",
                 "backend": "gpu" if use_gpu else "cpu-oracle",
                 "decision_p50_ms": round(p50_ms, 3),
