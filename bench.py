@@ -246,12 +246,9 @@ def main():
                     if args.bench_config == "gpu1k" else
                     "10k-pod mixed CPU/GPU burst, 3 heterogeneous agent pools (BASELINE config #3)"
                 ),
-                "pods_per_gpu": args.pods_per_gpu,
-                "nodes": sum(
-                    BURST1M_POOL_SIZES if resident
-                    else GPU1K_POOL_SIZES if args.bench_config == "gpu1k"
-                    else POOL_SIZES),
-                "pools": 1 if args.bench_config == "gpu1k" else 3,
+                "pods_per_gpu": snap.num_pods if maintain_mode else args.pods_per_gpu,
+                "nodes": int(snap.node_alloc.shape[0]),
+                "pools": len(snap.pool_names),
                 "gpu_pod_fraction": (
                     1.0 if args.bench_config == "gpu1k" else GPU_FRACTION),
                 "parallelism": f"pod-sharded controller x{world}" if world > 1 else "single",
