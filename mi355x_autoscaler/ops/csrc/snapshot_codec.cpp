@@ -791,9 +791,24 @@ void pack_into(pybind11::buffer buf, const std::vector<int64_t>& offsets,
     py::buffer src_b = py::cast<py::buffer>(arrays[i]);
     py::buffer_info src = src_b.request(false);
     int64_t nb = src.size * src.itemsize;
-    // contiguity: itemsize-stride rows (ascontiguousarray upstream)
     if (offsets[i] + nb > cap) throw std::runtime_error("pack overflow");
-    std::memcpy(base + offsets[i], src.ptr, (size_t)nb);
+    const bool strided2d =
+        src.ndim == 2 && src.strides[1] == src.itemsize &&
+        src.strides[0] != src.shape[1] * src.itemsize;
+    if (strided2d) {
+      // row-contiguous 2D view with a row gap (e.g. the narrowed
+      // pod_sel_pairs[:, :W] slice): gather rows in one pass instead
+      // of forcing an ascontiguousarray copy upstream
+      const int64_t rows = src.shape[0];
+      const size_t rowb = (size_t)(src.shape[1] * src.itemsize);
+      const char* sp = static_cast<const char*>(src.ptr);
+      char* dp = base + offsets[i];
+      for (int64_t r = 0; r < rows; ++r)
+        std::memcpy(dp + (size_t)r * rowb, sp + r * src.strides[0], rowb);
+    } else {
+      // contiguous (ascontiguousarray upstream)
+      std::memcpy(base + offsets[i], src.ptr, (size_t)nb);
+    }
   }
 }
 

@@ -137,6 +137,9 @@ class _PinnedPacker:
         self._total = 0
         self._wide_pairs: Optional[torch.Tensor] = None
         self._wide_selw = 0
+        self._shape_sig = None   # layout cache key (shapes fix the layout)
+        self._pack_offs = None   # cached [off per field] for pack_into
+        self._buf_np = None      # cached numpy view of the staging buffer
         self._ops = load_ops()
 
     def _layout(self, snap: ClusterSnapshot, overrides=None):
@@ -169,10 +172,23 @@ class _PinnedPacker:
         if P > 0:  # same path on CPU so tests cover the expand exactly
             selw = max(1, int(snap.pod_sel_count.max()))
             if selw < full_w:
-                overrides = {"pod_sel_pairs": np.ascontiguousarray(
-                    snap.pod_sel_pairs[:, :selw])}
+                # strided view: pack_into gathers the rows in C
+                overrides = {"pod_sel_pairs": snap.pod_sel_pairs[:, :selw]}
 
-        offsets, total = self._layout(snap, overrides)
+        # layout is a pure function of the field shapes: cache it (the
+        # per-field offset arithmetic was ~45 us/pack at 10k pods)
+        shape_sig = tuple(
+            (overrides["pod_sel_pairs"] if overrides and name == "pod_sel_pairs"
+             else getattr(snap, attr)).shape
+            for name, attr, _, _ in self.fields
+        )
+        if shape_sig == self._shape_sig:
+            offsets, total = self._offsets, self._total
+        else:
+            offsets, total = self._layout(snap, overrides)
+            self._offsets, self._total = offsets, total
+            self._shape_sig = shape_sig
+            self._pack_offs = [o for _, _, o, _, _, _ in offsets]
         sig = tuple(shape for _, _, _, _, _, shape in offsets)
 
         rebuild = (
@@ -184,13 +200,14 @@ class _PinnedPacker:
         if self.buf is None or self.buf.numel() != total:
             pin = torch.cuda.is_available()
             self.buf = torch.empty(total, dtype=torch.uint8, pin_memory=pin)
+            self._buf_np = self.buf.numpy()
             rebuild = True
         arrays = [
             overrides[name] if overrides and name in overrides
             else np.ascontiguousarray(getattr(snap, attr), dtype=dt)
             for name, attr, o, nb, dt, shape in offsets
         ]
-        self._ops.pack_into(self.buf.numpy(), [o for _, _, o, _, _, _ in offsets], arrays)
+        self._ops.pack_into(self._buf_np, self._pack_offs, arrays)
 
         if rebuild:
             # on CPU (tests) the views alias the staging buffer itself —

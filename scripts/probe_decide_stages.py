@@ -55,6 +55,21 @@ def main():
     t = ms(lambda: None, 1000)
     print(f"timer overhead     : {t:7.3f} ms")
 
+    # gpu1k shape (BASELINE config #2): 1k pods, 1 pool, 8 nodes — why
+    # is the tiny burst SLOWER per step than the 10k one?
+    pods_json, nodes_json = burst_trace(
+        1000, n_nodes_per_pool=(8,),
+        pools=(("gpupool", "Standard_ND96_MI355X_v1"),),
+        gpu_fraction=1.0, seed=0, bound_pods_per_node=2)
+    pods, nodes = wrap(pods_json, nodes_json)
+    snap1k = encode_snapshot(pods, nodes, group_nodes_into_pools(nodes))
+    for _ in range(3):
+        eng.decide(snap1k)
+    print(f"[gpu1k] full decide: {ms(lambda: eng.decide(snap1k)):7.3f} ms")
+    print(f"[gpu1k] pack       : {ms(lambda: eng.packer.pack(snap1k, eng.device)):7.3f} ms")
+    ds1 = eng.packer.pack(snap1k, eng.device)
+    print(f"[gpu1k] from ds    : {ms(lambda: eng._decide_fused(ds1, snap1k.pool_names)):7.3f} ms")
+
 
 if __name__ == "__main__":
     main()
