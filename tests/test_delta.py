@@ -62,6 +62,38 @@ def test_delta_add_modify_delete_converges():
     assert len(cache.drain_dirty()) == 0  # drained
 
 
+def test_delta_overflow_selector_rows_match_full_encode():
+    """Watch-added pods with >32-pair selectors take the poison path in
+    the cache's row encoder exactly as in encode_snapshot: same pairs
+    column (incl. the poison slot), same flags, same decision."""
+    from mi355x_autoscaler.snapshot.encode import SELECTOR_POISON, PodFlags
+
+    nodes = [KubeNode(make_node_json("agentpool1", i)) for i in range(2)]
+    base, pools = _base(nodes)
+    cache = DeltaSnapshotCache()
+    cache.prime(_full([], nodes), {n.name: i for i, n in enumerate(nodes)})
+
+    sel33 = {f"k{i:02d}": f"v{i}" for i in range(33)}
+    ovf = make_pod_json("ovf", cpu_milli=100, pending=True, node_selector=sel33)
+    plain = make_pod_json("plain", cpu_milli=100, pending=True)
+    cache.apply_event({"type": "ADDED", "object": ovf})
+    cache.apply_event({"type": "ADDED", "object": plain})
+
+    snap_delta = cache.build_snapshot(base)
+    snap_full = _full([ovf, plain], nodes)
+    rows = {u: i for i, u in enumerate(snap_delta.pod_uids)}
+    full_rows = {u: i for i, u in enumerate(snap_full.pod_uids)}
+    for uid in ("pod-default-ovf", "pod-default-plain"):
+        a = snap_delta.pod_sel_pairs[rows[uid]]
+        b = snap_full.pod_sel_pairs[full_rows[uid]]
+        np.testing.assert_array_equal(a, b, err_msg=uid)
+        assert snap_delta.pod_flags[rows[uid]] == snap_full.pod_flags[full_rows[uid]]
+    ovf_row = rows["pod-default-ovf"]
+    assert snap_delta.pod_sel_pairs[ovf_row][-1] == SELECTOR_POISON
+    assert snap_delta.pod_flags[ovf_row] & PodFlags.SELECTOR_OVERFLOW
+    _assert_equivalent_decisions(snap_delta, snap_full)
+
+
 def test_delta_growth_and_compaction():
     nodes = [KubeNode(make_node_json("agentpool1", 0))]
     base, pools = _base(nodes)
