@@ -424,6 +424,47 @@ __device__ inline void dev_binpack(
   }
   __syncthreads();
 
+  if (N <= WAVE) {
+    // single-wave fast path: one wave covers every node, so the
+    // G-serial walk needs NO block barriers — each iteration is a
+    // wave shuffle-scan instead of a 2-barrier LDS scan. At tiny N
+    // (the 1k-pod / 8-node BASELINE config #2 shape) the segment loop
+    // is latency-bound on exactly those barriers.
+    if (tid < WAVE) {
+      const int lane = tid;
+      for (int g = 0; g < G; ++g) {
+        const int64_t need = g < BP_PRELOAD ? s_cnt[g] : seg_count[g];
+        if (need <= 0) {
+          if (lane == 0) placed_out[g] = 0;
+          continue;
+        }
+        int32_t rq[NRES];
+#pragma unroll
+        for (int r = 0; r < NRES; ++r)
+          rq[r] = g < BP_PRELOAD ? s_req[g * NRES + r] : seg_req[g * NRES + r];
+        int64_t m = 0;
+        if (lane < N && ((elig[(int64_t)g * W] >> lane) & 1ull)) {
+          m = per_node_capacity(fr, rq);
+        }
+        const int64_t inc = wave_inclusive_scan_i64(m);
+        const int64_t chunk_total = __shfl(inc, WAVE - 1);
+        int64_t take = need - (inc - m);
+        take = take < 0 ? 0 : (take > m ? m : take);
+        if (take > 0) {
+#pragma unroll
+          for (int r = 0; r < NRES; ++r) fr[r] -= (int32_t)(take * rq[r]);
+        }
+        if (lane == 0) placed_out[g] = chunk_total < need ? chunk_total : need;
+      }
+      if (lane < N) {  // keep the mutated-free contract
+#pragma unroll
+        for (int r = 0; r < NRES; ++r) free_[(int64_t)lane * NRES + r] = fr[r];
+      }
+    }
+    __syncthreads();  // other waves rejoin; wave 0's writes visible
+    return;
+  }
+
   for (int g = 0; g < G; ++g) {
     const int64_t need = g < BP_PRELOAD ? s_cnt[g] : seg_count[g];
     if (need <= 0) {
