@@ -221,6 +221,8 @@ def main():
     pods_total = world * (snap.num_pods if maintain_mode else args.pods_per_gpu)
     pods_per_sec = pods_total / (total_s / args.steps)
     p50_ms = statistics.median(step_times) * 1000.0
+    p99_ms = sorted(step_times)[min(len(step_times) - 1,
+                                    int(0.99 * len(step_times)))] * 1000.0
 
     if rank == 0:
         result = {
@@ -254,6 +256,7 @@ def main():
                 "parallelism": f"pod-sharded controller x{world}" if world > 1 else "single",
                 "backend": "gpu" if use_gpu else "cpu-oracle",
                 "decision_p50_ms": round(p50_ms, 3),
+                "decision_p99_ms": round(p99_ms, 3),
                 **(
                     {"drain_candidates": len(decision.candidates)}
                     if maintain_mode else
