@@ -432,6 +432,25 @@ class Cluster:
                         logger.exception("orphan node cleanup failed for %s", node.name)
 
         drainable_states = (STATE_IDLE_SCHEDULABLE, STATE_IDLE_UNSCHEDULABLE, STATE_UNDER_UTILIZED_DRAINABLE)
+        # PDB budgets are consumed by THIS cycle's earlier drains too:
+        # kernel D's per-node pdb_ok is checked against the fresh API
+        # budget, so two candidates sharing one PDB could jointly
+        # violate it — track the remaining allowance across the serial
+        # drain loop (deterministic on every rank: counts come from the
+        # full actuation snapshot).
+        import numpy as np
+
+        pdb_left = actuation_snap.pdb_allowed.astype(np.int64).copy()
+        pod_pdb = actuation_snap.pod_pdb
+        pod_node_full = actuation_snap.pod_node
+
+        def _pdb_counts(j: int) -> Dict[int, int]:
+            rows = (pod_node_full == j) & (pod_pdb >= 0)
+            counts: Dict[int, int] = {}
+            for d in pod_pdb[rows]:
+                counts[int(d)] = counts.get(int(d), 0) + 1
+            return counts
+
         # drain lowest-utilization first (report.candidates is util-sorted)
         for cand in report.candidates:
             if len(drained) >= self.opts.max_drains_per_cycle:
@@ -443,11 +462,19 @@ class Cluster:
             node = node_by_name.get(name)
             if node is None:
                 continue
+            counts = _pdb_counts(j)
+            if any(c > pdb_left[d] for d, c in counts.items()):
+                logger.info("skipping drain of %s: PDB budget exhausted this cycle", name)
+                continue
             if not actuate:
+                for d, c in counts.items():
+                    pdb_left[d] -= c
                 drained.append(name)  # agreed decision; rank 0 executes
                 continue
             try:
                 self._drain_and_delete(actuation_snap, node, j)
+                for d, c in counts.items():
+                    pdb_left[d] -= c
                 drained.append(name)
                 self.metrics.count("scale_downs")
             except Exception:

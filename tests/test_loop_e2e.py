@@ -135,3 +135,30 @@ def test_instance_terminated_orphan_cleanup(server):
     cluster.loop_logic()
     assert orphan in server.state.deleted_nodes
     assert orphan not in server.state.nodes
+
+
+def test_shared_pdb_budget_across_serial_drains(server):
+    """Two under-utilized nodes each host one pod guarded by the SAME
+    PDB with disruptionsAllowed=1: kernel D's per-node check passes for
+    both, but the cycle-level drain loop must consume the shared budget
+    and drain only ONE node this cycle (the second retries next cycle,
+    after the API serves a fresh budget)."""
+    nodes = [make_node_json("agentpool1", i) for i in range(3)]
+    busy = make_pod_json("busy", cpu_milli=1500,
+                         node_name=nodes[0]["metadata"]["name"])
+    g1 = make_pod_json("g1", cpu_milli=100,
+                       node_name=nodes[1]["metadata"]["name"],
+                       labels={"app": "guarded"})
+    g2 = make_pod_json("g2", cpu_milli=100,
+                       node_name=nodes[2]["metadata"]["name"],
+                       labels={"app": "guarded"})
+    pdb = {"metadata": {"name": "guard"},
+           "spec": {"selector": {"matchLabels": {"app": "guarded"}}},
+           "status": {"disruptionsAllowed": 1}}
+    server.state.load([busy, g1, g2], nodes, pdbs=[pdb])
+    cluster, provider = _cluster(server, no_scale=True, max_drains_per_cycle=10)
+
+    res = cluster.loop_logic()
+    assert len(res.drained_nodes) == 1
+    assert res.drained_nodes[0] in (nodes[1]["metadata"]["name"],
+                                    nodes[2]["metadata"]["name"])
