@@ -443,13 +443,20 @@ class Cluster:
         pdb_left = actuation_snap.pdb_allowed.astype(np.int64).copy()
         pod_pdb = actuation_snap.pod_pdb
         pod_node_full = actuation_snap.pod_node
+        # one vectorized pass: (node, pdb) -> pod count for every
+        # guarded bound pod (a per-candidate O(P) scan would cost
+        # candidates x milliseconds at 1M pods)
+        D = max(1, len(pdb_left))
+        guarded = (pod_pdb >= 0) & (pod_node_full >= 0)
+        node_pdb_counts: Dict[int, Dict[int, int]] = {}
+        if guarded.any():
+            key = pod_node_full[guarded].astype(np.int64) * D + pod_pdb[guarded]
+            uniq, cnts = np.unique(key, return_counts=True)
+            for k, c in zip(uniq, cnts):
+                node_pdb_counts.setdefault(int(k) // D, {})[int(k) % D] = int(c)
 
         def _pdb_counts(j: int) -> Dict[int, int]:
-            rows = (pod_node_full == j) & (pod_pdb >= 0)
-            counts: Dict[int, int] = {}
-            for d in pod_pdb[rows]:
-                counts[int(d)] = counts.get(int(d), 0) + 1
-            return counts
+            return node_pdb_counts.get(j, {})
 
         # drain lowest-utilization first (report.candidates is util-sorted)
         for cand in report.candidates:

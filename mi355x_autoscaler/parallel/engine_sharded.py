@@ -186,7 +186,7 @@ class ShardedEngine(SimulationEngine):
                     gsegs = self._rows_to_device_segments(merged, device)
                     decision = self.inner.decide_device(
                         ds, gsegs, snap.pool_names, over_provision=over_provision,
-                        free_hook=free_hook,
+                        free_hook=hook,
                     )
         if self.active and self.check_agreement:
             t0 = time.perf_counter()
