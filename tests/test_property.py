@@ -84,3 +84,60 @@ def test_pack_key_order_preserving(a, b):
     arr = np.array([a, b], np.int64)
     ks = pack_req_keys(arr)
     assert int(ks[0]) == ka and int(ks[1]) == kb
+
+
+@st.composite
+def heterogeneous_case(draw):
+    """Wider shapes: two pools (CPU + MI355X), selectors that match one
+    pool / neither, gpu requests, criticals, bound pods, zero requests."""
+    n_cpu = draw(st.integers(1, 4))
+    n_gpu = draw(st.integers(0, 3))
+    nodes = [make_node_json("agentpool1", i,
+                            cpu_milli=draw(st.sampled_from([2000, 4000])))
+             for i in range(n_cpu)]
+    nodes += [make_node_json("gpupool", i, instance_type="Standard_ND96_MI355X_v1",
+                             cpu_milli=96000, mem_mib=65536, gpu=8)
+              for i in range(n_gpu)]
+    sel_menu = [
+        None,
+        {"accelerator": "amd-mi355x"},            # matches gpupool only
+        {"agentpool": "agentpool1"},              # matches the CPU pool
+        {"nosuch": "label"},                      # matches nothing
+    ]
+    pods = []
+    for i in range(draw(st.integers(0, 50))):
+        gpu = draw(st.sampled_from([0, 0, 0, 1, 2]))
+        pods.append(make_pod_json(
+            f"p{i}",
+            cpu_milli=draw(st.sampled_from([0, 100, 1500, 3000])),
+            mem_mib=draw(st.sampled_from([0, 256, 2048])),
+            gpu=gpu,
+            pending=True,
+            node_selector=draw(st.sampled_from(sel_menu)),
+            annotations=(
+                {"scheduler.alpha.kubernetes.io/critical-pod": "true"}
+                if draw(st.booleans()) else None),
+        ))
+    # a few bound pods consume node capacity
+    for b in range(draw(st.integers(0, 4))):
+        j = draw(st.integers(0, len(nodes) - 1))
+        pods.append(make_pod_json(
+            f"b{b}", cpu_milli=draw(st.sampled_from([200, 900])),
+            node_name=nodes[j]["metadata"]["name"]))
+    max_size = draw(st.integers(1, 8))
+    return pods, nodes, max_size
+
+
+@settings(max_examples=60, deadline=None)
+@given(heterogeneous_case())
+def test_oracle_equals_naive_heterogeneous(case):
+    pods_json, nodes_json, max_size = case
+    pods, nodes = wrap(pods_json, nodes_json)
+    pools = group_nodes_into_pools(nodes, max_size=max_size)
+    snap = encode_snapshot(pods, nodes, pools)
+    d = OracleSimulator(snap).fulfill_pending()
+    naive = naive_fulfill(snap)
+    assert d.pods_placed_existing == naive["placed_existing"]
+    assert d.pods_placed_new == naive["placed_new"]
+    assert d.pods_unfulfillable == naive["unfulfillable"]
+    assert d.units_needed == naive["units_needed"]
