@@ -93,6 +93,10 @@ void fused_decide_launch(
   const int64_t K = pool_unit.size(0);
   const int64_t N = node_alloc.size(0);
   const int64_t P = pod_flags.size(0);
+  // dev_units keeps per-pool state in fixed LDS (UNITS_MAX_POOLS=64);
+  // without this check K>64 would write out of LDS bounds. The Python
+  // engine falls back to the CPU oracle before reaching here.
+  TORCH_CHECK(K <= 64, "fused decide supports at most 64 agent pools, got ", K);
 
   // small/medium-P regime: ONE cooperative launch runs every phase
   // (mega_decide.hip); falls through to the classic multi-launch

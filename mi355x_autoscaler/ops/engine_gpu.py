@@ -44,6 +44,10 @@ from ..snapshot.encode import MAX_SELECTOR_PAIRS as _MAXSEL
 from ..snapshot.encode import ClusterSnapshot
 from . import load_ops
 
+import logging
+
+logger = logging.getLogger(__name__)
+
 _PACK_MAX = 1 << 56
 _FASTPATH_WI: dict = {}  # (device, M) -> cached segsum chunk work items
 _SEGSUM_CHUNK = 256  # exactness bound for the f32 MFMA limb chains
@@ -797,6 +801,20 @@ class GpuEngine(SimulationEngine):
 
     # ------------------------------------------------------------------
     def decide(self, snap: ClusterSnapshot, over_provision: int = 0) -> ScaleDecision:
+        if snap.num_pools > 64:
+            # the units kernel keeps per-pool FFD state in fixed LDS
+            # (UNITS_MAX_POOLS); >64 agent pools is outside every ACS
+            # deployment shape we target — decide on the CPU oracle
+            # (bit-identical by construction) instead of crashing.
+            if not getattr(self, "_warned_many_pools", False):
+                logger.warning(
+                    "%d agent pools exceeds the GPU units kernel's 64-pool "
+                    "LDS budget; deciding on the CPU oracle", snap.num_pools)
+                self._warned_many_pools = True
+            from ..sim.oracle import OracleSimulator
+
+            return OracleSimulator(
+                snap, over_provision=over_provision).fulfill_pending()
         ds = self.packer.pack(snap, self.device)
         if not self.force_torch_segbuild:
             decision = self._decide_fused(ds, snap.pool_names, over_provision)
@@ -1164,6 +1182,12 @@ class GpuEngine(SimulationEngine):
         node/pool-side snapshot (zero pods)."""
         import types
 
+        if base.num_pools > 64:  # same LDS bound as decide(); oracle path
+            from ..sim.oracle import OracleSimulator
+
+            snap = cache.build_snapshot(base)
+            return OracleSimulator(
+                snap, over_provision=over_provision).fulfill_pending()
         cap = cache.cap
         res = self._resident
         if res is None or res["cap"] != cap:

@@ -487,6 +487,18 @@ def test_wide_selector_decide_matches_oracle(engine):
     _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))
 
 
+def test_more_than_64_pools_falls_back_to_oracle(engine):
+    """>64 agent pools exceeds the units kernel's fixed LDS budget: the
+    engine must decide on the CPU oracle (bit-identical), not corrupt
+    LDS or crash (the orchestrator TORCH_CHECKs as a second fence)."""
+    nodes_json = [make_node_json(f"pool{i:02d}", 0) for i in range(66)]
+    pods = [make_pod_json(f"p{i}", cpu_milli=500, pending=True)
+            for i in range(40)]
+    snap = _snap(pods, nodes_json)
+    assert snap.num_pools == 66
+    _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))
+
+
 def test_overflow_selector_matches_nothing_on_gpu(engine):
     """>32-pair selectors are poisoned at encode (exact matches-nothing
     semantics): the GPU pipeline must agree with the oracle and place
