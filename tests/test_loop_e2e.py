@@ -162,3 +162,17 @@ def test_shared_pdb_budget_across_serial_drains(server):
     assert len(res.drained_nodes) == 1
     assert res.drained_nodes[0] in (nodes[1]["metadata"]["name"],
                                     nodes[2]["metadata"]["name"])
+
+
+def test_over_provision_pads_scale_up(server):
+    """--over-provision N pads every scale-up by N extra agents (capped
+    at the pool max) [R: reference main.py --over-provision]."""
+    pods, nodes = burst_trace(20, n_nodes_per_pool=(2,), seed=7)
+    server.state.load(pods, nodes)
+    plain, p_plain = _cluster(server, no_maintenance=True)
+    base_units = plain.loop_logic().decision.units_needed["agentpool1"]
+
+    server.state.load(pods, nodes)  # reset
+    padded, p_pad = _cluster(server, no_maintenance=True, over_provision=2)
+    padded_units = padded.loop_logic().decision.units_needed["agentpool1"]
+    assert padded_units == base_units + 2
