@@ -73,3 +73,51 @@ def test_prometheus_export_starts():
     body = urllib.request.urlopen(f"http://127.0.0.1:{port}/metrics", timeout=5).read().decode()
     assert "autoscaler_cycles_total" in body
     assert "autoscaler_pods_pending 3.0" in body
+
+
+def test_kubeconfig_parsing(tmp_path):
+    """from_kubeconfig resolves current-context -> cluster/user, token
+    auth, and the TLS-verify setting [R: reference used pykube.KubeConfig
+    — reconstructed]."""
+    import yaml
+
+    from mi355x_autoscaler.kube.client import KubeApiClient
+
+    cfg = {
+        "current-context": "prod",
+        "contexts": [
+            {"name": "dev", "context": {"cluster": "c-dev", "user": "u-dev"}},
+            {"name": "prod", "context": {"cluster": "c-prod", "user": "u-prod"}},
+        ],
+        "clusters": [
+            {"name": "c-dev", "cluster": {"server": "https://dev:6443"}},
+            {"name": "c-prod", "cluster": {"server": "https://prod:6443",
+                                           "insecure-skip-tls-verify": True}},
+        ],
+        "users": [
+            {"name": "u-dev", "user": {}},
+            {"name": "u-prod", "user": {"token": "sekrit"}},
+        ],
+    }
+    p = tmp_path / "kubeconfig"
+    p.write_text(yaml.safe_dump(cfg))
+    kube = KubeApiClient.from_kubeconfig(str(p))
+    assert kube.base_url.startswith("https://prod:6443")
+    assert kube.session.headers["Authorization"] == "Bearer sekrit"
+    assert kube.session.verify is False
+
+
+def test_in_cluster_reads_service_account(tmp_path, monkeypatch):
+    import mi355x_autoscaler.kube.client as kc
+
+    sa = tmp_path / "sa"
+    sa.mkdir()
+    (sa / "token").write_text("tok123\n")
+    (sa / "ca.crt").write_text("CERT")
+    monkeypatch.setattr(kc, "_SA_DIR", str(sa))
+    monkeypatch.setenv("KUBERNETES_SERVICE_HOST", "10.0.0.1")
+    monkeypatch.setenv("KUBERNETES_SERVICE_PORT", "443")
+    kube = kc.KubeApiClient.in_cluster()
+    assert kube.base_url == "https://10.0.0.1:443"
+    assert kube.session.headers["Authorization"] == "Bearer tok123"
+    assert str(kube.session.verify).endswith("ca.crt")
