@@ -142,6 +142,10 @@ class KubePod:
             return False  # restarts with the node; does not pin it
         return not self.is_drainable
 
+    def delete(self, kube, grace_seconds: int = 30) -> None:
+        """[R: autoscaler/kube.py · KubePod.delete — reconstructed]"""
+        kube.delete_pod(self.namespace, self.name, grace_seconds=grace_seconds)
+
     def __repr__(self):
         return f"KubePod({self.namespace}/{self.name})"
 
@@ -188,6 +192,32 @@ class KubeNode:
             if cond.get("type") == "Ready":
                 return cond.get("status") == "True"
         return False
+
+    # -- actuation verbs (the reference put these on KubeNode; here
+    # they are thin delegations to the injected API client so the data
+    # object stays transport-free)
+    # [R: autoscaler/kube.py · cordon/uncordon/drain/delete — reconstructed]
+    def cordon(self, kube) -> None:
+        kube.set_unschedulable(self.name, True)
+        self.unschedulable = True
+
+    def uncordon(self, kube) -> None:
+        kube.set_unschedulable(self.name, False)
+        self.unschedulable = False
+
+    def drain(self, kube, pods, grace_seconds: int = 30) -> int:
+        """Evict the given resident pods (mirror/daemonset pods are
+        skipped, as `kubectl drain` does). Returns pods deleted."""
+        n = 0
+        for p in pods:
+            if p.is_mirrored or p.is_daemonset:
+                continue
+            kube.delete_pod(p.namespace, p.name, grace_seconds=grace_seconds)
+            n += 1
+        return n
+
+    def delete(self, kube) -> None:
+        kube.delete_node(self.name)
 
     def __repr__(self):
         return f"KubeNode({self.name})"

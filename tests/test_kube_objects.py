@@ -54,3 +54,36 @@ def test_node_pool_parse_and_provider_id():
     assert n.is_ready
     master = KubeNode({"metadata": {"name": "k8s-master-12345678-0x"}, "spec": {}, "status": {}})
     assert master.pool_name is None
+
+
+def test_node_actuation_verbs_against_fake_api():
+    """KubeNode.cordon/uncordon/drain/delete and KubePod.delete — the
+    reference's verb surface — against the in-process API server."""
+    from mi355x_autoscaler.kube.client import KubeApiClient
+    from mi355x_autoscaler.kube.fake_api import FakeKubeServer
+    from mi355x_autoscaler.sim.synth import make_node_json, make_pod_json
+
+    with FakeKubeServer() as srv:
+        nj = make_node_json("agentpool1", 0)
+        name = nj["metadata"]["name"]
+        pods = [
+            make_pod_json("w1", cpu_milli=100, node_name=name),
+            make_pod_json("ds", cpu_milli=50, node_name=name, owner_kind="DaemonSet"),
+        ]
+        srv.state.load(pods, [nj])
+        kube = KubeApiClient(srv.base_url)
+
+        node = KubeNode(nj)
+        node.cordon(kube)
+        assert srv.state.nodes[name]["spec"]["unschedulable"] is True
+        assert node.unschedulable
+        node.uncordon(kube)
+        assert not srv.state.nodes[name]["spec"].get("unschedulable")
+
+        wrapped = [KubePod(p) for p in pods]
+        n = node.drain(kube, wrapped)
+        assert n == 1  # daemonset pod skipped, like kubectl drain
+        assert "default/w1" not in srv.state.pods
+
+        node.delete(kube)
+        assert name not in srv.state.nodes
