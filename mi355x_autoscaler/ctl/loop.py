@@ -495,10 +495,15 @@ class Cluster:
             logger.info("[dry-run] would drain %s", node.name)
             return
         self.kube.set_unschedulable(node.name, True)
+        from ..snapshot.encode import PodFlags
+
+        skip = PodFlags.MIRROR | PodFlags.DAEMONSET
         n_drained = 0
         for i in range(snap.num_pods):
             if int(snap.pod_node[i]) != j:
                 continue
+            if int(snap.pod_flags[i]) & skip:
+                continue  # mirror/daemonset pods survive, as kubectl drain
             ns, name = snap.pod_refs[i]
             self.kube.delete_pod(ns, name, grace_seconds=self.opts.drain_grace_seconds)
             n_drained += 1
