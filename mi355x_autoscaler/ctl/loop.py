@@ -26,6 +26,8 @@ import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
 
+import numpy as np
+
 from ..cloud.provider import CloudProvider
 from ..kube.client import KubeApiClient
 from ..kube.objects import PodDisruptionBudget
@@ -226,9 +228,20 @@ class Cluster:
                 self.provider.scale_pools(target, dry_run=self.opts.dry_run)
             self.metrics.count("scale_ups")
             self.notifier.notify_scale_up(decision.units_needed, result.pods_pending)
-        if is_rank0:
+        if is_rank0 and decision.unfulfillable_detail:
+            # name one pending pod per unfulfillable selector group so
+            # the notification is actionable (the lattice carries only
+            # the group hash)
+            examples: Dict[int, str] = {}
+            pending_rows = np.nonzero((snap.pod_flags & PodFlags.PENDING) != 0)[0]
+            for i in pending_rows:
+                h = int(snap.pod_sel_hash[i]) & ((1 << 64) - 1)
+                if h not in examples and snap.pod_refs:
+                    ns, name = snap.pod_refs[int(i)]
+                    examples[h] = f"{ns}/{name}"
             for sel_hash, count in decision.unfulfillable_detail:
-                self.notifier.notify_failed_to_scale(sel_hash, count)
+                self.notifier.notify_failed_to_scale(
+                    sel_hash, count, example_pod=examples.get(sel_hash, ""))
 
         # ---- scale-down ------------------------------------------------
         suppress = decision.wants_scale_up() and not self.opts.scale_down_during_scale_up
@@ -438,8 +451,6 @@ class Cluster:
         # violate it — track the remaining allowance across the serial
         # drain loop (deterministic on every rank: counts come from the
         # full actuation snapshot).
-        import numpy as np
-
         pdb_left = actuation_snap.pdb_allowed.astype(np.int64).copy()
         pod_pdb = actuation_snap.pod_pdb
         pod_node_full = actuation_snap.pod_node

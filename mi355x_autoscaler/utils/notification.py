@@ -43,11 +43,17 @@ class Notifier:
         lines = ", ".join(f"{pool}: +{n}" for pool, n in units_needed.items())
         self._post(f":chart_with_upwards_trend: scaling up ({lines}) for {pods_pending} pending pods")
 
-    def notify_failed_to_scale(self, selector_hash: int, pods: int):
-        self._record("failed_to_scale", selector_hash=selector_hash, pods=pods)
+    def notify_failed_to_scale(self, selector_hash: int, pods: int,
+                               example_pod: str = ""):
+        """[R: reference formatted the pending pod list into the Slack
+        message; the lattice carries hashes, so we name an example pod
+        from the group instead.]"""
+        self._record("failed_to_scale", selector_hash=selector_hash, pods=pods,
+                     example_pod=example_pod)
+        eg = f", e.g. {example_pod}" if example_pod else ""
         self._post(
-            f":warning: failed to scale: {pods} pods (selector group {selector_hash:#x}) "
-            "fit no pool or pool at max"
+            f":warning: failed to scale: {pods} pods (selector group "
+            f"{selector_hash:#x}{eg}) fit no pool or pool at max"
         )
 
     def notify_scale_down(self, node_name: str, drained_pods: int):

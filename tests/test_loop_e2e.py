@@ -176,3 +176,16 @@ def test_over_provision_pads_scale_up(server):
     padded, p_pad = _cluster(server, no_maintenance=True, over_provision=2)
     padded_units = padded.loop_logic().decision.units_needed["agentpool1"]
     assert padded_units == base_units + 2
+
+
+def test_failed_to_scale_names_an_example_pod(server):
+    """Unfulfillable selector groups notify with a concrete pod name
+    [R: reference Slack messages formatted the pod list]."""
+    nodes = [make_node_json("agentpool1", 0)]
+    orphan = make_pod_json("orphan", cpu_milli=100, pending=True,
+                           node_selector={"nosuch": "pool"})
+    server.state.load([orphan], nodes)
+    cluster, _ = _cluster(server, no_maintenance=True)
+    cluster.loop_logic()
+    evts = [e for e in cluster.notifier.events if e["kind"] == "failed_to_scale"]
+    assert evts and evts[0]["example_pod"] == "default/orphan"
