@@ -231,17 +231,22 @@ class Cluster:
         if is_rank0 and decision.unfulfillable_detail:
             # name one pending pod per unfulfillable selector group so
             # the notification is actionable (the lattice carries only
-            # the group hash)
-            examples: Dict[int, str] = {}
-            pending_rows = np.nonzero((snap.pod_flags & PodFlags.PENDING) != 0)[0]
-            for i in pending_rows:
-                h = int(snap.pod_sel_hash[i]) & ((1 << 64) - 1)
-                if h not in examples and snap.pod_refs:
-                    ns, name = snap.pod_refs[int(i)]
-                    examples[h] = f"{ns}/{name}"
-            for sel_hash, count in decision.unfulfillable_detail:
+            # the group hash). One vectorized scan per group, groups
+            # capped — never a per-pod Python loop (1M-pod snapshots).
+            pending_mask = (snap.pod_flags & PodFlags.PENDING) != 0
+            hashes_u = snap.pod_sel_hash.astype(np.uint64)
+            for sel_hash, count in decision.unfulfillable_detail[:20]:
+                example = ""
+                if snap.pod_refs:
+                    rows = np.nonzero(
+                        pending_mask & (hashes_u == np.uint64(sel_hash)))[0]
+                    if rows.size:
+                        ns, name = snap.pod_refs[int(rows[0])]
+                        example = f"{ns}/{name}"
                 self.notifier.notify_failed_to_scale(
-                    sel_hash, count, example_pod=examples.get(sel_hash, ""))
+                    sel_hash, count, example_pod=example)
+            for sel_hash, count in decision.unfulfillable_detail[20:]:
+                self.notifier.notify_failed_to_scale(sel_hash, count)
 
         # ---- scale-down ------------------------------------------------
         suppress = decision.wants_scale_up() and not self.opts.scale_down_during_scale_up
