@@ -16,22 +16,25 @@ from mi355x_autoscaler.snapshot.encode import encode_snapshot
 WORLD = 2
 
 
-def _snap_for_rank(rank, skewed=True):
+def _snap_for_rank(rank, skewed=True, world=WORLD):
     pods_json, nodes_json = burst_trace(400, n_nodes_per_pool=(4,), seed=8,
                                         bound_pods_per_node=2)
     if skewed:
-        # rank 0 gets 90% of the pods
+        # rank 0 gets ~everything; the tail splits over the others
         cut = int(len(pods_json) * 0.9)
-        pods_json = pods_json[:cut] if rank == 0 else pods_json[cut:]
+        if rank == 0:
+            pods_json = pods_json[:cut]
+        else:
+            pods_json = pods_json[cut:][rank - 1 :: world - 1]
     else:
-        pods_json = pods_json[rank::WORLD]
+        pods_json = pods_json[rank::world]
     pods, nodes = wrap(pods_json, nodes_json)
     return encode_snapshot(pods, nodes, group_nodes_into_pools(nodes))
 
 
-def _worker(rank, q):
-    os.environ.update(WORLD_SIZE=str(WORLD), RANK=str(rank), LOCAL_RANK=str(rank),
-                      MASTER_ADDR="127.0.0.1", MASTER_PORT="29631")
+def _worker(rank, q, world=WORLD, port="29631"):
+    os.environ.update(WORLD_SIZE=str(world), RANK=str(rank), LOCAL_RANK=str(rank),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=port)
     import torch.distributed as dist
 
     from mi355x_autoscaler.parallel.engine_sharded import ShardedEngine
@@ -39,7 +42,7 @@ def _worker(rank, q):
 
     try:
         engine = ShardedEngine(CpuEngine())
-        snap = _snap_for_rank(rank)
+        snap = _snap_for_rank(rank, world=world)
         before = snap.num_pods
         d_before = engine.decide(snap)
         snap2, did = rebalance_shard(snap, skew_threshold=1.2)
@@ -53,20 +56,25 @@ def _worker(rank, q):
         dist.destroy_process_group()
 
 
-def test_rebalance_evens_shards_and_preserves_decision():
+def _run_world(world, port):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_worker, args=(r, q)) for r in range(WORLD)]
+    procs = [ctx.Process(target=_worker, args=(r, q, world, port))
+             for r in range(world)]
     for p in procs:
         p.start()
     res = {}
-    for _ in range(WORLD):
+    for _ in range(world):
         rank, blob = q.get(timeout=180)
         res[rank] = pickle.loads(blob)
     for p in procs:
         p.join(timeout=30)
         assert p.exitcode == 0
+    return res
 
+
+def test_rebalance_evens_shards_and_preserves_decision():
+    res = _run_world(WORLD, "29631")
     assert res[0]["did"] and res[1]["did"]
     total = res[0]["before"] + res[1]["before"]
     assert res[0]["after"] + res[1]["after"] == total
@@ -74,6 +82,18 @@ def test_rebalance_evens_shards_and_preserves_decision():
     # ownership-invariance of the global decision
     assert res[0]["units_b"] == res[0]["units_a"] == res[1]["units_a"]
     assert res[0]["pe_b"] == res[0]["pe_a"]
+
+
+def test_rebalance_world4():
+    """Four-rank all_to_all redistribution: balanced shards, conserved
+    pod count, decision unchanged on every rank."""
+    res = _run_world(4, "29673")
+    total = sum(res[r]["before"] for r in range(4))
+    after = [res[r]["after"] for r in range(4)]
+    assert sum(after) == total
+    assert max(after) - min(after) <= 2
+    for r in range(4):
+        assert res[r]["units_a"] == res[0]["units_b"]
 
 
 def test_rebalance_noop_when_balanced():
