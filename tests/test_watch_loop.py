@@ -160,3 +160,31 @@ def test_remap_nodes_unit():
     assert pn[2] == 1 and (fl[2] & PodFlags.BOUND)
     rows = cache.drain_dirty()
     assert set(rows.tolist()) >= {0, 1, 2}
+
+
+def test_pdb_set_change_reprimes(server):
+    """A changed PDB set invalidates the cached pod_pdb column: the
+    watch cache re-primes and drain scoring honors the new budget."""
+    nodes = [make_node_json("agentpool1", i) for i in range(3)]
+    busy = make_pod_json("busy", cpu_milli=1500,
+                         node_name=nodes[0]["metadata"]["name"])
+    guarded = make_pod_json("g", cpu_milli=100,
+                            node_name=nodes[1]["metadata"]["name"],
+                            labels={"app": "guarded"})
+    server.state.load([busy, guarded], nodes)
+    cluster, provider = _cluster(server, no_scale=True, max_drains_per_cycle=10)
+
+    res1 = cluster.loop_logic()
+    cache1 = cluster._watch_state["cache"]
+    # both idle-ish nodes drained (no PDB yet)
+    assert nodes[1]["metadata"]["name"] in res1.drained_nodes
+
+    # re-load state with a zero-budget PDB guarding the pod
+    server.state.load([busy, guarded], nodes)
+    server.state.pdbs = {"guard": {
+        "metadata": {"name": "guard"},
+        "spec": {"selector": {"matchLabels": {"app": "guarded"}}},
+        "status": {"disruptionsAllowed": 0}}}
+    res2 = cluster.loop_logic()
+    assert cluster._watch_state["cache"] is not cache1  # re-primed
+    assert nodes[1]["metadata"]["name"] not in res2.drained_nodes
