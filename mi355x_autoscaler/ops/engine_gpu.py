@@ -1,5 +1,7 @@
 """GPU simulation engine: the HIP kernel pipeline on a device-resident
-cluster snapshot.
+cluster snapshot — the MI355X-native form of the reference's scale-up
+and scale-down hot loops [R: autoscaler/cluster.py · fulfill_pending +
+maintain — reconstructed, mount empty; see SURVEY provenance].
 
 decide() pipeline — every stage on-device, ONE H2D copy in, ONE small
 D2H readback out:

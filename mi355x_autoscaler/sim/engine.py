@@ -28,8 +28,10 @@ class SimulationEngine:
 
 
 class CpuEngine(SimulationEngine):
-    """Oracle-backed engine — the reference-equivalent slow path, and
-    the stand-in 'published number' per BASELINE.md."""
+    """Oracle-backed engine — the reference-equivalent slow path
+    [R: autoscaler/cluster.py · fulfill_pending/maintain semantics —
+    reconstructed], and the stand-in 'published number' per
+    BASELINE.md."""
 
     name = "cpu"
 
