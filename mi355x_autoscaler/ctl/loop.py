@@ -294,10 +294,17 @@ class Cluster:
             logger.warning("watch stream dropped; re-priming from a full relist")
             ws = self._watch_state = None
         if ws is not None and ws.get("pdb_fp") != pdb_fp:
-            # PDB selectors feed the cached pod_pdb column; a changed
-            # set invalidates it (rare — full re-prime)
-            logger.info("PDB set changed; re-priming the watch cache")
-            ws = self._watch_state = None
+            # PDB selectors feed the cached pod_pdb column. Re-derive
+            # it in place from the cached pod LABEL pair hashes; fall
+            # back to a full re-prime only when an exact rematch is
+            # impossible (labels wider than the lattice, old cache).
+            n = ws["cache"].reassign_pdbs([dict(p.match_labels) for p in pdbs])
+            if n is None:
+                logger.info("PDB set changed; re-priming the watch cache")
+                ws = self._watch_state = None
+            else:
+                logger.info("PDB set changed; reassigned %d pod rows in place", n)
+                ws["pdb_fp"] = pdb_fp
         if ws is not None and ws["node_names"] != node_names:
             # node set changed (scale-up materialized / node deleted):
             # remap the cached pod->node indices in place — NO pod

@@ -40,6 +40,9 @@ constexpr int NRES = 4;
 constexpr int MAXSEL = 32;
 // int64 view of uint64 max — mirrors encode.py::SELECTOR_POISON
 constexpr int64_t SELECTOR_POISON = -1;
+// pod LABEL pair width — mirrors encode.py::MAX_POD_LABEL_PAIRS
+// (host-only column: PDB re-matching in the watch cache)
+constexpr int MAXPODLBL = 16;
 constexpr int64_t MIB = 1024 * 1024;
 
 constexpr uint16_t PF_PENDING = 1 << 0;
@@ -459,6 +462,8 @@ struct Columns {
   std::vector<int32_t> pod_sel_count;
   std::vector<int32_t> pod_node;
   std::vector<int32_t> pod_pdb;
+  std::vector<int64_t> pod_label_pairs;
+  std::vector<int32_t> pod_label_count;
   std::vector<std::string> uids, ns, names;
 
   void reserve(size_t n) {
@@ -469,6 +474,8 @@ struct Columns {
     pod_sel_count.reserve(n);
     pod_node.reserve(n);
     pod_pdb.reserve(n);
+    pod_label_pairs.reserve(n * MAXPODLBL);
+    pod_label_count.reserve(n);
     uids.reserve(n);
     ns.reserve(n);
     names.reserve(n);
@@ -571,10 +578,27 @@ void encode_one(const JValue* pod, const Ctx& ctx, Columns& out) {
   out.pod_node.push_back(nidx);
   out.pod_flags.push_back(f);
 
+  // pod LABEL pair hashes (host-only: PDB re-matching in the watch
+  // cache) — sorted uint64 FNV pairs, first MAXPODLBL stored, RAW
+  // count kept so >width rows are detectable (mirrors encode.py)
+  const JValue* labels = get(meta, "labels");
+  {
+    std::vector<uint64_t> lh;
+    if (labels && labels->kind == JValue::Obj) {
+      for (auto& m : labels->members)
+        if (m.second->kind == JValue::Str)
+          lh.push_back(hash_label_pair(m.first, m.second->str));
+    }
+    std::sort(lh.begin(), lh.end());
+    const int lcnt = (int)lh.size();
+    for (int q = 0; q < MAXPODLBL; ++q)
+      out.pod_label_pairs.push_back(q < lcnt ? (int64_t)lh[q] : 0);
+    out.pod_label_count.push_back((int32_t)lcnt);
+  }
+
   // PDB assignment: first PDB whose matchLabels are all present
   int32_t pdb = -1;
   if (!ctx.pdbs.empty()) {
-    const JValue* labels = get(meta, "labels");
     for (size_t d = 0; d < ctx.pdbs.size(); ++d) {
       if (ctx.pdbs[d].empty()) continue;
       bool all = true;
@@ -748,6 +772,8 @@ pybind11::dict encode_pods_json(pybind11::bytes body, pybind11::dict node_index,
   result["pod_sel_count"] = make([](Columns& c) -> auto& { return c.pod_sel_count; }, 1, int32_t{});
   result["pod_node"] = make([](Columns& c) -> auto& { return c.pod_node; }, 1, int32_t{});
   result["pod_pdb"] = make([](Columns& c) -> auto& { return c.pod_pdb; }, 1, int32_t{});
+  result["pod_label_pairs"] = make([](Columns& c) -> auto& { return c.pod_label_pairs; }, MAXPODLBL, int64_t{});
+  result["pod_label_count"] = make([](Columns& c) -> auto& { return c.pod_label_count; }, 1, int32_t{});
   PyObject* uids = PyList_New((Py_ssize_t)P);
   PyObject* refs = PyList_New((Py_ssize_t)P);
   Py_ssize_t w = 0;
@@ -906,6 +932,8 @@ pybind11::dict decode_watch_events(pybind11::bytes body,
       app(out.pod_sel_count, sh.cols.pod_sel_count);
       app(out.pod_node, sh.cols.pod_node);
       app(out.pod_pdb, sh.cols.pod_pdb);
+      app(out.pod_label_pairs, sh.cols.pod_label_pairs);
+      app(out.pod_label_count, sh.cols.pod_label_count);
       for (auto& u : sh.cols.uids) out.uids.push_back(std::move(u));
       for (auto& x : sh.cols.ns) out.ns.push_back(std::move(x));
       for (auto& x : sh.cols.names) out.names.push_back(std::move(x));
@@ -930,6 +958,8 @@ pybind11::dict decode_watch_events(pybind11::bytes body,
   result["pod_sel_count"] = make(out.pod_sel_count, 1, int32_t{});
   result["pod_node"] = make(out.pod_node, 1, int32_t{});
   result["pod_pdb"] = make(out.pod_pdb, 1, int32_t{});
+  result["pod_label_pairs"] = make(out.pod_label_pairs, MAXPODLBL, int64_t{});
+  result["pod_label_count"] = make(out.pod_label_count, 1, int32_t{});
   PyObject* uids = PyList_New(P);
   PyObject* refs = PyList_New(P);
   for (ssize_t i = 0; i < P; ++i) {

@@ -58,6 +58,7 @@ def shard_snapshot(snap: ClusterSnapshot, rank: int, world: int) -> ClusterSnaps
     if world <= 1:
         return snap
     idx = shard_indices(snap.pod_uids, rank, world)
+    has_lbl = snap.pod_label_pairs.shape[0] == snap.num_pods
     sub = dataclasses.replace(
         snap,
         pod_req=snap.pod_req[idx],
@@ -67,6 +68,10 @@ def shard_snapshot(snap: ClusterSnapshot, rank: int, world: int) -> ClusterSnaps
         pod_sel_count=snap.pod_sel_count[idx],
         pod_node=snap.pod_node[idx],
         pod_pdb=snap.pod_pdb[idx],
+        pod_label_pairs=(snap.pod_label_pairs[idx] if has_lbl
+                         else snap.pod_label_pairs),
+        pod_label_count=(snap.pod_label_count[idx] if has_lbl
+                         else snap.pod_label_count),
         pod_uids=[snap.pod_uids[i] for i in idx],
         pod_refs=[snap.pod_refs[i] for i in idx],
     )

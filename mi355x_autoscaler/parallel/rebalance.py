@@ -27,7 +27,9 @@ from . import dist as pdist
 
 # packed row: req(4) + flags + sel_hash + sel_pairs(8) + sel_count +
 # node + pdb = 16 int64 columns (uids/refs ride in a side channel)
-_ROW64 = 4 + 1 + 1 + MAX_SELECTOR_PAIRS + 1 + 1 + 1
+_SEL0 = 6                              # first selector-pair column
+_TAIL = _SEL0 + MAX_SELECTOR_PAIRS     # sel_count, node, pdb follow
+_ROW64 = _TAIL + 3
 
 
 def pack_rows(snap: ClusterSnapshot, idx: np.ndarray) -> np.ndarray:
@@ -35,10 +37,10 @@ def pack_rows(snap: ClusterSnapshot, idx: np.ndarray) -> np.ndarray:
     rows[:, 0:4] = snap.pod_req[idx]
     rows[:, 4] = snap.pod_flags[idx]
     rows[:, 5] = snap.pod_sel_hash[idx]
-    rows[:, 6 : 6 + MAX_SELECTOR_PAIRS] = snap.pod_sel_pairs[idx]
-    rows[:, 14] = snap.pod_sel_count[idx]
-    rows[:, 15] = snap.pod_node[idx]
-    rows[:, 16] = snap.pod_pdb[idx]  # pdb list is replicated across ranks
+    rows[:, _SEL0:_TAIL] = snap.pod_sel_pairs[idx]
+    rows[:, _TAIL] = snap.pod_sel_count[idx]
+    rows[:, _TAIL + 1] = snap.pod_node[idx]
+    rows[:, _TAIL + 2] = snap.pod_pdb[idx]  # pdb list replicated across ranks
     return rows
 
 
@@ -47,10 +49,10 @@ def unpack_rows(rows: np.ndarray):
         pod_req=rows[:, 0:4].astype(np.int32),
         pod_flags=rows[:, 4].astype(np.int32),
         pod_sel_hash=rows[:, 5].copy(),
-        pod_sel_pairs=rows[:, 6 : 6 + MAX_SELECTOR_PAIRS].copy(),
-        pod_sel_count=rows[:, 14].astype(np.int32),
-        pod_node=rows[:, 15].astype(np.int32),
-        pod_pdb=rows[:, 16].astype(np.int32),
+        pod_sel_pairs=rows[:, _SEL0:_TAIL].copy(),
+        pod_sel_count=rows[:, _TAIL].astype(np.int32),
+        pod_node=rows[:, _TAIL + 1].astype(np.int32),
+        pod_pdb=rows[:, _TAIL + 2].astype(np.int32),
     )
 
 
@@ -151,6 +153,11 @@ def rebalance_shard(
         pod_sel_count=np.concatenate([snap.pod_sel_count[:keep], cols["pod_sel_count"]]),
         pod_node=np.concatenate([snap.pod_node[:keep], cols["pod_node"]]),
         pod_pdb=np.concatenate([snap.pod_pdb[:keep], cols["pod_pdb"]]),
+        # label pairs are NOT shipped (pod_pdb already travels; the
+        # delta cache is never primed from a rebalanced shard) — mark
+        # the column absent instead of leaving a wrong-length array
+        pod_label_pairs=snap.pod_label_pairs[:0],
+        pod_label_count=snap.pod_label_count[:0],
         pod_uids=snap.pod_uids[:keep] + my_uids,
         pod_refs=snap.pod_refs[:keep] + my_refs,
     )

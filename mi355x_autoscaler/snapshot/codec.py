@@ -60,6 +60,8 @@ def encode_snapshot_fast(
         pod_sel_count=cols["pod_sel_count"],
         pod_node=cols["pod_node"],
         pod_pdb=cols["pod_pdb"],
+        pod_label_pairs=cols["pod_label_pairs"],
+        pod_label_count=cols["pod_label_count"],
         node_alloc=base.node_alloc,
         node_free=base.node_alloc.copy(),
         node_labels=base.node_labels,

@@ -20,13 +20,20 @@ from typing import Dict, List, Optional, Set, Tuple
 import numpy as np
 
 from ..models.kube import KubePod
-from .encode import MAX_SELECTOR_PAIRS, ClusterSnapshot, PodFlags, encode_snapshot
+from .encode import (
+    MAX_POD_LABEL_PAIRS,
+    MAX_SELECTOR_PAIRS,
+    ClusterSnapshot,
+    PodFlags,
+    encode_snapshot,
+)
 
 _GROW = 1.5
 _COMPACT_DEAD_FRACTION = 0.5
 
 _POD_COLS = ("pod_req", "pod_flags", "pod_sel_hash", "pod_sel_pairs",
-             "pod_sel_count", "pod_node", "pod_pdb")
+             "pod_sel_count", "pod_node", "pod_pdb",
+             "pod_label_pairs", "pod_label_count")
 
 
 class DeltaSnapshotCache:
@@ -48,6 +55,7 @@ class DeltaSnapshotCache:
         self.pod_refs: List[Tuple[str, str]] = []
         self.node_index: Dict[str, int] = {}
         self.pdb_selectors: List[Dict[str, str]] = []
+        self.has_label_pairs = False  # set by prime()/load()
 
     # ------------------------------------------------------------------
     def prime(self, snap: ClusterSnapshot, node_index: Dict[str, int],
@@ -56,8 +64,14 @@ class DeltaSnapshotCache:
         self.size = snap.num_pods
         self.cap = max(64, int(self.size * _GROW))
         self.arrays = {}
+        # pod_label_* may be absent on the priming snapshot (synthetic
+        # SoA traces, shards): cache zeros and refuse reassign_pdbs.
+        self.has_label_pairs = (
+            snap.pod_label_pairs.shape[0] == self.size)
         for col in _POD_COLS:
             src = getattr(snap, col)
+            if src.shape[0] != self.size:  # absent optional column
+                src = np.zeros((self.size,) + src.shape[1:], src.dtype)
             shape = (self.cap,) + src.shape[1:]
             a = np.zeros(shape, src.dtype)
             a[: self.size] = src
@@ -81,6 +95,7 @@ class DeltaSnapshotCache:
         req = np.array(quantize_request(pod.resources), np.int32)
         pairs, cnt, combined = _hash_pairs(
             pod.selectors, MAX_SELECTOR_PAIRS, poison_overflow=True)
+        lpairs, lcnt, _ = _hash_pairs(pod.labels, MAX_POD_LABEL_PAIRS)
         f = 0
         if pod.is_pending_unschedulable:
             f |= PodFlags.PENDING
@@ -107,10 +122,10 @@ class DeltaSnapshotCache:
             if sel and all(pod.labels.get(k) == v for k, v in sel.items()):
                 pdb = d
                 break
-        return req, pairs, min(cnt, MAX_SELECTOR_PAIRS), combined, f, node, pdb
+        return req, pairs, min(cnt, MAX_SELECTOR_PAIRS), combined, f, node, pdb, lpairs, lcnt
 
     def _write_row(self, row: int, pod: KubePod):
-        req, pairs, cnt, combined, f, node, pdb = self._encode_row(pod)
+        req, pairs, cnt, combined, f, node, pdb, lpairs, lcnt = self._encode_row(pod)
         a = self.arrays
         a["pod_req"][row] = req
         a["pod_sel_pairs"][row] = pairs
@@ -119,6 +134,8 @@ class DeltaSnapshotCache:
         a["pod_flags"][row] = f
         a["pod_node"][row] = node
         a["pod_pdb"][row] = pdb
+        a["pod_label_pairs"][row] = lpairs
+        a["pod_label_count"][row] = lcnt
         self.pod_uids[row] = pod.uid
         self.pod_refs[row] = (pod.namespace, pod.name)
         self.dirty.add(row)
@@ -212,6 +229,9 @@ class DeltaSnapshotCache:
                 a["pod_sel_count"][rows] = decoded["pod_sel_count"][sel]
                 a["pod_node"][rows] = decoded["pod_node"][sel]
                 a["pod_pdb"][rows] = decoded["pod_pdb"][sel]
+                if "pod_label_pairs" in decoded:
+                    a["pod_label_pairs"][rows] = decoded["pod_label_pairs"][sel]
+                    a["pod_label_count"][rows] = decoded["pod_label_count"][sel]
                 self.dirty.update(rows.tolist())
             return n
         return self._apply_batch_py(decoded)
@@ -263,6 +283,9 @@ class DeltaSnapshotCache:
             a["pod_sel_count"][dst] = decoded["pod_sel_count"][sel]
             a["pod_node"][dst] = decoded["pod_node"][sel]
             a["pod_pdb"][dst] = decoded["pod_pdb"][sel]
+            if "pod_label_pairs" in decoded:
+                a["pod_label_pairs"][dst] = decoded["pod_label_pairs"][sel]
+                a["pod_label_count"][dst] = decoded["pod_label_count"][sel]
         return n
 
     def drain_dirty(self) -> np.ndarray:
@@ -310,6 +333,52 @@ class DeltaSnapshotCache:
         self.node_index = dict(new_node_index)
         return nchanged
 
+    # ------------------------------------------------------------------
+    def reassign_pdbs(self, pdb_selectors: List[Dict[str, str]]) -> Optional[int]:
+        """The PDB SET changed: re-derive the pod->PDB column in place
+        from the cached pod LABEL pair hashes (subset match, same FNV
+        convention as selectors) instead of re-priming from a full pod
+        relist. Returns rows changed, or None when an exact in-place
+        rematch is impossible (no label data cached, or a pod/selector
+        wider than the MAX_POD_LABEL_PAIRS lattice) — the caller falls
+        back to the re-prime."""
+        if not self.has_label_pairs:
+            return None
+        from .encode import _hash_pairs
+
+        n = self.size
+        lp = self.arrays["pod_label_pairs"][:n]
+        lc = self.arrays["pod_label_count"][:n]
+        live = np.array([bool(u) for u in self.pod_uids[:n]], bool)
+        if live.any() and (lc[live] > MAX_POD_LABEL_PAIRS).any():
+            return None  # truncated label sets: can't rematch exactly
+        new_pdb = np.full(n, -1, np.int32)
+        unassigned = live.copy()
+        for d, sel in enumerate(pdb_selectors):
+            if not sel:
+                continue  # empty matchLabels never matches (encode parity)
+            pairs, cnt, _ = _hash_pairs(sel, MAX_POD_LABEL_PAIRS)
+            if cnt > MAX_POD_LABEL_PAIRS:
+                return None
+            m = unassigned.copy()
+            for h in pairs[:cnt]:
+                m &= (lp == h).any(axis=1)
+                if not m.any():
+                    break
+            new_pdb[m] = d
+            unassigned &= ~m
+        old = self.arrays["pod_pdb"][:n]
+        # tombstones keep their -1 (new_pdb is -1 there: not live)
+        changed = np.nonzero(new_pdb != old)[0]
+        if changed.size:
+            old[changed] = new_pdb[changed]
+            if changed.size > max(1024, n // 8):
+                self.dirty_all = True
+            else:
+                self.dirty.update(int(r) for r in changed)
+        self.pdb_selectors = [dict(s) for s in pdb_selectors]
+        return int(changed.size)
+
     @property
     def dead_fraction(self) -> float:
         return len(self.free_rows) / max(1, self.size)
@@ -350,6 +419,8 @@ class DeltaSnapshotCache:
             pod_sel_count=self.arrays["pod_sel_count"][: self.size],
             pod_node=self.arrays["pod_node"][: self.size],
             pod_pdb=self.arrays["pod_pdb"][: self.size],
+            pod_label_pairs=self.arrays["pod_label_pairs"][: self.size],
+            pod_label_count=self.arrays["pod_label_count"][: self.size],
             pod_uids=self.pod_uids[: self.size],
             pod_refs=self.pod_refs[: self.size],
         )
@@ -375,8 +446,17 @@ class DeltaSnapshotCache:
         self.size = int(data["size"])
         self.cap = max(64, int(self.size * _GROW))
         self.arrays = {}
+        from .encode import MAX_POD_LABEL_PAIRS as _LW
+
+        _DEFAULTS = {"pod_label_pairs": ((_LW,), np.int64),
+                     "pod_label_count": ((), np.int32)}
+        self.has_label_pairs = "pod_label_pairs" in data
         for col in _POD_COLS:
-            src = data[col]
+            if col in data:
+                src = data[col]
+            else:  # checkpoint predates the column: zeros, reassign off
+                tail, dt = _DEFAULTS[col]
+                src = np.zeros((self.size,) + tail, dt)
             a = np.zeros((self.cap,) + src.shape[1:], src.dtype)
             a[: self.size] = src
             if col in ("pod_node", "pod_pdb"):

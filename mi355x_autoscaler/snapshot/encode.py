@@ -49,6 +49,14 @@ MIB = 1024 * 1024
 MAX_SELECTOR_PAIRS = 32
 MAX_LABEL_PAIRS = 32
 
+# Pod LABEL pair hashes (host-only column, never uploaded): lets the
+# watch cache re-derive the pod->PDB mapping in place when the PDB set
+# changes (PDB matchLabels ⊆ pod labels, same FNV-pair subset
+# convention as selectors). Pods with more labels than this width are
+# flagged via pod_label_count > MAX_POD_LABEL_PAIRS and force the old
+# full-re-prime path (exactness over speed on the rare row).
+MAX_POD_LABEL_PAIRS = 16
+
 # uint64 0xFFFF…F as int64: sorts LAST in the unsigned pair order, and no
 # real FNV-1a label hash equals it in practice, so a selector carrying it
 # fails every subset walk (host selector_subset_match and device
@@ -170,6 +178,14 @@ class ClusterSnapshot:
 
     pdb_allowed: np.ndarray = field(default_factory=lambda: np.zeros(0, np.int32))
 
+    # host-only pod LABEL pair hashes (PDB re-matching in the watch
+    # cache; NOT part of the device upload). Zero-length when the
+    # producer doesn't track them (synthetic SoA traces, shards).
+    pod_label_pairs: np.ndarray = field(
+        default_factory=lambda: np.zeros((0, MAX_POD_LABEL_PAIRS), np.int64))
+    pod_label_count: np.ndarray = field(
+        default_factory=lambda: np.zeros(0, np.int32))
+
     # bookkeeping for decoding decisions back to objects
     pod_uids: List[str] = field(default_factory=list)
     pod_refs: List[Tuple[str, str]] = field(default_factory=list)  # (namespace, name)
@@ -238,6 +254,8 @@ def encode_snapshot(
     pod_sel_hash = np.zeros(P, np.int64)
     pod_sel_pairs = np.zeros((P, MAX_SELECTOR_PAIRS), np.int64)
     pod_sel_count = np.zeros(P, np.int32)
+    pod_label_pairs = np.zeros((P, MAX_POD_LABEL_PAIRS), np.int64)
+    pod_label_count = np.zeros(P, np.int32)
     pod_node = np.full(P, -1, np.int32)
     pod_pdb = np.full(P, -1, np.int32)
 
@@ -271,6 +289,9 @@ def encode_snapshot(
         pod_req[i] = quantize_request(p.resources)
         pairs, cnt, combined = _hash_pairs(
             p.selectors, MAX_SELECTOR_PAIRS, poison_overflow=True)
+        lpairs, lcnt, _ = _hash_pairs(p.labels, MAX_POD_LABEL_PAIRS)
+        pod_label_pairs[i] = lpairs
+        pod_label_count[i] = lcnt  # RAW count: >width flags overflow
         pod_sel_pairs[i] = pairs
         pod_sel_count[i] = min(cnt, MAX_SELECTOR_PAIRS)
         pod_sel_hash[i] = combined
@@ -336,6 +357,8 @@ def encode_snapshot(
         pool_label_count=pool_label_count,
         pool_names=pool_names,
         pdb_allowed=pdb_allowed,
+        pod_label_pairs=pod_label_pairs,
+        pod_label_count=pod_label_count,
         pod_uids=[p.uid for p in pods],
         pod_refs=[(p.namespace, p.name) for p in pods],
         node_names=[n.name for n in nodes],

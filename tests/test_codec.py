@@ -15,7 +15,8 @@ from mi355x_autoscaler.sim.synth import burst_trace, make_node_json, make_pod_js
 pytestmark = pytest.mark.skipif(not codec_available(), reason="native codec not built")
 
 POD_COLS = ("pod_req", "pod_flags", "pod_sel_hash", "pod_sel_pairs",
-            "pod_sel_count", "pod_node", "pod_pdb")
+            "pod_sel_count", "pod_node", "pod_pdb",
+            "pod_label_pairs", "pod_label_count")
 
 
 def _compare(pods_json, nodes_json, pdbs=None):

@@ -133,3 +133,60 @@ def test_delta_checkpoint_resume(tmp_path):
     snap = fresh.build_snapshot(base)
     snap_full = _full(pods + more, nodes)
     _assert_equivalent_decisions(snap, snap_full)
+
+
+def test_reassign_pdbs_matches_fresh_encode():
+    """PDB-set change: the in-place rematch from cached label pairs
+    must equal a fresh encode with the new PDBs, only changed rows go
+    dirty, and overflow label sets refuse (caller re-primes)."""
+    from mi355x_autoscaler.kube.objects import PodDisruptionBudget
+    from mi355x_autoscaler.snapshot.encode import MAX_POD_LABEL_PAIRS
+
+    nodes = [KubeNode(make_node_json("agentpool1", 0))]
+    pods_json = [
+        make_pod_json("a", cpu_milli=100, pending=True, labels={"app": "x"}),
+        make_pod_json("b", cpu_milli=100, pending=True,
+                      labels={"app": "y", "tier": "db"}),
+        make_pod_json("c", cpu_milli=100, pending=True),  # no labels
+    ]
+    base, pools = _base(nodes)
+    cache = DeltaSnapshotCache()
+    pdbs0 = [PodDisruptionBudget(name="p0", match_labels={"app": "x"},
+                                 disruptions_allowed=1)]
+    snap0 = encode_snapshot([KubePod(p) for p in pods_json], nodes, pools, pdbs0)
+    cache.prime(snap0, {nodes[0].name: 0}, [dict(p.match_labels) for p in pdbs0])
+    cache.drain_dirty()
+
+    # new PDB set: p0 now guards app=y+tier=db; p1 guards app=x
+    pdbs1 = [
+        PodDisruptionBudget(name="n0", match_labels={"app": "y", "tier": "db"},
+                            disruptions_allowed=2),
+        PodDisruptionBudget(name="n1", match_labels={"app": "x"},
+                            disruptions_allowed=0),
+    ]
+    n = cache.reassign_pdbs([dict(p.match_labels) for p in pdbs1])
+    assert n == 2  # rows a (0 -> 1) and b (-1 -> 0) changed
+    fresh = encode_snapshot([KubePod(p) for p in pods_json], nodes, pools, pdbs1)
+    rows = {u: i for i, u in enumerate(cache.pod_uids[: cache.size])}
+    for i, u in enumerate(fresh.pod_uids):
+        assert cache.arrays["pod_pdb"][rows[u]] == fresh.pod_pdb[i], u
+    dirty = set(cache.drain_dirty().tolist())
+    assert len(dirty) == 2
+
+    # label overflow: >MAX_POD_LABEL_PAIRS labels on a live pod
+    wide = make_pod_json("wide", cpu_milli=50, pending=True,
+                         labels={f"l{i:02d}": "v"
+                                 for i in range(MAX_POD_LABEL_PAIRS + 1)})
+    cache.apply_event({"type": "ADDED", "object": wide})
+    assert cache.reassign_pdbs([{"l00": "v"}]) is None
+
+
+def test_reassign_pdbs_refuses_without_label_data():
+    from mi355x_autoscaler.snapshot.delta import DeltaSnapshotCache
+    from mi355x_autoscaler.sim.synth_soa import synth_burst_snapshot
+
+    snap = synth_burst_snapshot(100, n_nodes_per_pool=(2,), seed=0)
+    cache = DeltaSnapshotCache()
+    cache.prime(snap, {})
+    assert not cache.has_label_pairs
+    assert cache.reassign_pdbs([{"app": "x"}]) is None

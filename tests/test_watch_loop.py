@@ -162,9 +162,10 @@ def test_remap_nodes_unit():
     assert set(rows.tolist()) >= {0, 1, 2}
 
 
-def test_pdb_set_change_reprimes(server):
-    """A changed PDB set invalidates the cached pod_pdb column: the
-    watch cache re-primes and drain scoring honors the new budget."""
+def test_pdb_set_change_reassigns_in_place(server):
+    """A changed PDB set re-derives the cached pod_pdb column from the
+    cached pod LABEL pair hashes — NO re-prime (pod relist) — and
+    drain scoring honors the new budget."""
     nodes = [make_node_json("agentpool1", i) for i in range(3)]
     busy = make_pod_json("busy", cpu_milli=1500,
                          node_name=nodes[0]["metadata"]["name"])
@@ -174,17 +175,18 @@ def test_pdb_set_change_reprimes(server):
     server.state.load([busy, guarded], nodes)
     cluster, provider = _cluster(server, no_scale=True, max_drains_per_cycle=10)
 
-    res1 = cluster.loop_logic()
-    cache1 = cluster._watch_state["cache"]
-    # both idle-ish nodes drained (no PDB yet)
-    assert nodes[1]["metadata"]["name"] in res1.drained_nodes
-
-    # re-load state with a zero-budget PDB guarding the pod
-    server.state.load([busy, guarded], nodes)
+    # prime WITH a zero-budget PDB guarding the pod: no drain
     server.state.pdbs = {"guard": {
         "metadata": {"name": "guard"},
         "spec": {"selector": {"matchLabels": {"app": "guarded"}}},
         "status": {"disruptionsAllowed": 0}}}
+    res1 = cluster.loop_logic()
+    cache1 = cluster._watch_state["cache"]
+    assert nodes[1]["metadata"]["name"] not in res1.drained_nodes
+
+    # PDB set changes (removed): the pod un-guards IN PLACE — same
+    # cache object, no pod relist — and the node drains next cycle
+    server.state.pdbs = {}
     res2 = cluster.loop_logic()
-    assert cluster._watch_state["cache"] is not cache1  # re-primed
-    assert nodes[1]["metadata"]["name"] not in res2.drained_nodes
+    assert cluster._watch_state["cache"] is cache1  # reassigned in place
+    assert nodes[1]["metadata"]["name"] in res2.drained_nodes
