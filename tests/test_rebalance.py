@@ -102,3 +102,28 @@ def test_rebalance_noop_when_balanced():
     snap = _snap_for_rank(0, skewed=False)
     out, did = rebalance_shard(snap)  # world == 1 here
     assert not did and out is snap
+
+
+def test_pack_rows_roundtrip_wide_selectors():
+    """Regression: pack_rows once hardcoded 8-pair-era column offsets,
+    so sel_count/node/pdb stomped selector pairs 8-10 on the wire.
+    Round-trip a pod with 12 selector pairs and assert every column."""
+    from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools
+    from mi355x_autoscaler.models.kube import KubeNode, KubePod
+    from mi355x_autoscaler.parallel.rebalance import pack_rows, unpack_rows
+    from mi355x_autoscaler.sim.synth import make_node_json, make_pod_json
+    from mi355x_autoscaler.snapshot.encode import encode_snapshot
+
+    sel = {f"k{i:02d}": f"v{i}" for i in range(12)}
+    node = make_node_json("agentpool1", 0)
+    node["metadata"]["labels"].update(sel)
+    nodes = [KubeNode(node)]
+    pods = [KubePod(make_pod_json("w", cpu_milli=100, pending=True,
+                                  node_selector=sel))]
+    snap = encode_snapshot(pods, nodes, group_nodes_into_pools(nodes))
+    cols = unpack_rows(pack_rows(snap, np.arange(1)))
+    np.testing.assert_array_equal(cols["pod_sel_pairs"], snap.pod_sel_pairs)
+    assert cols["pod_sel_count"][0] == 12
+    np.testing.assert_array_equal(cols["pod_req"], snap.pod_req)
+    assert cols["pod_node"][0] == snap.pod_node[0]
+    assert cols["pod_pdb"][0] == snap.pod_pdb[0]
