@@ -157,7 +157,8 @@ def test_decode_watch_events_batch_matches_per_event():
     events = []
     for i in range(30):
         events.append({"type": "ADDED", "object": make_pod_json(
-            f"n{i}", cpu_milli=200 + i, mem_mib=64, pending=True)})
+            f"n{i}", cpu_milli=200 + i, mem_mib=64, pending=True,
+            labels={"app": f"w{i % 4}", "tier": "web"} if i % 3 else None)})
     events.append({"type": "MODIFIED", "object": make_pod_json(
         "p0", cpu_milli=999, node_name=nodes[0].name)})
     events.append({"type": "DELETED", "object": initial[1]})
