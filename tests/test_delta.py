@@ -190,3 +190,18 @@ def test_reassign_pdbs_refuses_without_label_data():
     cache.prime(snap, {})
     assert not cache.has_label_pairs
     assert cache.reassign_pdbs([{"app": "x"}]) is None
+
+
+def test_randomized_delta_fuzz():
+    """Randomized watch-event/PDB-change/compaction sequences converge
+    to the full re-encode, column-exact (scripts/fuzz_delta_cpu.py
+    runs the longer form; 200 iterations green at commit time)."""
+    import importlib.util
+    import os
+
+    spec = importlib.util.spec_from_file_location(
+        "fuzz_delta_cpu", os.path.join(os.path.dirname(__file__), "..",
+                                       "scripts", "fuzz_delta_cpu.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    mod.main(iters=8)
