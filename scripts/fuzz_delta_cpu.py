@@ -11,7 +11,14 @@ import sys
 
 import numpy as np
 
-ITERS = int(sys.argv[1]) if len(sys.argv) > 1 else 30
+def _argv_iters():
+    try:
+        return int(sys.argv[1])
+    except (IndexError, ValueError):  # imported under pytest etc.
+        return 30
+
+
+ITERS = _argv_iters()
 
 sys.path.insert(0, __file__.rsplit("/", 2)[0])
 
