@@ -78,8 +78,11 @@ class DeltaSnapshotCache:
             if col in ("pod_node", "pod_pdb"):
                 a[self.size :] = -1  # padding rows must be UNBOUND
             self.arrays[col] = a
-        self.pod_uids = list(snap.pod_uids) + [""] * (self.cap - self.size)
-        self.pod_refs = list(snap.pod_refs) + [("", "")] * (self.cap - self.size)
+        # pad by ACTUAL list length: snapshots without uid bookkeeping
+        # (synthetic SoA) would otherwise leave the lists shorter than
+        # cap and row writes past len() would raise
+        self.pod_uids = list(snap.pod_uids) + [""] * (self.cap - len(snap.pod_uids))
+        self.pod_refs = list(snap.pod_refs) + [("", "")] * (self.cap - len(snap.pod_refs))
         self.uid_row = {u: i for i, u in enumerate(snap.pod_uids) if u}
         self.free_rows = []
         self.dirty = set()
