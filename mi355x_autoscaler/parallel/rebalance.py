@@ -25,8 +25,8 @@ import torch
 from ..snapshot.encode import MAX_SELECTOR_PAIRS, ClusterSnapshot
 from . import dist as pdist
 
-# packed row: req(4) + flags + sel_hash + sel_pairs(8) + sel_count +
-# node + pdb = 16 int64 columns (uids/refs ride in a side channel)
+# packed row: req(4) + flags + sel_hash + sel_pairs(MAX_SELECTOR_PAIRS)
+# + sel_count + node + pdb int64 columns (uids/refs in a side channel)
 _SEL0 = 6                              # first selector-pair column
 _TAIL = _SEL0 + MAX_SELECTOR_PAIRS     # sel_count, node, pdb follow
 _ROW64 = _TAIL + 3
