@@ -269,7 +269,7 @@ class Cluster:
                  frozenset(n.labels.items()))
                 for n in nodes
             ),
-            tuple((p.name, p.disruptions_allowed,
+            tuple((p.namespace, p.name, p.disruptions_allowed,
                    frozenset(p.match_labels.items())) for p in pdbs),
         ))
         if self._base_cache is not None and self._base_cache[0] == fp:
@@ -288,7 +288,8 @@ class Cluster:
         from ..snapshot.delta import DeltaSnapshotCache
 
         node_names = tuple(n.name for n in nodes)
-        pdb_fp = tuple((p.name, frozenset(p.match_labels.items())) for p in pdbs)
+        pdb_fp = tuple((p.namespace, p.name,
+                        frozenset(p.match_labels.items())) for p in pdbs)
         ws = self._watch_state
         if ws is not None and ws["broken"]:
             logger.warning("watch stream dropped; re-priming from a full relist")
@@ -298,7 +299,8 @@ class Cluster:
             # it in place from the cached pod LABEL pair hashes; fall
             # back to a full re-prime only when an exact rematch is
             # impossible (labels wider than the lattice, old cache).
-            n = ws["cache"].reassign_pdbs([dict(p.match_labels) for p in pdbs])
+            n = ws["cache"].reassign_pdbs(
+                [(p.namespace, dict(p.match_labels)) for p in pdbs])
             if n is None:
                 logger.info("PDB set changed; re-priming the watch cache")
                 ws = self._watch_state = None
@@ -326,7 +328,7 @@ class Cluster:
             cache.prime(
                 snap,
                 {n.name: i for i, n in enumerate(nodes)},
-                [dict(p.match_labels) for p in pdbs],
+                [(p.namespace, dict(p.match_labels)) for p in pdbs],
             )
             q: _q.SimpleQueue = _q.SimpleQueue()
             th = threading.Thread(target=self._watch_pump, args=(q,), daemon=True)

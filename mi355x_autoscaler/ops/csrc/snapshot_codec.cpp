@@ -449,10 +449,37 @@ bool ieq_true(std::string_view s) {
 // -> column concatenation.
 // ---------------------------------------------------------------------------
 
+struct PdbMatcher {
+  std::string ns;  // a v1 PDB selects pods in ITS OWN namespace only
+  std::vector<std::pair<std::string, std::string>> labels;
+};
+
 struct Ctx {
   std::unordered_map<std::string, int32_t> node_idx;
-  std::vector<std::vector<std::pair<std::string, std::string>>> pdbs;
+  std::vector<PdbMatcher> pdbs;
 };
+
+// Parse the Python-side matcher list: entries are (namespace, labels)
+// tuples, or bare label dicts (legacy — namespace 'default'); mirrors
+// snapshot/delta.py::_norm_matchers.
+static void parse_pdb_matchers(pybind11::list pdb_selectors, Ctx& ctx) {
+  namespace py = pybind11;
+  for (auto sel : pdb_selectors) {
+    PdbMatcher m;
+    py::object labels_obj = py::reinterpret_borrow<py::object>(sel);
+    if (py::isinstance<py::tuple>(labels_obj) || py::isinstance<py::list>(labels_obj)) {
+      auto seq = labels_obj.cast<py::sequence>();
+      m.ns = py::cast<std::string>(seq[0]);
+      labels_obj = py::reinterpret_borrow<py::object>(seq[1]);
+    } else {
+      m.ns = "default";
+    }
+    for (auto kv : labels_obj.cast<py::dict>())
+      m.labels.emplace_back(py::cast<std::string>(kv.first),
+                            py::cast<std::string>(kv.second));
+    ctx.pdbs.push_back(std::move(m));
+  }
+}
 
 struct Columns {
   std::vector<int32_t> pod_req;
@@ -596,13 +623,17 @@ void encode_one(const JValue* pod, const Ctx& ctx, Columns& out) {
     out.pod_label_count.push_back((int32_t)lcnt);
   }
 
-  // PDB assignment: first PDB whose matchLabels are all present
+  std::string ns(get_str(meta, "namespace"));
+  if (ns.empty()) ns = "default";
+
+  // PDB assignment: first SAME-NAMESPACE PDB whose matchLabels are
+  // all present (v1 PDB semantics; mirrors encode.py/delta.py)
   int32_t pdb = -1;
   if (!ctx.pdbs.empty()) {
     for (size_t d = 0; d < ctx.pdbs.size(); ++d) {
-      if (ctx.pdbs[d].empty()) continue;
+      if (ctx.pdbs[d].labels.empty() || ctx.pdbs[d].ns != ns) continue;
       bool all = true;
-      for (auto& [k, v] : ctx.pdbs[d]) {
+      for (auto& [k, v] : ctx.pdbs[d].labels) {
         const JValue* lv = labels ? get(labels, k) : nullptr;
         if (!lv || lv->kind != JValue::Str || lv->str != v) {
           all = false;
@@ -618,8 +649,6 @@ void encode_one(const JValue* pod, const Ctx& ctx, Columns& out) {
   out.pod_pdb.push_back(pdb);
 
   out.uids.emplace_back(get_str(meta, "uid"));
-  std::string ns(get_str(meta, "namespace"));
-  if (ns.empty()) ns = "default";
   out.ns.push_back(std::move(ns));
   out.names.emplace_back(get_str(meta, "name"));
 }
@@ -701,12 +730,7 @@ pybind11::dict encode_pods_json(pybind11::bytes body, pybind11::dict node_index,
   Ctx ctx;
   for (auto item : node_index)
     ctx.node_idx[py::cast<std::string>(item.first)] = py::cast<int32_t>(item.second);
-  for (auto sel : pdb_selectors) {
-    std::vector<std::pair<std::string, std::string>> pairs;
-    for (auto kv : py::cast<py::dict>(sel))
-      pairs.emplace_back(py::cast<std::string>(kv.first), py::cast<std::string>(kv.second));
-    ctx.pdbs.push_back(std::move(pairs));
-  }
+  parse_pdb_matchers(pdb_selectors, ctx);
 
   const bool dbg = getenv("MI355X_CODEC_DEBUG") != nullptr;
   auto t0 = std::chrono::steady_clock::now();
@@ -853,12 +877,7 @@ pybind11::dict decode_watch_events(pybind11::bytes body,
   Ctx ctx;
   for (auto item : node_index)
     ctx.node_idx[py::cast<std::string>(item.first)] = py::cast<int32_t>(item.second);
-  for (auto sel : pdb_selectors) {
-    std::vector<std::pair<std::string, std::string>> pairs;
-    for (auto kv : py::cast<py::dict>(sel))
-      pairs.emplace_back(py::cast<std::string>(kv.first), py::cast<std::string>(kv.second));
-    ctx.pdbs.push_back(std::move(pairs));
-  }
+  parse_pdb_matchers(pdb_selectors, ctx);
 
   struct EvShard {
     Columns cols;

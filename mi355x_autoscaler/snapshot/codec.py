@@ -49,7 +49,11 @@ def encode_snapshot_fast(
     # node/pool side via the canonical encoder (zero pods)
     base = encode_snapshot([], nodes, pools, pdbs)
     node_index = {n.name: i for i, n in enumerate(nodes)}
-    pdb_selectors = [dict(getattr(p, "match_labels", {}) or {}) for p in (pdbs or [])]
+    pdb_selectors = [
+        (getattr(p, "namespace", "default"),
+         dict(getattr(p, "match_labels", {}) or {}))
+        for p in (pdbs or [])
+    ]
     cols = _hip_ops.encode_pods_json(bytes(pods_body), node_index, pdb_selectors)
 
     snap = ClusterSnapshot(

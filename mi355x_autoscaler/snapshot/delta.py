@@ -36,6 +36,20 @@ _POD_COLS = ("pod_req", "pod_flags", "pod_sel_hash", "pod_sel_pairs",
              "pod_label_pairs", "pod_label_count")
 
 
+def _norm_matchers(seq) -> List[Tuple[str, Dict[str, str]]]:
+    """Normalize PDB matchers to [(namespace, matchLabels)] — a v1 PDB
+    selects pods in ITS OWN namespace only. Bare label dicts (legacy
+    call shape) normalize to namespace 'default'."""
+    out: List[Tuple[str, Dict[str, str]]] = []
+    for s in (seq or []):
+        if isinstance(s, dict):
+            out.append(("default", dict(s)))
+        else:
+            ns, labels = s
+            out.append((str(ns), dict(labels)))
+    return out
+
+
 class DeltaSnapshotCache:
     """Maintains the pod half of a ClusterSnapshot under watch events.
 
@@ -88,7 +102,7 @@ class DeltaSnapshotCache:
         self.dirty = set()
         self.dirty_all = False
         self.node_index = dict(node_index)
-        self.pdb_selectors = list(pdb_selectors or [])
+        self.pdb_selectors = _norm_matchers(pdb_selectors)
 
     # ------------------------------------------------------------------
     def _encode_row(self, pod: KubePod):
@@ -121,8 +135,9 @@ class DeltaSnapshotCache:
             node = self.node_index[pod.node_name]
             f |= PodFlags.BOUND
         pdb = -1
-        for d, sel in enumerate(self.pdb_selectors):
-            if sel and all(pod.labels.get(k) == v for k, v in sel.items()):
+        for d, (pns, sel) in enumerate(self.pdb_selectors):
+            if (sel and pns == pod.namespace
+                    and all(pod.labels.get(k) == v for k, v in sel.items())):
                 pdb = d
                 break
         return req, pairs, min(cnt, MAX_SELECTOR_PAIRS), combined, f, node, pdb, lpairs, lcnt
@@ -349,25 +364,27 @@ class DeltaSnapshotCache:
             return None
         from .encode import _hash_pairs
 
+        matchers = _norm_matchers(pdb_selectors)
         n = self.size
         lp = self.arrays["pod_label_pairs"][:n]
         lc = self.arrays["pod_label_count"][:n]
         live = np.array([bool(u) for u in self.pod_uids[:n]], bool)
         if live.any() and (lc[live] > MAX_POD_LABEL_PAIRS).any():
             return None  # truncated label sets: can't rematch exactly
+        pod_ns = np.array([r[0] for r in self.pod_refs[:n]], object)
         new_pdb = np.full(n, -1, np.int32)
         unassigned = live.copy()
-        for d, sel in enumerate(pdb_selectors):
+        for d, (pns, sel) in enumerate(matchers):
             if not sel:
                 continue  # empty matchLabels never matches (encode parity)
             pairs, cnt, _ = _hash_pairs(sel, MAX_POD_LABEL_PAIRS)
             if cnt > MAX_POD_LABEL_PAIRS:
                 return None
-            m = unassigned.copy()
+            m = unassigned & (pod_ns == pns)
             for h in pairs[:cnt]:
-                m &= (lp == h).any(axis=1)
                 if not m.any():
                     break
+                m &= (lp == h).any(axis=1)
             new_pdb[m] = d
             unassigned &= ~m
         old = self.arrays["pod_pdb"][:n]
@@ -379,7 +396,7 @@ class DeltaSnapshotCache:
                 self.dirty_all = True
             else:
                 self.dirty.update(int(r) for r in changed)
-        self.pdb_selectors = [dict(s) for s in pdb_selectors]
+        self.pdb_selectors = matchers
         return int(changed.size)
 
     @property
@@ -474,4 +491,4 @@ class DeltaSnapshotCache:
         self.dirty = set()
         self.dirty_all = True
         self.node_index = dict(node_index)
-        self.pdb_selectors = list(pdb_selectors or [])
+        self.pdb_selectors = _norm_matchers(pdb_selectors)

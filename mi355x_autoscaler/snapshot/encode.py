@@ -277,13 +277,16 @@ def encode_snapshot(
         if n.pool_name in pool_index:
             node_pool[i] = pool_index[n.pool_name]
 
-    # PDB matching: a PDB object carries {selector labels, allowed}.
+    # PDB matching: a PDB object carries {namespace, selector labels,
+    # allowed}. A PDB selects pods in ITS OWN namespace only (v1
+    # semantics) — label reuse across namespaces must not cross-guard.
     pdbs = list(pdbs or [])
     pdb_allowed = np.zeros(len(pdbs), np.int32)
-    pdb_selectors: List[Dict[str, str]] = []
+    pdb_matchers: List[Tuple[str, Dict[str, str]]] = []
     for d, pdb in enumerate(pdbs):
         pdb_allowed[d] = int(getattr(pdb, "disruptions_allowed", 0))
-        pdb_selectors.append(dict(getattr(pdb, "match_labels", {}) or {}))
+        pdb_matchers.append((getattr(pdb, "namespace", "default"),
+                             dict(getattr(pdb, "match_labels", {}) or {})))
 
     for i, p in enumerate(pods):
         pod_req[i] = quantize_request(p.resources)
@@ -316,8 +319,9 @@ def encode_snapshot(
             pod_node[i] = node_idx[p.node_name]
             f |= PodFlags.BOUND
         pod_flags[i] = f
-        for d, sel in enumerate(pdb_selectors):
-            if sel and all(p.labels.get(k) == v for k, v in sel.items()):
+        for d, (pns, sel) in enumerate(pdb_matchers):
+            if (sel and pns == p.namespace
+                    and all(p.labels.get(k) == v for k, v in sel.items())):
                 pod_pdb[i] = d
                 break
 

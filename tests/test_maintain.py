@@ -94,3 +94,54 @@ def test_scaledown_trace_500_nodes():
     for c in rep.candidates:
         assert rep.node_blocked[c.node_index] == 0
         assert rep.node_pdb_ok[c.node_index] == 1
+
+
+def test_pdb_namespace_scoping():
+    """A v1 PDB selects pods in ITS OWN namespace only: identical
+    labels in another namespace must not be guarded — consistently
+    across the Python encoder, the C++ codec, and the watch cache."""
+    import json
+
+    from mi355x_autoscaler.kube.objects import PodDisruptionBudget
+    from mi355x_autoscaler.models.kube import KubeNode, KubePod
+    from mi355x_autoscaler.snapshot.codec import codec_available, encode_snapshot_fast
+    from mi355x_autoscaler.snapshot.delta import DeltaSnapshotCache
+
+    node = make_node_json("agentpool1", 0)
+    nodes = [KubeNode(node)]
+    pools = group_nodes_into_pools(nodes)
+    nn = node["metadata"]["name"]
+    guarded = make_pod_json("g", cpu_milli=100, node_name=nn,
+                            labels={"app": "web"})
+    other = make_pod_json("o", cpu_milli=100, node_name=nn,
+                          namespace="other", labels={"app": "web"})
+    pdb = PodDisruptionBudget(name="guard", match_labels={"app": "web"},
+                              disruptions_allowed=0, namespace="default")
+
+    pods_json = [guarded, other]
+    snap = encode_snapshot([KubePod(p) for p in pods_json], nodes, pools, [pdb])
+    by_uid = {u: i for i, u in enumerate(snap.pod_uids)}
+    assert snap.pod_pdb[by_uid["pod-default-g"]] == 0
+    assert snap.pod_pdb[by_uid["pod-other-o"]] == -1  # different namespace
+
+    if codec_available():
+        body = json.dumps({"items": pods_json}).encode()
+        fast = encode_snapshot_fast(body, nodes, pools, [pdb])
+        fby = {u: i for i, u in enumerate(fast.pod_uids)}
+        assert fast.pod_pdb[fby["pod-default-g"]] == 0
+        assert fast.pod_pdb[fby["pod-other-o"]] == -1
+
+    # watch cache: per-event encode + in-place reassign
+    cache = DeltaSnapshotCache()
+    cache.prime(encode_snapshot([], nodes, pools, [pdb]), {nn: 0},
+                [(pdb.namespace, dict(pdb.match_labels))])
+    cache.apply_event({"type": "ADDED", "object": guarded})
+    cache.apply_event({"type": "ADDED", "object": other})
+    rows = {u: i for i, u in enumerate(cache.pod_uids[: cache.size])}
+    assert cache.arrays["pod_pdb"][rows["pod-default-g"]] == 0
+    assert cache.arrays["pod_pdb"][rows["pod-other-o"]] == -1
+    # reassign with the PDB moved to namespace "other"
+    n = cache.reassign_pdbs([("other", {"app": "web"})])
+    assert n == 2  # g unguarded, o guarded
+    assert cache.arrays["pod_pdb"][rows["pod-default-g"]] == -1
+    assert cache.arrays["pod_pdb"][rows["pod-other-o"]] == 0
