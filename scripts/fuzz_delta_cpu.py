@@ -50,7 +50,8 @@ def rand_pod(rng, i, node_names):
             labels.update({f"l{j:02d}": "v" for j in range(18)})  # overflow
     node = rng.choice(node_names + [None, None])
     return make_pod_json(
-        f"f{i}", cpu_milli=rng.choice([0, 100, 900]),
+        f"f{i}", namespace=rng.choice(["default", "default", "team-a"]),
+        cpu_milli=rng.choice([0, 100, 900]),
         mem_mib=rng.choice([0, 128, 2048]),
         gpu=rng.choice([0, 0, 1]),
         pending=node is None, node_name=node,
@@ -66,6 +67,7 @@ def pdb_menu(rng):
         if rng.random() < 0.6:
             out.append(PodDisruptionBudget(
                 name=name, match_labels=sel,
+                namespace=rng.choice(["default", "team-a"]),
                 disruptions_allowed=rng.randint(0, 2)))
     return out
 
@@ -110,7 +112,7 @@ def main(iters=None):
         snap0 = encode_snapshot([KubePod(p) for p in initial], nodes, pools, pdbs)
         cache = DeltaSnapshotCache()
         cache.prime(snap0, {n: i for i, n in enumerate(node_names)},
-                    [dict(p.match_labels) for p in pdbs])
+                    [(p.namespace, dict(p.match_labels)) for p in pdbs])
 
         next_id = len(initial)
         for step in range(rng.randint(1, 6)):
@@ -137,13 +139,13 @@ def main(iters=None):
                         cache.apply_event({"type": "DELETED", "object": live.pop(uid)})
             elif op < 0.75:  # PDB-set change
                 pdbs = pdb_menu(rng)
-                n = cache.reassign_pdbs([dict(p.match_labels) for p in pdbs])
+                n = cache.reassign_pdbs([(p.namespace, dict(p.match_labels)) for p in pdbs])
                 if n is None:  # overflow rows: re-prime like the loop does
                     snapn = encode_snapshot([KubePod(p) for p in live.values()],
                                             nodes, pools, pdbs)
                     cache = DeltaSnapshotCache()
                     cache.prime(snapn, {n2: i for i, n2 in enumerate(node_names)},
-                                [dict(p.match_labels) for p in pdbs])
+                                [(p.namespace, dict(p.match_labels)) for p in pdbs])
             else:
                 cache.maybe_compact()
             base = encode_snapshot([], nodes, pools, pdbs)
