@@ -42,7 +42,7 @@ from ..snapshot.encode import MAX_SELECTOR_PAIRS, ClusterSnapshot
 from . import dist as pdist
 
 _U64 = (1 << 64) - 1
-_ROW_COLS = MAX_SELECTOR_PAIRS + 8  # pairs + count + req(4) + seg_count + sel_hash + req_key (+pad)
+_ROW_COLS = MAX_SELECTOR_PAIRS + 8  # pairs + count + req(4) + seg_count + sel_hash + req_key
 
 
 def _segments_to_rows(segs: List[Segment]) -> np.ndarray:
