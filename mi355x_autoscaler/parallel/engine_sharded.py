@@ -7,7 +7,7 @@ Protocol per decide():
      segments (kernel C + device radix sort — the P-proportional work,
      which is what scales with ranks);
   2. `allgather_rows` of the per-rank segment rows (sel pairs + request
-     + count: 15 x i64 per segment, KB-sized, latency-bound);
+     + count: MAX_SELECTOR_PAIRS+8 i64 per segment, KB-sized, latency-bound);
   3. every rank merges counts by key into the IDENTICAL global segment
      list (same order as sim/oracle.py::merge_segment_counts);
   4. every rank runs the node-table placement (kernels A/B) on the
