@@ -454,6 +454,9 @@ class DeltaSnapshotCache:
     def save(self, path: str):
         np.savez_compressed(
             path,
+            version=2,  # v2 = + pod_label_pairs/count (v1 loads fine:
+            #             load() zero-fills absent columns and disables
+            #             in-place PDB reassignment)
             size=self.size,
             uids=np.array(self.pod_uids[: self.size], dtype=object),
             refs=np.array(self.pod_refs[: self.size], dtype=object),
