@@ -195,7 +195,7 @@ def test_reassign_pdbs_refuses_without_label_data():
 def test_randomized_delta_fuzz():
     """Randomized watch-event/PDB-change/compaction sequences converge
     to the full re-encode, column-exact (scripts/fuzz_delta_cpu.py
-    runs the longer form; 200 iterations green at commit time)."""
+    runs the longer form; 1000 iterations green at commit time)."""
     import importlib.util
     import os
 
