@@ -67,8 +67,12 @@ ext = CUDAExtension(
 # LD_PRELOAD=$(gcc -print-file-name=libasan.so).
 
 setup(
-    name="mi355x_autoscaler_ops",
+    name="mi355x-autoscaler",
     version="0.1.0",
+    description=("MI355X-native Kubernetes GPU-node autoscaler for "
+                 "ACS-Engine agent pools (CDNA4 HIP scheduling simulation, "
+                 "RCCL-sharded controller)"),
+    packages=["mi355x_autoscaler"],
     ext_modules=[ext],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
 )
