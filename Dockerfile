@@ -13,5 +13,5 @@ RUN python setup.py build_ext --inplace
 
 # one GPU is enough for the controller; sharded mode is launched via
 # torchrun (see deploy/autoscaler.yaml)
-ENTRYPOINT ["python", "-m", "mi355x_autoscaler.cli"]
+ENTRYPOINT ["python", "-m", "mi355x_autoscaler"]
 CMD ["--backend", "gpu", "--sleep", "10", "-v"]
