@@ -70,6 +70,8 @@ class CycleMetrics:
                 "pods_scheduled": Counter("autoscaler_pods_scheduled_total", "pods placed by simulation"),
                 "scale_ups": Counter("autoscaler_scale_ups_total", "scale-up actuations"),
                 "scale_downs": Counter("autoscaler_scale_downs_total", "nodes removed"),
+                "watch_events": Counter("autoscaler_watch_events_total",
+                                        "pod watch events applied to the delta cache"),
                 "decision_latency": Histogram(
                     "autoscaler_decision_latency_seconds",
                     "pod-pending -> decision latency",
