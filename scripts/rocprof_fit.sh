@@ -8,6 +8,8 @@
 #   scripts/rocprof_fit.sh [outdir] [bench args...]
 # PMC counters (MFMA util, LDS conflicts) need a SEPARATE run:
 #   scripts/rocprof_fit.sh --pmc
+# If rocprofv3 segfaults on the box image (seen late round 1), use the
+# event-timed fallback: python scripts/probe_phase_times.py
 set -euo pipefail
 cd "$(dirname "$0")/.."
 REPO="$PWD"
