@@ -472,7 +472,14 @@ class Cluster:
         # guarded bound pod (a per-candidate O(P) scan would cost
         # candidates x milliseconds at 1M pods)
         D = max(1, len(pdb_left))
-        guarded = (pod_pdb >= 0) & (pod_node_full >= 0)
+        from ..snapshot.encode import PodFlags
+
+        # only pods a drain actually evicts consume budget: mirror/
+        # daemonset pods survive _drain_and_delete, so counting them
+        # here would over-decrement pdb_left and skip later drains
+        evictable = (actuation_snap.pod_flags
+                     & (PodFlags.MIRROR | PodFlags.DAEMONSET)) == 0
+        guarded = (pod_pdb >= 0) & (pod_node_full >= 0) & evictable
         node_pdb_counts: Dict[int, Dict[int, int]] = {}
         if guarded.any():
             key = pod_node_full[guarded].astype(np.int64) * D + pod_pdb[guarded]
@@ -523,12 +530,13 @@ class Cluster:
         from ..snapshot.encode import PodFlags
 
         skip = PodFlags.MIRROR | PodFlags.DAEMONSET
+        # vectorized resident-pod lookup: a Python for-loop over
+        # snap.num_pods costs seconds per drain at the 1M-pod scale
+        resident = np.nonzero(
+            (snap.pod_node == j) & ((snap.pod_flags & skip) == 0))[0]
         n_drained = 0
-        for i in range(snap.num_pods):
-            if int(snap.pod_node[i]) != j:
-                continue
-            if int(snap.pod_flags[i]) & skip:
-                continue  # mirror/daemonset pods survive, as kubectl drain
+        for i in resident.tolist():
+            # mirror/daemonset pods already excluded, as kubectl drain
             ns, name = snap.pod_refs[i]
             self.kube.delete_pod(ns, name, grace_seconds=self.opts.drain_grace_seconds)
             n_drained += 1

@@ -627,11 +627,13 @@ void encode_one(const JValue* pod, const Ctx& ctx, Columns& out) {
   if (ns.empty()) ns = "default";
 
   // PDB assignment: first SAME-NAMESPACE PDB whose matchLabels are
-  // all present (v1 PDB semantics; mirrors encode.py/delta.py)
+  // all present (v1 PDB semantics; mirrors encode.py/delta.py).
+  // policy/v1: EMPTY matchLabels selects every pod in the namespace
+  // (the inner loop is vacuously true).
   int32_t pdb = -1;
   if (!ctx.pdbs.empty()) {
     for (size_t d = 0; d < ctx.pdbs.size(); ++d) {
-      if (ctx.pdbs[d].labels.empty() || ctx.pdbs[d].ns != ns) continue;
+      if (ctx.pdbs[d].ns != ns) continue;
       bool all = true;
       for (auto& [k, v] : ctx.pdbs[d].labels) {
         const JValue* lv = labels ? get(labels, k) : nullptr;

@@ -36,9 +36,12 @@ extern "C" __global__ void drain_flags_kernel(
     if (!(f & (PF_MIRROR | PF_DAEMONSET))) {
       atomicAdd(&nontrivial[j], 1);
       if (f & PF_DRAINABLE) atomicAdd(&drainable[j], 1);
+      // a drain never evicts mirror/daemonset pods, so only evictable
+      // pods consume PDB budget (matches the host-side cross-drain
+      // accounting and the Python oracle)
+      const int d = pod_pdb[i];
+      if (d >= 0 && pdb_allowed[d] < 1) atomicOr(&pdb_bad[j], 1);
     }
-    const int d = pod_pdb[i];
-    if (d >= 0 && pdb_allowed[d] < 1) atomicOr(&pdb_bad[j], 1);
   }
 }
 

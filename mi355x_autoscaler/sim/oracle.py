@@ -339,9 +339,11 @@ class OracleSimulator:
                 nontrivial[j] += 1
                 if f & PodFlags.DRAINABLE:
                     drainable_pods[j] += 1
-            d = int(snap.pod_pdb[i])
-            if d >= 0 and int(snap.pdb_allowed[d]) < 1:
-                pdb_bad[j] = 1
+                # drains never evict mirror/daemonset pods: only
+                # evictable pods consume PDB budget (kernel D parity)
+                d = int(snap.pod_pdb[i])
+                if d >= 0 and int(snap.pdb_allowed[d]) < 1:
+                    pdb_bad[j] = 1
         return dict(util=util, blocked=blocked, pdb_bad=pdb_bad,
                     nontrivial=nontrivial, drainable=drainable_pods)
 

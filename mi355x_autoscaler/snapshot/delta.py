@@ -15,6 +15,7 @@ to an .npz so a 1M-pod replay can resume mid-burst.
 
 from __future__ import annotations
 
+import logging
 from typing import Dict, List, Optional, Set, Tuple
 
 import numpy as np
@@ -27,6 +28,8 @@ from .encode import (
     PodFlags,
     encode_snapshot,
 )
+
+logger = logging.getLogger(__name__)
 
 _GROW = 1.5
 _COMPACT_DEAD_FRACTION = 0.5
@@ -212,6 +215,15 @@ class DeltaSnapshotCache:
         n = int(types.shape[0])
         if n == 0:
             return 0
+        if self.has_label_pairs and "pod_label_pairs" not in decoded:
+            # a batch without label columns (mismatched/absent compiled
+            # codec) would leave stale label hashes that reassign_pdbs
+            # trusts — disable in-place reassign, forcing the re-prime
+            # fallback instead of silently wrong PDB matching
+            logger.warning(
+                "watch batch lacks pod label columns; disabling in-place "
+                "PDB reassignment (re-prime fallback)")
+            self.has_label_pairs = False
         resolve = None
         try:
             import torch  # noqa: F401  (extension links against libtorch)
@@ -375,8 +387,9 @@ class DeltaSnapshotCache:
         new_pdb = np.full(n, -1, np.int32)
         unassigned = live.copy()
         for d, (pns, sel) in enumerate(matchers):
-            if not sel:
-                continue  # empty matchLabels never matches (encode parity)
+            # policy/v1: empty matchLabels selects every pod in the
+            # PDB's namespace — the pair loop below is vacuous, so m
+            # stays unassigned & same-namespace (encode/codec parity)
             pairs, cnt, _ = _hash_pairs(sel, MAX_POD_LABEL_PAIRS)
             if cnt > MAX_POD_LABEL_PAIRS:
                 return None

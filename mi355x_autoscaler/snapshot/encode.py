@@ -320,7 +320,10 @@ def encode_snapshot(
             f |= PodFlags.BOUND
         pod_flags[i] = f
         for d, (pns, sel) in enumerate(pdb_matchers):
-            if (sel and pns == p.namespace
+            # policy/v1: an EMPTY selector selects every pod in the
+            # PDB's namespace (all() is vacuously true) — matched
+            # identically in the C++ codec and delta.reassign_pdbs
+            if (pns == p.namespace
                     and all(p.labels.get(k) == v for k, v in sel.items())):
                 pod_pdb[i] = d
                 break

@@ -8,7 +8,7 @@ with the repo snapshot to GPU boxes (no JIT cache dependence).
 
 import os
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -72,7 +72,9 @@ setup(
     description=("MI355X-native Kubernetes GPU-node autoscaler for "
                  "ACS-Engine agent pools (CDNA4 HIP scheduling simulation, "
                  "RCCL-sharded controller)"),
-    packages=["mi355x_autoscaler"],
+    # every subpackage (ops, snapshot, ctl, ...) must ship in a wheel
+    # install, not just the source-tree build_ext --inplace path
+    packages=find_packages(include=["mi355x_autoscaler", "mi355x_autoscaler.*"]),
     ext_modules=[ext],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
 )

@@ -205,3 +205,40 @@ def test_randomized_delta_fuzz():
     mod = importlib.util.module_from_spec(spec)
     spec.loader.exec_module(mod)
     mod.main(iters=8)
+
+
+def test_apply_batch_without_label_columns_disables_reassign():
+    """A decoded batch lacking the label columns (mismatched compiled
+    codec) must clear has_label_pairs so reassign_pdbs falls back to
+    the re-prime instead of trusting stale label hashes."""
+    import numpy as np
+
+    from mi355x_autoscaler.snapshot.encode import (
+        MAX_POD_LABEL_PAIRS, MAX_SELECTOR_PAIRS, NUM_RESOURCES)
+
+    nodes = [KubeNode(make_node_json("agentpool1", 0))]
+    base, pools = _base(nodes)
+    pods_json = [make_pod_json("a", cpu_milli=100, pending=True,
+                               labels={"app": "x"})]
+    snap = encode_snapshot([KubePod(p) for p in pods_json], nodes, pools)
+    cache = DeltaSnapshotCache()
+    cache.prime(snap, {nodes[0].name: 0})
+    assert cache.has_label_pairs
+
+    n = 1
+    decoded = {
+        "types": np.zeros(n, np.int32),
+        "pod_uids": ["uid-batch-a"],
+        "pod_refs": [("default", "batch-a")],
+        "pod_req": np.zeros((n, NUM_RESOURCES), np.int32),
+        "pod_flags": np.zeros(n, np.int32),
+        "pod_sel_hash": np.zeros(n, np.int64),
+        "pod_sel_pairs": np.zeros((n, MAX_SELECTOR_PAIRS), np.int64),
+        "pod_sel_count": np.zeros(n, np.int32),
+        "pod_node": np.full(n, -1, np.int32),
+        "pod_pdb": np.full(n, -1, np.int32),
+        # NOTE: no pod_label_pairs / pod_label_count columns
+    }
+    cache.apply_batch(decoded)
+    assert not cache.has_label_pairs
+    assert cache.reassign_pdbs([{"app": "x"}]) is None

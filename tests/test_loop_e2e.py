@@ -189,3 +189,28 @@ def test_failed_to_scale_names_an_example_pod(server):
     cluster.loop_logic()
     evts = [e for e in cluster.notifier.events if e["kind"] == "failed_to_scale"]
     assert evts and evts[0]["example_pod"] == "default/orphan"
+
+
+def test_pdb_budget_ignores_daemonset_pods(server):
+    """The cross-drain PDB budget must count only pods a drain actually
+    evicts: a guarded DAEMONSET pod sharing the PDB must not consume
+    budget and starve a later drain in the same cycle."""
+    nodes = [make_node_json("agentpool1", i) for i in range(3)]
+    busy = make_pod_json("busy", cpu_milli=1500,
+                         node_name=nodes[0]["metadata"]["name"])
+    ds = make_pod_json("infra-ds", cpu_milli=50,
+                       node_name=nodes[1]["metadata"]["name"],
+                       owner_kind="DaemonSet", labels={"app": "guarded"})
+    g = make_pod_json("g", cpu_milli=100,
+                      node_name=nodes[2]["metadata"]["name"],
+                      labels={"app": "guarded"})
+    pdb = {"metadata": {"name": "guard"},
+           "spec": {"selector": {"matchLabels": {"app": "guarded"}}},
+           "status": {"disruptionsAllowed": 1}}
+    server.state.load([busy, ds, g], nodes, pdbs=[pdb])
+    cluster, provider = _cluster(server, no_scale=True, max_drains_per_cycle=10)
+
+    res = cluster.loop_logic()
+    # both idle nodes drain: the daemonset pod consumed no budget
+    assert sorted(res.drained_nodes) == sorted(
+        [nodes[1]["metadata"]["name"], nodes[2]["metadata"]["name"]])

@@ -145,3 +145,61 @@ def test_pdb_namespace_scoping():
     assert n == 2  # g unguarded, o guarded
     assert cache.arrays["pod_pdb"][rows["pod-default-g"]] == -1
     assert cache.arrays["pod_pdb"][rows["pod-other-o"]] == 0
+
+
+def test_pdb_empty_selector_selects_whole_namespace():
+    """policy/v1: a PDB with EMPTY matchLabels selects every pod in its
+    namespace — consistent across the Python encoder, the C++ codec,
+    and the watch cache's in-place reassign."""
+    import json
+
+    from mi355x_autoscaler.models.kube import KubeNode, KubePod
+    from mi355x_autoscaler.snapshot.codec import codec_available, encode_snapshot_fast
+    from mi355x_autoscaler.snapshot.delta import DeltaSnapshotCache
+
+    node = make_node_json("agentpool1", 0)
+    nodes = [KubeNode(node)]
+    pools = group_nodes_into_pools(nodes)
+    nn = node["metadata"]["name"]
+    a = make_pod_json("a", cpu_milli=100, node_name=nn, labels={"app": "x"})
+    b = make_pod_json("b", cpu_milli=100, node_name=nn)  # no labels at all
+    other = make_pod_json("o", cpu_milli=100, node_name=nn, namespace="other")
+    pdb = PodDisruptionBudget(name="all", match_labels={},
+                              disruptions_allowed=0, namespace="default")
+
+    pods_json = [a, b, other]
+    snap = encode_snapshot([KubePod(p) for p in pods_json], nodes, pools, [pdb])
+    by_uid = {u: i for i, u in enumerate(snap.pod_uids)}
+    assert snap.pod_pdb[by_uid["pod-default-a"]] == 0
+    assert snap.pod_pdb[by_uid["pod-default-b"]] == 0
+    assert snap.pod_pdb[by_uid["pod-other-o"]] == -1  # other namespace
+
+    if codec_available():
+        body = json.dumps({"items": pods_json}).encode()
+        fast = encode_snapshot_fast(body, nodes, pools, [pdb])
+        fby = {u: i for i, u in enumerate(fast.pod_uids)}
+        for u, want in (("pod-default-a", 0), ("pod-default-b", 0),
+                        ("pod-other-o", -1)):
+            assert fast.pod_pdb[fby[u]] == want, u
+
+    # in-place reassign from cached label hashes must agree
+    cache = DeltaSnapshotCache()
+    cache.prime(snap, {nn: 0}, [("default", {})])
+    changed = cache.reassign_pdbs([("default", {})])
+    assert changed == 0  # already assigned identically by encode
+
+
+def test_pdb_on_daemonset_pod_does_not_block_drain():
+    """A zero-budget PDB guarding only a DAEMONSET pod must not mark
+    the node pdb_bad: drains never evict daemonset/mirror pods, so no
+    disruption is consumed (kernel D + oracle + host accounting)."""
+    node = make_node_json("agentpool1", 0)
+    ds = make_pod_json("ds", cpu_milli=100, node_name=node["metadata"]["name"],
+                       owner_kind="DaemonSet", labels={"app": "infra"})
+    pdb0 = PodDisruptionBudget("g", match_labels={"app": "infra"},
+                               disruptions_allowed=0)
+    snap = _snap([ds], [node], [pdb0])
+    assert snap.pod_pdb[0] == 0  # the pod IS guarded...
+    rep = OracleSimulator(snap).maintain_report()
+    assert rep.node_pdb_ok[0] == 1  # ...but consumes no budget
+    assert len(rep.candidates) == 1
