@@ -206,6 +206,15 @@ def main():
     sync()
     t_all1 = time.perf_counter()
 
+    if use_gpu:
+        # the benched configs must run the HIP fast path end to end —
+        # a silent capacity-cliff fallback would invalidate the number
+        from mi355x_autoscaler.ops.engine_gpu import FALLBACK_COUNTS
+
+        assert not FALLBACK_COUNTS, (
+            f"GPU fast-path fallback engaged during bench: "
+            f"{dict(FALLBACK_COUNTS)}")
+
     total_s_local = t_all1 - t_all0
     # MAX over ranks of the timed window
     if world > 1:
@@ -219,7 +228,15 @@ def main():
 
     ms_per_step = total_s / args.steps * 1000.0
     pods_total = world * (snap.num_pods if maintain_mode else args.pods_per_gpu)
+    # three rates so the headline can't be misread:
+    #   pods_scheduled_per_sec — pods PROCESSED per decision-second
+    #     (the whole burst is classified/fitted/packed each step, incl.
+    #     pods that end up unfulfillable at pool caps)
+    #   pods_placed_per_sec    — pods actually PLACED (existing + new
+    #     nodes) per decision-second
+    #   decisions_per_sec      — global scale decisions per second
     pods_per_sec = pods_total / (total_s / args.steps)
+    decisions_per_sec = args.steps / total_s
     p50_ms = statistics.median(step_times) * 1000.0
     p99_ms = sorted(step_times)[min(len(step_times) - 1,
                                     int(0.99 * len(step_times)))] * 1000.0
@@ -257,11 +274,18 @@ def main():
                 "backend": "gpu" if use_gpu else "cpu-oracle",
                 "decision_p50_ms": round(p50_ms, 3),
                 "decision_p99_ms": round(p99_ms, 3),
+                "decisions_per_sec": round(decisions_per_sec, 1),
                 **(
                     {"drain_candidates": len(decision.candidates)}
                     if maintain_mode else
                     {
+                        # sharded decisions merge the GLOBAL segment
+                        # list, so placed counts are already whole-job
                         "pods_placed": decision.pods_placed_existing + decision.pods_placed_new,
+                        "pods_placed_per_sec": round(
+                            (decision.pods_placed_existing
+                             + decision.pods_placed_new)
+                            / (total_s / args.steps), 1),
                         "pods_unfulfillable": decision.pods_unfulfillable,
                         "units_needed": decision.units_needed,
                     }

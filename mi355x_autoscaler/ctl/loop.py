@@ -258,6 +258,16 @@ class Cluster:
 
         result.phase_times = t.end_cycle()
         self.metrics.count("cycles")
+        # loud fallback telemetry: capacity-cliff fallbacks in the GPU
+        # engine (>64 pools, hash overflow, merged-row overflow) are
+        # exported every cycle so operators see a lost GPU path
+        try:
+            from ..ops.engine_gpu import FALLBACK_COUNTS
+
+            for kind, n in FALLBACK_COUNTS.items():
+                self.metrics.gauge(f"gpu_fallback_{kind}", n)
+        except ImportError:
+            pass
         return result
 
     def _base_snapshot(self, nodes, pools, pdbs) -> ClusterSnapshot:

@@ -57,6 +57,23 @@ SEG_MAX = 2048
 HASH_SLOTS = 8192
 _ALIGN = 64
 
+# Loud fallback telemetry: every capacity-cliff
+# fallback (>64 pools -> CPU oracle, hash/sort overflow -> torch-sort
+# segbuild, merged rows > workspace) bumps a counter and WARNs on first
+# occurrence, so a 65-pool cluster can't silently lose the GPU path.
+# The control loop exports these as gauges; bench.py asserts the
+# headline configs stay clean.
+from collections import Counter as _Counter
+
+FALLBACK_COUNTS: "_Counter[str]" = _Counter()
+
+
+def _note_fallback(kind: str, detail: str = ""):
+    FALLBACK_COUNTS[kind] += 1
+    if FALLBACK_COUNTS[kind] == 1:
+        logger.warning("GPU fast-path fallback engaged: %s%s", kind,
+                       f" ({detail})" if detail else "")
+
 
 @dataclass
 class DeviceSnapshot:
@@ -799,6 +816,8 @@ class GpuEngine(SimulationEngine):
             segs = self.build_segments_hash(ds)
             if segs is not None:
                 return segs
+            _note_fallback("segbuild_overflow_torch_sort",
+                           f">{SEG_MAX} segments or >{HASH_SLOTS} keys")
         return build_segments_device(ds, self.ops)
 
     # ------------------------------------------------------------------
@@ -808,11 +827,8 @@ class GpuEngine(SimulationEngine):
             # (UNITS_MAX_POOLS); >64 agent pools is outside every ACS
             # deployment shape we target — decide on the CPU oracle
             # (bit-identical by construction) instead of crashing.
-            if not getattr(self, "_warned_many_pools", False):
-                logger.warning(
-                    "%d agent pools exceeds the GPU units kernel's 64-pool "
-                    "LDS budget; deciding on the CPU oracle", snap.num_pools)
-                self._warned_many_pools = True
+            _note_fallback("many_pools_oracle",
+                           f"{snap.num_pools} pools > 64-pool LDS budget")
             from ..sim.oracle import OracleSimulator
 
             return OracleSimulator(
@@ -1004,6 +1020,8 @@ class GpuEngine(SimulationEngine):
             d.pool_residual_totals = np.zeros((ds.K, 4), np.int64)
             return d
         if G > SEG_MAX:
+            _note_fallback("merged_rows_overflow",
+                           f"{G} merged segments > {SEG_MAX} workspace")
             return None
         device = self.device
         N, K = ds.N, ds.K
@@ -1185,6 +1203,8 @@ class GpuEngine(SimulationEngine):
         import types
 
         if base.num_pools > 64:  # same LDS bound as decide(); oracle path
+            _note_fallback("many_pools_oracle",
+                           f"{base.num_pools} pools > 64-pool LDS budget")
             from ..sim.oracle import OracleSimulator
 
             snap = cache.build_snapshot(base)

@@ -496,7 +496,12 @@ def test_more_than_64_pools_falls_back_to_oracle(engine):
             for i in range(40)]
     snap = _snap(pods, nodes_json)
     assert snap.num_pools == 66
+    from mi355x_autoscaler.ops.engine_gpu import FALLBACK_COUNTS
+
+    before = FALLBACK_COUNTS["many_pools_oracle"]
     _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))
+    # the fallback must be LOUD: counter bumped (exported as a gauge)
+    assert FALLBACK_COUNTS["many_pools_oracle"] == before + 1
 
 
 def test_overflow_selector_matches_nothing_on_gpu(engine):
