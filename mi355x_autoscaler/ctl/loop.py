@@ -187,10 +187,17 @@ class Cluster:
 
         world = pdist.world_size()
         rank = pdist.rank()
+        ws = self._watch_state
         if world > 1:
-            from ..parallel.elastic import shard_snapshot
+            if self.opts.watch and ws is not None and ws.get("shard_cache") is not None:
+                # sharded resident mode: this rank's shard cache is
+                # maintained under watch events — no O(P) re-mask
+                base = self._base_snapshot(ws["nodes"], ws["pools"], ws["pdbs"])
+                engine_snap = ws["shard_cache"].build_snapshot(base)
+            else:
+                from ..parallel.elastic import shard_snapshot
 
-            engine_snap = shard_snapshot(snap, rank, world)
+                engine_snap = shard_snapshot(snap, rank, world)
         else:
             engine_snap = snap
         is_rank0 = rank == 0
@@ -198,18 +205,20 @@ class Cluster:
         # ---- scale-up --------------------------------------------------
         t_decide0 = time.perf_counter()
         with t.phase("decide"):
-            ws = self._watch_state
             if (
                 self.opts.watch
                 and ws is not None
-                and world == 1
                 and hasattr(self.engine, "decide_resident")
+                and (world == 1 or ws.get("shard_cache") is not None)
             ):
                 # resident-controller path: pod SoA stays in HBM; watch
-                # deltas scatter in (kernel E), node/pool side re-uploads
+                # deltas scatter in (kernel E), node/pool side re-uploads.
+                # At world>1 each rank's resident SoA is its uid-hash
+                # shard; the engine's row gather agrees globally.
                 base = self._base_snapshot(ws["nodes"], ws["pools"], ws["pdbs"])
+                rcache = ws["shard_cache"] if world > 1 else ws["cache"]
                 decision = self.engine.decide_resident(
-                    ws["cache"], base, over_provision=self.opts.over_provision
+                    rcache, base, over_provision=self.opts.over_provision
                 )
             else:
                 decision = self.engine.decide(engine_snap, over_provision=self.opts.over_provision)
@@ -295,8 +304,11 @@ class Cluster:
         import queue as _q
         import threading
 
+        from ..parallel import dist as pdist
         from ..snapshot.delta import DeltaSnapshotCache
 
+        world = pdist.world_size()
+        rank = pdist.rank()
         node_names = tuple(n.name for n in nodes)
         pdb_fp = tuple((p.namespace, p.name,
                         frozenset(p.match_labels.items())) for p in pdbs)
@@ -309,9 +321,11 @@ class Cluster:
             # it in place from the cached pod LABEL pair hashes; fall
             # back to a full re-prime only when an exact rematch is
             # impossible (labels wider than the lattice, old cache).
-            n = ws["cache"].reassign_pdbs(
-                [(p.namespace, dict(p.match_labels)) for p in pdbs])
-            if n is None:
+            sel = [(p.namespace, dict(p.match_labels)) for p in pdbs]
+            n = ws["cache"].reassign_pdbs(sel)
+            sc = ws.get("shard_cache")
+            n2 = sc.reassign_pdbs(sel) if (n is not None and sc is not None) else 0
+            if n is None or n2 is None:
                 logger.info("PDB set changed; re-priming the watch cache")
                 ws = self._watch_state = None
             else:
@@ -322,7 +336,10 @@ class Cluster:
             # remap the cached pod->node indices in place — NO pod
             # relist. New nodes list at the end in the common case, so
             # the remap usually touches zero rows.
-            n = ws["cache"].remap_nodes({nd.name: i for i, nd in enumerate(nodes)})
+            idx = {nd.name: i for i, nd in enumerate(nodes)}
+            n = ws["cache"].remap_nodes(idx)
+            if ws.get("shard_cache") is not None:
+                ws["shard_cache"].remap_nodes(idx)
             logger.info("node set changed; remapped %d pod bindings in place", n)
             ws["node_names"] = node_names
 
@@ -334,12 +351,22 @@ class Cluster:
             else:
                 pods = [KubePod(p) for p in self.kube.list_pods()]
                 snap = encode_snapshot(pods, nodes, pools, pdbs)
+            node_index = {n.name: i for i, n in enumerate(nodes)}
+            pdb_sel = [(p.namespace, dict(p.match_labels)) for p in pdbs]
             cache = DeltaSnapshotCache()
-            cache.prime(
-                snap,
-                {n.name: i for i, n in enumerate(nodes)},
-                [(p.namespace, dict(p.match_labels)) for p in pdbs],
-            )
+            cache.prime(snap, node_index, pdb_sel)
+            shard_cache = None
+            if world > 1 and hasattr(self.engine, "decide_resident"):
+                # sharded resident mode: a second cache holding only
+                # this rank's uid-hash pod shard feeds the device-
+                # resident decide; the full cache stays host-side for
+                # actuation, reporting and the cross-drain PDB budget.
+                from ..parallel.elastic import shard_snapshot
+
+                shard_cache = DeltaSnapshotCache()
+                shard_cache.set_shard(rank, world)
+                shard_cache.prime(shard_snapshot(snap, rank, world),
+                                  node_index, pdb_sel)
             q: _q.SimpleQueue = _q.SimpleQueue()
             th = threading.Thread(target=self._watch_pump, args=(q,), daemon=True)
             th.start()
@@ -347,10 +374,12 @@ class Cluster:
                 "cache": cache, "q": q, "thread": th,
                 "node_names": node_names, "pdb_fp": pdb_fp, "broken": False,
                 "nodes": nodes, "pools": pools, "pdbs": pdbs,
+                "shard_cache": shard_cache,
             }
             return snap
 
         cache = ws["cache"]
+        shard_cache = ws.get("shard_cache")
         ws["nodes"], ws["pools"], ws["pdbs"] = nodes, pools, pdbs
         applied = 0
         raw_lines: List[bytes] = []
@@ -366,6 +395,8 @@ class Cluster:
                 ws["broken"] = True
                 break
             cache.apply_event(ev)
+            if shard_cache is not None:
+                shard_cache.apply_event(ev)  # self-filters by uid hash
             applied += 1
         if raw_lines:
             # C++ batch decode of the whole cycle's events at once
@@ -376,7 +407,11 @@ class Cluster:
                 list(cache.pdb_selectors),
             )
             applied += cache.apply_batch(decoded)
+            if shard_cache is not None:
+                shard_cache.apply_batch(decoded)
         cache.maybe_compact()
+        if shard_cache is not None:
+            shard_cache.maybe_compact()
         self.metrics.count("watch_events", applied)
         base = self._base_snapshot(nodes, pools, pdbs)
         return cache.build_snapshot(base)

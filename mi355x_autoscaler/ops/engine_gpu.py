@@ -1193,23 +1193,14 @@ class GpuEngine(SimulationEngine):
         )
 
     # ------------------------------------------------------------------
-    def decide_resident(self, cache, base: ClusterSnapshot,
-                        over_provision: int = 0) -> ScaleDecision:
-        """Resident-controller decide: the pod SoA stays in HBM at the
-        watch cache's CAPACITY (tombstone rows are inert); per cycle the
-        dirty rows scatter in (kernel E) and the small node/pool side
-        re-uploads — no full snapshot upload. `base` is the per-cycle
-        node/pool-side snapshot (zero pods)."""
+    def resident_ds(self, cache, base: ClusterSnapshot) -> "DeviceSnapshot":
+        """Scatter the watch cache's dirty rows into the HBM-resident
+        pod SoA and pair it with a freshly-uploaded node/pool side;
+        returns the composed DeviceSnapshot. Shared by the world-1
+        decide_resident and the sharded controller's resident step
+        (each rank's cache holds its uid-hash shard there)."""
         import types
 
-        if base.num_pools > 64:  # same LDS bound as decide(); oracle path
-            _note_fallback("many_pools_oracle",
-                           f"{base.num_pools} pools > 64-pool LDS budget")
-            from ..sim.oracle import OracleSimulator
-
-            snap = cache.build_snapshot(base)
-            return OracleSimulator(
-                snap, over_provision=over_provision).fulfill_pending()
         cap = cache.cap
         res = self._resident
         if res is None or res["cap"] != cap:
@@ -1231,11 +1222,28 @@ class GpuEngine(SimulationEngine):
             self._scatter_rows(res["pod_views"], cache, cache.drain_dirty())
 
         rest_views = self._rest_packer.pack_views(base, self.device)
-        ds = DeviceSnapshot(
+        return DeviceSnapshot(
             **res["pod_views"], **rest_views,
             pool_max_h=np.ascontiguousarray(base.pool_max, dtype=np.int32).copy(),
             pool_cur_h=np.ascontiguousarray(base.pool_cur, dtype=np.int32).copy(),
         )
+
+    def decide_resident(self, cache, base: ClusterSnapshot,
+                        over_provision: int = 0) -> ScaleDecision:
+        """Resident-controller decide: the pod SoA stays in HBM at the
+        watch cache's CAPACITY (tombstone rows are inert); per cycle the
+        dirty rows scatter in (kernel E) and the small node/pool side
+        re-uploads — no full snapshot upload. `base` is the per-cycle
+        node/pool-side snapshot (zero pods)."""
+        if base.num_pools > 64:  # same LDS bound as decide(); oracle path
+            _note_fallback("many_pools_oracle",
+                           f"{base.num_pools} pools > 64-pool LDS budget")
+            from ..sim.oracle import OracleSimulator
+
+            snap = cache.build_snapshot(base)
+            return OracleSimulator(
+                snap, over_provision=over_provision).fulfill_pending()
+        ds = self.resident_ds(cache, base)
         decision = self._decide_fused(ds, base.pool_names, over_provision,
                                       stable_pods=True)
         if decision is None:  # hash overflow: fall back via a full snapshot

@@ -91,7 +91,9 @@ def run_elastic_cycle(engine, snap: ClusterSnapshot, over_provision: int = 0):
         return engine.decide(snap, over_provision=over_provision)
     except RuntimeError as e:
         msg = str(e).lower()
-        if any(k in msg for k in ("nccl", "rccl", "gloo", "connection", "timed out", "timeout")):
+        if any(k in msg for k in ("nccl", "rccl", "gloo", "connection",
+                                  "timed out", "timeout", "closed", "peer",
+                                  "aborted", "econnreset", "broken pipe")):
             if dist.is_available() and dist.is_initialized():
                 try:
                     dist.destroy_process_group()

@@ -73,6 +73,25 @@ class DeltaSnapshotCache:
         self.node_index: Dict[str, int] = {}
         self.pdb_selectors: List[Dict[str, str]] = []
         self.has_label_pairs = False  # set by prime()/load()
+        # uid-hash shard filter (sharded resident controller): when
+        # set, only pods with shard_for(uid) == shard_rank are cached;
+        # watch events for other ranks' pods are dropped at apply.
+        self.shard_rank = 0
+        self.shard_world = 1
+
+    def set_shard(self, rank: int, world: int):
+        """Restrict this cache to one uid-hash shard (prime() must be
+        fed the already-sharded snapshot — parallel.elastic
+        .shard_snapshot uses the same fnv1a64(uid) %% world hash)."""
+        self.shard_rank = int(rank)
+        self.shard_world = max(1, int(world))
+
+    def _in_shard(self, uid: str) -> bool:
+        if self.shard_world <= 1:
+            return True
+        from ..parallel.elastic import shard_for
+
+        return shard_for(uid, self.shard_world) == self.shard_rank
 
     # ------------------------------------------------------------------
     def prime(self, snap: ClusterSnapshot, node_index: Dict[str, int],
@@ -187,6 +206,8 @@ class DeltaSnapshotCache:
         etype = event.get("type")
         obj = event.get("object") or {}
         pod = KubePod(obj)
+        if not self._in_shard(pod.uid):
+            return  # another rank's pod (sharded resident mode)
         if etype in ("ADDED", "MODIFIED"):
             row = self.uid_row.get(pod.uid)
             if row is None:
@@ -215,6 +236,12 @@ class DeltaSnapshotCache:
         n = int(types.shape[0])
         if n == 0:
             return 0
+        if self.shard_world > 1:
+            decoded = self._filter_shard(decoded)
+            types = decoded["types"]
+            n = int(types.shape[0])
+            if n == 0:
+                return 0
         if self.has_label_pairs and "pod_label_pairs" not in decoded:
             # a batch without label columns (mismatched/absent compiled
             # codec) would leave stale label hashes that reassign_pdbs
@@ -265,6 +292,28 @@ class DeltaSnapshotCache:
                 self.dirty.update(rows.tolist())
             return n
         return self._apply_batch_py(decoded)
+
+    def _filter_shard(self, decoded: Dict) -> Dict:
+        """Drop decoded watch rows whose pod uid hashes to another
+        rank's shard (sharded resident mode)."""
+        from ..parallel.elastic import shard_for
+
+        uids = decoded["pod_uids"]
+        keep = np.fromiter(
+            (shard_for(u, self.shard_world) == self.shard_rank for u in uids),
+            bool, count=len(uids))
+        if keep.all():
+            return decoded
+        idx = np.nonzero(keep)[0]
+        out = {}
+        for k, v in decoded.items():
+            if isinstance(v, np.ndarray) and v.shape[:1] == (len(uids),):
+                out[k] = v[idx]
+            elif isinstance(v, list) and len(v) == len(uids):
+                out[k] = [v[i] for i in idx]
+            else:
+                out[k] = v
+        return out
 
     def _apply_batch_py(self, decoded: Dict) -> int:
         """Pure-Python reference for apply_batch's row resolution."""
