@@ -177,6 +177,16 @@ def main():
         def decide_once():
             segs = inner.build_segments(ds)
             return inner.decide_device(ds, segs, snap.pool_names)
+    elif resident and use_gpu:
+        # config #5 sharded: each rank's 1M-pod shard stays resident in
+        # its HBM; a step is local segment build + fixed-size row
+        # gather (RCCL) + device merge + replicated placement — no
+        # per-step snapshot upload at any world size.
+        ds = inner.packer.pack(snap, inner.device)
+        torch.cuda.synchronize()
+
+        def decide_once():
+            return engine.decide_packed(ds, snap.pool_names)
     else:
         def decide_once():
             return engine.decide(snap)

@@ -208,6 +208,21 @@ class ShardedEngine(SimulationEngine):
         return decision
 
     # ------------------------------------------------------------------
+    def decide_packed(self, ds, pool_names: List[str],
+                      over_provision: int = 0) -> ScaleDecision:
+        """Sharded decide over an ALREADY-PACKED device shard (HBM-
+        resident benchmarks / callers that manage their own upload):
+        local segment build -> fixed-size row gather -> device merge ->
+        replicated placement, with the usual agreement check."""
+        decision = self._decide_ds(ds, pool_names, over_provision)
+        if self.active and self.check_agreement:
+            t0 = time.perf_counter()
+            self._assert_agreement(decision, len(pool_names), pool_names)
+            self.comm_stats["allreduce_s"] += time.perf_counter() - t0
+        self.comm_stats["calls"] += 1
+        return decision
+
+    # ------------------------------------------------------------------
     def decide_resident(self, cache, base: ClusterSnapshot,
                         over_provision: int = 0) -> ScaleDecision:
         """Sharded RESIDENT decide (VERDICT r01 #1b): each rank's watch

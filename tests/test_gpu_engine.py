@@ -326,6 +326,21 @@ def test_sharded_engine_world1_gpu_path(engine):
     rep2 = engine.maintain(snap)
     np.testing.assert_array_equal(rep1.node_util_permille, rep2.node_util_permille)
 
+    # the packed-resident entry (bench world>1 path) and the cache-
+    # resident entry must produce the same decision as plain decide
+    ds = engine.packer.pack(snap, engine.device)
+    _assert_decisions_equal(sharded.decide_packed(ds, snap.pool_names), d2)
+
+    from mi355x_autoscaler.snapshot.delta import DeltaSnapshotCache
+    from mi355x_autoscaler.snapshot.encode import encode_snapshot as _enc
+
+    pods, nodes = wrap(pods_json, nodes_json)
+    pools = group_nodes_into_pools(nodes)
+    base = _enc([], nodes, pools)
+    cache = DeltaSnapshotCache()
+    cache.prime(snap, {n.name: i for i, n in enumerate(nodes)})
+    _assert_decisions_equal(sharded.decide_resident(cache, base), d2)
+
 
 @pytest.mark.parametrize("n_pods,n_nodes", [
     (1, 1), (64, 3), (1000, 500), (10000, 10000), (100000, 10000),
