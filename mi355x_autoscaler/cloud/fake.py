@@ -76,7 +76,8 @@ class FakeCloudProvider(CloudProvider):
         with self.kube_state.lock:
             for i in range(new - old):
                 obj = make_node_json(pool, base + 1 + i, **tmpl)
-                self.kube_state.nodes[obj["metadata"]["name"]] = obj
+                # via add_node so node watchers see the ADDED event
+                self.kube_state.add_node(obj)
 
     def delete_node_resources(self, node_name: str, dry_run: bool = False) -> bool:
         with self.lock:

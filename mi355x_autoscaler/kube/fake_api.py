@@ -26,6 +26,10 @@ class FakeKubeState:
 
     def __init__(self):
         self.lock = threading.RLock()
+        # watchers block on this instead of sleep-polling: every emit
+        # wakes the streams, so watch latency is scheduler-bound (the
+        # reaction-time benchmark measures this path)
+        self.cond = threading.Condition(self.lock)
         self.pods: Dict[str, Dict] = {}   # "<ns>/<name>" -> obj
         self.nodes: Dict[str, Dict] = {}
         self.pdbs: Dict[str, Dict] = {}
@@ -37,6 +41,7 @@ class FakeKubeState:
         self.patched_nodes: List[str] = []
         self.resource_version: int = 0
         self.pod_events: List[Dict] = []  # [{rv, type, object}]
+        self.node_events: List[Dict] = []
 
     def load(self, pods: List[Dict], nodes: List[Dict], pdbs: Optional[List[Dict]] = None):
         with self.lock:
@@ -47,7 +52,9 @@ class FakeKubeState:
             self.nodes = {n["metadata"]["name"]: copy.deepcopy(n) for n in nodes}
             self.pdbs = {d["metadata"]["name"]: copy.deepcopy(d) for d in (pdbs or [])}
             self.pod_events = []
+            self.node_events = []
             self.resource_version += 1
+            self.cond.notify_all()
 
     # -- watch-visible mutations ---------------------------------------
     def _emit(self, etype: str, obj: Dict):
@@ -57,6 +64,21 @@ class FakeKubeState:
         self.pod_events.append(
             {"rv": self.resource_version, "type": etype, "object": obj}
         )
+        self.cond.notify_all()
+
+    def _emit_node(self, etype: str, obj: Dict):
+        self.resource_version += 1
+        obj = copy.deepcopy(obj)
+        obj.setdefault("metadata", {})["resourceVersion"] = str(self.resource_version)
+        self.node_events.append(
+            {"rv": self.resource_version, "type": etype, "object": obj}
+        )
+        self.cond.notify_all()
+
+    def add_node(self, node: Dict):
+        with self.lock:
+            self.nodes[node["metadata"]["name"]] = copy.deepcopy(node)
+            self._emit_node("ADDED", node)
 
     def add_pod(self, pod: Dict):
         with self.lock:
@@ -76,6 +98,29 @@ class FakeKubeState:
             pod = self.pods.pop(key, None)
             if pod is not None:
                 self._emit("DELETED", pod)
+
+
+def _event_stream(state: FakeKubeState, log: str, start_rv: int):
+    """JSONL watch stream over an event log; wakes on state.cond
+    (no poll-sleep on the reaction path), honors drop_watch_after."""
+    import json as _json
+
+    sent = 0
+    last_rv = start_rv
+    while True:
+        batch = []
+        with state.lock:
+            for e in getattr(state, log):
+                if e["rv"] > last_rv:
+                    batch.append(e)
+            if not batch:
+                state.cond.wait(timeout=0.5)
+        for e in batch:
+            if state.drop_watch_after >= 0 and sent >= state.drop_watch_after:
+                return  # injected disconnect
+            last_rv = e["rv"]
+            sent += 1
+            yield _json.dumps({"type": e["type"], "object": e["object"]}) + "\n"
 
 
 def build_app(state: FakeKubeState) -> FastAPI:
@@ -98,27 +143,8 @@ def build_app(state: FakeKubeState) -> FastAPI:
 
             start_rv = int(resourceVersion or 0)
 
-            def stream():
-                import json as _json
-
-                sent = 0
-                last_rv = start_rv
-                while True:
-                    batch = []
-                    with state.lock:
-                        for e in state.pod_events:
-                            if e["rv"] > last_rv:
-                                batch.append(e)
-                    for e in batch:
-                        if state.drop_watch_after >= 0 and sent >= state.drop_watch_after:
-                            return  # injected disconnect
-                        last_rv = e["rv"]
-                        sent += 1
-                        yield _json.dumps({"type": e["type"], "object": e["object"]}) + "\n"
-                    if not batch:
-                        time.sleep(0.02)
-
-            return StreamingResponse(stream(), media_type="application/json")
+            return StreamingResponse(_event_stream(state, "pod_events", start_rv),
+                                     media_type="application/json")
         with state.lock:
             return {
                 "kind": "PodList",
@@ -127,9 +153,16 @@ def build_app(state: FakeKubeState) -> FastAPI:
             }
 
     @app.get("/api/v1/nodes")
-    def list_nodes():
+    def list_nodes(watch: bool = False, resourceVersion: str = "0"):
+        if watch:
+            from starlette.responses import StreamingResponse
+
+            return StreamingResponse(
+                _event_stream(state, "node_events", int(resourceVersion or 0)),
+                media_type="application/json")
         with state.lock:
-            return {"kind": "NodeList", "items": list(state.nodes.values()), "metadata": {}}
+            return {"kind": "NodeList", "items": list(state.nodes.values()),
+                    "metadata": {"resourceVersion": str(state.resource_version)}}
 
     @app.get("/apis/policy/v1/poddisruptionbudgets")
     def list_pdbs():
@@ -147,6 +180,7 @@ def build_app(state: FakeKubeState) -> FastAPI:
             if unsched is not None:
                 node.setdefault("spec", {})["unschedulable"] = bool(unsched)
             state.patched_nodes.append(name)
+            state._emit_node("MODIFIED", node)
             return node
 
     @app.delete("/api/v1/namespaces/{ns}/pods/{name}")
@@ -162,7 +196,9 @@ def build_app(state: FakeKubeState) -> FastAPI:
     @app.delete("/api/v1/nodes/{name}")
     def delete_node(name: str):
         with state.lock:
-            state.nodes.pop(name, None)
+            node = state.nodes.pop(name, None)
+            if node is not None:
+                state._emit_node("DELETED", node)
             state.deleted_nodes.append(name)
             return {"status": "Success"}
 

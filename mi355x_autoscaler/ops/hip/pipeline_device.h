@@ -535,6 +535,17 @@ __device__ inline void dev_binpack(
 // epilogue. Block size must be UNITS_BLOCK (== 256). `lds_arena`
 // (lds_cap runs * 6 i64) may be null when lds_cap == 0 — then the
 // global scratch is used unconditionally.
+//
+// Run store (round 2): runs live in an ARRAY indexed
+// by an explicit per-pool visit-order list `ord` (replacing the
+// round-1 linked list whose single-lane walk dominated config #2).
+// Per segment the whole wave evaluates run capacities in visit order,
+// a wave scan turns FFD into the same scan-threshold uniform step the
+// node FFD uses, full runs update in parallel, and the one boundary
+// run splits with a wave-parallel ord shift. Semantics are the
+// oracle's exactly: units visited in creation order, splits keep the
+// [full-takers, partial, untouched] adjacency (sim/oracle.py
+// fulfill_pending phase 2).
 #define UNITS_MAX_POOLS 64
 #define UNITS_BLOCK 256
 #define UNITS_NW (UNITS_BLOCK / WAVE)
@@ -557,13 +568,14 @@ __device__ inline void dev_units(
   const int lane = tid & (WAVE - 1);
   const int w = tid / WAVE;
 
-  __shared__ int32_t head[UNITS_MAX_POOLS];
-  __shared__ int32_t tailp[UNITS_MAX_POOLS];
+  __shared__ int32_t pool_nseg[UNITS_MAX_POOLS];   // segments per pool
+  __shared__ int32_t ord_base_s[UNITS_MAX_POOLS];  // per-pool ord region
+  __shared__ int32_t pool_R[UNITS_MAX_POOLS];      // live runs per pool
   __shared__ int64_t opened[UNITS_MAX_POOLS];
   __shared__ int64_t resid[UNITS_MAX_POOLS * NRES];
   __shared__ int32_t wave_off[UNITS_NW + 1];
-  // the serial walks are latency-bound on chained global loads of the
-  // segment metadata; stage it in LDS cooperatively first
+  __shared__ int32_t wave_alloc[UNITS_NW];         // run-slot allocator
+  // the metadata loads are latency-bound; stage them in LDS first
 #define U_PRELOAD 512
   __shared__ int32_t u_pool[U_PRELOAD];
   __shared__ int64_t u_cnt[U_PRELOAD];
@@ -571,8 +583,8 @@ __device__ inline void dev_units(
   __shared__ int32_t u_req[U_PRELOAD * NRES];
 
   for (int k = tid; k < K; k += UNITS_BLOCK) {
-    head[k] = -1;
-    tailp[k] = -1;
+    pool_nseg[k] = 0;
+    pool_R[k] = 0;
     opened[k] = 0;
   }
   for (int q = tid; q < K * NRES; q += UNITS_BLOCK) resid[q] = 0;
@@ -584,179 +596,266 @@ __device__ inline void dev_units(
     for (int r = 0; r < NRES; ++r) u_req[g * NRES + r] = seg_req[g * NRES + r];
   }
   __syncthreads();
-
-  // per-wave arena need: 4 runs per owned segment + fixed slack
-  int cnt = 0;
-  for (int g = lane; g < G; g += WAVE) {
+  for (int g = tid; g < G; g += UNITS_BLOCK) {
     const int k = g < U_PRELOAD ? u_pool[g] : seg_pool[g];
-    if (k >= 0 && (k % UNITS_NW) == w) ++cnt;
+    if (k >= 0) atomicAdd(&pool_nseg[k], 1);
   }
-#pragma unroll
-  for (int off = WAVE / 2; off; off >>= 1) cnt += __shfl_down(cnt, off, WAVE);
-  if (lane == 0) wave_off[w] = 4 * cnt + UNITS_PAD;
   __syncthreads();
   if (tid == 0) {
+    // wave regions sized 4 runs per owned segment + fixed slack; the
+    // per-pool ord (visit-order) regions are carved inside them —
+    // runs per pool <= 4 * its segments (<=2 split + <=2 new-unit
+    // runs each), so ord and run-slot usage both fit the region.
+    int woff[UNITS_NW];
+    for (int i = 0; i < UNITS_NW; ++i) woff[i] = 0;
+    for (int k = 0; k < K; ++k) woff[k % UNITS_NW] += 4 * pool_nseg[k];
     int off = 0;
     for (int i = 0; i < UNITS_NW; ++i) {
-      const int n = wave_off[i];
       wave_off[i] = off;
-      off += n;
+      wave_alloc[i] = off;
+      off += woff[i] + UNITS_PAD;
     }
     wave_off[UNITS_NW] = off;
+    int cur[UNITS_NW];
+    for (int i = 0; i < UNITS_NW; ++i) cur[i] = wave_off[i];
+    for (int k = 0; k < K; ++k) {
+      const int i = k % UNITS_NW;
+      ord_base_s[k] = cur[i];
+      cur[i] += 4 * pool_nseg[k];
+    }
   }
   __syncthreads();
   const int total_need = wave_off[UNITS_NW];
   const bool use_lds = lds_arena != nullptr && total_need <= lds_cap;
   const int cap = use_lds ? lds_cap : RTOT;
   int64_t* arena = use_lds ? lds_arena : scratch;
-  int64_t* run_res = arena;               // [cap][NRES]
-  int64_t* run_cnt = arena + (int64_t)cap * NRES;
-  int64_t* run_next = run_cnt + cap;
+  int64_t* run_res = arena;                        // [cap][NRES]
+  int64_t* run_cnt = arena + (int64_t)cap * NRES;  // [cap]
+  int64_t* ord = run_cnt + cap;                    // [cap] visit order
   const int a_hi = wave_off[w + 1];
 
-  if (lane == 0) {
-    int alloc = wave_off[w];
-    for (int g = 0; g < G; ++g) {
-      const bool pre = g < U_PRELOAD;
-      const int k = pre ? u_pool[g] : seg_pool[g];
-      const bool mine = (k >= 0) ? ((k % UNITS_NW) == w) : (w == 0);
-      if (!mine) continue;
-      int64_t rem = (pre ? u_cnt[g] : seg_count[g]) -
-                    (pre ? u_placed[g] : placed[g]);
+  for (int g = 0; g < G; ++g) {
+    const bool pre = g < U_PRELOAD;
+    const int k = pre ? u_pool[g] : seg_pool[g];
+    if (k < 0) {
+      if (w == 0 && lane == 0) {
+        const int64_t rem0 = (pre ? u_cnt[g] : seg_count[g]) -
+                             (pre ? u_placed[g] : placed[g]);
+        placed_new[g] = 0;
+        unful[g] = rem0 > 0 ? rem0 : 0;
+      }
+      continue;
+    }
+    if ((k % UNITS_NW) != w) continue;
+    int64_t rem = (pre ? u_cnt[g] : seg_count[g]) -
+                  (pre ? u_placed[g] : placed[g]);
+    if (lane == 0) {
       placed_new[g] = 0;
       unful[g] = 0;
-      if (rem <= 0) continue;
-      if (k < 0) {
-        unful[g] = rem;
-        continue;
-      }
-      if (alloc + 4 > a_hi) {  // arena bound: a segment allocates <=4 runs
-        unful[g] = rem;        // (unreachable when the scratch honors the
-        continue;              //  4G+pad sizing; kept so a bound bug can
-      }                        //  never corrupt the lists)
-      int64_t rq[NRES];
+    }
+    if (rem <= 0) continue;
+    if (wave_alloc[w] + 4 > a_hi) {  // arena bound: <=4 new runs/segment
+      if (lane == 0) unful[g] = rem; // (unreachable when the scratch
+      continue;                      //  honors 4G+pad; kept as a fence)
+    }
+    int64_t rq[NRES];
 #pragma unroll
-      for (int r = 0; r < NRES; ++r)
-        rq[r] = pre ? u_req[g * NRES + r] : seg_req[g * NRES + r];
+    for (int r = 0; r < NRES; ++r)
+      rq[r] = pre ? u_req[g * NRES + r] : seg_req[g * NRES + r];
+    if (lane == 0) {
 #pragma unroll
       for (int r = 0; r < NRES; ++r) resid[k * NRES + r] += rq[r] * rem;
-      const int64_t need0 = rem;
+    }
+    const int64_t need0 = rem;
+    const int ob = ord_base_s[k];
+    int R = pool_R[k];
 
-      // 1) walk open runs in creation order
-      for (int idx = head[k]; idx != -1 && rem > 0; idx = (int)run_next[idx]) {
-        int64_t m = (int64_t)1 << 31;
+    // 1) wave-parallel FFD over open runs in visit order: capacity
+    //    scan -> threshold take (the same uniform step as the node
+    //    FFD); full runs update in parallel, the single boundary run
+    //    records itself for the split below.
+    int64_t carry = 0;
+    int b_have = 0, b_slot = -1, b_pos = -1;
+    int64_t b_take = 0, b_m = 0, b_cnt = 0, b_res[NRES] = {0};
+    for (int base = 0; base < R && carry < rem; base += WAVE) {
+      const int p = base + lane;
+      int slot = -1;
+      int64_t m = 0, c = 0, fr[NRES] = {0};
+      if (p < R) {
+        slot = (int)ord[ob + p];
+        c = run_cnt[slot];
         bool neg = false;
+        m = (int64_t)1 << 31;
 #pragma unroll
         for (int r = 0; r < NRES; ++r) {
-          const int64_t fr = run_res[(int64_t)idx * NRES + r];
-          if (fr < 0) neg = true;
+          fr[r] = run_res[(int64_t)slot * NRES + r];
+          if (fr[r] < 0) neg = true;
           if (rq[r] > 0) {
-            const int64_t q = fr / rq[r];
+            const int64_t q = fr[r] / rq[r];
             m = q < m ? q : m;
           }
         }
-        if (neg || m <= 0) continue;
-        const int64_t c = run_cnt[idx];
-        if (rem >= m * c) {  // every unit of the run takes m
+        if (neg || m < 0) m = 0;
+      }
+      // run capacity, clamped to rem+1 so the 64-lane scan cannot
+      // overflow i64; take/boundary math is unchanged (a clamped run
+      // can never appear fully consumed: take <= rem < rem+1)
+      const int64_t t_true = m * c;
+      const int64_t t = t_true > rem + 1 ? rem + 1 : t_true;
+      const int64_t inc = wave_inclusive_scan_i64(t);
+      const int64_t S = carry + inc - t;
+      int64_t take = rem - S;
+      take = take < 0 ? 0 : (take > t ? t : take);
+      if (t > 0 && take == t && t == t_true) {
+        // fully consumed: every node of the run takes m
 #pragma unroll
-          for (int r = 0; r < NRES; ++r) run_res[(int64_t)idx * NRES + r] -= m * rq[r];
-          rem -= m * c;
-          continue;
-        }
-        // split: nf full takers, maybe one partial, rest untouched
-        const int64_t nf = rem / m;
-        const int64_t part = rem - nf * m;
-        int64_t untouched = c - nf - (part > 0 ? 1 : 0);
-        // node idx becomes the full-takers run (or partial if nf==0)
-        const int64_t nxt = run_next[idx];
-        int cur = idx;
+        for (int r = 0; r < NRES; ++r)
+          run_res[(int64_t)slot * NRES + r] = fr[r] - m * rq[r];
+      } else if (take > 0) {
+        b_have = 1;
+        b_slot = slot;
+        b_pos = p;
+        b_take = take;
+        b_m = m;
+        b_cnt = c;
+#pragma unroll
+        for (int r = 0; r < NRES; ++r) b_res[r] = fr[r];
+      }
+      carry += __shfl(inc, WAVE - 1);
+    }
+    const int64_t placed_runs = carry < rem ? carry : rem;
+
+    // 2) boundary split (at most one run): nf full takers keep the
+    //    slot; partial/untouched allocate new runs inserted right
+    //    after it in the visit order (wave-parallel ord shift) —
+    //    exactly the oracle's unit adjacency.
+    const uint64_t bmask = __ballot(b_have);
+    if (bmask) {
+      const int bl = __ffsll((unsigned long long)bmask) - 1;
+      const int slot = __shfl(b_slot, bl);
+      const int pos = __shfl(b_pos, bl);
+      const int64_t take = __shfl(b_take, bl);
+      const int64_t m = __shfl(b_m, bl);
+      const int64_t c = __shfl(b_cnt, bl);
+      int64_t res0[NRES];
+#pragma unroll
+      for (int r = 0; r < NRES; ++r) res0[r] = __shfl(b_res[r], bl);
+      const int64_t nf = m > 0 ? take / m : 0;
+      const int64_t part = take - nf * m;
+      const int64_t untouched = c - nf - (part > 0 ? 1 : 0);
+      const int ins =
+          ((nf > 0 && part > 0) ? 1 : 0) + (untouched > 0 ? 1 : 0);
+      int s1 = -1, s2 = -1;
+      if (lane == 0) {
+        int a = wave_alloc[w];
         if (nf > 0) {
-          run_cnt[cur] = nf;
+          run_cnt[slot] = nf;
 #pragma unroll
-          for (int r = 0; r < NRES; ++r) run_res[(int64_t)cur * NRES + r] -= m * rq[r];
+          for (int r = 0; r < NRES; ++r)
+            run_res[(int64_t)slot * NRES + r] = res0[r] - m * rq[r];
+          if (part > 0) {
+            s1 = a++;
+            run_cnt[s1] = 1;
+#pragma unroll
+            for (int r = 0; r < NRES; ++r)
+              run_res[(int64_t)s1 * NRES + r] = res0[r] - part * rq[r];
+          }
+        } else if (part > 0) {
+          run_cnt[slot] = 1;
+#pragma unroll
+          for (int r = 0; r < NRES; ++r)
+            run_res[(int64_t)slot * NRES + r] = res0[r] - part * rq[r];
         }
-        if (part > 0) {
+        if (untouched > 0) {
+          s2 = a++;
+          run_cnt[s2] = untouched;
+#pragma unroll
+          for (int r = 0; r < NRES; ++r)
+            run_res[(int64_t)s2 * NRES + r] = res0[r];
+        }
+        wave_alloc[w] = a;
+      }
+      s1 = __shfl(s1, 0);
+      s2 = __shfl(s2, 0);
+      if (ins > 0) {
+        // shift ord[pos+1 .. R-1] right by ins. High-to-low chunks:
+        // within a chunk the wave's lockstep load completes for every
+        // lane before the store issues; across chunks the higher one
+        // finished first and the ranges don't collide.
+        const int tail = R - 1 - (pos + 1);
+        for (int base = (tail / WAVE) * WAVE; base >= 0; base -= WAVE) {
+          const int i = pos + 1 + base + lane;
+          int64_t v = 0;
+          const bool act = i >= pos + 1 && i < R;
+          if (act) v = ord[ob + i];
+          __builtin_amdgcn_wave_barrier();
+          if (act) ord[ob + i + ins] = v;
+          __builtin_amdgcn_wave_barrier();
+        }
+        if (lane == 0) {
+          int qp = pos + 1;
+          if (s1 >= 0) ord[ob + qp++] = s1;
+          if (s2 >= 0) ord[ob + qp++] = s2;
+        }
+        R += ins;
+      }
+    }
+    rem -= placed_runs;
+
+    // 3) open new virtual units while the pool cap allows (O(1):
+    //    one full-takers run + at most one partial run)
+    if (lane == 0) {
+      if (rem > 0) {
+        int64_t per_unit = (int64_t)1 << 31;
+        bool fits = true;
+#pragma unroll
+        for (int r = 0; r < NRES; ++r) {
+          const int64_t ucap = pool_unit[k * NRES + r];
+          if (rq[r] > ucap) fits = false;
+          if (rq[r] > 0) {
+            const int64_t q = ucap / rq[r];
+            per_unit = q < per_unit ? q : per_unit;
+          }
+        }
+        if (fits && per_unit > 0) {
+          int32_t mn = pool_max[k] - pool_cur[k];  // inline max-new
+          if (mn < 0) mn = 0;
+          int64_t room = (int64_t)mn - opened[k];
+          if (room < 0) room = 0;
+          int64_t nf = rem / per_unit;
+          nf = nf < room ? nf : room;
+          int a = wave_alloc[w];
           if (nf > 0) {
-            const int nn = alloc++;
-            run_next[cur] = nn;
-            run_next[nn] = nxt;
+            const int nn = a++;
 #pragma unroll
             for (int r = 0; r < NRES; ++r)
               run_res[(int64_t)nn * NRES + r] =
-                  run_res[(int64_t)idx * NRES + r] + m * rq[r] - part * rq[r];
-            run_cnt[nn] = 1;
-            if (tailp[k] == cur) tailp[k] = nn;
-            cur = nn;
-          } else {
-            run_cnt[cur] = 1;
-#pragma unroll
-            for (int r = 0; r < NRES; ++r) run_res[(int64_t)cur * NRES + r] -= part * rq[r];
+                  (int64_t)pool_unit[k * NRES + r] - per_unit * rq[r];
+            run_cnt[nn] = nf;
+            ord[ob + R++] = nn;
+            opened[k] += nf;
+            rem -= nf * per_unit;
+            room -= nf;
           }
-        }
-        if (untouched > 0) {
-          const int nn = alloc++;
-          run_next[cur] = nn;
-          run_next[nn] = nxt;
+          if (rem > 0 && room > 0) {  // one partial unit
+            const int nn = a++;
 #pragma unroll
-          for (int r = 0; r < NRES; ++r)
-            run_res[(int64_t)nn * NRES + r] =
-                run_res[(int64_t)idx * NRES + r] +
-                ((cur == idx && part > 0) ? part : m) * rq[r];
-          run_cnt[nn] = untouched;
-          if (tailp[k] == cur) tailp[k] = nn;
-        }
-        rem = 0;
-        break;
-      }
-
-      // 2) open new units while the pool cap allows
-      int64_t per_unit = (int64_t)1 << 31;
-      bool fits = true;
-#pragma unroll
-      for (int r = 0; r < NRES; ++r) {
-        const int64_t ucap = pool_unit[k * NRES + r];  // not the arena cap
-        if (rq[r] > ucap) fits = false;
-        if (rq[r] > 0) {
-          const int64_t q = ucap / rq[r];
-          per_unit = q < per_unit ? q : per_unit;
-        }
-      }
-      if (rem > 0 && fits && per_unit > 0) {
-        int32_t mn = pool_max[k] - pool_cur[k];  // inline max-new
-        if (mn < 0) mn = 0;
-        int64_t room = (int64_t)mn - opened[k];
-        if (room < 0) room = 0;
-        int64_t nf = rem / per_unit;
-        nf = nf < room ? nf : room;
-        if (nf > 0) {
-          const int nn = alloc++;
-#pragma unroll
-          for (int r = 0; r < NRES; ++r)
-            run_res[(int64_t)nn * NRES + r] = (int64_t)pool_unit[k * NRES + r] - per_unit * rq[r];
-          run_cnt[nn] = nf;
-          run_next[nn] = -1;
-          if (tailp[k] == -1) head[k] = nn; else run_next[tailp[k]] = nn;
-          tailp[k] = nn;
-          opened[k] += nf;
-          rem -= nf * per_unit;
-          room -= nf;
-        }
-        if (rem > 0 && room > 0) {  // one partial unit
-          const int nn = alloc++;
-#pragma unroll
-          for (int r = 0; r < NRES; ++r)
-            run_res[(int64_t)nn * NRES + r] = (int64_t)pool_unit[k * NRES + r] - rem * rq[r];
-          run_cnt[nn] = 1;
-          run_next[nn] = -1;
-          if (tailp[k] == -1) head[k] = nn; else run_next[tailp[k]] = nn;
-          tailp[k] = nn;
-          opened[k] += 1;
-          rem = 0;
+            for (int r = 0; r < NRES; ++r)
+              run_res[(int64_t)nn * NRES + r] =
+                  (int64_t)pool_unit[k * NRES + r] - rem * rq[r];
+            run_cnt[nn] = 1;
+            ord[ob + R++] = nn;
+            opened[k] += 1;
+            rem = 0;
+          }
+          wave_alloc[w] = a;
         }
       }
       placed_new[g] = need0 - rem;
       unful[g] = rem;
+      pool_R[k] = R;
     }
+    __builtin_amdgcn_wave_barrier();
   }
   __syncthreads();
   for (int k = tid; k < K; k += UNITS_BLOCK) units_cnt[k] = opened[k];
