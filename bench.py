@@ -99,6 +99,111 @@ def build_snapshot_gpu1k(rank: int, pods_per_rank: int):
     return encode_snapshot(pods, nodes, pools)
 
 
+def run_e2e_reaction(args):
+    """END-TO-END reaction benchmark: pods POSTed
+    to the live in-process fake k8s API -> watch stream -> C++ batch
+    decode -> delta cache -> decision. One step = one 10k-pod burst;
+    the timed quantity is pod-POST to decision-in-hand, i.e. the
+    BASELINE 'scale-up reaction' INCLUDING ingest, not just decide."""
+    import statistics as _stats
+
+    from mi355x_autoscaler.cloud.fake import FakeCloudProvider
+    from mi355x_autoscaler.ctl.loop import Cluster, ClusterOptions
+    from mi355x_autoscaler.kube.client import KubeApiClient
+    from mi355x_autoscaler.kube.fake_api import FakeKubeServer
+    from mi355x_autoscaler.sim.synth import make_pod_json
+    from mi355x_autoscaler.utils.notification import Notifier
+
+    use_gpu = args.backend != "cpu" and torch.cuda.is_available()
+    wave_n = args.pods_per_gpu or 10_000
+    _, nodes_json = burst_trace(
+        0, n_nodes_per_pool=POOL_SIZES, pools=POOLS, seed=7)
+
+    with FakeKubeServer() as srv:
+        srv.state.load([], nodes_json)
+        kube = KubeApiClient(srv.base_url)
+        provider = FakeCloudProvider(kube_state=srv.state)
+        if use_gpu:
+            from mi355x_autoscaler.ops.engine_gpu import GpuEngine
+
+            engine = GpuEngine("cuda:0")
+        else:
+            engine = CpuEngine()
+        opts = ClusterOptions(sleep_seconds=0, no_scale=True,
+                              no_maintenance=True, watch=True,
+                              idle_threshold_seconds=10**9)
+        cluster = Cluster(kube, provider, engine, options=opts,
+                          notifier=Notifier(dry_run=True))
+        cluster.loop_logic()  # primes pod + node watches
+
+        rng_waves = []
+        for s in range(args.warmup + args.steps):
+            wave = [make_pod_json(f"burst{s}-{i}", cpu_milli=100,
+                                  mem_mib=256, pending=True,
+                                  gpu=1 if i % 20 == 0 else 0)
+                    for i in range(wave_n)]
+            # the POST body a real client would send, prepared untimed
+            lines = [json.dumps({"type": "ADDED", "object": p}) + "\n"
+                     for p in wave]
+            rng_waves.append((wave, lines))
+
+        reactions = []
+        ingest_ms = decide_ms = 0.0
+        for s, (wave, lines) in enumerate(rng_waves):
+            t0 = time.perf_counter()
+            srv.state.bulk_add_pods(wave, copy_objs=False, lines=lines)
+            while True:
+                res = cluster.loop_logic()
+                if res.pods_pending >= wave_n:
+                    break
+            dt = time.perf_counter() - t0
+            if s >= args.warmup:
+                reactions.append(dt)
+                ingest_ms += res.phase_times.get("ingest", 0.0) * 1e3
+                ingest_ms += res.phase_times.get("encode", 0.0) * 1e3
+                decide_ms += res.phase_times.get("decide", 0.0) * 1e3
+            # untimed cleanup: delete the wave, absorb the deletions
+            srv.state.bulk_remove_pods(
+                [f"default/{p['metadata']['name']}" for p in wave])
+            deadline = time.perf_counter() + 60
+            while cluster.loop_logic().pods_pending > 0:
+                if time.perf_counter() > deadline:
+                    raise RuntimeError("cleanup did not converge in 60 s")
+
+        p50 = _stats.median(reactions) * 1e3
+        p99 = sorted(reactions)[min(len(reactions) - 1,
+                                    int(0.99 * len(reactions)))] * 1e3
+        mean = sum(reactions) / len(reactions)
+        result = {
+            "metric": "e2e_reaction_pods_per_sec",
+            "value": round(wave_n / mean, 1),
+            "unit": "pods/s",
+            "n_gpus": 1 if use_gpu else 0,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(mean * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "int32+exact-f32-mfma",
+            "data": "synthetic",
+            "config": {
+                "model": f"{wave_n}-pod burst POSTed to live fake API -> "
+                         "watch -> decision (reaction incl. ingest)",
+                "pods_per_gpu": wave_n,
+                "nodes": len(nodes_json),
+                "pools": len(POOLS),
+                "parallelism": "single",
+                "backend": "gpu" if use_gpu else "cpu-oracle",
+                "reaction_p50_ms": round(p50, 3),
+                "reaction_p99_ms": round(p99, 3),
+                "phase_ingest_ms_mean": round(ingest_ms / len(reactions), 3),
+                "phase_decide_ms_mean": round(decide_ms / len(reactions), 3),
+            },
+        }
+        print(json.dumps(result))
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -107,13 +212,19 @@ def main():
     ap.add_argument("--pods-per-gpu", type=int, default=None)
     ap.add_argument("--backend", choices=["auto", "gpu", "cpu"], default="auto")
     ap.add_argument("--bench-config",
-                    choices=["burst10k", "burst1m", "maintain500", "gpu1k"],
+                    choices=["burst10k", "burst1m", "maintain500", "gpu1k",
+                             "e2e10k"],
                     default="burst10k",
                     help="burst10k = BASELINE config #3 (default); burst1m = config #5 "
                          "(1M pods, HBM-resident snapshot); maintain500 = config #4 "
                          "(drain scoring over 500 under-utilized MI355X nodes with PDBs); "
-                         "gpu1k = config #2 (1k amd.com/gpu=1 pods, single MI355X pool)")
+                         "gpu1k = config #2 (1k amd.com/gpu=1 pods, single MI355X pool); "
+                         "e2e10k = end-to-end reaction: pod-POST on the live fake API "
+                         "-> watch ingest -> decision")
     args = ap.parse_args()
+    if args.bench_config == "e2e10k":
+        run_e2e_reaction(args)
+        return
     if args.pods_per_gpu is None:
         args.pods_per_gpu = {"burst1m": 1_000_000, "gpu1k": 1_000}.get(
             args.bench_config, 10_000)

@@ -121,6 +121,7 @@ class Cluster:
         self._idle_since: Dict[str, float] = {}
         self._cycle = 0
         self._watch_state: Optional[Dict] = None
+        self._node_watch: Optional[Dict] = None
         # node/pool-side base-snapshot cache for the watch-resident path
         # (re-encoding 1k+ nodes in Python costs ~100 ms/cycle otherwise)
         self._base_cache: Optional[tuple] = None
@@ -150,10 +151,20 @@ class Cluster:
         result = CycleResult()
 
         use_codec = self.opts.native_codec and _codec_available()
+        from ..parallel import dist as _pdist
+
         with t.phase("ingest"):
-            nodes_json = self.kube.list_nodes()
+            if self.opts.watch and _pdist.world_size() == 1:
+                # node watch: one relist, then events
+                # patch the cached node set — no per-cycle node list
+                # round trip. world>1 keeps the relist: nodes are a
+                # REPLICATED input, and per-rank event timing must not
+                # diverge the replicated tables mid-cycle.
+                nodes = self._nodes_via_watch()
+            else:
+                nodes_json = self.kube.list_nodes()
+                nodes = [KubeNode(n) for n in nodes_json]
             pdbs_json = self.kube.list_pdbs()
-            nodes = [KubeNode(n) for n in nodes_json]
             pdbs = [PodDisruptionBudget.from_json(d) for d in pdbs_json]
             pools = group_nodes_into_pools(
                 nodes, ignore_pools=self.opts.ignore_pools, max_size=self.opts.max_pool_size
@@ -415,6 +426,60 @@ class Cluster:
         self.metrics.count("watch_events", applied)
         base = self._base_snapshot(nodes, pools, pdbs)
         return cache.build_snapshot(base)
+
+    def _nodes_via_watch(self) -> List[KubeNode]:
+        """Node set maintained under watch events (world-1): one full
+        relist primes the cache; ADDED/MODIFIED/DELETED events patch
+        it. Wrapped KubeNode list is rebuilt only when something
+        changed (re-wrapping 1k+ nodes costs ~ms per cycle)."""
+        import queue as _q
+        import threading
+
+        nw = self._node_watch
+        if nw is not None and nw["broken"]:
+            logger.warning("node watch dropped; re-priming from a relist")
+            nw = self._node_watch = None
+        if nw is None:
+            nodes_json = self.kube.list_nodes()
+            cache = {n["metadata"]["name"]: n for n in nodes_json}
+            q: _q.SimpleQueue = _q.SimpleQueue()
+            th = threading.Thread(target=self._node_pump, args=(q,), daemon=True)
+            th.start()
+            self._node_watch = nw = {
+                "q": q, "thread": th, "cache": cache, "broken": False,
+                "wrapped": None,
+            }
+        changed = False
+        while True:
+            try:
+                ev = nw["q"].get_nowait()
+            except _q.Empty:
+                break
+            if ev.get("type") == "__STREAM_END__":
+                nw["broken"] = True  # serve the cache; re-prime next cycle
+                break
+            obj = ev.get("object") or {}
+            name = (obj.get("metadata") or {}).get("name")
+            if not name:
+                continue
+            if ev.get("type") == "DELETED":
+                nw["cache"].pop(name, None)
+            else:
+                nw["cache"][name] = obj
+            changed = True
+        if changed:
+            self.metrics.count("node_watch_events")
+        if changed or nw["wrapped"] is None:
+            nw["wrapped"] = [KubeNode(n) for n in nw["cache"].values()]
+        return nw["wrapped"]
+
+    def _node_pump(self, q):
+        try:
+            for ev in self.kube.watch("/api/v1/nodes", resource_version="0"):
+                q.put(ev)
+        except Exception:
+            logger.exception("node watch stream error")
+        q.put({"type": "__STREAM_END__"})
 
     def _watch_pump(self, q):
         """Background thread: stream pod watch events into the queue —

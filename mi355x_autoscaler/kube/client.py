@@ -129,7 +129,9 @@ class KubeApiClient:
         resp = self.session.get(
             f"{self.base_url}{path}", params=params, stream=True, timeout=None
         )
-        for line in resp.iter_lines():
+        # 1 MiB read chunks: a 10k-pod burst (~5 MB of JSONL) is a few
+        # socket reads, not thousands of 512 B ones
+        for line in resp.iter_lines(chunk_size=1 << 20):
             if line:
                 yield line
 

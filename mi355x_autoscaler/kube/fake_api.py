@@ -58,11 +58,14 @@ class FakeKubeState:
 
     # -- watch-visible mutations ---------------------------------------
     def _emit(self, etype: str, obj: Dict):
+        import json as _json
+
         self.resource_version += 1
         obj = copy.deepcopy(obj)
         obj.setdefault("metadata", {})["resourceVersion"] = str(self.resource_version)
         self.pod_events.append(
-            {"rv": self.resource_version, "type": etype, "object": obj}
+            {"rv": self.resource_version, "type": etype, "object": obj,
+             "line": _json.dumps({"type": etype, "object": obj}) + "\n"}
         )
         self.cond.notify_all()
 
@@ -79,6 +82,60 @@ class FakeKubeState:
         with self.lock:
             self.nodes[node["metadata"]["name"]] = copy.deepcopy(node)
             self._emit_node("ADDED", node)
+
+    def bulk_add_pods(self, pods: List[Dict], copy_objs: bool = True,
+                      lines: Optional[List[str]] = None):
+        """Burst injection under ONE lock/notify (the reaction-time
+        benchmark's POST); copy_objs=False trusts the caller not to
+        mutate the objects afterwards. `lines` are optional
+        pre-serialized watch lines ('{"type":"ADDED","object":...}\n')
+        — the real apiserver receives serialized bodies on the wire,
+        so a POSTing client can hand them over instead of paying a
+        Python re-serialization here (those lines then lack the bumped
+        resourceVersion, which watchers don't consume)."""
+        import copy as _copy
+        import json as _json
+
+        with self.lock:
+            for i, pod in enumerate(pods):
+                obj = _copy.deepcopy(pod) if copy_objs else pod
+                key = (f"{obj['metadata'].get('namespace', 'default')}/"
+                       f"{obj['metadata']['name']}")
+                self.pods[key] = obj
+                self.resource_version += 1
+                ev = {"rv": self.resource_version, "type": "ADDED",
+                      "object": obj}
+                if lines is not None:
+                    ev["line"] = lines[i]
+                else:
+                    obj.setdefault("metadata", {})["resourceVersion"] = str(
+                        self.resource_version)
+                    ev["line"] = _json.dumps(
+                        {"type": "ADDED", "object": obj}) + "\n"
+                self.pod_events.append(ev)
+            self.cond.notify_all()
+
+    def bulk_remove_pods(self, keys: List[str]):
+        """Delete many pods under one lock ('<ns>/<name>' keys)."""
+        with self.lock:
+            import json as _json
+
+            for key in keys:
+                pod = self.pods.pop(key, None)
+                if pod is not None:
+                    self.resource_version += 1
+                    self.pod_events.append(
+                        {"rv": self.resource_version, "type": "DELETED",
+                         "object": pod,
+                         "line": _json.dumps({"type": "DELETED",
+                                              "object": pod}) + "\n"})
+            self.cond.notify_all()
+
+    def remove_node(self, name: str):
+        with self.lock:
+            node = self.nodes.pop(name, None)
+            if node is not None:
+                self._emit_node("DELETED", node)
 
     def add_pod(self, pod: Dict):
         with self.lock:
@@ -101,26 +158,39 @@ class FakeKubeState:
 
 
 def _event_stream(state: FakeKubeState, log: str, start_rv: int):
-    """JSONL watch stream over an event log; wakes on state.cond
-    (no poll-sleep on the reaction path), honors drop_watch_after."""
+    """JSONL watch stream over an event log; wakes on state.cond (no
+    poll-sleep on the reaction path), scans the log positionally
+    (O(new events), not O(log)), serializes outside the lock, and
+    yields one CHUNK per batch so a 10k-pod burst is a handful of
+    writes, not 10k. Honors drop_watch_after fault injection."""
     import json as _json
 
     sent = 0
     last_rv = start_rv
+    pos = 0
     while True:
-        batch = []
         with state.lock:
-            for e in getattr(state, log):
-                if e["rv"] > last_rv:
-                    batch.append(e)
+            log_list = getattr(state, log)
+            if pos > len(log_list):
+                pos = 0  # log was reset by load()
+            batch = [e for e in log_list[pos:] if e["rv"] > last_rv]
+            pos = len(log_list)
             if not batch:
                 state.cond.wait(timeout=0.5)
-        for e in batch:
-            if state.drop_watch_after >= 0 and sent >= state.drop_watch_after:
+                continue
+        if state.drop_watch_after >= 0:
+            room = state.drop_watch_after - sent
+            if room <= 0:
                 return  # injected disconnect
-            last_rv = e["rv"]
-            sent += 1
-            yield _json.dumps({"type": e["type"], "object": e["object"]}) + "\n"
+            batch = batch[: room]
+        last_rv = batch[-1]["rv"]
+        sent += len(batch)
+        yield "".join(
+            e.get("line") or
+            _json.dumps({"type": e["type"], "object": e["object"]}) + "\n"
+            for e in batch)
+        if state.drop_watch_after >= 0 and sent >= state.drop_watch_after:
+            return
 
 
 def build_app(state: FakeKubeState) -> FastAPI:

@@ -87,8 +87,9 @@ def test_watch_node_churn_remaps_in_place(server):
     cluster.loop_logic()
     first_state = cluster._watch_state
     first_cache = first_state["cache"]
-    # add a node out of band: the watch cache must survive
-    server.state.nodes["k8s-agentpool1-12345678-9"] = make_node_json("agentpool1", 9)
+    # add a node (watch-visible): the POD cache must survive un-primed
+    server.state.add_node(make_node_json("agentpool1", 9))
+    time.sleep(0.3)  # let the node watch deliver
     res = cluster.loop_logic()
     assert cluster._watch_state is first_state  # NOT re-primed
     assert cluster._watch_state["cache"] is first_cache
@@ -98,7 +99,8 @@ def test_watch_node_churn_remaps_in_place(server):
     # remove a node that has bound pods: indices of later nodes shift
     # and the orphaned bindings must clear, exactly as a relist would
     victim = sorted(server.state.nodes)[0]
-    del server.state.nodes[victim]
+    server.state.remove_node(victim)
+    time.sleep(0.3)  # let the node watch deliver
     res2 = cluster.loop_logic()
     assert cluster._watch_state is first_state
     snap = res2.snapshot
@@ -236,3 +238,29 @@ def test_watch_mode_native_codec_pump(server):
     import numpy as np
 
     assert int((snap.pod_pdb >= 0).sum()) == 5  # the guarded late pods
+
+
+def test_node_watch_caches_and_tracks(server):
+    """World-1 watch mode keeps the node set under a watch too: the
+    wrapped node list is reused while nothing changes (no per-cycle
+    relist/re-wrap) and watch events update it within a cycle."""
+    pods, nodes = burst_trace(5, n_nodes_per_pool=(3,), seed=4)
+    server.state.load(pods, nodes)
+    cluster, _ = _cluster(server, no_maintenance=True, no_scale=True)
+    cluster.loop_logic()
+    nw = cluster._node_watch
+    assert nw is not None and not nw["broken"]
+    wrapped1 = nw["wrapped"]
+    assert len(wrapped1) == 3
+    cluster.loop_logic()
+    assert cluster._node_watch["wrapped"] is wrapped1  # reused, no rebuild
+
+    server.state.add_node(make_node_json("agentpool1", 7))
+    time.sleep(0.3)
+    res = cluster.loop_logic()
+    assert res.nodes_total == 4
+    assert cluster._node_watch["wrapped"] is not wrapped1
+
+    server.state.remove_node("k8s-agentpool1-12345678-7")
+    time.sleep(0.3)
+    assert cluster.loop_logic().nodes_total == 3
