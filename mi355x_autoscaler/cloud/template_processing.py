@@ -70,9 +70,10 @@ def set_pool_counts(
 
     for pool, size in new_sizes.items():
         _set(f"{pool}Count", size)
-        offset = current_sizes.get(pool, 0)
-        if size > offset:
-            _set(f"{pool}Offset", offset)
+        # ARM deploys agent indices [offset, count): offset = current
+        # so only NEW VMs deploy; min() makes an equal-size resubmit a
+        # true no-op (idempotent retry)
+        _set(f"{pool}Offset", min(current_sizes.get(pool, 0), size))
     return out
 
 
