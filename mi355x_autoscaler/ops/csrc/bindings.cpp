@@ -106,6 +106,19 @@ void fused_decide_launch(
     torch::Tensor seg_pool, torch::Tensor run_scratch,
     torch::Tensor blob, torch::Tensor pod_pool, int64_t hash_slots,
     int64_t seg_max, int64_t want_pod_pool);
+#include "decide_plan.h"
+std::shared_ptr<DecidePlan> create_decide_plan(
+    torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
+    torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
+    torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
+    torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
+    torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
+    torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
+    torch::Tensor, torch::Tensor, torch::Tensor, int64_t, int64_t, int64_t);
+void run_decide_plan(const std::shared_ptr<DecidePlan>&);
+pybind11::tuple finish_decision(torch::Tensor blob_host, int64_t K,
+                                int64_t seg_max, int64_t over_provision,
+                                torch::Tensor max_new, torch::Tensor pool_cur);
 void snapshot_scatter(torch::Tensor rows, torch::Tensor u_req,
                       torch::Tensor u_flags, torch::Tensor u_sel_hash,
                       torch::Tensor u_sel_pairs, torch::Tensor u_sel_count,
@@ -171,4 +184,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "enqueue the whole scale-up pipeline in one call");
   m.def("fused_tail_launch", &fused_tail_launch,
         "enqueue the post-segment-build pipeline (sharded merged path)");
+  m.def("finish_decision", &finish_decision,
+        "host tail: pinned blob -> packed decision numbers (one C pass)");
+  pybind11::class_<DecidePlan, std::shared_ptr<DecidePlan>>(m, "DecidePlan");
+  m.def("create_decide_plan", &create_decide_plan,
+        "bind the fused-decide tensor set once");
+  m.def("run_decide_plan", &run_decide_plan,
+        "re-enqueue the fused decide with zero argument conversion");
 }
