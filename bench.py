@@ -281,13 +281,17 @@ def main():
             return engine.maintain(snap)
     elif resident and use_gpu and world == 1:
         # config #5: the snapshot stays resident in HBM; each step is a
-        # full re-decision (segment build + fit + pack) over it.
+        # full re-decision over it via the fused single-call pipeline
+        # (the same path the resident watch loop takes — all kernels
+        # enqueued back-to-back from C++, no per-op dispatch gaps).
         ds = inner.packer.pack(snap, inner.device)
         torch.cuda.synchronize()
 
         def decide_once():
-            segs = inner.build_segments(ds)
-            return inner.decide_device(ds, segs, snap.pool_names)
+            d = inner._decide_fused(ds, snap.pool_names, stable_pods=True)
+            if d is None:  # hash overflow: not expected in this config
+                raise RuntimeError("fused decide overflowed in burst1m")
+            return d
     elif resident and use_gpu:
         # config #5 sharded: each rank's 1M-pod shard stays resident in
         # its HBM; a step is local segment build + fixed-size row

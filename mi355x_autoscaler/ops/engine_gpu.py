@@ -439,6 +439,7 @@ class GpuEngine(SimulationEngine):
         self._side_stream = torch.cuda.Stream(self.device) if torch.cuda.is_available() else None
         self._ev_fork = None
         self._ev_join = None
+        self._pp_guard: Dict[bool, object] = {}  # flip -> segsum-done event
         self._side_done: Optional[torch.Tensor] = None  # cuda Event when pending
         d = self.device
         # persistent hash workspace; the mutable i32 parts live in ONE
@@ -655,6 +656,13 @@ class GpuEngine(SimulationEngine):
         if stable_pods and want_pod_pool:
             self._pp_flip = not getattr(self, "_pp_flip", False)
             pod_pool = bufs["pod_pool_b"] if self._pp_flip else bufs["pod_pool"]
+            # A-B-A fence: before the pipeline rewrites THIS pod_pool
+            # buffer, the side-stream segsum that read it two decides
+            # ago must be done (stream-level wait, no host sync; only
+            # bites when >2 unsynced decides stack up back-to-back)
+            guard = self._pp_guard.get(bool(self._pp_flip))
+            if guard is not None:
+                torch.cuda.current_stream().wait_event(guard)
         else:
             pod_pool = bufs["pod_pool"]
 
@@ -716,6 +724,13 @@ class GpuEngine(SimulationEngine):
                                                out=dbuf)
                     self._ev_join.record(self._side_stream)
                     self._side_done = self._ev_join
+                    if stable_pods:
+                        # per-buffer read-done marker for the A-B-A fence
+                        fl = bool(self._pp_flip)
+                        gev = self._pp_guard.get(fl)
+                        if gev is None:
+                            self._pp_guard[fl] = gev = torch.cuda.Event()
+                        gev.record(self._side_stream)
                 self._last_demand.record_stream(self._side_stream)
             else:
                 self._last_demand = segsum(ds.pod_req, pod_pool, K, ops)

@@ -465,6 +465,99 @@ __device__ inline void dev_binpack(
     return;
   }
 
+  // Large-N fast path (round 2, the 1M-config N=4000 regime): wave-
+  // owned node chunks with the free table REGISTER-resident across
+  // the whole segment walk and ONE block barrier per segment (the
+  // round-1 chunked loop paid 2 barriers per chunk per segment —
+  // ~208 barriers at N=4000/G=26 — plus a free reload per chunk).
+  // Wave w owns chunks {w, w+NW, ...}; only the per-chunk capacity
+  // totals cross waves (parity-double-buffered LDS row), so the FFD
+  // scan-threshold math is bit-identical to the serial node order.
+  const int nchunks = (N + WAVE - 1) / WAVE;
+  if constexpr (BLOCK >= 1024) {
+   if (N > BLOCK && nchunks <= WAVE) {
+    constexpr int NW_ = BLOCK / WAVE;
+    constexpr int CPW = (WAVE + NW_ - 1) / NW_;  // chunks per wave
+    const int lane = tid & (WAVE - 1);
+    const int wv = tid / WAVE;
+    __shared__ int64_t s_ctot[2][WAVE];
+    // my chunk ids: wv, wv+NW_, ... ; node j = chunk*WAVE + lane
+    int32_t fr2[CPW][NRES];
+    int cid[CPW];
+#pragma unroll
+    for (int q = 0; q < CPW; ++q) {
+      cid[q] = wv + q * NW_;
+      const int j = cid[q] * WAVE + lane;
+      if (cid[q] < nchunks && j < N) {
+#pragma unroll
+        for (int r = 0; r < NRES; ++r)
+          fr2[q][r] = free_[(int64_t)j * NRES + r];
+      }
+    }
+    for (int g = 0; g < G; ++g) {
+      const int64_t need = g < BP_PRELOAD ? s_cnt[g] : seg_count[g];
+      const int par = g & 1;
+      if (need <= 0) {
+        if (tid == 0) placed_out[g] = 0;
+        // keep the barrier structure uniform across the block
+        __syncthreads();
+        continue;
+      }
+      int32_t rq[NRES];
+#pragma unroll
+      for (int r = 0; r < NRES; ++r)
+        rq[r] = g < BP_PRELOAD ? s_req[g * NRES + r] : seg_req[g * NRES + r];
+      const uint64_t* erow = elig + (int64_t)g * W;
+      int64_t m_q[CPW], inc_q[CPW];
+#pragma unroll
+      for (int q = 0; q < CPW; ++q) {
+        const int j = cid[q] * WAVE + lane;
+        int64_t m = 0;
+        if (cid[q] < nchunks && j < N &&
+            ((erow[cid[q]] >> lane) & 1ull)) {
+          m = per_node_capacity(fr2[q], rq);
+        }
+        m_q[q] = m;
+        inc_q[q] = wave_inclusive_scan_i64(m);
+        if (cid[q] < nchunks && lane == WAVE - 1)
+          s_ctot[par][cid[q]] = inc_q[q];
+      }
+      __syncthreads();  // publish chunk totals
+      // every wave redundantly scans the totals row (<= 64 entries)
+      const int64_t v = lane < nchunks ? s_ctot[par][lane] : 0;
+      const int64_t cinc = wave_inclusive_scan_i64(v);
+      const int64_t total_all = __shfl(cinc, nchunks - 1, WAVE);
+#pragma unroll
+      for (int q = 0; q < CPW; ++q) {
+        if (cid[q] >= nchunks) continue;
+        const int64_t cpref = __shfl(cinc - v, cid[q], WAVE);
+        const int64_t S = cpref + inc_q[q] - m_q[q];
+        int64_t take = need - S;
+        take = take < 0 ? 0 : (take > m_q[q] ? m_q[q] : take);
+        if (take > 0) {
+#pragma unroll
+          for (int r = 0; r < NRES; ++r)
+            fr2[q][r] -= (int32_t)(take * rq[r]);
+        }
+      }
+      if (tid == 0) placed_out[g] = total_all < need ? total_all : need;
+      // no second barrier: the next segment's totals go to the other
+      // parity row, and free is only touched by its owner lane
+    }
+#pragma unroll
+    for (int q = 0; q < CPW; ++q) {  // keep the mutated-free contract
+      const int j = cid[q] * WAVE + lane;
+      if (cid[q] < nchunks && j < N) {
+#pragma unroll
+        for (int r = 0; r < NRES; ++r)
+          free_[(int64_t)j * NRES + r] = fr2[q][r];
+      }
+    }
+    __syncthreads();
+    return;
+   }
+  }
+
   for (int g = 0; g < G; ++g) {
     const int64_t need = g < BP_PRELOAD ? s_cnt[g] : seg_count[g];
     if (need <= 0) {

@@ -30,7 +30,9 @@
 
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
-extern "C" __global__ __launch_bounds__(WAVE) void segsum_mfma_kernel(
+#define SS_WAVES 4  // waves per block, each owning its own chunk + LDS
+
+extern "C" __global__ __launch_bounds__(WAVE * SS_WAVES) void segsum_mfma_kernel(
     const int32_t* __restrict__ values,   // [M, NRES] sorted by seg
     const int32_t* __restrict__ seg_of,   // [M] ascending
     const int32_t* __restrict__ wi_win,   // [T] window per work item
@@ -45,10 +47,19 @@ extern "C" __global__ __launch_bounds__(WAVE) void segsum_mfma_kernel(
   // together), then the MFMA loop feeds from LDS broadcasts — the
   // fragment-at-a-time global loads of v1 serialized on L2 latency
   // (220 us at 1M pods).
-  __shared__ int32_t l_vals[256 * NRES];
-  __shared__ int32_t l_segs[256];
+  //
+  // Round 2: 4 waves per block (each with a private LDS region and its
+  // own grid-strided work-item walk — no cross-wave barriers) and FOUR
+  // independent MFMA accumulator chains per chunk. The v2 single-wave
+  // blocks left 3/4 of the SIMDs idle and the lone dependent MFMA
+  // chain exposed the full 16x16x4 latency every step (74.9 us at 1M);
+  // exactness is unchanged — each chain sums <= 64 limb values < 2^22,
+  // recombined through the same int64 window partials.
+  __shared__ int32_t l_vals[SS_WAVES][256 * NRES];
+  __shared__ int32_t l_segs[SS_WAVES][256];
 
   const int lane = threadIdx.x & (WAVE - 1);
+  const int wv = threadIdx.x / WAVE;
   const int col = lane & 15;      // A row index i / B,D column j
   const int kidx = lane >> 4;     // K slot 0..3 (one pod each)
 
@@ -83,7 +94,10 @@ extern "C" __global__ __launch_bounds__(WAVE) void segsum_mfma_kernel(
     for (int reg = 0; reg < 4; ++reg) p64[reg] = 0;
   };
 
-  for (int64_t t = blockIdx.x; t < T; t += gridDim.x) {
+  int32_t* lv = l_vals[wv];
+  int32_t* ls = l_segs[wv];
+  const int64_t wstride = (int64_t)gridDim.x * SS_WAVES;
+  for (int64_t t = (int64_t)blockIdx.x * SS_WAVES + wv; t < T; t += wstride) {
     const int win = wi_win[t];
     if (win != cur_win) {
       if (cur_win >= 0) flush(cur_win);
@@ -100,30 +114,51 @@ extern "C" __global__ __launch_bounds__(WAVE) void segsum_mfma_kernel(
       const int64_t p = p_begin + pl;
       if (p < p_end) {
         const int4 v = reinterpret_cast<const int4*>(values)[p];
-        reinterpret_cast<int4*>(l_vals)[pl] = v;
-        l_segs[pl] = seg_of[p] - seg_base;
+        reinterpret_cast<int4*>(lv)[pl] = v;
+        ls[pl] = seg_of[p] - seg_base;
       } else {
-        reinterpret_cast<int4*>(l_vals)[pl] = int4{0, 0, 0, 0};
-        l_segs[pl] = -1;  // no column matches: contributes zero
+        reinterpret_cast<int4*>(lv)[pl] = int4{0, 0, 0, 0};
+        ls[pl] = -1;  // no column matches: contributes zero
       }
     }
-    __syncthreads();
+    // private LDS region: only this wave's in-order LDS ops touch it
+    __builtin_amdgcn_wave_barrier();
 
     const int steps = (int)((p_end - p_begin + 3) >> 2);
-    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    for (int sstep = 0; sstep < steps; ++sstep) {
+    f32x4 acc[4] = {{0.f, 0.f, 0.f, 0.f},
+                    {0.f, 0.f, 0.f, 0.f},
+                    {0.f, 0.f, 0.f, 0.f},
+                    {0.f, 0.f, 0.f, 0.f}};
+    int sstep = 0;
+    for (; sstep + 4 <= steps; sstep += 4) {
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        const int pl = (sstep + u) * 4 + kidx;
+        const float b = (ls[pl] == col) ? 1.f : 0.f;
+        float a = 0.f;
+        if (live_row) {
+          const uint32_t v = (uint32_t)lv[pl * NRES + res];
+          a = (float)(limb ? (v >> 16) : (v & 0xffffu));
+        }
+        acc[u] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[u], 0, 0, 0);
+      }
+    }
+    for (; sstep < steps; ++sstep) {
       const int pl = sstep * 4 + kidx;
-      const float b = (l_segs[pl] == col) ? 1.f : 0.f;
+      const float b = (ls[pl] == col) ? 1.f : 0.f;
       float a = 0.f;
       if (live_row) {
-        const uint32_t v = (uint32_t)l_vals[pl * NRES + res];
+        const uint32_t v = (uint32_t)lv[pl * NRES + res];
         a = (float)(limb ? (v >> 16) : (v & 0xffffu));
       }
-      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+      acc[0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[0], 0, 0, 0);
     }
 #pragma unroll
-    for (int reg = 0; reg < 4; ++reg) p64[reg] += (long long)acc[reg];
-    __syncthreads();  // LDS reused by the next chunk
+    for (int u = 0; u < 4; ++u) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) p64[reg] += (long long)acc[u][reg];
+    }
+    __builtin_amdgcn_wave_barrier();  // LDS region reused next chunk
   }
   if (cur_win >= 0) flush(cur_win);
 }
@@ -141,10 +176,12 @@ void segsum_mfma(torch::Tensor values, torch::Tensor seg_of,
   const int64_t T = wi_win.size(0);
   if (T == 0) return;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  // enough single-wave blocks to fill the SIMD slots; each walks
-  // work items grid-strided with register-accumulated partials
-  const uint32_t grid = (uint32_t)std::min<int64_t>(T, 1024);
-  hipLaunchKernelGGL(segsum_mfma_kernel, dim3(grid), dim3(WAVE), 0,
+  // SS_WAVES waves per block, each walking work items grid-strided
+  // with register-accumulated partials (fills 4x the SIMD slots of
+  // the round-1 single-wave blocks)
+  const uint32_t grid =
+      (uint32_t)std::min<int64_t>((T + SS_WAVES - 1) / SS_WAVES, 1024);
+  hipLaunchKernelGGL(segsum_mfma_kernel, dim3(grid), dim3(WAVE * SS_WAVES), 0,
                      stream.stream(), values.data_ptr<int32_t>(),
                      seg_of.data_ptr<int32_t>(), wi_win.data_ptr<int32_t>(),
                      wi_start.data_ptr<int64_t>(), wi_end.data_ptr<int64_t>(),
