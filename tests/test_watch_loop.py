@@ -264,3 +264,36 @@ def test_node_watch_caches_and_tracks(server):
     server.state.remove_node("k8s-agentpool1-12345678-7")
     time.sleep(0.3)
     assert cluster.loop_logic().nodes_total == 3
+
+
+def test_bulk_injection_reaches_watchers(server):
+    """bulk_add_pods (pre-serialized POST bodies) + the batched watch
+    stream + bulk_remove_pods — the e2e reaction bench's path — must
+    deliver every event to a watch-mode loop."""
+    import json as _json
+
+    pods, nodes = burst_trace(3, n_nodes_per_pool=(3,), seed=8)
+    server.state.load(pods, nodes)
+    cluster, _ = _cluster(server, no_maintenance=True, no_scale=True)
+    cluster.loop_logic()  # prime
+
+    wave = [make_pod_json(f"bulk{i}", cpu_milli=50, pending=True)
+            for i in range(40)]
+    lines = [_json.dumps({"type": "ADDED", "object": p}) + "\n" for p in wave]
+    server.state.bulk_add_pods(wave, copy_objs=False, lines=lines)
+    deadline = time.time() + 10
+    while time.time() < deadline:
+        res = cluster.loop_logic()
+        if res.pods_pending >= 43:
+            break
+        time.sleep(0.05)
+    assert res.pods_pending == 43
+
+    server.state.bulk_remove_pods([f"default/bulk{i}" for i in range(40)])
+    deadline = time.time() + 10
+    while time.time() < deadline:
+        res = cluster.loop_logic()
+        if res.pods_pending <= 3:
+            break
+        time.sleep(0.05)
+    assert res.pods_pending == 3
