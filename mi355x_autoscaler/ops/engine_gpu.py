@@ -1340,10 +1340,12 @@ class GpuEngine(SimulationEngine):
         ops = self.ops
         device = self.device
         N = ds.N
-        blocked = torch.zeros(N, dtype=torch.int32, device=device)
-        pdb_bad = torch.zeros(N, dtype=torch.int32, device=device)
-        nontrivial = torch.zeros(N, dtype=torch.int32, device=device)
-        drainable = torch.zeros(N, dtype=torch.int32, device=device)
+        # one stacked [5, N] i32 buffer so the report needs ONE D2H
+        # sync instead of five .cpu() round trips (row 4 is the
+        # finalize's util_permille output); rows are contiguous views,
+        # so the sharded all-reduce path is unchanged
+        mt = torch.zeros((5, max(N, 1)), dtype=torch.int32, device=device)
+        blocked, pdb_bad, nontrivial, drainable = mt[0], mt[1], mt[2], mt[3]
         pdb_allowed = ds.pdb_allowed if ds.pdb_allowed.numel() else torch.zeros(
             1, dtype=torch.int32, device=device
         )
@@ -1358,7 +1360,7 @@ class GpuEngine(SimulationEngine):
             ops.node_usage_sum(ds.pod_node, ds.pod_req, util)
 
         tensors = dict(util=util, blocked=blocked, pdb_bad=pdb_bad,
-                       nontrivial=nontrivial, drainable=drainable)
+                       nontrivial=nontrivial, drainable=drainable, _mt=mt)
         rep = self.finalize_maintain(ds, tensors, util_threshold_permille)
         return rep, tensors
 
@@ -1366,20 +1368,29 @@ class GpuEngine(SimulationEngine):
         ops = self.ops
         N = ds.N
         device = self.device
-        util_permille = torch.zeros(N, dtype=torch.int32, device=device)
+        mt = tensors.get("_mt")
+        if mt is None or mt.shape[1] < max(N, 1):  # sharded caller built its own dict
+            mt = torch.zeros((5, max(N, 1)), dtype=torch.int32, device=device)
+            if N > 0:
+                mt[0].copy_(tensors["blocked"])
+                mt[1].copy_(tensors["pdb_bad"])
+                mt[2].copy_(tensors["nontrivial"])
+                mt[3].copy_(tensors["drainable"])
+        util_permille = mt[4]
         candidate = torch.zeros(N, dtype=torch.int32, device=device)
         if N > 0:
             ops.drain_finalize(
                 tensors["util"], ds.node_alloc, ds.node_pool, tensors["blocked"],
                 tensors["pdb_bad"], tensors["nontrivial"],
-                util_threshold_permille, util_permille, candidate,
+                util_threshold_permille, util_permille[:N], candidate,
             )
+        mt_h = mt[:, :N].cpu().numpy() if N > 0 else np.zeros((5, 0), np.int32)
         return MaintainReport(
-            node_util_permille=util_permille.cpu().numpy(),
-            node_blocked=(tensors["blocked"].cpu().numpy() != 0).astype(np.uint8),
-            node_pdb_ok=(tensors["pdb_bad"].cpu().numpy() == 0).astype(np.uint8),
-            node_idle=(tensors["nontrivial"].cpu().numpy() == 0).astype(np.uint8),
-            node_drainable_pods=tensors["drainable"].cpu().numpy().astype(np.int32),
+            node_util_permille=mt_h[4].copy(),
+            node_blocked=(mt_h[0] != 0).astype(np.uint8),
+            node_pdb_ok=(mt_h[1] == 0).astype(np.uint8),
+            node_idle=(mt_h[2] == 0).astype(np.uint8),
+            node_drainable_pods=mt_h[3].copy(),
         )
 
     @staticmethod
