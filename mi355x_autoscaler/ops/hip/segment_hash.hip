@@ -64,12 +64,24 @@ extern "C" __global__ __launch_bounds__(256) void segment_hash_build(
   // pass 1: aggregate this block's pods into the LDS table; stash the
   // LDS slot id (or direct-global fallback) in slot_of_pod for pass 2.
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  // per-lane 1-entry memo: bursts are runs of identical (selector,
-  // request) keys, so most pods resolve with two register compares —
-  // no LDS probe at all (round-2: the probe chain + per-lane LDS
-  // atomics were 74.7M VALU / 218k bank conflicts at 1M pods)
+  // per-lane 1-entry memo WITH local run counting: bursts are runs of
+  // identical (selector, request) keys, so the steady state per pod is
+  // two register compares + a register increment — no LDS probe and NO
+  // LDS atomic at all; counts/min-ids flush once per key RUN (memo
+  // switch or loop end). Round-1 profile: the probe chain + per-pod
+  // LDS atomics were 74.7M VALU / 218k bank conflicts at 1M pods.
   int64_t memo_ks = 0, memo_kr = 0;
   int memo_enc = INT32_MIN;
+  int32_t memo_cnt = 0;
+  int32_t memo_min_i0 = 0;
+  auto memo_flush = [&]() {
+    if (memo_enc >= 0 && memo_cnt > 0) {
+      __hip_atomic_fetch_add(&l_count[memo_enc], memo_cnt, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_WORKGROUP);
+      atomicMax(&l_first[memo_enc], INT32_MAX - memo_min_i0);
+    }
+    memo_cnt = 0;
+  };
   for (int64_t i0 = (int64_t)blockIdx.x * blockDim.x + tid; ; i0 += stride) {
     if (__all(i0 >= P)) break;
     const int pf = i0 < P ? pod_flags[i0] : 0;
@@ -85,15 +97,18 @@ extern "C" __global__ __launch_bounds__(256) void segment_hash_build(
       slot_of_pod[i0] = -1;
     }
 
-    // fast path: memo hit, else per-lane DIRECT probe of already-
-    // claimed LDS slots — reads are deadlock-free (only claims need
-    // the leader-at-a-time discipline); after the first occurrence of
-    // each key per block every pod hits here, skipping the serialized
-    // ballot loop
+    // fast path: memo hit (count locally), else per-lane DIRECT probe
+    // of already-claimed LDS slots — reads are deadlock-free (only
+    // claims need the leader-at-a-time discipline); after the first
+    // occurrence of each key per block every pod hits here
     int my_enc = -1;
     if (live) {
       if (memo_enc != INT32_MIN && ks == memo_ks && kr == memo_kr) {
         my_enc = memo_enc;
+        if (memo_enc >= 0) {
+          if (memo_cnt == 0) memo_min_i0 = (int32_t)i0;
+          ++memo_cnt;
+        }
       } else {
         uint32_t h = mix_hash(ks, kr) & (LDS_SLOTS - 1);
         for (int probe = 0; probe < 8; ++probe) {
@@ -112,28 +127,14 @@ extern "C" __global__ __launch_bounds__(256) void segment_hash_build(
           h = (h + 1u) & (LDS_SLOTS - 1);
         }
         if (my_enc >= 0) {
+          memo_flush();
           memo_ks = ks;
           memo_kr = kr;
           memo_enc = my_enc;
+          memo_cnt = 1;
+          memo_min_i0 = (int32_t)i0;
         }
       }
-    }
-    // wave-aggregated accumulation: ONE l_count/l_first atomic pair
-    // per distinct key per wave-iteration instead of per pod (the
-    // leader is the lowest matched lane, so its i0 IS the wave min)
-    uint64_t agg = __ballot(live && my_enc >= 0);
-    while (agg) {
-      const int leader = __ffsll((unsigned long long)agg) - 1;
-      const int e0 = __shfl(my_enc, leader, WAVE);
-      const uint64_t mm = __ballot(live && my_enc == e0) & agg;
-      if (lane == leader) {
-        __hip_atomic_fetch_add(&l_count[e0],
-                               __popcll((unsigned long long)mm),
-                               __ATOMIC_RELAXED,
-                               __HIP_MEMORY_SCOPE_WORKGROUP);
-        atomicMax(&l_first[e0], INT32_MAX - (int32_t)i0);
-      }
-      agg &= ~mm;
     }
 
     uint64_t pending_lanes = __ballot(live && my_enc < 0);
@@ -197,11 +198,22 @@ extern "C" __global__ __launch_bounds__(256) void segment_hash_build(
         }
       }
       enc = __shfl(enc, leader, WAVE);
-      if (match) my_enc = enc;
+      if (match) {
+        my_enc = enc;
+        if (enc >= 0) {
+          // already counted by the leader; seed the memo so the run
+          // continues on the fast path (cnt starts at 0)
+          memo_flush();
+          memo_ks = ks;
+          memo_kr = kr;
+          memo_enc = enc;
+        }
+      }
       pending_lanes &= ~match_mask;
     }
     if (i0 < P && live) slot_of_pod[i0] = my_enc;
   }
+  memo_flush();
   __syncthreads();
 
   // flush: wave 0 resolves each occupied LDS slot to a global slot,

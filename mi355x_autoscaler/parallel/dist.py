@@ -57,7 +57,10 @@ def allgather_rows(rows: np.ndarray, device=None) -> np.ndarray:
         return rows
     assert rows.ndim == 2
     C = rows.shape[1]
-    use_cuda = device is not None and str(device).startswith("cuda")
+    # device-side gather only when the backend actually supports CUDA
+    # tensors (nccl/RCCL); gloo with GPU engines gathers on the host
+    use_cuda = (device is not None and str(device).startswith("cuda")
+                and backend() == "nccl")
     dev = device if use_cuda else torch.device("cpu")
     t = torch.from_numpy(np.ascontiguousarray(rows)).to(dev)
 
@@ -115,9 +118,16 @@ def allgather_fixed_device(t: torch.Tensor) -> torch.Tensor:
 
 
 def allreduce_tensor(t: torch.Tensor, op: str = "sum") -> torch.Tensor:
-    """In-place all-reduce; returns t. ops: sum | max | min."""
+    """In-place all-reduce; returns t. ops: sum | max | min. CUDA
+    tensors bounce through the host when the backend can't take them
+    (gloo test rigs driving real GPU engines)."""
     if world_size() == 1:
         return t
     red = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX, "min": dist.ReduceOp.MIN}[op]
+    if t.is_cuda and backend() != "nccl":
+        h = t.cpu()
+        dist.all_reduce(h, op=red)
+        t.copy_(h)
+        return t
     dist.all_reduce(t, op=red)
     return t

@@ -1,0 +1,119 @@
+"""Sharded GPU engine on HARDWARE (world 2, gloo, both ranks on
+cuda:0): the ds-packed and resident sharded decide paths run real HIP
+kernels per rank with cross-rank row merge + agreement — the closest
+1-GPU-box approximation of the driver's 8-GPU run (the nccl device
+merge swaps in by backend switch only; its row protocol is covered by
+test_decide_rows_device_* in test_gpu_engine.py)."""
+
+import multiprocessing as mp
+import os
+import pickle
+
+import pytest
+
+from mi355x_autoscaler.models.agent_pool import group_nodes_into_pools
+from mi355x_autoscaler.sim.engine import CpuEngine
+from mi355x_autoscaler.sim.synth import burst_trace, make_pod_json, wrap
+from mi355x_autoscaler.snapshot.encode import encode_snapshot
+
+pytestmark = pytest.mark.gpu
+
+WORLD = 2
+
+
+def _trace():
+    return burst_trace(
+        4000, n_nodes_per_pool=(20, 10),
+        pools=(("agentpool1", "Standard_D2_v2"),
+               ("gpupool", "Standard_ND96_MI355X_v1")),
+        gpu_fraction=0.4, seed=77, bound_pods_per_node=3)
+
+
+def _full_snap():
+    pods_json, nodes_json = _trace()
+    pods, nodes = wrap(pods_json, nodes_json)
+    return encode_snapshot(pods, nodes, group_nodes_into_pools(nodes))
+
+
+def _rank(rank, q, mode):
+    os.environ.update(WORLD_SIZE=str(WORLD), RANK=str(rank),
+                      LOCAL_RANK="0",  # both ranks share the one GPU
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT="29715")
+    import torch.distributed as dist
+
+    from mi355x_autoscaler.ops.engine_gpu import GpuEngine
+    from mi355x_autoscaler.parallel.elastic import shard_snapshot
+    from mi355x_autoscaler.parallel.engine_sharded import ShardedEngine
+    from mi355x_autoscaler.snapshot.delta import DeltaSnapshotCache
+
+    try:
+        snap = _full_snap()
+        shard = shard_snapshot(snap, rank, WORLD)
+        engine = ShardedEngine(GpuEngine("cuda:0"), check_agreement=True)
+        if mode == "decide":
+            d = engine.decide(shard)
+        else:  # resident: per-rank shard cache + churn
+            pods_json, nodes_json = _trace()
+            pods, nodes = wrap(pods_json, nodes_json)
+            nodes_idx = {n.name: i for i, n in enumerate(nodes)}
+            pools = group_nodes_into_pools(nodes)
+            base = encode_snapshot([], nodes, pools)
+            cache = DeltaSnapshotCache()
+            cache.set_shard(rank, WORLD)
+            cache.prime(shard, nodes_idx)
+            for i in range(60):  # same stream on every rank; self-filtered
+                cache.apply_event({"type": "ADDED",
+                                   "object": make_pod_json(
+                                       f"late{i}", cpu_milli=300,
+                                       pending=True)})
+            d = engine.decide_resident(cache, base)
+        rep = engine.maintain(shard)
+        q.put((rank, pickle.dumps(dict(
+            units=d.units_needed, pe=d.pods_placed_existing,
+            pn=d.pods_placed_new, un=d.pods_unfulfillable,
+            cands=[c.node_name for c in rep.candidates]))))
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def _run(mode):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_rank, args=(r, q, mode)) for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, blob = q.get(timeout=240)
+        results[rank] = pickle.loads(blob)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    return results
+
+
+def test_sharded_gpu_decide_matches_single_process():
+    results = _run("decide")
+    assert results[0] == results[1]
+    d = CpuEngine().decide(_full_snap())
+    assert results[0]["units"] == d.units_needed
+    assert results[0]["pe"] == d.pods_placed_existing
+    assert results[0]["un"] == d.pods_unfulfillable
+    rep = CpuEngine().maintain(_full_snap())
+    assert results[0]["cands"] == [c.node_name for c in rep.candidates]
+
+
+def test_sharded_gpu_resident_decide_matches_full():
+    results = _run("resident")
+    assert results[0] == results[1]
+    # reference: full churned pod set, single process
+    pods_json, nodes_json = _trace()
+    extra = [make_pod_json(f"late{i}", cpu_milli=300, pending=True)
+             for i in range(60)]
+    pods, nodes = wrap(pods_json + extra, nodes_json)
+    ref = CpuEngine().decide(
+        encode_snapshot(pods, nodes, group_nodes_into_pools(nodes)))
+    assert results[0]["units"] == ref.units_needed
+    assert results[0]["pe"] == ref.pods_placed_existing
+    assert results[0]["un"] == ref.pods_unfulfillable
