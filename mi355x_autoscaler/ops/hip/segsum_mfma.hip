@@ -57,6 +57,18 @@ extern "C" __global__ __launch_bounds__(WAVE * SS_WAVES) void segsum_mfma_kernel
   // recombined through the same int64 window partials.
   __shared__ int32_t l_vals[SS_WAVES][256 * NRES];
   __shared__ int32_t l_segs[SS_WAVES][256];
+  // single-window case (S <= 16, the per-pool demand sum): waves
+  // accumulate into BLOCK LDS first and one wave flushes per block —
+  // direct per-wave global atomics serialized ~8k adds onto S*NRES
+  // hot words at 1M pods (the whole kernel was flush-bound)
+  __shared__ unsigned long long s_acc[16 * NRES];
+
+  const bool single_win = S <= 16;
+  if (single_win) {
+    for (int idx = threadIdx.x; idx < 16 * NRES; idx += blockDim.x)
+      s_acc[idx] = 0ull;
+    __syncthreads();
+  }
 
   const int lane = threadIdx.x & (WAVE - 1);
   const int wv = threadIdx.x / WAVE;
@@ -83,9 +95,13 @@ extern "C" __global__ __launch_bounds__(WAVE * SS_WAVES) void segsum_mfma_kernel
           const int olimb = (orow >> 2) & 1;
           const unsigned long long partial = (unsigned long long)p64[reg];
           if (partial) {
-            atomicAdd(reinterpret_cast<unsigned long long*>(
-                          &out[(int64_t)s * NRES + ores]),
-                      olimb ? (partial << 16) : partial);
+            const unsigned long long v = olimb ? (partial << 16) : partial;
+            if (single_win) {
+              atomicAdd(&s_acc[s * NRES + ores], v);  // LDS, block-local
+            } else {
+              atomicAdd(reinterpret_cast<unsigned long long*>(
+                            &out[(int64_t)s * NRES + ores]), v);
+            }
           }
         }
       }
@@ -161,6 +177,15 @@ extern "C" __global__ __launch_bounds__(WAVE * SS_WAVES) void segsum_mfma_kernel
     __builtin_amdgcn_wave_barrier();  // LDS region reused next chunk
   }
   if (cur_win >= 0) flush(cur_win);
+  if (single_win) {
+    __syncthreads();  // all waves' LDS partials in
+    for (int idx = threadIdx.x; idx < S * NRES; idx += blockDim.x) {
+      const unsigned long long v = s_acc[idx];
+      if (v) {
+        atomicAdd(reinterpret_cast<unsigned long long*>(out) + idx, v);
+      }
+    }
+  }
 }
 
 #include <torch/extension.h>

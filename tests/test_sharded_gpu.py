@@ -46,6 +46,10 @@ def _rank(rank, q, mode):
     from mi355x_autoscaler.parallel.engine_sharded import ShardedEngine
     from mi355x_autoscaler.snapshot.delta import DeltaSnapshotCache
 
+    # explicit gloo: both ranks share ONE GPU, which RCCL refuses
+    # ("Duplicate GPU detected") — init before ShardedEngine so
+    # init_from_env sees the group and keeps it
+    dist.init_process_group(backend="gloo")
     try:
         snap = _full_snap()
         shard = shard_snapshot(snap, rank, WORLD)
