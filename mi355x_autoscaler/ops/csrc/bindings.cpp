@@ -116,6 +116,7 @@ std::shared_ptr<DecidePlan> create_decide_plan(
     torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
     torch::Tensor, torch::Tensor, torch::Tensor, int64_t, int64_t, int64_t);
 void run_decide_plan(const std::shared_ptr<DecidePlan>&);
+void run_decide_plan_phase(const std::shared_ptr<DecidePlan>&, int64_t);
 pybind11::tuple finish_decision(torch::Tensor blob_host, int64_t K,
                                 int64_t seg_max, int64_t over_provision,
                                 torch::Tensor max_new, torch::Tensor pool_cur);
@@ -191,4 +192,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "bind the fused-decide tensor set once");
   m.def("run_decide_plan", &run_decide_plan,
         "re-enqueue the fused decide with zero argument conversion");
+  m.def("run_decide_plan_phase", &run_decide_plan_phase,
+        "classic pipeline phase: 1=head (through pod_pool_map), 2=tail");
 }

@@ -67,6 +67,16 @@ bool mega_decide_launch(
     torch::Tensor blob, torch::Tensor pod_pool, int64_t hash_slots,
     int64_t seg_max, int64_t want_pod_pool);
 
+void fused_classic_phase(
+    torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
+    torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
+    torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
+    torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
+    torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
+    torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
+    torch::Tensor, torch::Tensor, torch::Tensor, int64_t, int64_t, int64_t,
+    int64_t);
+
 void fused_decide_launch(
     // snapshot (device views)
     torch::Tensor pod_req, torch::Tensor pod_flags, torch::Tensor pod_sel_hash,
@@ -123,15 +133,42 @@ void fused_decide_launch(
     }
   }
 
-  // workspace reinit; the blob needs zeroing only when a producer
-  // kernel is skipped (binpack writes placed[:G], units writes the
-  // rest + the readback epilogue)
-  hipMemsetAsync(ws_i32.data_ptr<int32_t>(), 0,
-                 (size_t)ws_i32.numel() * sizeof(int32_t), stream);
-  if (N == 0 || K == 0) {
-    hipMemsetAsync(blob.data_ptr<int64_t>(), 0,
-                   (size_t)blob.numel() * sizeof(int64_t), stream);
-  }
+  fused_classic_phase(pod_req, pod_flags, pod_sel_hash, pod_sel_pairs,
+                      pod_sel_count, pod_node, node_alloc, node_labels,
+                      node_label_count, node_unsched, pool_unit, pool_labels,
+                      pool_label_count, pool_max, pool_cur, ws_i32, slot_ks,
+                      slot_kr, slot2seg, seg_sel_pairs, seg_sel_count,
+                      seg_sel_hash, seg_req, seg_count, seg_key, seg_first,
+                      slot_of_pod, free, elig, seg_pool, run_scratch, blob,
+                      pod_pool, hash_slots, seg_max, want_pod_pool, 3);
+}
+
+// Classic multi-launch pipeline with a PHASE mask: 1 = head (hash ->
+// compact -> free -> eligibility -> route -> pod_pool_map), 2 = tail
+// (FFD place + units/epilogue), 3 = both. The 1M resident path runs
+// head, forks the side-stream MFMA demand segsum off pod_pool, then
+// runs the tail — the single-workgroup place/units kernels leave
+// 255/256 CUs idle, exactly where the segsum overlaps.
+void fused_classic_phase(
+    torch::Tensor pod_req, torch::Tensor pod_flags, torch::Tensor pod_sel_hash,
+    torch::Tensor pod_sel_pairs, torch::Tensor pod_sel_count,
+    torch::Tensor pod_node, torch::Tensor node_alloc, torch::Tensor node_labels,
+    torch::Tensor node_label_count, torch::Tensor node_unsched,
+    torch::Tensor pool_unit, torch::Tensor pool_labels,
+    torch::Tensor pool_label_count, torch::Tensor pool_max, torch::Tensor pool_cur,
+    torch::Tensor ws_i32, torch::Tensor slot_ks, torch::Tensor slot_kr,
+    torch::Tensor slot2seg, torch::Tensor seg_sel_pairs,
+    torch::Tensor seg_sel_count, torch::Tensor seg_sel_hash,
+    torch::Tensor seg_req, torch::Tensor seg_count, torch::Tensor seg_key,
+    torch::Tensor seg_first, torch::Tensor slot_of_pod, torch::Tensor free,
+    torch::Tensor elig, torch::Tensor seg_pool, torch::Tensor run_scratch,
+    torch::Tensor blob, torch::Tensor pod_pool, int64_t hash_slots,
+    int64_t seg_max, int64_t want_pod_pool, int64_t phase) {
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  const int64_t H = hash_slots;
+  const int64_t SM = seg_max;
+  const int64_t K = pool_unit.size(0);
+  const int64_t N = node_alloc.size(0);
 
   auto slot_state = ws_i32.narrow(0, 0, H);
   auto slot_count = ws_i32.narrow(0, H, H);
@@ -140,53 +177,67 @@ void fused_decide_launch(
   auto g_dev = stat.narrow(0, 0, 1);
   auto overflow = stat.narrow(0, 2, 1);
 
-  segment_hash(pod_flags, pod_sel_hash, pod_req, slot_ks, slot_kr, slot_state,
-               slot_count, slot_first, slot_of_pod, overflow);
-  segment_compact(slot_ks, slot_kr, slot_state, slot_count, slot_first,
-                  pod_sel_pairs, pod_sel_count, pod_req, seg_sel_pairs,
-                  seg_sel_count, seg_sel_hash, seg_req, seg_count, seg_key,
-                  seg_first, slot2seg, stat.narrow(0, 0, 2), c10::nullopt);
+  if (phase & 1) {
+    // workspace reinit; the blob needs zeroing only when a producer
+    // kernel is skipped (binpack writes placed[:G], units writes the
+    // rest + the readback epilogue)
+    hipMemsetAsync(ws_i32.data_ptr<int32_t>(), 0,
+                   (size_t)ws_i32.numel() * sizeof(int32_t), stream);
+    if (N == 0 || K == 0) {
+      hipMemsetAsync(blob.data_ptr<int64_t>(), 0,
+                     (size_t)blob.numel() * sizeof(int64_t), stream);
+    }
 
-  // free = alloc - bound requests
-  hipMemcpyAsync(free.data_ptr<int32_t>(), node_alloc.data_ptr<int32_t>(),
-                 (size_t)node_alloc.numel() * sizeof(int32_t),
-                 hipMemcpyDeviceToDevice, stream);
-  node_free_subtract(pod_node, pod_req, free);
+    segment_hash(pod_flags, pod_sel_hash, pod_req, slot_ks, slot_kr, slot_state,
+                 slot_count, slot_first, slot_of_pod, overflow);
+    segment_compact(slot_ks, slot_kr, slot_state, slot_count, slot_first,
+                    pod_sel_pairs, pod_sel_count, pod_req, seg_sel_pairs,
+                    seg_sel_count, seg_sel_hash, seg_req, seg_count, seg_key,
+                    seg_first, slot2seg, stat.narrow(0, 0, 2), c10::nullopt);
 
-  if (N > 0) {
-    fit_eligibility(node_labels, node_label_count, node_unsched, seg_sel_pairs,
-                    seg_sel_count, g_dev, elig);
-  }
-  seg_pool_route(seg_sel_pairs, seg_sel_count, seg_req, pool_labels,
-                 pool_label_count, pool_unit, g_dev, seg_pool);
-  if (want_pod_pool) {
-    pod_pool_map(slot_of_pod, slot2seg, seg_pool, pod_pool);
-  }
-
-  auto placed = blob.narrow(0, 0, SM);
-  auto placed_new = blob.narrow(0, SM, SM);
-  auto unful = blob.narrow(0, 2 * SM, SM);
-  if (N > 0) {
-    binpack_place(free, elig, seg_req, seg_count, g_dev, placed);
-  }
-  if (K > 0) {
-    auto units_cnt = blob.narrow(0, 3 * SM, K);
-    auto pool_residual = blob.narrow(0, 3 * SM + K, 4 * K);
-    // units also runs the readback epilogue: segment sel-hash column +
-    // 3-word stat (G, sort-ovf, hash-ovf) land on the blob tail so the
-    // host needs exactly ONE readback (the sel-hash feeds
-    // unfulfillable_detail — without the mirror that path cost a
-    // second D2H sync per decide)
-    binpack_units(seg_pool, seg_req, seg_count, placed, pool_unit, pool_max,
-                  pool_cur, g_dev, run_scratch, units_cnt, placed_new, unful,
-                  pool_residual, seg_sel_hash.narrow(0, 0, SM), stat, blob);
-  } else {
-    hipMemcpyAsync(blob.data_ptr<int64_t>() + (blob.numel() - 2 - SM),
-                   seg_sel_hash.data_ptr<int64_t>(), (size_t)SM * sizeof(int64_t),
+    // free = alloc - bound requests
+    hipMemcpyAsync(free.data_ptr<int32_t>(), node_alloc.data_ptr<int32_t>(),
+                   (size_t)node_alloc.numel() * sizeof(int32_t),
                    hipMemcpyDeviceToDevice, stream);
-    hipMemcpyAsync(blob.data_ptr<int64_t>() + (blob.numel() - 2),
-                   stat.data_ptr<int32_t>(), 3 * sizeof(int32_t),
-                   hipMemcpyDeviceToDevice, stream);
+    node_free_subtract(pod_node, pod_req, free);
+
+    if (N > 0) {
+      fit_eligibility(node_labels, node_label_count, node_unsched, seg_sel_pairs,
+                      seg_sel_count, g_dev, elig);
+    }
+    seg_pool_route(seg_sel_pairs, seg_sel_count, seg_req, pool_labels,
+                   pool_label_count, pool_unit, g_dev, seg_pool);
+    if (want_pod_pool) {
+      pod_pool_map(slot_of_pod, slot2seg, seg_pool, pod_pool);
+    }
+  }
+
+  if (phase & 2) {
+    auto placed = blob.narrow(0, 0, SM);
+    auto placed_new = blob.narrow(0, SM, SM);
+    auto unful = blob.narrow(0, 2 * SM, SM);
+    if (N > 0) {
+      binpack_place(free, elig, seg_req, seg_count, g_dev, placed);
+    }
+    if (K > 0) {
+      auto units_cnt = blob.narrow(0, 3 * SM, K);
+      auto pool_residual = blob.narrow(0, 3 * SM + K, 4 * K);
+      // units also runs the readback epilogue: segment sel-hash column +
+      // 3-word stat (G, sort-ovf, hash-ovf) land on the blob tail so the
+      // host needs exactly ONE readback (the sel-hash feeds
+      // unfulfillable_detail — without the mirror that path cost a
+      // second D2H sync per decide)
+      binpack_units(seg_pool, seg_req, seg_count, placed, pool_unit, pool_max,
+                    pool_cur, g_dev, run_scratch, units_cnt, placed_new, unful,
+                    pool_residual, seg_sel_hash.narrow(0, 0, SM), stat, blob);
+    } else {
+      hipMemcpyAsync(blob.data_ptr<int64_t>() + (blob.numel() - 2 - SM),
+                     seg_sel_hash.data_ptr<int64_t>(), (size_t)SM * sizeof(int64_t),
+                     hipMemcpyDeviceToDevice, stream);
+      hipMemcpyAsync(blob.data_ptr<int64_t>() + (blob.numel() - 2),
+                     stat.data_ptr<int32_t>(), 3 * sizeof(int32_t),
+                     hipMemcpyDeviceToDevice, stream);
+    }
   }
 }
 
@@ -338,6 +389,16 @@ std::shared_ptr<DecidePlan> create_decide_plan(
   p->seg_max = seg_max;
   p->want_pod_pool = want_pod_pool;
   return p;
+}
+
+void run_decide_plan_phase(const std::shared_ptr<DecidePlan>& p,
+                           int64_t phase) {
+  auto& t = p->t;
+  fused_classic_phase(t[0], t[1], t[2], t[3], t[4], t[5], t[6], t[7], t[8],
+                      t[9], t[10], t[11], t[12], t[13], t[14], t[15], t[16],
+                      t[17], t[18], t[19], t[20], t[21], t[22], t[23], t[24],
+                      t[25], t[26], t[27], t[28], t[29], t[30], t[31], t[32],
+                      p->hash_slots, p->seg_max, p->want_pod_pool, phase);
 }
 
 void run_decide_plan(const std::shared_ptr<DecidePlan>& p) {

@@ -380,6 +380,16 @@ def segsum_workitems(seg_of: torch.Tensor, num_segments: int):
     return wi_win.to(torch.int32), wi_start, wi_end
 
 
+def _mega_max_p() -> int:
+    """P above which the orchestrator takes the classic multi-launch
+    pipeline instead of the cooperative mega kernel (mirrors
+    orchestrator.cpp's MI355X_AS_MEGA_MAX_P dispatch)."""
+    try:
+        return int(os.environ.get("MI355X_AS_MEGA_MAX_P", "131072"))
+    except ValueError:
+        return 131072
+
+
 def segsum(values: torch.Tensor, seg_of: torch.Tensor, num_segments: int, ops,
            sorted_ids: bool = True, out: Optional[torch.Tensor] = None):
     """MFMA segmented sum -> [S,4] int64 (values >= 0; ids with -1 are
@@ -440,6 +450,7 @@ class GpuEngine(SimulationEngine):
         self._ev_fork = None
         self._ev_join = None
         self._pp_guard: Dict[bool, object] = {}  # flip -> segsum-done event
+        self._demand_forked = False
         self._side_done: Optional[torch.Tensor] = None  # cuda Event when pending
         d = self.device
         # persistent hash workspace; the mutable i32 parts live in ONE
@@ -693,45 +704,32 @@ class GpuEngine(SimulationEngine):
                         self._plan_cache.clear()
                     plan = ops.create_decide_plan(*tensors, *scalars)
                     self._plan_cache[pkey] = plan
-                ops.run_decide_plan(plan)
+                # large-P classic path: run the HEAD (through
+                # pod_pool_map), fork the side-stream MFMA demand
+                # segsum, then the TAIL — the single-workgroup FFD
+                # place/units leave 255/256 CUs idle, which is exactly
+                # where the segsum overlaps (the mega path at small P
+                # is one kernel and keeps the post-pipeline fork)
+                split = (want_pod_pool and self._side_stream is not None
+                         and ds.P > _mega_max_p())
+                if split:
+                    ops.run_decide_plan_phase(plan, 1)
+                    self._fork_demand_segsum(ds, pod_pool, K, stable_pods)
+                    self._demand_forked = True
+                    ops.run_decide_plan_phase(plan, 2)
+                else:
+                    ops.run_decide_plan(plan)
 
         # MFMA per-pool demand (zero-sort path; skipped for >16 pools) —
-        # enqueued on the side stream AFTER the pipeline's pod_pool map,
-        # overlapping the blob readback + host decision finish. The next
+        # enqueued on the side stream AFTER the pipeline's pod_pool map
+        # (mid-pipeline for the split large-P path above), overlapping
+        # the FFD tail / blob readback / host decision finish. The next
         # decide waits on it (stream-level) before rewriting pod_pool.
         if want_pod_pool:
-            if self._side_stream is not None:
-                # persistent fork/join events: re-recording a consumed
-                # event is safe (the wait precedes the re-record in
-                # program order) and saves two allocations per decide
-                if self._ev_fork is None:
-                    self._ev_fork = torch.cuda.Event()
-                    self._ev_join = torch.cuda.Event()
-                self._ev_fork.record()
-                # double-buffered demand output: reuse allocations but
-                # never overwrite the tensor pool_demand() handed out
-                # for the PREVIOUS decide
-                self._demand_flip = not getattr(self, "_demand_flip", False)
-                dkey = (K, self._demand_flip)
-                dbuf = self._demand_bufs.get(dkey)
-                if dbuf is None:
-                    dbuf = torch.zeros((K, 4), dtype=torch.int64,
-                                       device=self.device)
-                    self._demand_bufs[dkey] = dbuf
-                with torch.cuda.stream(self._side_stream):
-                    self._side_stream.wait_event(self._ev_fork)
-                    self._last_demand = segsum(ds.pod_req, pod_pool, K, ops,
-                                               out=dbuf)
-                    self._ev_join.record(self._side_stream)
-                    self._side_done = self._ev_join
-                    if stable_pods:
-                        # per-buffer read-done marker for the A-B-A fence
-                        fl = bool(self._pp_flip)
-                        gev = self._pp_guard.get(fl)
-                        if gev is None:
-                            self._pp_guard[fl] = gev = torch.cuda.Event()
-                        gev.record(self._side_stream)
-                self._last_demand.record_stream(self._side_stream)
+            if self._demand_forked:
+                self._demand_forked = False  # split path already forked
+            elif self._side_stream is not None:
+                self._fork_demand_segsum(ds, pod_pool, K, stable_pods)
             else:
                 self._last_demand = segsum(ds.pod_req, pod_pool, K, ops)
         else:
@@ -794,6 +792,39 @@ class GpuEngine(SimulationEngine):
             max_new_h, ds.pool_cur_h, K, pool_names, over_provision,
             lambda: blob_h[-2 - SEG_MAX : -2],
         )
+
+    def _fork_demand_segsum(self, ds, pod_pool, K, stable_pods):
+        """Enqueue the MFMA demand segsum on the side stream, ordered
+        after whatever the MAIN stream has enqueued so far (pod_pool
+        must be final at that point)."""
+        ops = self.ops
+        # persistent fork/join events: re-recording a consumed event is
+        # safe (the wait precedes the re-record in program order)
+        if self._ev_fork is None:
+            self._ev_fork = torch.cuda.Event()
+            self._ev_join = torch.cuda.Event()
+        self._ev_fork.record()
+        # double-buffered demand output: reuse allocations but never
+        # overwrite the tensor pool_demand() handed out last decide
+        self._demand_flip = not getattr(self, "_demand_flip", False)
+        dkey = (K, self._demand_flip)
+        dbuf = self._demand_bufs.get(dkey)
+        if dbuf is None:
+            dbuf = torch.zeros((K, 4), dtype=torch.int64, device=self.device)
+            self._demand_bufs[dkey] = dbuf
+        with torch.cuda.stream(self._side_stream):
+            self._side_stream.wait_event(self._ev_fork)
+            self._last_demand = segsum(ds.pod_req, pod_pool, K, ops, out=dbuf)
+            self._ev_join.record(self._side_stream)
+            self._side_done = self._ev_join
+            if stable_pods:
+                # per-buffer read-done marker for the A-B-A fence
+                fl = bool(self._pp_flip)
+                gev = self._pp_guard.get(fl)
+                if gev is None:
+                    self._pp_guard[fl] = gev = torch.cuda.Event()
+                gev.record(self._side_stream)
+        self._last_demand.record_stream(self._side_stream)
 
     def pool_demand(self) -> Optional[torch.Tensor]:
         """Per-pool pending demand totals from the MFMA segsum of the

@@ -481,6 +481,17 @@ __device__ inline void dev_binpack(
     const int lane = tid & (WAVE - 1);
     const int wv = tid / WAVE;
     __shared__ int64_t s_ctot[2][WAVE];
+    // stage the whole eligibility bitmask in LDS when it fits: the
+    // per-chunk erow loads were dependent L2/HBM round trips on the
+    // single resident workgroup (PMC: 692k VALU yet ~70 us — pure
+    // latency), and every segment touches every chunk word
+#define BP_ELIG_LDS 2048
+    __shared__ uint64_t s_elig[BP_ELIG_LDS];
+    const bool stage_elig = (int64_t)G * W <= BP_ELIG_LDS;
+    if (stage_elig) {
+      for (int idx = tid; idx < G * W; idx += BLOCK) s_elig[idx] = elig[idx];
+    }
+    __syncthreads();
     // my chunk ids: wv, wv+NW_, ... ; node j = chunk*WAVE + lane
     int32_t fr2[CPW][NRES];
     int cid[CPW];
@@ -507,7 +518,8 @@ __device__ inline void dev_binpack(
 #pragma unroll
       for (int r = 0; r < NRES; ++r)
         rq[r] = g < BP_PRELOAD ? s_req[g * NRES + r] : seg_req[g * NRES + r];
-      const uint64_t* erow = elig + (int64_t)g * W;
+      const uint64_t* erow =
+          stage_elig ? s_elig + (int64_t)g * W : elig + (int64_t)g * W;
       int64_t m_q[CPW], inc_q[CPW];
 #pragma unroll
       for (int q = 0; q < CPW; ++q) {
