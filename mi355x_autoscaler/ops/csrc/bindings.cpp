@@ -117,6 +117,12 @@ std::shared_ptr<DecidePlan> create_decide_plan(
     torch::Tensor, torch::Tensor, torch::Tensor, int64_t, int64_t, int64_t);
 void run_decide_plan(const std::shared_ptr<DecidePlan>&);
 void run_decide_plan_phase(const std::shared_ptr<DecidePlan>&, int64_t);
+void set_plan_side(const std::shared_ptr<DecidePlan>&, torch::Tensor,
+                   torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
+                   int64_t);
+int64_t run_decide_plan_auto(const std::shared_ptr<DecidePlan>&, int64_t);
+void main_wait_side();
+void side_sync();
 pybind11::tuple finish_decision(torch::Tensor blob_host, int64_t K,
                                 int64_t seg_max, int64_t over_provision,
                                 torch::Tensor max_new, torch::Tensor pool_cur);
@@ -194,4 +200,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "re-enqueue the fused decide with zero argument conversion");
   m.def("run_decide_plan_phase", &run_decide_plan_phase,
         "classic pipeline phase: 1=head (through pod_pool_map), 2=tail");
+  m.def("set_plan_side", &set_plan_side,
+        "bind the side-stream demand segsum (work items + double buffers)");
+  m.def("run_decide_plan_auto", &run_decide_plan_auto,
+        "pipeline + side-stream demand segsum in ONE call; returns buffer idx");
+  m.def("main_wait_side", &main_wait_side,
+        "main stream waits for the last side segsum (stream-level)");
+  m.def("side_sync", &side_sync, "host-block until the last side segsum");
 }

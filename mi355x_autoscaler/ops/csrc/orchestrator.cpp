@@ -409,3 +409,91 @@ void run_decide_plan(const std::shared_ptr<DecidePlan>& p) {
                       t[25], t[26], t[27], t[28], t[29], t[30], t[31], t[32],
                       p->hash_slots, p->seg_max, p->want_pod_pool);
 }
+
+
+// ---------------------------------------------------------------------------
+// Orchestrator-owned side stream for the demand segsum. Statics are
+// per process (one GPU per process in every deployment shape); the
+// wait-on-unrecorded-event is a defined no-op, so the first call needs
+// no special casing.
+void segsum_mfma_on(torch::Tensor values, torch::Tensor seg_of,
+                    torch::Tensor wi_win, torch::Tensor wi_start,
+                    torch::Tensor wi_end, int64_t num_segments,
+                    torch::Tensor out, hipStream_t stream);
+
+static hipStream_t g_side_stream = nullptr;
+static hipEvent_t g_fork_ev = nullptr;
+static hipEvent_t g_join_ev = nullptr;
+
+static void side_init() {
+  if (g_side_stream == nullptr) {
+    hipStreamCreateWithFlags(&g_side_stream, hipStreamNonBlocking);
+    hipEventCreateWithFlags(&g_fork_ev, hipEventDisableTiming);
+    hipEventCreateWithFlags(&g_join_ev, hipEventDisableTiming);
+  }
+}
+
+void set_plan_side(const std::shared_ptr<DecidePlan>& p,
+                   torch::Tensor wi_win, torch::Tensor wi_start,
+                   torch::Tensor wi_end, torch::Tensor demand_a,
+                   torch::Tensor demand_b, int64_t K) {
+  p->wi_win = wi_win;
+  p->wi_start = wi_start;
+  p->wi_end = wi_end;
+  p->demand_a = demand_a;
+  p->demand_b = demand_b;
+  p->K = K;
+  p->has_side = true;
+}
+
+// Run the whole decide INCLUDING the side-stream demand segsum in one
+// Python->C++ call. `split` selects the classic head/fork/tail overlap
+// (large P); otherwise the full pipeline (mega at small P) runs first
+// and the segsum forks after it. Returns which demand buffer was
+// written (0 = a, 1 = b), or -1 when the plan has no side work.
+int64_t run_decide_plan_auto(const std::shared_ptr<DecidePlan>& p,
+                             int64_t split) {
+  auto& t = p->t;
+  if (!p->has_side) {
+    run_decide_plan(p);
+    return -1;
+  }
+  side_init();
+  auto main_stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  // order THIS pipeline (which rewrites pod_pool / pod columns) after
+  // the previous decide's segsum — stream-level, normally long past
+  hipStreamWaitEvent(main_stream, g_join_ev, 0);
+  if (split) {
+    run_decide_plan_phase(p, 1);  // classic head: through pod_pool_map
+  } else {
+    run_decide_plan(p);           // mega or full classic
+  }
+  hipEventRecord(g_fork_ev, main_stream);
+  hipStreamWaitEvent(g_side_stream, g_fork_ev, 0);
+  const int use_b = p->flip;
+  p->flip ^= 1;
+  torch::Tensor out = use_b ? p->demand_b : p->demand_a;
+  hipMemsetAsync(out.data_ptr<int64_t>(), 0,
+                 (size_t)out.numel() * sizeof(int64_t), g_side_stream);
+  segsum_mfma_on(t[0], t[32], p->wi_win, p->wi_start, p->wi_end, p->K, out,
+                 g_side_stream);
+  hipEventRecord(g_join_ev, g_side_stream);
+  if (split) {
+    run_decide_plan_phase(p, 2);  // classic tail overlaps the segsum
+  }
+  return use_b;
+}
+
+// main stream orders after the last side segsum (resident scatter
+// rewrites pod columns the segsum may still read)
+void main_wait_side() {
+  side_init();
+  hipStreamWaitEvent(at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream(),
+                     g_join_ev, 0);
+}
+
+// host-blocking: pool_demand() hands the tensor out
+void side_sync() {
+  side_init();
+  hipEventSynchronize(g_join_ev);
+}

@@ -390,6 +390,24 @@ def _mega_max_p() -> int:
         return 131072
 
 
+def fastpath_workitems(device, M: int):
+    """(win, start, end) work items for the zero-sort S<=16 segsum
+    fast path, cached per (device, M). Tensors referenced by cached
+    DecidePlans stay alive through the plan's C++ references even if
+    the cache rotates."""
+    cached = _FASTPATH_WI.get((device, M))
+    if cached is None:
+        T = (M + _SEGSUM_CHUNK - 1) // _SEGSUM_CHUNK
+        wi_win = torch.zeros(T, dtype=torch.int32, device=device)
+        wi_start = torch.arange(T, dtype=torch.int64, device=device) * _SEGSUM_CHUNK
+        wi_end = torch.minimum(wi_start + _SEGSUM_CHUNK,
+                               torch.tensor(M, dtype=torch.int64, device=device))
+        _FASTPATH_WI.clear()
+        cached = (wi_win, wi_start, wi_end)
+        _FASTPATH_WI[(device, M)] = cached
+    return cached
+
+
 def segsum(values: torch.Tensor, seg_of: torch.Tensor, num_segments: int, ops,
            sorted_ids: bool = True, out: Optional[torch.Tensor] = None):
     """MFMA segmented sum -> [S,4] int64 (values >= 0; ids with -1 are
@@ -406,17 +424,7 @@ def segsum(values: torch.Tensor, seg_of: torch.Tensor, num_segments: int, ops,
     if M == 0 or num_segments == 0:
         return out
     if num_segments <= 16:
-        T = (M + _SEGSUM_CHUNK - 1) // _SEGSUM_CHUNK
-        cached = _FASTPATH_WI.get((device, M))
-        if cached is None:
-            wi_win = torch.zeros(T, dtype=torch.int32, device=device)
-            wi_start = torch.arange(T, dtype=torch.int64, device=device) * _SEGSUM_CHUNK
-            wi_end = torch.minimum(wi_start + _SEGSUM_CHUNK,
-                                   torch.tensor(M, dtype=torch.int64, device=device))
-            _FASTPATH_WI.clear()
-            _FASTPATH_WI[(device, M)] = (wi_win, wi_start, wi_end)
-        else:
-            wi_win, wi_start, wi_end = cached
+        wi_win, wi_start, wi_end = fastpath_workitems(device, M)
     else:
         assert sorted_ids, "segsum with >16 segments needs sorted ids"
         wi_win, wi_start, wi_end = segsum_workitems(seg_of, num_segments)
@@ -451,6 +459,8 @@ class GpuEngine(SimulationEngine):
         self._ev_join = None
         self._pp_guard: Dict[bool, object] = {}  # flip -> segsum-done event
         self._demand_forked = False
+        self._side_native = False  # last segsum ran on the C++ side stream
+        self._plan_demand: Dict[tuple, tuple] = {}
         self._side_done: Optional[torch.Tensor] = None  # cuda Event when pending
         d = self.device
         # persistent hash workspace; the mutable i32 parts live in ONE
@@ -657,7 +667,8 @@ class GpuEngine(SimulationEngine):
             # stream after the previous cycle's side-stream segsum.
             # (Resident mode double-buffers pod_pool and waits at the
             # scatter instead, where the wait has ~a full readback of
-            # natural slack.)
+            # natural slack. The native plan path orders itself inside
+            # run_decide_plan_auto.)
             torch.cuda.current_stream().wait_event(self._side_done)
             self._side_done = None
         bufs = self._decide_buffers(ds.P, K, want_pod_pool)
@@ -702,23 +713,32 @@ class GpuEngine(SimulationEngine):
                 if plan is None:
                     if len(self._plan_cache) > 8:
                         self._plan_cache.clear()
+                        self._plan_demand.clear()
                     plan = ops.create_decide_plan(*tensors, *scalars)
+                    if want_pod_pool:
+                        # bind the side-stream demand segsum into the
+                        # plan: the whole decide becomes ONE C++ call
+                        # (the Python stream/event dance cost ~28 us)
+                        wi = fastpath_workitems(self.device, ds.P)
+                        da = torch.zeros((K, 4), dtype=torch.int64,
+                                         device=self.device)
+                        db = torch.zeros((K, 4), dtype=torch.int64,
+                                         device=self.device)
+                        ops.set_plan_side(plan, *wi, da, db, K)
+                        self._plan_demand[pkey] = (da, db)
                     self._plan_cache[pkey] = plan
-                # large-P classic path: run the HEAD (through
-                # pod_pool_map), fork the side-stream MFMA demand
-                # segsum, then the TAIL — the single-workgroup FFD
-                # place/units leave 255/256 CUs idle, which is exactly
-                # where the segsum overlaps (the mega path at small P
-                # is one kernel and keeps the post-pipeline fork)
-                split = (want_pod_pool and self._side_stream is not None
-                         and ds.P > _mega_max_p())
-                if split:
-                    ops.run_decide_plan_phase(plan, 1)
-                    self._fork_demand_segsum(ds, pod_pool, K, stable_pods)
+                # large-P classic path: head (through pod_pool_map) ->
+                # fork segsum -> tail, so the single-workgroup FFD
+                # place/units (255/256 CUs idle) overlap the segsum;
+                # at mega sizes the full kernel runs first
+                split = want_pod_pool and ds.P > _mega_max_p()
+                flip = ops.run_decide_plan_auto(plan, 1 if split else 0)
+                if flip >= 0:
+                    da, db = self._plan_demand[pkey]
+                    self._last_demand = db if flip else da
                     self._demand_forked = True
-                    ops.run_decide_plan_phase(plan, 2)
-                else:
-                    ops.run_decide_plan(plan)
+                    self._side_native = True
+                    self._side_done = None
 
         # MFMA per-pool demand (zero-sort path; skipped for >16 pools) —
         # enqueued on the side stream AFTER the pipeline's pod_pool map
@@ -727,7 +747,7 @@ class GpuEngine(SimulationEngine):
         # decide waits on it (stream-level) before rewriting pod_pool.
         if want_pod_pool:
             if self._demand_forked:
-                self._demand_forked = False  # split path already forked
+                self._demand_forked = False  # plan path already forked
             elif self._side_stream is not None:
                 self._fork_demand_segsum(ds, pod_pool, K, stable_pods)
             else:
@@ -794,6 +814,7 @@ class GpuEngine(SimulationEngine):
         )
 
     def _fork_demand_segsum(self, ds, pod_pool, K, stable_pods):
+        self._side_native = False
         """Enqueue the MFMA demand segsum on the side stream, ordered
         after whatever the MAIN stream has enqueued so far (pod_pool
         must be final at that point)."""
@@ -830,7 +851,9 @@ class GpuEngine(SimulationEngine):
         """Per-pool pending demand totals from the MFMA segsum of the
         last decide (None when >16 pools). Synchronizes the side
         stream before handing the tensor out."""
-        if self._side_done is not None:
+        if self._side_native:
+            self.ops.side_sync()
+        elif self._side_done is not None:
             self._side_done.synchronize()
         return self._last_demand
 
@@ -1317,10 +1340,11 @@ class GpuEngine(SimulationEngine):
                                     "pod_views": pod_views}
             cache.drain_dirty()
         else:
-            if self._side_done is not None:
+            if self._side_native:
                 # the scatter rewrites pod columns the side-stream
-                # demand segsum may still read; stream-level wait here
-                # (never a host sync)
+                # demand segsum may still read; stream-level wait
+                self.ops.main_wait_side()
+            elif self._side_done is not None:
                 torch.cuda.current_stream().wait_event(self._side_done)
                 self._side_done = None
             self._scatter_rows(res["pod_views"], cache, cache.drain_dirty())

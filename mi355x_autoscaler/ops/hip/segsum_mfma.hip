@@ -194,13 +194,12 @@ extern "C" __global__ __launch_bounds__(WAVE * SS_WAVES) void segsum_mfma_kernel
 // values/seg_of sorted by segment; work items precomputed on device by
 // ops/engine_gpu.py (window = seg // 16, chunks of <=256 elements that
 // never cross a window boundary). out must be zeroed.
-void segsum_mfma(torch::Tensor values, torch::Tensor seg_of,
-                 torch::Tensor wi_win, torch::Tensor wi_start,
-                 torch::Tensor wi_end, int64_t num_segments,
-                 torch::Tensor out) {
+void segsum_mfma_on(torch::Tensor values, torch::Tensor seg_of,
+                    torch::Tensor wi_win, torch::Tensor wi_start,
+                    torch::Tensor wi_end, int64_t num_segments,
+                    torch::Tensor out, hipStream_t stream) {
   const int64_t T = wi_win.size(0);
   if (T == 0) return;
-  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   // SS_WAVES waves per block; cap at 256 blocks = 1024 waves (one per
   // SIMD). MORE waves would hand each wave ~1 chunk and collapse the
   // flush-once-per-window amortization back into flush-per-chunk
@@ -209,9 +208,17 @@ void segsum_mfma(torch::Tensor values, torch::Tensor seg_of,
   const uint32_t grid =
       (uint32_t)std::min<int64_t>((T + SS_WAVES - 1) / SS_WAVES, 256);
   hipLaunchKernelGGL(segsum_mfma_kernel, dim3(grid), dim3(WAVE * SS_WAVES), 0,
-                     stream.stream(), values.data_ptr<int32_t>(),
+                     stream, values.data_ptr<int32_t>(),
                      seg_of.data_ptr<int32_t>(), wi_win.data_ptr<int32_t>(),
                      wi_start.data_ptr<int64_t>(), wi_end.data_ptr<int64_t>(),
                      (int32_t)num_segments, T,
                      reinterpret_cast<uint64_t*>(out.data_ptr<int64_t>()));
+}
+
+void segsum_mfma(torch::Tensor values, torch::Tensor seg_of,
+                 torch::Tensor wi_win, torch::Tensor wi_start,
+                 torch::Tensor wi_end, int64_t num_segments,
+                 torch::Tensor out) {
+  segsum_mfma_on(values, seg_of, wi_win, wi_start, wi_end, num_segments, out,
+                 at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream());
 }
