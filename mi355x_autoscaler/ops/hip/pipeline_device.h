@@ -83,9 +83,13 @@ __device__ inline int probe_claim(int64_t ks, int64_t kr, int32_t H,
   return -1;
 }
 
-// Wave-aggregated direct-global hash build (the small-P form): one
-// probe + one count/first atomic pair per distinct key per wave.
-// Grid-strided over pods; safe at any grid shape.
+// Wave-aggregated direct-global hash build (the small-P form).
+// Round 2: per-lane 1-entry memo with LOCAL run counting — a burst is
+// runs of identical (selector, request) keys, so the steady state per
+// pod is two register compares + an increment; counts/min-ids flush
+// once per key RUN (memo switch or loop end) instead of one global
+// atomic pair per distinct key per wave-iteration. Grid-strided over
+// pods; safe at any grid shape.
 __device__ inline void dev_hash_build_small_ex(
     const int32_t* __restrict__ pod_flags, const int64_t* __restrict__ sel_hash,
     const int32_t* __restrict__ pod_req, int64_t P, int32_t H,
@@ -94,6 +98,17 @@ __device__ inline void dev_hash_build_small_ex(
     int32_t* __restrict__ slot_first, int32_t* __restrict__ slot_of_pod,
     int32_t* __restrict__ overflow, int64_t start, int64_t stride) {
   const int lane = threadIdx.x & (WAVE - 1);
+  int64_t memo_ks = 0, memo_kr = 0;
+  int memo_slot = INT32_MIN;
+  int32_t memo_cnt = 0;
+  int32_t memo_min_i0 = 0;
+  auto memo_flush = [&]() {
+    if (memo_slot >= 0 && memo_cnt > 0) {
+      atomicAdd(&slot_count[memo_slot], memo_cnt);
+      atomicMax(&slot_first[memo_slot], INT32_MAX - memo_min_i0);
+    }
+    memo_cnt = 0;
+  };
   for (int64_t i0 = start; ; i0 += stride) {
     if (__all(i0 >= P)) break;
     const int pf = i0 < P ? pod_flags[i0] : 0;
@@ -109,13 +124,23 @@ __device__ inline void dev_hash_build_small_ex(
       slot_of_pod[i0] = -1;
     }
 
-    uint64_t pending_lanes = __ballot(live);
     int my_slot = -1;
+    bool hit = false;
+    if (live && memo_slot != INT32_MIN && ks == memo_ks && kr == memo_kr) {
+      my_slot = memo_slot;
+      hit = true;
+      if (memo_slot >= 0) {
+        if (memo_cnt == 0) memo_min_i0 = (int32_t)i0;
+        ++memo_cnt;
+      }
+    }
+
+    uint64_t pending_lanes = __ballot(live && !hit);
     while (pending_lanes) {
       const int leader = __ffsll((unsigned long long)pending_lanes) - 1;
       const int64_t k0 = __shfl(ks, leader, WAVE);
       const int64_t r0 = __shfl(kr, leader, WAVE);
-      const bool match = live && ks == k0 && kr == r0;
+      const bool match = live && !hit && ks == k0 && kr == r0;
       const uint64_t match_mask = __ballot(match);
       int slot = -1;
       if (lane == leader) {
@@ -128,11 +153,20 @@ __device__ inline void dev_hash_build_small_ex(
         }
       }
       slot = __shfl(slot, leader, WAVE);
-      if (match) my_slot = slot;
+      if (match) {
+        my_slot = slot;
+        // counted by the leader already: seed the memo with cnt 0 so
+        // the run continues on the register fast path
+        memo_flush();
+        memo_ks = ks;
+        memo_kr = kr;
+        memo_slot = slot;
+      }
       pending_lanes &= ~match_mask;
     }
     if (i0 < P && live) slot_of_pod[i0] = my_slot;
   }
+  memo_flush();
 }
 
 __device__ inline void dev_hash_build_small(
