@@ -745,7 +745,9 @@ pybind11::dict encode_pods_json(pybind11::bytes body, pybind11::dict node_index,
     const size_t M = spans.size();
 
     unsigned hw = std::thread::hardware_concurrency();
-    size_t T = std::max<size_t>(1, std::min<size_t>({hw ? hw : 1, 8, (M + 4095) / 4096}));
+    // up to 16 threads, >=1024 pods each: the 8-thread/4096-line cap
+    // left most of the box idle on 10k-100k bursts
+    size_t T = std::max<size_t>(1, std::min<size_t>({hw ? hw : 1, 16, (M + 1023) / 1024}));
     shards.resize(T);
     std::vector<std::thread> threads;
     std::atomic<bool> failed{false};
@@ -899,7 +901,9 @@ pybind11::dict decode_watch_events(pybind11::bytes body,
     }
     const size_t M = lines.size();
     unsigned hw = std::thread::hardware_concurrency();
-    size_t T = std::max<size_t>(1, std::min<size_t>({hw ? hw : 1, 8, (M + 2047) / 2048}));
+    // up to 16 threads, >=512 events each (a 10k-event burst ran on
+    // 5 threads under the old 2048-line shard floor)
+    size_t T = std::max<size_t>(1, std::min<size_t>({hw ? hw : 1, 16, (M + 511) / 512}));
     shards.resize(T);
     std::vector<std::thread> threads;
     std::atomic<bool> failed{false};
