@@ -608,3 +608,51 @@ def test_randomized_differential_fuzz(engine):
     spec.loader.exec_module(mod)
     mod.ITERS = 12
     mod.main()
+
+
+def test_units_long_run_lists_and_splits(engine):
+    """Stress the round-2 array-backed units run store: one pool, ~700
+    distinct request shapes over tiny nodes forces hundreds of open
+    runs (multi-chunk wave scans), repeated mid-list boundary splits
+    and deep ord shifts — bit-equal to the oracle."""
+    import random
+
+    rng = random.Random(7)
+    nodes = [make_node_json("agentpool1", i) for i in range(2)]
+    pods = []
+    for i in range(1400):
+        # alternate wide/narrow shapes so later segments partially
+        # consume earlier runs (split + insert), not just append
+        cpu = rng.choice([3, 7, 11, 37, 101, 397, 998])
+        mem = rng.randrange(1, 900)
+        for j in range(rng.randrange(1, 4)):
+            pods.append(make_pod_json(f"u{i}-{j}", cpu_milli=cpu, mem_mib=mem,
+                                      pending=True))
+    snap = _snap(pods, nodes, max_size=4096)
+    assert snap.pool_max[0] >= 4096
+    d_gpu = engine.decide(snap)
+    d_cpu = CpuEngine().decide(snap)
+    _assert_decisions_equal(d_gpu, d_cpu)
+    assert sum(d_cpu.units_needed.values()) > 64  # many runs really opened
+
+
+def test_units_interleaved_pools_long_lists(engine):
+    """Same stress across 3 pools (waves own different pools) with
+    selector-pinned segments so each pool's ord list grows and splits
+    independently."""
+    import random
+
+    rng = random.Random(11)
+    pools = ["agentpool1", "bigpool", "gpupool"]
+    nodes_json = []
+    for p in pools:
+        nodes_json.append(make_node_json(p, 0))
+    pods = []
+    for i in range(900):
+        p = pools[i % 3]
+        pods.append(make_pod_json(
+            f"q{i}", cpu_milli=rng.choice([5, 13, 57, 333, 777]),
+            mem_mib=rng.randrange(1, 600), pending=True,
+            node_selector={"agentpool": p}))
+    snap = _snap(pods, nodes_json, max_size=4096)
+    _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))
