@@ -240,11 +240,14 @@ def main():
 
     import torch.distributed as dist
 
+    dist_backend = os.environ.get(
+        "MI355X_AS_DIST_BACKEND") or ("nccl" if use_gpu else "gloo")
     if world > 1:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        dist.init_process_group(backend="nccl" if use_gpu else "gloo")
+        dist.init_process_group(backend=dist_backend)
         if use_gpu:
-            torch.cuda.set_device(local_rank)
+            # modulo so a gloo readiness run can put 2 ranks on 1 GPU
+            torch.cuda.set_device(local_rank % torch.cuda.device_count())
 
     if use_gpu:
         from mi355x_autoscaler.ops.engine_gpu import GpuEngine
@@ -344,7 +347,7 @@ def main():
     # MAX over ranks of the timed window
     if world > 1:
         t = torch.tensor([total_s_local], dtype=torch.float64)
-        if use_gpu:
+        if use_gpu and dist_backend == "nccl":
             t = t.cuda()
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         total_s = float(t.item())
