@@ -252,7 +252,7 @@ def main():
     if use_gpu:
         from mi355x_autoscaler.ops.engine_gpu import GpuEngine
 
-        inner = GpuEngine(f"cuda:{local_rank}")
+        inner = GpuEngine(f"cuda:{local_rank % torch.cuda.device_count()}")
     else:
         inner = CpuEngine()
 
