@@ -22,8 +22,9 @@
 // N==0, and as the fallback when cooperative launch is unavailable.
 //
 // LDS: one dynamic carve, phase-exclusive — compact needs 40 KiB,
-// the eligibility tile 65 KiB; the units arena uses global scratch
-// here (its LDS tier only pays in the standalone launch).
+// the eligibility tile 65 KiB; by P4 that carve is idle and doubles
+// as the units run arena (round 2 — the round-1 global-scratch choice
+// predates the wave-parallel array run store).
 #include "pipeline_device.h"
 
 #include <hip/hip_cooperative_groups.h>
@@ -134,9 +135,13 @@ extern "C" __global__ __launch_bounds__(MEGA_BLOCK) void mega_decide_kernel(
     __shared__ int64_t lds_scan[MEGA_BLOCK / WAVE + 1];
     dev_binpack<MEGA_BLOCK>(free_, elig, seg_req, seg_count, N, G, blob,
                             lds_scan);
+    // the 65 KiB dynamic carve is idle by P4 (fit tiles done): hand
+    // it to the units run arena (1386 runs) instead of global scratch
     dev_units(seg_pool, seg_req, seg_count, /*placed=*/blob, pool_unit,
-              pool_max, pool_cur, G, K, RTOT, /*lds_cap=*/0, run_scratch,
-              /*lds_arena=*/nullptr,
+              pool_max, pool_cur, G, K, RTOT,
+              /*lds_cap=*/(int32_t)(MEGA_SMEM / (6 * sizeof(int64_t))),
+              run_scratch,
+              /*lds_arena=*/reinterpret_cast<int64_t*>(smem),
               /*units_cnt=*/blob + 3 * (int64_t)SM,
               /*placed_new=*/blob + SM,
               /*unful=*/blob + 2 * (int64_t)SM,
