@@ -23,7 +23,7 @@ from mi355x_autoscaler.sim.engine import CpuEngine
 from mi355x_autoscaler.sim.synth import burst_trace, wrap
 from mi355x_autoscaler.snapshot.encode import encode_snapshot
 
-ITERS = 60
+ITERS = int(__import__("os").environ.get("FUZZ_ITERS", "60"))
 
 POOL_MENU = [
     ("agentpool1", "Standard_D2_v2"),
