@@ -128,11 +128,24 @@ class Cluster:
 
     # ------------------------------------------------------------------
     def loop(self, max_cycles: Optional[int] = None, debug: bool = False):
-        """Sleep-loop driver [R: main.py while True — reconstructed]."""
+        """Sleep-loop driver [R: main.py while True — reconstructed].
+        Ordinary cycle failures retry next cycle (the reference's
+        crash-loop resilience: every cycle re-lists the world); a LOST
+        PEER RANK is different — retrying against a dead communicator
+        spins forever — so collective failures at world>1 surface as
+        RankLossError for the supervisor to relaunch the rank set
+        (SURVEY §5 failure detection)."""
+        from ..parallel import dist as pdist
+        from ..parallel.elastic import RankLossError, is_collective_failure
+
         while True:
             try:
                 self.loop_logic()
             except Exception as e:
+                if pdist.world_size() > 1 and is_collective_failure(e):
+                    logger.error("peer rank lost mid-cycle: %s", e)
+                    self.notifier.notify_error(f"rank loss: {e}")
+                    raise RankLossError(str(e)) from e
                 logger.exception("reconcile cycle failed")
                 self.notifier.notify_error(str(e))
                 if debug:

@@ -81,6 +81,18 @@ def shard_snapshot(snap: ClusterSnapshot, rank: int, world: int) -> ClusterSnaps
     return sub
 
 
+_COLLECTIVE_FAILURE_KEYWORDS = (
+    "nccl", "rccl", "gloo", "connection", "timed out", "timeout",
+    "closed", "peer", "aborted", "econnreset", "broken pipe")
+
+
+def is_collective_failure(exc: BaseException) -> bool:
+    """Does this exception look like a lost peer / dead communicator
+    (as opposed to an ordinary bug)?"""
+    msg = str(exc).lower()
+    return any(k in msg for k in _COLLECTIVE_FAILURE_KEYWORDS)
+
+
 def run_elastic_cycle(engine, snap: ClusterSnapshot, over_provision: int = 0):
     """One decide() with rank-loss translation: collective failures
     (RCCL aborts, gloo timeouts) surface as RankLossError after tearing
@@ -90,10 +102,7 @@ def run_elastic_cycle(engine, snap: ClusterSnapshot, over_provision: int = 0):
     try:
         return engine.decide(snap, over_provision=over_provision)
     except RuntimeError as e:
-        msg = str(e).lower()
-        if any(k in msg for k in ("nccl", "rccl", "gloo", "connection",
-                                  "timed out", "timeout", "closed", "peer",
-                                  "aborted", "econnreset", "broken pipe")):
+        if is_collective_failure(e):
             if dist.is_available() and dist.is_initialized():
                 try:
                     dist.destroy_process_group()

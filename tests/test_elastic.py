@@ -66,3 +66,50 @@ def test_rank_loss_error_translation():
         assert False
     except RuntimeError as e:
         assert "unrelated" in str(e)
+
+
+def test_loop_driver_translates_rank_loss(monkeypatch):
+    """The sleep-loop retries ordinary cycle failures but must NOT
+    spin against a dead communicator: at world>1 a collective failure
+    surfaces as RankLossError for the supervisor to relaunch."""
+    from mi355x_autoscaler.ctl.loop import Cluster, ClusterOptions
+    from mi355x_autoscaler.parallel import dist as pdist
+    from mi355x_autoscaler.parallel.elastic import RankLossError
+
+    cluster = Cluster.__new__(Cluster)
+    cluster.opts = ClusterOptions(sleep_seconds=0)
+    cluster._cycle = 0
+
+    class _N:
+        def notify_error(self, msg):
+            pass
+
+    cluster.notifier = _N()
+
+    def dead_cycle(now=None):
+        raise RuntimeError("Connection closed by peer")
+
+    cluster.loop_logic = dead_cycle
+
+    # world 1: an ordinary failure — retried, loop terminates by count
+    monkeypatch.setattr(pdist, "world_size", lambda: 1)
+    cluster.loop(max_cycles=2)
+    assert cluster._cycle == 2
+
+    # world 4: the same error is a lost peer — loop must raise
+    monkeypatch.setattr(pdist, "world_size", lambda: 4)
+    cluster._cycle = 0
+    try:
+        cluster.loop(max_cycles=2)
+        assert False, "expected RankLossError"
+    except RankLossError:
+        pass
+
+    # a non-collective bug at world 4 still retries
+    def buggy_cycle(now=None):
+        raise RuntimeError("some unrelated bug")
+
+    cluster.loop_logic = buggy_cycle
+    cluster._cycle = 0
+    cluster.loop(max_cycles=2)
+    assert cluster._cycle == 2
