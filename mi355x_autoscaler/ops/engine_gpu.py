@@ -436,6 +436,49 @@ def segsum(values: torch.Tensor, seg_of: torch.Tensor, num_segments: int, ops,
     return out
 
 
+class _CandidateList:
+    """Lazy sequence of DrainCandidate (util-ascending): maintain at
+    500 nodes yields ~250 candidates but the loop drains at most
+    max_drains_per_cycle — eagerly constructing 250 dataclasses cost
+    more host time than the drain kernels. Materializes per access;
+    equal iteration/indexing semantics to the eager list."""
+
+    __slots__ = ("_idx", "_rep", "_names")
+
+    def __init__(self, idx, rep, names):
+        self._idx = idx
+        self._rep = rep
+        self._names = names
+
+    def __len__(self):
+        return int(self._idx.size)
+
+    def _make(self, j: int) -> DrainCandidate:
+        rep = self._rep
+        return DrainCandidate(
+            node_index=j,
+            node_name=self._names[j] if self._names else str(j),
+            util_permille=int(rep.node_util_permille[j]),
+            drainable_pods=int(rep.node_drainable_pods[j]),
+            idle=bool(rep.node_idle[j]),
+        )
+
+    def __getitem__(self, i):
+        if isinstance(i, slice):
+            return [self._make(int(j)) for j in self._idx[i]]
+        return self._make(int(self._idx[i]))
+
+    def __iter__(self):
+        for j in self._idx:
+            yield self._make(int(j))
+
+    def __eq__(self, other):
+        try:
+            return list(self) == list(other)
+        except TypeError:
+            return NotImplemented
+
+
 class GpuEngine(SimulationEngine):
     name = "gpu"
 
@@ -1504,16 +1547,4 @@ class GpuEngine(SimulationEngine):
             & ((rep.node_idle == 1) | (rep.node_util_permille < util_threshold_permille))
         )
         order = np.argsort(rep.node_util_permille, kind="stable")
-        out = []
-        for j in order[ok[order]]:
-            j = int(j)
-            out.append(
-                DrainCandidate(
-                    node_index=j,
-                    node_name=node_names[j] if node_names else str(j),
-                    util_permille=int(rep.node_util_permille[j]),
-                    drainable_pods=int(rep.node_drainable_pods[j]),
-                    idle=bool(rep.node_idle[j]),
-                )
-            )
-        return out
+        return _CandidateList(order[ok[order]].astype(np.int64), rep, node_names)
