@@ -841,71 +841,29 @@ void pack_into(pybind11::buffer buf, const std::vector<int64_t>& offsets,
   const int64_t cap = dst.size * dst.itemsize;
   const size_t n = py::len(arrays);
   if (offsets.size() != n) throw std::runtime_error("offsets/arrays mismatch");
-  struct Spec {
-    const char* sp;
-    char* dp;
-    int64_t rows, row_stride;
-    size_t rowb, nb;
-    bool strided;
-  };
-  std::vector<Spec> specs(n);
-  size_t total = 0;
   for (size_t i = 0; i < n; ++i) {
     py::buffer src_b = py::cast<py::buffer>(arrays[i]);
     py::buffer_info src = src_b.request(false);
     int64_t nb = src.size * src.itemsize;
     if (offsets[i] + nb > cap) throw std::runtime_error("pack overflow");
-    Spec& sp = specs[i];
-    sp.sp = static_cast<const char*>(src.ptr);
-    sp.dp = base + offsets[i];
-    sp.nb = (size_t)nb;
-    sp.strided = src.ndim == 2 && src.strides[1] == src.itemsize &&
-                 src.strides[0] != src.shape[1] * src.itemsize;
-    if (sp.strided) {
-      sp.rows = src.shape[0];
-      sp.rowb = (size_t)(src.shape[1] * src.itemsize);
-      sp.row_stride = src.strides[0];
-    }
-    total += sp.nb;
-  }
-  auto run = [](const Spec& sp) {
-    if (sp.strided) {
+    const bool strided2d =
+        src.ndim == 2 && src.strides[1] == src.itemsize &&
+        src.strides[0] != src.shape[1] * src.itemsize;
+    if (strided2d) {
       // row-contiguous 2D view with a row gap (e.g. the narrowed
       // pod_sel_pairs[:, :W] slice): gather rows in one pass instead
       // of forcing an ascontiguousarray copy upstream
-      for (int64_t r = 0; r < sp.rows; ++r)
-        std::memcpy(sp.dp + (size_t)r * sp.rowb, sp.sp + r * sp.row_stride,
-                    sp.rowb);
+      const int64_t rows = src.shape[0];
+      const size_t rowb = (size_t)(src.shape[1] * src.itemsize);
+      const char* sp = static_cast<const char*>(src.ptr);
+      char* dp = base + offsets[i];
+      for (int64_t r = 0; r < rows; ++r)
+        std::memcpy(dp + (size_t)r * rowb, sp + r * src.strides[0], rowb);
     } else {
-      std::memcpy(sp.dp, sp.sp, sp.nb);
+      // contiguous (ascontiguousarray upstream)
+      std::memcpy(base + offsets[i], src.ptr, (size_t)nb);
     }
-  };
-  // single-thread for small packs; field-partitioned threads when the
-  // memcpy volume dominates (the 10k-pod staging was ~30 us serial)
-  if (total < (256u << 10) || n < 2) {
-    for (const Spec& sp : specs) run(sp);
-    return;
   }
-  py::gil_scoped_release nogil;
-  unsigned hw = std::thread::hardware_concurrency();
-  const size_t T = std::max<size_t>(1, std::min<size_t>({hw ? hw : 1, 4, n}));
-  // greedy largest-first assignment to the least-loaded thread
-  std::vector<size_t> order(n);
-  for (size_t i = 0; i < n; ++i) order[i] = i;
-  std::sort(order.begin(), order.end(),
-            [&](size_t a, size_t b) { return specs[a].nb > specs[b].nb; });
-  std::vector<std::vector<size_t>> bins(T);
-  std::vector<size_t> load(T, 0);
-  for (size_t i : order) {
-    size_t t = std::min_element(load.begin(), load.end()) - load.begin();
-    bins[t].push_back(i);
-    load[t] += specs[i].nb;
-  }
-  std::vector<std::thread> threads;
-  for (size_t t = 1; t < T; ++t)
-    threads.emplace_back([&, t]() { for (size_t i : bins[t]) run(specs[i]); });
-  for (size_t i : bins[0]) run(specs[i]);
-  for (auto& th : threads) th.join();
 }
 
 
