@@ -135,7 +135,16 @@ def main(ctx,
     logger.warning(
         "autoscaler starting: backend=%s cloud=%s sleep=%ss dry_run=%s", backend, cloud, sleep, dry_run
     )
-    cluster.loop(max_cycles=max_cycles, debug=debug)
+    from .parallel.elastic import RankLossError
+
+    try:
+        cluster.loop(max_cycles=max_cycles, debug=debug)
+    except RankLossError as e:
+        # distinct exit code so torchrun --max-restarts / a pod
+        # restartPolicy relaunches the rank set (stateless by design:
+        # the next incarnation re-lists the world)
+        logger.error("exiting for supervisor relaunch after rank loss: %s", e)
+        raise SystemExit(75)  # EX_TEMPFAIL
 
 
 def _azure_provider(resource_group, app_id, secret, tenant):
