@@ -122,6 +122,7 @@ class Cluster:
         self._cycle = 0
         self._watch_state: Optional[Dict] = None
         self._node_watch: Optional[Dict] = None
+        self._pdb_watch: Optional[Dict] = None
         # node/pool-side base-snapshot cache for the watch-resident path
         # (re-encoding 1k+ nodes in Python costs ~100 ms/cycle otherwise)
         self._base_cache: Optional[tuple] = None
@@ -168,17 +169,18 @@ class Cluster:
 
         with t.phase("ingest"):
             if self.opts.watch and _pdist.world_size() == 1:
-                # node watch: one relist, then events
-                # patch the cached node set — no per-cycle node list
-                # round trip. world>1 keeps the relist: nodes are a
-                # REPLICATED input, and per-rank event timing must not
-                # diverge the replicated tables mid-cycle.
+                # node + PDB watches: one relist each,
+                # then events patch the cached sets — ZERO per-cycle
+                # list round trips at world 1. world>1 keeps the
+                # relists: nodes/PDBs are REPLICATED inputs, and
+                # per-rank event timing must not diverge them mid-cycle.
                 nodes = self._nodes_via_watch()
+                pdbs = self._pdbs_via_watch()
             else:
                 nodes_json = self.kube.list_nodes()
                 nodes = [KubeNode(n) for n in nodes_json]
-            pdbs_json = self.kube.list_pdbs()
-            pdbs = [PodDisruptionBudget.from_json(d) for d in pdbs_json]
+                pdbs_json = self.kube.list_pdbs()
+                pdbs = [PodDisruptionBudget.from_json(d) for d in pdbs_json]
             pools = group_nodes_into_pools(
                 nodes, ignore_pools=self.opts.ignore_pools, max_size=self.opts.max_pool_size
             )
@@ -440,58 +442,72 @@ class Cluster:
         base = self._base_snapshot(nodes, pools, pdbs)
         return cache.build_snapshot(base)
 
-    def _nodes_via_watch(self) -> List[KubeNode]:
-        """Node set maintained under watch events (world-1): one full
-        relist primes the cache; ADDED/MODIFIED/DELETED events patch
-        it. Wrapped KubeNode list is rebuilt only when something
-        changed (re-wrapping 1k+ nodes costs ~ms per cycle)."""
+    def _watched_objects(self, attr: str, list_fn, watch_path: str,
+                         wrap_fn, counter: str):
+        """Generic watched object set (world-1 nodes / PDBs): one full
+        relist primes a name-keyed cache; ADDED/MODIFIED/DELETED watch
+        events patch it; the wrapped list rebuilds only on change
+        (re-wrapping 1k+ objects costs ~ms per cycle). A dropped
+        stream serves the cache once and re-primes next cycle."""
         import queue as _q
         import threading
 
-        nw = self._node_watch
-        if nw is not None and nw["broken"]:
-            logger.warning("node watch dropped; re-priming from a relist")
-            nw = self._node_watch = None
-        if nw is None:
-            nodes_json = self.kube.list_nodes()
-            cache = {n["metadata"]["name"]: n for n in nodes_json}
+        st = getattr(self, attr)
+        if st is not None and st["broken"]:
+            logger.warning("%s watch dropped; re-priming from a relist", counter)
+            st = None
+            setattr(self, attr, None)
+        if st is None:
+            objs = list_fn()
+            cache = {o["metadata"]["name"]: o for o in objs}
             q: _q.SimpleQueue = _q.SimpleQueue()
-            th = threading.Thread(target=self._node_pump, args=(q,), daemon=True)
+            th = threading.Thread(target=self._obj_pump,
+                                  args=(q, watch_path), daemon=True)
             th.start()
-            self._node_watch = nw = {
-                "q": q, "thread": th, "cache": cache, "broken": False,
-                "wrapped": None,
-            }
+            st = {"q": q, "thread": th, "cache": cache, "broken": False,
+                  "wrapped": None}
+            setattr(self, attr, st)
         changed = False
         while True:
             try:
-                ev = nw["q"].get_nowait()
+                ev = st["q"].get_nowait()
             except _q.Empty:
                 break
             if ev.get("type") == "__STREAM_END__":
-                nw["broken"] = True  # serve the cache; re-prime next cycle
+                st["broken"] = True  # serve the cache; re-prime next cycle
                 break
             obj = ev.get("object") or {}
             name = (obj.get("metadata") or {}).get("name")
             if not name:
                 continue
             if ev.get("type") == "DELETED":
-                nw["cache"].pop(name, None)
+                st["cache"].pop(name, None)
             else:
-                nw["cache"][name] = obj
+                st["cache"][name] = obj
             changed = True
         if changed:
-            self.metrics.count("node_watch_events")
-        if changed or nw["wrapped"] is None:
-            nw["wrapped"] = [KubeNode(n) for n in nw["cache"].values()]
-        return nw["wrapped"]
+            self.metrics.count(counter)
+        if changed or st["wrapped"] is None:
+            st["wrapped"] = [wrap_fn(o) for o in st["cache"].values()]
+        return st["wrapped"]
 
-    def _node_pump(self, q):
+    def _nodes_via_watch(self) -> List[KubeNode]:
+        return self._watched_objects(
+            "_node_watch", self.kube.list_nodes, "/api/v1/nodes", KubeNode,
+            "node_watch_events")
+
+    def _pdbs_via_watch(self) -> List[PodDisruptionBudget]:
+        return self._watched_objects(
+            "_pdb_watch", self.kube.list_pdbs,
+            "/apis/policy/v1/poddisruptionbudgets",
+            PodDisruptionBudget.from_json, "pdb_watch_events")
+
+    def _obj_pump(self, q, path: str):
         try:
-            for ev in self.kube.watch("/api/v1/nodes", resource_version="0"):
+            for ev in self.kube.watch(path, resource_version="0"):
                 q.put(ev)
         except Exception:
-            logger.exception("node watch stream error")
+            logger.exception("watch stream error for %s", path)
         q.put({"type": "__STREAM_END__"})
 
     def _watch_pump(self, q):

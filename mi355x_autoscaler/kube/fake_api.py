@@ -42,6 +42,7 @@ class FakeKubeState:
         self.resource_version: int = 0
         self.pod_events: List[Dict] = []  # [{rv, type, object}]
         self.node_events: List[Dict] = []
+        self.pdb_events: List[Dict] = []
 
     def load(self, pods: List[Dict], nodes: List[Dict], pdbs: Optional[List[Dict]] = None):
         with self.lock:
@@ -53,6 +54,7 @@ class FakeKubeState:
             self.pdbs = {d["metadata"]["name"]: copy.deepcopy(d) for d in (pdbs or [])}
             self.pod_events = []
             self.node_events = []
+            self.pdb_events = []
             self.resource_version += 1
             self.cond.notify_all()
 
@@ -77,6 +79,31 @@ class FakeKubeState:
             {"rv": self.resource_version, "type": etype, "object": obj}
         )
         self.cond.notify_all()
+
+    def _emit_pdb(self, etype: str, obj: Dict):
+        self.resource_version += 1
+        obj = copy.deepcopy(obj)
+        obj.setdefault("metadata", {})["resourceVersion"] = str(self.resource_version)
+        self.pdb_events.append(
+            {"rv": self.resource_version, "type": etype, "object": obj}
+        )
+        self.cond.notify_all()
+
+    def add_pdb(self, pdb: Dict):
+        with self.lock:
+            self.pdbs[pdb["metadata"]["name"]] = copy.deepcopy(pdb)
+            self._emit_pdb("ADDED", pdb)
+
+    def modify_pdb(self, pdb: Dict):
+        with self.lock:
+            self.pdbs[pdb["metadata"]["name"]] = copy.deepcopy(pdb)
+            self._emit_pdb("MODIFIED", pdb)
+
+    def remove_pdb(self, name: str):
+        with self.lock:
+            pdb = self.pdbs.pop(name, None)
+            if pdb is not None:
+                self._emit_pdb("DELETED", pdb)
 
     def add_node(self, node: Dict):
         with self.lock:
@@ -235,9 +262,17 @@ def build_app(state: FakeKubeState) -> FastAPI:
                     "metadata": {"resourceVersion": str(state.resource_version)}}
 
     @app.get("/apis/policy/v1/poddisruptionbudgets")
-    def list_pdbs():
+    def list_pdbs(watch: bool = False, resourceVersion: str = "0"):
+        if watch:
+            from starlette.responses import StreamingResponse
+
+            return StreamingResponse(
+                _event_stream(state, "pdb_events", int(resourceVersion or 0)),
+                media_type="application/json")
         with state.lock:
-            return {"kind": "PodDisruptionBudgetList", "items": list(state.pdbs.values()), "metadata": {}}
+            return {"kind": "PodDisruptionBudgetList",
+                    "items": list(state.pdbs.values()),
+                    "metadata": {"resourceVersion": str(state.resource_version)}}
 
     @app.patch("/api/v1/nodes/{name}")
     async def patch_node(name: str, request: Request):

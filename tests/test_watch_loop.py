@@ -178,17 +178,18 @@ def test_pdb_set_change_reassigns_in_place(server):
     cluster, provider = _cluster(server, no_scale=True, max_drains_per_cycle=10)
 
     # prime WITH a zero-budget PDB guarding the pod: no drain
-    server.state.pdbs = {"guard": {
+    server.state.add_pdb({
         "metadata": {"name": "guard"},
         "spec": {"selector": {"matchLabels": {"app": "guarded"}}},
-        "status": {"disruptionsAllowed": 0}}}
+        "status": {"disruptionsAllowed": 0}})
     res1 = cluster.loop_logic()
     cache1 = cluster._watch_state["cache"]
     assert nodes[1]["metadata"]["name"] not in res1.drained_nodes
 
-    # PDB set changes (removed): the pod un-guards IN PLACE — same
-    # cache object, no pod relist — and the node drains next cycle
-    server.state.pdbs = {}
+    # PDB set changes (removed, watch-visible): the pod un-guards IN
+    # PLACE — same cache object, no pod relist — and the node drains
+    server.state.remove_pdb("guard")
+    time.sleep(0.3)  # let the PDB watch deliver
     res2 = cluster.loop_logic()
     assert cluster._watch_state["cache"] is cache1  # reassigned in place
     assert nodes[1]["metadata"]["name"] in res2.drained_nodes
@@ -226,12 +227,13 @@ def test_watch_mode_native_codec_pump(server):
     assert res2.pods_pending == 11
     assert cluster._watch_state["cache"] is cache1
 
-    # PDB appears: the C++-decoded rows carry label pairs, so the
-    # reassign happens in place even for watch-added pods
-    server.state.pdbs = {"guard": {
+    # PDB appears (watch-visible): the C++-decoded rows carry label
+    # pairs, so the reassign happens in place even for watch-added pods
+    server.state.add_pdb({
         "metadata": {"name": "guard"},
         "spec": {"selector": {"matchLabels": {"app": "guarded"}}},
-        "status": {"disruptionsAllowed": 0}}}
+        "status": {"disruptionsAllowed": 0}})
+    time.sleep(0.3)  # let the PDB watch deliver
     res3 = cluster.loop_logic()
     assert cluster._watch_state["cache"] is cache1  # no re-prime
     snap = res3.snapshot
@@ -297,3 +299,37 @@ def test_bulk_injection_reaches_watchers(server):
             break
         time.sleep(0.05)
     assert res.pods_pending == 3
+
+
+def test_pdb_watch_caches_and_tracks(server):
+    """PDBs are watched too at world 1: the wrapped list is reused
+    while nothing changes (no per-cycle PDB list round trip) and a
+    MODIFIED budget reaches the drain decision via the watch."""
+    nodes = [make_node_json("agentpool1", i) for i in range(3)]
+    busy = make_pod_json("busy", cpu_milli=1500,
+                         node_name=nodes[0]["metadata"]["name"])
+    g = make_pod_json("g", cpu_milli=100,
+                      node_name=nodes[1]["metadata"]["name"],
+                      labels={"app": "db"})
+    pdb0 = {"metadata": {"name": "dbguard"},
+            "spec": {"selector": {"matchLabels": {"app": "db"}}},
+            "status": {"disruptionsAllowed": 0}}
+    server.state.load([busy, g], nodes, pdbs=[pdb0])
+    cluster, _ = _cluster(server, no_scale=True, max_drains_per_cycle=10)
+    res = cluster.loop_logic()
+    # the guarded node cannot drain while the zero-budget PDB exists
+    assert nodes[1]["metadata"]["name"] not in res.drained_nodes
+    assert cluster._pdb_watch is not None
+    wrapped1 = cluster._pdb_watch["wrapped"]
+    res = cluster.loop_logic()
+    assert cluster._pdb_watch["wrapped"] is wrapped1  # reused, no relist
+    assert nodes[1]["metadata"]["name"] not in res.drained_nodes
+
+    server.state.modify_pdb({
+        "metadata": {"name": "dbguard"},
+        "spec": {"selector": {"matchLabels": {"app": "db"}}},
+        "status": {"disruptionsAllowed": 3}})
+    time.sleep(0.3)
+    res2 = cluster.loop_logic()
+    assert cluster._pdb_watch["wrapped"] is not wrapped1  # event applied
+    assert nodes[1]["metadata"]["name"] in res2.drained_nodes
