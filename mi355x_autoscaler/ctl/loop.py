@@ -393,11 +393,28 @@ class Cluster:
                 shard_cache.set_shard(rank, world)
                 shard_cache.prime(shard_snapshot(snap, rank, world),
                                   node_index, pdb_sel)
-            q: _q.SimpleQueue = _q.SimpleQueue()
-            th = threading.Thread(target=self._watch_pump, args=(q,), daemon=True)
-            th.start()
+            q_out: _q.SimpleQueue = _q.SimpleQueue()
+            if use_codec:
+                # decode-on-arrival pipeline: the pump streams raw
+                # JSONL into q_raw; a decoder thread batch-decodes
+                # (GIL released, thread-parallel C++) while the loop
+                # does other work, tagging each batch with the
+                # node-index / PDB-selector identities it used so a
+                # concurrent remap/reassign falls back to a fresh
+                # re-decode of the carried raw lines.
+                q_raw: _q.SimpleQueue = _q.SimpleQueue()
+                th = threading.Thread(target=self._watch_pump,
+                                      args=(q_raw,), daemon=True)
+                dec = threading.Thread(target=self._decoder_pump,
+                                       args=(q_raw, q_out), daemon=True)
+                th.start()
+                dec.start()
+            else:
+                th = threading.Thread(target=self._watch_pump, args=(q_out,),
+                                      daemon=True)
+                th.start()
             self._watch_state = {
-                "cache": cache, "q": q, "thread": th,
+                "cache": cache, "q": q_out, "thread": th,
                 "node_names": node_names, "pdb_fp": pdb_fp, "broken": False,
                 "nodes": nodes, "pools": pools, "pdbs": pdbs,
                 "shard_cache": shard_cache,
@@ -409,6 +426,25 @@ class Cluster:
         ws["nodes"], ws["pools"], ws["pdbs"] = nodes, pools, pdbs
         applied = 0
         raw_lines: List[bytes] = []
+
+        def _decode_apply(lines):
+            from mi355x_autoscaler import _hip_ops
+
+            decoded = _hip_ops.decode_watch_events(
+                b"\n".join(lines) + b"\n", cache.node_index,
+                list(cache.pdb_selectors),
+            )
+            n = cache.apply_batch(decoded)
+            if shard_cache is not None:
+                shard_cache.apply_batch(decoded)
+            return n
+
+        def _flush_raw():
+            nonlocal applied
+            if raw_lines:
+                applied += _decode_apply(raw_lines)
+                raw_lines.clear()
+
         while True:
             try:
                 ev = ws["q"].get_nowait()
@@ -417,6 +453,20 @@ class Cluster:
             if isinstance(ev, bytes):
                 raw_lines.append(ev)
                 continue
+            if isinstance(ev, tuple):  # pre-decoded by the decoder pump
+                _flush_raw()  # preserve event order
+                _, ni_id, sel_id, decoded, batch_raw = ev
+                if (ni_id == id(cache.node_index)
+                        and sel_id == id(cache.pdb_selectors)):
+                    applied += cache.apply_batch(decoded)
+                    if shard_cache is not None:
+                        shard_cache.apply_batch(decoded)
+                else:
+                    # node remap / PDB reassign landed after the pump
+                    # decoded: its node/pdb ids are stale — re-decode
+                    # the carried raw lines with the fresh mappings
+                    applied += _decode_apply(batch_raw)
+                continue
             if ev.get("type") == "__STREAM_END__":
                 ws["broken"] = True
                 break
@@ -424,17 +474,7 @@ class Cluster:
             if shard_cache is not None:
                 shard_cache.apply_event(ev)  # self-filters by uid hash
             applied += 1
-        if raw_lines:
-            # C++ batch decode of the whole cycle's events at once
-            from mi355x_autoscaler import _hip_ops
-
-            decoded = _hip_ops.decode_watch_events(
-                b"\n".join(raw_lines) + b"\n", cache.node_index,
-                list(cache.pdb_selectors),
-            )
-            applied += cache.apply_batch(decoded)
-            if shard_cache is not None:
-                shard_cache.apply_batch(decoded)
+        _flush_raw()
         cache.maybe_compact()
         if shard_cache is not None:
             shard_cache.maybe_compact()
@@ -509,6 +549,52 @@ class Cluster:
         except Exception:
             logger.exception("watch stream error for %s", path)
         q.put({"type": "__STREAM_END__"})
+
+    def _decoder_pump(self, q_raw, q_out):
+        """Pipeline stage between the raw watch pump and the loop:
+        drains whatever raw lines have arrived, batch-decodes them
+        with the C++ decoder (GIL released) and forwards
+        ('decoded', node_index_id, pdb_sel_id, decoded, raw_batch)
+        tuples; non-bytes items (sentinels) pass through in order."""
+        import queue as _q
+
+        from mi355x_autoscaler import _hip_ops
+
+        while True:
+            item = q_raw.get()
+            batch: List[bytes] = []
+            passthrough = None
+            while True:
+                if isinstance(item, bytes):
+                    batch.append(item)
+                else:
+                    passthrough = item
+                    break
+                try:
+                    item = q_raw.get_nowait()
+                except _q.Empty:
+                    break
+            if batch:
+                ws = self._watch_state
+                cache = ws["cache"] if ws else None
+                if cache is None:
+                    for line in batch:
+                        q_out.put(line)
+                else:
+                    ni = cache.node_index
+                    sel = cache.pdb_selectors
+                    try:
+                        decoded = _hip_ops.decode_watch_events(
+                            b"\n".join(batch) + b"\n", ni, list(sel))
+                        q_out.put(("decoded", id(ni), id(sel), decoded, batch))
+                    except Exception:
+                        logger.exception("decoder pump failed; raw fallback")
+                        for line in batch:
+                            q_out.put(line)
+            if passthrough is not None:
+                q_out.put(passthrough)
+                if passthrough.get("type") == "__STREAM_END__":
+                    return
 
     def _watch_pump(self, q):
         """Background thread: stream pod watch events into the queue —
