@@ -333,3 +333,56 @@ def test_pdb_watch_caches_and_tracks(server):
     res2 = cluster.loop_logic()
     assert cluster._pdb_watch["wrapped"] is not wrapped1  # event applied
     assert nodes[1]["metadata"]["name"] in res2.drained_nodes
+
+
+def test_decoder_pipeline_stale_batch_redecodes(server):
+    """A pre-decoded batch whose node-index/PDB identities no longer
+    match the cache (a remap/reassign landed in between) must be
+    re-decoded from the carried raw lines, not applied stale."""
+    import json as _json
+
+    from mi355x_autoscaler.snapshot.codec import codec_available
+
+    if not codec_available():
+        pytest.skip("native codec not built")
+    import mi355x_autoscaler._hip_ops as _hip_ops
+
+    pods, nodes = burst_trace(4, n_nodes_per_pool=(3,), seed=6)
+    server.state.load(pods, nodes)
+    kube = KubeApiClient(server.base_url)
+    provider = FakeCloudProvider(initial_pool_sizes={"agentpool1": 3},
+                                 kube_state=server.state)
+    opts = ClusterOptions(sleep_seconds=0, idle_threshold_seconds=10**9,
+                          spare_agents=0, watch=True, native_codec=True,
+                          no_scale=True, no_maintenance=True)
+    cluster = Cluster(kube, provider, CpuEngine(), options=opts)
+    cluster.loop_logic()  # prime
+    ws = cluster._watch_state
+    cache = ws["cache"]
+
+    # a pod BOUND to node index 2 under the CURRENT index
+    target = sorted(server.state.nodes)[2]
+    pod = make_pod_json("bound-late", cpu_milli=100, node_name=target)
+    raw = [_json.dumps({"type": "ADDED", "object": pod}).encode()]
+    stale_index = dict(cache.node_index)  # decoded against THIS map
+    decoded = _hip_ops.decode_watch_events(
+        b"\n".join(raw) + b"\n", stale_index, list(cache.pdb_selectors))
+    # remap AFTER the decode: drop node 0, shifting the target 2 -> 1
+    new_index = {n: i for i, n in
+                 enumerate(nm for nm in sorted(cache.node_index) if
+                           nm != sorted(cache.node_index)[0])}
+    cache.remap_nodes(new_index)
+    ws["node_names"] = tuple(sorted(new_index, key=new_index.get))
+    # feed the now-STALE tuple as the decoder pump would have
+    ws["q"].put(("decoded", id(stale_index), id(cache.pdb_selectors),
+                 decoded, raw))
+    server.state.remove_node(sorted(server.state.nodes)[0])
+    import time as _t
+
+    _t.sleep(0.3)  # node watch delivers the matching node set
+    res = cluster.loop_logic()
+    snap = res.snapshot
+    rows = {u: i for i, u in enumerate(snap.pod_uids)}
+    r = rows["pod-default-bound-late"]
+    # correct binding under the NEW index (re-decoded), not the stale 2
+    assert snap.node_names[int(snap.pod_node[r])] == target
