@@ -5,10 +5,14 @@ one namespace, drains, watch deltas) can skew shard sizes, and the
 controller's per-cycle cost is max-over-ranks. This module measures
 the skew with one tiny all-gather and, past a threshold, exchanges
 surplus pod SoA rows with `all_to_all_single` — the bandwidth-bound
-collective of the design: on MI355X it runs over RCCL/xGMI (7 p2p
-links per GPU), on the CPU test rig over gloo. Row payloads are the
-packed 27-int64 pod columns, so a 1M-pod rebalance moves ~200 MB
-chunked by the collective itself.
+collective of the design: on MI355X it runs over RCCL/xGMI, on the
+CPU test rig over gloo. All-to-all is PAIRWISE p2p, so a single call
+already drives every xGMI link concurrently (each rank streams
+distinct slices to up to 7 peers at ~153 GB/s each) — unlike ring
+collectives, no multi-ring chunking is needed for link engagement.
+Row payloads are the packed int64 pod columns (~200 MB for a 1M-pod
+rebalance); overlapping the host pack/unpack with the exchange is the
+remaining (unmeasured, multi-GPU-gated) tuning headroom.
 
 The decision layer never depends on WHICH rank owns a pod (segment
 counts are all-reduced), so rebalancing is pure load-balancing — the
