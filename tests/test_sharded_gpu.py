@@ -195,3 +195,92 @@ def test_rccl_world1_collective_paths():
     assert res["u3"] == ref.units_needed
     assert res["pe"] == ref.pods_placed_existing
     assert res["un"] == ref.pods_unfulfillable
+
+
+def _loop_server(q_url, q_go, q_done):
+    from mi355x_autoscaler.kube.fake_api import FakeKubeServer
+
+    pods_json, nodes_json = _trace()
+    with FakeKubeServer() as srv:
+        srv.state.load(pods_json, nodes_json)
+        q_url.put(srv.base_url)
+        q_go.get(timeout=180)  # both ranks primed
+        for i in range(40):
+            srv.state.add_pod(make_pod_json(f"wave{i}", cpu_milli=400,
+                                            pending=True))
+        q_done.get(timeout=300)
+
+
+def _loop_rank(rank, url, q_out, q_prime):
+    os.environ.update(WORLD_SIZE=str(WORLD), RANK=str(rank),
+                      LOCAL_RANK="0", MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT="29735")
+    import time
+
+    import torch.distributed as dist
+
+    from mi355x_autoscaler.cloud.fake import FakeCloudProvider
+    from mi355x_autoscaler.ctl.loop import Cluster, ClusterOptions
+    from mi355x_autoscaler.kube.client import KubeApiClient
+    from mi355x_autoscaler.ops.engine_gpu import GpuEngine
+    from mi355x_autoscaler.parallel.engine_sharded import ShardedEngine
+
+    dist.init_process_group(backend="gloo")  # one shared GPU: no RCCL
+    try:
+        kube = KubeApiClient(url)
+        provider = FakeCloudProvider(initial_pool_sizes={"agentpool1": 20})
+        opts = ClusterOptions(sleep_seconds=0, idle_threshold_seconds=10**9,
+                              spare_agents=0, watch=True)
+        cluster = Cluster(kube, provider,
+                          ShardedEngine(GpuEngine("cuda:0")), options=opts)
+        res1 = cluster.loop_logic()  # primes watch + shard caches
+        assert cluster._watch_state.get("shard_cache") is not None
+        q_prime.put(rank)
+        dist.barrier()
+        time.sleep(1.0)  # churn wave arrives via watch
+        res2 = cluster.loop_logic()  # RESIDENT sharded decide on GPU
+        q_out.put((rank, pickle.dumps(dict(
+            u1=res1.decision.units_needed, u2=res2.decision.units_needed,
+            p2=res2.decision.pods_placed_existing,
+            total2=res2.pods_total,
+            events=[e.kind for e in provider.events]))))
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_sharded_gpu_watch_loop_world2():
+    """Capstone: the FULL watch-mode control loop at world 2 with real
+    GPU engines on hardware — prime, churn via watch, per-rank
+    resident shard decide, identical decisions, rank-0-only
+    actuation. gloo carries the collectives (one shared GPU); the
+    8-GPU driver run swaps in RCCL by backend only."""
+    ctx = mp.get_context("spawn")
+    q_url, q_go, q_done, q_out, q_prime = (ctx.Queue() for _ in range(5))
+    server = ctx.Process(target=_loop_server, args=(q_url, q_go, q_done))
+    server.start()
+    url = q_url.get(timeout=120)
+    ranks = [ctx.Process(target=_loop_rank, args=(r, url, q_out, q_prime))
+             for r in range(WORLD)]
+    for p in ranks:
+        p.start()
+    results = {}
+    try:
+        for _ in range(WORLD):
+            q_prime.get(timeout=240)
+        q_go.put(True)
+        for _ in range(WORLD):
+            rank, blob = q_out.get(timeout=240)
+            results[rank] = pickle.loads(blob)
+    finally:
+        q_done.put(True)
+        for p in ranks:
+            p.join(timeout=60)
+        server.join(timeout=60)
+    assert all(p.exitcode == 0 for p in ranks), [p.exitcode for p in ranks]
+    assert results[0]["u1"] == results[1]["u1"]
+    assert results[0]["u2"] == results[1]["u2"]
+    assert results[0]["p2"] == results[1]["p2"]
+    assert results[0]["total2"] == 4000 + 40  # churn reached both ranks
+    assert results[0]["events"], "rank 0 must actuate"
+    assert results[1]["events"] == []
