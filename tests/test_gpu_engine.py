@@ -184,7 +184,12 @@ def test_gpu_pipeline_scales_10k(engine):
         gpu_fraction=0.35, seed=7, bound_pods_per_node=2,
     )
     snap = _snap(pods_json, nodes_json)
+    from mi355x_autoscaler.ops.engine_gpu import FALLBACK_COUNTS
+
+    before = dict(FALLBACK_COUNTS)
     _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))
+    # the headline shape must run the HIP fast path end to end
+    assert dict(FALLBACK_COUNTS) == before
 
 
 def test_segment_hash_path_equals_torch_path(engine):
