@@ -45,7 +45,8 @@ import json as _json
 
 import mi355x_autoscaler._hip_ops as _ops
 
-for cycle in range(15):
+CYCLES = int(__import__("os").environ.get("CYCLES", "15"))
+for cycle in range(CYCLES):
     pods = [make_pod_json(f"churn-{cycle}-{e}", cpu_milli=rng.choice([100, 500]),
                           mem_mib=rng.choice([128, 1024]), pending=True)
             for e in range(EVENTS)]
