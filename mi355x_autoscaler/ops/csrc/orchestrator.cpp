@@ -65,7 +65,14 @@ bool mega_decide_launch(
     torch::Tensor seg_first, torch::Tensor slot_of_pod, torch::Tensor free,
     torch::Tensor elig, torch::Tensor seg_pool, torch::Tensor run_scratch,
     torch::Tensor blob, torch::Tensor pod_pool, int64_t hash_slots,
-    int64_t seg_max, int64_t want_pod_pool);
+    int64_t seg_max, int64_t want_pod_pool, uint64_t* demand_out);
+
+// set by run_decide_plan_auto for the next fused launch: the mega
+// kernel computes the per-pool demand segsum in-kernel (idle blocks
+// during the serial tail) instead of a side-stream launch. Cleared by
+// fused_decide_launch; g_mega_demand_used reports consumption.
+static uint64_t* g_mega_demand = nullptr;
+static bool g_mega_demand_used = false;
 
 void fused_classic_phase(
     torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
@@ -119,6 +126,9 @@ void fused_decide_launch(
     const char* mp = getenv("MI355X_AS_MEGA_MAX_P");
     mega_max_p = mp ? atoll(mp) : 131072;
   }
+  uint64_t* demand = g_mega_demand;
+  g_mega_demand = nullptr;
+  g_mega_demand_used = false;
   if (mega_env && N > 0 && K > 0 && P > 0 && P <= mega_max_p && N <= 65536) {
     if (mega_decide_launch(pod_req, pod_flags, pod_sel_hash, pod_sel_pairs,
                            pod_sel_count, pod_node, node_alloc, node_labels,
@@ -128,7 +138,8 @@ void fused_decide_launch(
                            seg_sel_count, seg_sel_hash, seg_req, seg_count,
                            seg_key, seg_first, slot_of_pod, free, elig,
                            seg_pool, run_scratch, blob, pod_pool, hash_slots,
-                           seg_max, want_pod_pool)) {
+                           seg_max, want_pod_pool, demand)) {
+      g_mega_demand_used = (demand != nullptr && want_pod_pool);
       return;
     }
   }
@@ -463,16 +474,28 @@ int64_t run_decide_plan_auto(const std::shared_ptr<DecidePlan>& p,
   // order THIS pipeline (which rewrites pod_pool / pod columns) after
   // the previous decide's segsum — stream-level, normally long past
   hipStreamWaitEvent(main_stream, g_join_ev, 0);
-  if (split) {
-    run_decide_plan_phase(p, 1);  // classic head: through pod_pool_map
-  } else {
-    run_decide_plan(p);           // mega or full classic
-  }
-  hipEventRecord(g_fork_ev, main_stream);
-  hipStreamWaitEvent(g_side_stream, g_fork_ev, 0);
   const int use_b = p->flip;
   p->flip ^= 1;
   torch::Tensor out = use_b ? p->demand_b : p->demand_a;
+  if (split) {
+    run_decide_plan_phase(p, 1);  // classic head: through pod_pool_map
+  } else {
+    // mega path: hand the demand buffer to the kernel — its idle
+    // blocks compute the segsum during the serial tail (no side
+    // stream, no extra launch). Classic fallback leaves it unused.
+    hipMemsetAsync(out.data_ptr<int64_t>(), 0,
+                   (size_t)out.numel() * sizeof(int64_t), main_stream);
+    g_mega_demand = reinterpret_cast<uint64_t*>(out.data_ptr<int64_t>());
+    run_decide_plan(p);           // mega or full classic
+    if (g_mega_demand_used) {
+      // demand produced on the MAIN stream; keep the join-event
+      // contract for pool_demand()/resident-scatter ordering
+      hipEventRecord(g_join_ev, main_stream);
+      return use_b;
+    }
+  }
+  hipEventRecord(g_fork_ev, main_stream);
+  hipStreamWaitEvent(g_side_stream, g_fork_ev, 0);
   hipMemsetAsync(out.data_ptr<int64_t>(), 0,
                  (size_t)out.numel() * sizeof(int64_t), g_side_stream);
   segsum_mfma_on(t[0], t[32], p->wi_win, p->wi_start, p->wi_end, p->K, out,

@@ -58,7 +58,7 @@ extern "C" __global__ __launch_bounds__(MEGA_BLOCK) void mega_decide_kernel(
     int32_t* __restrict__ seg_pool, int64_t* __restrict__ run_scratch,
     int64_t* __restrict__ blob, int32_t* __restrict__ pod_pool,
     int64_t P, int32_t N, int32_t K, int32_t H, int32_t SM, int64_t nblob,
-    int32_t RTOT, int32_t want_pod_pool)
+    int32_t RTOT, int32_t want_pod_pool, uint64_t* __restrict__ demand_out)
 {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   cg::grid_group grid = cg::this_grid();
@@ -129,8 +129,24 @@ extern "C" __global__ __launch_bounds__(MEGA_BLOCK) void mega_decide_kernel(
   }
   grid.sync();
 
-  // P4: block 0 runs the serial tail (place -> units -> epilogue);
-  // the other blocks concurrently build the pod -> pool map
+  // P4: block 0 runs the serial tail (place -> units -> epilogue).
+  // With demand_out set (round 2), ALL blocks first build the
+  // pod -> pool map, grid.sync, and the non-zero blocks then compute
+  // the MFMA per-pool demand segsum DURING the single-workgroup tail
+  // (they were idle after their grid-strided map share); no separate
+  // segsum launch or side stream is needed on the mega path.
+  if (want_pod_pool && demand_out != nullptr) {
+    for (int64_t i = gtid; i < P; i += gsz) {
+      const int s = slot_of_pod[i];
+      pod_pool[i] = (s >= 0) ? seg_pool[slot2seg[s]] : -1;
+    }
+    grid.sync();
+    if (blockIdx.x > 0 && gridDim.x > 1) {
+      dev_demand_segsum_fastpath(
+          pod_req, pod_pool, P, K, demand_out, smem,
+          (int64_t)(blockIdx.x - 1) * 4, (int64_t)(gridDim.x - 1));
+    }
+  }
   if (blockIdx.x == 0) {
     __shared__ int64_t lds_scan[MEGA_BLOCK / WAVE + 1];
     dev_binpack<MEGA_BLOCK>(free_, elig, seg_req, seg_count, N, G, blob,
@@ -149,7 +165,7 @@ extern "C" __global__ __launch_bounds__(MEGA_BLOCK) void mega_decide_kernel(
               seg_sel_hash_ws, stat,
               /*blob_mirror=*/blob + (nblob - 2 - SM),
               /*blob_stat=*/reinterpret_cast<int32_t*>(blob + (nblob - 2)), SM);
-  } else if (want_pod_pool) {
+  } else if (want_pod_pool && demand_out == nullptr) {
     for (int64_t i = (int64_t)(blockIdx.x - 1) * blockDim.x + tid; i < P;
          i += (int64_t)(gridDim.x - 1) * blockDim.x) {
       const int s = slot_of_pod[i];
@@ -177,7 +193,7 @@ bool mega_decide_launch(
     torch::Tensor seg_first, torch::Tensor slot_of_pod, torch::Tensor free,
     torch::Tensor elig, torch::Tensor seg_pool, torch::Tensor run_scratch,
     torch::Tensor blob, torch::Tensor pod_pool, int64_t hash_slots,
-    int64_t seg_max, int64_t want_pod_pool) {
+    int64_t seg_max, int64_t want_pod_pool, uint64_t* demand_out) {
   static int coop_ok = -1;
   static int max_coop_blocks = 0;
   if (coop_ok < 0) {
@@ -250,6 +266,7 @@ bool mega_decide_launch(
   int64_t* p_scratch = run_scratch.data_ptr<int64_t>();
   int64_t* p_blob = blob.data_ptr<int64_t>();
   int32_t* p_ppool = wpp ? pod_pool.data_ptr<int32_t>() : nullptr;
+  uint64_t* p_demand = wpp ? demand_out : nullptr;
 
   void* args[] = {
       (void*)&p_pod_req, (void*)&p_pod_flags, (void*)&p_sel_hash,
@@ -262,7 +279,7 @@ bool mega_decide_launch(
       (void*)&p_sop, (void*)&p_free, (void*)&p_elig, (void*)&p_spool,
       (void*)&p_scratch, (void*)&p_blob, (void*)&p_ppool,
       (void*)&P, (void*)&N, (void*)&K, (void*)&H, (void*)&SM, (void*)&nblob,
-      (void*)&RTOT, (void*)&wpp,
+      (void*)&RTOT, (void*)&wpp, (void*)&p_demand,
   };
   hipError_t e = hipLaunchCooperativeKernel(
       reinterpret_cast<const void*>(mega_decide_kernel), dim3(grid),

@@ -669,6 +669,119 @@ __device__ inline void dev_binpack(
   }
 }
 
+// Single-window MFMA demand segsum as a DEVICE function (round 2):
+// the mega kernel's non-zero blocks run it over pod_pool during the
+// single-workgroup FFD tail (which otherwise leaves them idle), so
+// the per-pool demand needs no separate launch or side stream on the
+// mega path. Same exact-f32 limb math as segsum_mfma.hip (chunks of
+// 256, partials < 2^24); S <= 16 (one window); out accumulated with
+// device atomics (commutative integer adds — deterministic).
+typedef float bp_f32x4 __attribute__((ext_vector_type(4)));
+
+__device__ inline void dev_demand_segsum_fastpath(
+    const int32_t* __restrict__ values,  // [M, NRES]
+    const int32_t* __restrict__ seg_of,  // [M] in [-1, S)
+    int64_t M, int32_t S,
+    uint64_t* __restrict__ out,          // [S, NRES] zeroed by caller
+    char* smem_base,                     // >= 4*(256*NRES+256)*4 + 16*NRES*8
+    int64_t chunk_start, int64_t chunk_stride) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wv = tid / WAVE;  // 4 waves per 256-thread block
+  int32_t* l_vals = reinterpret_cast<int32_t*>(smem_base) + wv * 256 * NRES;
+  int32_t* l_segs = reinterpret_cast<int32_t*>(smem_base) + 4 * 256 * NRES +
+                    wv * 256;
+  unsigned long long* s_acc = reinterpret_cast<unsigned long long*>(
+      smem_base + 4 * (256 * NRES + 256) * sizeof(int32_t));
+
+  for (int idx = tid; idx < 16 * NRES; idx += blockDim.x) s_acc[idx] = 0ull;
+  __syncthreads();
+
+  const int col = lane & 15;
+  const int kidx = lane >> 4;
+  const int row = col;
+  const int res = row & 3;
+  const int limb = (row >> 2) & 1;
+  const bool live_row = row < 8;
+  const int64_t T = (M + 255) >> 8;
+
+  long long p64[4] = {0, 0, 0, 0};
+  for (int64_t t = chunk_start + wv; t < T; t += chunk_stride * 4) {
+    const int64_t p_begin = t << 8;
+    const int64_t p_end = p_begin + 256 < M ? p_begin + 256 : M;
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int pl = lane + 64 * q;
+      const int64_t pp = p_begin + pl;
+      if (pp < p_end) {
+        const int4 v = reinterpret_cast<const int4*>(values)[pp];
+        reinterpret_cast<int4*>(l_vals)[pl] = v;
+        l_segs[pl] = seg_of[pp];
+      } else {
+        reinterpret_cast<int4*>(l_vals)[pl] = int4{0, 0, 0, 0};
+        l_segs[pl] = -1;
+      }
+    }
+    __builtin_amdgcn_wave_barrier();
+    const int steps = (int)((p_end - p_begin + 3) >> 2);
+    bp_f32x4 acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
+                       {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+    int sstep = 0;
+    for (; sstep + 4 <= steps; sstep += 4) {
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        const int pl = (sstep + u) * 4 + kidx;
+        const float b = (l_segs[pl] == col) ? 1.f : 0.f;
+        float a = 0.f;
+        if (live_row) {
+          const uint32_t v = (uint32_t)l_vals[pl * NRES + res];
+          a = (float)(limb ? (v >> 16) : (v & 0xffffu));
+        }
+        acc[u] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[u], 0, 0, 0);
+      }
+    }
+    for (; sstep < steps; ++sstep) {
+      const int pl = sstep * 4 + kidx;
+      const float b = (l_segs[pl] == col) ? 1.f : 0.f;
+      float a = 0.f;
+      if (live_row) {
+        const uint32_t v = (uint32_t)l_vals[pl * NRES + res];
+        a = (float)(limb ? (v >> 16) : (v & 0xffffu));
+      }
+      acc[0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[0], 0, 0, 0);
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) p64[reg] += (long long)acc[u][reg];
+    }
+    __builtin_amdgcn_wave_barrier();
+  }
+
+  // flush window partials into block LDS, then one global add set
+  const int s = col;
+  if (s < S) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int orow = (lane >> 4) * 4 + reg;
+      if (orow < 8) {
+        const int ores = orow & 3;
+        const int olimb = (orow >> 2) & 1;
+        const unsigned long long partial = (unsigned long long)p64[reg];
+        if (partial) {
+          atomicAdd(&s_acc[s * NRES + ores],
+                    olimb ? (partial << 16) : partial);
+        }
+      }
+    }
+  }
+  __syncthreads();
+  for (int idx = tid; idx < S * NRES; idx += blockDim.x) {
+    const unsigned long long v = s_acc[idx];
+    if (v) atomicAdd(reinterpret_cast<unsigned long long*>(out) + idx, v);
+  }
+}
+
 // Units phase (kernel B'): run-length FFD into virtual new nodes,
 // pools wave-parallel, inline max-new, optional fused readback
 // epilogue. Block size must be UNITS_BLOCK (== 256). `lds_arena`

@@ -661,3 +661,33 @@ def test_units_interleaved_pools_long_lists(engine):
             node_selector={"agentpool": p}))
     snap = _snap(pods, nodes_json, max_size=4096)
     _assert_decisions_equal(engine.decide(snap), CpuEngine().decide(snap))
+
+
+def test_pool_demand_matches_numpy_both_paths(engine):
+    """pool_demand() — the MFMA demand segsum — must equal the numpy
+    per-pool pending request totals on BOTH execution paths: the
+    in-mega idle-block segsum (small P) and the classic side-stream
+    fork (P above the mega threshold)."""
+    for n_pods in (8000, 150_000):  # mega path / classic split path
+        pods_json, nodes_json = burst_trace(
+            n_pods, n_nodes_per_pool=(20, 10),
+            pools=(("agentpool1", "Standard_D2_v2"),
+                   ("gpupool", "Standard_ND96_MI355X_v1")),
+            gpu_fraction=0.4, seed=3, bound_pods_per_node=2)
+        snap = _snap(pods_json, nodes_json)
+        engine.decide(snap)
+        demand = engine.pool_demand()
+        assert demand is not None
+        got = demand.cpu().numpy()
+        assert got.shape == (snap.num_pools, 4)
+
+        # CPU reference: each segment routes to its first candidate
+        # pool (seg_pool_route semantics); sum its pods' requests there.
+        sim = OracleSimulator(snap)
+        ref = np.zeros((snap.num_pools, 4), np.int64)
+        for seg in build_segments(snap):
+            cands = sim.pool_candidates(seg)
+            if cands:
+                ref[cands[0]] += snap.pod_req[seg.pod_indices].astype(
+                    np.int64).sum(axis=0)
+        np.testing.assert_array_equal(got, ref)
