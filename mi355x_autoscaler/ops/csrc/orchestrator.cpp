@@ -66,6 +66,21 @@ bool mega_decide_launch(
     torch::Tensor elig, torch::Tensor seg_pool, torch::Tensor run_scratch,
     torch::Tensor blob, torch::Tensor pod_pool, int64_t hash_slots,
     int64_t seg_max, int64_t want_pod_pool, uint64_t* demand_out);
+bool mega_decide_single_launch(
+    torch::Tensor pod_req, torch::Tensor pod_flags, torch::Tensor pod_sel_hash,
+    torch::Tensor pod_sel_pairs, torch::Tensor pod_sel_count,
+    torch::Tensor pod_node, torch::Tensor node_alloc, torch::Tensor node_labels,
+    torch::Tensor node_label_count, torch::Tensor node_unsched,
+    torch::Tensor pool_unit, torch::Tensor pool_labels,
+    torch::Tensor pool_label_count, torch::Tensor pool_max,
+    torch::Tensor pool_cur, torch::Tensor ws_i32, torch::Tensor slot_ks,
+    torch::Tensor slot_kr, torch::Tensor slot2seg, torch::Tensor seg_sel_pairs,
+    torch::Tensor seg_sel_count, torch::Tensor seg_sel_hash,
+    torch::Tensor seg_req, torch::Tensor seg_count, torch::Tensor seg_key,
+    torch::Tensor seg_first, torch::Tensor slot_of_pod, torch::Tensor free,
+    torch::Tensor elig, torch::Tensor seg_pool, torch::Tensor run_scratch,
+    torch::Tensor blob, torch::Tensor pod_pool, int64_t hash_slots,
+    int64_t seg_max, int64_t want_pod_pool, uint64_t* demand_out);
 
 // set by run_decide_plan_auto for the next fused launch: the mega
 // kernel computes the per-pool demand segsum in-kernel (idle blocks
@@ -121,14 +136,34 @@ void fused_decide_launch(
   // launch is unavailable. MI355X_AS_NO_MEGA=1 forces classic.
   static int mega_env = -1;
   static int64_t mega_max_p = 0;
+  static int64_t single_max_p = 0;
   if (mega_env < 0) {
     mega_env = getenv("MI355X_AS_NO_MEGA") ? 0 : 1;
     const char* mp = getenv("MI355X_AS_MEGA_MAX_P");
     mega_max_p = mp ? atoll(mp) : 131072;
+    // single-workgroup regime: regular launch, __syncthreads-only
+    // barriers. 0 disables. Crossover measured on MI355X (see
+    // profiles/): default covers config #2 (1k pods, 8 nodes).
+    const char* sp = getenv("MI355X_AS_SINGLE_MAX_P");
+    single_max_p = sp ? atoll(sp) : 2048;
   }
   uint64_t* demand = g_mega_demand;
   g_mega_demand = nullptr;
   g_mega_demand_used = false;
+  if (mega_env && single_max_p > 0 && N > 0 && K > 0 && P > 0 &&
+      P <= single_max_p && N <= 2048) {
+    if (mega_decide_single_launch(
+            pod_req, pod_flags, pod_sel_hash, pod_sel_pairs, pod_sel_count,
+            pod_node, node_alloc, node_labels, node_label_count, node_unsched,
+            pool_unit, pool_labels, pool_label_count, pool_max, pool_cur,
+            ws_i32, slot_ks, slot_kr, slot2seg, seg_sel_pairs, seg_sel_count,
+            seg_sel_hash, seg_req, seg_count, seg_key, seg_first, slot_of_pod,
+            free, elig, seg_pool, run_scratch, blob, pod_pool, hash_slots,
+            seg_max, want_pod_pool, demand)) {
+      g_mega_demand_used = (demand != nullptr && want_pod_pool);
+      return;
+    }
+  }
   if (mega_env && N > 0 && K > 0 && P > 0 && P <= mega_max_p && N <= 65536) {
     if (mega_decide_launch(pod_req, pod_flags, pod_sel_hash, pod_sel_pairs,
                            pod_sel_count, pod_node, node_alloc, node_labels,

@@ -34,7 +34,23 @@ namespace cg = cooperative_groups;
 #define MEGA_BLOCK 256
 #define MEGA_SMEM (FIT_TILE * MAXLAB * 8 + FIT_TILE * 4)  // 66560 B
 
-extern "C" __global__ __launch_bounds__(MEGA_BLOCK) void mega_decide_kernel(
+// SINGLE=true compiles the same pipeline as ONE regular (non-
+// cooperative) workgroup: every grid.sync() degenerates to
+// __syncthreads() and the launch skips hipLaunchCooperativeKernel's
+// host-side cost. At small P (config #2: 1k pods, 8 nodes) the
+// cooperative grid spends more time in its 4 grid-wide barriers and
+// launch setup than in work; one block owns every phase instead.
+template <bool SINGLE> struct mega_sync_t;
+template <> struct mega_sync_t<false> {
+  cg::grid_group g = cg::this_grid();
+  __device__ void operator()() { g.sync(); }
+};
+template <> struct mega_sync_t<true> {
+  __device__ void operator()() { __syncthreads(); }
+};
+
+template <bool SINGLE>
+__global__ __launch_bounds__(MEGA_BLOCK) void mega_decide_kernel_t(
     const int32_t* __restrict__ pod_req, const int32_t* __restrict__ pod_flags,
     const int64_t* __restrict__ pod_sel_hash,
     const int64_t* __restrict__ pod_sel_pairs,
@@ -61,7 +77,7 @@ extern "C" __global__ __launch_bounds__(MEGA_BLOCK) void mega_decide_kernel(
     int32_t RTOT, int32_t want_pod_pool, uint64_t* __restrict__ demand_out)
 {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  cg::grid_group grid = cg::this_grid();
+  mega_sync_t<SINGLE> gsync;
   const int tid = threadIdx.x;
   const int64_t gsz = (int64_t)gridDim.x * blockDim.x;
   const int64_t gtid = (int64_t)blockIdx.x * blockDim.x + tid;
@@ -74,7 +90,7 @@ extern "C" __global__ __launch_bounds__(MEGA_BLOCK) void mega_decide_kernel(
   // P0: workspace zero + free init
   for (int64_t i = gtid; i < 3 * (int64_t)H + 3; i += gsz) ws_i32[i] = 0;
   for (int64_t i = gtid; i < (int64_t)N * NRES; i += gsz) free_[i] = node_alloc[i];
-  grid.sync();
+  gsync();
 
   // P1: hash-aggregate + usage subtraction. The hash serializes on
   // per-key atomics, so only ~1 block per 1024 pods participates
@@ -93,7 +109,7 @@ extern "C" __global__ __launch_bounds__(MEGA_BLOCK) void mega_decide_kernel(
                               (int64_t)hb * blockDim.x);
     }
     dev_node_usage_subtract(pod_node, pod_req, P, free_);
-    grid.sync();
+    gsync();
   }
 
   // P2: compact + sort (block 0; LDS carve: 40 KiB)
@@ -108,7 +124,7 @@ extern "C" __global__ __launch_bounds__(MEGA_BLOCK) void mega_decide_kernel(
                      seg_key, seg_first, slot2seg, stat, c_ks, c_kr, c_slot,
                      c_n);
   }
-  grid.sync();
+  gsync();
   const int G = stat[0];
 
   // P3: eligibility tiles (LDS carve: 65 KiB) + routing on the last block
@@ -127,7 +143,7 @@ extern "C" __global__ __launch_bounds__(MEGA_BLOCK) void mega_decide_kernel(
       dev_route_one(g, K, seg_sel_pairs, seg_sel_count, seg_req, pool_labels,
                     pool_label_count, pool_unit, seg_pool);
   }
-  grid.sync();
+  gsync();
 
   // P4: block 0 runs the serial tail (place -> units -> epilogue).
   // With demand_out set (round 2), ALL blocks first build the
@@ -140,11 +156,28 @@ extern "C" __global__ __launch_bounds__(MEGA_BLOCK) void mega_decide_kernel(
       const int s = slot_of_pod[i];
       pod_pool[i] = (s >= 0) ? seg_pool[slot2seg[s]] : -1;
     }
-    grid.sync();
-    if (blockIdx.x > 0 && gridDim.x > 1) {
+    gsync();
+    if constexpr (SINGLE) {
+      // one block: demand first (tiny at small P), then the tail —
+      // the segsum's LDS carve is reused by units right after
+      dev_demand_segsum_fastpath(pod_req, pod_pool, P, K, demand_out, smem,
+                                 /*chunk_start=*/0, /*chunk_stride=*/1);
+      __syncthreads();
+    } else if (blockIdx.x > 0 && gridDim.x > 1) {
       dev_demand_segsum_fastpath(
           pod_req, pod_pool, P, K, demand_out, smem,
           (int64_t)(blockIdx.x - 1) * 4, (int64_t)(gridDim.x - 1));
+    }
+  }
+  if constexpr (SINGLE) {
+    // no idle blocks exist: the lone block maps pod -> pool itself
+    // before the tail when the side-stream segsum will consume it
+    if (want_pod_pool && demand_out == nullptr) {
+      for (int64_t i = gtid; i < P; i += gsz) {
+        const int s = slot_of_pod[i];
+        pod_pool[i] = (s >= 0) ? seg_pool[slot2seg[s]] : -1;
+      }
+      __syncthreads();
     }
   }
   if (blockIdx.x == 0) {
@@ -177,9 +210,11 @@ extern "C" __global__ __launch_bounds__(MEGA_BLOCK) void mega_decide_kernel(
 #include <torch/extension.h>
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
-// Returns true when the cooperative launch was enqueued; false routes
-// the caller to the classic multi-launch pipeline.
-bool mega_decide_launch(
+// Returns true when the launch was enqueued; false routes the caller
+// to the classic multi-launch pipeline (or, for the single-block
+// variant, to the cooperative mega attempt).
+static bool mega_decide_launch_impl(
+    bool single,
     torch::Tensor pod_req, torch::Tensor pod_flags, torch::Tensor pod_sel_hash,
     torch::Tensor pod_sel_pairs, torch::Tensor pod_sel_count,
     torch::Tensor pod_node, torch::Tensor node_alloc, torch::Tensor node_labels,
@@ -196,26 +231,29 @@ bool mega_decide_launch(
     int64_t seg_max, int64_t want_pod_pool, uint64_t* demand_out) {
   static int coop_ok = -1;
   static int max_coop_blocks = 0;
-  if (coop_ok < 0) {
-    int v = 0;
-    if (hipDeviceGetAttribute(&v, hipDeviceAttributeCooperativeLaunch, 0) !=
-        hipSuccess) v = 0;
-    if (v) {
-      int per_cu = 0, ncu = 0;
-      if (hipOccupancyMaxActiveBlocksPerMultiprocessor(
-              &per_cu, reinterpret_cast<const void*>(mega_decide_kernel),
-              MEGA_BLOCK, MEGA_SMEM) != hipSuccess ||
-          hipDeviceGetAttribute(&ncu, hipDeviceAttributeMultiprocessorCount,
-                                0) != hipSuccess || per_cu <= 0) {
-        v = 0;
-      } else {
-        max_coop_blocks = per_cu * ncu;
+  if (!single) {
+    if (coop_ok < 0) {
+      int v = 0;
+      if (hipDeviceGetAttribute(&v, hipDeviceAttributeCooperativeLaunch, 0) !=
+          hipSuccess) v = 0;
+      if (v) {
+        int per_cu = 0, ncu = 0;
+        if (hipOccupancyMaxActiveBlocksPerMultiprocessor(
+                &per_cu,
+                reinterpret_cast<const void*>(&mega_decide_kernel_t<false>),
+                MEGA_BLOCK, MEGA_SMEM) != hipSuccess ||
+            hipDeviceGetAttribute(&ncu, hipDeviceAttributeMultiprocessorCount,
+                                  0) != hipSuccess || per_cu <= 0) {
+          v = 0;
+        } else {
+          max_coop_blocks = per_cu * ncu;
+        }
       }
+      (void)hipGetLastError();  // clear any probe error
+      coop_ok = v;
     }
-    (void)hipGetLastError();  // clear any probe error
-    coop_ok = v;
+    if (!coop_ok || max_coop_blocks < 2) return false;
   }
-  if (!coop_ok || max_coop_blocks < 2) return false;
 
   const int64_t P = pod_flags.size(0);
   int32_t N = (int32_t)node_alloc.size(0);
@@ -230,7 +268,7 @@ bool mega_decide_launch(
   const int ntiles = (N + FIT_TILE - 1) / FIT_TILE;
   if (grid < ntiles + 1) grid = ntiles + 1;
   if (grid < 8) grid = 8;
-  if (grid > max_coop_blocks) grid = max_coop_blocks;
+  if (!single && grid > max_coop_blocks) grid = max_coop_blocks;
 
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const int32_t* p_pod_req = pod_req.data_ptr<int32_t>();
@@ -281,8 +319,22 @@ bool mega_decide_launch(
       (void*)&P, (void*)&N, (void*)&K, (void*)&H, (void*)&SM, (void*)&nblob,
       (void*)&RTOT, (void*)&wpp, (void*)&p_demand,
   };
+  if (single) {
+    static int single_ok = -1;
+    if (single_ok == 0) return false;
+    hipError_t e = hipLaunchKernel(
+        reinterpret_cast<const void*>(&mega_decide_kernel_t<true>), dim3(1),
+        dim3(MEGA_BLOCK), args, MEGA_SMEM, stream.stream());
+    if (e != hipSuccess) {
+      (void)hipGetLastError();
+      single_ok = 0;  // don't retry a failing configuration
+      return false;
+    }
+    single_ok = 1;
+    return true;
+  }
   hipError_t e = hipLaunchCooperativeKernel(
-      reinterpret_cast<const void*>(mega_decide_kernel), dim3(grid),
+      reinterpret_cast<const void*>(&mega_decide_kernel_t<false>), dim3(grid),
       dim3(MEGA_BLOCK), args, MEGA_SMEM, stream.stream());
   if (e != hipSuccess) {
     (void)hipGetLastError();
@@ -290,4 +342,41 @@ bool mega_decide_launch(
     return false;
   }
   return true;
+}
+
+#define MEGA_FWD_ARGS                                                        \
+  pod_req, pod_flags, pod_sel_hash, pod_sel_pairs, pod_sel_count, pod_node,  \
+      node_alloc, node_labels, node_label_count, node_unsched, pool_unit,    \
+      pool_labels, pool_label_count, pool_max, pool_cur, ws_i32, slot_ks,    \
+      slot_kr, slot2seg, seg_sel_pairs, seg_sel_count, seg_sel_hash,         \
+      seg_req, seg_count, seg_key, seg_first, slot_of_pod, free, elig,       \
+      seg_pool, run_scratch, blob, pod_pool, hash_slots, seg_max,            \
+      want_pod_pool, demand_out
+
+#define MEGA_FWD_PARAMS                                                       \
+  torch::Tensor pod_req, torch::Tensor pod_flags, torch::Tensor pod_sel_hash, \
+      torch::Tensor pod_sel_pairs, torch::Tensor pod_sel_count,               \
+      torch::Tensor pod_node, torch::Tensor node_alloc,                       \
+      torch::Tensor node_labels, torch::Tensor node_label_count,              \
+      torch::Tensor node_unsched, torch::Tensor pool_unit,                    \
+      torch::Tensor pool_labels, torch::Tensor pool_label_count,              \
+      torch::Tensor pool_max, torch::Tensor pool_cur, torch::Tensor ws_i32,   \
+      torch::Tensor slot_ks, torch::Tensor slot_kr, torch::Tensor slot2seg,   \
+      torch::Tensor seg_sel_pairs, torch::Tensor seg_sel_count,               \
+      torch::Tensor seg_sel_hash, torch::Tensor seg_req,                      \
+      torch::Tensor seg_count, torch::Tensor seg_key, torch::Tensor seg_first,\
+      torch::Tensor slot_of_pod, torch::Tensor free, torch::Tensor elig,      \
+      torch::Tensor seg_pool, torch::Tensor run_scratch, torch::Tensor blob,  \
+      torch::Tensor pod_pool, int64_t hash_slots, int64_t seg_max,            \
+      int64_t want_pod_pool, uint64_t* demand_out
+
+bool mega_decide_launch(MEGA_FWD_PARAMS) {
+  return mega_decide_launch_impl(false, MEGA_FWD_ARGS);
+}
+
+// Single-workgroup variant: regular (non-cooperative) launch, one
+// block runs every phase with __syncthreads barriers. For the small-P
+// regime where grid-wide sync + cooperative-launch setup dominate.
+bool mega_decide_single_launch(MEGA_FWD_PARAMS) {
+  return mega_decide_launch_impl(true, MEGA_FWD_ARGS);
 }
