@@ -143,15 +143,18 @@ void fused_decide_launch(
     mega_max_p = mp ? atoll(mp) : 131072;
     // single-workgroup regime: regular launch, __syncthreads-only
     // barriers. 0 disables. Crossover measured on MI355X (see
-    // profiles/): default covers config #2 (1k pods, 8 nodes).
+    // profiles/r02_single_*): ~P 1000-1500, earlier on node-heavy
+    // shapes — single wins at (1000 pods, 8 nodes) and (500, 100)
+    // but loses at (1000, 100) and (2000, any), so the gate is
+    // (P <= max && N <= 32) || P <= max/2.
     const char* sp = getenv("MI355X_AS_SINGLE_MAX_P");
-    single_max_p = sp ? atoll(sp) : 2048;
+    single_max_p = sp ? atoll(sp) : 1024;
   }
   uint64_t* demand = g_mega_demand;
   g_mega_demand = nullptr;
   g_mega_demand_used = false;
   if (mega_env && single_max_p > 0 && N > 0 && K > 0 && P > 0 &&
-      P <= single_max_p && N <= 2048) {
+      ((P <= single_max_p && N <= 32) || P <= single_max_p / 2) && N <= 2048) {
     if (mega_decide_single_launch(
             pod_req, pod_flags, pod_sel_hash, pod_sel_pairs, pod_sel_count,
             pod_node, node_alloc, node_labels, node_label_count, node_unsched,

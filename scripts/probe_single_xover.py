@@ -16,12 +16,20 @@ from mi355x_autoscaler.snapshot.encode import encode_snapshot
 
 def main():
     P = int(sys.argv[1])
+    shape = sys.argv[2] if len(sys.argv) > 2 else "a"
     eng = GpuEngine("cuda:0")
-    pods_json, nodes_json = burst_trace(
-        P, n_nodes_per_pool=(40, 30, 30),
-        pools=(("agentpool1", "Standard_D2_v2"), ("bigpool", "Standard_D4_v2"),
-               ("gpupool", "Standard_ND96_MI355X_v1")),
-        gpu_fraction=0.35, seed=0, bound_pods_per_node=3)
+    if shape == "a":  # config #3 shape: 100 nodes, 3 pools
+        pods_json, nodes_json = burst_trace(
+            P, n_nodes_per_pool=(40, 30, 30),
+            pools=(("agentpool1", "Standard_D2_v2"),
+                   ("bigpool", "Standard_D4_v2"),
+                   ("gpupool", "Standard_ND96_MI355X_v1")),
+            gpu_fraction=0.35, seed=0, bound_pods_per_node=3)
+    else:  # config #2 shape: 8 nodes, 1 GPU pool
+        pods_json, nodes_json = burst_trace(
+            P, n_nodes_per_pool=(8,),
+            pools=(("gpupool", "Standard_ND96_MI355X_v1"),),
+            gpu_fraction=1.0, seed=0, bound_pods_per_node=3)
     pods, nodes = wrap(pods_json, nodes_json)
     snap = encode_snapshot(pods, nodes, group_nodes_into_pools(nodes))
     for _ in range(10):
@@ -34,7 +42,7 @@ def main():
         ts.append(time.perf_counter() - t0)
     ts.sort()
     mode = os.environ.get("MI355X_AS_SINGLE_MAX_P", "2048")
-    print(f"P={P} single_max_p={mode} p50={ts[30]*1e3:.3f} ms")
+    print(f"P={P} shape={shape} single_max_p={mode} p50={ts[30]*1e3:.3f} ms")
 
 
 main()
