@@ -41,7 +41,7 @@ def main():
         eng.decide(snap)
         ts.append(time.perf_counter() - t0)
     ts.sort()
-    mode = os.environ.get("MI355X_AS_SINGLE_MAX_P", "2048")
+    mode = os.environ.get("MI355X_AS_SINGLE_MAX_P", "default(1024)")
     print(f"P={P} shape={shape} single_max_p={mode} p50={ts[30]*1e3:.3f} ms")
 
 
